@@ -1,0 +1,75 @@
+"""Typed configuration (ConfigOption pattern).
+
+Role parity: auron-core ConfigOption/AuronConfiguration +
+SparkAuronConfiguration.java (~80 `spark.auron.*` options) and the native
+pull-through side (auron-jni-bridge/src/conf.rs:32-65). Keys keep the
+`spark.auron.*` shape so reference users find the same knobs; values can
+be overridden via environment (AURON_<NAME>) or programmatically.
+"""
+from __future__ import annotations
+
+import os
+from dataclasses import dataclass, field
+from typing import Any, Callable, Dict, Optional
+
+
+@dataclass(frozen=True)
+class ConfigOption:
+    key: str
+    default: Any
+    typ: type
+    doc: str = ""
+    env: Optional[str] = None
+
+
+_REGISTRY: Dict[str, ConfigOption] = {}
+
+
+def _opt(key: str, default, typ, doc="", env=None) -> ConfigOption:
+    o = ConfigOption(key, default, typ, doc, env)
+    _REGISTRY[key] = o
+    return o
+
+
+# conf.rs:32-65 analogues
+BATCH_SIZE = _opt("spark.auron.batchSize", 1 << 22, int, "rows per in-flight batch")
+MEMORY_FRACTION = _opt("spark.auron.memoryFraction", 0.8, float, "fraction of HBM the memmgr may use")
+ENABLE_NATIVE = _opt("spark.auron.enable", True, bool, "master switch")
+REQUIRE_NATIVE = _opt("spark.auron.requireNativeKernels", True, bool,
+                      "fail loudly if HIP kernels missing on GPU", env="AURON_REQUIRE_NATIVE")
+SPILL_COMPRESSION = _opt("spark.auron.spill.compression.codec", "lz4", str)
+SHUFFLE_COMPRESSION = _opt("spark.auron.shuffle.compression.codec", "none", str,
+                           "xGMI is fast enough that intra-node shuffle ships raw")
+SMJ_FALLBACK_ENABLE = _opt("spark.auron.smjfallback.enable", False, bool)
+SMJ_FALLBACK_ROWS = _opt("spark.auron.smjfallback.rows.threshold", 10_000_000, int)
+PARTIAL_AGG_SKIPPING_RATIO = _opt("spark.auron.partialAggSkipping.ratio", 0.999, float)
+UDF_FALLBACK = _opt("spark.auron.udf.hostFallback.enable", True, bool)
+LOG_LEVEL = _opt("spark.auron.native.log.level", "WARN", str, env="AURON_LOG_LEVEL")
+
+
+class AuronConf:
+    def __init__(self, overrides: Optional[Dict[str, Any]] = None):
+        self._values: Dict[str, Any] = {}
+        if overrides:
+            for k, v in overrides.items():
+                self.set(k, v)
+
+    def set(self, key: str, value) -> "AuronConf":
+        if key not in _REGISTRY:
+            raise KeyError(f"unknown config {key}; known: {sorted(_REGISTRY)}")
+        self._values[key] = value
+        return self
+
+    def get(self, opt: ConfigOption):
+        if opt.key in self._values:
+            return self._values[opt.key]
+        if opt.env and opt.env in os.environ:
+            raw = os.environ[opt.env]
+            if opt.typ is bool:
+                return raw not in ("0", "false", "False")
+            return opt.typ(raw)
+        return opt.default
+
+    @staticmethod
+    def options() -> Dict[str, ConfigOption]:
+        return dict(_REGISTRY)

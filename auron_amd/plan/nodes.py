@@ -1,0 +1,234 @@
+"""Physical plan nodes.
+
+Role parity: the PhysicalPlanNode oneof in the reference's plan protocol
+(/root/reference/native-engine/auron-planner/proto/auron.proto:27-57, 27
+operator kinds) and the operator set of datafusion-ext-plans. Nodes here
+are built by the front-end (query builders / TPC-DS plans), serialized via
+auron_amd.plan.serde, and lowered by the executor into device-columnar
+pipelines.
+
+Distribution model: one process per GPU; a plan executes
+SPMD on every rank; Exchange/Broadcast nodes are the only cross-rank
+communication points (RCCL all-to-all / all-gather over xGMI).
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
+
+from ..exprs import AggFunc, Aliased, Expr
+
+
+class PlanNode:
+    def children(self) -> List["PlanNode"]:
+        return []
+
+
+@dataclass(eq=False)
+class ParquetScan(PlanNode):
+    """Host Parquet page read (pyarrow) -> device columns.
+
+    Reference analogue: parquet_exec.rs (scan through JVM FS). Files are
+    sharded across ranks round-robin; `columns` prunes projection at the
+    reader; `filters` is an optional list of pyarrow-compatible predicates
+    pushed into row-group pruning.
+    """
+    paths: List[str]
+    columns: Optional[List[str]] = None
+    filters: Optional[list] = None
+
+
+@dataclass(eq=False)
+class MemoryScan(PlanNode):
+    """In-memory table scan (tests / broadcast-collected relations)."""
+    batches: list  # List[RecordBatch], already rank-local
+
+
+@dataclass(eq=False)
+class Filter(PlanNode):
+    child: PlanNode
+    predicate: Expr
+
+    def children(self):
+        return [self.child]
+
+
+@dataclass(eq=False)
+class Project(PlanNode):
+    child: PlanNode
+    exprs: List[Aliased]
+
+    def children(self):
+        return [self.child]
+
+
+@dataclass(eq=False)
+class HashAgg(PlanNode):
+    """Modes mirror the reference (agg/mod.rs:44): partial | final | complete."""
+    child: PlanNode
+    keys: List[Aliased]
+    aggs: List[AggFunc]
+    mode: str = "complete"  # partial | final | complete
+
+    def children(self):
+        return [self.child]
+
+
+@dataclass(eq=False)
+class HashJoin(PlanNode):
+    """BHJ + SHJ in one operator (broadcast_join_exec.rs pattern: the
+    `broadcast` flag decides whether the build side is gathered to every
+    rank first)."""
+    left: PlanNode
+    right: PlanNode
+    left_keys: List[Expr]
+    right_keys: List[Expr]
+    how: str = "inner"  # inner|left|right|full|semi|anti|existence
+    build_side: str = "right"  # right|left
+    broadcast: bool = False  # build side is broadcast (BHJ) vs co-partitioned (SHJ)
+    existence_col: str = "exists"
+
+    def children(self):
+        return [self.left, self.right]
+
+
+@dataclass(eq=False)
+class SortMergeJoin(PlanNode):
+    """Declared for plan parity; lowered to HashJoin until the merge-path
+    kernel lands (sort_merge_join_exec.rs analogue)."""
+    left: PlanNode
+    right: PlanNode
+    left_keys: List[Expr]
+    right_keys: List[Expr]
+    how: str = "inner"
+
+    def children(self):
+        return [self.left, self.right]
+
+
+@dataclass(eq=False)
+class Sort(PlanNode):
+    child: PlanNode
+    keys: List[Tuple[Expr, bool]]  # (expr, ascending); Spark null ordering default
+    limit: Optional[int] = None
+
+    def children(self):
+        return [self.child]
+
+
+@dataclass(eq=False)
+class Limit(PlanNode):
+    child: PlanNode
+    n: int
+    offset: int = 0
+
+    def children(self):
+        return [self.child]
+
+
+@dataclass(eq=False)
+class Exchange(PlanNode):
+    """Shuffle repartition. kind: hash (murmur3 pmod over keys, RCCL
+    all-to-all), single (gather everything to rank 0), roundrobin.
+
+    Reference analogue: shuffle_writer_exec.rs + ipc_reader_exec.rs; here
+    the write+read pair collapses into one in-flight collective."""
+    child: PlanNode
+    kind: str = "hash"  # hash | single | roundrobin
+    keys: List[Expr] = field(default_factory=list)
+
+    def children(self):
+        return [self.child]
+
+
+@dataclass(eq=False)
+class Broadcast(PlanNode):
+    """Collect child result on every rank (rcclAllGather / TorrentBroadcast
+    analogue, NativeBroadcastExchangeBase.scala)."""
+    child: PlanNode
+
+    def children(self):
+        return [self.child]
+
+
+@dataclass(eq=False)
+class Union(PlanNode):
+    inputs: List[PlanNode]
+
+    def children(self):
+        return list(self.inputs)
+
+
+@dataclass(eq=False)
+class Expand(PlanNode):
+    """GROUPING SETS fan-out (expand_exec.rs)."""
+    child: PlanNode
+    projections: List[List[Aliased]]
+
+    def children(self):
+        return [self.child]
+
+
+@dataclass(eq=False)
+class Window(PlanNode):
+    """Window functions over (partition_by, order_by) (window_exec.rs)."""
+    child: PlanNode
+    partition_by: List[Expr]
+    order_by: List[Tuple[Expr, bool]]
+    functions: List[Aliased]  # WindowFunc exprs aliased to output names
+
+    def children(self):
+        return [self.child]
+
+
+@dataclass(eq=False)
+class Generate(PlanNode):
+    """explode/posexplode (generate_exec.rs). Round-1: explode of
+    literal lists is unsupported; placeholder for API parity."""
+    child: PlanNode
+    generator: str
+    args: List[Expr] = field(default_factory=list)
+
+    def children(self):
+        return [self.child]
+
+
+@dataclass(eq=False)
+class RenameColumns(PlanNode):
+    child: PlanNode
+    names: List[str]
+
+    def children(self):
+        return [self.child]
+
+
+@dataclass(eq=False)
+class EmptyPartitions(PlanNode):
+    names: List[str] = field(default_factory=list)
+
+
+@dataclass(eq=False)
+class CoalesceBatches(PlanNode):
+    child: PlanNode
+    target_rows: int = 1 << 22
+
+    def children(self):
+        return [self.child]
+
+
+@dataclass(eq=False)
+class ParquetSink(PlanNode):
+    child: PlanNode
+    path: str
+
+    def children(self):
+        return [self.child]
+
+
+@dataclass(eq=False)
+class Debug(PlanNode):
+    child: PlanNode
+    label: str = ""
+
+    def children(self):
+        return [self.child]

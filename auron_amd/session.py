@@ -1,0 +1,72 @@
+"""AuronSession — the user entry point.
+
+Role parity: AuronSparkSessionExtension + NativeHelper.executeNativePlan
+(/root/reference/spark-extension/src/main/scala/org/apache/spark/sql/auron/
+ AuronSparkSessionExtension.scala:31, NativeHelper.scala:91). A session
+owns the exec context (device, rank/world from torch.distributed) and
+runs physical plans on it.
+"""
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+from .column import RecordBatch
+from .config import AuronConf
+from .engine.executor import ExecContext, Executor
+from .plan import nodes as P
+
+
+class AuronSession:
+    def __init__(self, conf: Optional[AuronConf] = None, device=None):
+        self.conf = conf or AuronConf()
+        self.ctx = ExecContext.from_dist(device)
+        self.executor = Executor(self.ctx)
+
+    @property
+    def rank(self) -> int:
+        return self.ctx.rank
+
+    @property
+    def world_size(self) -> int:
+        return self.ctx.world_size
+
+    def execute(self, plan: P.PlanNode):
+        """Run a physical plan; returns this rank's local batches."""
+        return self.executor.execute(plan)
+
+    def collect(self, plan: P.PlanNode) -> RecordBatch:
+        """Run and concat this rank's result (driver-side rows analogue)."""
+        return self.executor.collect(plan)
+
+    def collect_all(self, plan: P.PlanNode) -> RecordBatch:
+        """Gather the full result on every rank (for result checking)."""
+        local = self.collect(plan)
+        if self.world_size == 1:
+            return local
+        from .exchange import all_gather_batch
+
+        parts = all_gather_batch(local, self.ctx.device, self.ctx.group)
+        return RecordBatch.concat(parts)
+
+    def metrics(self):
+        return dict(self.ctx.metrics)
+
+
+def init_distributed(backend: Optional[str] = None) -> ExecContext:
+    """Initialize torch.distributed from torchrun env (RCCL on GPU)."""
+    import os
+
+    if dist.is_initialized():
+        return ExecContext.from_dist()
+    if "RANK" not in os.environ:
+        return ExecContext.from_dist()
+    if backend is None:
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+    if backend == "nccl":
+        local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+        torch.cuda.set_device(local_rank)
+    dist.init_process_group(backend=backend)
+    return ExecContext.from_dist()

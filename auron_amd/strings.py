@@ -1,0 +1,241 @@
+"""Vectorized string kernels over Arrow-layout (offsets+bytes) columns.
+
+Role parity: spark_strings.rs / StringStartsWith/EndsWith/Contains exprs in
+/root/reference/native-engine/datafusion-ext-functions and -ext-exprs.
+
+All ops are expressed as tensor ops over the byte/offset buffers so the
+same code runs on CPU (CI) and on device (HIP via torch). Byte-wise
+comparison == UTF-8 codepoint order, matching Spark's binary collation.
+Dedicated HIP kernels replace the padded-matrix path for hot ops via
+auron_amd.ops when profitable.
+"""
+from __future__ import annotations
+
+from typing import List, Optional
+
+import torch
+
+from . import dtypes
+from .column import Column
+
+_MAX_PAD = 256
+
+
+def lengths(c: Column) -> torch.Tensor:
+    off = c.offsets.to(torch.int64)
+    return off[1:] - off[:-1]
+
+
+def to_padded(c: Column, width: Optional[int] = None) -> torch.Tensor:
+    """[n, W] uint8 matrix, zero-padded (0 sorts below every valid byte)."""
+    n = len(c)
+    lens = lengths(c)
+    maxlen = int(lens.max().item()) if n and lens.numel() else 0
+    W = width if width is not None else min(max(maxlen, 1), _MAX_PAD)
+    if n == 0:
+        return torch.zeros((0, W), dtype=torch.uint8, device=c.device)
+    off = c.offsets.to(torch.int64)
+    pos = off[:-1].unsqueeze(1) + torch.arange(W, dtype=torch.int64, device=c.device).unsqueeze(0)
+    valid = pos < off[1:].unsqueeze(1)
+    pos = torch.where(valid, pos, torch.zeros_like(pos))
+    if c.data.numel() == 0:
+        return torch.zeros((n, W), dtype=torch.uint8, device=c.device)
+    flat = c.data[pos.reshape(-1)].reshape(n, W)
+    return torch.where(valid, flat, torch.zeros_like(flat))
+
+
+def _pad_pair(l: Column, r: Column):
+    W = int(max(lengths(l).max().item() if len(l) else 0, lengths(r).max().item() if len(r) else 0, 1))
+    W = min(W, _MAX_PAD)
+    return to_padded(l, W), to_padded(r, W)
+
+
+def compare(l: Column, r: Column, op: str) -> torch.Tensor:
+    A, B = _pad_pair(l, r)
+    if op == "==":
+        return (A == B).all(dim=1) & (lengths(l) == lengths(r))
+    if op == "!=":
+        return ~compare(l, r, "==")
+    # lexicographic
+    lt = torch.zeros(A.shape[0], dtype=torch.bool, device=A.device)
+    gt = torch.zeros_like(lt)
+    for j in range(A.shape[1]):
+        a = A[:, j]
+        b = B[:, j]
+        und = ~lt & ~gt
+        lt = lt | (und & (a < b))
+        gt = gt | (und & (a > b))
+    if op == "<":
+        return lt
+    if op == ">":
+        return gt
+    if op == "<=":
+        return ~gt
+    if op == ">=":
+        return ~lt
+    raise ValueError(op)
+
+
+def _pat_tensor(pattern: str, device) -> torch.Tensor:
+    b = pattern.encode("utf-8")
+    return torch.tensor(list(b), dtype=torch.uint8, device=device)
+
+
+def eq_literal(c: Column, s: str) -> torch.Tensor:
+    pat = _pat_tensor(s, c.device)
+    m = len(pat)
+    lens = lengths(c)
+    ok = lens == m
+    if m == 0:
+        return ok
+    A = to_padded(c, max(m, 1))
+    return ok & (A[:, :m] == pat).all(dim=1)
+
+
+def startswith(c: Column, s: str) -> torch.Tensor:
+    pat = _pat_tensor(s, c.device)
+    m = len(pat)
+    if m == 0:
+        return torch.ones(len(c), dtype=torch.bool, device=c.device)
+    lens = lengths(c)
+    A = to_padded(c, max(m, 1))
+    return (lens >= m) & (A[:, :m] == pat).all(dim=1)
+
+
+def endswith(c: Column, s: str) -> torch.Tensor:
+    pat = _pat_tensor(s, c.device)
+    m = len(pat)
+    if m == 0:
+        return torch.ones(len(c), dtype=torch.bool, device=c.device)
+    lens = lengths(c)
+    off = c.offsets.to(torch.int64)
+    n = len(c)
+    starts = off[1:] - m
+    ok = lens >= m
+    starts = starts.clamp(min=0)
+    pos = starts.unsqueeze(1) + torch.arange(m, device=c.device).unsqueeze(0)
+    pos = pos.clamp(max=max(int(c.data.numel()) - 1, 0))
+    if c.data.numel() == 0:
+        return torch.zeros(n, dtype=torch.bool, device=c.device)
+    tail = c.data[pos.reshape(-1)].reshape(n, m)
+    return ok & (tail == pat).all(dim=1)
+
+
+def contains(c: Column, s: str) -> torch.Tensor:
+    pat = _pat_tensor(s, c.device)
+    m = len(pat)
+    if m == 0:
+        return torch.ones(len(c), dtype=torch.bool, device=c.device)
+    lens = lengths(c)
+    maxlen = int(lens.max().item()) if len(c) else 0
+    if maxlen < m:
+        return torch.zeros(len(c), dtype=torch.bool, device=c.device)
+    A = to_padded(c, maxlen)
+    win = A.unfold(1, m, 1)  # [n, maxlen-m+1, m]
+    hit = (win == pat).all(dim=2)  # [n, nwin]
+    starts = torch.arange(hit.shape[1], device=c.device).unsqueeze(0)
+    valid_win = starts + m <= lens.unsqueeze(1)
+    return (hit & valid_win).any(dim=1)
+
+
+def like(c: Column, pattern: str) -> torch.Tensor:
+    parts = pattern.split("%")
+    if len(parts) == 1:
+        return eq_literal(c, pattern)
+    if "_" in pattern:
+        return _like_host(c, pattern)
+    head, tail = parts[0], parts[-1]
+    mids = [p for p in parts[1:-1] if p]
+    ok = torch.ones(len(c), dtype=torch.bool, device=c.device)
+    if head:
+        ok = ok & startswith(c, head)
+    if tail:
+        ok = ok & endswith(c, tail)
+    min_len = len(head) + len(tail) + sum(len(m) for m in mids)
+    ok = ok & (lengths(c) >= min_len)
+    # middle fragments: each must appear (ordered check approximated by
+    # containment; exact ordered multi-fragment LIKE falls to host)
+    if len(mids) == 1 and not head and not tail:
+        return contains(c, mids[0])
+    if mids:
+        return _like_host(c, pattern)
+    return ok
+
+
+def _like_host(c: Column, pattern: str) -> torch.Tensor:
+    import re
+
+    rx = re.compile("^" + re.escape(pattern).replace("%", ".*").replace("_", ".") + "$", re.S)
+    vals = c.to_pylist()
+    out = [bool(v is not None and rx.match(v)) for v in vals]
+    return torch.tensor(out, dtype=torch.bool, device=c.device)
+
+
+def isin(c: Column, values: List[str]) -> torch.Tensor:
+    out = torch.zeros(len(c), dtype=torch.bool, device=c.device)
+    for v in values:
+        out = out | eq_literal(c, v)
+    return out
+
+
+def substr(c: Column, start: int, length: int) -> Column:
+    """SQL substring: 1-based start, byte-based (ASCII-safe)."""
+    off = c.offsets.to(torch.int64)
+    lens = off[1:] - off[:-1]
+    s0 = max(start - 1, 0)
+    new_lens = (lens - s0).clamp(min=0).clamp(max=length)
+    new_off = torch.zeros(len(c) + 1, dtype=torch.int64, device=c.device)
+    torch.cumsum(new_lens, 0, out=new_off[1:])
+    total = int(new_off[-1].item())
+    if total == 0:
+        data = torch.empty(0, dtype=torch.uint8, device=c.device)
+    else:
+        row = torch.repeat_interleave(new_lens)
+        pos = torch.arange(total, dtype=torch.int64, device=c.device)
+        byte_idx = pos - new_off[:-1][row] + off[:-1][row] + s0
+        data = c.data[byte_idx]
+    return Column(dtypes.string, data, c.validity, new_off.to(torch.int32))
+
+
+def concat(cols: List[Column]) -> Column:
+    n = len(cols[0])
+    device = cols[0].device
+    lens = [lengths(c) for c in cols]
+    tot_lens = torch.zeros(n, dtype=torch.int64, device=device)
+    for L in lens:
+        tot_lens = tot_lens + L
+    new_off = torch.zeros(n + 1, dtype=torch.int64, device=device)
+    torch.cumsum(tot_lens, 0, out=new_off[1:])
+    total = int(new_off[-1].item())
+    data = torch.zeros(total, dtype=torch.uint8, device=device)
+    cursor = new_off[:-1].clone()
+    for c, L in zip(cols, lens):
+        tot = int(L.sum().item())
+        if tot == 0:
+            continue
+        off = c.offsets.to(torch.int64)
+        row = torch.repeat_interleave(L)
+        pos = torch.arange(tot, dtype=torch.int64, device=device)
+        excl = torch.cat([torch.zeros(1, dtype=torch.int64, device=device), torch.cumsum(L, 0)[:-1]])
+        within = pos - excl[row]
+        src = off[:-1][row] + within
+        dst = cursor[row] + within
+        data[dst] = c.data[src]
+        cursor = cursor + L
+    validity = None
+    for c in cols:
+        if c.validity is not None:
+            validity = c.validity if validity is None else (validity & c.validity)
+    return Column(dtypes.string, data, validity, new_off.to(torch.int32))
+
+
+def upper(c: Column) -> Column:
+    is_lower = (c.data >= 97) & (c.data <= 122)
+    data = torch.where(is_lower, c.data - 32, c.data)
+    return Column(dtypes.string, data, c.validity, c.offsets)
+
+
+def lower(c: Column) -> Column:
+    is_upper = (c.data >= 65) & (c.data <= 90)
+    data = torch.where(is_upper, c.data + 32, c.data)
+    return Column(dtypes.string, data, c.validity, c.offsets)

@@ -1,0 +1,278 @@
+"""Operator correctness vs pandas oracle (the QueryResultComparator
+pattern from the reference's integration harness: vanilla-engine results
+are ground truth, compared row-wise with float tolerance)."""
+import math
+
+import pandas as pd
+import pytest
+import torch
+
+from auron_amd import AggFunc, AuronSession, Col, col, dtypes, exprs, lit
+from auron_amd.column import RecordBatch
+from auron_amd.plan import nodes as P
+
+T = {"k": dtypes.string, "g": dtypes.int32, "x": dtypes.int64, "f": dtypes.float64,
+     "d": dtypes.date32}
+DATA = {
+    "k": ["a", "b", "a", None, "c", "b", "a"],
+    "g": [1, 2, 1, 2, 3, None, 1],
+    "x": [10, 20, 30, 40, None, 60, 70],
+    "f": [1.5, 2.5, None, 4.0, 5.5, 6.0, 7.25],
+    "d": [19000, 19001, 19002, 19003, 19004, 19005, 19006],
+}
+
+
+def scan(device="cpu"):
+    return P.MemoryScan([RecordBatch.from_pydict(DATA, T, device)])
+
+
+def df():
+    return pd.DataFrame(DATA)
+
+
+def sorted_rows(d: dict):
+    rows = list(zip(*d.values()))
+    return sorted(rows, key=lambda r: tuple((v is None, str(v)) for v in r))
+
+
+def assert_rows_equal(got: dict, want: dict):
+    gr, wr = sorted_rows(got), sorted_rows(want)
+    assert len(gr) == len(wr), f"{len(gr)} rows != {len(wr)}\n{gr}\n{wr}"
+    for a, b in zip(gr, wr):
+        for x, y in zip(a, b):
+            if isinstance(x, float) and isinstance(y, float):
+                assert math.isclose(x, y, rel_tol=1e-9, abs_tol=1e-9), (a, b)
+            else:
+                assert x == y, (a, b)
+
+
+def test_filter_project():
+    s = AuronSession()
+    plan = P.Project(
+        P.Filter(scan(), (col("x") > 15) & col("k").is_not_null()),
+        [exprs.Aliased(col("k"), "k"), exprs.Aliased(col("x") * 2, "x2")],
+    )
+    out = s.collect(plan).to_pydict()
+    want = df()
+    want = want[(want.x > 15) & want.k.notna()]
+    assert_rows_equal(out, {"k": want.k.tolist(), "x2": (want.x * 2).tolist()})
+
+
+def test_filter_null_semantics():
+    s = AuronSession()
+    # x > 15 is NULL for the null row -> dropped
+    plan = P.Filter(scan(), col("x") > 15)
+    out = s.collect(plan)
+    # x = [10,20,30,40,None,60,70]: null comparison row dropped -> 5 rows
+    assert out.num_rows == 5
+    assert sorted(out.to_pydict()["x"]) == [20, 30, 40, 60, 70]
+
+
+def _oracle_group_agg():
+    """dict-based oracle (pandas groupby-with-None quirks avoided)."""
+    from collections import defaultdict
+
+    rows = list(zip(*DATA.values()))
+    groups = defaultdict(list)
+    for r in rows:
+        groups[r[0]].append(dict(zip(DATA.keys(), r)))
+    want = {"k": [], "sx": [], "cx": [], "af": [], "mn": [], "mx": []}
+    for k, rs in groups.items():
+        xs = [r["x"] for r in rs if r["x"] is not None]
+        fs = [r["f"] for r in rs if r["f"] is not None]
+        want["k"].append(k)
+        want["sx"].append(sum(xs) if xs else None)
+        want["cx"].append(len(xs))
+        want["af"].append(sum(fs) / len(fs) if fs else None)
+        want["mn"].append(min(xs) if xs else None)
+        want["mx"].append(max(xs) if xs else None)
+    return want
+
+
+def test_agg_complete_group():
+    s = AuronSession()
+    plan = P.HashAgg(
+        scan(), [exprs.Aliased(col("k"), "k")],
+        [AggFunc("sum", col("x"), name="sx"), AggFunc("count", col("x"), name="cx"),
+         AggFunc("avg", col("f"), name="af"), AggFunc("min", col("x"), name="mn"),
+         AggFunc("max", col("x"), name="mx")],
+        mode="complete",
+    )
+    out = s.collect(plan).to_pydict()
+    want = _oracle_group_agg()
+    assert_rows_equal(out, want)
+
+
+def test_agg_global_no_keys():
+    s = AuronSession()
+    plan = P.HashAgg(scan(), [], [AggFunc("count_star", None, name="n"),
+                                  AggFunc("sum", col("f"), name="sf")], mode="complete")
+    out = s.collect(plan).to_pydict()
+    assert out["n"] == [7]
+    assert abs(out["sf"][0] - df().f.sum()) < 1e-9
+
+
+def test_agg_partial_final_roundtrip():
+    s = AuronSession()
+    partial = P.HashAgg(scan(), [exprs.Aliased(col("k"), "k")],
+                        [AggFunc("sum", col("x"), name="sx"),
+                         AggFunc("avg", col("f"), name="af")], mode="partial")
+    final = P.HashAgg(partial, [exprs.Aliased(col("k"), "k")],
+                      [AggFunc("sum", col("x"), name="sx"),
+                       AggFunc("avg", col("f"), name="af")], mode="final")
+    out = s.collect(final).to_pydict()
+    w = _oracle_group_agg()
+    want = {"k": w["k"], "sx": w["sx"], "af": w["af"]}
+    assert_rows_equal(out, want)
+
+
+def test_count_distinct():
+    s = AuronSession()
+    plan = P.HashAgg(scan(), [exprs.Aliased(col("k"), "k")],
+                     [AggFunc("count_distinct", col("g"), name="dg")], mode="complete")
+    out = s.collect(plan).to_pydict()
+    from collections import defaultdict
+    gs = defaultdict(set)
+    for k, g in zip(DATA["k"], DATA["g"]):
+        if g is not None:
+            gs[k].add(g)
+    got = dict(zip(out["k"], out["dg"]))
+    for k in set(DATA["k"]):
+        assert got[k] == len(gs[k]), (k, got[k], gs[k])
+
+
+JL = {"id": [1, 2, 3, 4, None], "lv": ["a", "b", "c", "d", "e"]}
+JR = {"id": [2, 3, 3, None, 6], "rv": ["x", "y", "z", "w", "v"]}
+JT = {"id": dtypes.int64, "lv": dtypes.string, "rv": dtypes.string}
+
+
+def _join_plan(how, build_side="right", broadcast=False):
+    l = P.MemoryScan([RecordBatch.from_pydict(JL, {"id": dtypes.int64, "lv": dtypes.string})])
+    r = P.MemoryScan([RecordBatch.from_pydict(
+        {"rid": JR["id"], "rv": JR["rv"]}, {"rid": dtypes.int64, "rv": dtypes.string})])
+    return P.HashJoin(l, r, [col("id")], [col("rid")], how=how,
+                      build_side=build_side, broadcast=broadcast)
+
+
+def _sql_join_oracle(how):
+    """SQL semantics: NULL keys never match (pandas merge wrongly matches
+    NaN==NaN, so the oracle is explicit)."""
+    lrows = list(zip(JL["id"], JL["lv"]))
+    rrows = list(zip(JR["id"], JR["rv"]))
+    out = {"id": [], "lv": [], "rid": [], "rv": []}
+    rmatched = [False] * len(rrows)
+    for lid, lv in lrows:
+        hit = False
+        for j, (rid, rv) in enumerate(rrows):
+            if lid is not None and rid is not None and lid == rid:
+                out["id"].append(lid); out["lv"].append(lv)
+                out["rid"].append(rid); out["rv"].append(rv)
+                hit = True
+                rmatched[j] = True
+        if not hit and how in ("left", "full"):
+            out["id"].append(lid); out["lv"].append(lv)
+            out["rid"].append(None); out["rv"].append(None)
+    if how in ("right", "full"):
+        for j, (rid, rv) in enumerate(rrows):
+            if not rmatched[j]:
+                out["id"].append(None); out["lv"].append(None)
+                out["rid"].append(rid); out["rv"].append(rv)
+    return out
+
+
+@pytest.mark.parametrize("how", ["inner", "left", "right", "full"])
+@pytest.mark.parametrize("build_side", ["right", "left"])
+def test_hash_join(how, build_side):
+    s = AuronSession()
+    out = s.collect(_join_plan(how, build_side)).to_pydict()
+    assert_rows_equal(out, _sql_join_oracle(how))
+
+
+def test_semi_anti_join():
+    s = AuronSession()
+    semi = s.collect(_join_plan("semi")).to_pydict()
+    assert sorted(semi["lv"]) == ["b", "c"]
+    anti = s.collect(_join_plan("anti")).to_pydict()
+    assert sorted(anti["lv"]) == ["a", "d", "e"]  # null-key left row kept by anti
+
+
+def test_existence_join():
+    s = AuronSession()
+    out = s.collect(_join_plan("existence")).to_pydict()
+    m = dict(zip(out["lv"], out["exists"]))
+    assert m == {"a": False, "b": True, "c": True, "d": False, "e": False}
+
+
+def test_sort_limit():
+    s = AuronSession()
+    plan = P.Sort(scan(), [(col("x"), False)], limit=3)
+    out = s.collect(plan).to_pydict()
+    assert out["x"] == [70, 60, 40]
+
+
+def test_sort_nulls_first_asc():
+    s = AuronSession()
+    plan = P.Sort(scan(), [(col("x"), True)])
+    out = s.collect(plan).to_pydict()
+    assert out["x"] == [None, 10, 20, 30, 40, 60, 70]
+
+
+def test_union_expand_limit():
+    s = AuronSession()
+    u = P.Union([scan(), scan()])
+    out = s.collect(u)
+    assert out.num_rows == 14
+    e = P.Expand(scan(), [
+        [exprs.Aliased(col("x"), "v"), exprs.Aliased(lit(0), "tag")],
+        [exprs.Aliased(col("g").cast(dtypes.int64), "v"), exprs.Aliased(lit(1), "tag")],
+    ])
+    out = s.collect(e)
+    assert out.num_rows == 14
+    lim = P.Limit(scan(), 3, offset=2)
+    assert s.collect(lim).num_rows == 3
+
+
+def test_case_when_and_dates():
+    s = AuronSession()
+    plan = P.Project(scan(), [
+        exprs.Aliased(exprs.CaseWhen([(col("x") > 30, lit(1))], lit(0)), "c"),
+        exprs.Aliased(exprs.year(col("d")), "y"),
+        exprs.Aliased(exprs.month(col("d")), "m"),
+    ])
+    out = s.collect(plan).to_pydict()
+    d = pd.to_datetime(pd.Series(DATA["d"]), unit="D", origin="unix")
+    assert out["y"] == d.dt.year.tolist()
+    assert out["m"] == d.dt.month.tolist()
+
+
+def test_window_row_number_rank():
+    s = AuronSession()
+    plan = P.Window(
+        scan(), [col("k")], [(col("x"), True)],
+        [exprs.Aliased(exprs.WindowFunc("row_number"), "rn"),
+         exprs.Aliased(exprs.WindowFunc("sum", col("x")), "sx")],
+    )
+    out = s.collect(plan).to_pydict()
+    d = df()
+    d["rn"] = d.sort_values("x").groupby("k", dropna=False).cumcount() + 1
+    got = {(k, x): rn for k, x, rn in zip(out["k"], out["x"], out["rn"])}
+    for _, row in d.iterrows():
+        k = None if isinstance(row.k, float) and math.isnan(row.k) else row.k
+        x = None if math.isnan(row.x) else int(row.x)
+        # nulls-first ordering differs from pandas NaN-last; only check non-null x
+        if x is not None and k is not None:
+            pass  # rank check below via direct reconstruction
+    # direct check: within each partition rn is 1..n and ordered by x (nulls first)
+    from collections import defaultdict
+
+    parts = defaultdict(list)
+    for k, x, rn in zip(out["k"], out["x"], out["rn"]):
+        parts[k].append((rn, x))
+    for k, rows in parts.items():
+        rows.sort()
+        assert [r for r, _ in rows] == list(range(1, len(rows) + 1))
+        xs = [x for _, x in rows]
+        non_null = [x for x in xs if x is not None]
+        assert non_null == sorted(non_null)
+        if None in xs:
+            assert xs[0] is None  # asc -> nulls first
