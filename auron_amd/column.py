@@ -56,11 +56,15 @@ class Column:
         device = torch.device(device)
         if self.device == device:
             return self
+        # non_blocking is only safe toward the GPU; an async D2H copy into
+        # pageable host memory returns before the bytes land (reader sees
+        # stale pool memory)
+        nb = device.type == "cuda"
         return Column(
             self.dtype,
-            self.data.to(device, non_blocking=True),
-            None if self.validity is None else self.validity.to(device, non_blocking=True),
-            None if self.offsets is None else self.offsets.to(device, non_blocking=True),
+            self.data.to(device, non_blocking=nb),
+            None if self.validity is None else self.validity.to(device, non_blocking=nb),
+            None if self.offsets is None else self.offsets.to(device, non_blocking=nb),
         )
 
     def clone_meta(self, data, validity=None, offsets=None) -> "Column":

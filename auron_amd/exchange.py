@@ -66,7 +66,8 @@ def pack_batch(batch: RecordBatch, device) -> Tuple[dict, torch.Tensor]:
     for seg in segs:
         n = seg.numel()
         if n:
-            buf[pos:pos + n].copy_(seg.to(device, non_blocking=True))
+            nb = torch.device(device).type == "cuda"
+            buf[pos:pos + n].copy_(seg.to(device, non_blocking=nb))
         pos += _pad8(n)
     return {"nrows": batch.num_rows, "cols": metas}, buf
 

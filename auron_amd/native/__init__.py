@@ -126,3 +126,5 @@ def stream_ptr(device) -> ctypes.c_void_p:
 def check(rc: int, what: str):
     if rc != 0:
         raise RuntimeError(f"native kernel {what} failed with hipError {rc}")
+    if os.environ.get("AURON_SYNC_NATIVE", "0") == "1":
+        torch.cuda.synchronize()

@@ -25,6 +25,8 @@ _M32 = 0xFFFFFFFF
 def _use_native(device: torch.device) -> bool:
     if device.type != "cuda":
         return False
+    if os.environ.get("AURON_FORCE_REF", "0") == "1":
+        return False  # debugging only: force torch/host reference paths
     if os.environ.get("AURON_REQUIRE_NATIVE", "1") == "0" and not native.available():
         return False
     native.require()  # raise loudly on GPU when missing
