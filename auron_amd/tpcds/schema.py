@@ -157,12 +157,16 @@ FACT_TABLES = {"store_sales", "store_returns", "catalog_sales", "catalog_returns
 SCALED_DIMS = {"item": 2.0, "customer": 3.0, "customer_address": 3.0}
 
 
+_DIM_MIN = {"item": 900, "customer": 5000, "customer_address": 2500,
+            "customer_demographics": 48020}
+
+
 def row_count(table: str, sf: float) -> int:
     base = BASE_ROWS[table]
     if table in FACT_TABLES:
         return max(int(base * sf), 1000)
+    if sf < 1 and table in _DIM_MIN:
+        return max(int(base * sf), _DIM_MIN[table])
     if table in SCALED_DIMS and sf > 1:
-        import math
-
         return int(base * min(sf ** 0.5 * SCALED_DIMS[table] / 2, sf))
     return base

@@ -1,0 +1,68 @@
+"""TPC-DS query correctness vs pandas oracle at small SF (oracle (b) of
+SURVEY.md §4: vanilla-engine results are ground truth)."""
+import math
+import os
+
+import numpy as np
+import pytest
+
+from auron_amd import AuronSession
+from auron_amd.tpcds import datagen
+from auron_amd.tpcds.oracle import ORACLES
+from auron_amd.tpcds.queries import QUERIES, Catalog
+
+SF = 0.01
+ROOT = os.path.join(os.path.dirname(__file__), "..", ".tpcds_cache")
+
+
+@pytest.fixture(scope="module")
+def dataset():
+    datagen.write_dataset(ROOT, SF)
+    return ROOT
+
+
+def rows_of(df):
+    out = []
+    for _, r in df.iterrows():
+        row = []
+        for v in r:
+            if isinstance(v, float) and math.isnan(v):
+                row.append(None)
+            elif isinstance(v, (np.integer,)):
+                row.append(int(v))
+            elif isinstance(v, (np.floating,)):
+                row.append(float(v))
+            else:
+                row.append(v)
+        out.append(tuple(row))
+    return out
+
+
+def assert_result_matches(batch, df):
+    got_d = batch.to_pydict()
+    got_cols = list(got_d.keys())
+    want_cols = list(df.columns)
+    assert got_cols == want_cols, f"{got_cols} != {want_cols}"
+    got_rows = sorted(zip(*got_d.values()),
+                      key=lambda r: tuple((v is None, str(v)) for v in r))
+    want_rows = sorted(rows_of(df), key=lambda r: tuple((v is None, str(v)) for v in r))
+    assert len(got_rows) == len(want_rows), \
+        f"{len(got_rows)} rows != {len(want_rows)}\n{got_rows[:5]}\n{want_rows[:5]}"
+    for a, b in zip(got_rows, want_rows):
+        for x, y in zip(a, b):
+            if isinstance(x, float) and isinstance(y, float):
+                assert math.isclose(x, y, rel_tol=1e-6, abs_tol=1e-6), (a, b)
+            elif isinstance(x, float) or isinstance(y, float):
+                assert x is not None and y is not None and math.isclose(float(x), float(y), rel_tol=1e-6), (a, b)
+            else:
+                assert x == y, (a, b)
+
+
+@pytest.mark.parametrize("qname", sorted(QUERIES.keys()))
+def test_query_vs_oracle(dataset, qname):
+    s = AuronSession()
+    cat = Catalog(dataset, SF)
+    plan = QUERIES[qname](cat, s)
+    got = s.collect(plan)
+    want = ORACLES[qname](dataset, SF)
+    assert_result_matches(got, want)
