@@ -76,6 +76,7 @@ def _empty_like(batch: RecordBatch) -> RecordBatch:
 class Executor:
     def __init__(self, ctx: Optional[ExecContext] = None):
         self.ctx = ctx or ExecContext.from_dist()
+        self._child_time = [0.0]
 
     # ------------------------------------------------------------- dispatch
     def execute(self, node: P.PlanNode) -> List[RecordBatch]:
@@ -83,8 +84,14 @@ class Executor:
         fn = getattr(self, f"_exec_{name}", None)
         if fn is None:
             raise NotImplementedError(f"operator {name}")
-        with self.ctx.timeit(f"op.{name}"):
-            out = fn(node)
+        t0 = time.perf_counter()
+        self._child_time.append(0.0)
+        out = fn(node)
+        dt = time.perf_counter() - t0
+        child = self._child_time.pop()
+        self._child_time[-1] += dt
+        m = self.ctx.metrics
+        m[f"op.{name}"] = m.get(f"op.{name}", 0.0) + (dt - child)
         assert out, f"{name} returned no batches"
         return out
 

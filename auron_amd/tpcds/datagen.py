@@ -451,10 +451,13 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
 
 def _nparts_for(table: str, sf: float) -> int:
     rows = row_count(table, sf)
-    return max(1, min(64, (rows + 2_000_000 - 1) // 2_000_000)) if table in (
-        "store_sales", "catalog_sales", "web_sales", "inventory",
-        "store_returns", "catalog_returns", "web_returns", "customer_demographics",
-    ) else 1
+    if table in ("store_sales", "catalog_sales", "web_sales", "inventory",
+                 "store_returns", "catalog_returns", "web_returns"):
+        # >=8 parts so an 8-GPU node shards fact scans evenly
+        return max(8, min(256, (rows + 2_000_000 - 1) // 2_000_000))
+    if table == "customer_demographics":
+        return max(1, min(16, (rows + 2_000_000 - 1) // 2_000_000))
+    return 1
 
 
 def dataset_root(root: str, sf: float) -> str:

@@ -101,6 +101,9 @@ def main():
         elapsed = float(t.item())
 
     suite_seconds = elapsed / args.steps
+    if rank == 0 and os.environ.get("AURON_BENCH_METRICS", "0") == "1":
+        for k, v in sorted(session.metrics().items(), key=lambda kv: -kv[1]):
+            log(f"metric {k}: {v:.3f}s")
     if rank == 0:
         out = {
             "metric": "tpcds_suite_seconds",
