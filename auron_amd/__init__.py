@@ -9,6 +9,11 @@ scatter) are hand-written HIP, and exchange runs over RCCL on xGMI via
 torch.distributed. See ARCHITECTURE.md for the component map against the
 reference.
 """
+import warnings as _warnings
+
+# scans wrap arrow buffers zero-copy; columns are never mutated in place
+_warnings.filterwarnings("ignore", message="The given NumPy array is not writable")
+
 from . import dtypes, exprs, ops
 from .column import Column, RecordBatch
 from .config import AuronConf

@@ -210,36 +210,37 @@ class Column:
             validity = torch.from_numpy(arr.is_valid().to_numpy(zero_copy_only=False))
         if dt.is_string:
             arr = arr.cast(pa.string())
-            # offsets/values buffers
+            # offsets/values buffers (zero-copy views into the arrow buffers;
+            # the arrow table is kept alive by the returned tensors' base)
             arr = arr.combine_chunks() if isinstance(arr, pa.ChunkedArray) else arr
-            offsets = torch.from_numpy(
-                np.frombuffer(arr.buffers()[1], dtype=np.int32, count=len(arr) + 1, offset=arr.offset * 4).copy()
-            )
-            start = int(offsets[0].item())
-            end = int(offsets[-1].item())
+            off_np = np.frombuffer(arr.buffers()[1], dtype=np.int32,
+                                   count=len(arr) + 1, offset=arr.offset * 4)
+            start = int(off_np[0])
+            end = int(off_np[-1])
             buf = arr.buffers()[2]
             if buf is None or end == start:
                 data = torch.empty(0, dtype=torch.uint8)
                 offsets = torch.zeros(len(arr) + 1, dtype=torch.int32)
             else:
-                data = torch.from_numpy(np.frombuffer(buf, dtype=np.uint8, count=end - start, offset=start).copy())
-                offsets = offsets - start
+                data = torch.from_numpy(np.frombuffer(buf, dtype=np.uint8, count=end - start, offset=start))
+                offsets = torch.from_numpy(off_np) if start == 0 else torch.from_numpy((off_np - start))
             col = Column(dt, data, validity, offsets)
         elif dt.code == dtypes.DECIMAL64:
             np_vals = arr.cast(pa.decimal128(dt.precision, dt.scale)).to_numpy(zero_copy_only=False)
             scaled = np.round(np_vals.astype(np.float64) * (10 ** dt.scale)).astype(np.int64)
             col = Column(dt, torch.from_numpy(scaled), validity, None)
         else:
-            np_arr = arr.to_numpy(zero_copy_only=False)
-            if np_arr.dtype == object or np_arr.dtype.kind in "OM":
-                if dt.code == dtypes.DATE32:
-                    np_arr = np.asarray(arr.cast(pa.int32()).to_numpy(zero_copy_only=False), dtype=np.int32)
-                else:
-                    np_arr = np.asarray(np_arr, dtype=np.float64)
+            if dt.code == dtypes.DATE32:
+                arr = arr.cast(pa.int32())
+            if arr.null_count:
+                arr = arr.fill_null(0)  # validity carries the null mask
+            np_arr = arr.to_numpy(zero_copy_only=arr.type != pa.bool_())
             if np_arr.dtype == np.bool_:
-                col = Column(dt, torch.from_numpy(np_arr.copy()), validity, None)
-            else:
-                col = Column(dt, torch.from_numpy(np.ascontiguousarray(np_arr)).to(dt.torch_dtype), validity, None)
+                np_arr = np.ascontiguousarray(np_arr)
+            t = torch.from_numpy(np_arr)
+            if t.dtype != dt.torch_dtype:
+                t = t.to(dt.torch_dtype)
+            col = Column(dt, t, validity, None)
         return col.to(device)
 
     def to_arrow(self):

@@ -105,17 +105,22 @@ class Executor:
 
     def _exec_ParquetScan(self, node: P.ParquetScan) -> List[RecordBatch]:
         import pyarrow.parquet as pq
+        from concurrent.futures import ThreadPoolExecutor
 
         my_files = node.paths[self.ctx.rank::self.ctx.world_size]
-        out = []
-        for f in my_files:
-            t = pq.read_table(f, columns=node.columns, filters=node.filters)
-            out.append(RecordBatch.from_arrow(t, self.ctx.device))
-        if not out:
+        if not my_files:
             # keep the >=1 batch invariant: 0-row batch with the file schema
             t = pq.read_table(node.paths[0], columns=node.columns).slice(0, 0)
-            out.append(RecordBatch.from_arrow(t, self.ctx.device))
-        return out
+            return [RecordBatch.from_arrow(t, self.ctx.device)]
+        if len(my_files) == 1:
+            t = pq.read_table(my_files[0], columns=node.columns, filters=node.filters)
+            return [RecordBatch.from_arrow(t, self.ctx.device)]
+        # overlap host page reads across files; convert+upload on this thread
+        with ThreadPoolExecutor(max_workers=min(8, len(my_files))) as pool:
+            tables = pool.map(
+                lambda f: pq.read_table(f, columns=node.columns, filters=node.filters),
+                my_files)
+            return [RecordBatch.from_arrow(t, self.ctx.device) for t in tables]
 
     # ------------------------------------------------------- row operators
     def _exec_Filter(self, node: P.Filter) -> List[RecordBatch]:

@@ -295,6 +295,16 @@ class Cmp(Expr):
     right: Expr
 
     def eval(self, batch: RecordBatch) -> Column:
+        # literal-string fast path: compare against the pattern directly
+        # instead of materializing a repeated literal column
+        if (self.op in ("==", "!=") and isinstance(self.right, Literal)
+                and isinstance(self.right.value, str)):
+            l = self.left.eval(batch)
+            if l.dtype.is_string:
+                m = strings.eq_literal(l, self.right.value)
+                if self.op == "!=":
+                    m = ~m
+                return Column(dtypes.bool_, m, l.validity)
         l = self.left.eval(batch)
         r = self.right.eval(batch)
         if l.dtype.is_string or r.dtype.is_string:
