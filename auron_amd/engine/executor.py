@@ -103,6 +103,18 @@ class Executor:
         assert node.batches
         return [b.to(self.ctx.device) for b in node.batches]
 
+    def _read_parquet(self, path: str, columns, filters) -> RecordBatch:
+        import pyarrow.parquet as pq
+
+        if filters is None and columns is not None:
+            from .. import parquet_native
+
+            cols = parquet_native.read_columns_native(path, columns, self.ctx.device)
+            if cols is not None:
+                return RecordBatch(list(columns), [cols[c] for c in columns])
+        t = pq.read_table(path, columns=columns, filters=filters)
+        return RecordBatch.from_arrow(t, self.ctx.device)
+
     def _exec_ParquetScan(self, node: P.ParquetScan) -> List[RecordBatch]:
         import pyarrow.parquet as pq
         from concurrent.futures import ThreadPoolExecutor
@@ -113,14 +125,11 @@ class Executor:
             t = pq.read_table(node.paths[0], columns=node.columns).slice(0, 0)
             return [RecordBatch.from_arrow(t, self.ctx.device)]
         if len(my_files) == 1:
-            t = pq.read_table(my_files[0], columns=node.columns, filters=node.filters)
-            return [RecordBatch.from_arrow(t, self.ctx.device)]
-        # overlap host page reads across files; convert+upload on this thread
+            return [self._read_parquet(my_files[0], node.columns, node.filters)]
+        # overlap host page reads / chunk staging across files
         with ThreadPoolExecutor(max_workers=min(8, len(my_files))) as pool:
-            tables = pool.map(
-                lambda f: pq.read_table(f, columns=node.columns, filters=node.filters),
-                my_files)
-            return [RecordBatch.from_arrow(t, self.ctx.device) for t in tables]
+            return list(pool.map(
+                lambda f: self._read_parquet(f, node.columns, node.filters), my_files))
 
     # ------------------------------------------------------- row operators
     def _exec_Filter(self, node: P.Filter) -> List[RecordBatch]:

@@ -12,7 +12,7 @@ import sys
 HERE = os.path.dirname(os.path.abspath(__file__))
 CSRC = os.path.join(HERE, "csrc")
 OUT = os.path.join(HERE, "libauron_hip.so")
-SOURCES = [os.path.join(CSRC, "kernels.hip")]
+SOURCES = [os.path.join(CSRC, "kernels.hip"), os.path.join(CSRC, "parquet.hip")]
 HIPCC = os.environ.get("HIPCC", "hipcc")
 ARCH = os.environ.get("AURON_OFFLOAD_ARCH", "gfx950")
 

@@ -1,0 +1,209 @@
+// GPU Parquet page decode — gfx950.
+//
+// Role parity: the reference's CPU parquet decode (parquet_exec.rs via
+// arrow-rs). Here the host ships raw uncompressed column-chunk bytes to
+// HBM and these kernels do the decode work:
+//  - k_pq_rle1:      RLE/bit-packed-hybrid definition levels (bit width 1)
+//                    -> validity bytes, one workgroup per page
+//  - k_pq_scatter:   PLAIN non-null values scattered to row slots using a
+//                    precomputed validity prefix sum
+//  - k_pq_copy:      PLAIN values for required (no-null) chunks, segmented
+//                    vectorized copy
+//
+// Page descriptor layout (int64 x 6, packed by parquet_native.py):
+//   [0]=def_off [1]=def_len [2]=values_off [3]=n_values [4]=row_start [5]=pad
+
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+
+#define AU_EXPORT extern "C" __attribute__((visibility("default")))
+
+struct PqPage {
+  int64_t def_off;
+  int64_t def_len;
+  int64_t values_off;
+  int64_t n_values;
+  int64_t row_start;
+  int64_t pad;
+};
+
+// ------------------------------------------------------------------- RLE1
+// Two-phase per block: lane 0 parses run headers into an LDS window, then
+// all 256 threads expand the window in parallel. Runs are (value,count) or
+// bit-packed literal groups.
+struct Run {
+  int32_t out_start;  // relative to page
+  int32_t count;
+  int32_t src_off;    // for literal runs: byte offset of packed bits
+  int32_t rep_val;    // for repeat runs: value; -1 => literal
+};
+
+#define RLE_WINDOW 512
+
+__global__ void k_pq_rle1(const PqPage* pages, const uint8_t* buf, uint8_t* out) {
+  const PqPage p = pages[blockIdx.x];
+  const uint8_t* src = buf + p.def_off;
+  int64_t src_len = p.def_len;
+  uint8_t* dst = out + p.row_start;
+  const int64_t n = p.n_values;
+
+  __shared__ Run runs[RLE_WINDOW];
+  __shared__ int nruns;
+  __shared__ int64_t s_pos, s_i;
+
+  if (threadIdx.x == 0) {
+    s_pos = 0;
+    s_i = 0;
+  }
+  __syncthreads();
+
+  while (true) {
+    if (threadIdx.x == 0) {
+      int64_t pos = s_pos;
+      int64_t i = s_i;
+      int r = 0;
+      while (r < RLE_WINDOW && i < n && pos < src_len) {
+        // uvarint
+        uint64_t header = 0;
+        int shift = 0;
+        while (true) {
+          uint8_t b = src[pos++];
+          header |= (uint64_t)(b & 0x7F) << shift;
+          if (!(b & 0x80)) break;
+          shift += 7;
+        }
+        if (header & 1) {  // literal: (header>>1) groups of 8 bit-packed
+          int64_t ngroups = header >> 1;
+          int64_t nvals = ngroups * 8;
+          if (nvals > n - i) nvals = n - i;
+          runs[r].out_start = (int32_t)i;
+          runs[r].count = (int32_t)nvals;
+          runs[r].src_off = (int32_t)pos;
+          runs[r].rep_val = -1;
+          pos += ngroups;
+          i += nvals;
+        } else {
+          int64_t cnt = header >> 1;
+          if (cnt > n - i) cnt = n - i;
+          runs[r].out_start = (int32_t)i;
+          runs[r].count = (int32_t)cnt;
+          runs[r].rep_val = src[pos] & 1;
+          pos += 1;
+          i += cnt;
+        }
+        r++;
+      }
+      nruns = r;
+      s_pos = pos;
+      s_i = i;
+    }
+    __syncthreads();
+    int count = nruns;
+    if (count == 0) break;
+    // expand: threads stride over runs
+    for (int r = threadIdx.x / 64; r < count; r += blockDim.x / 64) {
+      const Run run = runs[r];
+      int lane = threadIdx.x & 63;
+      if (run.rep_val >= 0) {
+        for (int j = lane; j < run.count; j += 64) dst[run.out_start + j] = (uint8_t)run.rep_val;
+      } else {
+        for (int j = lane; j < run.count; j += 64) {
+          dst[run.out_start + j] = (src[run.src_off + (j >> 3)] >> (j & 7)) & 1;
+        }
+      }
+    }
+    __syncthreads();
+    if (count < RLE_WINDOW) break;
+  }
+}
+
+AU_EXPORT int au_pq_rle1(const void* pages_dev, int npages, const void* buf,
+                         uint8_t* out, void* stream) {
+  if (npages == 0) return 0;
+  hipLaunchKernelGGL(k_pq_rle1, dim3(npages), dim3(256), 0, (hipStream_t)stream,
+                     (const PqPage*)pages_dev, (const uint8_t*)buf, out);
+  return (int)hipGetLastError();
+}
+
+// ----------------------------------------------------------------- scatter
+// out[row] = valid(row) ? values[prefix[row]-1 - page_base] : 0
+// prefix = inclusive cumsum of validity over the whole chunk (int64).
+template <typename T>
+__global__ void k_pq_scatter(const PqPage* pages, int npages, const uint8_t* buf,
+                             const uint8_t* validity, const int64_t* prefix,
+                             T* out, int64_t total) {
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < total;
+       row += (int64_t)gridDim.x * blockDim.x) {
+    // binary search the page containing `row`
+    int lo = 0, hi = npages - 1;
+    while (lo < hi) {
+      int mid = (lo + hi + 1) >> 1;
+      if (pages[mid].row_start <= row) lo = mid; else hi = mid - 1;
+    }
+    const PqPage p = pages[lo];
+    if (!validity[row]) {
+      out[row] = (T)0;
+      continue;
+    }
+    int64_t page_base = p.row_start == 0 ? 0 : prefix[p.row_start - 1];
+    int64_t k = prefix[row] - 1 - page_base;
+    T v;
+    __builtin_memcpy(&v, buf + p.values_off + k * (int64_t)sizeof(T), sizeof(T));
+    out[row] = v;
+  }
+}
+
+// ---------------------------------------------------------------- copy (no nulls)
+template <typename T>
+__global__ void k_pq_copy(const PqPage* pages, int npages, const uint8_t* buf,
+                          T* out, int64_t total) {
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < total;
+       row += (int64_t)gridDim.x * blockDim.x) {
+    int lo = 0, hi = npages - 1;
+    while (lo < hi) {
+      int mid = (lo + hi + 1) >> 1;
+      if (pages[mid].row_start <= row) lo = mid; else hi = mid - 1;
+    }
+    const PqPage p = pages[lo];
+    int64_t k = row - p.row_start;
+    T v;
+    __builtin_memcpy(&v, buf + p.values_off + k * (int64_t)sizeof(T), sizeof(T));
+    out[row] = v;
+  }
+}
+
+static inline int pq_grid(int64_t n) {
+  int64_t g = (n + 255) / 256;
+  if (g > 2048) g = 2048;
+  if (g < 1) g = 1;
+  return (int)g;
+}
+
+AU_EXPORT int au_pq_scatter(const void* pages_dev, int npages, const void* buf,
+                            const uint8_t* validity, const int64_t* prefix,
+                            void* out, int esize, int64_t total, void* stream) {
+  if (total == 0) return 0;
+  hipStream_t s = (hipStream_t)stream;
+  dim3 g(pq_grid(total)), b(256);
+  if (esize == 4)
+    hipLaunchKernelGGL((k_pq_scatter<uint32_t>), g, b, 0, s, (const PqPage*)pages_dev,
+                       npages, (const uint8_t*)buf, validity, prefix, (uint32_t*)out, total);
+  else
+    hipLaunchKernelGGL((k_pq_scatter<uint64_t>), g, b, 0, s, (const PqPage*)pages_dev,
+                       npages, (const uint8_t*)buf, validity, prefix, (uint64_t*)out, total);
+  return (int)hipGetLastError();
+}
+
+AU_EXPORT int au_pq_copy_plain(const void* pages_dev, int npages, const void* buf,
+                               void* out, int esize, int64_t total, void* stream) {
+  if (total == 0) return 0;
+  hipStream_t s = (hipStream_t)stream;
+  dim3 g(pq_grid(total)), b(256);
+  if (esize == 4)
+    hipLaunchKernelGGL((k_pq_copy<uint32_t>), g, b, 0, s, (const PqPage*)pages_dev,
+                       npages, (const uint8_t*)buf, (uint32_t*)out, total);
+  else
+    hipLaunchKernelGGL((k_pq_copy<uint64_t>), g, b, 0, s, (const PqPage*)pages_dev,
+                       npages, (const uint8_t*)buf, (uint64_t*)out, total);
+  return (int)hipGetLastError();
+}

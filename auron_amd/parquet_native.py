@@ -1,0 +1,414 @@
+"""Native Parquet read path: host page-header parse -> device page decode.
+
+Role parity: the reference's parquet scan (parquet_exec.rs + arrow-rs
+parquet decoder, CPU). MI355X design: the host only touches metadata —
+footer via pyarrow, page headers via a minimal thrift-compact parser —
+and ships the raw column-chunk bytes to HBM once; gfx950 kernels decode
+RLE/bit-packed definition levels and scatter PLAIN values into device
+columns (csrc/parquet.hip). Unsupported shapes (strings, dictionary,
+compressed pages, nested) fall back to the pyarrow host path per column.
+
+Supported fast path: UNCOMPRESSED column chunks, PLAIN-encoded
+INT32/INT64/FLOAT/DOUBLE, optional-level (max_def_level<=1) columns,
+data page v1 — exactly what the TPC-DS fact tables are written as.
+"""
+from __future__ import annotations
+
+import os
+from dataclasses import dataclass
+from typing import Dict, List, Optional, Tuple
+
+import numpy as np
+import torch
+
+from . import dtypes
+from .column import Column, RecordBatch
+
+# ---------------------------------------------------------------- thrift
+# Minimal thrift compact-protocol reader (PageHeader only).
+_CT_STOP = 0
+_CT_BOOL_TRUE = 1
+_CT_BOOL_FALSE = 2
+_CT_BYTE = 3
+_CT_I16 = 4
+_CT_I32 = 5
+_CT_I64 = 6
+_CT_DOUBLE = 7
+_CT_BINARY = 8
+_CT_LIST = 9
+_CT_SET = 10
+_CT_MAP = 11
+_CT_STRUCT = 12
+
+
+def _uvarint(buf: memoryview, pos: int) -> Tuple[int, int]:
+    result = 0
+    shift = 0
+    while True:
+        b = buf[pos]
+        pos += 1
+        result |= (b & 0x7F) << shift
+        if not b & 0x80:
+            return result, pos
+        shift += 7
+
+
+def _zigzag(v: int) -> int:
+    return (v >> 1) ^ -(v & 1)
+
+
+def _skip(buf: memoryview, pos: int, ctype: int) -> int:
+    if ctype in (_CT_BOOL_TRUE, _CT_BOOL_FALSE):
+        return pos
+    if ctype == _CT_BYTE:
+        return pos + 1
+    if ctype in (_CT_I16, _CT_I32, _CT_I64):
+        v, pos = _uvarint(buf, pos)
+        return pos
+    if ctype == _CT_DOUBLE:
+        return pos + 8
+    if ctype == _CT_BINARY:
+        n, pos = _uvarint(buf, pos)
+        return pos + n
+    if ctype in (_CT_LIST, _CT_SET):
+        b = buf[pos]
+        pos += 1
+        size = b >> 4
+        etype = b & 0x0F
+        if size == 15:
+            size, pos = _uvarint(buf, pos)
+        for _ in range(size):
+            pos = _skip(buf, pos, etype)
+        return pos
+    if ctype == _CT_STRUCT:
+        fid = 0
+        while True:
+            b = buf[pos]
+            pos += 1
+            if b == _CT_STOP:
+                return pos
+            delta = b >> 4
+            ft = b & 0x0F
+            if delta == 0:
+                v, pos = _uvarint(buf, pos)
+                fid = _zigzag(v)
+            else:
+                fid += delta
+            pos = _skip(buf, pos, ft)
+    if ctype == _CT_MAP:
+        n, pos = _uvarint(buf, pos)
+        if n:
+            kv = buf[pos]
+            pos += 1
+            for _ in range(n):
+                pos = _skip(buf, pos, kv >> 4)
+                pos = _skip(buf, pos, kv & 0x0F)
+        return pos
+    raise ValueError(f"thrift type {ctype}")
+
+
+def _read_struct_fields(buf: memoryview, pos: int, want: Dict[int, str]) -> Tuple[dict, int]:
+    """Read a compact struct, capturing integer fields in `want` and the
+    byte range of any wanted struct fields."""
+    out: dict = {}
+    fid = 0
+    while True:
+        b = buf[pos]
+        pos += 1
+        if b == _CT_STOP:
+            return out, pos
+        delta = b >> 4
+        ft = b & 0x0F
+        if delta == 0:
+            v, pos = _uvarint(buf, pos)
+            fid = _zigzag(v)
+        else:
+            fid += delta
+        name = want.get(fid)
+        if name is None:
+            pos = _skip(buf, pos, ft)
+        elif ft in (_CT_I16, _CT_I32, _CT_I64):
+            v, pos = _uvarint(buf, pos)
+            out[name] = _zigzag(v)
+        elif ft == _CT_STRUCT:
+            start = pos
+            pos = _skip(buf, pos, _CT_STRUCT)
+            out[name] = (start, pos)
+        elif ft in (_CT_BOOL_TRUE, _CT_BOOL_FALSE):
+            out[name] = ft == _CT_BOOL_TRUE
+        else:
+            pos = _skip(buf, pos, ft)
+    return out, pos
+
+
+@dataclass
+class PageDesc:
+    n_values: int
+    def_off: int  # byte offset of RLE def-level payload (after 4-byte len), -1 if none
+    def_len: int
+    values_off: int  # byte offset of PLAIN values
+    values_len: int
+    row_start: int
+
+
+def parse_pages(buf: np.ndarray, chunk_off: int, chunk_len: int,
+                num_values: int, has_def: bool) -> Optional[List[PageDesc]]:
+    """Parse v1 data page headers of one uncompressed column chunk."""
+    mv = memoryview(buf)
+    pos = chunk_off
+    end = chunk_off + chunk_len
+    pages: List[PageDesc] = []
+    row = 0
+    while row < num_values and pos < end:
+        hdr, pos2 = _read_struct_fields(mv, pos, {
+            1: "type", 2: "uncompressed_page_size", 3: "compressed_page_size",
+            5: "data_page_header", 7: "dictionary_page_header",
+        })
+        page_data = pos2
+        page_len = hdr.get("compressed_page_size", 0)
+        ptype = hdr.get("type", -1)
+        if ptype == 2:  # dictionary page -> unsupported fast path
+            return None
+        if ptype != 0:  # v2 or index page
+            return None
+        dph_range = hdr.get("data_page_header")
+        if dph_range is None:
+            return None
+        dph, _ = _read_struct_fields(mv, dph_range[0], {
+            1: "num_values", 2: "encoding", 3: "def_enc", 4: "rep_enc"})
+        if dph.get("encoding", 0) != 0:  # PLAIN only
+            return None
+        nv = dph["num_values"]
+        if has_def:
+            dlen = int.from_bytes(bytes(mv[page_data:page_data + 4]), "little")
+            def_off = page_data + 4
+            values_off = def_off + dlen
+            values_len = page_len - 4 - dlen
+            pages.append(PageDesc(nv, def_off, dlen, values_off, values_len, row))
+        else:
+            pages.append(PageDesc(nv, -1, 0, page_data, page_len, row))
+        row += nv
+        pos = page_data + page_len
+    if row != num_values:
+        return None
+    return pages
+
+
+# ------------------------------------------------------- numpy reference
+def rle1_decode_np(data: np.ndarray, n: int) -> np.ndarray:
+    """RLE/bit-packed hybrid, bit width 1 (definition levels)."""
+    out = np.zeros(n, dtype=np.uint8)
+    mv = memoryview(data)
+    pos = 0
+    i = 0
+    while i < n:
+        header, pos = _uvarint(mv, pos)
+        if header & 1:  # bit-packed literal run: (header>>1) groups of 8
+            ngroups = header >> 1
+            bits = np.unpackbits(data[pos:pos + ngroups], bitorder="little")
+            take = min(ngroups * 8, n - i)
+            out[i:i + take] = bits[:take]
+            pos += ngroups
+            i += take
+        else:  # repeated run
+            cnt = header >> 1
+            val = data[pos]
+            pos += 1
+            out[i:i + cnt] = val & 1
+            i += cnt
+    return out
+
+
+_PHYS_NP = {"INT32": np.int32, "INT64": np.int64, "FLOAT": np.float32, "DOUBLE": np.float64}
+_PHYS_CODE = {"INT32": 0, "INT64": 1, "FLOAT": 2, "DOUBLE": 3}
+
+
+def decode_chunk_np(buf: np.ndarray, pages: List[PageDesc], num_values: int,
+                    phys: str) -> Tuple[np.ndarray, Optional[np.ndarray]]:
+    """Host reference decoder (correctness oracle for the HIP kernels)."""
+    npdt = _PHYS_NP[phys]
+    esize = npdt().itemsize
+    out = np.zeros(num_values, dtype=npdt)
+    validity = None
+    has_def = pages[0].def_off >= 0
+    if has_def:
+        validity = np.zeros(num_values, dtype=np.uint8)
+    for p in pages:
+        if has_def:
+            v = rle1_decode_np(buf[p.def_off:p.def_off + p.def_len], p.n_values)
+            validity[p.row_start:p.row_start + p.n_values] = v
+            nvalid = int(v.sum())
+            vals = np.frombuffer(buf, dtype=npdt, count=nvalid, offset=p.values_off)
+            out[p.row_start:p.row_start + p.n_values][v.astype(bool)] = vals
+        else:
+            vals = np.frombuffer(buf, dtype=npdt, count=p.n_values, offset=p.values_off)
+            out[p.row_start:p.row_start + p.n_values] = vals
+    return out, validity
+
+
+# ------------------------------------------------------------ file reader
+_ARROW_TO_AURON = {
+    "int32": dtypes.int32, "int64": dtypes.int64, "float": dtypes.float32,
+    "double": dtypes.float64, "date32[day]": dtypes.date32,
+}
+
+
+class NativeParquetFile:
+    """Per-file metadata: which columns take the device fast path."""
+
+    def __init__(self, path: str):
+        import pyarrow.parquet as pq
+
+        self.path = path
+        self.pf = pq.ParquetFile(path)
+        self.md = self.pf.metadata
+        self.schema = self.pf.schema
+        self.names = [self.schema.column(i).name for i in range(self.md.num_columns)]
+
+    def column_supported(self, name: str) -> bool:
+        i = self.names.index(name)
+        sc = self.schema.column(i)
+        if sc.max_definition_level > 1 or sc.max_repetition_level > 0:
+            return False
+        if sc.physical_type not in _PHYS_NP:
+            return False
+        for rg in range(self.md.num_row_groups):
+            cm = self.md.row_group(rg).column(i)
+            if cm.compression != "UNCOMPRESSED":
+                return False
+            if getattr(cm, "has_dictionary_page", False) or cm.dictionary_page_offset is not None:
+                return False
+        return True
+
+
+def _chunk_meta(md, rg: int, ci: int):
+    cm = md.row_group(rg).column(ci)
+    off = cm.data_page_offset
+    if cm.dictionary_page_offset is not None:
+        off = min(off, cm.dictionary_page_offset)
+    return off, cm.total_compressed_size, cm.num_values, cm.physical_type, cm.statistics
+
+
+def read_columns_native(path: str, columns: List[str], device,
+                        _np_only: bool = False) -> Optional[Dict[str, Column]]:
+    """Decode `columns` of `path` on `device`. Returns None if any column
+    is unsupported (caller falls back wholesale for simplicity).
+
+    IO strategy: memmap the file, copy ONLY the needed column-chunk byte
+    ranges into one compact host buffer, upload it to HBM once, then run
+    the decode kernels against device-resident page bytes."""
+    nf = NativeParquetFile(path)
+    md = nf.md
+    for c in columns:
+        if c not in nf.names or not nf.column_supported(c):
+            return None
+
+    mm = np.memmap(path, dtype=np.uint8, mode="r")
+    # gather needed chunk ranges -> compact buffer
+    ranges = []  # (cname, rg, src_off, length, nvals, phys, stats)
+    for cname in columns:
+        ci = nf.names.index(cname)
+        for rg in range(md.num_row_groups):
+            off, clen, nvals, phys, stats = _chunk_meta(md, rg, ci)
+            ranges.append((cname, rg, off, clen, nvals, phys, stats))
+    total = sum(r[3] for r in ranges)
+    buf = np.empty(total, dtype=np.uint8)
+    new_off = {}
+    pos = 0
+    for (cname, rg, off, clen, *_rest) in ranges:
+        buf[pos:pos + clen] = mm[off:off + clen]
+        new_off[(cname, rg)] = pos
+        pos += clen
+
+    use_gpu = (not _np_only) and torch.device(device).type == "cuda"
+    dbuf = torch.from_numpy(buf).to(device) if use_gpu else None
+
+    out: Dict[str, Column] = {}
+    for cname in columns:
+        ci = nf.names.index(cname)
+        sc = nf.schema.column(ci)
+        has_def = sc.max_definition_level == 1
+        logical = str(sc.logical_type).lower()
+        parts_data = []
+        parts_valid = []
+        any_nulls = False
+        for (cn, rg, off, clen, nvals, phys, stats) in ranges:
+            if cn != cname:
+                continue
+            pages = parse_pages(buf, new_off[(cname, rg)], clen, nvals, has_def)
+            if pages is None:
+                return None
+            if use_gpu:
+                data_t, valid_t = _decode_chunk_gpu(dbuf, pages, nvals, phys, device)
+            else:
+                data_np, valid_np = decode_chunk_np(buf, pages, nvals, phys)
+                data_t = torch.from_numpy(data_np)
+                valid_t = torch.from_numpy(valid_np).to(torch.bool) if valid_np is not None else None
+            parts_data.append(data_t)
+            if valid_t is not None:
+                parts_valid.append(valid_t)
+                if stats is None or not stats.has_null_count or stats.null_count > 0:
+                    any_nulls = True
+        data = torch.cat(parts_data) if len(parts_data) > 1 else parts_data[0]
+        validity = None
+        if parts_valid and len(parts_valid) == len(parts_data):
+            validity = torch.cat(parts_valid) if len(parts_valid) > 1 else parts_valid[0]
+            if not any_nulls:
+                validity = None
+            elif not use_gpu and bool(validity.all()):
+                validity = None
+        phys = sc.physical_type
+        if "date" in logical:
+            dt = dtypes.date32
+        elif phys == "INT32":
+            dt = dtypes.int32
+        elif phys == "INT64":
+            dt = dtypes.int64
+        elif phys == "FLOAT":
+            dt = dtypes.float32
+        else:
+            dt = dtypes.float64
+        if not use_gpu:
+            data = data.to(device)
+            if validity is not None:
+                validity = validity.to(device)
+        out[cname] = Column(dt, data, validity)
+    return out
+
+
+def _decode_chunk_gpu(dbuf: torch.Tensor, pages: List[PageDesc], num_values: int,
+                      phys: str, device):
+    """HIP kernel path: RLE def-levels + PLAIN value scatter on device."""
+    from . import native
+
+    lib = native.require()
+    import ctypes
+
+    esize = _PHYS_NP[phys]().itemsize
+    has_def = pages[0].def_off >= 0
+    npages = len(pages)
+    # descriptor layout (int64 x 6): def_off, def_len, values_off, n, row_start, pad
+    arr = np.zeros((npages, 6), dtype=np.int64)
+    for i, p in enumerate(pages):
+        arr[i] = (p.def_off, p.def_len, p.values_off, p.n_values, p.row_start, 0)
+    darr = torch.from_numpy(arr.reshape(-1)).to(device)
+    sp = native.stream_ptr(device)
+
+    tdt = {"INT32": torch.int32, "INT64": torch.int64,
+           "FLOAT": torch.float32, "DOUBLE": torch.float64}[phys]
+    out = torch.empty(num_values, dtype=tdt, device=device)
+    if not has_def:
+        rc = lib.au_pq_copy_plain(darr.data_ptr(), npages, dbuf.data_ptr(),
+                                  out.data_ptr(), esize, num_values, sp)
+        native.check(rc, "au_pq_copy_plain")
+        return out, None
+    validity = torch.empty(num_values, dtype=torch.uint8, device=device)
+    rc = lib.au_pq_rle1(darr.data_ptr(), npages, dbuf.data_ptr(),
+                        validity.data_ptr(), sp)
+    native.check(rc, "au_pq_rle1")
+    prefix = torch.cumsum(validity.to(torch.int64), 0)
+    rc = lib.au_pq_scatter(darr.data_ptr(), npages, dbuf.data_ptr(),
+                           validity.data_ptr(), prefix.data_ptr(),
+                           out.data_ptr(), esize, num_values, sp)
+    native.check(rc, "au_pq_scatter")
+    return out, validity.to(torch.bool)

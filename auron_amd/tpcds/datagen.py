@@ -16,7 +16,7 @@ from typing import List, Optional
 
 import numpy as np
 
-from .schema import BASE_ROWS, SCHEMAS, row_count
+from .schema import BASE_ROWS, FACT_TABLES, SCHEMAS, row_count
 
 JULIAN_BASE = 2415022  # d_date_sk of calendar index 0 == 1900-01-01
 EPOCH_IDX = 25567  # calendar index of 1970-01-01
@@ -490,8 +490,18 @@ def write_dataset(root: str, sf: float, tables: Optional[List[str]] = None,
             if job % world == rank:
                 if force or not os.path.exists(path):
                     tbl = generate_table(t, sf, p, nparts)
-                    # lz4 decompresses ~5x faster than zstd; scan is the
-                    # critical path (fact scans are re-read per query)
-                    pq.write_table(tbl, path, compression="lz4", row_group_size=1 << 20)
+                    if t in FACT_TABLES:
+                        # uncompressed PLAIN pages: the GPU parquet decoder
+                        # (parquet_native.py + csrc/parquet.hip) consumes the
+                        # page bytes directly in HBM — no host decode at all
+                        pq.write_table(tbl, path, compression="NONE",
+                                       use_dictionary=False,
+                                       data_page_version="1.0",
+                                       row_group_size=1 << 20)
+                    else:
+                        # dims carry strings; host (pyarrow) path, lz4 for
+                        # fast repeated decompress
+                        pq.write_table(tbl, path, compression="lz4",
+                                       row_group_size=1 << 20)
             job += 1
     return base
