@@ -109,9 +109,16 @@ class Executor:
         if filters is None and columns is not None:
             from .. import parquet_native
 
-            cols = parquet_native.read_columns_native(path, columns, self.ctx.device)
-            if cols is not None:
-                return RecordBatch(list(columns), [cols[c] for c in columns])
+            native_cols, host_cols = parquet_native.split_supported(path, columns)
+            if native_cols:
+                got = parquet_native.read_columns_native(path, native_cols, self.ctx.device)
+                if got is not None:
+                    if host_cols:
+                        t = pq.read_table(path, columns=host_cols)
+                        hb = RecordBatch.from_arrow(t, self.ctx.device)
+                        for n, c in zip(hb.names, hb.columns):
+                            got[n] = c
+                    return RecordBatch(list(columns), [got[c] for c in columns])
         t = pq.read_table(path, columns=columns, filters=filters)
         return RecordBatch.from_arrow(t, self.ctx.device)
 

@@ -252,6 +252,12 @@ _ARROW_TO_AURON = {
     "double": dtypes.float64, "date32[day]": dtypes.date32,
 }
 
+# footer + page-header metadata cache, keyed by (path, mtime, columns).
+# Metadata only (never data): the Spark-side analogue is the parquet
+# footer cache; pages are re-read and re-decoded on every scan.
+_META_CACHE: Dict[tuple, object] = {}
+_META_CACHE_MAX = 4096
+
 
 class NativeParquetFile:
     """Per-file metadata: which columns take the device fast path."""
@@ -289,74 +295,37 @@ def _chunk_meta(md, rg: int, ci: int):
     return off, cm.total_compressed_size, cm.num_values, cm.physical_type, cm.statistics
 
 
-def read_columns_native(path: str, columns: List[str], device,
-                        _np_only: bool = False) -> Optional[Dict[str, Column]]:
-    """Decode `columns` of `path` on `device`. Returns None if any column
-    is unsupported (caller falls back wholesale for simplicity).
+@dataclass
+class _ColMeta:
+    name: str
+    dtype: object = None  # auron DataType
+    phys: str = ""
+    has_def: bool = False
+    chunks: list = None  # [(new_off, clen, nvals, any_nulls)]
+    pages: list = None  # [[PageDesc]] rebased to compact buffer; lazy
 
-    IO strategy: memmap the file, copy ONLY the needed column-chunk byte
-    ranges into one compact host buffer, upload it to HBM once, then run
-    the decode kernels against device-resident page bytes."""
+
+@dataclass
+class _FileMeta:
+    total: int
+    ranges: list  # [(src_off, clen, new_off)]
+    cols: list  # [_ColMeta]
+    parsed: bool = False
+
+
+def _build_meta(path: str, columns: List[str]) -> Optional[_FileMeta]:
     nf = NativeParquetFile(path)
     md = nf.md
     for c in columns:
         if c not in nf.names or not nf.column_supported(c):
             return None
-
-    mm = np.memmap(path, dtype=np.uint8, mode="r")
-    # gather needed chunk ranges -> compact buffer
-    ranges = []  # (cname, rg, src_off, length, nvals, phys, stats)
-    for cname in columns:
-        ci = nf.names.index(cname)
-        for rg in range(md.num_row_groups):
-            off, clen, nvals, phys, stats = _chunk_meta(md, rg, ci)
-            ranges.append((cname, rg, off, clen, nvals, phys, stats))
-    total = sum(r[3] for r in ranges)
-    buf = np.empty(total, dtype=np.uint8)
-    new_off = {}
+    ranges = []
+    cols = []
     pos = 0
-    for (cname, rg, off, clen, *_rest) in ranges:
-        buf[pos:pos + clen] = mm[off:off + clen]
-        new_off[(cname, rg)] = pos
-        pos += clen
-
-    use_gpu = (not _np_only) and torch.device(device).type == "cuda"
-    dbuf = torch.from_numpy(buf).to(device) if use_gpu else None
-
-    out: Dict[str, Column] = {}
     for cname in columns:
         ci = nf.names.index(cname)
         sc = nf.schema.column(ci)
-        has_def = sc.max_definition_level == 1
         logical = str(sc.logical_type).lower()
-        parts_data = []
-        parts_valid = []
-        any_nulls = False
-        for (cn, rg, off, clen, nvals, phys, stats) in ranges:
-            if cn != cname:
-                continue
-            pages = parse_pages(buf, new_off[(cname, rg)], clen, nvals, has_def)
-            if pages is None:
-                return None
-            if use_gpu:
-                data_t, valid_t = _decode_chunk_gpu(dbuf, pages, nvals, phys, device)
-            else:
-                data_np, valid_np = decode_chunk_np(buf, pages, nvals, phys)
-                data_t = torch.from_numpy(data_np)
-                valid_t = torch.from_numpy(valid_np).to(torch.bool) if valid_np is not None else None
-            parts_data.append(data_t)
-            if valid_t is not None:
-                parts_valid.append(valid_t)
-                if stats is None or not stats.has_null_count or stats.null_count > 0:
-                    any_nulls = True
-        data = torch.cat(parts_data) if len(parts_data) > 1 else parts_data[0]
-        validity = None
-        if parts_valid and len(parts_valid) == len(parts_data):
-            validity = torch.cat(parts_valid) if len(parts_valid) > 1 else parts_valid[0]
-            if not any_nulls:
-                validity = None
-            elif not use_gpu and bool(validity.all()):
-                validity = None
         phys = sc.physical_type
         if "date" in logical:
             dt = dtypes.date32
@@ -368,11 +337,107 @@ def read_columns_native(path: str, columns: List[str], device,
             dt = dtypes.float32
         else:
             dt = dtypes.float64
+        cm = _ColMeta(cname, dt, phys, sc.max_definition_level == 1, [], None)
+        for rg in range(md.num_row_groups):
+            off, clen, nvals, _phys, stats = _chunk_meta(md, rg, ci)
+            any_nulls = stats is None or not stats.has_null_count or stats.null_count > 0
+            ranges.append((off, clen, pos))
+            cm.chunks.append((pos, clen, nvals, any_nulls))
+            pos += clen
+        cols.append(cm)
+    return _FileMeta(pos, ranges, cols)
+
+
+def _get_meta(path: str, columns: List[str]) -> Optional[_FileMeta]:
+    key = (path, os.path.getmtime(path), tuple(columns))
+    if key in _META_CACHE:
+        return _META_CACHE[key]
+    meta = _build_meta(path, columns)
+    if len(_META_CACHE) > _META_CACHE_MAX:
+        _META_CACHE.clear()
+    _META_CACHE[key] = meta
+    return meta
+
+
+def split_supported(path: str, columns: List[str]) -> Tuple[List[str], List[str]]:
+    """Partition `columns` into (native-decodable, host-fallback)."""
+    key = (path, "split", tuple(columns))
+    if key in _META_CACHE:
+        return _META_CACHE[key]
+    try:
+        nf = NativeParquetFile(path)
+        ok = [c for c in columns if c in nf.names and nf.column_supported(c)]
+        rest = [c for c in columns if c not in ok]
+    except Exception:
+        ok, rest = [], list(columns)
+    _META_CACHE[key] = (ok, rest)
+    return ok, rest
+
+
+def read_columns_native(path: str, columns: List[str], device,
+                        _np_only: bool = False) -> Optional[Dict[str, Column]]:
+    """Decode `columns` of `path` on `device`. Returns None if any column
+    is unsupported (caller falls back wholesale for simplicity).
+
+    IO strategy: memmap the file, copy ONLY the needed column-chunk byte
+    ranges into one pinned host buffer, upload it to HBM once, then run
+    the decode kernels against device-resident page bytes. Footer and
+    page-header metadata is cached per (path, mtime)."""
+    columns = list(columns)
+    meta = _get_meta(path, columns)
+    if meta is None:
+        return None
+    use_gpu = (not _np_only) and torch.device(device).type == "cuda"
+
+    mm = np.memmap(path, dtype=np.uint8, mode="r")
+    buf_t = torch.empty(meta.total, dtype=torch.uint8, pin_memory=use_gpu)
+    buf = buf_t.numpy()
+    for (src, clen, dst) in meta.ranges:
+        buf[dst:dst + clen] = mm[src:src + clen]
+
+    if not meta.parsed:
+        for cm in meta.cols:
+            cm.pages = []
+            for (new_off, clen, nvals, _an) in cm.chunks:
+                pages = parse_pages(buf, new_off, clen, nvals, cm.has_def)
+                if pages is None:
+                    return None
+                cm.pages.append(pages)
+        meta.parsed = True
+
+    dbuf = buf_t.to(device, non_blocking=True) if use_gpu else None
+
+    out: Dict[str, Column] = {}
+    for cm in meta.cols:
+        parts_data = []
+        parts_valid = []
+        any_nulls = False
+        for (chunk, pages) in zip(cm.chunks, cm.pages):
+            (_off, _clen, nvals, chunk_nulls) = chunk
+            if use_gpu:
+                data_t, valid_t = _decode_chunk_gpu(dbuf, pages, nvals, cm.phys, device)
+            else:
+                data_np, valid_np = decode_chunk_np(buf, pages, nvals, cm.phys)
+                data_t = torch.from_numpy(data_np)
+                valid_t = torch.from_numpy(valid_np).to(torch.bool) if valid_np is not None else None
+            parts_data.append(data_t)
+            if valid_t is not None:
+                parts_valid.append(valid_t)
+                if chunk_nulls:
+                    any_nulls = True
+        data = torch.cat(parts_data) if len(parts_data) > 1 else parts_data[0]
+        validity = None
+        if parts_valid and len(parts_valid) == len(parts_data):
+            validity = torch.cat(parts_valid) if len(parts_valid) > 1 else parts_valid[0]
+            if not any_nulls:
+                validity = None
+            elif not use_gpu and bool(validity.all()):
+                validity = None
         if not use_gpu:
             data = data.to(device)
             if validity is not None:
                 validity = validity.to(device)
-        out[cname] = Column(dt, data, validity)
+        out[cm.name] = Column(cm.dtype, data, validity)
     return out
 
 

@@ -490,18 +490,13 @@ def write_dataset(root: str, sf: float, tables: Optional[List[str]] = None,
             if job % world == rank:
                 if force or not os.path.exists(path):
                     tbl = generate_table(t, sf, p, nparts)
-                    if t in FACT_TABLES:
-                        # uncompressed PLAIN pages: the GPU parquet decoder
-                        # (parquet_native.py + csrc/parquet.hip) consumes the
-                        # page bytes directly in HBM — no host decode at all
-                        pq.write_table(tbl, path, compression="NONE",
-                                       use_dictionary=False,
-                                       data_page_version="1.0",
-                                       row_group_size=1 << 20)
-                    else:
-                        # dims carry strings; host (pyarrow) path, lz4 for
-                        # fast repeated decompress
-                        pq.write_table(tbl, path, compression="lz4",
-                                       row_group_size=1 << 20)
+                    # uncompressed PLAIN pages: the GPU parquet decoder
+                    # (parquet_native.py + csrc/parquet.hip) consumes numeric
+                    # page bytes directly in HBM — no host decode at all;
+                    # string columns fall back to the pyarrow host path
+                    pq.write_table(tbl, path, compression="NONE",
+                                   use_dictionary=False,
+                                   data_page_version="1.0",
+                                   row_group_size=1 << 20)
             job += 1
     return base
