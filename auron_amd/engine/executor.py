@@ -37,6 +37,13 @@ class ExecContext:
     group: object = None
     batch_rows: int = 1 << 22
     metrics: Dict[str, float] = field(default_factory=dict)
+    memmgr: object = None  # auron_amd.memory.MemManager
+
+    def __post_init__(self):
+        if self.memmgr is None:
+            from ..memory import MemManager
+
+            self.memmgr = MemManager()
 
     @staticmethod
     def from_dist(device=None) -> "ExecContext":
@@ -425,18 +432,30 @@ class Executor:
     # ------------------------------------------------------------ hash join
     def _exec_HashJoin(self, node: P.HashJoin) -> List[RecordBatch]:
         left_bs = self.execute(node.left)
+        # the materialized left side is spillable while the right side runs
+        # (memmgr may push it to host/disk under pressure, lib.rs:308 pattern)
+        holder = self.ctx.memmgr.register("join-left", left_bs)
+        try:
+            return self._exec_hash_join_inner(node, holder)
+        finally:
+            holder.release()
+
+    def _exec_hash_join_inner(self, node: P.HashJoin, left_holder) -> List[RecordBatch]:
+        def left_batches():
+            return left_holder.batches()
+
         if node.broadcast and self.ctx.world_size > 1:
             if node.build_side == "right":
                 rb = _concat(self.execute(node.right))
                 right = _concat(all_gather_batch(rb, self.ctx.device, self.ctx.group))
-                left = _concat(left_bs)
+                left = _concat(left_batches())
             else:
-                lb = _concat(left_bs)
+                lb = _concat(left_batches())
                 left = _concat(all_gather_batch(lb, self.ctx.device, self.ctx.group))
                 right = _concat(self.execute(node.right))
         else:
-            left = _concat(left_bs)
             right = _concat(self.execute(node.right))
+            left = _concat(left_batches())
         lkeys = [k.eval(left) for k in node.left_keys]
         rkeys = [k.eval(right) for k in node.right_keys]
         how = node.how

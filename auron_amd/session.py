@@ -37,6 +37,14 @@ class AuronSession:
         """Run a physical plan; returns this rank's local batches."""
         return self.executor.execute(plan)
 
+    def execute_serialized(self, task_bytes: bytes):
+        """JniBridge.callNative analogue: run a serialized TaskDefinition
+        (plan/serde.py) and return this rank's batches."""
+        from .plan import serde
+
+        task_id, stage_id, partition, plan = serde.deserialize_task(task_bytes)
+        return self.executor.execute(plan)
+
     def collect(self, plan: P.PlanNode) -> RecordBatch:
         """Run and concat this rank's result (driver-side rows analogue)."""
         return self.executor.collect(plan)
