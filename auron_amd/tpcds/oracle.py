@@ -218,3 +218,303 @@ def q68(root, sf):
 
 ORACLES = {"q1": q1, "q3": q3, "q6": q6, "q7": q7, "q19": q19, "q42": q42,
            "q52": q52, "q55": q55, "q68": q68, "q96": q96}
+
+
+# ------------------------------------------------- batch 2: q25..q88
+def _q34_q73_oracle(root, sf, dom_pred, hd_pred, counties, cnt_lo, cnt_hi):
+    ss = _read(root, sf, "store_sales", ["ss_sold_date_sk", "ss_store_sk", "ss_hdemo_sk",
+                                         "ss_ticket_number", "ss_customer_sk"])
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_dom", "d_year"])
+    dd = dd[dom_pred(dd) & dd.d_year.isin([1999, 2000, 2001])]
+    st = _read(root, sf, "store", ["s_store_sk", "s_county"])
+    st = st[st.s_county.isin(counties)]
+    hd = _read(root, sf, "household_demographics")
+    hd = hd[hd_pred(hd)]
+    j = _merge(ss, dd, "ss_sold_date_sk", "d_date_sk")
+    j = _merge(j, st, "ss_store_sk", "s_store_sk")
+    j = _merge(j, hd, "ss_hdemo_sk", "hd_demo_sk")
+    g = j.groupby(["ss_ticket_number", "ss_customer_sk"], dropna=False).size().reset_index(name="cnt")
+    g = g[(g.cnt >= cnt_lo) & (g.cnt <= cnt_hi)]
+    cust = _read(root, sf, "customer", ["c_customer_sk", "c_last_name", "c_first_name",
+                                        "c_salutation", "c_preferred_cust_flag"])
+    j2 = _merge(g, cust, "ss_customer_sk", "c_customer_sk")
+    return j2[["c_last_name", "c_first_name", "c_salutation",
+               "c_preferred_cust_flag", "ss_ticket_number", "cnt"]].reset_index(drop=True)
+
+
+def q34(root, sf):
+    return _q34_q73_oracle(
+        root, sf,
+        lambda d: ((d.d_dom >= 1) & (d.d_dom <= 3)) | ((d.d_dom >= 25) & (d.d_dom <= 28)),
+        lambda h: h.hd_buy_potential.isin([">10000", "unknown"]) & (h.hd_vehicle_count > 0)
+        & ((h.hd_dep_count / h.hd_vehicle_count) > 1.2),
+        ["Williamson County"], 15, 20)
+
+
+def q73(root, sf):
+    return _q34_q73_oracle(
+        root, sf,
+        lambda d: (d.d_dom >= 1) & (d.d_dom <= 2),
+        lambda h: h.hd_buy_potential.isin([">10000", "unknown"]) & (h.hd_vehicle_count > 0)
+        & ((h.hd_dep_count / h.hd_vehicle_count) > 1.0),
+        ["Williamson County", "Franklin Parish", "Bronx County", "Orange County"], 1, 5)
+
+
+def q43(root, sf):
+    ss = _read(root, sf, "store_sales", ["ss_sold_date_sk", "ss_store_sk", "ss_sales_price"])
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year", "d_day_name"])
+    dd = dd[dd.d_year == 2000]
+    st = _read(root, sf, "store", ["s_store_sk", "s_store_name", "s_store_id", "s_gmt_offset"])
+    st = st[st.s_gmt_offset == -5.0]
+    j = _merge(ss, dd, "ss_sold_date_sk", "d_date_sk")
+    j = _merge(j, st, "ss_store_sk", "s_store_sk")
+    days = [("sun", "Sunday"), ("mon", "Monday"), ("tue", "Tuesday"),
+            ("wed", "Wednesday"), ("thu", "Thursday"), ("fri", "Friday"), ("sat", "Saturday")]
+    for tag, day in days:
+        j[f"{tag}_sales"] = j.ss_sales_price.where(j.d_day_name == day)
+    g = j.groupby(["s_store_name", "s_store_id"], dropna=False).agg(
+        **{f"{t}_sales": (f"{t}_sales", lambda x: x.sum(min_count=1)) for t, _ in days}).reset_index()
+    g = g.sort_values(["s_store_name", "s_store_id"] + [f"{t}_sales" for t, _ in days]).head(100)
+    return g.reset_index(drop=True)
+
+
+def q46(root, sf):
+    ss = _read(root, sf, "store_sales", ["ss_sold_date_sk", "ss_store_sk", "ss_hdemo_sk",
+                                         "ss_addr_sk", "ss_customer_sk", "ss_ticket_number",
+                                         "ss_coupon_amt", "ss_net_profit"])
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_dow", "d_year"])
+    dd = dd[dd.d_dow.isin([6, 0]) & dd.d_year.isin([1999, 2000, 2001])]
+    st = _read(root, sf, "store", ["s_store_sk", "s_city"])
+    st = st[st.s_city.isin(["Fairview", "Midway"])]
+    hd = _read(root, sf, "household_demographics")
+    hd = hd[(hd.hd_dep_count == 4) | (hd.hd_vehicle_count == 3)]
+    ca = _read(root, sf, "customer_address", ["ca_address_sk", "ca_city"])
+    j = _merge(ss, dd, "ss_sold_date_sk", "d_date_sk")
+    j = _merge(j, st, "ss_store_sk", "s_store_sk")
+    j = _merge(j, hd, "ss_hdemo_sk", "hd_demo_sk")
+    j = _merge(j, ca, "ss_addr_sk", "ca_address_sk")
+    g = j.groupby(["ss_ticket_number", "ss_customer_sk", "ss_addr_sk", "ca_city"],
+                  dropna=False).agg(amt=("ss_coupon_amt", lambda x: x.sum(min_count=1)),
+                                    profit=("ss_net_profit", lambda x: x.sum(min_count=1))).reset_index()
+    g = g.rename(columns={"ca_city": "bought_city"})
+    cust = _read(root, sf, "customer", ["c_customer_sk", "c_current_addr_sk",
+                                        "c_first_name", "c_last_name"])
+    j2 = _merge(g, cust, "ss_customer_sk", "c_customer_sk")
+    ca2 = ca.rename(columns={"ca_address_sk": "cur_addr_sk"})
+    j3 = _merge(j2, ca2, "c_current_addr_sk", "cur_addr_sk")
+    f = j3[(j3.ca_city != j3.bought_city) & j3.ca_city.notna() & j3.bought_city.notna()]
+    out = f[["c_last_name", "c_first_name", "ca_city", "bought_city",
+             "ss_ticket_number", "amt", "profit"]]
+    out = out.sort_values(["c_last_name", "c_first_name", "ca_city", "bought_city",
+                           "ss_ticket_number"], na_position="first").head(100)
+    return out.reset_index(drop=True)
+
+
+def q48(root, sf):
+    import pandas as pd
+
+    ss = _read(root, sf, "store_sales", ["ss_store_sk", "ss_sold_date_sk", "ss_cdemo_sk",
+                                         "ss_addr_sk", "ss_quantity", "ss_sales_price",
+                                         "ss_net_profit"])
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year"])
+    dd = dd[dd.d_year == 2001]
+    cd = _read(root, sf, "customer_demographics",
+               ["cd_demo_sk", "cd_marital_status", "cd_education_status"])
+    ca = _read(root, sf, "customer_address", ["ca_address_sk", "ca_country", "ca_state"])
+    ca = ca[ca.ca_country == "United States"]
+    st = _read(root, sf, "store", ["s_store_sk"])
+    j = _merge(ss, dd, "ss_sold_date_sk", "d_date_sk")
+    j = _merge(j, st, "ss_store_sk", "s_store_sk")
+    j = _merge(j, cd, "ss_cdemo_sk", "cd_demo_sk")
+    j = _merge(j, ca, "ss_addr_sk", "ca_address_sk")
+    c1 = (((j.cd_marital_status == "M") & (j.cd_education_status == "4 yr Degree")
+           & j.ss_sales_price.between(100.0, 150.0))
+          | ((j.cd_marital_status == "D") & (j.cd_education_status == "2 yr Degree")
+             & j.ss_sales_price.between(50.0, 100.0))
+          | ((j.cd_marital_status == "S") & (j.cd_education_status == "College")
+             & j.ss_sales_price.between(150.0, 200.0)))
+    c2 = ((j.ca_state.isin(["CO", "OH", "TX"]) & j.ss_net_profit.between(0, 2000))
+          | (j.ca_state.isin(["OR", "MN", "KY"]) & j.ss_net_profit.between(150, 3000))
+          | (j.ca_state.isin(["VA", "CA", "MS"]) & j.ss_net_profit.between(50, 25000)))
+    f = j[c1.fillna(False) & c2.fillna(False)]
+    v = f.ss_quantity.sum(min_count=1)
+    return pd.DataFrame({"s": [None if pd.isna(v) else int(v)]})
+
+
+def q65(root, sf):
+    ss = _read(root, sf, "store_sales", ["ss_sold_date_sk", "ss_store_sk", "ss_item_sk",
+                                         "ss_sales_price"])
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_month_seq"])
+    dd = dd[dd.d_month_seq.between(1176, 1187)]
+    j = _merge(ss, dd, "ss_sold_date_sk", "d_date_sk")
+    sc = j.groupby(["ss_store_sk", "ss_item_sk"], dropna=False) \
+          .ss_sales_price.sum(min_count=1).reset_index()
+    sc.columns = ["ss_store_sk", "ss_item_sk", "revenue"]
+    sb = sc.groupby("ss_store_sk", dropna=False).revenue.mean().reset_index()
+    sb.columns = ["sb_store_sk", "ave"]
+    j2 = _merge(sc, sb, "ss_store_sk", "sb_store_sk")
+    f = j2[j2.revenue <= 0.1 * j2.ave]
+    st = _read(root, sf, "store", ["s_store_sk", "s_store_name"])
+    it = _read(root, sf, "item", ["i_item_sk", "i_item_desc", "i_current_price", "i_brand"])
+    j3 = _merge(f, st, "ss_store_sk", "s_store_sk")
+    j4 = _merge(j3, it, "ss_item_sk", "i_item_sk")
+    out = j4[["s_store_name", "i_item_desc", "revenue", "i_current_price", "i_brand"]]
+    out = out.sort_values(["s_store_name", "i_item_desc"], na_position="first").head(100)
+    return out.reset_index(drop=True)
+
+
+def q79(root, sf):
+    ss = _read(root, sf, "store_sales", ["ss_sold_date_sk", "ss_store_sk", "ss_hdemo_sk",
+                                         "ss_addr_sk", "ss_customer_sk", "ss_ticket_number",
+                                         "ss_coupon_amt", "ss_net_profit"])
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_dow", "d_year"])
+    dd = dd[(dd.d_dow == 1) & dd.d_year.isin([1999, 2000, 2001])]
+    st = _read(root, sf, "store", ["s_store_sk", "s_number_employees", "s_city"])
+    st = st[st.s_number_employees.between(200, 295)]
+    hd = _read(root, sf, "household_demographics")
+    hd = hd[(hd.hd_dep_count == 6) | (hd.hd_vehicle_count > 2)]
+    j = _merge(ss, dd, "ss_sold_date_sk", "d_date_sk")
+    j = _merge(j, st, "ss_store_sk", "s_store_sk")
+    j = _merge(j, hd, "ss_hdemo_sk", "hd_demo_sk")
+    g = j.groupby(["ss_ticket_number", "ss_customer_sk", "ss_addr_sk", "s_city"],
+                  dropna=False).agg(amt=("ss_coupon_amt", lambda x: x.sum(min_count=1)),
+                                    profit=("ss_net_profit", lambda x: x.sum(min_count=1))).reset_index()
+    cust = _read(root, sf, "customer", ["c_customer_sk", "c_last_name", "c_first_name"])
+    j2 = _merge(g, cust, "ss_customer_sk", "c_customer_sk")
+    j2["s_city30"] = j2.s_city.str[:30]
+    out = j2[["c_last_name", "c_first_name", "s_city30", "ss_ticket_number", "amt", "profit"]]
+    out = out.sort_values(["c_last_name", "c_first_name", "s_city30",
+                           "ss_ticket_number", "profit"], na_position="first").head(100)
+    return out.reset_index(drop=True)
+
+
+def _q25_q29_oracle(root, sf, d1p, d2p, d3p, ss_m, sr_m, cs_m, n1, n2, n3):
+    ss = _read(root, sf, "store_sales", ["ss_sold_date_sk", "ss_item_sk", "ss_store_sk",
+                                         "ss_customer_sk", "ss_ticket_number", ss_m])
+    sr = _read(root, sf, "store_returns", ["sr_returned_date_sk", "sr_item_sk",
+                                           "sr_customer_sk", "sr_ticket_number", sr_m])
+    cs = _read(root, sf, "catalog_sales", ["cs_sold_date_sk", "cs_bill_customer_sk",
+                                           "cs_item_sk", cs_m])
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_moy", "d_year"])
+    d1 = dd[d1p(dd)]
+    d2 = dd[d2p(dd)]
+    d3 = dd[d3p(dd)]
+    j_ss = _merge(ss, d1[["d_date_sk"]], "ss_sold_date_sk", "d_date_sk")
+    j_sr = _merge(sr, d2[["d_date_sk"]].rename(columns={"d_date_sk": "d2_sk"}), "sr_returned_date_sk", "d2_sk")
+    j_cs = _merge(cs, d3[["d_date_sk"]].rename(columns={"d_date_sk": "d3_sk"}), "cs_sold_date_sk", "d3_sk")
+    j1 = j_ss.dropna(subset=["ss_customer_sk", "ss_item_sk", "ss_ticket_number"]).merge(
+        j_sr.dropna(subset=["sr_customer_sk", "sr_item_sk", "sr_ticket_number"]),
+        left_on=["ss_customer_sk", "ss_item_sk", "ss_ticket_number"],
+        right_on=["sr_customer_sk", "sr_item_sk", "sr_ticket_number"])
+    j2 = j1.merge(j_cs.dropna(subset=["cs_bill_customer_sk", "cs_item_sk"]),
+                  left_on=["sr_customer_sk", "sr_item_sk"],
+                  right_on=["cs_bill_customer_sk", "cs_item_sk"])
+    st = _read(root, sf, "store", ["s_store_sk", "s_store_id", "s_store_name"])
+    it = _read(root, sf, "item", ["i_item_sk", "i_item_id", "i_item_desc"])
+    j3 = _merge(j2, st, "ss_store_sk", "s_store_sk")
+    j4 = _merge(j3, it, "ss_item_sk", "i_item_sk")
+    g = j4.groupby(["i_item_id", "i_item_desc", "s_store_id", "s_store_name"], dropna=False).agg(
+        **{n1: (ss_m, lambda x: x.sum(min_count=1)),
+           n2: (sr_m, lambda x: x.sum(min_count=1)),
+           n3: (cs_m, lambda x: x.sum(min_count=1))}).reset_index()
+    g = g.sort_values(["i_item_id", "i_item_desc", "s_store_id", "s_store_name"]).head(100)
+    return g.reset_index(drop=True)
+
+
+def q25(root, sf):
+    return _q25_q29_oracle(
+        root, sf,
+        lambda d: (d.d_moy == 4) & (d.d_year == 2001),
+        lambda d: d.d_moy.between(4, 10) & (d.d_year == 2001),
+        lambda d: d.d_moy.between(4, 10) & (d.d_year == 2001),
+        "ss_net_profit", "sr_net_loss", "cs_net_profit",
+        "store_sales_profit", "store_returns_loss", "catalog_sales_profit")
+
+
+def q29(root, sf):
+    return _q25_q29_oracle(
+        root, sf,
+        lambda d: (d.d_moy == 9) & (d.d_year == 1999),
+        lambda d: d.d_moy.between(9, 12) & (d.d_year == 1999),
+        lambda d: d.d_year.isin([1999, 2000, 2001]),
+        "ss_quantity", "sr_return_quantity", "cs_quantity",
+        "store_sales_quantity", "store_returns_quantity", "catalog_sales_quantity")
+
+
+def q72(root, sf):
+    import pandas as pd
+
+    cs = _read(root, sf, "catalog_sales", ["cs_item_sk", "cs_order_number", "cs_bill_cdemo_sk",
+                                           "cs_bill_hdemo_sk", "cs_sold_date_sk",
+                                           "cs_ship_date_sk", "cs_promo_sk", "cs_quantity"])
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_week_seq", "d_date", "d_year"])
+    dd["d_date_i"] = pd.to_datetime(dd.d_date).map(lambda x: x.toordinal() - 719163)
+    d1 = dd[dd.d_year == 1999]
+    cd = _read(root, sf, "customer_demographics", ["cd_demo_sk", "cd_marital_status"])
+    cd = cd[cd.cd_marital_status == "D"]
+    hd = _read(root, sf, "household_demographics", ["hd_demo_sk", "hd_buy_potential"])
+    hd = hd[hd.hd_buy_potential == ">10000"]
+    j = _merge(cs, d1[["d_date_sk", "d_week_seq", "d_date_i"]].rename(
+        columns={"d_date_sk": "d1_sk", "d_week_seq": "d1_week_seq", "d_date_i": "d1_date"}),
+        "cs_sold_date_sk", "d1_sk")
+    j = _merge(j, cd, "cs_bill_cdemo_sk", "cd_demo_sk")
+    j = _merge(j, hd, "cs_bill_hdemo_sk", "hd_demo_sk")
+    d3 = dd[["d_date_sk", "d_date_i"]].rename(columns={"d_date_sk": "d3_sk", "d_date_i": "d3_date"})
+    j = _merge(j, d3, "cs_ship_date_sk", "d3_sk")
+    j = j[j.d3_date > j.d1_date + 5]
+    inv = _read(root, sf, "inventory")
+    d2 = dd[["d_date_sk", "d_week_seq"]].rename(columns={"d_date_sk": "d2_sk", "d_week_seq": "d2_week_seq"})
+    inv_j = _merge(inv, d2, "inv_date_sk", "d2_sk")
+    big = j.dropna(subset=["cs_item_sk", "d1_week_seq"]).merge(
+        inv_j.dropna(subset=["inv_item_sk", "d2_week_seq"]),
+        left_on=["cs_item_sk", "d1_week_seq"], right_on=["inv_item_sk", "d2_week_seq"])
+    big = big[big.inv_quantity_on_hand < big.cs_quantity]
+    wh = _read(root, sf, "warehouse", ["w_warehouse_sk", "w_warehouse_name"])
+    it = _read(root, sf, "item", ["i_item_sk", "i_item_desc"])
+    big = _merge(big, wh, "inv_warehouse_sk", "w_warehouse_sk")
+    big = _merge(big, it, "cs_item_sk", "i_item_sk")
+    pr = _read(root, sf, "promotion", ["p_promo_sk"])
+    big = big.merge(pr.dropna(), left_on="cs_promo_sk", right_on="p_promo_sk", how="left")
+    cr = _read(root, sf, "catalog_returns", ["cr_item_sk", "cr_order_number"])
+    big = big.merge(cr.dropna(subset=["cr_item_sk", "cr_order_number"]),
+                    left_on=["cs_item_sk", "cs_order_number"],
+                    right_on=["cr_item_sk", "cr_order_number"], how="left")
+    g = big.groupby(["i_item_desc", "w_warehouse_name", "d1_week_seq"], dropna=False).agg(
+        no_promo=("p_promo_sk", "size"), promo=("p_promo_sk", "size"),
+        total_cnt=("p_promo_sk", "size")).reset_index()
+    g = g.sort_values(["total_cnt", "i_item_desc", "w_warehouse_name", "d1_week_seq"],
+                      ascending=[False, True, True, True]).head(100)
+    return g.reset_index(drop=True)
+
+
+def q88(root, sf):
+    import pandas as pd
+
+    ss = _read(root, sf, "store_sales", ["ss_sold_time_sk", "ss_hdemo_sk", "ss_store_sk"])
+    hd = _read(root, sf, "household_demographics")
+    hd = hd[((hd.hd_dep_count == 4) & (hd.hd_vehicle_count <= 6))
+            | ((hd.hd_dep_count == 2) & (hd.hd_vehicle_count <= 4))
+            | ((hd.hd_dep_count == 0) & (hd.hd_vehicle_count <= 2))]
+    td = _read(root, sf, "time_dim")
+    st = _read(root, sf, "store", ["s_store_sk", "s_store_name"])
+    st = st[st.s_store_name == "ese"]
+    base = _merge(ss, hd, "ss_hdemo_sk", "hd_demo_sk")
+    base = _merge(base, st, "ss_store_sk", "s_store_sk")
+    slots = [(8, 30, None), (9, None, 30), (9, 30, None), (10, None, 30),
+             (10, 30, None), (11, None, 30), (11, 30, None), (12, None, 30)]
+    names = ["h8_30_to_9", "h9_to_9_30", "h9_30_to_10", "h10_to_10_30",
+             "h10_30_to_11", "h11_to_11_30", "h11_30_to_12", "h12_to_12_30"]
+    out = {}
+    for (h, lo, hi), nm in zip(slots, names):
+        t = td[td.t_hour == h]
+        if lo is not None:
+            t = t[t.t_minute >= lo]
+        if hi is not None:
+            t = t[t.t_minute < hi]
+        out[nm] = [len(_merge(base, t, "ss_sold_time_sk", "t_time_sk"))]
+    return pd.DataFrame(out)
+
+
+ORACLES.update({"q25": q25, "q29": q29, "q34": q34, "q43": q43, "q46": q46,
+                "q48": q48, "q65": q65, "q72": q72, "q73": q73, "q79": q79,
+                "q88": q88})

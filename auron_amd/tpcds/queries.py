@@ -287,3 +287,373 @@ QUERIES = {
     "q1": q1, "q3": q3, "q6": q6, "q7": q7, "q19": q19, "q42": q42,
     "q52": q52, "q55": q55, "q68": q68, "q96": q96,
 }
+
+
+# ------------------------------------------------- batch 2: q25..q88
+def _rename_scan(cat, table, cols, prefix):
+    sc = cat.scan(table, cols)
+    return P.Project(sc, [_a(col(c), f"{prefix}{c}") for c in cols])
+
+
+def shj(left, right, lkeys, rkeys, how="inner"):
+    """Shuffled hash join: co-partition both sides by the join keys over
+    RCCL all-to-all, then join locally (SHJ leg of broadcast_join_exec)."""
+    lex = P.Exchange(left, "hash", [col(k) for k in lkeys])
+    rex = P.Exchange(right, "hash", [col(k) for k in rkeys])
+    return P.HashJoin(lex, rex, [col(k) for k in lkeys], [col(k) for k in rkeys],
+                      how=how, build_side="right", broadcast=False)
+
+
+def _q34_q73(cat, s, dom_lo, dom_hi, dom_or=None, ratio=1.2, counties=None,
+             cnt_lo=15, cnt_hi=20, order_desc_cnt=False):
+    ss = cat.scan("store_sales", ["ss_sold_date_sk", "ss_store_sk", "ss_hdemo_sk",
+                                  "ss_ticket_number", "ss_customer_sk"])
+    dom_f = (col("d_dom") >= dom_lo) & (col("d_dom") <= dom_hi)
+    if dom_or:
+        dom_f = dom_f | ((col("d_dom") >= dom_or[0]) & (col("d_dom") <= dom_or[1]))
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_dom", "d_year"]),
+                  dom_f & col("d_year").isin([1999, 2000, 2001]))
+    st = P.Filter(cat.scan("store", ["s_store_sk", "s_county"]),
+                  col("s_county").isin(counties or ["Williamson County"]))
+    from ..exprs import CaseWhen, Literal
+
+    hd = P.Filter(
+        cat.scan("household_demographics",
+                 ["hd_demo_sk", "hd_buy_potential", "hd_dep_count", "hd_vehicle_count"]),
+        (col("hd_buy_potential").isin([">10000", "unknown"]))
+        & (col("hd_vehicle_count") > 0)
+        & (CaseWhen([(col("hd_vehicle_count") > 0,
+                      col("hd_dep_count").cast(dtypes.float64) / col("hd_vehicle_count").cast(dtypes.float64))],
+                    Literal(None, dtypes.float64)) > ratio))
+    j = bhj(ss, dd, ["ss_sold_date_sk"], ["d_date_sk"])
+    j = bhj(j, st, ["ss_store_sk"], ["s_store_sk"])
+    j = bhj(j, hd, ["ss_hdemo_sk"], ["hd_demo_sk"])
+    a = agg2(j, ["ss_ticket_number", "ss_customer_sk"],
+             [AggFunc("count_star", None, name="cnt")])
+    f = P.Filter(a, (col("cnt") >= cnt_lo) & (col("cnt") <= cnt_hi))
+    cust = cat.scan("customer", ["c_customer_sk", "c_last_name", "c_first_name",
+                                 "c_salutation", "c_preferred_cust_flag"])
+    j2 = bhj(f, cust, ["ss_customer_sk"], ["c_customer_sk"])
+    proj = P.Project(j2, [_a(col("c_last_name"), "c_last_name"),
+                          _a(col("c_first_name"), "c_first_name"),
+                          _a(col("c_salutation"), "c_salutation"),
+                          _a(col("c_preferred_cust_flag"), "c_preferred_cust_flag"),
+                          _a(col("ss_ticket_number"), "ss_ticket_number"),
+                          _a(col("cnt"), "cnt")])
+    if order_desc_cnt:
+        return topk(proj, [(col("cnt"), False)], 100000)
+    return topk(proj, [(col("c_last_name"), True), (col("c_first_name"), True),
+                       (col("c_salutation"), True), (col("c_preferred_cust_flag"), True),
+                       (col("ss_ticket_number"), False)], 100000)
+
+
+def q34(cat, s):
+    return _q34_q73(cat, s, 1, 3, dom_or=(25, 28), ratio=1.2,
+                    counties=["Williamson County"], cnt_lo=15, cnt_hi=20)
+
+
+def q73(cat, s):
+    return _q34_q73(cat, s, 1, 2, ratio=1.0,
+                    counties=["Williamson County", "Franklin Parish",
+                              "Bronx County", "Orange County"],
+                    cnt_lo=1, cnt_hi=5, order_desc_cnt=True)
+
+
+def q43(cat, s):
+    from ..exprs import CaseWhen, Literal
+
+    ss = cat.scan("store_sales", ["ss_sold_date_sk", "ss_store_sk", "ss_sales_price"])
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_year", "d_day_name"]),
+                  col("d_year") == 2000)
+    st = P.Filter(cat.scan("store", ["s_store_sk", "s_store_name", "s_store_id", "s_gmt_offset"]),
+                  col("s_gmt_offset") == -5.0)
+    j = bhj(ss, dd, ["ss_sold_date_sk"], ["d_date_sk"])
+    j = bhj(j, st, ["ss_store_sk"], ["s_store_sk"])
+    days = [("sun", "Sunday"), ("mon", "Monday"), ("tue", "Tuesday"),
+            ("wed", "Wednesday"), ("thu", "Thursday"), ("fri", "Friday"),
+            ("sat", "Saturday")]
+    aggs = [AggFunc("sum", CaseWhen([(col("d_day_name") == lit(day), col("ss_sales_price"))],
+                                    Literal(None, dtypes.float64)), name=f"{tag}_sales")
+            for tag, day in days]
+    a = agg2(j, ["s_store_name", "s_store_id"], aggs)
+    keys = [(col("s_store_name"), True), (col("s_store_id"), True)] + \
+        [(col(f"{t}_sales"), True) for t, _ in days]
+    return topk(a, keys, 100)
+
+
+def q46(cat, s):
+    ss = cat.scan("store_sales", ["ss_sold_date_sk", "ss_store_sk", "ss_hdemo_sk",
+                                  "ss_addr_sk", "ss_customer_sk", "ss_ticket_number",
+                                  "ss_coupon_amt", "ss_net_profit"])
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_dow", "d_year"]),
+                  col("d_dow").isin([6, 0]) & col("d_year").isin([1999, 2000, 2001]))
+    st = P.Filter(cat.scan("store", ["s_store_sk", "s_city"]),
+                  col("s_city").isin(["Fairview", "Midway"]))
+    hd = P.Filter(cat.scan("household_demographics",
+                           ["hd_demo_sk", "hd_dep_count", "hd_vehicle_count"]),
+                  (col("hd_dep_count") == 4) | (col("hd_vehicle_count") == 3))
+    ca = cat.scan("customer_address", ["ca_address_sk", "ca_city"])
+    j = bhj(ss, dd, ["ss_sold_date_sk"], ["d_date_sk"])
+    j = bhj(j, st, ["ss_store_sk"], ["s_store_sk"])
+    j = bhj(j, hd, ["ss_hdemo_sk"], ["hd_demo_sk"])
+    j = bhj(j, ca, ["ss_addr_sk"], ["ca_address_sk"])
+    a = agg2(j, ["ss_ticket_number", "ss_customer_sk", "ss_addr_sk", "bought_city"],
+             [AggFunc("sum", col("ss_coupon_amt"), name="amt"),
+              AggFunc("sum", col("ss_net_profit"), name="profit")],
+             key_exprs=[col("ss_ticket_number"), col("ss_customer_sk"),
+                        col("ss_addr_sk"), col("ca_city")])
+    cust = cat.scan("customer", ["c_customer_sk", "c_current_addr_sk",
+                                 "c_first_name", "c_last_name"])
+    j2 = bhj(a, cust, ["ss_customer_sk"], ["c_customer_sk"])
+    ca2 = P.Project(cat.scan("customer_address", ["ca_address_sk", "ca_city"]),
+                    [_a(col("ca_address_sk"), "cur_addr_sk"), _a(col("ca_city"), "ca_city")])
+    j3 = bhj(j2, ca2, ["c_current_addr_sk"], ["cur_addr_sk"])
+    f = P.Filter(j3, ~(col("ca_city") == col("bought_city")))
+    proj = P.Project(f, [_a(col("c_last_name"), "c_last_name"),
+                         _a(col("c_first_name"), "c_first_name"),
+                         _a(col("ca_city"), "ca_city"),
+                         _a(col("bought_city"), "bought_city"),
+                         _a(col("ss_ticket_number"), "ss_ticket_number"),
+                         _a(col("amt"), "amt"), _a(col("profit"), "profit")])
+    return topk(proj, [(col("c_last_name"), True), (col("c_first_name"), True),
+                       (col("ca_city"), True), (col("bought_city"), True),
+                       (col("ss_ticket_number"), True)], 100)
+
+
+def q48(cat, s):
+    ss = cat.scan("store_sales", ["ss_store_sk", "ss_sold_date_sk", "ss_cdemo_sk",
+                                  "ss_addr_sk", "ss_quantity", "ss_sales_price",
+                                  "ss_net_profit"])
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_year"]), col("d_year") == 2001)
+    st = cat.scan("store", ["s_store_sk"])
+    cd = cat.scan("customer_demographics",
+                  ["cd_demo_sk", "cd_marital_status", "cd_education_status"])
+    ca = P.Filter(cat.scan("customer_address", ["ca_address_sk", "ca_country", "ca_state"]),
+                  col("ca_country") == lit("United States"))
+    j = bhj(ss, dd, ["ss_sold_date_sk"], ["d_date_sk"])
+    j = bhj(j, st, ["ss_store_sk"], ["s_store_sk"])
+    j = bhj(j, cd, ["ss_cdemo_sk"], ["cd_demo_sk"])
+    j = bhj(j, ca, ["ss_addr_sk"], ["ca_address_sk"])
+    sp = col("ss_sales_price")
+    np_ = col("ss_net_profit")
+    cond1 = (((col("cd_marital_status") == lit("M")) & (col("cd_education_status") == lit("4 yr Degree")) & sp.between(100.0, 150.0))
+             | ((col("cd_marital_status") == lit("D")) & (col("cd_education_status") == lit("2 yr Degree")) & sp.between(50.0, 100.0))
+             | ((col("cd_marital_status") == lit("S")) & (col("cd_education_status") == lit("College")) & sp.between(150.0, 200.0)))
+    cond2 = ((col("ca_state").isin(["CO", "OH", "TX"]) & np_.between(0.0, 2000.0))
+             | (col("ca_state").isin(["OR", "MN", "KY"]) & np_.between(150.0, 3000.0))
+             | (col("ca_state").isin(["VA", "CA", "MS"]) & np_.between(50.0, 25000.0)))
+    f = P.Filter(j, cond1 & cond2)
+    partial = P.HashAgg(f, [], [AggFunc("sum", col("ss_quantity"), name="s")], mode="partial")
+    return P.HashAgg(P.Exchange(partial, "single"), [],
+                     [AggFunc("sum", col("ss_quantity"), name="s")], mode="final")
+
+
+def q65(cat, s):
+    ss = cat.scan("store_sales", ["ss_sold_date_sk", "ss_store_sk", "ss_item_sk",
+                                  "ss_sales_price"])
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_month_seq"]),
+                  col("d_month_seq").between(1176, 1187))
+    j = bhj(ss, dd, ["ss_sold_date_sk"], ["d_date_sk"])
+    sc_plan = agg2(j, ["ss_store_sk", "ss_item_sk"],
+                   [AggFunc("sum", col("ss_sales_price"), name="revenue")])
+    sc_batches = s.execute(sc_plan)
+    sc1 = P.MemoryScan(sc_batches)
+    sb_partial = P.HashAgg(P.MemoryScan(sc_batches),
+                           [_a(col("ss_store_sk"), "sb_store_sk")],
+                           [AggFunc("avg", col("revenue"), name="ave")], mode="partial")
+    sb = P.Broadcast(P.HashAgg(P.Exchange(sb_partial, "hash", [col("sb_store_sk")]),
+                               [_a(col("sb_store_sk"), "sb_store_sk")],
+                               [AggFunc("avg", col("revenue"), name="ave")], mode="final"))
+    j2 = P.HashJoin(sc1, sb, [col("ss_store_sk")], [col("sb_store_sk")],
+                    how="inner", build_side="right")
+    f = P.Filter(j2, col("revenue") <= col("ave") * lit(0.1))
+    st = cat.scan("store", ["s_store_sk", "s_store_name"])
+    it = cat.scan("item", ["i_item_sk", "i_item_desc", "i_current_price", "i_brand"])
+    j3 = bhj(f, st, ["ss_store_sk"], ["s_store_sk"])
+    j4 = bhj(j3, it, ["ss_item_sk"], ["i_item_sk"])
+    proj = P.Project(j4, [_a(col("s_store_name"), "s_store_name"),
+                          _a(col("i_item_desc"), "i_item_desc"),
+                          _a(col("revenue"), "revenue"),
+                          _a(col("i_current_price"), "i_current_price"),
+                          _a(col("i_brand"), "i_brand")])
+    return topk(proj, [(col("s_store_name"), True), (col("i_item_desc"), True)], 100)
+
+
+def q79(cat, s):
+    from ..exprs import Substr
+
+    ss = cat.scan("store_sales", ["ss_sold_date_sk", "ss_store_sk", "ss_hdemo_sk",
+                                  "ss_addr_sk", "ss_customer_sk", "ss_ticket_number",
+                                  "ss_coupon_amt", "ss_net_profit"])
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_dow", "d_year"]),
+                  (col("d_dow") == 1) & col("d_year").isin([1999, 2000, 2001]))
+    st = P.Filter(cat.scan("store", ["s_store_sk", "s_number_employees", "s_city"]),
+                  col("s_number_employees").between(200, 295))
+    hd = P.Filter(cat.scan("household_demographics",
+                           ["hd_demo_sk", "hd_dep_count", "hd_vehicle_count"]),
+                  (col("hd_dep_count") == 6) | (col("hd_vehicle_count") > 2))
+    j = bhj(ss, dd, ["ss_sold_date_sk"], ["d_date_sk"])
+    j = bhj(j, st, ["ss_store_sk"], ["s_store_sk"])
+    j = bhj(j, hd, ["ss_hdemo_sk"], ["hd_demo_sk"])
+    a = agg2(j, ["ss_ticket_number", "ss_customer_sk", "ss_addr_sk", "s_city"],
+             [AggFunc("sum", col("ss_coupon_amt"), name="amt"),
+              AggFunc("sum", col("ss_net_profit"), name="profit")])
+    cust = cat.scan("customer", ["c_customer_sk", "c_last_name", "c_first_name"])
+    j2 = bhj(a, cust, ["ss_customer_sk"], ["c_customer_sk"])
+    proj = P.Project(j2, [_a(col("c_last_name"), "c_last_name"),
+                          _a(col("c_first_name"), "c_first_name"),
+                          _a(Substr(col("s_city"), 1, 30), "s_city30"),
+                          _a(col("ss_ticket_number"), "ss_ticket_number"),
+                          _a(col("amt"), "amt"), _a(col("profit"), "profit")])
+    return topk(proj, [(col("c_last_name"), True), (col("c_first_name"), True),
+                       (col("s_city30"), True), (col("ss_ticket_number"), True),
+                       (col("profit"), True)], 100)
+
+
+def _q25_q29(cat, s, d1f, d2f, d3f, measures):
+    (ss_m, sr_m, cs_m), (n1, n2, n3) = measures
+    ss = cat.scan("store_sales", ["ss_sold_date_sk", "ss_item_sk", "ss_store_sk",
+                                  "ss_customer_sk", "ss_ticket_number", ss_m])
+    sr = cat.scan("store_returns", ["sr_returned_date_sk", "sr_item_sk",
+                                    "sr_customer_sk", "sr_ticket_number", sr_m])
+    cs = cat.scan("catalog_sales", ["cs_sold_date_sk", "cs_bill_customer_sk",
+                                    "cs_item_sk", cs_m])
+    dd_cols = ["d_date_sk", "d_moy", "d_year"]
+    d1 = P.Filter(cat.scan("date_dim", dd_cols), d1f)
+    d2 = P.Filter(cat.scan("date_dim", dd_cols), d2f)
+    d3 = P.Filter(cat.scan("date_dim", dd_cols), d3f)
+    j_ss = bhj(ss, P.Project(d1, [_a(col("d_date_sk"), "d1_sk")]), ["ss_sold_date_sk"], ["d1_sk"])
+    j_sr = bhj(sr, P.Project(d2, [_a(col("d_date_sk"), "d2_sk")]), ["sr_returned_date_sk"], ["d2_sk"])
+    j_cs = bhj(cs, P.Project(d3, [_a(col("d_date_sk"), "d3_sk")]), ["cs_sold_date_sk"], ["d3_sk"])
+    # fact-fact joins co-partitioned over RCCL
+    j1 = shj(j_ss, j_sr,
+             ["ss_customer_sk", "ss_item_sk", "ss_ticket_number"],
+             ["sr_customer_sk", "sr_item_sk", "sr_ticket_number"])
+    j2 = shj(j1, j_cs, ["sr_customer_sk", "sr_item_sk"],
+             ["cs_bill_customer_sk", "cs_item_sk"])
+    st = cat.scan("store", ["s_store_sk", "s_store_id", "s_store_name"])
+    it = cat.scan("item", ["i_item_sk", "i_item_id", "i_item_desc"])
+    j3 = bhj(j2, st, ["ss_store_sk"], ["s_store_sk"])
+    j4 = bhj(j3, it, ["ss_item_sk"], ["i_item_sk"])
+    a = agg2(j4, ["i_item_id", "i_item_desc", "s_store_id", "s_store_name"],
+             [AggFunc("sum", col(ss_m), name=n1),
+              AggFunc("sum", col(sr_m), name=n2),
+              AggFunc("sum", col(cs_m), name=n3)])
+    return topk(a, [(col("i_item_id"), True), (col("i_item_desc"), True),
+                    (col("s_store_id"), True), (col("s_store_name"), True)], 100)
+
+
+def q25(cat, s):
+    return _q25_q29(
+        cat, s,
+        (col("d_moy") == 4) & (col("d_year") == 2001),
+        col("d_moy").between(4, 10) & (col("d_year") == 2001),
+        col("d_moy").between(4, 10) & (col("d_year") == 2001),
+        (("ss_net_profit", "sr_net_loss", "cs_net_profit"),
+         ("store_sales_profit", "store_returns_loss", "catalog_sales_profit")))
+
+
+def q29(cat, s):
+    return _q25_q29(
+        cat, s,
+        (col("d_moy") == 9) & (col("d_year") == 1999),
+        col("d_moy").between(9, 12) & (col("d_year") == 1999),
+        col("d_year").isin([1999, 2000, 2001]),
+        (("ss_quantity", "sr_return_quantity", "cs_quantity"),
+         ("store_sales_quantity", "store_returns_quantity", "catalog_sales_quantity")))
+
+
+def q72(cat, s):
+    from ..exprs import CaseWhen, IsNull, Literal, Not
+
+    cs = cat.scan("catalog_sales", ["cs_item_sk", "cs_order_number", "cs_bill_cdemo_sk",
+                                    "cs_bill_hdemo_sk", "cs_sold_date_sk",
+                                    "cs_ship_date_sk", "cs_promo_sk", "cs_quantity"])
+    d1 = P.Project(P.Filter(cat.scan("date_dim", ["d_date_sk", "d_week_seq", "d_date", "d_year"]),
+                            col("d_year") == 1999),
+                   [_a(col("d_date_sk"), "d1_sk"), _a(col("d_week_seq"), "d1_week_seq"),
+                    _a(col("d_date"), "d1_date")])
+    cd = P.Filter(cat.scan("customer_demographics", ["cd_demo_sk", "cd_marital_status"]),
+                  col("cd_marital_status") == lit("D"))
+    hd = P.Filter(cat.scan("household_demographics", ["hd_demo_sk", "hd_buy_potential"]),
+                  col("hd_buy_potential") == lit(">10000"))
+    d3 = P.Project(cat.scan("date_dim", ["d_date_sk", "d_date"]),
+                   [_a(col("d_date_sk"), "d3_sk"), _a(col("d_date"), "d3_date")])
+    j = bhj(cs, d1, ["cs_sold_date_sk"], ["d1_sk"])
+    j = bhj(j, cd, ["cs_bill_cdemo_sk"], ["cd_demo_sk"])
+    j = bhj(j, hd, ["cs_bill_hdemo_sk"], ["hd_demo_sk"])
+    j = bhj(j, d3, ["cs_ship_date_sk"], ["d3_sk"])
+    j = P.Filter(j, col("d3_date").cast(dtypes.int64) > col("d1_date").cast(dtypes.int64) + lit(5))
+    inv = cat.scan("inventory", ["inv_item_sk", "inv_warehouse_sk", "inv_date_sk",
+                                 "inv_quantity_on_hand"])
+    d2 = P.Project(cat.scan("date_dim", ["d_date_sk", "d_week_seq"]),
+                   [_a(col("d_date_sk"), "d2_sk"), _a(col("d_week_seq"), "d2_week_seq")])
+    inv_j = bhj(inv, d2, ["inv_date_sk"], ["d2_sk"])
+    # the classic q72 explosion guard: week equality rides the join key
+    big = shj(j, inv_j, ["cs_item_sk", "d1_week_seq"], ["inv_item_sk", "d2_week_seq"])
+    big = P.Filter(big, col("inv_quantity_on_hand") < col("cs_quantity"))
+    wh = cat.scan("warehouse", ["w_warehouse_sk", "w_warehouse_name"])
+    it = cat.scan("item", ["i_item_sk", "i_item_desc"])
+    j2 = bhj(big, wh, ["inv_warehouse_sk"], ["w_warehouse_sk"])
+    j2 = bhj(j2, it, ["cs_item_sk"], ["i_item_sk"])
+    pr = cat.scan("promotion", ["p_promo_sk"])
+    j2 = P.HashJoin(j2, P.Broadcast(pr), [col("cs_promo_sk")], [col("p_promo_sk")],
+                    how="left", build_side="right")
+    cr = cat.scan("catalog_returns", ["cr_item_sk", "cr_order_number"])
+    j3 = shj(j2, cr, ["cs_item_sk", "cs_order_number"],
+             ["cr_item_sk", "cr_order_number"], how="left")
+    a = agg2(j3, ["i_item_desc", "w_warehouse_name", "d1_week_seq"],
+             [AggFunc("count", CaseWhen([(IsNull(col("p_promo_sk")), lit(1))], lit(0)), name="no_promo"),
+              AggFunc("count", CaseWhen([(Not(IsNull(col("p_promo_sk"))), lit(1))], lit(0)), name="promo"),
+              AggFunc("count_star", None, name="total_cnt")])
+    return topk(a, [(col("total_cnt"), False), (col("i_item_desc"), True),
+                    (col("w_warehouse_name"), True), (col("d1_week_seq"), True)], 100)
+
+
+def _q88_count(cat, s, hour, minute_lo, minute_hi):
+    ss = cat.scan("store_sales", ["ss_sold_time_sk", "ss_hdemo_sk", "ss_store_sk"])
+    tf = col("t_hour") == hour
+    if minute_lo is not None:
+        tf = tf & (col("t_minute") >= minute_lo)
+    if minute_hi is not None:
+        tf = tf & (col("t_minute") < minute_hi)
+    td = P.Filter(cat.scan("time_dim", ["t_time_sk", "t_hour", "t_minute"]), tf)
+    hd = P.Filter(cat.scan("household_demographics",
+                           ["hd_demo_sk", "hd_dep_count", "hd_vehicle_count"]),
+                  ((col("hd_dep_count") == 4) & (col("hd_vehicle_count") <= 6))
+                  | ((col("hd_dep_count") == 2) & (col("hd_vehicle_count") <= 4))
+                  | ((col("hd_dep_count") == 0) & (col("hd_vehicle_count") <= 2)))
+    st = P.Filter(cat.scan("store", ["s_store_sk", "s_store_name"]),
+                  col("s_store_name") == lit("ese"))
+    j = bhj(ss, td, ["ss_sold_time_sk"], ["t_time_sk"])
+    j = bhj(j, hd, ["ss_hdemo_sk"], ["hd_demo_sk"])
+    j = bhj(j, st, ["ss_store_sk"], ["s_store_sk"])
+    partial = P.HashAgg(j, [], [AggFunc("count_star", None, name="c")], mode="partial")
+    plan = P.HashAgg(P.Exchange(partial, "single"), [],
+                     [AggFunc("count_star", None, name="c")], mode="final")
+    return scalar(s, plan)
+
+
+def q88(cat, s):
+    slots = [(8, 30, None), (9, None, 30), (9, 30, None), (10, None, 30),
+             (10, 30, None), (11, None, 30), (11, 30, None), (12, None, 30)]
+    names = ["h8_30_to_9", "h9_to_9_30", "h9_30_to_10", "h10_to_10_30",
+             "h10_30_to_11", "h11_to_11_30", "h11_30_to_12", "h12_to_12_30"]
+    vals = [_q88_count(cat, s, *slot) for slot in slots]
+    # cross join of eight single-row aggregates -> one row (on rank 0)
+    n = 1 if s.rank == 0 else 0
+    data = {nm: [v] * n for nm, v in zip(names, vals)}
+    types = {nm: dtypes.int64 for nm in names}
+    from ..column import RecordBatch
+
+    if n == 0:
+        b = RecordBatch.from_pydict({nm: [0] for nm in names}, types).slice(0, 0)
+    else:
+        b = RecordBatch.from_pydict(data, types)
+    return P.MemoryScan([b])
+
+
+QUERIES.update({
+    "q25": q25, "q29": q29, "q34": q34, "q43": q43, "q46": q46, "q48": q48,
+    "q65": q65, "q72": q72, "q73": q73, "q79": q79, "q88": q88,
+})
