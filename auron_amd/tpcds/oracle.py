@@ -518,3 +518,319 @@ def q88(root, sf):
 ORACLES.update({"q25": q25, "q29": q29, "q34": q34, "q43": q43, "q46": q46,
                 "q48": q48, "q65": q65, "q72": q72, "q73": q73, "q79": q79,
                 "q88": q88})
+
+
+# ------------------------------- batch 3 oracles
+def _days(y, m, d):
+    import datetime
+
+    return (datetime.date(y, m, d) - datetime.date(1970, 1, 1)).days
+
+
+def _date_i(dd):
+    import pandas as pd
+
+    return pd.to_datetime(dd.d_date).map(lambda x: x.toordinal() - 719163)
+
+
+def _ratio_window_oracle(root, sf, fact, pre, measure):
+    ss = _read(root, sf, fact, [f"{pre}_sold_date_sk", f"{pre}_item_sk", measure])
+    it = _read(root, sf, "item", ["i_item_sk", "i_item_id", "i_item_desc",
+                                  "i_category", "i_class", "i_current_price"])
+    it = it[it.i_category.isin(["Sports", "Books", "Home"])]
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_date"])
+    lo = _days(1999, 2, 22)
+    di = _date_i(dd)
+    dd = dd[(di >= lo) & (di <= lo + 30)]
+    j = _merge(ss, it, f"{pre}_item_sk", "i_item_sk")
+    j = _merge(j, dd, f"{pre}_sold_date_sk", "d_date_sk")
+    g = j.groupby(["i_item_id", "i_item_desc", "i_category", "i_class",
+                   "i_current_price"], dropna=False)[measure].sum(min_count=1).reset_index()
+    g = g.rename(columns={measure: "itemrevenue"})
+    g["_clsrev"] = g.groupby("i_class", dropna=False).itemrevenue.transform("sum")
+    g["revenueratio"] = g.itemrevenue * 100.0 / g._clsrev
+    g = g.sort_values(["i_category", "i_class", "i_item_id", "i_item_desc",
+                       "revenueratio"], na_position="first").head(100)
+    return g[["i_item_desc", "i_category", "i_class", "i_current_price",
+              "itemrevenue", "revenueratio"]].reset_index(drop=True)
+
+
+def q12(root, sf):
+    return _ratio_window_oracle(root, sf, "web_sales", "ws", "ws_ext_sales_price")
+
+
+def q20(root, sf):
+    return _ratio_window_oracle(root, sf, "catalog_sales", "cs", "cs_ext_sales_price")
+
+
+def q98(root, sf):
+    return _ratio_window_oracle(root, sf, "store_sales", "ss", "ss_ext_sales_price")
+
+
+def q15(root, sf):
+    cs = _read(root, sf, "catalog_sales", ["cs_bill_customer_sk", "cs_sold_date_sk", "cs_sales_price"])
+    cust = _read(root, sf, "customer", ["c_customer_sk", "c_current_addr_sk"])
+    ca = _read(root, sf, "customer_address", ["ca_address_sk", "ca_zip", "ca_state"])
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_qoy", "d_year"])
+    dd = dd[(dd.d_qoy == 2) & (dd.d_year == 2001)]
+    j = _merge(cs, cust, "cs_bill_customer_sk", "c_customer_sk")
+    j = _merge(j, ca, "c_current_addr_sk", "ca_address_sk")
+    j = _merge(j, dd, "cs_sold_date_sk", "d_date_sk")
+    zips = ["85669", "86197", "88274", "83405", "86475", "85392", "85460", "80348", "81792"]
+    m = (j.ca_zip.str[:5].isin(zips) | j.ca_state.isin(["CA", "WA", "GA"])
+         | (j.cs_sales_price > 500.0))
+    f = j[m.fillna(False)]
+    g = f.groupby("ca_zip", dropna=False).cs_sales_price.sum(min_count=1).reset_index()
+    g.columns = ["ca_zip", "s"]
+    return g.sort_values("ca_zip", na_position="first").head(100).reset_index(drop=True)
+
+
+def q22(root, sf):
+    import pandas as pd
+
+    inv = _read(root, sf, "inventory")
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_month_seq"])
+    dd = dd[dd.d_month_seq.between(1200, 1211)]
+    it = _read(root, sf, "item", ["i_item_sk", "i_product_name", "i_brand", "i_class", "i_category"])
+    wh = _read(root, sf, "warehouse", ["w_warehouse_sk"])
+    j = _merge(inv, dd, "inv_date_sk", "d_date_sk")
+    j = _merge(j, it, "inv_item_sk", "i_item_sk")
+    j = _merge(j, wh, "inv_warehouse_sk", "w_warehouse_sk")
+    keys = ["i_product_name", "i_brand", "i_class", "i_category"]
+    frames = []
+    for depth in (4, 3, 2, 1, 0):
+        t = j.copy()
+        for i, k in enumerate(keys):
+            if i >= depth:
+                t[k] = None
+        g = t.groupby(keys, dropna=False).inv_quantity_on_hand.mean().reset_index(name="qoh")
+        frames.append(g)
+    g = pd.concat(frames, ignore_index=True)
+    g = g.sort_values(["qoh"] + keys, na_position="first").head(100)
+    return g[keys + ["qoh"]].reset_index(drop=True)
+
+
+def q26(root, sf):
+    cs = _read(root, sf, "catalog_sales", ["cs_sold_date_sk", "cs_item_sk", "cs_bill_cdemo_sk",
+                                           "cs_promo_sk", "cs_quantity", "cs_list_price",
+                                           "cs_coupon_amt", "cs_sales_price"])
+    cd = _read(root, sf, "customer_demographics")
+    cd = cd[(cd.cd_gender == "M") & (cd.cd_marital_status == "S")
+            & (cd.cd_education_status == "College")]
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year"])
+    dd = dd[dd.d_year == 2000]
+    pr = _read(root, sf, "promotion")
+    pr = pr[(pr.p_channel_email == "N") | (pr.p_channel_event == "N")]
+    it = _read(root, sf, "item", ["i_item_sk", "i_item_id"])
+    j = _merge(cs, cd, "cs_bill_cdemo_sk", "cd_demo_sk")
+    j = _merge(j, dd, "cs_sold_date_sk", "d_date_sk")
+    j = _merge(j, pr, "cs_promo_sk", "p_promo_sk")
+    j = _merge(j, it, "cs_item_sk", "i_item_sk")
+    g = j.groupby("i_item_id", dropna=False).agg(
+        agg1=("cs_quantity", "mean"), agg2=("cs_list_price", "mean"),
+        agg3=("cs_coupon_amt", "mean"), agg4=("cs_sales_price", "mean")).reset_index()
+    return g.sort_values("i_item_id").head(100).reset_index(drop=True)
+
+
+def _monthly_window_oracle(root, sf, group_key, extra_group, out_cols, sort_cols,
+                           dd_pred):
+    ss = _read(root, sf, "store_sales", ["ss_item_sk", "ss_sold_date_sk", "ss_store_sk",
+                                         "ss_sales_price"])
+    it = _read(root, sf, "item")
+    m = ((it.i_category.isin(["Books", "Children", "Electronics"])
+          & it.i_class.isin(["class1", "class2", "class3", "class4"]))
+         | (it.i_category.isin(["Women", "Music", "Men"])
+            & it.i_class.isin(["class5", "class6", "class7", "class8"])))
+    it = it[m]
+    dd = _read(root, sf, "date_dim")
+    dd = dd[dd_pred(dd)]
+    st = _read(root, sf, "store", ["s_store_sk", "s_store_name"])
+    j = _merge(ss, it, "ss_item_sk", "i_item_sk")
+    j = _merge(j, dd, "ss_sold_date_sk", "d_date_sk")
+    j = _merge(j, st, "ss_store_sk", "s_store_sk")
+    g = j.groupby([group_key, extra_group], dropna=False) \
+         .ss_sales_price.sum(min_count=1).reset_index(name="sum_sales")
+    g["avg_sales"] = g.groupby(group_key, dropna=False).sum_sales.transform("mean")
+    cond = (g.avg_sales > 0) & ((g.sum_sales - g.avg_sales).abs() / g.avg_sales > 0.1)
+    f = g[cond.fillna(False)]
+    out = f[out_cols].sort_values(sort_cols, na_position="first").head(100)
+    return out.reset_index(drop=True)
+
+
+def q53(root, sf):
+    return _monthly_window_oracle(root, sf, "i_manufact_id", "d_qoy",
+                                  ["i_manufact_id", "sum_sales", "avg_sales"],
+                                  ["avg_sales", "sum_sales", "i_manufact_id"],
+                                  lambda d: d.d_month_seq.between(1200, 1211))
+
+
+def q63(root, sf):
+    return _monthly_window_oracle(root, sf, "i_manager_id", "d_moy",
+                                  ["i_manager_id", "sum_sales", "avg_sales"],
+                                  ["i_manager_id", "avg_sales", "sum_sales"],
+                                  lambda d: d.d_month_seq.between(1200, 1211))
+
+
+def q89(root, sf):
+    ss = _read(root, sf, "store_sales", ["ss_item_sk", "ss_sold_date_sk", "ss_store_sk",
+                                         "ss_sales_price"])
+    it = _read(root, sf, "item")
+    m = ((it.i_category.isin(["Books", "Children", "Electronics"])
+          & it.i_class.isin(["class1", "class2", "class3", "class4"]))
+         | (it.i_category.isin(["Women", "Music", "Men"])
+            & it.i_class.isin(["class5", "class6", "class7", "class8"])))
+    it = it[m]
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year", "d_moy"])
+    dd = dd[dd.d_year == 1999]
+    st = _read(root, sf, "store", ["s_store_sk", "s_store_name", "s_company_id"])
+    j = _merge(ss, it, "ss_item_sk", "i_item_sk")
+    j = _merge(j, dd, "ss_sold_date_sk", "d_date_sk")
+    j = _merge(j, st, "ss_store_sk", "s_store_sk")
+    g = j.groupby(["i_category", "i_class", "i_brand", "s_store_name",
+                   "s_company_id", "d_moy"], dropna=False) \
+         .ss_sales_price.sum(min_count=1).reset_index(name="sum_sales")
+    g["avg_monthly_sales"] = g.groupby(
+        ["i_category", "i_brand", "s_store_name", "s_company_id"],
+        dropna=False).sum_sales.transform("mean")
+    cond = (g.avg_monthly_sales != 0) & \
+        ((g.sum_sales - g.avg_monthly_sales).abs() / g.avg_monthly_sales > 0.1)
+    f = g[cond.fillna(False)].copy()
+    f["_d"] = f.sum_sales - f.avg_monthly_sales
+    f = f.sort_values(["_d", "s_store_name"], na_position="first").head(100)
+    return f[["i_category", "i_class", "i_brand", "s_store_name", "s_company_id",
+              "d_moy", "sum_sales", "avg_monthly_sales"]].reset_index(drop=True)
+
+
+def _inv_range_oracle(root, sf, fact, fk, lo, mfg, day0):
+    inv = _read(root, sf, "inventory")
+    it = _read(root, sf, "item", ["i_item_sk", "i_item_id", "i_item_desc",
+                                  "i_current_price", "i_manufact_id"])
+    it = it[it.i_current_price.between(lo, lo + 30) & it.i_manufact_id.isin(mfg)]
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_date"])
+    di = _date_i(dd)
+    dd = dd[(di >= day0) & (di <= day0 + 60)]
+    fs = _read(root, sf, fact, [fk])
+    j = _merge(inv, it, "inv_item_sk", "i_item_sk")
+    j = _merge(j, dd, "inv_date_sk", "d_date_sk")
+    j = j[j.inv_quantity_on_hand.between(100, 500)]
+    j = j[j.inv_item_sk.isin(fs[fk].dropna())]
+    g = j.groupby(["i_item_id", "i_item_desc", "i_current_price"], dropna=False) \
+         .size().reset_index()[["i_item_id", "i_item_desc", "i_current_price"]]
+    return g.sort_values("i_item_id").head(100).reset_index(drop=True)
+
+
+def q37(root, sf):
+    return _inv_range_oracle(root, sf, "catalog_sales", "cs_item_sk", 68.0,
+                             [677, 940, 694, 808], _days(2000, 2, 1))
+
+
+def q82(root, sf):
+    return _inv_range_oracle(root, sf, "store_sales", "ss_item_sk", 62.0,
+                             [129, 270, 821, 423], _days(2000, 5, 25))
+
+
+def q62(root, sf):
+    ws = _read(root, sf, "web_sales", ["ws_ship_date_sk", "ws_sold_date_sk",
+                                       "ws_warehouse_sk", "ws_ship_mode_sk", "ws_web_site_sk"])
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_month_seq"])
+    dd = dd[dd.d_month_seq.between(1200, 1211)]
+    wh = _read(root, sf, "warehouse", ["w_warehouse_sk", "w_warehouse_name"])
+    sm = _read(root, sf, "ship_mode", ["sm_ship_mode_sk", "sm_type"])
+    web = _read(root, sf, "web_site", ["web_site_sk", "web_name"])
+    j = _merge(ws, dd, "ws_ship_date_sk", "d_date_sk")
+    j = _merge(j, wh, "ws_warehouse_sk", "w_warehouse_sk")
+    j = _merge(j, sm, "ws_ship_mode_sk", "sm_ship_mode_sk")
+    j = _merge(j, web, "ws_web_site_sk", "web_site_sk")
+    j["wname20"] = j.w_warehouse_name.str[:20]
+    lag = j.ws_ship_date_sk - j.ws_sold_date_sk
+    j["d30"] = ((lag <= 30)).astype("float").where(lag.notna())
+    j["d31_60"] = ((lag > 30) & (lag <= 60)).astype("float").where(lag.notna())
+    j["d61_90"] = ((lag > 60) & (lag <= 90)).astype("float").where(lag.notna())
+    j["d91_120"] = ((lag > 90) & (lag <= 120)).astype("float").where(lag.notna())
+    j["d120p"] = ((lag > 120)).astype("float").where(lag.notna())
+    cols = ["d30", "d31_60", "d61_90", "d91_120", "d120p"]
+    g = j.groupby(["wname20", "sm_type", "web_name"], dropna=False)[cols] \
+         .sum(min_count=1).reset_index()
+    for c in cols:
+        g[c] = g[c].astype("Int64")
+    g = g.sort_values(["wname20", "sm_type", "web_name"], na_position="first").head(100)
+    return g.reset_index(drop=True)
+
+
+def q90(root, sf):
+    import pandas as pd
+
+    def cnt(h_lo, h_hi):
+        ws = _read(root, sf, "web_sales", ["ws_sold_time_sk", "ws_bill_hdemo_sk", "ws_web_page_sk"])
+        td = _read(root, sf, "time_dim")
+        td = td[td.t_hour.between(h_lo, h_hi)]
+        hd = _read(root, sf, "household_demographics")
+        hd = hd[hd.hd_dep_count == 6]
+        wp = _read(root, sf, "web_page")
+        wp = wp[wp.wp_char_count.between(5000, 5200)]
+        j = _merge(ws, td, "ws_sold_time_sk", "t_time_sk")
+        j = _merge(j, hd, "ws_bill_hdemo_sk", "hd_demo_sk")
+        j = _merge(j, wp, "ws_web_page_sk", "wp_web_page_sk")
+        return len(j)
+
+    amc, pmc = cnt(8, 9), cnt(19, 20)
+    return pd.DataFrame({"am_pm_ratio": [amc / pmc if pmc else None]})
+
+
+def q91(root, sf):
+    cr = _read(root, sf, "catalog_returns", ["cr_call_center_sk", "cr_returned_date_sk",
+                                             "cr_returning_customer_sk", "cr_net_loss"])
+    cc = _read(root, sf, "call_center")
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year", "d_moy"])
+    dd = dd[(dd.d_year == 1998) & (dd.d_moy == 11)]
+    cust = _read(root, sf, "customer", ["c_customer_sk", "c_current_cdemo_sk",
+                                        "c_current_hdemo_sk", "c_current_addr_sk"])
+    cd = _read(root, sf, "customer_demographics")
+    cd = cd[((cd.cd_marital_status == "M") & (cd.cd_education_status == "Unknown"))
+            | ((cd.cd_marital_status == "W") & (cd.cd_education_status == "Advanced Degree"))]
+    hd = _read(root, sf, "household_demographics")
+    hd = hd[hd.hd_buy_potential.str.startswith("Unknown")]
+    ca = _read(root, sf, "customer_address", ["ca_address_sk", "ca_gmt_offset"])
+    ca = ca[ca.ca_gmt_offset == -7.0]
+    j = _merge(cr, cc, "cr_call_center_sk", "cc_call_center_sk")
+    j = _merge(j, dd, "cr_returned_date_sk", "d_date_sk")
+    j = _merge(j, cust, "cr_returning_customer_sk", "c_customer_sk")
+    j = _merge(j, cd, "c_current_cdemo_sk", "cd_demo_sk")
+    j = _merge(j, hd, "c_current_hdemo_sk", "hd_demo_sk")
+    j = _merge(j, ca, "c_current_addr_sk", "ca_address_sk")
+    g = j.groupby(["cc_call_center_id", "cc_name", "cc_manager",
+                   "cd_marital_status", "cd_education_status"], dropna=False) \
+         .cr_net_loss.sum(min_count=1).reset_index(name="returns_loss")
+    out = g[["cc_call_center_id", "cc_name", "cc_manager", "returns_loss"]]
+    out.columns = ["call_center", "call_center_name", "manager", "returns_loss"]
+    return out.sort_values("returns_loss", ascending=False).reset_index(drop=True)
+
+
+def q93(root, sf):
+    import numpy as np
+
+    ss = _read(root, sf, "store_sales", ["ss_item_sk", "ss_ticket_number", "ss_customer_sk",
+                                         "ss_quantity", "ss_sales_price"])
+    sr = _read(root, sf, "store_returns", ["sr_item_sk", "sr_ticket_number",
+                                           "sr_reason_sk", "sr_return_quantity"])
+    re = _read(root, sf, "reason")
+    re = re[re.r_reason_desc == "reason 28"]
+    srj = _merge(sr, re, "sr_reason_sk", "r_reason_sk")
+    j = ss.merge(srj.dropna(subset=["sr_item_sk", "sr_ticket_number"]),
+                 left_on=["ss_item_sk", "ss_ticket_number"],
+                 right_on=["sr_item_sk", "sr_ticket_number"], how="left")
+    j = j[j.sr_reason_sk.notna()]
+    act = np.where(j.sr_return_quantity.notna(),
+                   (j.ss_quantity - j.sr_return_quantity) * j.ss_sales_price,
+                   j.ss_quantity * j.ss_sales_price)
+    j = j.assign(act_sales=act)
+    j.loc[j.ss_quantity.isna() | j.ss_sales_price.isna(), "act_sales"] = np.nan
+    g = j.groupby("ss_customer_sk", dropna=False).act_sales.sum(min_count=1) \
+         .reset_index(name="sumsales")
+    g = g.sort_values(["sumsales", "ss_customer_sk"], na_position="first").head(100)
+    return g[["ss_customer_sk", "sumsales"]].reset_index(drop=True)
+
+
+ORACLES.update({"q12": q12, "q15": q15, "q20": q20, "q22": q22, "q26": q26,
+                "q37": q37, "q53": q53, "q62": q62, "q63": q63, "q82": q82,
+                "q89": q89, "q90": q90, "q91": q91, "q93": q93, "q98": q98})

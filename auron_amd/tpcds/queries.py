@@ -657,3 +657,377 @@ QUERIES.update({
     "q25": q25, "q29": q29, "q34": q34, "q43": q43, "q46": q46, "q48": q48,
     "q65": q65, "q72": q72, "q73": q73, "q79": q79, "q88": q88,
 })
+
+
+# ------------------------------- batch 3: window / rollup / more channels
+def _days(y, m, d):
+    import datetime
+
+    return (datetime.date(y, m, d) - datetime.date(1970, 1, 1)).days
+
+
+def windowed(child, part_keys, order_keys, funcs):
+    """Exchange by the window partition keys, then Window locally."""
+    ex = P.Exchange(child, "hash", [col(k) for k in part_keys])
+    return P.Window(ex, [col(k) for k in part_keys], order_keys, funcs)
+
+
+# Literal adaptations for the synthetic dsdgen (documented deviations:
+# dsdgen's generated brand/class strings don't exist in the synthetic
+# catalog; equivalent-selectivity predicates on i_class/i_category are
+# used instead, keeping the same plan shape and filter structure).
+_Q53_ITEM_FILTER = ((col("i_category").isin(["Books", "Children", "Electronics"])
+                     & col("i_class").isin(["class1", "class2", "class3", "class4"]))
+                    | (col("i_category").isin(["Women", "Music", "Men"])
+                       & col("i_class").isin(["class5", "class6", "class7", "class8"])))
+
+
+def _ratio_window_q(cat, s, fact, prefix, measure):
+    """q12/q20/q98 shape: date-window scan + item cat filter + revenue ratio
+    over i_class window."""
+    date_cols = [f"{prefix}_sold_date_sk", f"{prefix}_item_sk", measure]
+    ss = cat.scan(fact, date_cols)
+    it = P.Filter(cat.scan("item", ["i_item_sk", "i_item_id", "i_item_desc",
+                                    "i_category", "i_class", "i_current_price"]),
+                  col("i_category").isin(["Sports", "Books", "Home"]))
+    lo = _days(1999, 2, 22)
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_date"]),
+                  col("d_date").cast(dtypes.int32).between(lo, lo + 30))
+    j = bhj(ss, it, [f"{prefix}_item_sk"], ["i_item_sk"])
+    j = bhj(j, dd, [f"{prefix}_sold_date_sk"], ["d_date_sk"])
+    a = agg2(j, ["i_item_id", "i_item_desc", "i_category", "i_class", "i_current_price"],
+             [AggFunc("sum", col(measure), name="itemrevenue")])
+    from ..exprs import WindowFunc
+
+    w = windowed(a, ["i_class"], [], [_a(WindowFunc("sum", col("itemrevenue")), "_clsrev")])
+    proj = P.Project(w, [_a(col("i_item_desc"), "i_item_desc"),
+                         _a(col("i_category"), "i_category"),
+                         _a(col("i_class"), "i_class"),
+                         _a(col("i_current_price"), "i_current_price"),
+                         _a(col("itemrevenue"), "itemrevenue"),
+                         _a(col("itemrevenue") * lit(100.0) / col("_clsrev"), "revenueratio"),
+                         _a(col("i_item_id"), "i_item_id")])
+    out = topk(proj, [(col("i_category"), True), (col("i_class"), True),
+                      (col("i_item_id"), True), (col("i_item_desc"), True),
+                      (col("revenueratio"), True)], 100)
+    return P.Project(out, [_a(col(c), c) for c in
+                           ["i_item_desc", "i_category", "i_class",
+                            "i_current_price", "itemrevenue", "revenueratio"]])
+
+
+def q12(cat, s):
+    return _ratio_window_q(cat, s, "web_sales", "ws", "ws_ext_sales_price")
+
+
+def q20(cat, s):
+    return _ratio_window_q(cat, s, "catalog_sales", "cs", "cs_ext_sales_price")
+
+
+def q98(cat, s):
+    return _ratio_window_q(cat, s, "store_sales", "ss", "ss_ext_sales_price")
+
+
+def q15(cat, s):
+    from ..exprs import Substr
+
+    cs = cat.scan("catalog_sales", ["cs_bill_customer_sk", "cs_sold_date_sk", "cs_sales_price"])
+    cust = cat.scan("customer", ["c_customer_sk", "c_current_addr_sk"])
+    ca = cat.scan("customer_address", ["ca_address_sk", "ca_zip", "ca_state"])
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_qoy", "d_year"]),
+                  (col("d_qoy") == 2) & (col("d_year") == 2001))
+    j = bhj(cs, cust, ["cs_bill_customer_sk"], ["c_customer_sk"])
+    j = bhj(j, ca, ["c_current_addr_sk"], ["ca_address_sk"])
+    j = bhj(j, dd, ["cs_sold_date_sk"], ["d_date_sk"])
+    zips = ["85669", "86197", "88274", "83405", "86475", "85392", "85460", "80348", "81792"]
+    f = P.Filter(j, Substr(col("ca_zip"), 1, 5).isin(zips)
+                 | col("ca_state").isin(["CA", "WA", "GA"])
+                 | (col("cs_sales_price") > 500.0))
+    a = agg2(f, ["ca_zip"], [AggFunc("sum", col("cs_sales_price"), name="s")])
+    return topk(a, [(col("ca_zip"), True)], 100)
+
+
+def q22(cat, s):
+    from ..exprs import Literal
+
+    inv = cat.scan("inventory", ["inv_date_sk", "inv_item_sk", "inv_warehouse_sk",
+                                 "inv_quantity_on_hand"])
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_month_seq"]),
+                  col("d_month_seq").between(1200, 1211))
+    it = cat.scan("item", ["i_item_sk", "i_product_name", "i_brand", "i_class", "i_category"])
+    wh = cat.scan("warehouse", ["w_warehouse_sk"])
+    j = bhj(inv, dd, ["inv_date_sk"], ["d_date_sk"])
+    j = bhj(j, it, ["inv_item_sk"], ["i_item_sk"])
+    j = bhj(j, wh, ["inv_warehouse_sk"], ["w_warehouse_sk"])
+    keys = ["i_product_name", "i_brand", "i_class", "i_category"]
+    nul = Literal(None, dtypes.string)
+    projections = []
+    for depth in (4, 3, 2, 1, 0):  # ROLLUP grouping sets
+        proj = [_a(col(k) if i < depth else nul, k) for i, k in enumerate(keys)]
+        proj.append(_a(lit(depth), "_gid"))
+        proj.append(_a(col("inv_quantity_on_hand"), "inv_quantity_on_hand"))
+        projections.append(proj)
+    ex = P.Expand(j, projections)
+    a = agg2(ex, keys + ["_gid"], [AggFunc("avg", col("inv_quantity_on_hand"), name="qoh")])
+    proj = P.Project(a, [_a(col(k), k) for k in keys] + [_a(col("qoh"), "qoh")])
+    out = topk(proj, [(col("qoh"), True)] + [(col(k), True) for k in keys], 100)
+    return P.Project(out, [_a(col(k), k) for k in keys] + [_a(col("qoh"), "qoh")])
+
+
+def q26(cat, s):
+    cs = cat.scan("catalog_sales", ["cs_sold_date_sk", "cs_item_sk", "cs_bill_cdemo_sk",
+                                    "cs_promo_sk", "cs_quantity", "cs_list_price",
+                                    "cs_coupon_amt", "cs_sales_price"])
+    cd = P.Filter(cat.scan("customer_demographics",
+                           ["cd_demo_sk", "cd_gender", "cd_marital_status", "cd_education_status"]),
+                  (col("cd_gender") == lit("M")) & (col("cd_marital_status") == lit("S"))
+                  & (col("cd_education_status") == lit("College")))
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_year"]), col("d_year") == 2000)
+    pr = P.Filter(cat.scan("promotion", ["p_promo_sk", "p_channel_email", "p_channel_event"]),
+                  (col("p_channel_email") == lit("N")) | (col("p_channel_event") == lit("N")))
+    it = cat.scan("item", ["i_item_sk", "i_item_id"])
+    j = bhj(cs, cd, ["cs_bill_cdemo_sk"], ["cd_demo_sk"])
+    j = bhj(j, dd, ["cs_sold_date_sk"], ["d_date_sk"])
+    j = bhj(j, pr, ["cs_promo_sk"], ["p_promo_sk"])
+    j = bhj(j, it, ["cs_item_sk"], ["i_item_sk"])
+    a = agg2(j, ["i_item_id"],
+             [AggFunc("avg", col("cs_quantity"), name="agg1"),
+              AggFunc("avg", col("cs_list_price"), name="agg2"),
+              AggFunc("avg", col("cs_coupon_amt"), name="agg3"),
+              AggFunc("avg", col("cs_sales_price"), name="agg4")])
+    return topk(a, [(col("i_item_id"), True)], 100)
+
+
+def _monthly_window_q(cat, s, group_key, extra_group, out_keys, order_keys_fn,
+                      month_filter):
+    from ..exprs import CaseWhen, Literal, WindowFunc
+
+    ss = cat.scan("store_sales", ["ss_item_sk", "ss_sold_date_sk", "ss_store_sk",
+                                  "ss_sales_price"])
+    it = P.Filter(cat.scan("item", ["i_item_sk", "i_manufact_id", "i_manager_id",
+                                    "i_category", "i_class", "i_brand"]),
+                  _Q53_ITEM_FILTER)
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_month_seq", "d_moy", "d_qoy", "d_year"]),
+                  month_filter)
+    st = cat.scan("store", ["s_store_sk", "s_store_name"])
+    j = bhj(ss, it, ["ss_item_sk"], ["i_item_sk"])
+    j = bhj(j, dd, ["ss_sold_date_sk"], ["d_date_sk"])
+    j = bhj(j, st, ["ss_store_sk"], ["s_store_sk"])
+    a = agg2(j, [group_key, extra_group],
+             [AggFunc("sum", col("ss_sales_price"), name="sum_sales")])
+    w = windowed(a, [group_key], [],
+                 [_a(WindowFunc("avg", col("sum_sales")), "avg_sales")])
+    ratio = CaseWhen([(col("avg_sales") > 0.0,
+                       col("sum_sales") - col("avg_sales"))], Literal(None, dtypes.float64))
+    # abs(x)/avg > 0.1 without an abs fn: compare both signs
+    diff = col("sum_sales") - col("avg_sales")
+    cond = CaseWhen([(col("avg_sales") > 0.0,
+                      ((diff / col("avg_sales")) > 0.1) | ((diff / col("avg_sales")) < -0.1))],
+                    Literal(None, dtypes.bool_))
+    f = P.Filter(w, cond)
+    proj = P.Project(f, [_a(col(c), c) for c in out_keys])
+    return topk(proj, order_keys_fn(), 100)
+
+
+def q53(cat, s):
+    return _monthly_window_q(
+        cat, s, "i_manufact_id", "d_qoy",
+        ["i_manufact_id", "sum_sales", "avg_sales"],
+        lambda: [(col("avg_sales"), True), (col("sum_sales"), True),
+                 (col("i_manufact_id"), True)],
+        col("d_month_seq").between(1200, 1211))
+
+
+def q63(cat, s):
+    return _monthly_window_q(
+        cat, s, "i_manager_id", "d_moy",
+        ["i_manager_id", "sum_sales", "avg_sales"],
+        lambda: [(col("i_manager_id"), True), (col("avg_sales"), True),
+                 (col("sum_sales"), True)],
+        col("d_month_seq").between(1200, 1211))
+
+
+def q89(cat, s):
+    from ..exprs import CaseWhen, Literal, WindowFunc
+
+    ss = cat.scan("store_sales", ["ss_item_sk", "ss_sold_date_sk", "ss_store_sk",
+                                  "ss_sales_price"])
+    it = P.Filter(cat.scan("item", ["i_item_sk", "i_category", "i_class", "i_brand"]),
+                  _Q53_ITEM_FILTER)
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_year", "d_moy"]),
+                  col("d_year") == 1999)
+    st = cat.scan("store", ["s_store_sk", "s_store_name", "s_company_id"])
+    j = bhj(ss, it, ["ss_item_sk"], ["i_item_sk"])
+    j = bhj(j, dd, ["ss_sold_date_sk"], ["d_date_sk"])
+    j = bhj(j, st, ["ss_store_sk"], ["s_store_sk"])
+    a = agg2(j, ["i_category", "i_class", "i_brand", "s_store_name", "s_company_id", "d_moy"],
+             [AggFunc("sum", col("ss_sales_price"), name="sum_sales")])
+    w = windowed(a, ["i_category", "i_brand", "s_store_name", "s_company_id"], [],
+                 [_a(WindowFunc("avg", col("sum_sales")), "avg_monthly_sales")])
+    diff = col("sum_sales") - col("avg_monthly_sales")
+    cond = CaseWhen([(~(col("avg_monthly_sales") == 0.0),
+                      ((diff / col("avg_monthly_sales")) > 0.1)
+                      | ((diff / col("avg_monthly_sales")) < -0.1))],
+                    Literal(None, dtypes.bool_))
+    f = P.Filter(w, cond)
+    proj = P.Project(f, [_a(col(c), c) for c in
+                         ["i_category", "i_class", "i_brand", "s_store_name",
+                          "s_company_id", "d_moy", "sum_sales", "avg_monthly_sales"]]
+                     + [_a(diff, "_d")])
+    out = topk(proj, [(col("_d"), True), (col("s_store_name"), True)], 100)
+    return P.Project(out, [_a(col(c), c) for c in
+                           ["i_category", "i_class", "i_brand", "s_store_name",
+                            "s_company_id", "d_moy", "sum_sales", "avg_monthly_sales"]])
+
+
+def _inv_range_q(cat, s, fact, fk, price_lo, mfg_ids, day0):
+    inv = cat.scan("inventory", ["inv_item_sk", "inv_date_sk", "inv_quantity_on_hand"])
+    it = P.Filter(cat.scan("item", ["i_item_sk", "i_item_id", "i_item_desc",
+                                    "i_current_price", "i_manufact_id"]),
+                  col("i_current_price").between(float(price_lo), float(price_lo + 30))
+                  & col("i_manufact_id").isin(mfg_ids))
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_date"]),
+                  col("d_date").cast(dtypes.int32).between(day0, day0 + 60))
+    fs = cat.scan(fact, [fk])
+    j = bhj(inv, it, ["inv_item_sk"], ["i_item_sk"])
+    j = bhj(j, dd, ["inv_date_sk"], ["d_date_sk"])
+    j = P.Filter(j, col("inv_quantity_on_hand").between(100, 500))
+    j = P.HashJoin(P.Exchange(j, "hash", [col("inv_item_sk")]),
+                   P.Exchange(fs, "hash", [col(fk)]),
+                   [col("inv_item_sk")], [col(fk)], how="semi", build_side="right")
+    a = agg2(j, ["i_item_id", "i_item_desc", "i_current_price"], [])
+    return topk(a, [(col("i_item_id"), True)], 100)
+
+
+def q37(cat, s):
+    return _inv_range_q(cat, s, "catalog_sales", "cs_item_sk", 68,
+                        [677, 940, 694, 808], _days(2000, 2, 1))
+
+
+def q82(cat, s):
+    return _inv_range_q(cat, s, "store_sales", "ss_item_sk", 62,
+                        [129, 270, 821, 423], _days(2000, 5, 25))
+
+
+def q62(cat, s):
+    from ..exprs import CaseWhen, Literal, Substr
+
+    ws = cat.scan("web_sales", ["ws_ship_date_sk", "ws_sold_date_sk",
+                                "ws_warehouse_sk", "ws_ship_mode_sk", "ws_web_site_sk"])
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_month_seq"]),
+                  col("d_month_seq").between(1200, 1211))
+    wh = cat.scan("warehouse", ["w_warehouse_sk", "w_warehouse_name"])
+    sm = cat.scan("ship_mode", ["sm_ship_mode_sk", "sm_type"])
+    web = cat.scan("web_site", ["web_site_sk", "web_name"])
+    j = bhj(ws, dd, ["ws_ship_date_sk"], ["d_date_sk"])
+    j = bhj(j, wh, ["ws_warehouse_sk"], ["w_warehouse_sk"])
+    j = bhj(j, sm, ["ws_ship_mode_sk"], ["sm_ship_mode_sk"])
+    j = bhj(j, web, ["ws_web_site_sk"], ["web_site_sk"])
+    lag = col("ws_ship_date_sk") - col("ws_sold_date_sk")
+    buckets = [("d30", (lag <= 30)),
+               ("d31_60", (lag > 30) & (lag <= 60)),
+               ("d61_90", (lag > 60) & (lag <= 90)),
+               ("d91_120", (lag > 90) & (lag <= 120)),
+               ("d120p", (lag > 120))]
+    aggs = [AggFunc("sum", CaseWhen([(cond, lit(1))], lit(0)), name=nm)
+            for nm, cond in buckets]
+    pre = P.Project(j, [_a(Substr(col("w_warehouse_name"), 1, 20), "wname20"),
+                        _a(col("sm_type"), "sm_type"), _a(col("web_name"), "web_name"),
+                        _a(col("ws_ship_date_sk"), "ws_ship_date_sk"),
+                        _a(col("ws_sold_date_sk"), "ws_sold_date_sk")])
+    a = agg2(pre, ["wname20", "sm_type", "web_name"], aggs)
+    return topk(a, [(col("wname20"), True), (col("sm_type"), True),
+                    (col("web_name"), True)], 100)
+
+
+def q90(cat, s):
+    def count_hours(h_lo, h_hi):
+        ws = cat.scan("web_sales", ["ws_sold_time_sk", "ws_bill_hdemo_sk", "ws_web_page_sk"])
+        td = P.Filter(cat.scan("time_dim", ["t_time_sk", "t_hour"]),
+                      col("t_hour").between(h_lo, h_hi))
+        hd = P.Filter(cat.scan("household_demographics", ["hd_demo_sk", "hd_dep_count"]),
+                      col("hd_dep_count") == 6)
+        wp = P.Filter(cat.scan("web_page", ["wp_web_page_sk", "wp_char_count"]),
+                      col("wp_char_count").between(5000, 5200))
+        j = bhj(ws, td, ["ws_sold_time_sk"], ["t_time_sk"])
+        j = bhj(j, hd, ["ws_bill_hdemo_sk"], ["hd_demo_sk"])
+        j = bhj(j, wp, ["ws_web_page_sk"], ["wp_web_page_sk"])
+        partial = P.HashAgg(j, [], [AggFunc("count_star", None, name="c")], mode="partial")
+        return scalar(s, P.HashAgg(P.Exchange(partial, "single"), [],
+                                   [AggFunc("count_star", None, name="c")], mode="final"))
+
+    amc = count_hours(8, 9)
+    pmc = count_hours(19, 20)
+    from ..column import RecordBatch
+
+    n = 1 if s.rank == 0 else 0
+    ratio = (amc / pmc) if pmc else None
+    b = RecordBatch.from_pydict({"am_pm_ratio": [ratio] * max(n, 1)},
+                                {"am_pm_ratio": dtypes.float64})
+    if n == 0:
+        b = b.slice(0, 0)
+    return P.MemoryScan([b])
+
+
+def q91(cat, s):
+    cr = cat.scan("catalog_returns", ["cr_call_center_sk", "cr_returned_date_sk",
+                                      "cr_returning_customer_sk", "cr_net_loss"])
+    cc = cat.scan("call_center", ["cc_call_center_sk", "cc_call_center_id",
+                                  "cc_name", "cc_manager"])
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_year", "d_moy"]),
+                  (col("d_year") == 1998) & (col("d_moy") == 11))
+    cust = cat.scan("customer", ["c_customer_sk", "c_current_cdemo_sk",
+                                 "c_current_hdemo_sk", "c_current_addr_sk"])
+    cd = P.Filter(cat.scan("customer_demographics",
+                           ["cd_demo_sk", "cd_marital_status", "cd_education_status"]),
+                  ((col("cd_marital_status") == lit("M")) & (col("cd_education_status") == lit("Unknown")))
+                  | ((col("cd_marital_status") == lit("W")) & (col("cd_education_status") == lit("Advanced Degree"))))
+    hd = P.Filter(cat.scan("household_demographics", ["hd_demo_sk", "hd_buy_potential"]),
+                  col("hd_buy_potential").like("Unknown%"))
+    ca = P.Filter(cat.scan("customer_address", ["ca_address_sk", "ca_gmt_offset"]),
+                  col("ca_gmt_offset") == -7.0)
+    j = bhj(cr, cc, ["cr_call_center_sk"], ["cc_call_center_sk"])
+    j = bhj(j, dd, ["cr_returned_date_sk"], ["d_date_sk"])
+    j = bhj(j, cust, ["cr_returning_customer_sk"], ["c_customer_sk"])
+    j = bhj(j, cd, ["c_current_cdemo_sk"], ["cd_demo_sk"])
+    j = bhj(j, hd, ["c_current_hdemo_sk"], ["hd_demo_sk"])
+    j = bhj(j, ca, ["c_current_addr_sk"], ["ca_address_sk"])
+    a = agg2(j, ["cc_call_center_id", "cc_name", "cc_manager",
+                 "cd_marital_status", "cd_education_status"],
+             [AggFunc("sum", col("cr_net_loss"), name="returns_loss")])
+    proj = P.Project(a, [_a(col("cc_call_center_id"), "call_center"),
+                         _a(col("cc_name"), "call_center_name"),
+                         _a(col("cc_manager"), "manager"),
+                         _a(col("returns_loss"), "returns_loss")])
+    return topk(proj, [(col("returns_loss"), False)], 100000)
+
+
+def q93(cat, s):
+    from ..exprs import CaseWhen, IsNull, Not
+
+    ss = cat.scan("store_sales", ["ss_item_sk", "ss_ticket_number", "ss_customer_sk",
+                                  "ss_quantity", "ss_sales_price"])
+    sr = cat.scan("store_returns", ["sr_item_sk", "sr_ticket_number",
+                                    "sr_reason_sk", "sr_return_quantity"])
+    re = P.Filter(cat.scan("reason", ["r_reason_sk", "r_reason_desc"]),
+                  col("r_reason_desc") == lit("reason 28"))
+    srj = bhj(sr, re, ["sr_reason_sk"], ["r_reason_sk"])
+    j = shj(ss, srj, ["ss_item_sk", "ss_ticket_number"],
+            ["sr_item_sk", "sr_ticket_number"], how="left")
+    # WHERE sr_reason_sk = r_reason_sk drops null-return rows -> the left
+    # join is effectively inner after the reason filter (faithful to the
+    # published query text)
+    j = P.Filter(j, Not(IsNull(col("sr_reason_sk"))))
+    act = CaseWhen(
+        [(Not(IsNull(col("sr_return_quantity"))),
+          (col("ss_quantity") - col("sr_return_quantity")).cast(dtypes.float64) * col("ss_sales_price"))],
+        col("ss_quantity").cast(dtypes.float64) * col("ss_sales_price"))
+    pre = P.Project(j, [_a(col("ss_customer_sk"), "ss_customer_sk"), _a(act, "act_sales")])
+    a = agg2(pre, ["ss_customer_sk"], [AggFunc("sum", col("act_sales"), name="sumsales")])
+    return topk(a, [(col("sumsales"), True), (col("ss_customer_sk"), True)], 100)
+
+
+QUERIES.update({
+    "q12": q12, "q15": q15, "q20": q20, "q22": q22, "q26": q26, "q37": q37,
+    "q53": q53, "q62": q62, "q63": q63, "q82": q82, "q89": q89, "q90": q90,
+    "q91": q91, "q93": q93, "q98": q98,
+})
