@@ -145,6 +145,64 @@ class Executor:
             return list(pool.map(
                 lambda f: self._read_parquet(f, node.columns, node.filters), my_files))
 
+    def _exec_OrcScan(self, node: P.OrcScan) -> List[RecordBatch]:
+        import pyarrow.orc as orc
+
+        my_files = node.paths[self.ctx.rank::self.ctx.world_size]
+        out = []
+        for f in my_files:
+            t = orc.ORCFile(f).read(columns=node.columns)
+            out.append(RecordBatch.from_arrow(t, self.ctx.device))
+        if not out:
+            t = orc.ORCFile(node.paths[0]).read(columns=node.columns).slice(0, 0)
+            out.append(RecordBatch.from_arrow(t, self.ctx.device))
+        return out
+
+    def _exec_ParquetSink(self, node: P.ParquetSink) -> List[RecordBatch]:
+        import os
+
+        import pyarrow.parquet as pq
+
+        os.makedirs(node.path, exist_ok=True)
+        rows = 0
+        for i, b in enumerate(self.execute(node.child)):
+            t = b.to("cpu").to_arrow()
+            pq.write_table(t, os.path.join(node.path, f"part-{self.ctx.rank:04d}-{i:04d}.parquet"))
+            rows += b.num_rows
+        return [RecordBatch.from_pydict({"rows_written": [rows]},
+                                        {"rows_written": dtypes.int64},
+                                        self.ctx.device)]
+
+    def _exec_OrcSink(self, node: P.OrcSink) -> List[RecordBatch]:
+        import os
+
+        import pyarrow.orc as orc
+
+        os.makedirs(node.path, exist_ok=True)
+        rows = 0
+        for i, b in enumerate(self.execute(node.child)):
+            t = b.to("cpu").to_arrow()
+            orc.write_table(t, os.path.join(node.path, f"part-{self.ctx.rank:04d}-{i:04d}.orc"))
+            rows += b.num_rows
+        return [RecordBatch.from_pydict({"rows_written": [rows]},
+                                        {"rows_written": dtypes.int64},
+                                        self.ctx.device)]
+
+    def _exec_PyUdf(self, node: P.PyUdf) -> List[RecordBatch]:
+        # device -> host, evaluate foreign function, host -> device
+        # (the unavoidable FFI bounce of spark_udf_wrapper.rs:207)
+        out = []
+        for b in self.execute(node.child):
+            host = b.to("cpu")
+            res = node.fn(host)
+            names = list(b.names)
+            cols = list(b.columns)
+            for name, (values, dt) in res.items():
+                names.append(name)
+                cols.append(Column.from_pylist(values, dt, str(self.ctx.device)))
+            out.append(RecordBatch(names, cols))
+        return out
+
     # ------------------------------------------------------- row operators
     def _exec_Filter(self, node: P.Filter) -> List[RecordBatch]:
         out = []
@@ -174,6 +232,38 @@ class Executor:
         for b in bs:
             print(f"[debug {node.label}] rank={self.ctx.rank} {b}")
         return bs
+
+    def _exec_Generate(self, node: P.Generate) -> List[RecordBatch]:
+        """explode/posexplode (generate_exec.rs). Without a list dtype the
+        exploded source is a delimited string column: generator
+        'explode_split' / 'posexplode_split' with args [expr, lit(delim)]."""
+        if node.generator not in ("explode_split", "posexplode_split"):
+            raise NotImplementedError(f"generator {node.generator}")
+        delim = node.args[1].value if len(node.args) > 1 else ","
+        out = []
+        for b in self.execute(node.child):
+            c = node.args[0].eval(b).to("cpu")
+            vals = c.to_pylist()
+            rows, toks, poss = [], [], []
+            for i, v in enumerate(vals):
+                if v is None:
+                    continue
+                for p, tok in enumerate(v.split(delim)):
+                    rows.append(i)
+                    toks.append(tok)
+                    poss.append(p)
+            idx = torch.tensor(rows, dtype=torch.int64, device=b.device)
+            base = b.gather(idx)
+            names = list(base.names)
+            cols = list(base.columns)
+            if node.generator == "posexplode_split":
+                names.append("pos")
+                cols.append(Column(dtypes.int32,
+                                   torch.tensor(poss, dtype=torch.int32, device=b.device)))
+            names.append("col")
+            cols.append(Column.from_pylist(toks, dtypes.string, str(b.device)))
+            out.append(RecordBatch(names, cols))
+        return out
 
     def _exec_Union(self, node: P.Union) -> List[RecordBatch]:
         out = []

@@ -217,9 +217,41 @@ class CoalesceBatches(PlanNode):
 
 
 @dataclass(eq=False)
+class OrcScan(PlanNode):
+    """ORC scan via host pyarrow.orc (orc_exec.rs analogue); device decode
+    follows the parquet path's trajectory."""
+    paths: List[str]
+    columns: Optional[List[str]] = None
+
+
+@dataclass(eq=False)
 class ParquetSink(PlanNode):
+    """Partitioned parquet writer (parquet_sink_exec.rs analogue). Each
+    rank writes part-{rank}-{seq}.parquet under `path`."""
     child: PlanNode
     path: str
+
+    def children(self):
+        return [self.child]
+
+
+@dataclass(eq=False)
+class OrcSink(PlanNode):
+    child: PlanNode
+    path: str
+
+    def children(self):
+        return [self.child]
+
+
+@dataclass(eq=False)
+class PyUdf(PlanNode):
+    """Host-evaluated UDF projection (spark_udf_wrapper.rs analogue: the
+    device->host->device bounce for engine-foreign functions). `fn` maps
+    a host RecordBatch to a dict of {name: (values, DataType)}."""
+    child: PlanNode
+    fn: object
+    names: List[str] = field(default_factory=list)
 
     def children(self):
         return [self.child]
