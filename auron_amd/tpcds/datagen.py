@@ -339,9 +339,9 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
         put(f"{pre}_bill_addr_sk", a, av)
         c2, c2v = fk(n_cust)
         put(f"{pre}_ship_customer_sk", c2, c2v)
+        a2, a2v = fk(n_addr)
+        put(f"{pre}_ship_addr_sk", a2, a2v)
         if pre == "cs":
-            a2, a2v = fk(n_addr)
-            put("cs_ship_addr_sk", a2, a2v)
             cc, ccv = fk(BASE_ROWS["call_center"], 0.02)
             put("cs_call_center_sk", cc, ccv)
             cp, cpv = fk(BASE_ROWS["catalog_page"], 0.02)
@@ -460,8 +460,11 @@ def _nparts_for(table: str, sf: float) -> int:
     return 1
 
 
+DATAGEN_VERSION = 2
+
+
 def dataset_root(root: str, sf: float) -> str:
-    tag = f"sf{sf:g}"
+    tag = f"sf{sf:g}-v{DATAGEN_VERSION}"
     return os.path.join(root, tag)
 
 

@@ -834,3 +834,170 @@ def q93(root, sf):
 ORACLES.update({"q12": q12, "q15": q15, "q20": q20, "q22": q22, "q26": q26,
                 "q37": q37, "q53": q53, "q62": q62, "q63": q63, "q82": q82,
                 "q89": q89, "q90": q90, "q91": q91, "q93": q93, "q98": q98})
+
+
+# ------------------------------- batch 4 oracles
+def _ship_oracle(root, sf, fact, pre, date0, state, site, rets, rpre):
+    import pandas as pd
+
+    lo = _days(*date0)
+    cols = [f"{pre}_ship_date_sk", f"{pre}_ship_addr_sk", f"{pre}_order_number",
+            f"{pre}_warehouse_sk", f"{pre}_ext_ship_cost", f"{pre}_net_profit"]
+    if site:
+        cols.append(site[0])
+    fs = _read(root, sf, fact, cols)
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_date"])
+    di = _date_i(dd)
+    dd = dd[(di >= lo) & (di <= lo + 60)]
+    ca = _read(root, sf, "customer_address", ["ca_address_sk", "ca_state"])
+    ca = ca[ca.ca_state == state]
+    j = _merge(fs, dd, f"{pre}_ship_date_sk", "d_date_sk")
+    j = _merge(j, ca, f"{pre}_ship_addr_sk", "ca_address_sk")
+    if site:
+        fk, table, tkey, scol, sval = site
+        stab = _read(root, sf, table)
+        stab = stab[stab[scol] == sval]
+        j = _merge(j, stab, fk, tkey)
+    allf = _read(root, sf, fact, [f"{pre}_order_number", f"{pre}_warehouse_sk"])
+    wh_per_order = allf.dropna().drop_duplicates().groupby(f"{pre}_order_number").size()
+    multi = set(wh_per_order[wh_per_order > 1].index)
+    j = j[j[f"{pre}_order_number"].isin(multi)]
+    ret = _read(root, sf, rets, [f"{rpre}_order_number"])
+    j = j[~j[f"{pre}_order_number"].isin(set(ret[f"{rpre}_order_number"].dropna()))]
+    return pd.DataFrame({
+        "order_count": [j[f"{pre}_order_number"].nunique()],
+        "total_shipping_cost": [j[f"{pre}_ext_ship_cost"].sum(min_count=1)],
+        "total_net_profit": [j[f"{pre}_net_profit"].sum(min_count=1)],
+    })
+
+
+def q16(root, sf):
+    return _ship_oracle(root, sf, "catalog_sales", "cs", (2002, 2, 1), "GA",
+                        ("cs_call_center_sk", "call_center", "cc_call_center_sk",
+                         "cc_county", "Williamson County"), "catalog_returns", "cr")
+
+
+def q94(root, sf):
+    return _ship_oracle(root, sf, "web_sales", "ws", (1999, 2, 1), "IL",
+                        ("ws_web_site_sk", "web_site", "web_site_sk",
+                         "web_name", "site_1"), "web_returns", "wr")
+
+
+def _discount_oracle(root, sf, fact, pre, mfg, date0):
+    import pandas as pd
+
+    lo = _days(*date0)
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_date"])
+    di = _date_i(dd)
+    dd = dd[(di >= lo) & (di <= lo + 90)]
+    fs = _read(root, sf, fact, [f"{pre}_item_sk", f"{pre}_sold_date_sk",
+                                f"{pre}_ext_discount_amt"])
+    win = _merge(fs, dd, f"{pre}_sold_date_sk", "d_date_sk")
+    av = win.groupby(f"{pre}_item_sk", dropna=False)[f"{pre}_ext_discount_amt"] \
+            .mean().reset_index(name="av")
+    it = _read(root, sf, "item", ["i_item_sk", "i_manufact_id"])
+    it = it[it.i_manufact_id == mfg]
+    j = _merge(win, it, f"{pre}_item_sk", "i_item_sk")
+    j = _merge(j, av.rename(columns={f"{pre}_item_sk": "av_item"}), f"{pre}_item_sk", "av_item")
+    f = j[j[f"{pre}_ext_discount_amt"] > 1.3 * j.av]
+    v = f[f"{pre}_ext_discount_amt"].sum(min_count=1)
+    return pd.DataFrame({"excess": [None if pd.isna(v) else v]})
+
+
+def q32(root, sf):
+    return _discount_oracle(root, sf, "catalog_sales", "cs", 269, (1998, 3, 18))
+
+
+def q92(root, sf):
+    return _discount_oracle(root, sf, "web_sales", "ws", 350, (2000, 1, 27))
+
+
+def q40(root, sf):
+    import numpy as np
+
+    pivot = _days(2000, 3, 11)
+    cs = _read(root, sf, "catalog_sales", ["cs_order_number", "cs_item_sk",
+                                           "cs_warehouse_sk", "cs_sold_date_sk",
+                                           "cs_sales_price"])
+    cr = _read(root, sf, "catalog_returns", ["cr_order_number", "cr_item_sk",
+                                             "cr_refunded_cash"])
+    j = cs.merge(cr.dropna(subset=["cr_order_number", "cr_item_sk"]),
+                 left_on=["cs_order_number", "cs_item_sk"],
+                 right_on=["cr_order_number", "cr_item_sk"], how="left")
+    wh = _read(root, sf, "warehouse", ["w_warehouse_sk", "w_state"])
+    it = _read(root, sf, "item", ["i_item_sk", "i_item_id", "i_current_price"])
+    it = it[it.i_current_price.between(0.99, 1.49)]
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_date"])
+    di = _date_i(dd)
+    dd = dd.assign(d_i=di)
+    dd = dd[(di >= pivot - 30) & (di <= pivot + 30)]
+    j = _merge(j, wh, "cs_warehouse_sk", "w_warehouse_sk")
+    j = _merge(j, it, "cs_item_sk", "i_item_sk")
+    j = _merge(j, dd, "cs_sold_date_sk", "d_date_sk")
+    net = j.cs_sales_price - j.cr_refunded_cash.fillna(0.0)
+    j = j.assign(b=np.where(j.d_i < pivot, net, 0.0), a=np.where(j.d_i >= pivot, net, 0.0))
+    j.loc[j.cs_sales_price.isna(), ["b", "a"]] = np.nan
+    g = j.groupby(["w_state", "i_item_id"], dropna=False).agg(
+        sales_before=("b", lambda x: x.sum(min_count=1)),
+        sales_after=("a", lambda x: x.sum(min_count=1))).reset_index()
+    g = g.sort_values(["w_state", "i_item_id"], na_position="first").head(100)
+    return g.reset_index(drop=True)
+
+
+def q45(root, sf):
+    it_all = _read(root, sf, "item", ["i_item_sk", "i_item_id"])
+    ids = set(it_all[it_all.i_item_sk.isin([2, 3, 5, 7, 11, 13, 17, 19, 23, 29])].i_item_id)
+    ws = _read(root, sf, "web_sales", ["ws_bill_customer_sk", "ws_item_sk",
+                                       "ws_sold_date_sk", "ws_sales_price"])
+    cust = _read(root, sf, "customer", ["c_customer_sk", "c_current_addr_sk"])
+    ca = _read(root, sf, "customer_address", ["ca_address_sk", "ca_zip", "ca_city"])
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_qoy", "d_year"])
+    dd = dd[(dd.d_qoy == 2) & (dd.d_year == 2001)]
+    j = _merge(ws, cust, "ws_bill_customer_sk", "c_customer_sk")
+    j = _merge(j, ca, "c_current_addr_sk", "ca_address_sk")
+    j = _merge(j, dd, "ws_sold_date_sk", "d_date_sk")
+    j = _merge(j, it_all, "ws_item_sk", "i_item_sk")
+    zips = ["85669", "86197", "88274", "83405", "86475", "85392", "85460", "80348", "81792"]
+    m = j.ca_zip.str[:5].isin(zips) | j.i_item_id.isin(ids)
+    f = j[m.fillna(False)]
+    g = f.groupby(["ca_zip", "ca_city"], dropna=False).ws_sales_price \
+         .sum(min_count=1).reset_index(name="s")
+    g = g.sort_values(["ca_zip", "ca_city"], na_position="first").head(100)
+    return g.reset_index(drop=True)
+
+
+def q50(root, sf):
+    import numpy as np
+
+    ss = _read(root, sf, "store_sales", ["ss_ticket_number", "ss_item_sk", "ss_customer_sk",
+                                         "ss_sold_date_sk", "ss_store_sk"])
+    sr = _read(root, sf, "store_returns", ["sr_ticket_number", "sr_item_sk", "sr_customer_sk",
+                                           "sr_returned_date_sk"])
+    d2 = _read(root, sf, "date_dim", ["d_date_sk", "d_year", "d_moy"])
+    d2 = d2[(d2.d_year == 2001) & (d2.d_moy == 8)]
+    srj = _merge(sr, d2, "sr_returned_date_sk", "d_date_sk")
+    j = ss.dropna(subset=["ss_ticket_number", "ss_item_sk", "ss_customer_sk"]).merge(
+        srj.dropna(subset=["sr_ticket_number", "sr_item_sk", "sr_customer_sk"]),
+        left_on=["ss_ticket_number", "ss_item_sk", "ss_customer_sk"],
+        right_on=["sr_ticket_number", "sr_item_sk", "sr_customer_sk"])
+    st_cols = ["s_store_name", "s_company_id", "s_street_name", "s_city",
+               "s_county", "s_state", "s_zip"]
+    st = _read(root, sf, "store", ["s_store_sk"] + st_cols)
+    j = _merge(j, st, "ss_store_sk", "s_store_sk")
+    lag = j.sr_returned_date_sk - j.ss_sold_date_sk
+    j = j.assign(
+        d30=np.where(lag.notna(), (lag <= 30).astype(float), np.nan),
+        d31_60=np.where(lag.notna(), ((lag > 30) & (lag <= 60)).astype(float), np.nan),
+        d61_90=np.where(lag.notna(), ((lag > 60) & (lag <= 90)).astype(float), np.nan),
+        d91_120=np.where(lag.notna(), ((lag > 90) & (lag <= 120)).astype(float), np.nan),
+        d120p=np.where(lag.notna(), (lag > 120).astype(float), np.nan))
+    cols = ["d30", "d31_60", "d61_90", "d91_120", "d120p"]
+    g = j.groupby(st_cols, dropna=False)[cols].sum(min_count=1).reset_index()
+    for c in cols:
+        g[c] = g[c].astype("Int64")
+    g = g.sort_values(st_cols, na_position="first").head(100)
+    return g.reset_index(drop=True)
+
+
+ORACLES.update({"q16": q16, "q32": q32, "q40": q40, "q45": q45, "q50": q50,
+                "q92": q92, "q94": q94})

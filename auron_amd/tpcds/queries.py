@@ -1031,3 +1031,177 @@ QUERIES.update({
     "q53": q53, "q62": q62, "q63": q63, "q82": q82, "q89": q89, "q90": q90,
     "q91": q91, "q93": q93, "q98": q98,
 })
+
+
+# ------------------------------- batch 4: exists/threshold/buckets
+def _global_agg(child, aggs):
+    """single exchange -> complete agg (for small filtered inputs needing
+    count_distinct alongside sums)."""
+    return P.HashAgg(P.Exchange(child, "single"), [], aggs, mode="complete")
+
+
+def _ship_q(cat, s, fact, pre, date0, state, site_join, returns_table, ret_pre):
+    lo = _days(*date0)
+    fs = cat.scan(fact, [f"{pre}_ship_date_sk", f"{pre}_ship_addr_sk", f"{pre}_order_number",
+                         f"{pre}_warehouse_sk", f"{pre}_ext_ship_cost", f"{pre}_net_profit"]
+                  + ([site_join[0]] if site_join else []))
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_date"]),
+                  col("d_date").cast(dtypes.int32).between(lo, lo + 60))
+    ca = P.Filter(cat.scan("customer_address", ["ca_address_sk", "ca_state"]),
+                  col("ca_state") == lit(state))
+    j = bhj(fs, dd, [f"{pre}_ship_date_sk"], ["d_date_sk"])
+    j = bhj(j, ca, [f"{pre}_ship_addr_sk"], ["ca_address_sk"])
+    if site_join:
+        fk, table, tkey, pred = site_join
+        site = P.Filter(cat.scan(table, [tkey] + pred[0]), pred[1])
+        j = bhj(j, site, [fk], [tkey])
+    # EXISTS (same order, different warehouse): orders with >1 distinct wh
+    all_orders = cat.scan(fact, [f"{pre}_order_number", f"{pre}_warehouse_sk"])
+    ord_wh = agg2(P.Project(all_orders, [_a(col(f"{pre}_order_number"), "o"),
+                                         _a(col(f"{pre}_warehouse_sk"), "w")]),
+                  ["o", "w"], [])
+    multi = P.Filter(
+        P.HashAgg(ord_wh, [_a(col("o"), "o")],
+                  [AggFunc("count_star", None, name="nwh")], mode="complete"),
+        col("nwh") > 1)
+    j = P.HashJoin(P.Exchange(j, "hash", [col(f"{pre}_order_number")]),
+                   P.Exchange(multi, "hash", [col("o")]),
+                   [col(f"{pre}_order_number")], [col("o")], how="semi", build_side="right")
+    # NOT EXISTS returns
+    ret = cat.scan(returns_table, [f"{ret_pre}_order_number"])
+    j = P.HashJoin(j, P.Exchange(ret, "hash", [col(f"{ret_pre}_order_number")]),
+                   [col(f"{pre}_order_number")], [col(f"{ret_pre}_order_number")],
+                   how="anti", build_side="right")
+    return _global_agg(j, [
+        AggFunc("count_distinct", col(f"{pre}_order_number"), name="order_count"),
+        AggFunc("sum", col(f"{pre}_ext_ship_cost"), name="total_shipping_cost"),
+        AggFunc("sum", col(f"{pre}_net_profit"), name="total_net_profit")])
+
+
+def q16(cat, s):
+    return _ship_q(cat, s, "catalog_sales", "cs", (2002, 2, 1), "GA",
+                   ("cs_call_center_sk", "call_center", "cc_call_center_sk",
+                    (["cc_county"], col("cc_county") == lit("Williamson County"))),
+                   "catalog_returns", "cr")
+
+
+def q94(cat, s):
+    # adapted literal: web_company_name -> web_name (synthetic catalog)
+    return _ship_q(cat, s, "web_sales", "ws", (1999, 2, 1), "IL",
+                   ("ws_web_site_sk", "web_site", "web_site_sk",
+                    (["web_name"], col("web_name") == lit("site_1"))),
+                   "web_returns", "wr")
+
+
+def _discount_q(cat, s, fact, pre, mfg, date0):
+    lo = _days(*date0)
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_date"]),
+                  col("d_date").cast(dtypes.int32).between(lo, lo + 90))
+    fs = cat.scan(fact, [f"{pre}_item_sk", f"{pre}_sold_date_sk", f"{pre}_ext_discount_amt"])
+    win = bhj(fs, dd, [f"{pre}_sold_date_sk"], ["d_date_sk"])
+    avg_by_item = agg2(P.Project(win, [_a(col(f"{pre}_item_sk"), "av_item"),
+                                       _a(col(f"{pre}_ext_discount_amt"), "amt")]),
+                       ["av_item"], [AggFunc("avg", col("amt"), name="av")])
+    it = P.Filter(cat.scan("item", ["i_item_sk", "i_manufact_id"]),
+                  col("i_manufact_id") == mfg)
+    j = bhj(win, it, [f"{pre}_item_sk"], ["i_item_sk"])
+    j = P.HashJoin(P.Exchange(j, "hash", [col(f"{pre}_item_sk")]),
+                   avg_by_item, [col(f"{pre}_item_sk")], [col("av_item")],
+                   how="inner", build_side="right", broadcast=False)
+    f = P.Filter(j, col(f"{pre}_ext_discount_amt") > lit(1.3) * col("av"))
+    partial = P.HashAgg(f, [], [AggFunc("sum", col(f"{pre}_ext_discount_amt"),
+                                        name="excess")], mode="partial")
+    return P.HashAgg(P.Exchange(partial, "single"), [],
+                     [AggFunc("sum", col(f"{pre}_ext_discount_amt"), name="excess")],
+                     mode="final")
+
+
+def q32(cat, s):
+    return _discount_q(cat, s, "catalog_sales", "cs", 269, (1998, 3, 18))
+
+
+def q92(cat, s):
+    return _discount_q(cat, s, "web_sales", "ws", 350, (2000, 1, 27))
+
+
+def q40(cat, s):
+    from ..exprs import CaseWhen, Coalesce, Literal
+
+    pivot = _days(2000, 3, 11)
+    cs = cat.scan("catalog_sales", ["cs_order_number", "cs_item_sk", "cs_warehouse_sk",
+                                    "cs_sold_date_sk", "cs_sales_price"])
+    cr = cat.scan("catalog_returns", ["cr_order_number", "cr_item_sk", "cr_refunded_cash"])
+    j = shj(cs, cr, ["cs_order_number", "cs_item_sk"],
+            ["cr_order_number", "cr_item_sk"], how="left")
+    wh = cat.scan("warehouse", ["w_warehouse_sk", "w_state"])
+    it = P.Filter(cat.scan("item", ["i_item_sk", "i_item_id", "i_current_price"]),
+                  col("i_current_price").between(0.99, 1.49))
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_date"]),
+                  col("d_date").cast(dtypes.int32).between(pivot - 30, pivot + 30))
+    j = bhj(j, wh, ["cs_warehouse_sk"], ["w_warehouse_sk"])
+    j = bhj(j, it, ["cs_item_sk"], ["i_item_sk"])
+    j = bhj(j, dd, ["cs_sold_date_sk"], ["d_date_sk"])
+    net = col("cs_sales_price") - Coalesce([col("cr_refunded_cash"), lit(0.0)])
+    before = CaseWhen([(col("d_date").cast(dtypes.int32) < pivot, net)], lit(0.0))
+    after = CaseWhen([(col("d_date").cast(dtypes.int32) >= pivot, net)], lit(0.0))
+    pre = P.Project(j, [_a(col("w_state"), "w_state"), _a(col("i_item_id"), "i_item_id"),
+                        _a(before, "b"), _a(after, "a")])
+    agg = agg2(pre, ["w_state", "i_item_id"],
+               [AggFunc("sum", col("b"), name="sales_before"),
+                AggFunc("sum", col("a"), name="sales_after")])
+    return topk(agg, [(col("w_state"), True), (col("i_item_id"), True)], 100)
+
+
+def q45(cat, s):
+    from ..exprs import Substr
+
+    item_ids = s.collect_all(P.Project(
+        P.Filter(cat.scan("item", ["i_item_sk", "i_item_id"]),
+                 col("i_item_sk").isin([2, 3, 5, 7, 11, 13, 17, 19, 23, 29])),
+        [_a(col("i_item_id"), "iid")])).to_pydict()["iid"]
+    ws = cat.scan("web_sales", ["ws_bill_customer_sk", "ws_item_sk",
+                                "ws_sold_date_sk", "ws_sales_price"])
+    cust = cat.scan("customer", ["c_customer_sk", "c_current_addr_sk"])
+    ca = cat.scan("customer_address", ["ca_address_sk", "ca_zip", "ca_city"])
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_qoy", "d_year"]),
+                  (col("d_qoy") == 2) & (col("d_year") == 2001))
+    it = cat.scan("item", ["i_item_sk", "i_item_id"])
+    j = bhj(ws, cust, ["ws_bill_customer_sk"], ["c_customer_sk"])
+    j = bhj(j, ca, ["c_current_addr_sk"], ["ca_address_sk"])
+    j = bhj(j, dd, ["ws_sold_date_sk"], ["d_date_sk"])
+    j = bhj(j, it, ["ws_item_sk"], ["i_item_sk"])
+    zips = ["85669", "86197", "88274", "83405", "86475", "85392", "85460", "80348", "81792"]
+    f = P.Filter(j, Substr(col("ca_zip"), 1, 5).isin(zips)
+                 | col("i_item_id").isin(item_ids))
+    a = agg2(f, ["ca_zip", "ca_city"], [AggFunc("sum", col("ws_sales_price"), name="s")])
+    return topk(a, [(col("ca_zip"), True), (col("ca_city"), True)], 100)
+
+
+def q50(cat, s):
+    from ..exprs import CaseWhen
+
+    ss = cat.scan("store_sales", ["ss_ticket_number", "ss_item_sk", "ss_customer_sk",
+                                  "ss_sold_date_sk", "ss_store_sk"])
+    sr = cat.scan("store_returns", ["sr_ticket_number", "sr_item_sk", "sr_customer_sk",
+                                    "sr_returned_date_sk"])
+    d2 = P.Project(P.Filter(cat.scan("date_dim", ["d_date_sk", "d_year", "d_moy"]),
+                            (col("d_year") == 2001) & (col("d_moy") == 8)),
+                   [_a(col("d_date_sk"), "d2_sk")])
+    srj = bhj(sr, d2, ["sr_returned_date_sk"], ["d2_sk"])
+    j = shj(ss, srj, ["ss_ticket_number", "ss_item_sk", "ss_customer_sk"],
+            ["sr_ticket_number", "sr_item_sk", "sr_customer_sk"])
+    st_cols = ["s_store_name", "s_company_id", "s_street_name", "s_city",
+               "s_county", "s_state", "s_zip"]
+    st = cat.scan("store", ["s_store_sk"] + st_cols)
+    j = bhj(j, st, ["ss_store_sk"], ["s_store_sk"])
+    lag = col("sr_returned_date_sk") - col("ss_sold_date_sk")
+    buckets = [("d30", lag <= 30), ("d31_60", (lag > 30) & (lag <= 60)),
+               ("d61_90", (lag > 60) & (lag <= 90)),
+               ("d91_120", (lag > 90) & (lag <= 120)), ("d120p", lag > 120)]
+    aggs = [AggFunc("sum", CaseWhen([(c, lit(1))], lit(0)), name=n) for n, c in buckets]
+    a = agg2(j, st_cols, aggs)
+    return topk(a, [(col(c), True) for c in st_cols], 100)
+
+
+QUERIES.update({"q16": q16, "q32": q32, "q40": q40, "q45": q45, "q50": q50,
+                "q92": q92, "q94": q94})

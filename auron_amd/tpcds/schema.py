@@ -113,7 +113,7 @@ SCHEMAS = {
     "web_sales": {
         "ws_sold_date_sk": i64, "ws_sold_time_sk": i64, "ws_ship_date_sk": i64,
         "ws_item_sk": i64, "ws_bill_customer_sk": i64, "ws_bill_cdemo_sk": i64,
-        "ws_bill_hdemo_sk": i64, "ws_bill_addr_sk": i64, "ws_ship_customer_sk": i64,
+        "ws_bill_hdemo_sk": i64, "ws_bill_addr_sk": i64, "ws_ship_customer_sk": i64, "ws_ship_addr_sk": i64,
         "ws_web_page_sk": i64, "ws_web_site_sk": i64, "ws_ship_mode_sk": i64,
         "ws_warehouse_sk": i64, "ws_promo_sk": i64, "ws_order_number": i64,
         "ws_quantity": i32, "ws_wholesale_cost": f64, "ws_list_price": f64,
