@@ -1001,3 +1001,205 @@ def q50(root, sf):
 
 ORACLES.update({"q16": q16, "q32": q32, "q40": q40, "q45": q45, "q50": q50,
                 "q92": q92, "q94": q94})
+
+
+# ------------------------------- batch 5 oracles
+def _channel_attr_sum_oracle(root, sf, fact, pre, addr_fk, attr, cats, year, moy):
+    fs = _read(root, sf, fact, [f"{pre}_item_sk", f"{pre}_sold_date_sk", addr_fk,
+                                f"{pre}_ext_sales_price"])
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year", "d_moy"])
+    dd = dd[(dd.d_year == year) & (dd.d_moy == moy)]
+    ca = _read(root, sf, "customer_address", ["ca_address_sk", "ca_gmt_offset"])
+    ca = ca[ca.ca_gmt_offset == -5.0]
+    it = _read(root, sf, "item", ["i_item_sk", attr, "i_category"])
+    sub_vals = set(it[it.i_category.isin(cats)][attr].dropna())
+    it_f = it[it[attr].isin(sub_vals)]
+    j = _merge(fs, dd, f"{pre}_sold_date_sk", "d_date_sk")
+    j = _merge(j, ca, addr_fk, "ca_address_sk")
+    j = _merge(j, it_f, f"{pre}_item_sk", "i_item_sk")
+    return j.groupby(attr, dropna=False)[f"{pre}_ext_sales_price"] \
+            .sum(min_count=1).reset_index(name="total_sales")
+
+
+def q33(root, sf):
+    import pandas as pd
+
+    parts = [_channel_attr_sum_oracle(root, sf, f, p, a, "i_manufact_id",
+                                      ["Electronics"], 1998, 5)
+             for f, p, a in [("store_sales", "ss", "ss_addr_sk"),
+                             ("catalog_sales", "cs", "cs_bill_addr_sk"),
+                             ("web_sales", "ws", "ws_bill_addr_sk")]]
+    u = pd.concat(parts, ignore_index=True)
+    g = u.groupby("i_manufact_id", dropna=False).total_sales.sum(min_count=1) \
+         .reset_index(name="total_sales")
+    g = g.sort_values("total_sales", na_position="first").head(100)
+    return g[["i_manufact_id", "total_sales"]].reset_index(drop=True)
+
+
+def q60(root, sf):
+    import pandas as pd
+
+    parts = [_channel_attr_sum_oracle(root, sf, f, p, a, "i_item_id",
+                                      ["Music"], 1998, 9)
+             for f, p, a in [("store_sales", "ss", "ss_addr_sk"),
+                             ("catalog_sales", "cs", "cs_bill_addr_sk"),
+                             ("web_sales", "ws", "ws_bill_addr_sk")]]
+    u = pd.concat(parts, ignore_index=True)
+    g = u.groupby("i_item_id", dropna=False).total_sales.sum(min_count=1) \
+         .reset_index(name="total_sales")
+    g = g.sort_values(["i_item_id", "total_sales"], na_position="first").head(100)
+    return g.reset_index(drop=True)
+
+
+def _distinct_cd_oracle(root, sf, fact, pre, cust_fk):
+    fs = _read(root, sf, fact, [f"{pre}_sold_date_sk", cust_fk])
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_date", "d_month_seq"])
+    dd = dd[dd.d_month_seq.between(1200, 1211)]
+    cust = _read(root, sf, "customer", ["c_customer_sk", "c_last_name", "c_first_name"])
+    j = _merge(fs, dd, f"{pre}_sold_date_sk", "d_date_sk")
+    j = _merge(j, cust, cust_fk, "c_customer_sk")
+    return j[["c_last_name", "c_first_name", "d_date"]].drop_duplicates()
+
+
+def q38(root, sf):
+    import pandas as pd
+
+    a = _distinct_cd_oracle(root, sf, "store_sales", "ss", "ss_customer_sk")
+    b = _distinct_cd_oracle(root, sf, "catalog_sales", "cs", "cs_bill_customer_sk")
+    c = _distinct_cd_oracle(root, sf, "web_sales", "ws", "ws_bill_customer_sk")
+    keys = ["c_last_name", "c_first_name", "d_date"]
+    ab = a.merge(b, on=keys)
+    abc = ab.merge(c, on=keys)
+    return pd.DataFrame({"cnt": [len(abc)]})
+
+
+def q87(root, sf):
+    import pandas as pd
+
+    a = _distinct_cd_oracle(root, sf, "store_sales", "ss", "ss_customer_sk")
+    b = _distinct_cd_oracle(root, sf, "catalog_sales", "cs", "cs_bill_customer_sk")
+    c = _distinct_cd_oracle(root, sf, "web_sales", "ws", "ws_bill_customer_sk")
+    keys = ["c_last_name", "c_first_name", "d_date"]
+    bk = set(map(tuple, b[keys].itertuples(index=False)))
+    ck = set(map(tuple, c[keys].itertuples(index=False)))
+    rows = [t for t in map(tuple, a[keys].itertuples(index=False))
+            if t not in bk and t not in ck]
+    return pd.DataFrame({"cnt": [len(rows)]})
+
+
+def q11(root, sf):
+    def year_total(fact, pre, cust_fk, lp, da):
+        fs = _read(root, sf, fact, [cust_fk, f"{pre}_sold_date_sk", lp, da])
+        dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year"])
+        dd = dd[dd.d_year.isin([2001, 2002])]
+        cust = _read(root, sf, "customer", ["c_customer_sk", "c_customer_id",
+                                            "c_first_name", "c_last_name",
+                                            "c_preferred_cust_flag"])
+        j = _merge(fs, dd, f"{pre}_sold_date_sk", "d_date_sk")
+        j = _merge(j, cust, cust_fk, "c_customer_sk")
+        j["v"] = j[lp] - j[da]
+        g = j.groupby(["c_customer_id", "c_first_name", "c_last_name",
+                       "c_preferred_cust_flag", "d_year"], dropna=False) \
+             .v.sum(min_count=1).reset_index(name="year_total")
+        return g
+
+    ss = year_total("store_sales", "ss", "ss_customer_sk",
+                    "ss_ext_list_price", "ss_ext_discount_amt")
+    ws = year_total("web_sales", "ws", "ws_bill_customer_sk",
+                    "ws_ext_list_price", "ws_ext_discount_amt")
+    s1 = ss[(ss.d_year == 2001) & (ss.year_total > 0)]
+    s2 = ss[ss.d_year == 2002]
+    w1 = ws[(ws.d_year == 2001) & (ws.year_total > 0)]
+    w2 = ws[ws.d_year == 2002]
+    j = s1.merge(s2, on="c_customer_id", suffixes=("_sf", "_ssec"))
+    j = j.merge(w1[["c_customer_id", "year_total"]].rename(columns={"year_total": "wf"}),
+                on="c_customer_id")
+    j = j.merge(w2[["c_customer_id", "year_total"]].rename(columns={"year_total": "wsec"}),
+                on="c_customer_id")
+    f = j[(j.wsec / j.wf) > (j.year_total_ssec / j.year_total_sf)]
+    out = f[["c_preferred_cust_flag_ssec"]].rename(
+        columns={"c_preferred_cust_flag_ssec": "customer_preferred_cust_flag"})
+    out = out.sort_values("customer_preferred_cust_flag", na_position="first").head(100)
+    return out.reset_index(drop=True)
+
+
+def q31(root, sf):
+    def county(fact, pre, addr_fk):
+        fs = _read(root, sf, fact, [f"{pre}_sold_date_sk", addr_fk, f"{pre}_ext_sales_price"])
+        dd = _read(root, sf, "date_dim", ["d_date_sk", "d_qoy", "d_year"])
+        dd = dd[(dd.d_year == 2000) & dd.d_qoy.isin([1, 2, 3])]
+        ca = _read(root, sf, "customer_address", ["ca_address_sk", "ca_county"])
+        j = _merge(fs, dd, f"{pre}_sold_date_sk", "d_date_sk")
+        j = _merge(j, ca, addr_fk, "ca_address_sk")
+        return j.groupby(["ca_county", "d_qoy"], dropna=False)[f"{pre}_ext_sales_price"] \
+                .sum(min_count=1).reset_index(name="v")
+
+    ssx = county("store_sales", "ss", "ss_addr_sk")
+    wsx = county("web_sales", "ws", "ws_bill_addr_sk")
+
+    def inst(df, q, name):
+        d = df[df.d_qoy == q][["ca_county", "v"]].rename(columns={"v": name})
+        return d.dropna(subset=["ca_county"])
+
+    j = inst(ssx, 1, "ss1").merge(inst(ssx, 2, "ss2"), on="ca_county") \
+        .merge(inst(ssx, 3, "ss3"), on="ca_county") \
+        .merge(inst(wsx, 1, "ws1"), on="ca_county") \
+        .merge(inst(wsx, 2, "ws2"), on="ca_county") \
+        .merge(inst(wsx, 3, "ws3"), on="ca_county")
+    import numpy as np
+
+    wr1 = np.where(j.ws1 > 0, j.ws2 / j.ws1, np.nan)
+    sr1 = np.where(j.ss1 > 0, j.ss2 / j.ss1, np.nan)
+    wr2 = np.where(j.ws2 > 0, j.ws3 / j.ws2, np.nan)
+    sr2 = np.where(j.ss2 > 0, j.ss3 / j.ss2, np.nan)
+    m = (wr1 > sr1) & (wr2 > sr2)
+    f = j[m].copy()
+    f["d_year"] = 2000
+    f["web_q1_q2_increase"] = f.ws2 / f.ws1
+    f["store_q1_q2_increase"] = f.ss2 / f.ss1
+    f["web_q2_q3_increase"] = f.ws3 / f.ws2
+    f["store_q2_q3_increase"] = f.ss3 / f.ss2
+    out = f[["ca_county", "d_year", "web_q1_q2_increase", "store_q1_q2_increase",
+             "web_q2_q3_increase", "store_q2_q3_increase"]]
+    return out.sort_values("ca_county").reset_index(drop=True)
+
+
+def q23(root, sf):
+    import pandas as pd
+
+    ss = _read(root, sf, "store_sales", ["ss_sold_date_sk", "ss_item_sk",
+                                         "ss_customer_sk", "ss_quantity", "ss_sales_price"])
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_date", "d_year", "d_moy"])
+    dd4 = dd[dd.d_year.isin([2000, 2001, 2002, 2003])]
+    it = _read(root, sf, "item", ["i_item_sk", "i_item_desc"])
+    j = _merge(ss, dd4, "ss_sold_date_sk", "d_date_sk")
+    j = _merge(j, it, "ss_item_sk", "i_item_sk")
+    j["itemdesc"] = j.i_item_desc.str[:30]
+    freq = j.groupby(["itemdesc", "ss_item_sk", "d_date"], dropna=False) \
+            .size().reset_index(name="cnt")
+    freq_items = set(freq[freq.cnt > 4].ss_item_sk.dropna())
+    cs_cust = _merge(ss, dd4, "ss_sold_date_sk", "d_date_sk")
+    cs_cust = cs_cust.assign(v=cs_cust.ss_quantity * cs_cust.ss_sales_price)
+    csales = cs_cust.groupby("ss_customer_sk", dropna=False).v.sum(min_count=1)
+    cmax = csales.max()
+    ss_all = _read(root, sf, "store_sales", ["ss_customer_sk", "ss_quantity", "ss_sales_price"])
+    ss_all = ss_all.assign(v=ss_all.ss_quantity * ss_all.ss_sales_price)
+    ssales = ss_all.groupby("ss_customer_sk", dropna=False).v.sum(min_count=1)
+    best = set(ssales[ssales > 0.5 * cmax].index.dropna())
+    ddm = dd[(dd.d_year == 2000) & (dd.d_moy == 2)]
+
+    def channel(fact, pre, cust_fk):
+        fs = _read(root, sf, fact, [f"{pre}_sold_date_sk", cust_fk, f"{pre}_item_sk",
+                                    f"{pre}_quantity", f"{pre}_list_price"])
+        jj = _merge(fs, ddm, f"{pre}_sold_date_sk", "d_date_sk")
+        jj = jj[jj[f"{pre}_item_sk"].isin(freq_items)]
+        jj = jj[jj[cust_fk].isin(best)]
+        return (jj[f"{pre}_quantity"] * jj[f"{pre}_list_price"]).sum(min_count=1)
+
+    total = pd.Series([channel("catalog_sales", "cs", "cs_bill_customer_sk"),
+                       channel("web_sales", "ws", "ws_bill_customer_sk")]).sum(min_count=1)
+    return pd.DataFrame({"s": [None if pd.isna(total) else total]})
+
+
+ORACLES.update({"q11": q11, "q23": q23, "q31": q31, "q33": q33, "q38": q38,
+                "q60": q60, "q87": q87})

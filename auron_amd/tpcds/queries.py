@@ -1205,3 +1205,257 @@ def q50(cat, s):
 
 QUERIES.update({"q16": q16, "q32": q32, "q40": q40, "q45": q45, "q50": q50,
                 "q92": q92, "q94": q94})
+
+
+# ------------------------------- batch 5: big CTE queries
+def _channel_attr_sum(cat, s, fact, pre, addr_fk, attr, attr_src, it_filter_attr_vals,
+                      year, moy):
+    """q33/q60 channel CTE: sum ext_sales_price by an item attribute for
+    items whose attribute appears in a filtered item subset."""
+    fs = cat.scan(fact, [f"{pre}_item_sk", f"{pre}_sold_date_sk", addr_fk,
+                         f"{pre}_ext_sales_price"])
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_year", "d_moy"]),
+                  (col("d_year") == year) & (col("d_moy") == moy))
+    ca = P.Filter(cat.scan("customer_address", ["ca_address_sk", "ca_gmt_offset"]),
+                  col("ca_gmt_offset") == -5.0)
+    it = cat.scan("item", ["i_item_sk", attr_src])
+    sub = P.Filter(cat.scan("item", [attr_src, "i_category"]),
+                   col("i_category").isin(it_filter_attr_vals))
+    it_f = P.HashJoin(it, P.Project(sub, [_a(col(attr_src), "_sub_attr")]),
+                      [col(attr_src)], [col("_sub_attr")], how="semi",
+                      build_side="right", broadcast=True)
+    j = bhj(fs, dd, [f"{pre}_sold_date_sk"], ["d_date_sk"])
+    j = bhj(j, ca, [addr_fk], ["ca_address_sk"])
+    j = bhj(j, it_f, [f"{pre}_item_sk"], ["i_item_sk"])
+    return agg2(j, [attr], [AggFunc("sum", col(f"{pre}_ext_sales_price"),
+                                    name="total_sales")],
+                key_exprs=[col(attr_src)])
+
+
+def _union_channel_q(cat, s, attr, attr_src, categories, year, moy, order_by_attr):
+    u = P.Union([
+        _channel_attr_sum(cat, s, "store_sales", "ss", "ss_addr_sk", attr, attr_src,
+                          categories, year, moy),
+        _channel_attr_sum(cat, s, "catalog_sales", "cs", "cs_bill_addr_sk", attr, attr_src,
+                          categories, year, moy),
+        _channel_attr_sum(cat, s, "web_sales", "ws", "ws_bill_addr_sk", attr, attr_src,
+                          categories, year, moy),
+    ])
+    a = agg2(u, [attr], [AggFunc("sum", col("total_sales"), name="total_sales")])
+    keys = ([(col(attr), True), (col("total_sales"), True)] if order_by_attr
+            else [(col("total_sales"), True)])
+    return topk(a, keys, 100)
+
+
+def q33(cat, s):
+    return _union_channel_q(cat, s, "i_manufact_id", "i_manufact_id",
+                            ["Electronics"], 1998, 5, order_by_attr=False)
+
+
+def q60(cat, s):
+    return _union_channel_q(cat, s, "i_item_id", "i_item_id",
+                            ["Music"], 1998, 9, order_by_attr=True)
+
+
+def _distinct_cust_dates(cat, s, fact, pre, cust_fk):
+    fs = cat.scan(fact, [f"{pre}_sold_date_sk", cust_fk])
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_date", "d_month_seq"]),
+                  col("d_month_seq").between(1200, 1211))
+    cust = cat.scan("customer", ["c_customer_sk", "c_last_name", "c_first_name"])
+    j = bhj(fs, dd, [f"{pre}_sold_date_sk"], ["d_date_sk"])
+    j = bhj(j, cust, [cust_fk], ["c_customer_sk"])
+    return agg2(j, ["c_last_name", "c_first_name", "d_date"], [])
+
+
+def q38(cat, s):
+    a = _distinct_cust_dates(cat, s, "store_sales", "ss", "ss_customer_sk")
+    b = _distinct_cust_dates(cat, s, "catalog_sales", "cs", "cs_bill_customer_sk")
+    c = _distinct_cust_dates(cat, s, "web_sales", "ws", "ws_bill_customer_sk")
+    keys = ["c_last_name", "c_first_name", "d_date"]
+    ab = P.HashJoin(P.Exchange(a, "hash", [col(k) for k in keys]),
+                    P.Exchange(b, "hash", [col(k) for k in keys]),
+                    [col(k) for k in keys], [col(k) for k in keys],
+                    how="semi", build_side="right")
+    abc = P.HashJoin(ab, P.Exchange(c, "hash", [col(k) for k in keys]),
+                     [col(k) for k in keys], [col(k) for k in keys],
+                     how="semi", build_side="right")
+    partial = P.HashAgg(abc, [], [AggFunc("count_star", None, name="cnt")], mode="partial")
+    return P.HashAgg(P.Exchange(partial, "single"), [],
+                     [AggFunc("count_star", None, name="cnt")], mode="final")
+
+
+def q87(cat, s):
+    # EXCEPT chain: store minus catalog minus web (anti joins)
+    a = _distinct_cust_dates(cat, s, "store_sales", "ss", "ss_customer_sk")
+    b = _distinct_cust_dates(cat, s, "catalog_sales", "cs", "cs_bill_customer_sk")
+    c = _distinct_cust_dates(cat, s, "web_sales", "ws", "ws_bill_customer_sk")
+    keys = ["c_last_name", "c_first_name", "d_date"]
+    ab = P.HashJoin(P.Exchange(a, "hash", [col(k) for k in keys]),
+                    P.Exchange(b, "hash", [col(k) for k in keys]),
+                    [col(k) for k in keys], [col(k) for k in keys],
+                    how="anti", build_side="right")
+    abc = P.HashJoin(ab, P.Exchange(c, "hash", [col(k) for k in keys]),
+                     [col(k) for k in keys], [col(k) for k in keys],
+                     how="anti", build_side="right")
+    partial = P.HashAgg(abc, [], [AggFunc("count_star", None, name="cnt")], mode="partial")
+    return P.HashAgg(P.Exchange(partial, "single"), [],
+                     [AggFunc("count_star", None, name="cnt")], mode="final")
+
+
+def _year_total(cat, s, fact, pre, cust_fk, measure_expr, tag):
+    fs_cols = {"ss": ["ss_customer_sk", "ss_sold_date_sk", "ss_ext_list_price",
+                      "ss_ext_discount_amt", "ss_ext_wholesale_cost", "ss_ext_sales_price"],
+               "ws": ["ws_bill_customer_sk", "ws_sold_date_sk", "ws_ext_list_price",
+                      "ws_ext_discount_amt", "ws_ext_wholesale_cost", "ws_ext_sales_price"],
+               "cs": ["cs_bill_customer_sk", "cs_sold_date_sk", "cs_ext_list_price",
+                      "cs_ext_discount_amt", "cs_ext_wholesale_cost", "cs_ext_sales_price"]}[pre]
+    fs = cat.scan(fact, fs_cols)
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_year"]),
+                  col("d_year").isin([2001, 2002]))
+    cust = cat.scan("customer", ["c_customer_sk", "c_customer_id", "c_first_name",
+                                 "c_last_name", "c_preferred_cust_flag", "c_birth_country",
+                                 "c_email_address"])
+    j = bhj(fs, dd, [f"{pre}_sold_date_sk"], ["d_date_sk"])
+    j = bhj(j, cust, [cust_fk], ["c_customer_sk"])
+    a = agg2(j, ["customer_id", "customer_first_name", "customer_last_name",
+                 "customer_preferred_cust_flag", "dyear"],
+             [AggFunc("sum", measure_expr, name="year_total")],
+             key_exprs=[col("c_customer_id"), col("c_first_name"), col("c_last_name"),
+                        col("c_preferred_cust_flag"), col("d_year")])
+    return s.execute(a)
+
+
+def q11(cat, s):
+    ss_batches = _year_total(cat, s, "store_sales", "ss", "ss_customer_sk",
+                             col("ss_ext_list_price") - col("ss_ext_discount_amt"), "s")
+    ws_batches = _year_total(cat, s, "web_sales", "ws", "ws_bill_customer_sk",
+                             col("ws_ext_list_price") - col("ws_ext_discount_amt"), "w")
+
+    def inst(batches, year, prefix):
+        scan = P.MemoryScan(batches)
+        f = P.Filter(scan, col("dyear") == year)
+        cols = [("customer_id", f"{prefix}_id"), ("year_total", f"{prefix}_total"),
+                ("customer_preferred_cust_flag", f"{prefix}_flag")]
+        return P.Project(f, [_a(col(a), b) for a, b in cols])
+
+    s1 = inst(ss_batches, 2001, "sf")   # store first year
+    s2 = inst(ss_batches, 2002, "ssec")
+    w1 = inst(ws_batches, 2001, "wf")
+    w2 = inst(ws_batches, 2002, "wsec")
+    j = shj(P.Filter(s1, col("sf_total") > 0.0), s2, ["sf_id"], ["ssec_id"])
+    j = P.HashJoin(j, P.Exchange(P.Filter(w1, col("wf_total") > 0.0), "hash", [col("wf_id")]),
+                   [col("sf_id")], [col("wf_id")], how="inner", build_side="right")
+    j = P.HashJoin(j, P.Exchange(w2, "hash", [col("wsec_id")]),
+                   [col("sf_id")], [col("wsec_id")], how="inner", build_side="right")
+    f = P.Filter(j, (col("wsec_total") / col("wf_total")) > (col("ssec_total") / col("sf_total")))
+    proj = P.Project(f, [_a(col("ssec_flag"), "customer_preferred_cust_flag")])
+    return topk(proj, [(col("customer_preferred_cust_flag"), True)], 100)
+
+
+def q31(cat, s):
+    def county_q(fact, pre, addr_fk):
+        fs = cat.scan(fact, [f"{pre}_sold_date_sk", addr_fk, f"{pre}_ext_sales_price"])
+        dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_qoy", "d_year"]),
+                      (col("d_year") == 2000) & col("d_qoy").isin([1, 2, 3]))
+        ca = cat.scan("customer_address", ["ca_address_sk", "ca_county"])
+        j = bhj(fs, dd, [f"{pre}_sold_date_sk"], ["d_date_sk"])
+        j = bhj(j, ca, [addr_fk], ["ca_address_sk"])
+        return s.execute(agg2(j, ["ca_county", "d_qoy"],
+                              [AggFunc("sum", col(f"{pre}_ext_sales_price"), name="v")]))
+
+    ss_b = county_q("store_sales", "ss", "ss_addr_sk")
+    ws_b = county_q("web_sales", "ws", "ws_bill_addr_sk")
+
+    def inst(batches, q, name):
+        f = P.Filter(P.MemoryScan(batches), col("d_qoy") == q)
+        return P.Broadcast(P.Project(f, [_a(col("ca_county"), f"{name}_county"),
+                                         _a(col("v"), name)]))
+
+    base = inst(ss_b, 1, "ss1")
+    j = P.HashJoin(base, inst(ss_b, 2, "ss2"), [col("ss1_county")], [col("ss2_county")],
+                   how="inner", build_side="right")
+    j = P.HashJoin(j, inst(ss_b, 3, "ss3"), [col("ss1_county")], [col("ss3_county")],
+                   how="inner", build_side="right")
+    j = P.HashJoin(j, inst(ws_b, 1, "ws1"), [col("ss1_county")], [col("ws1_county")],
+                   how="inner", build_side="right")
+    j = P.HashJoin(j, inst(ws_b, 2, "ws2"), [col("ss1_county")], [col("ws2_county")],
+                   how="inner", build_side="right")
+    j = P.HashJoin(j, inst(ws_b, 3, "ws3"), [col("ss1_county")], [col("ws3_county")],
+                   how="inner", build_side="right")
+    from ..exprs import CaseWhen, Literal
+
+    wr1 = CaseWhen([(col("ws1") > 0.0, col("ws2") / col("ws1"))], Literal(None, dtypes.float64))
+    sr1 = CaseWhen([(col("ss1") > 0.0, col("ss2") / col("ss1"))], Literal(None, dtypes.float64))
+    wr2 = CaseWhen([(col("ws2") > 0.0, col("ws3") / col("ws2"))], Literal(None, dtypes.float64))
+    sr2 = CaseWhen([(col("ss2") > 0.0, col("ss3") / col("ss2"))], Literal(None, dtypes.float64))
+    f = P.Filter(j, (wr1 > sr1) & (wr2 > sr2))
+    proj = P.Project(f, [_a(col("ss1_county"), "ca_county"), _a(lit(2000), "d_year"),
+                         _a(col("ws2") / col("ws1"), "web_q1_q2_increase"),
+                         _a(col("ss2") / col("ss1"), "store_q1_q2_increase"),
+                         _a(col("ws3") / col("ws2"), "web_q2_q3_increase"),
+                         _a(col("ss3") / col("ss2"), "store_q2_q3_increase")])
+    return topk(proj, [(col("ca_county"), True)], 100000)
+
+
+def q23(cat, s):
+    from ..exprs import Substr
+
+    # frequent items: sold >4 times on one day over 4 years
+    ssc = cat.scan("store_sales", ["ss_sold_date_sk", "ss_item_sk"])
+    dd4 = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_date", "d_year"]),
+                   col("d_year").isin([2000, 2001, 2002, 2003]))
+    it = cat.scan("item", ["i_item_sk", "i_item_desc"])
+    j = bhj(ssc, dd4, ["ss_sold_date_sk"], ["d_date_sk"])
+    j = bhj(j, it, ["ss_item_sk"], ["i_item_sk"])
+    freq = P.Filter(
+        agg2(j, ["itemdesc", "item_sk", "solddate"],
+             [AggFunc("count_star", None, name="cnt")],
+             key_exprs=[Substr(col("i_item_desc"), 1, 30), col("ss_item_sk"), col("d_date")]),
+        col("cnt") > 4)
+    freq_items = P.HashAgg(P.Exchange(P.Project(freq, [_a(col("item_sk"), "item_sk")]),
+                                      "hash", [col("item_sk")]),
+                           [_a(col("item_sk"), "item_sk")], [], mode="complete")
+
+    # best customers: > 50% of the max per-customer sales
+    ss2 = cat.scan("store_sales", ["ss_customer_sk", "ss_sold_date_sk",
+                                   "ss_quantity", "ss_sales_price"])
+    j2 = bhj(ss2, dd4, ["ss_sold_date_sk"], ["d_date_sk"])
+    csales = agg2(P.Project(j2, [_a(col("ss_customer_sk"), "csk"),
+                                 _a(col("ss_quantity").cast(dtypes.float64) * col("ss_sales_price"), "v")]),
+                  ["csk"], [AggFunc("sum", col("v"), name="csales")])
+    csales_b = s.execute(csales)
+    tpcds_cmax = scalar(s, P.HashAgg(P.Exchange(P.MemoryScan(csales_b), "single"), [],
+                                     [AggFunc("max", col("csales"), name="m")],
+                                     mode="complete"))
+    ss3 = cat.scan("store_sales", ["ss_customer_sk", "ss_quantity", "ss_sales_price"])
+    ssales = agg2(P.Project(ss3, [_a(col("ss_customer_sk"), "csk"),
+                                  _a(col("ss_quantity").cast(dtypes.float64) * col("ss_sales_price"), "v")]),
+                  ["csk"], [AggFunc("sum", col("v"), name="ssales")])
+    best = P.Filter(ssales, col("ssales") > lit(0.5 * (tpcds_cmax or 0.0)))
+
+    dd_m = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_year", "d_moy"]),
+                    (col("d_year") == 2000) & (col("d_moy") == 2))
+
+    def channel(fact, pre, cust_fk):
+        fs = cat.scan(fact, [f"{pre}_sold_date_sk", cust_fk, f"{pre}_item_sk",
+                             f"{pre}_quantity", f"{pre}_list_price"])
+        jj = bhj(fs, dd_m, [f"{pre}_sold_date_sk"], ["d_date_sk"])
+        jj = P.HashJoin(P.Exchange(jj, "hash", [col(f"{pre}_item_sk")]),
+                        P.Exchange(freq_items, "hash", [col("item_sk")]),
+                        [col(f"{pre}_item_sk")], [col("item_sk")],
+                        how="semi", build_side="right")
+        jj = P.HashJoin(P.Exchange(jj, "hash", [col(cust_fk)]),
+                        P.Exchange(best, "hash", [col("csk")]),
+                        [col(cust_fk)], [col("csk")], how="semi", build_side="right")
+        return P.Project(jj, [_a(col(f"{pre}_quantity").cast(dtypes.float64)
+                                 * col(f"{pre}_list_price"), "sales")])
+
+    u = P.Union([channel("catalog_sales", "cs", "cs_bill_customer_sk"),
+                 channel("web_sales", "ws", "ws_bill_customer_sk")])
+    partial = P.HashAgg(u, [], [AggFunc("sum", col("sales"), name="s")], mode="partial")
+    return P.HashAgg(P.Exchange(partial, "single"), [],
+                     [AggFunc("sum", col("sales"), name="s")], mode="final")
+
+
+QUERIES.update({"q11": q11, "q23": q23, "q31": q31, "q33": q33, "q38": q38,
+                "q60": q60, "q87": q87})
