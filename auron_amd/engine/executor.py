@@ -86,7 +86,22 @@ class Executor:
         self._child_time = [0.0]
 
     # ------------------------------------------------------------- dispatch
+    def _rewrite(self, node: P.PlanNode) -> P.PlanNode:
+        """AQE-style local rewrites. At world_size==1 a
+        final-agg(exchange(partial-agg)) tower regroups the same rows
+        twice through an identity exchange — collapse it to one complete
+        agg (the AuronConvertStrategy removeInefficientConverts spirit)."""
+        if (self.ctx.world_size == 1 and isinstance(node, P.HashAgg)
+                and node.mode == "final" and isinstance(node.child, P.Exchange)
+                and isinstance(node.child.child, P.HashAgg)
+                and node.child.child.mode == "partial"
+                and len(node.keys) == len(node.child.child.keys)):
+            p = node.child.child
+            return P.HashAgg(p.child, p.keys, p.aggs, mode="complete")
+        return node
+
     def execute(self, node: P.PlanNode) -> List[RecordBatch]:
+        node = self._rewrite(node)
         name = type(node).__name__
         fn = getattr(self, f"_exec_{name}", None)
         if fn is None:
