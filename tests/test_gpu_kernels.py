@@ -148,3 +148,29 @@ def test_agg_avg_min_max_gpu_matches_cpu():
         a1, mn1, mx1 = r_cpu[g]
         a2, mn2, mx2 = r_gpu[g]
         assert abs(a1 - a2) < 1e-6 and mn1 == mn2 and mx1 == mx2
+
+
+def test_agg_scatter_native_vs_torch():
+    rng = np.random.default_rng(11)
+    n = 500_000
+    gids = torch.from_numpy(rng.integers(0, 1000, n)).to(torch.int64)
+    vals_np = rng.normal(100, 50, n)
+    validity = torch.from_numpy(rng.random(n) > 0.05)
+    col_cpu = Column(dtypes.float64, torch.from_numpy(vals_np), validity)
+    acc_ref, cnt_ref = ops.agg_scatter(gids, 1000, col_cpu, "sum")
+    col_gpu = col_cpu.to(DEV)
+    for fn in ("sum", "min", "max", "count"):
+        a_ref, c_ref = ops.agg_scatter(gids, 1000, col_cpu, fn)
+        a_gpu, c_gpu = ops.agg_scatter(gids.to(DEV), 1000, col_gpu, fn)
+        assert torch.equal(c_ref, c_gpu.cpu()), fn
+        if fn == "sum":
+            assert torch.allclose(a_ref, a_gpu.cpu(), rtol=1e-9, atol=1e-6), fn
+        elif fn != "count":
+            assert torch.equal(a_ref, a_gpu.cpu()), fn
+    # int64 path
+    icol_cpu = Column(dtypes.int64, torch.from_numpy(rng.integers(-10**6, 10**6, n)), validity)
+    icol_gpu = icol_cpu.to(DEV)
+    for fn in ("sum", "min", "max"):
+        a_ref, c_ref = ops.agg_scatter(gids, 1000, icol_cpu, fn)
+        a_gpu, c_gpu = ops.agg_scatter(gids.to(DEV), 1000, icol_gpu, fn)
+        assert torch.equal(a_ref, a_gpu.cpu()), fn
