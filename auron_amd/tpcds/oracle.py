@@ -1203,3 +1203,349 @@ def q23(root, sf):
 
 ORACLES.update({"q11": q11, "q23": q23, "q31": q31, "q33": q33, "q38": q38,
                 "q60": q60, "q87": q87})
+
+
+# ------------------------------- batch 6 oracles
+def _rollup2_oracle(u, k1, k2):
+    import pandas as pd
+
+    g2 = u.groupby([k1, k2], dropna=False)[["sales", "returns", "profit"]] \
+          .sum(min_count=1).reset_index()
+    g1 = u.groupby([k1], dropna=False)[["sales", "returns", "profit"]] \
+          .sum(min_count=1).reset_index()
+    g1[k2] = None
+    g0 = u[["sales", "returns", "profit"]].sum(min_count=1).to_frame().T
+    g0[k1] = None
+    g0[k2] = None
+    out = pd.concat([g2, g1, g0], ignore_index=True)
+    return out[[k1, k2, "sales", "returns", "profit"]]
+
+
+def q5(root, sf):
+    import pandas as pd
+
+    d0 = _days(2000, 8, 23)
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_date"])
+    di = _date_i(dd)
+    dd = dd[(di >= d0) & (di <= d0 + 14)]
+
+    def part(rows, dim, dim_key, id_col, tag):
+        j = _merge(rows, dd, "date_sk", "d_date_sk")
+        j = _merge(j, dim, "fk", dim_key)
+        a = j.groupby(id_col, dropna=False).agg(
+            sales=("sales_price", lambda x: x.sum(min_count=1)),
+            returns=("return_amt", lambda x: x.sum(min_count=1)),
+            p1=("profit", lambda x: x.sum(min_count=1)),
+            p2=("net_loss", lambda x: x.sum(min_count=1))).reset_index()
+        a["channel"] = f"{tag} channel"
+        a["id"] = tag.replace(" ", "_") + a[id_col].astype(str)
+        a["profit"] = a.p1 - a.p2
+        return a[["channel", "id", "sales", "returns", "profit"]]
+
+    def mk(df, fk, datec, sp=None, pr=None, ra=None, nl=None):
+        out = pd.DataFrame({
+            "fk": df[fk], "date_sk": df[datec],
+            "sales_price": df[sp] if sp else 0.0,
+            "profit": df[pr] if pr else 0.0,
+            "return_amt": df[ra] if ra else 0.0,
+            "net_loss": df[nl] if nl else 0.0})
+        return out
+
+    ss = _read(root, sf, "store_sales", ["ss_store_sk", "ss_sold_date_sk",
+                                         "ss_ext_sales_price", "ss_net_profit"])
+    sr = _read(root, sf, "store_returns", ["sr_store_sk", "sr_returned_date_sk",
+                                           "sr_return_amt", "sr_net_loss"])
+    st = _read(root, sf, "store", ["s_store_sk", "s_store_id"])
+    ssr = part(pd.concat([mk(ss, "ss_store_sk", "ss_sold_date_sk", sp="ss_ext_sales_price", pr="ss_net_profit"),
+                          mk(sr, "sr_store_sk", "sr_returned_date_sk", ra="sr_return_amt", nl="sr_net_loss")],
+                         ignore_index=True), st, "s_store_sk", "s_store_id", "store")
+    cs = _read(root, sf, "catalog_sales", ["cs_catalog_page_sk", "cs_sold_date_sk",
+                                           "cs_ext_sales_price", "cs_net_profit"])
+    cr = _read(root, sf, "catalog_returns", ["cr_catalog_page_sk", "cr_returned_date_sk",
+                                             "cr_return_amount", "cr_net_loss"])
+    cp = _read(root, sf, "catalog_page", ["cp_catalog_page_sk", "cp_catalog_page_id"])
+    csr = part(pd.concat([mk(cs, "cs_catalog_page_sk", "cs_sold_date_sk", sp="cs_ext_sales_price", pr="cs_net_profit"),
+                          mk(cr, "cr_catalog_page_sk", "cr_returned_date_sk", ra="cr_return_amount", nl="cr_net_loss")],
+                         ignore_index=True), cp, "cp_catalog_page_sk", "cp_catalog_page_id", "catalog page")
+    ws = _read(root, sf, "web_sales", ["ws_web_site_sk", "ws_sold_date_sk",
+                                       "ws_ext_sales_price", "ws_net_profit",
+                                       "ws_item_sk", "ws_order_number"])
+    wr = _read(root, sf, "web_returns", ["wr_item_sk", "wr_order_number",
+                                         "wr_returned_date_sk", "wr_return_amt", "wr_net_loss"])
+    wrj = wr.merge(ws[["ws_item_sk", "ws_order_number", "ws_web_site_sk"]]
+                   .dropna(subset=["ws_item_sk", "ws_order_number"]),
+                   left_on=["wr_item_sk", "wr_order_number"],
+                   right_on=["ws_item_sk", "ws_order_number"], how="left")
+    web = _read(root, sf, "web_site", ["web_site_sk", "web_site_id"])
+    wsr = part(pd.concat([mk(ws, "ws_web_site_sk", "ws_sold_date_sk", sp="ws_ext_sales_price", pr="ws_net_profit"),
+                          mk(wrj, "ws_web_site_sk", "wr_returned_date_sk", ra="wr_return_amt", nl="wr_net_loss")],
+                         ignore_index=True), web, "web_site_sk", "web_site_id", "web site")
+    u = pd.concat([ssr, csr, wsr], ignore_index=True)
+    out = _rollup2_oracle(u, "channel", "id")
+    out = out.sort_values(["channel", "id"], na_position="first").head(100)
+    return out.reset_index(drop=True)
+
+
+def q77(root, sf):
+    import pandas as pd
+
+    d0 = _days(2000, 8, 3)
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_date"])
+    di = _date_i(dd)
+    dd = dd[(di >= d0) & (di <= d0 + 30)]
+
+    def cte(fact, datec, fkc, sc, pc):
+        fs = _read(root, sf, fact, [datec, fkc, sc, pc])
+        j = _merge(fs, dd, datec, "d_date_sk")
+        return j.groupby(fkc, dropna=False).agg(
+            sales=(sc, lambda x: x.sum(min_count=1)),
+            profit=(pc, lambda x: x.sum(min_count=1))).reset_index()
+
+    ss = cte("store_sales", "ss_sold_date_sk", "ss_store_sk",
+             "ss_ext_sales_price", "ss_net_profit")
+    srr = _read(root, sf, "store_returns", ["sr_returned_date_sk", "sr_store_sk",
+                                            "sr_return_amt", "sr_net_loss"])
+    srj = _merge(srr, dd, "sr_returned_date_sk", "d_date_sk")
+    sr = srj.groupby("sr_store_sk", dropna=False).agg(
+        returns=("sr_return_amt", lambda x: x.sum(min_count=1)),
+        profit_loss=("sr_net_loss", lambda x: x.sum(min_count=1))).reset_index()
+    store = ss.merge(sr.dropna(subset=["sr_store_sk"]), left_on="ss_store_sk",
+                     right_on="sr_store_sk", how="left")
+    store_rows = pd.DataFrame({
+        "channel": "store channel",
+        "id": store.ss_store_sk.map(lambda v: None if pd.isna(v) else str(int(v))),
+        "sales": store.sales,
+        "returns": store.returns.fillna(0.0),
+        "profit": store.profit - store.profit_loss.fillna(0.0)})
+
+    cs = cte("catalog_sales", "cs_sold_date_sk", "cs_call_center_sk",
+             "cs_ext_sales_price", "cs_net_profit")
+    crr = _read(root, sf, "catalog_returns", ["cr_returned_date_sk", "cr_return_amount",
+                                              "cr_net_loss"])
+    crj = _merge(crr, dd, "cr_returned_date_sk", "d_date_sk")
+    cr_ret = crj.cr_return_amount.sum(min_count=1)
+    cr_loss = crj.cr_net_loss.sum(min_count=1)
+    catalog_rows = pd.DataFrame({
+        "channel": "catalog channel",
+        "id": cs.cs_call_center_sk.map(lambda v: None if pd.isna(v) else str(int(v))),
+        "sales": cs.sales, "returns": float(cr_ret or 0.0),
+        "profit": cs.profit - float(cr_loss or 0.0)})
+
+    ws = cte("web_sales", "ws_sold_date_sk", "ws_web_page_sk",
+             "ws_ext_sales_price", "ws_net_profit")
+    wrr = _read(root, sf, "web_returns", ["wr_item_sk", "wr_order_number",
+                                          "wr_returned_date_sk", "wr_return_amt", "wr_net_loss"])
+    wrj = _merge(wrr, dd, "wr_returned_date_sk", "d_date_sk")
+    wss = _read(root, sf, "web_sales", ["ws_item_sk", "ws_order_number", "ws_web_page_sk"])
+    wrj = wrj.merge(wss.dropna(subset=["ws_item_sk", "ws_order_number"]),
+                    left_on=["wr_item_sk", "wr_order_number"],
+                    right_on=["ws_item_sk", "ws_order_number"], how="left")
+    wra = wrj.groupby("ws_web_page_sk", dropna=False).agg(
+        returns=("wr_return_amt", lambda x: x.sum(min_count=1)),
+        profit_loss=("wr_net_loss", lambda x: x.sum(min_count=1))).reset_index()
+    web = ws.merge(wra.dropna(subset=["ws_web_page_sk"]), on="ws_web_page_sk", how="left")
+    web_rows = pd.DataFrame({
+        "channel": "web channel",
+        "id": web.ws_web_page_sk.map(lambda v: None if pd.isna(v) else str(int(v))),
+        "sales": web.sales, "returns": web.returns.fillna(0.0),
+        "profit": web.profit - web.profit_loss.fillna(0.0)})
+
+    u = pd.concat([store_rows, catalog_rows, web_rows], ignore_index=True)
+    out = _rollup2_oracle(u, "channel", "id")
+    out = out.sort_values(["channel", "id", "sales"], na_position="first").head(100)
+    return out.reset_index(drop=True)
+
+
+def q80(root, sf):
+    import pandas as pd
+
+    d0 = _days(2000, 8, 23)
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_date"])
+    di = _date_i(dd)
+    dd = dd[(di >= d0) & (di <= d0 + 30)]
+    it = _read(root, sf, "item", ["i_item_sk", "i_current_price"])
+    it = it[it.i_current_price > 50.0]
+    pr = _read(root, sf, "promotion", ["p_promo_sk", "p_channel_tv"])
+    pr = pr[pr.p_channel_tv == "N"]
+
+    def channel(fact, pre, rett, rpre, k1, k2, rk1, rk2, dim, dim_key, fk, id_col, tag):
+        ret_amt = f"{rpre}_return_{'amount' if rpre == 'cr' else 'amt'}"
+        fs = _read(root, sf, fact, list(dict.fromkeys(
+            [f"{pre}_item_sk", f"{pre}_sold_date_sk", fk, f"{pre}_promo_sk",
+             f"{pre}_ext_sales_price", f"{pre}_net_profit", k1, k2])))
+        rt = _read(root, sf, rett, [rk1, rk2, ret_amt, f"{rpre}_net_loss"])
+        j = fs.merge(rt.dropna(subset=[rk1, rk2]), left_on=[k1, k2],
+                     right_on=[rk1, rk2], how="left")
+        j = _merge(j, dd, f"{pre}_sold_date_sk", "d_date_sk")
+        j = _merge(j, it, f"{pre}_item_sk", "i_item_sk")
+        j = _merge(j, pr, f"{pre}_promo_sk", "p_promo_sk")
+        j = _merge(j, dim, fk, dim_key)
+        j = j.assign(sales_v=j[f"{pre}_ext_sales_price"],
+                     ret_v=j[ret_amt].fillna(0.0),
+                     prof_v=j[f"{pre}_net_profit"] - j[f"{rpre}_net_loss"].fillna(0.0))
+        a = j.groupby(id_col, dropna=False).agg(
+            sales=("sales_v", lambda x: x.sum(min_count=1)),
+            returns=("ret_v", lambda x: x.sum(min_count=1)),
+            profit=("prof_v", lambda x: x.sum(min_count=1))).reset_index()
+        a["channel"] = f"{tag} channel"
+        a["id"] = tag + a[id_col].astype(str)
+        return a[["channel", "id", "sales", "returns", "profit"]]
+
+    ssr = channel("store_sales", "ss", "store_returns", "sr",
+                  "ss_item_sk", "ss_ticket_number", "sr_item_sk", "sr_ticket_number",
+                  _read(root, sf, "store", ["s_store_sk", "s_store_id"]),
+                  "s_store_sk", "ss_store_sk", "s_store_id", "store")
+    csr = channel("catalog_sales", "cs", "catalog_returns", "cr",
+                  "cs_item_sk", "cs_order_number", "cr_item_sk", "cr_order_number",
+                  _read(root, sf, "catalog_page", ["cp_catalog_page_sk", "cp_catalog_page_id"]),
+                  "cp_catalog_page_sk", "cs_catalog_page_sk", "cp_catalog_page_id", "catalog_page")
+    wsr = channel("web_sales", "ws", "web_returns", "wr",
+                  "ws_item_sk", "ws_order_number", "wr_item_sk", "wr_order_number",
+                  _read(root, sf, "web_site", ["web_site_sk", "web_site_id"]),
+                  "web_site_sk", "ws_web_site_sk", "web_site_id", "web_site")
+    u = pd.concat([ssr, csr, wsr], ignore_index=True)
+    out = _rollup2_oracle(u, "channel", "id")
+    out = out.sort_values(["channel", "id"], na_position="first").head(100)
+    return out.reset_index(drop=True)
+
+
+def _v1_window_oracle(root, sf, fact, pre, fk, dim, dim_key, dim_cols, measure, part4,
+                      order_out):
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year", "d_moy"])
+    dd = dd[(dd.d_year == 1999) | ((dd.d_year == 1998) & (dd.d_moy == 12))
+            | ((dd.d_year == 2000) & (dd.d_moy == 1))]
+    fs = _read(root, sf, fact, [f"{pre}_item_sk", f"{pre}_sold_date_sk", fk, measure])
+    it = _read(root, sf, "item", ["i_item_sk", "i_category", "i_brand"])
+    dimdf = _read(root, sf, dim, [dim_key] + dim_cols)
+    j = _merge(fs, it, f"{pre}_item_sk", "i_item_sk")
+    j = _merge(j, dd, f"{pre}_sold_date_sk", "d_date_sk")
+    j = _merge(j, dimdf, fk, dim_key)
+    keys = part4 + ["d_year", "d_moy"]
+    g = j.groupby(keys, dropna=False)[measure].sum(min_count=1).reset_index(name="sum_sales")
+    g["avg_monthly_sales"] = g.groupby(part4 + ["d_year"], dropna=False) \
+                              .sum_sales.transform("mean")
+    g = g.sort_values(part4 + ["d_year", "d_moy"])
+    g["psum"] = g.groupby(part4, dropna=False).sum_sales.shift(1)
+    g["nsum"] = g.groupby(part4, dropna=False).sum_sales.shift(-1)
+    cond = ((g.d_year == 1999) & (g.avg_monthly_sales > 0)
+            & ((g.sum_sales - g.avg_monthly_sales).abs() / g.avg_monthly_sales > 0.1)
+            & g.psum.notna() & g.nsum.notna())
+    f = g[cond.fillna(False)].copy()
+    f["_d"] = f.sum_sales - f.avg_monthly_sales
+    f = f.sort_values(["_d", order_out], na_position="first").head(100)
+    cols = part4 + ["d_year", "d_moy", "avg_monthly_sales", "sum_sales", "psum", "nsum"]
+    return f[cols].reset_index(drop=True)
+
+
+def q47(root, sf):
+    return _v1_window_oracle(root, sf, "store_sales", "ss", "ss_store_sk",
+                             "store", "s_store_sk", ["s_store_name", "s_company_id"],
+                             "ss_sales_price",
+                             ["i_category", "i_brand", "s_store_name", "s_company_id"],
+                             "s_store_name")
+
+
+def q57(root, sf):
+    return _v1_window_oracle(root, sf, "catalog_sales", "cs", "cs_call_center_sk",
+                             "call_center", "cc_call_center_sk", ["cc_name"],
+                             "cs_sales_price", ["i_category", "i_brand", "cc_name"],
+                             "cc_name")
+
+
+def q61(root, sf):
+    import pandas as pd
+
+    def total(with_promo):
+        ss = _read(root, sf, "store_sales", ["ss_sold_date_sk", "ss_store_sk", "ss_promo_sk",
+                                             "ss_customer_sk", "ss_item_sk", "ss_ext_sales_price"])
+        dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year", "d_moy"])
+        dd = dd[(dd.d_year == 1998) & (dd.d_moy == 11)]
+        st = _read(root, sf, "store", ["s_store_sk", "s_gmt_offset"])
+        st = st[st.s_gmt_offset == -5.0]
+        it = _read(root, sf, "item", ["i_item_sk", "i_category"])
+        it = it[it.i_category == "Jewelry"]
+        cust = _read(root, sf, "customer", ["c_customer_sk", "c_current_addr_sk"])
+        ca = _read(root, sf, "customer_address", ["ca_address_sk", "ca_gmt_offset"])
+        ca = ca[ca.ca_gmt_offset == -5.0]
+        j = _merge(ss, dd, "ss_sold_date_sk", "d_date_sk")
+        j = _merge(j, st, "ss_store_sk", "s_store_sk")
+        j = _merge(j, it, "ss_item_sk", "i_item_sk")
+        j = _merge(j, cust, "ss_customer_sk", "c_customer_sk")
+        j = _merge(j, ca, "c_current_addr_sk", "ca_address_sk")
+        if with_promo:
+            pr = _read(root, sf, "promotion")
+            pr = pr[(pr.p_channel_dmail == "Y") | (pr.p_channel_email == "Y")
+                    | (pr.p_channel_tv == "Y")]
+            j = _merge(j, pr, "ss_promo_sk", "p_promo_sk")
+        v = j.ss_ext_sales_price.sum(min_count=1)
+        return None if pd.isna(v) else v
+
+    p = total(True)
+    t = total(False)
+    return pd.DataFrame({"promotions": [p], "total": [t],
+                         "ratio": [p / t * 100.0 if (p is not None and t) else None]})
+
+
+def q99(root, sf):
+    cs = _read(root, sf, "catalog_sales", ["cs_ship_date_sk", "cs_sold_date_sk",
+                                           "cs_warehouse_sk", "cs_ship_mode_sk",
+                                           "cs_call_center_sk"])
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_month_seq"])
+    dd = dd[dd.d_month_seq.between(1200, 1211)]
+    wh = _read(root, sf, "warehouse", ["w_warehouse_sk", "w_warehouse_name"])
+    sm = _read(root, sf, "ship_mode", ["sm_ship_mode_sk", "sm_type"])
+    cc = _read(root, sf, "call_center", ["cc_call_center_sk", "cc_name"])
+    j = _merge(cs, dd, "cs_ship_date_sk", "d_date_sk")
+    j = _merge(j, wh, "cs_warehouse_sk", "w_warehouse_sk")
+    j = _merge(j, sm, "cs_ship_mode_sk", "sm_ship_mode_sk")
+    j = _merge(j, cc, "cs_call_center_sk", "cc_call_center_sk")
+    j["wname20"] = j.w_warehouse_name.str[:20]
+    lag = j.cs_ship_date_sk - j.cs_sold_date_sk
+    j["d30"] = ((lag <= 30)).astype("float").where(lag.notna())
+    j["d31_60"] = ((lag > 30) & (lag <= 60)).astype("float").where(lag.notna())
+    j["d61_90"] = ((lag > 60) & (lag <= 90)).astype("float").where(lag.notna())
+    j["d91_120"] = ((lag > 90) & (lag <= 120)).astype("float").where(lag.notna())
+    j["d120p"] = ((lag > 120)).astype("float").where(lag.notna())
+    cols = ["d30", "d31_60", "d61_90", "d91_120", "d120p"]
+    g = j.groupby(["wname20", "sm_type", "cc_name"], dropna=False)[cols] \
+         .sum(min_count=1).reset_index()
+    for c in cols:
+        g[c] = g[c].astype("Int64")
+    g = g.sort_values(["wname20", "sm_type", "cc_name"], na_position="first").head(100)
+    return g.reset_index(drop=True)
+
+
+def q69(root, sf):
+    cust = _read(root, sf, "customer", ["c_customer_sk", "c_current_addr_sk",
+                                        "c_current_cdemo_sk"])
+    ca = _read(root, sf, "customer_address", ["ca_address_sk", "ca_state"])
+    ca = ca[ca.ca_state.isin(["KY", "GA", "NM"])]
+    cd = _read(root, sf, "customer_demographics")
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year", "d_moy"])
+    dd = dd[(dd.d_year == 2001) & dd.d_moy.between(4, 6)]
+
+    def custs(fact, pre, fk):
+        fs = _read(root, sf, fact, [f"{pre}_sold_date_sk", fk])
+        jj = _merge(fs, dd, f"{pre}_sold_date_sk", "d_date_sk")
+        return set(jj[fk].dropna())
+
+    sset = custs("store_sales", "ss", "ss_customer_sk")
+    wset = custs("web_sales", "ws", "ws_bill_customer_sk")
+    cset = custs("catalog_sales", "cs", "cs_ship_customer_sk")
+    j = _merge(cust, ca, "c_current_addr_sk", "ca_address_sk")
+    j = j[j.c_customer_sk.isin(sset) & ~j.c_customer_sk.isin(wset)
+          & ~j.c_customer_sk.isin(cset)]
+    j = _merge(j, cd, "c_current_cdemo_sk", "cd_demo_sk")
+    keys = ["cd_gender", "cd_marital_status", "cd_education_status",
+            "cd_purchase_estimate", "cd_credit_rating"]
+    g = j.groupby(keys, dropna=False).size().reset_index(name="cnt1")
+    g["cnt2"] = g.cnt1
+    g["cnt3"] = g.cnt1
+    g = g.sort_values(keys, na_position="first").head(100)
+    out = g[["cd_gender", "cd_marital_status", "cd_education_status", "cnt1",
+             "cd_purchase_estimate", "cnt2", "cd_credit_rating", "cnt3"]]
+    return out.reset_index(drop=True)
+
+
+ORACLES.update({"q5": q5, "q47": q47, "q57": q57, "q61": q61, "q69": q69,
+                "q77": q77, "q80": q80, "q99": q99})

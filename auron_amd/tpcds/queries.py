@@ -1459,3 +1459,386 @@ def q23(cat, s):
 
 QUERIES.update({"q11": q11, "q23": q23, "q31": q31, "q33": q33, "q38": q38,
                 "q60": q60, "q87": q87})
+
+
+# ------------------------------- batch 6: channel reports / windows
+def rollup2(child, k1, k2, aggs, k2_dtype=dtypes.string):
+    """GROUP BY ROLLUP(k1, k2) via Expand (3 grouping sets)."""
+    from ..exprs import Literal
+
+    n1 = Literal(None, dtypes.string)
+    n2 = Literal(None, k2_dtype)
+    carry = ["sales", "returns", "profit"]
+    projections = [
+        [_a(col(k1), k1), _a(col(k2), k2), _a(lit(2), "_gid")] + [_a(col(c), c) for c in carry],
+        [_a(col(k1), k1), _a(n2, k2), _a(lit(1), "_gid")] + [_a(col(c), c) for c in carry],
+        [_a(n1, k1), _a(n2, k2), _a(lit(0), "_gid")] + [_a(col(c), c) for c in carry],
+    ]
+    ex = P.Expand(child, projections)
+    a = agg2(ex, [k1, k2, "_gid"],
+             [AggFunc("sum", col(c), name=c) for c in carry])
+    return P.Project(a, [_a(col(k1), k1), _a(col(k2), k2)]
+                     + [_a(col(c), c) for c in carry])
+
+
+def q5(cat, s):
+    from ..exprs import ConcatStr
+
+    d0 = _days(2000, 8, 23)
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_date"]),
+                  col("d_date").cast(dtypes.int32).between(d0, d0 + 14))
+
+    def part(rows_sales, rows_returns, dim_scan, dim_key, fk, id_col, tag):
+        u = P.Union([rows_sales, rows_returns])
+        j = bhj(u, dd, ["date_sk"], ["d_date_sk"])
+        j = bhj(j, dim_scan, [fk], [dim_key])
+        a = agg2(j, [id_col],
+                 [AggFunc("sum", col("sales_price"), name="sales"),
+                  AggFunc("sum", col("return_amt"), name="returns"),
+                  AggFunc("sum", col("profit"), name="p1"),
+                  AggFunc("sum", col("net_loss"), name="p2")])
+        return P.Project(a, [_a(lit(f"{tag} channel"), "channel"),
+                             _a(ConcatStr([lit(tag.replace(" ", "_")), col(id_col)]), "id"),
+                             _a(col("sales"), "sales"), _a(col("returns"), "returns"),
+                             _a(col("p1") - col("p2"), "profit")])
+
+    z = lit(0.0)
+
+    def sel(scan, fk_col, date_col, sp, pr, ra, nl):
+        return P.Project(scan, [_a(col(fk_col), "fk"), _a(col(date_col), "date_sk"),
+                                _a(sp, "sales_price"), _a(pr, "profit"),
+                                _a(ra, "return_amt"), _a(nl, "net_loss")])
+
+    ss_rows = sel(cat.scan("store_sales", ["ss_store_sk", "ss_sold_date_sk",
+                                           "ss_ext_sales_price", "ss_net_profit"]),
+                  "ss_store_sk", "ss_sold_date_sk",
+                  col("ss_ext_sales_price"), col("ss_net_profit"), z, z)
+    sr_rows = sel(cat.scan("store_returns", ["sr_store_sk", "sr_returned_date_sk",
+                                             "sr_return_amt", "sr_net_loss"]),
+                  "sr_store_sk", "sr_returned_date_sk", z, z,
+                  col("sr_return_amt"), col("sr_net_loss"))
+    st = cat.scan("store", ["s_store_sk", "s_store_id"])
+    ssr = part(ss_rows, sr_rows, st, "s_store_sk", "fk", "s_store_id", "store")
+
+    cs_rows = sel(cat.scan("catalog_sales", ["cs_catalog_page_sk", "cs_sold_date_sk",
+                                             "cs_ext_sales_price", "cs_net_profit"]),
+                  "cs_catalog_page_sk", "cs_sold_date_sk",
+                  col("cs_ext_sales_price"), col("cs_net_profit"), z, z)
+    cr_rows = sel(cat.scan("catalog_returns", ["cr_catalog_page_sk", "cr_returned_date_sk",
+                                               "cr_return_amount", "cr_net_loss"]),
+                  "cr_catalog_page_sk", "cr_returned_date_sk", z, z,
+                  col("cr_return_amount"), col("cr_net_loss"))
+    cp = cat.scan("catalog_page", ["cp_catalog_page_sk", "cp_catalog_page_id"])
+    csr = part(cs_rows, cr_rows, cp, "cp_catalog_page_sk", "fk", "cp_catalog_page_id", "catalog page")
+
+    ws_rows = sel(cat.scan("web_sales", ["ws_web_site_sk", "ws_sold_date_sk",
+                                         "ws_ext_sales_price", "ws_net_profit"]),
+                  "ws_web_site_sk", "ws_sold_date_sk",
+                  col("ws_ext_sales_price"), col("ws_net_profit"), z, z)
+    # web returns reach the site through the originating sale
+    wr = cat.scan("web_returns", ["wr_item_sk", "wr_order_number",
+                                  "wr_returned_date_sk", "wr_return_amt", "wr_net_loss"])
+    wsj = cat.scan("web_sales", ["ws_item_sk", "ws_order_number", "ws_web_site_sk"])
+    wrj = shj(wr, wsj, ["wr_item_sk", "wr_order_number"],
+              ["ws_item_sk", "ws_order_number"], how="left")
+    wr_rows = P.Project(wrj, [_a(col("ws_web_site_sk"), "fk"),
+                              _a(col("wr_returned_date_sk"), "date_sk"),
+                              _a(z, "sales_price"), _a(z, "profit"),
+                              _a(col("wr_return_amt"), "return_amt"),
+                              _a(col("wr_net_loss"), "net_loss")])
+    web = cat.scan("web_site", ["web_site_sk", "web_site_id"])
+    wsr = part(ws_rows, wr_rows, web, "web_site_sk", "fk", "web_site_id", "web site")
+
+    u = P.Union([ssr, csr, wsr])
+    r = rollup2(u, "channel", "id",
+                [AggFunc("sum", col(c), name=c) for c in ("sales", "returns", "profit")])
+    return topk(r, [(col("channel"), True), (col("id"), True)], 100)
+
+
+def q77(cat, s):
+    from ..exprs import Coalesce
+
+    d0 = _days(2000, 8, 3)
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_date"]),
+                  col("d_date").cast(dtypes.int32).between(d0, d0 + 30))
+
+    def cte(fact, pre, date_fk, group_fk, sales_col, profit_col, names):
+        fs = cat.scan(fact, [date_fk, group_fk, sales_col, profit_col])
+        j = bhj(fs, dd, [date_fk], ["d_date_sk"])
+        return agg2(j, [names[0]],
+                    [AggFunc("sum", col(sales_col), name=names[1]),
+                     AggFunc("sum", col(profit_col), name=names[2])],
+                    key_exprs=[col(group_fk)])
+
+    ss = cte("store_sales", "ss", "ss_sold_date_sk", "ss_store_sk",
+             "ss_ext_sales_price", "ss_net_profit", ["sk", "sales", "profit"])
+    sr = cte("store_returns", "sr", "sr_returned_date_sk", "sr_store_sk",
+             "sr_return_amt", "sr_net_loss", ["rsk", "returns", "profit_loss"])
+    store_part = P.HashJoin(ss, P.Broadcast(sr), [col("sk")], [col("rsk")],
+                            how="left", build_side="right")
+    store_rows = P.Project(store_part, [
+        _a(lit("store channel"), "channel"), _a(col("sk").cast(dtypes.string), "id"),
+        _a(col("sales"), "sales"),
+        _a(Coalesce([col("returns"), lit(0.0)]), "returns"),
+        _a(col("profit") - Coalesce([col("profit_loss"), lit(0.0)]), "profit")])
+
+    cs = cte("catalog_sales", "cs", "cs_sold_date_sk", "cs_call_center_sk",
+             "cs_ext_sales_price", "cs_net_profit", ["csk", "sales", "profit"])
+    cr_tot = _global_agg(bhj(cat.scan("catalog_returns",
+                                      ["cr_returned_date_sk", "cr_return_amount", "cr_net_loss"]),
+                             dd, ["cr_returned_date_sk"], ["d_date_sk"]),
+                         [AggFunc("sum", col("cr_return_amount"), name="returns"),
+                          AggFunc("sum", col("cr_net_loss"), name="profit_loss")])
+    cr_b = s.collect_all(cr_tot).to_pydict()
+    cr_ret = cr_b["returns"][0] if cr_b["returns"] else 0.0
+    cr_loss = cr_b["profit_loss"][0] if cr_b["profit_loss"] else 0.0
+    catalog_rows = P.Project(cs, [
+        _a(lit("catalog channel"), "channel"), _a(col("csk").cast(dtypes.string), "id"),
+        _a(col("sales"), "sales"), _a(lit(cr_ret or 0.0), "returns"),
+        _a(col("profit") - lit(cr_loss or 0.0), "profit")])
+
+    ws = cte("web_sales", "ws", "ws_sold_date_sk", "ws_web_page_sk",
+             "ws_ext_sales_price", "ws_net_profit", ["wsk", "sales", "profit"])
+    # web_returns has no page fk in the synthetic schema; returns reach the
+    # page through the originating sale (same shape as q5's wsr)
+    wr = cat.scan("web_returns", ["wr_item_sk", "wr_order_number",
+                                  "wr_returned_date_sk", "wr_return_amt", "wr_net_loss"])
+    wrj = bhj(wr, dd, ["wr_returned_date_sk"], ["d_date_sk"])
+    wsj = cat.scan("web_sales", ["ws_item_sk", "ws_order_number", "ws_web_page_sk"])
+    wrj = shj(wrj, wsj, ["wr_item_sk", "wr_order_number"],
+              ["ws_item_sk", "ws_order_number"], how="left")
+    wr_agg = agg2(wrj, ["wrsk"],
+                  [AggFunc("sum", col("wr_return_amt"), name="returns"),
+                   AggFunc("sum", col("wr_net_loss"), name="profit_loss")],
+                  key_exprs=[col("ws_web_page_sk")])
+    web_part = P.HashJoin(ws, P.Broadcast(wr_agg), [col("wsk")], [col("wrsk")],
+                          how="left", build_side="right")
+    web_rows = P.Project(web_part, [
+        _a(lit("web channel"), "channel"), _a(col("wsk").cast(dtypes.string), "id"),
+        _a(col("sales"), "sales"),
+        _a(Coalesce([col("returns"), lit(0.0)]), "returns"),
+        _a(col("profit") - Coalesce([col("profit_loss"), lit(0.0)]), "profit")])
+
+    u = P.Union([store_rows, catalog_rows, web_rows])
+    r = rollup2(u, "channel", "id", None)
+    return topk(r, [(col("channel"), True), (col("id"), True), (col("sales"), True)], 100)
+
+
+def q80(cat, s):
+    from ..exprs import Coalesce, ConcatStr
+
+    d0 = _days(2000, 8, 23)
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_date"]),
+                  col("d_date").cast(dtypes.int32).between(d0, d0 + 30))
+    it = P.Filter(cat.scan("item", ["i_item_sk", "i_current_price"]),
+                  col("i_current_price") > 50.0)
+    pr = P.Filter(cat.scan("promotion", ["p_promo_sk", "p_channel_tv"]),
+                  col("p_channel_tv") == lit("N"))
+
+    def channel(fact, pre, ret_table, rpre, join_keys, dim_scan, dim_key, fk, id_col, tag):
+        fs_cols = [f"{pre}_item_sk", f"{pre}_sold_date_sk", fk, f"{pre}_promo_sk",
+                   f"{pre}_ext_sales_price", f"{pre}_net_profit",
+                   join_keys[0][1], join_keys[1][1]]
+        fs = cat.scan(fact, list(dict.fromkeys(fs_cols)))
+        rt = cat.scan(ret_table, [join_keys[0][2], join_keys[1][2],
+                                  f"{rpre}_return_{'amount' if rpre == 'cr' else 'amt'}",
+                                  f"{rpre}_net_loss"])
+        j = shj(fs, rt, [join_keys[0][1], join_keys[1][1]],
+                [join_keys[0][2], join_keys[1][2]], how="left")
+        j = bhj(j, dd, [f"{pre}_sold_date_sk"], ["d_date_sk"])
+        j = bhj(j, it, [f"{pre}_item_sk"], ["i_item_sk"])
+        j = bhj(j, pr, [f"{pre}_promo_sk"], ["p_promo_sk"])
+        j = bhj(j, dim_scan, [fk], [dim_key])
+        ret_amt = f"{rpre}_return_{'amount' if rpre == 'cr' else 'amt'}"
+        pre_rows = P.Project(j, [
+            _a(col(id_col), "gid_col"),
+            _a(col(f"{pre}_ext_sales_price"), "sales_v"),
+            _a(Coalesce([col(ret_amt), lit(0.0)]), "ret_v"),
+            _a(col(f"{pre}_net_profit") - Coalesce([col(f"{rpre}_net_loss"), lit(0.0)]), "prof_v")])
+        a = agg2(pre_rows, ["gid_col"],
+                 [AggFunc("sum", col("sales_v"), name="sales"),
+                  AggFunc("sum", col("ret_v"), name="returns"),
+                  AggFunc("sum", col("prof_v"), name="profit")])
+        return P.Project(a, [_a(lit(f"{tag} channel"), "channel"),
+                             _a(ConcatStr([lit(tag), col("gid_col")]), "id"),
+                             _a(col("sales"), "sales"), _a(col("returns"), "returns"),
+                             _a(col("profit"), "profit")])
+
+    ssr = channel("store_sales", "ss", "store_returns", "sr",
+                  [("item", "ss_item_sk", "sr_item_sk"),
+                   ("ticket", "ss_ticket_number", "sr_ticket_number")],
+                  cat.scan("store", ["s_store_sk", "s_store_id"]),
+                  "s_store_sk", "ss_store_sk", "s_store_id", "store")
+    csr = channel("catalog_sales", "cs", "catalog_returns", "cr",
+                  [("item", "cs_item_sk", "cr_item_sk"),
+                   ("order", "cs_order_number", "cr_order_number")],
+                  cat.scan("catalog_page", ["cp_catalog_page_sk", "cp_catalog_page_id"]),
+                  "cp_catalog_page_sk", "cs_catalog_page_sk", "cp_catalog_page_id", "catalog_page")
+    wsr = channel("web_sales", "ws", "web_returns", "wr",
+                  [("item", "ws_item_sk", "wr_item_sk"),
+                   ("order", "ws_order_number", "wr_order_number")],
+                  cat.scan("web_site", ["web_site_sk", "web_site_id"]),
+                  "web_site_sk", "ws_web_site_sk", "web_site_id", "web_site")
+    u = P.Union([ssr, csr, wsr])
+    r = rollup2(u, "channel", "id", None)
+    return topk(r, [(col("channel"), True), (col("id"), True)], 100)
+
+
+def _v1_window_q(cat, s, dims, fact, pre, fact_fk_pairs, measure, part4, order_out):
+    """q47/q57 shape: monthly sums + avg over year window + lag/lead."""
+    from ..exprs import CaseWhen, Literal, WindowFunc
+
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_year", "d_moy"]),
+                  (col("d_year") == 1999)
+                  | ((col("d_year") == 1998) & (col("d_moy") == 12))
+                  | ((col("d_year") == 2000) & (col("d_moy") == 1)))
+    j = cat.scan(fact, [f"{pre}_item_sk", f"{pre}_sold_date_sk", fact_fk_pairs[0][0],
+                        measure])
+    j = bhj(j, cat.scan("item", ["i_item_sk", "i_category", "i_brand"]),
+            [f"{pre}_item_sk"], ["i_item_sk"])
+    j = bhj(j, dd, [f"{pre}_sold_date_sk"], ["d_date_sk"])
+    j = bhj(j, fact_fk_pairs[0][2], [fact_fk_pairs[0][0]], [fact_fk_pairs[0][1]])
+    keys = part4 + ["d_year", "d_moy"]
+    a = agg2(j, keys, [AggFunc("sum", col(measure), name="sum_sales")])
+    ex = P.Exchange(a, "hash", [col(k) for k in part4])
+    w1 = P.Window(ex, [col(k) for k in part4 + ["d_year"]], [],
+                  [_a(WindowFunc("avg", col("sum_sales")), "avg_monthly_sales")])
+    w2 = P.Window(w1, [col(k) for k in part4],
+                  [(col("d_year"), True), (col("d_moy"), True)],
+                  [_a(WindowFunc("lag", col("sum_sales")), "psum"),
+                   _a(WindowFunc("lead", col("sum_sales")), "nsum")])
+    diff = col("sum_sales") - col("avg_monthly_sales")
+    f = P.Filter(w2, (col("d_year") == 1999) & (col("avg_monthly_sales") > 0.0)
+                 & ((diff / col("avg_monthly_sales") > 0.1)
+                    | (diff / col("avg_monthly_sales") < -0.1))
+                 & col("psum").is_not_null() & col("nsum").is_not_null())
+    out_cols = part4 + ["d_year", "d_moy", "avg_monthly_sales", "sum_sales", "psum", "nsum"]
+    proj = P.Project(f, [_a(col(c), c) for c in out_cols] + [_a(diff, "_d")])
+    out = topk(proj, [(col("_d"), True), (col(order_out), True)], 100)
+    return P.Project(out, [_a(col(c), c) for c in out_cols])
+
+
+def q47(cat, s):
+    st = cat.scan("store", ["s_store_sk", "s_store_name", "s_company_id"])
+    return _v1_window_q(cat, s, None, "store_sales", "ss",
+                        [("ss_store_sk", "s_store_sk", st)], "ss_sales_price",
+                        ["i_category", "i_brand", "s_store_name", "s_company_id"],
+                        "s_store_name")
+
+
+def q57(cat, s):
+    cc = cat.scan("call_center", ["cc_call_center_sk", "cc_name"])
+    return _v1_window_q(cat, s, None, "catalog_sales", "cs",
+                        [("cs_call_center_sk", "cc_call_center_sk", cc)], "cs_sales_price",
+                        ["i_category", "i_brand", "cc_name"], "cc_name")
+
+
+def q61(cat, s):
+    def total(with_promo):
+        ss = cat.scan("store_sales", ["ss_sold_date_sk", "ss_store_sk", "ss_promo_sk",
+                                      "ss_customer_sk", "ss_item_sk", "ss_ext_sales_price"])
+        dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_year", "d_moy"]),
+                      (col("d_year") == 1998) & (col("d_moy") == 11))
+        st = P.Filter(cat.scan("store", ["s_store_sk", "s_gmt_offset"]),
+                      col("s_gmt_offset") == -5.0)
+        it = P.Filter(cat.scan("item", ["i_item_sk", "i_category"]),
+                      col("i_category") == lit("Jewelry"))
+        cust = cat.scan("customer", ["c_customer_sk", "c_current_addr_sk"])
+        ca = P.Filter(cat.scan("customer_address", ["ca_address_sk", "ca_gmt_offset"]),
+                      col("ca_gmt_offset") == -5.0)
+        j = bhj(ss, dd, ["ss_sold_date_sk"], ["d_date_sk"])
+        j = bhj(j, st, ["ss_store_sk"], ["s_store_sk"])
+        j = bhj(j, it, ["ss_item_sk"], ["i_item_sk"])
+        j = bhj(j, cust, ["ss_customer_sk"], ["c_customer_sk"])
+        j = bhj(j, ca, ["c_current_addr_sk"], ["ca_address_sk"])
+        if with_promo:
+            pr = P.Filter(cat.scan("promotion", ["p_promo_sk", "p_channel_dmail",
+                                                 "p_channel_email", "p_channel_tv"]),
+                          (col("p_channel_dmail") == lit("Y"))
+                          | (col("p_channel_email") == lit("Y"))
+                          | (col("p_channel_tv") == lit("Y")))
+            j = bhj(j, pr, ["ss_promo_sk"], ["p_promo_sk"])
+        partial = P.HashAgg(j, [], [AggFunc("sum", col("ss_ext_sales_price"), name="v")],
+                            mode="partial")
+        return scalar(s, P.HashAgg(P.Exchange(partial, "single"), [],
+                                   [AggFunc("sum", col("ss_ext_sales_price"), name="v")],
+                                   mode="final"))
+
+    promotions = total(True)
+    tot = total(False)
+    from ..column import RecordBatch
+
+    n = 1 if s.rank == 0 else 0
+    ratio = (promotions / tot * 100.0) if (promotions is not None and tot) else None
+    b = RecordBatch.from_pydict(
+        {"promotions": [promotions] * max(n, 1), "total": [tot] * max(n, 1),
+         "ratio": [ratio] * max(n, 1)},
+        {"promotions": dtypes.float64, "total": dtypes.float64, "ratio": dtypes.float64})
+    if n == 0:
+        b = b.slice(0, 0)
+    return P.MemoryScan([b])
+
+
+def q99(cat, s):
+    from ..exprs import CaseWhen, Substr
+
+    fs = cat.scan("catalog_sales", ["cs_ship_date_sk", "cs_sold_date_sk",
+                                    "cs_warehouse_sk", "cs_ship_mode_sk", "cs_call_center_sk"])
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_month_seq"]),
+                  col("d_month_seq").between(1200, 1211))
+    wh = cat.scan("warehouse", ["w_warehouse_sk", "w_warehouse_name"])
+    sm = cat.scan("ship_mode", ["sm_ship_mode_sk", "sm_type"])
+    cc = cat.scan("call_center", ["cc_call_center_sk", "cc_name"])
+    j = bhj(fs, dd, ["cs_ship_date_sk"], ["d_date_sk"])
+    j = bhj(j, wh, ["cs_warehouse_sk"], ["w_warehouse_sk"])
+    j = bhj(j, sm, ["cs_ship_mode_sk"], ["sm_ship_mode_sk"])
+    j = bhj(j, cc, ["cs_call_center_sk"], ["cc_call_center_sk"])
+    lag = col("cs_ship_date_sk") - col("cs_sold_date_sk")
+    buckets = [("d30", lag <= 30), ("d31_60", (lag > 30) & (lag <= 60)),
+               ("d61_90", (lag > 60) & (lag <= 90)),
+               ("d91_120", (lag > 90) & (lag <= 120)), ("d120p", lag > 120)]
+    aggs = [AggFunc("sum", CaseWhen([(c, lit(1))], lit(0)), name=n) for n, c in buckets]
+    pre = P.Project(j, [_a(Substr(col("w_warehouse_name"), 1, 20), "wname20"),
+                        _a(col("sm_type"), "sm_type"), _a(col("cc_name"), "cc_name"),
+                        _a(col("cs_ship_date_sk"), "cs_ship_date_sk"),
+                        _a(col("cs_sold_date_sk"), "cs_sold_date_sk")])
+    a = agg2(pre, ["wname20", "sm_type", "cc_name"], aggs)
+    return topk(a, [(col("wname20"), True), (col("sm_type"), True),
+                    (col("cc_name"), True)], 100)
+
+
+def q69(cat, s):
+    cust = cat.scan("customer", ["c_customer_sk", "c_current_addr_sk", "c_current_cdemo_sk"])
+    ca = P.Filter(cat.scan("customer_address", ["ca_address_sk", "ca_state"]),
+                  col("ca_state").isin(["KY", "GA", "NM"]))
+    cd = cat.scan("customer_demographics",
+                  ["cd_demo_sk", "cd_gender", "cd_marital_status", "cd_education_status",
+                   "cd_purchase_estimate", "cd_credit_rating"])
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_year", "d_moy"]),
+                  (col("d_year") == 2001) & col("d_moy").between(4, 6))
+
+    def channel_cust(fact, pre, fk):
+        fs = cat.scan(fact, [f"{pre}_sold_date_sk", fk])
+        jj = bhj(fs, dd, [f"{pre}_sold_date_sk"], ["d_date_sk"])
+        return P.Exchange(P.Project(jj, [_a(col(fk), "xck")]), "hash", [col("xck")])
+
+    j = bhj(cust, ca, ["c_current_addr_sk"], ["ca_address_sk"])
+    j = P.Exchange(j, "hash", [col("c_customer_sk")])
+    j = P.HashJoin(j, channel_cust("store_sales", "ss", "ss_customer_sk"),
+                   [col("c_customer_sk")], [col("xck")], how="semi", build_side="right")
+    j = P.HashJoin(j, channel_cust("web_sales", "ws", "ws_bill_customer_sk"),
+                   [col("c_customer_sk")], [col("xck")], how="anti", build_side="right")
+    j = P.HashJoin(j, channel_cust("catalog_sales", "cs", "cs_ship_customer_sk"),
+                   [col("c_customer_sk")], [col("xck")], how="anti", build_side="right")
+    j = bhj(j, cd, ["c_current_cdemo_sk"], ["cd_demo_sk"])
+    keys = ["cd_gender", "cd_marital_status", "cd_education_status",
+            "cd_purchase_estimate", "cd_credit_rating"]
+    a = agg2(j, keys, [AggFunc("count_star", None, name="cnt1")])
+    proj = P.Project(a, [_a(col(k), k) for k in keys[:3]] + [_a(col("cnt1"), "cnt1")]
+                     + [_a(col(keys[3]), keys[3]), _a(col("cnt1"), "cnt2"),
+                        _a(col(keys[4]), keys[4]), _a(col("cnt1"), "cnt3")])
+    return topk(proj, [(col(k), True) for k in keys], 100)
+
+
+QUERIES.update({"q5": q5, "q47": q47, "q57": q57, "q61": q61, "q69": q69,
+                "q77": q77, "q80": q80, "q99": q99})
