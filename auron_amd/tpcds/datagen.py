@@ -176,6 +176,7 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
         put("ca_county", [f"{_CITIES[int(v)]} County" for v in rng.integers(0, len(_CITIES), n)])
         put("ca_gmt_offset", rng.choice([-5.0, -6.0, -7.0, -8.0], n))
         put("ca_street_name", [f"{_LAST[int(v)]} St" for v in rng.integers(0, len(_LAST), n)])
+        put("ca_street_type", [["Street", "Avenue", "Blvd", "Court", "Lane"][int(v)] for v in rng.integers(0, 5, n)])
         put("ca_location_type", [["apartment", "condo", "single family"][int(v)] for v in rng.integers(0, 3, n)])
         put("ca_suite_number", [f"Suite {int(v)}" for v in rng.integers(0, 100, n)])
         put("ca_street_number", [str(int(v)) for v in rng.integers(1, 1000, n)])
@@ -393,6 +394,8 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
         put(f"{pre}_order_number", rng.integers(1, max(row_count(src_fact, sf) // 4, 2), n))
         c, cv = fk(n_cust)
         put(f"{pre}_returning_customer_sk", c, cv)
+        ra, rav = fk(n_addr)
+        put(f"{pre}_returning_addr_sk", ra, rav)
         if pre == "cr":
             cp, cpv = fk(BASE_ROWS["catalog_page"], 0.02)
             put("cr_catalog_page_sk", cp, cpv)
@@ -405,6 +408,7 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
         put(amt_name, amt, _with_nulls(rng, amt, 0.02)[1])
         if pre == "cr":
             put("cr_return_tax", np.round(amt * 0.09 * rng.random(n), 2))
+            put("cr_return_amt_inc_tax", np.round(amt * 1.05, 2))
         else:
             put("wr_fee", _money(rng, n, 0.5, 100))
         put(f"{pre}_net_loss", _money(rng, n, 0.5, 1000))
@@ -460,7 +464,7 @@ def _nparts_for(table: str, sf: float) -> int:
     return 1
 
 
-DATAGEN_VERSION = 2
+DATAGEN_VERSION = 4
 
 
 def dataset_root(root: str, sf: float) -> str:

@@ -1549,3 +1549,260 @@ def q69(root, sf):
 
 ORACLES.update({"q5": q5, "q47": q47, "q57": q57, "q61": q61, "q69": q69,
                 "q77": q77, "q80": q80, "q99": q99})
+
+
+# ------------------------------- batch 7 oracles
+def q13(root, sf):
+    import pandas as pd
+
+    ss = _read(root, sf, "store_sales")
+    st = _read(root, sf, "store", ["s_store_sk"])
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year"])
+    dd = dd[dd.d_year == 2001]
+    cd = _read(root, sf, "customer_demographics")
+    hd = _read(root, sf, "household_demographics")
+    ca = _read(root, sf, "customer_address")
+    ca = ca[ca.ca_country == "United States"]
+    j = _merge(ss, st, "ss_store_sk", "s_store_sk")
+    j = _merge(j, dd, "ss_sold_date_sk", "d_date_sk")
+    j = _merge(j, cd, "ss_cdemo_sk", "cd_demo_sk")
+    j = _merge(j, hd, "ss_hdemo_sk", "hd_demo_sk")
+    j = _merge(j, ca, "ss_addr_sk", "ca_address_sk")
+    c1 = (((j.cd_marital_status == "M") & (j.cd_education_status == "Advanced Degree")
+           & j.ss_sales_price.between(100, 150) & (j.hd_dep_count == 3))
+          | ((j.cd_marital_status == "S") & (j.cd_education_status == "College")
+             & j.ss_sales_price.between(50, 100) & (j.hd_dep_count == 1))
+          | ((j.cd_marital_status == "W") & (j.cd_education_status == "2 yr Degree")
+             & j.ss_sales_price.between(150, 200) & (j.hd_dep_count == 1)))
+    c2 = ((j.ca_state.isin(["TX", "OH"]) & j.ss_net_profit.between(100, 200))
+          | (j.ca_state.isin(["OR", "NM", "KY"]) & j.ss_net_profit.between(150, 300))
+          | (j.ca_state.isin(["VA", "TX", "MS"]) & j.ss_net_profit.between(50, 250)))
+    f = j[c1.fillna(False) & c2.fillna(False)]
+    sewc = f.ss_ext_wholesale_cost.sum(min_count=1)
+    return pd.DataFrame({
+        "avg_qty": [f.ss_quantity.mean()], "avg_esp": [f.ss_ext_sales_price.mean()],
+        "avg_ewc": [f.ss_ext_wholesale_cost.mean()],
+        "sum_ewc": [None if pd.isna(sewc) else sewc]})
+
+
+def q27(root, sf):
+    import pandas as pd
+
+    ss = _read(root, sf, "store_sales", ["ss_sold_date_sk", "ss_item_sk", "ss_store_sk",
+                                         "ss_cdemo_sk", "ss_quantity", "ss_list_price",
+                                         "ss_coupon_amt", "ss_sales_price"])
+    cd = _read(root, sf, "customer_demographics")
+    cd = cd[(cd.cd_gender == "M") & (cd.cd_marital_status == "S")
+            & (cd.cd_education_status == "College")]
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year"])
+    dd = dd[dd.d_year == 2002]
+    st = _read(root, sf, "store", ["s_store_sk", "s_state"])
+    st = st[st.s_state == "TN"]
+    it = _read(root, sf, "item", ["i_item_sk", "i_item_id"])
+    j = _merge(ss, cd, "ss_cdemo_sk", "cd_demo_sk")
+    j = _merge(j, dd, "ss_sold_date_sk", "d_date_sk")
+    j = _merge(j, st, "ss_store_sk", "s_store_sk")
+    j = _merge(j, it, "ss_item_sk", "i_item_sk")
+    frames = []
+    for depth, g_state in ((2, 0), (1, 1), (0, 1)):
+        t = j.copy()
+        if depth < 2:
+            t["s_state"] = None
+        if depth < 1:
+            t["i_item_id"] = None
+        g = t.groupby(["i_item_id", "s_state"], dropna=False).agg(
+            agg1=("ss_quantity", "mean"), agg2=("ss_list_price", "mean"),
+            agg3=("ss_coupon_amt", "mean"), agg4=("ss_sales_price", "mean")).reset_index()
+        g["g_state"] = g_state
+        frames.append(g)
+    out = pd.concat(frames, ignore_index=True)
+    out = out.sort_values(["i_item_id", "s_state"], na_position="first").head(100)
+    return out[["i_item_id", "s_state", "g_state", "agg1", "agg2", "agg3", "agg4"]] \
+        .reset_index(drop=True)
+
+
+def q36(root, sf):
+    import pandas as pd
+
+    ss = _read(root, sf, "store_sales", ["ss_sold_date_sk", "ss_item_sk", "ss_store_sk",
+                                         "ss_net_profit", "ss_ext_sales_price"])
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year"])
+    dd = dd[dd.d_year == 2001]
+    st = _read(root, sf, "store", ["s_store_sk", "s_state"])
+    st = st[st.s_state == "TN"]
+    it = _read(root, sf, "item", ["i_item_sk", "i_category", "i_class"])
+    j = _merge(ss, dd, "ss_sold_date_sk", "d_date_sk")
+    j = _merge(j, st, "ss_store_sk", "s_store_sk")
+    j = _merge(j, it, "ss_item_sk", "i_item_sk")
+    frames = []
+    for depth, loc in ((2, 0), (1, 1), (0, 2)):
+        t = j.copy()
+        if depth < 2:
+            t["i_class"] = None
+        if depth < 1:
+            t["i_category"] = None
+        g = t.groupby(["i_category", "i_class"], dropna=False).agg(
+            snp=("ss_net_profit", lambda x: x.sum(min_count=1)),
+            sesp=("ss_ext_sales_price", lambda x: x.sum(min_count=1))).reset_index()
+        g["lochierarchy"] = loc
+        frames.append(g)
+    out = pd.concat(frames, ignore_index=True)
+    out["gross_margin"] = out.snp / out.sesp
+    out["_pcat"] = out.i_category.where(out.lochierarchy == 0)
+    import numpy as np
+
+    out["_mkey"] = np.trunc(out.gross_margin * 1e6)
+    out["rank_within_parent"] = out.groupby(["lochierarchy", "_pcat"], dropna=False) \
+        ._mkey.rank(method="min")
+    out = out.sort_values(["lochierarchy", "_pcat", "rank_within_parent"],
+                          ascending=[False, True, True], na_position="first").head(100)
+    out["rank_within_parent"] = out.rank_within_parent.astype(int)
+    return out[["gross_margin", "i_category", "i_class", "lochierarchy",
+                "rank_within_parent"]].reset_index(drop=True)
+
+
+def q76(root, sf):
+    import pandas as pd
+
+    def chan(fact, pre, null_col, tag):
+        cols = list(dict.fromkeys([f"{pre}_sold_date_sk", f"{pre}_item_sk", null_col,
+                                   f"{pre}_ext_sales_price"]))
+        fs = _read(root, sf, fact, cols)
+        fs = fs[fs[null_col].isna()]
+        dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year", "d_qoy"])
+        it = _read(root, sf, "item", ["i_item_sk", "i_category"])
+        j = _merge(fs, dd, f"{pre}_sold_date_sk", "d_date_sk")
+        j = _merge(j, it, f"{pre}_item_sk", "i_item_sk")
+        j["channel"] = tag
+        j["col_name"] = null_col
+        j = j.rename(columns={f"{pre}_ext_sales_price": "ext_sales_price"})
+        return j[["channel", "col_name", "d_year", "d_qoy", "i_category", "ext_sales_price"]]
+
+    u = pd.concat([chan("store_sales", "ss", "ss_store_sk", "store"),
+                   chan("web_sales", "ws", "ws_ship_customer_sk", "web"),
+                   chan("catalog_sales", "cs", "cs_ship_addr_sk", "catalog")],
+                  ignore_index=True)
+    g = u.groupby(["channel", "col_name", "d_year", "d_qoy", "i_category"],
+                  dropna=False).agg(sales_cnt=("ext_sales_price", "size"),
+                                    sales_amt=("ext_sales_price", lambda x: x.sum(min_count=1))) \
+         .reset_index()
+    g = g.sort_values(["channel", "col_name", "d_year", "d_qoy", "i_category"],
+                      na_position="first").head(100)
+    return g.reset_index(drop=True)
+
+
+def _ctr_state_oracle(root, sf, rets, rpre, amt, year, out_cols):
+    cr = _read(root, sf, rets, [f"{rpre}_returned_date_sk", f"{rpre}_returning_customer_sk",
+                                f"{rpre}_returning_addr_sk", amt])
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year"])
+    dd = dd[dd.d_year == year]
+    ca = _read(root, sf, "customer_address", ["ca_address_sk", "ca_state"])
+    j = _merge(cr, dd, f"{rpre}_returned_date_sk", "d_date_sk")
+    j = _merge(j, ca, f"{rpre}_returning_addr_sk", "ca_address_sk")
+    ctr = j.groupby([f"{rpre}_returning_customer_sk", "ca_state"], dropna=False)[amt] \
+           .sum(min_count=1).reset_index()
+    ctr.columns = ["ctr_customer_sk", "ctr_state", "ctr_total_return"]
+    av = ctr.groupby("ctr_state", dropna=False).ctr_total_return.mean().reset_index(name="av")
+    j2 = _merge(ctr, av.rename(columns={"ctr_state": "av_state"}), "ctr_state", "av_state")
+    f = j2[j2.ctr_total_return > j2.av * 1.2]
+    cust = _read(root, sf, "customer")
+    ca2 = _read(root, sf, "customer_address")
+    ca2 = ca2[ca2.ca_state == "GA"].rename(columns={"ca_state": "ca2_state"})
+    j3 = _merge(f, cust, "ctr_customer_sk", "c_customer_sk")
+    j4 = _merge(j3, ca2, "c_current_addr_sk", "ca_address_sk")
+    out = j4[out_cols + ["ctr_total_return"]]
+    out = out.sort_values(out_cols + ["ctr_total_return"], na_position="first").head(100)
+    return out.reset_index(drop=True)
+
+
+def q81(root, sf):
+    return _ctr_state_oracle(root, sf, "catalog_returns", "cr", "cr_return_amt_inc_tax",
+                             2000,
+                             ["c_customer_id", "c_salutation", "c_first_name", "c_last_name",
+                              "ca_street_number", "ca_street_name", "ca_street_type",
+                              "ca_suite_number", "ca_city", "ca_county", "ca2_state",
+                              "ca_zip", "ca_country", "ca_gmt_offset", "ca_location_type"])
+
+
+def q30(root, sf):
+    return _ctr_state_oracle(root, sf, "web_returns", "wr", "wr_return_amt", 2002,
+                             ["c_customer_id", "c_salutation", "c_first_name", "c_last_name",
+                              "c_preferred_cust_flag", "c_birth_month", "c_birth_year",
+                              "c_birth_country", "c_email_address"])
+
+
+def _yoy_oracle(root, sf, channels, first_year, out_expr, out_name):
+    totals = {}
+    for tag, (fact, pre, fk, cols, mfn) in channels.items():
+        fs = _read(root, sf, fact, sorted({fk, f"{pre}_sold_date_sk"} | set(cols)))
+        dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year"])
+        dd = dd[dd.d_year.isin([first_year, first_year + 1])]
+        cust = _read(root, sf, "customer", ["c_customer_sk", "c_customer_id",
+                                            "c_preferred_cust_flag", "c_first_name",
+                                            "c_last_name"])
+        j = _merge(fs, dd, f"{pre}_sold_date_sk", "d_date_sk")
+        j = _merge(j, cust, fk, "c_customer_sk")
+        j["v"] = mfn(j)
+        totals[tag] = j.groupby(["c_customer_id", "c_preferred_cust_flag",
+                                 "c_first_name", "c_last_name", "d_year"], dropna=False) \
+                       .v.sum(min_count=1).reset_index(name="year_total")
+
+    tags = list(channels.keys())
+    base = totals[tags[0]]
+    s1 = base[(base.d_year == first_year) & (base.year_total > 0)]
+    s2 = base[base.d_year == first_year + 1]
+    j = s1.merge(s2, on="c_customer_id", suffixes=("_sf", "_ssec"))
+    keep = j
+    ratio_s = keep.year_total_ssec / keep.year_total_sf
+    mask = None
+    for i, tag in enumerate(tags[1:]):
+        t = totals[tag]
+        t1 = t[(t.d_year == first_year) & (t.year_total > 0)][["c_customer_id", "year_total"]] \
+            .rename(columns={"year_total": f"x{i}f"})
+        t2 = t[t.d_year == first_year + 1][["c_customer_id", "year_total"]] \
+            .rename(columns={"year_total": f"x{i}s"})
+        keep = keep.merge(t1, on="c_customer_id").merge(t2, on="c_customer_id")
+    ratio_s = keep.year_total_ssec / keep.year_total_sf
+    for i, tag in enumerate(tags[1:]):
+        m = (keep[f"x{i}s"] / keep[f"x{i}f"]) > ratio_s
+        mask = m if mask is None else (mask & m)
+    f = keep[mask.fillna(False)]
+    out = f[[out_expr]].rename(columns={out_expr: out_name})
+    out = out.sort_values(out_name, na_position="first").head(100)
+    return out.reset_index(drop=True)
+
+
+def q4(root, sf):
+    def half(lp, wc, da, sp):
+        return lambda j: (j[lp] - j[wc] - j[da] + j[sp]) / 2.0
+
+    return _yoy_oracle(root, sf, {
+        "s": ("store_sales", "ss", "ss_customer_sk",
+              ["ss_ext_list_price", "ss_ext_wholesale_cost", "ss_ext_discount_amt",
+               "ss_ext_sales_price"],
+              half("ss_ext_list_price", "ss_ext_wholesale_cost", "ss_ext_discount_amt",
+                   "ss_ext_sales_price")),
+        "c": ("catalog_sales", "cs", "cs_bill_customer_sk",
+              ["cs_ext_list_price", "cs_ext_wholesale_cost", "cs_ext_discount_amt",
+               "cs_ext_sales_price"],
+              half("cs_ext_list_price", "cs_ext_wholesale_cost", "cs_ext_discount_amt",
+                   "cs_ext_sales_price")),
+        "w": ("web_sales", "ws", "ws_bill_customer_sk",
+              ["ws_ext_list_price", "ws_ext_wholesale_cost", "ws_ext_discount_amt",
+               "ws_ext_sales_price"],
+              half("ws_ext_list_price", "ws_ext_wholesale_cost", "ws_ext_discount_amt",
+                   "ws_ext_sales_price")),
+    }, 2001, "c_preferred_cust_flag_ssec", "customer_preferred_cust_flag")
+
+
+def q74(root, sf):
+    return _yoy_oracle(root, sf, {
+        "s": ("store_sales", "ss", "ss_customer_sk", ["ss_net_paid"],
+              lambda j: j.ss_net_paid),
+        "w": ("web_sales", "ws", "ws_bill_customer_sk", ["ws_net_paid"],
+              lambda j: j.ws_net_paid),
+    }, 2001, "c_customer_id", "customer_id")
+
+
+ORACLES.update({"q4": q4, "q13": q13, "q27": q27, "q30": q30, "q36": q36,
+                "q74": q74, "q76": q76, "q81": q81})

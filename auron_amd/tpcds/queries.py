@@ -1842,3 +1842,289 @@ def q69(cat, s):
 
 QUERIES.update({"q5": q5, "q47": q47, "q57": q57, "q61": q61, "q69": q69,
                 "q77": q77, "q80": q80, "q99": q99})
+
+
+# ------------------------------- batch 7: rollup reports / YoY / profiles
+def rollup_expand(child, keys, key_dtypes, carry, with_gid=True):
+    """Generic ROLLUP(keys...) Expand: len(keys)+1 grouping sets."""
+    from ..exprs import Literal
+
+    projections = []
+    for depth in range(len(keys), -1, -1):
+        proj = [_a(col(k) if i < depth else Literal(None, dt), k)
+                for i, (k, dt) in enumerate(zip(keys, key_dtypes))]
+        if with_gid:
+            proj.append(_a(lit(len(keys) - depth), "_lochier"))
+        proj += [_a(col(c), c) for c in carry]
+        projections.append(proj)
+    return P.Expand(child, projections)
+
+
+def q13(cat, s):
+    ss = cat.scan("store_sales", ["ss_store_sk", "ss_sold_date_sk", "ss_hdemo_sk",
+                                  "ss_cdemo_sk", "ss_addr_sk", "ss_quantity",
+                                  "ss_ext_sales_price", "ss_ext_wholesale_cost",
+                                  "ss_sales_price", "ss_net_profit"])
+    st = cat.scan("store", ["s_store_sk"])
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_year"]), col("d_year") == 2001)
+    cd = cat.scan("customer_demographics",
+                  ["cd_demo_sk", "cd_marital_status", "cd_education_status"])
+    hd = cat.scan("household_demographics", ["hd_demo_sk", "hd_dep_count"])
+    ca = P.Filter(cat.scan("customer_address", ["ca_address_sk", "ca_country", "ca_state"]),
+                  col("ca_country") == lit("United States"))
+    j = bhj(ss, st, ["ss_store_sk"], ["s_store_sk"])
+    j = bhj(j, dd, ["ss_sold_date_sk"], ["d_date_sk"])
+    j = bhj(j, cd, ["ss_cdemo_sk"], ["cd_demo_sk"])
+    j = bhj(j, hd, ["ss_hdemo_sk"], ["hd_demo_sk"])
+    j = bhj(j, ca, ["ss_addr_sk"], ["ca_address_sk"])
+    sp = col("ss_sales_price")
+    np_ = col("ss_net_profit")
+    c1 = (((col("cd_marital_status") == lit("M")) & (col("cd_education_status") == lit("Advanced Degree"))
+           & sp.between(100.0, 150.0) & (col("hd_dep_count") == 3))
+          | ((col("cd_marital_status") == lit("S")) & (col("cd_education_status") == lit("College"))
+             & sp.between(50.0, 100.0) & (col("hd_dep_count") == 1))
+          | ((col("cd_marital_status") == lit("W")) & (col("cd_education_status") == lit("2 yr Degree"))
+             & sp.between(150.0, 200.0) & (col("hd_dep_count") == 1)))
+    c2 = ((col("ca_state").isin(["TX", "OH"]) & np_.between(100.0, 200.0))
+          | (col("ca_state").isin(["OR", "NM", "KY"]) & np_.between(150.0, 300.0))
+          | (col("ca_state").isin(["VA", "TX", "MS"]) & np_.between(50.0, 250.0)))
+    f = P.Filter(j, c1 & c2)
+    return _global_agg(f, [AggFunc("avg", col("ss_quantity"), name="avg_qty"),
+                           AggFunc("avg", col("ss_ext_sales_price"), name="avg_esp"),
+                           AggFunc("avg", col("ss_ext_wholesale_cost"), name="avg_ewc"),
+                           AggFunc("sum", col("ss_ext_wholesale_cost"), name="sum_ewc")])
+
+
+def q27(cat, s):
+    ss = cat.scan("store_sales", ["ss_sold_date_sk", "ss_item_sk", "ss_store_sk",
+                                  "ss_cdemo_sk", "ss_quantity", "ss_list_price",
+                                  "ss_coupon_amt", "ss_sales_price"])
+    cd = P.Filter(cat.scan("customer_demographics",
+                           ["cd_demo_sk", "cd_gender", "cd_marital_status", "cd_education_status"]),
+                  (col("cd_gender") == lit("M")) & (col("cd_marital_status") == lit("S"))
+                  & (col("cd_education_status") == lit("College")))
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_year"]), col("d_year") == 2002)
+    st = P.Filter(cat.scan("store", ["s_store_sk", "s_state"]), col("s_state") == lit("TN"))
+    it = cat.scan("item", ["i_item_sk", "i_item_id"])
+    j = bhj(ss, cd, ["ss_cdemo_sk"], ["cd_demo_sk"])
+    j = bhj(j, dd, ["ss_sold_date_sk"], ["d_date_sk"])
+    j = bhj(j, st, ["ss_store_sk"], ["s_store_sk"])
+    j = bhj(j, it, ["ss_item_sk"], ["i_item_sk"])
+    pre = P.Project(j, [_a(col("i_item_id"), "i_item_id"), _a(col("s_state"), "s_state"),
+                        _a(col("ss_quantity"), "q"), _a(col("ss_list_price"), "lp"),
+                        _a(col("ss_coupon_amt"), "cp"), _a(col("ss_sales_price"), "sp")])
+    ex = rollup_expand(pre, ["i_item_id", "s_state"], [dtypes.string, dtypes.string],
+                       ["q", "lp", "cp", "sp"])
+    a = agg2(ex, ["i_item_id", "s_state", "_lochier"],
+             [AggFunc("avg", col("q"), name="agg1"), AggFunc("avg", col("lp"), name="agg2"),
+              AggFunc("avg", col("cp"), name="agg3"), AggFunc("avg", col("sp"), name="agg4")])
+    from ..exprs import CaseWhen
+
+    g_state = CaseWhen([(col("_lochier") >= 1, lit(1))], lit(0))
+    proj = P.Project(a, [_a(col("i_item_id"), "i_item_id"), _a(col("s_state"), "s_state"),
+                         _a(g_state, "g_state"), _a(col("agg1"), "agg1"),
+                         _a(col("agg2"), "agg2"), _a(col("agg3"), "agg3"),
+                         _a(col("agg4"), "agg4")])
+    return topk(proj, [(col("i_item_id"), True), (col("s_state"), True)], 100)
+
+
+def q36(cat, s):
+    from ..exprs import CaseWhen, Literal, WindowFunc
+
+    ss = cat.scan("store_sales", ["ss_sold_date_sk", "ss_item_sk", "ss_store_sk",
+                                  "ss_net_profit", "ss_ext_sales_price"])
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_year"]), col("d_year") == 2001)
+    st = P.Filter(cat.scan("store", ["s_store_sk", "s_state"]), col("s_state") == lit("TN"))
+    it = cat.scan("item", ["i_item_sk", "i_category", "i_class"])
+    j = bhj(ss, dd, ["ss_sold_date_sk"], ["d_date_sk"])
+    j = bhj(j, st, ["ss_store_sk"], ["s_store_sk"])
+    j = bhj(j, it, ["ss_item_sk"], ["i_item_sk"])
+    pre = P.Project(j, [_a(col("i_category"), "i_category"), _a(col("i_class"), "i_class"),
+                        _a(col("ss_net_profit"), "np"), _a(col("ss_ext_sales_price"), "esp")])
+    ex = rollup_expand(pre, ["i_category", "i_class"], [dtypes.string, dtypes.string],
+                       ["np", "esp"])
+    a = agg2(ex, ["i_category", "i_class", "_lochier"],
+             [AggFunc("sum", col("np"), name="snp"), AggFunc("sum", col("esp"), name="sesp")])
+    margin = P.Project(a, [_a(col("snp") / col("sesp"), "gross_margin"),
+                           _a(col("i_category"), "i_category"),
+                           _a(col("i_class"), "i_class"),
+                           _a(col("_lochier"), "lochierarchy"),
+                           _a(CaseWhen([(col("_lochier") == 0, col("i_category"))],
+                                       Literal(None, dtypes.string)), "_pcat")])
+    # rank over a 1e-6-truncated margin: fp summation order differs between
+    # engines, so raw-float rank keys are not comparable across them
+    w = P.Window(P.Exchange(margin, "hash", [col("lochierarchy")]),
+                 [col("lochierarchy"), col("_pcat")],
+                 [((col("gross_margin") * lit(1000000.0)).cast(dtypes.int64), True)],
+                 [_a(WindowFunc("rank"), "rank_within_parent")])
+    proj = P.Project(w, [_a(col("gross_margin"), "gross_margin"),
+                         _a(col("i_category"), "i_category"),
+                         _a(col("i_class"), "i_class"),
+                         _a(col("lochierarchy"), "lochierarchy"),
+                         _a(col("rank_within_parent"), "rank_within_parent"),
+                         _a(col("_pcat"), "_pcat")])
+    out = topk(proj, [(col("lochierarchy"), False), (col("_pcat"), True),
+                      (col("rank_within_parent"), True)], 100)
+    return P.Project(out, [_a(col(c), c) for c in
+                           ["gross_margin", "i_category", "i_class", "lochierarchy",
+                            "rank_within_parent"]])
+
+
+def q76(cat, s):
+    def chan(fact, pre, null_col, tag):
+        cols = [f"{pre}_sold_date_sk", f"{pre}_item_sk", null_col, f"{pre}_ext_sales_price"]
+        fs = P.Filter(cat.scan(fact, list(dict.fromkeys(cols))),
+                      col(null_col).is_null())
+        dd = cat.scan("date_dim", ["d_date_sk", "d_year", "d_qoy"])
+        it = cat.scan("item", ["i_item_sk", "i_category"])
+        j = bhj(fs, dd, [f"{pre}_sold_date_sk"], ["d_date_sk"])
+        j = bhj(j, it, [f"{pre}_item_sk"], ["i_item_sk"])
+        return P.Project(j, [_a(lit(tag), "channel"), _a(lit(null_col), "col_name"),
+                             _a(col("d_year"), "d_year"), _a(col("d_qoy"), "d_qoy"),
+                             _a(col("i_category"), "i_category"),
+                             _a(col(f"{pre}_ext_sales_price"), "ext_sales_price")])
+
+    u = P.Union([chan("store_sales", "ss", "ss_store_sk", "store"),
+                 chan("web_sales", "ws", "ws_ship_customer_sk", "web"),
+                 chan("catalog_sales", "cs", "cs_ship_addr_sk", "catalog")])
+    a = agg2(u, ["channel", "col_name", "d_year", "d_qoy", "i_category"],
+             [AggFunc("count_star", None, name="sales_cnt"),
+              AggFunc("sum", col("ext_sales_price"), name="sales_amt")])
+    return topk(a, [(col("channel"), True), (col("col_name"), True),
+                    (col("d_year"), True), (col("d_qoy"), True),
+                    (col("i_category"), True)], 100)
+
+
+def _ctr_state_q(cat, s, rets, rpre, amt_col, year, out_cols):
+    """q81/q30 shape: per-customer-state returns vs state average."""
+    cr = cat.scan(rets, [f"{rpre}_returned_date_sk", f"{rpre}_returning_customer_sk",
+                         f"{rpre}_returning_addr_sk", amt_col])
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_year"]), col("d_year") == year)
+    ca = cat.scan("customer_address", ["ca_address_sk", "ca_state"])
+    j = bhj(cr, dd, [f"{rpre}_returned_date_sk"], ["d_date_sk"])
+    j = bhj(j, ca, [f"{rpre}_returning_addr_sk"], ["ca_address_sk"])
+    ctr = agg2(j, ["ctr_customer_sk", "ctr_state"],
+               [AggFunc("sum", col(amt_col), name="ctr_total_return")],
+               key_exprs=[col(f"{rpre}_returning_customer_sk"), col("ca_state")])
+    ctr_b = s.execute(ctr)
+    av = P.Broadcast(agg2(P.MemoryScan(ctr_b), ["av_state"],
+                          [AggFunc("avg", col("ctr_total_return"), name="av")],
+                          key_exprs=[col("ctr_state")]))
+    j2 = P.HashJoin(P.MemoryScan(ctr_b), av, [col("ctr_state")], [col("av_state")],
+                    how="inner", build_side="right")
+    f = P.Filter(j2, col("ctr_total_return") > col("av") * lit(1.2))
+    cust_cols = ["c_customer_sk", "c_customer_id", "c_salutation", "c_first_name",
+                 "c_last_name", "c_current_addr_sk", "c_preferred_cust_flag",
+                 "c_birth_month", "c_birth_year", "c_birth_country", "c_email_address"]
+    cust = cat.scan("customer", cust_cols)
+    ca2_cols = ["ca_address_sk", "ca_street_number", "ca_street_name", "ca_street_type",
+                "ca_suite_number", "ca_city", "ca_county", "ca_state", "ca_zip",
+                "ca_country", "ca_gmt_offset", "ca_location_type"]
+    ca2 = P.Filter(P.Project(cat.scan("customer_address", ca2_cols),
+                             [_a(col(c), c if c != "ca_state" else "ca2_state")
+                              for c in ca2_cols]),
+                   col("ca2_state") == lit("GA"))
+    j3 = bhj(f, cust, ["ctr_customer_sk"], ["c_customer_sk"])
+    j4 = bhj(j3, ca2, ["c_current_addr_sk"], ["ca_address_sk"])
+    proj = P.Project(j4, [_a(col(c), c) for c in out_cols] +
+                     [_a(col("ctr_total_return"), "ctr_total_return")])
+    return topk(proj, [(col(c), True) for c in out_cols]
+                + [(col("ctr_total_return"), True)], 100)
+
+
+def q81(cat, s):
+    return _ctr_state_q(cat, s, "catalog_returns", "cr", "cr_return_amt_inc_tax", 2000,
+                        ["c_customer_id", "c_salutation", "c_first_name", "c_last_name",
+                         "ca_street_number", "ca_street_name", "ca_street_type",
+                         "ca_suite_number", "ca_city", "ca_county", "ca2_state",
+                         "ca_zip", "ca_country", "ca_gmt_offset", "ca_location_type"])
+
+
+def q30(cat, s):
+    return _ctr_state_q(cat, s, "web_returns", "wr", "wr_return_amt", 2002,
+                        ["c_customer_id", "c_salutation", "c_first_name", "c_last_name",
+                         "c_preferred_cust_flag", "c_birth_month", "c_birth_year",
+                         "c_birth_country", "c_email_address"])
+
+
+def _yoy_q(cat, s, channels, first_year, out_col, out_name="customer_preferred_cust_flag"):
+    """q4/q74 shape: per-customer per-year totals, ratio comparison."""
+    totals = {}
+    for tag, (fact, pre, cust_fk, measure) in channels.items():
+        fs_cols = sorted({cust_fk, f"{pre}_sold_date_sk"} | set(measure[1]))
+        fs = cat.scan(fact, fs_cols)
+        dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_year"]),
+                      col("d_year").isin([first_year, first_year + 1]))
+        cust = cat.scan("customer", ["c_customer_sk", "c_customer_id",
+                                     "c_preferred_cust_flag", "c_first_name", "c_last_name"])
+        j = bhj(fs, dd, [f"{pre}_sold_date_sk"], ["d_date_sk"])
+        j = bhj(j, cust, [cust_fk], ["c_customer_sk"])
+        a = agg2(j, ["customer_id", "flag", "first_name", "last_name", "dyear"],
+                 [AggFunc("sum", measure[0], name="year_total")],
+                 key_exprs=[col("c_customer_id"), col("c_preferred_cust_flag"),
+                            col("c_first_name"), col("c_last_name"), col("d_year")])
+        totals[tag] = s.execute(a)
+
+    def inst(tag, year, prefix):
+        f = P.Filter(P.MemoryScan(totals[tag]), col("dyear") == year)
+        return P.Project(f, [_a(col("customer_id"), f"{prefix}_id"),
+                             _a(col("year_total"), f"{prefix}_total"),
+                             _a(col("flag"), f"{prefix}_flag")])
+
+    tags = list(channels.keys())
+    s1 = inst(tags[0], first_year, "sf")
+    s2 = inst(tags[0], first_year + 1, "ssec")
+    j = shj(P.Filter(s1, col("sf_total") > 0.0), s2, ["sf_id"], ["ssec_id"])
+    conds = []
+    for i, tag in enumerate(tags[1:]):
+        p1, p2 = f"x{i}f", f"x{i}s"
+        j = P.HashJoin(j, P.Exchange(P.Filter(inst(tag, first_year, p1),
+                                              col(f"{p1}_total") > 0.0),
+                                     "hash", [col(f"{p1}_id")]),
+                       [col("sf_id")], [col(f"{p1}_id")], how="inner", build_side="right")
+        j = P.HashJoin(j, P.Exchange(inst(tag, first_year + 1, p2),
+                                     "hash", [col(f"{p2}_id")]),
+                       [col("sf_id")], [col(f"{p2}_id")], how="inner", build_side="right")
+        conds.append((col(f"{p2}_total") / col(f"{p1}_total"))
+                     > (col("ssec_total") / col("sf_total")))
+    pred = conds[0]
+    for c in conds[1:]:
+        pred = pred & c
+    f = P.Filter(j, pred)
+    proj = P.Project(f, [_a(col(out_col), "out")])
+    out = topk(proj, [(col("out"), True)], 100)
+    return P.Project(out, [_a(col("out"), out_name)])
+
+
+def q4(cat, s):
+    half = lambda lp, wc, da, sp: (col(lp) - col(wc) - col(da) + col(sp)) / lit(2.0)
+    return _yoy_q(cat, s, {
+        "s": ("store_sales", "ss", "ss_customer_sk",
+              (half("ss_ext_list_price", "ss_ext_wholesale_cost",
+                    "ss_ext_discount_amt", "ss_ext_sales_price"),
+               ["ss_ext_list_price", "ss_ext_wholesale_cost",
+                "ss_ext_discount_amt", "ss_ext_sales_price"])),
+        "c": ("catalog_sales", "cs", "cs_bill_customer_sk",
+              (half("cs_ext_list_price", "cs_ext_wholesale_cost",
+                    "cs_ext_discount_amt", "cs_ext_sales_price"),
+               ["cs_ext_list_price", "cs_ext_wholesale_cost",
+                "cs_ext_discount_amt", "cs_ext_sales_price"])),
+        "w": ("web_sales", "ws", "ws_bill_customer_sk",
+              (half("ws_ext_list_price", "ws_ext_wholesale_cost",
+                    "ws_ext_discount_amt", "ws_ext_sales_price"),
+               ["ws_ext_list_price", "ws_ext_wholesale_cost",
+                "ws_ext_discount_amt", "ws_ext_sales_price"])),
+    }, 2001, "ssec_flag")
+
+
+def q74(cat, s):
+    return _yoy_q(cat, s, {
+        "s": ("store_sales", "ss", "ss_customer_sk",
+              (col("ss_net_paid"), ["ss_net_paid"])),
+        "w": ("web_sales", "ws", "ws_bill_customer_sk",
+              (col("ws_net_paid"), ["ws_net_paid"])),
+    }, 2001, "ssec_id", out_name="customer_id")
+
+
+QUERIES.update({"q4": q4, "q13": q13, "q27": q27, "q30": q30, "q36": q36,
+                "q74": q74, "q76": q76, "q81": q81})

@@ -36,7 +36,7 @@ SCHEMAS = {
     },
     "customer_address": {
         "ca_address_sk": i64, "ca_state": s, "ca_zip": s, "ca_country": s,
-        "ca_city": s, "ca_county": s, "ca_gmt_offset": f64, "ca_street_name": s,
+        "ca_city": s, "ca_county": s, "ca_gmt_offset": f64, "ca_street_name": s, "ca_street_type": s,
         "ca_location_type": s, "ca_suite_number": s, "ca_street_number": s,
     },
     "customer_demographics": {
@@ -105,8 +105,8 @@ SCHEMAS = {
     },
     "catalog_returns": {
         "cr_returned_date_sk": i64, "cr_item_sk": i64, "cr_order_number": i64,
-        "cr_returning_customer_sk": i64, "cr_catalog_page_sk": i64,
-        "cr_return_quantity": i32, "cr_return_amount": f64, "cr_return_tax": f64,
+        "cr_returning_customer_sk": i64, "cr_returning_addr_sk": i64, "cr_catalog_page_sk": i64,
+        "cr_return_quantity": i32, "cr_return_amount": f64, "cr_return_tax": f64, "cr_return_amt_inc_tax": f64,
         "cr_net_loss": f64, "cr_refunded_cash": f64, "cr_reversed_charge": f64,
         "cr_store_credit": f64, "cr_call_center_sk": i64,
     },
@@ -124,7 +124,7 @@ SCHEMAS = {
     },
     "web_returns": {
         "wr_returned_date_sk": i64, "wr_item_sk": i64, "wr_order_number": i64,
-        "wr_returning_customer_sk": i64, "wr_return_quantity": i32,
+        "wr_returning_customer_sk": i64, "wr_returning_addr_sk": i64, "wr_return_quantity": i32,
         "wr_return_amt": f64, "wr_net_loss": f64, "wr_fee": f64,
         "wr_refunded_cash": f64, "wr_reversed_charge": f64, "wr_account_credit": f64,
     },
