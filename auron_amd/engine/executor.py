@@ -647,10 +647,13 @@ class Executor:
                         r = torch.cumsum(incr, 0)
                         out = Column(dtypes.int64, r - r[seg_start][seg] + 1)
                     else:
-                        rn = pos_in_seg + 1
-                        newval = torch.where(same_as_prev, torch.zeros_like(rn), rn)
-                        r = torch.cummax(newval, 0).values
-                        out = Column(dtypes.int64, r)
+                        # rank = (index of last order-key change) - seg_start + 1;
+                        # seg starts count as changes, so the cummax of change
+                        # positions never leaks across partitions
+                        gpos = torch.arange(n, dtype=torch.int64, device=device)
+                        change_pos = torch.where(same_as_prev, torch.full_like(gpos, -1), gpos)
+                        lcp = torch.cummax(change_pos, 0).values
+                        out = Column(dtypes.int64, lcp - seg_start[seg] + 1)
                 else:
                     out = Column(dtypes.int64, torch.ones(n, dtype=torch.int64, device=device))
             elif wf.fn in ("sum", "avg", "count", "min", "max"):
