@@ -420,7 +420,14 @@ class Executor:
             out_keys = [c.gather(reps) for c in key_cols]
         else:
             gids = torch.zeros(n, dtype=torch.int64, device=device)
-            ngroups = 1
+            # a keyless (global) agg over an empty input yields one null row
+            # in SQL — but distributed, only the rank holding the gathered
+            # rows (rank 0 after an Exchange "single") may emit it, else
+            # every rank would contribute a duplicate
+            if n == 0 and node.mode in ("final", "complete") and self.ctx.rank != 0:
+                ngroups = 0
+            else:
+                ngroups = 1
             out_keys = []
         names = [a.name for a in node.keys]
         cols = list(out_keys)
