@@ -126,8 +126,13 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
     elif name == "time_dim":
         t = np.arange(n, dtype=np.int64) + lo
         put("t_time_sk", t)
-        put("t_hour", (t // 3600).astype(np.int32))
+        hour = (t // 3600).astype(np.int32)
+        put("t_hour", hour)
         put("t_minute", ((t % 3600) // 60).astype(np.int32))
+        meal = np.where(hour < 9, "breakfast",
+                        np.where((hour >= 11) & (hour < 14), "lunch",
+                                 np.where((hour >= 17) & (hour < 21), "dinner", "")))
+        put("t_meal_time", meal.tolist())
     elif name == "item":
         put("i_item_sk", sks)
         put("i_item_id", _id_str("AAAAAAAA", sks))
@@ -145,6 +150,9 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
         put("i_manager_id", rng.integers(1, 101, n).astype(np.int32))
         put("i_product_name", [f"product{int(s)}" for s in sks])
         put("i_item_desc", [f"the quite famous item number {int(s)} description" for s in sks])
+        _colors = ["red", "blue", "green", "yellow", "black", "white", "purple",
+                   "orange", "pink", "brown", "gray", "cyan"]
+        put("i_color", [_colors[int(v)] for v in rng.integers(0, 12, n)])
     elif name == "customer":
         put("c_customer_sk", sks)
         put("c_customer_id", _id_str("AAAAAAAA", sks))
@@ -243,6 +251,7 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
         put("web_site_sk", sks)
         put("web_site_id", _id_str("AAAAAAAA", sks))
         put("web_name", [f"site_{int(s) % 10}" for s in sks])
+        put("web_company_name", [["pri", "able", "ese", "anti", "cally"][int(s) % 5] for s in sks])
     elif name == "web_page":
         put("wp_web_page_sk", sks)
         put("wp_char_count", rng.integers(100, 8000, n).astype(np.int32))
@@ -464,7 +473,7 @@ def _nparts_for(table: str, sf: float) -> int:
     return 1
 
 
-DATAGEN_VERSION = 4
+DATAGEN_VERSION = 5
 
 
 def dataset_root(root: str, sf: float) -> str:

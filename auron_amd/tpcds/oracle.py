@@ -1806,3 +1806,267 @@ def q74(root, sf):
 
 ORACLES.update({"q4": q4, "q13": q13, "q27": q27, "q30": q30, "q36": q36,
                 "q74": q74, "q76": q76, "q81": q81})
+
+
+# ------------------------------- batch 8 oracles
+def q21(root, sf):
+    import numpy as np
+
+    pivot = _days(2000, 3, 11)
+    inv = _read(root, sf, "inventory")
+    wh = _read(root, sf, "warehouse", ["w_warehouse_sk", "w_warehouse_name"])
+    it = _read(root, sf, "item", ["i_item_sk", "i_item_id", "i_current_price"])
+    it = it[it.i_current_price.between(0.99, 1.49)]
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_date"])
+    di = _date_i(dd)
+    dd = dd.assign(d_i=di)
+    dd = dd[(di >= pivot - 30) & (di <= pivot + 30)]
+    j = _merge(inv, wh, "inv_warehouse_sk", "w_warehouse_sk")
+    j = _merge(j, it, "inv_item_sk", "i_item_sk")
+    j = _merge(j, dd, "inv_date_sk", "d_date_sk")
+    j = j.assign(b=np.where(j.d_i < pivot, j.inv_quantity_on_hand, 0),
+                 a=np.where(j.d_i >= pivot, j.inv_quantity_on_hand, 0))
+    j.loc[j.inv_quantity_on_hand.isna(), ["b", "a"]] = np.nan
+    g = j.groupby(["w_warehouse_name", "i_item_id"], dropna=False).agg(
+        inv_before=("b", lambda x: x.sum(min_count=1)),
+        inv_after=("a", lambda x: x.sum(min_count=1))).reset_index()
+    r = g.inv_after / g.inv_before
+    f = g[((g.inv_before > 0) & (r >= 2.0 / 3.0) & (r <= 1.5)).fillna(False)].copy()
+    for c in ("inv_before", "inv_after"):
+        f[c] = f[c].astype("Int64")
+    f = f.sort_values(["w_warehouse_name", "i_item_id"], na_position="first").head(100)
+    return f.reset_index(drop=True)
+
+
+def q28(root, sf):
+    import pandas as pd
+
+    ss = _read(root, sf, "store_sales", ["ss_quantity", "ss_list_price",
+                                         "ss_coupon_amt", "ss_wholesale_cost"])
+    buckets = [
+        ("b1", 0, 5, 8.0, 459.0, 57.0), ("b2", 6, 10, 90.0, 2323.0, 31.0),
+        ("b3", 11, 15, 142.0, 12214.0, 79.0), ("b4", 16, 20, 135.0, 6071.0, 38.0),
+        ("b5", 21, 25, 122.0, 836.0, 17.0), ("b6", 26, 30, 154.0, 7326.0, 7.0),
+    ]
+    data = {}
+    for name, qlo, qhi, lp, cp, wc in buckets:
+        f = ss[(ss.ss_quantity.between(qlo, qhi)
+                & (ss.ss_list_price.between(lp, lp + 10)
+                   | ss.ss_coupon_amt.between(cp, cp + 1000)
+                   | ss.ss_wholesale_cost.between(wc, wc + 20))).fillna(False)]
+        data[f"{name}_lp"] = [f.ss_list_price.mean() if len(f) else None]
+        data[f"{name}_cnt"] = [int(f.ss_list_price.count())]
+        data[f"{name}_cntd"] = [int(f.ss_list_price.nunique())]
+    return pd.DataFrame(data)
+
+
+def q35(root, sf):
+    cust = _read(root, sf, "customer", ["c_customer_sk", "c_current_addr_sk",
+                                        "c_current_cdemo_sk"])
+    ca = _read(root, sf, "customer_address", ["ca_address_sk", "ca_state"])
+    cd = _read(root, sf, "customer_demographics")
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year", "d_qoy"])
+    dd = dd[(dd.d_year == 2002) & (dd.d_qoy < 4)]
+
+    def custs(fact, pre, fk):
+        fs = _read(root, sf, fact, [f"{pre}_sold_date_sk", fk])
+        jj = _merge(fs, dd, f"{pre}_sold_date_sk", "d_date_sk")
+        return set(jj[fk].dropna())
+
+    sset = custs("store_sales", "ss", "ss_customer_sk")
+    wset = custs("web_sales", "ws", "ws_bill_customer_sk")
+    cset = custs("catalog_sales", "cs", "cs_ship_customer_sk")
+    j = _merge(cust, ca, "c_current_addr_sk", "ca_address_sk")
+    j = j[j.c_customer_sk.isin(sset) & (j.c_customer_sk.isin(wset) | j.c_customer_sk.isin(cset))]
+    j = _merge(j, cd, "c_current_cdemo_sk", "cd_demo_sk")
+    keys = ["ca_state", "cd_gender", "cd_marital_status", "cd_dep_count",
+            "cd_dep_employed_count", "cd_dep_college_count"]
+    g = j.groupby(keys, dropna=False).size().reset_index(name="cnt1")
+    g["cnt2"] = g.cnt1
+    g["cnt3"] = g.cnt1
+    g = g.sort_values(keys, na_position="first").head(100)
+    return g[["ca_state", "cd_gender", "cd_marital_status", "cnt1", "cd_dep_count",
+              "cd_dep_employed_count", "cnt2", "cd_dep_college_count", "cnt3"]] \
+        .reset_index(drop=True)
+
+
+def q56(root, sf):
+    import pandas as pd
+
+    it_all = _read(root, sf, "item", ["i_item_sk", "i_item_id", "i_color"])
+    ids = set(it_all[it_all.i_color.isin(["slate", "blanched", "burnished",
+                                          "red", "blue", "green"])].i_item_id)
+
+    def chan(fact, pre, addr_fk):
+        fs = _read(root, sf, fact, [f"{pre}_item_sk", f"{pre}_sold_date_sk", addr_fk,
+                                    f"{pre}_ext_sales_price"])
+        dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year", "d_moy"])
+        dd = dd[(dd.d_year == 2001) & (dd.d_moy == 2)]
+        ca = _read(root, sf, "customer_address", ["ca_address_sk", "ca_gmt_offset"])
+        ca = ca[ca.ca_gmt_offset == -5.0]
+        j = _merge(fs, dd, f"{pre}_sold_date_sk", "d_date_sk")
+        j = _merge(j, ca, addr_fk, "ca_address_sk")
+        j = _merge(j, it_all[it_all.i_item_id.isin(ids)], f"{pre}_item_sk", "i_item_sk")
+        return j.groupby("i_item_id", dropna=False)[f"{pre}_ext_sales_price"] \
+                .sum(min_count=1).reset_index(name="total_sales")
+
+    u = pd.concat([chan("store_sales", "ss", "ss_addr_sk"),
+                   chan("catalog_sales", "cs", "cs_bill_addr_sk"),
+                   chan("web_sales", "ws", "ws_bill_addr_sk")], ignore_index=True)
+    g = u.groupby("i_item_id", dropna=False).total_sales.sum(min_count=1) \
+         .reset_index(name="total_sales")
+    g = g.sort_values(["total_sales", "i_item_id"], na_position="first").head(100)
+    return g.reset_index(drop=True)
+
+
+def q59(root, sf):
+    ss = _read(root, sf, "store_sales", ["ss_sold_date_sk", "ss_store_sk", "ss_sales_price"])
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_week_seq", "d_day_name", "d_month_seq"])
+    j = _merge(ss, dd, "ss_sold_date_sk", "d_date_sk")
+    days = [("sun", "Sunday"), ("mon", "Monday"), ("tue", "Tuesday"),
+            ("wed", "Wednesday"), ("thu", "Thursday"), ("fri", "Friday"),
+            ("sat", "Saturday")]
+    for t, day in days:
+        j[f"{t}_sales"] = j.ss_sales_price.where(j.d_day_name == day)
+    wss = j.groupby(["d_week_seq", "ss_store_sk"], dropna=False).agg(
+        **{f"{t}_sales": (f"{t}_sales", lambda x: x.sum(min_count=1)) for t, _ in days}) \
+        .reset_index()
+    st = _read(root, sf, "store", ["s_store_sk", "s_store_name", "s_store_id"])
+    weeks = dd[["d_week_seq", "d_month_seq"]].drop_duplicates()
+
+    def block(mlo):
+        wk = set(weeks[weeks.d_month_seq.between(mlo, mlo + 11)].d_week_seq)
+        b = wss[wss.d_week_seq.isin(wk)]
+        return _merge(b, st, "ss_store_sk", "s_store_sk")
+
+    y = block(1212)
+    x = block(1224).copy()
+    x["wk_join"] = x.d_week_seq - 52
+    m = y.merge(x, left_on=["s_store_id", "d_week_seq"],
+                right_on=["s_store_id", "wk_join"], suffixes=("1", "2"))
+    out = {"s_store_name1": m.s_store_name1, "s_store_id1": m.s_store_id,
+           "d_week_seq1": m.d_week_seq1}
+    for t, _ in days:
+        out[f"{t}_ratio"] = m[f"{t}_sales1"] / m[f"{t}_sales2"]
+    import pandas as pd
+
+    # full (pre-LIMIT) result: the ORDER BY keys tie across store_sk pairs
+    # sharing a store_id, so which 100 rows survive the LIMIT is
+    # engine-dependent; the test checks subset membership instead
+    o = pd.DataFrame(out).sort_values(["s_store_name1", "s_store_id1", "d_week_seq1"],
+                                      na_position="first")
+    return o.reset_index(drop=True)
+
+
+def q71(root, sf):
+    import pandas as pd
+
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_moy", "d_year"])
+    dd = dd[(dd.d_moy == 11) & (dd.d_year == 1999)]
+
+    def chan(fact, pre):
+        fs = _read(root, sf, fact, [f"{pre}_ext_sales_price", f"{pre}_sold_date_sk",
+                                    f"{pre}_item_sk", f"{pre}_sold_time_sk"])
+        j = _merge(fs, dd, f"{pre}_sold_date_sk", "d_date_sk")
+        return pd.DataFrame({"ext_price": j[f"{pre}_ext_sales_price"],
+                             "sold_item_sk": j[f"{pre}_item_sk"],
+                             "time_sk": j[f"{pre}_sold_time_sk"]})
+
+    u = pd.concat([chan("web_sales", "ws"), chan("catalog_sales", "cs"),
+                   chan("store_sales", "ss")], ignore_index=True)
+    it = _read(root, sf, "item", ["i_item_sk", "i_brand_id", "i_brand", "i_manager_id"])
+    it = it[it.i_manager_id == 1]
+    td = _read(root, sf, "time_dim")
+    td = td[td.t_meal_time.isin(["breakfast", "dinner"])]
+    j = _merge(u, it, "sold_item_sk", "i_item_sk")
+    j = _merge(j, td, "time_sk", "t_time_sk")
+    g = j.groupby(["i_brand_id", "i_brand", "t_hour", "t_minute"], dropna=False) \
+         .ext_price.sum(min_count=1).reset_index(name="ext_price")
+    g = g.rename(columns={"i_brand_id": "brand_id", "i_brand": "brand"})
+    g = g.sort_values(["ext_price", "brand_id"], ascending=[False, True],
+                      na_position="first")
+    return g[["brand_id", "brand", "t_hour", "t_minute", "ext_price"]].reset_index(drop=True)
+
+
+def q84(root, sf):
+    cust = _read(root, sf, "customer", ["c_customer_sk", "c_customer_id", "c_first_name",
+                                        "c_last_name", "c_current_addr_sk",
+                                        "c_current_cdemo_sk", "c_current_hdemo_sk"])
+    ca = _read(root, sf, "customer_address", ["ca_address_sk", "ca_city"])
+    ca = ca[ca.ca_city == "Fairview"]
+    ib = _read(root, sf, "income_band")
+    ib = ib[(ib.ib_lower_bound >= 38128) & (ib.ib_upper_bound <= 88128)]
+    hd = _read(root, sf, "household_demographics", ["hd_demo_sk", "hd_income_band_sk"])
+    sr = _read(root, sf, "store_returns", ["sr_cdemo_sk"])
+    j = _merge(cust, ca, "c_current_addr_sk", "ca_address_sk")
+    j = _merge(j, hd, "c_current_hdemo_sk", "hd_demo_sk")
+    j = _merge(j, ib, "hd_income_band_sk", "ib_income_band_sk")
+    j = _merge(j, sr, "c_current_cdemo_sk", "sr_cdemo_sk")
+    j["customername"] = j.c_last_name + ", " + j.c_first_name
+    out = j[["c_customer_id", "customername"]].rename(columns={"c_customer_id": "customer_id"})
+    out = out.sort_values("customer_id", na_position="first").head(100)
+    return out.reset_index(drop=True)
+
+
+def q86(root, sf):
+    import numpy as np
+    import pandas as pd
+
+    ws = _read(root, sf, "web_sales", ["ws_sold_date_sk", "ws_item_sk", "ws_net_paid"])
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_month_seq"])
+    dd = dd[dd.d_month_seq.between(1200, 1211)]
+    it = _read(root, sf, "item", ["i_item_sk", "i_category", "i_class"])
+    j = _merge(ws, dd, "ws_sold_date_sk", "d_date_sk")
+    j = _merge(j, it, "ws_item_sk", "i_item_sk")
+    frames = []
+    for depth, loc in ((2, 0), (1, 1), (0, 2)):
+        t = j.copy()
+        if depth < 2:
+            t["i_class"] = None
+        if depth < 1:
+            t["i_category"] = None
+        g = t.groupby(["i_category", "i_class"], dropna=False).ws_net_paid \
+             .sum(min_count=1).reset_index(name="total_sum")
+        g["lochierarchy"] = loc
+        frames.append(g)
+    out = pd.concat(frames, ignore_index=True)
+    out["_pcat"] = out.i_category.where(out.lochierarchy == 0)
+    out["_key"] = np.trunc(out.total_sum * 100.0 + 0.5)
+    out["rank_within_parent"] = out.groupby(["lochierarchy", "_pcat"], dropna=False) \
+        ._key.rank(method="min", ascending=False)
+    out = out.sort_values(["lochierarchy", "_pcat", "rank_within_parent"],
+                          ascending=[False, True, True], na_position="first").head(100)
+    out["rank_within_parent"] = out.rank_within_parent.astype(int)
+    return out[["total_sum", "i_category", "i_class", "lochierarchy",
+                "rank_within_parent"]].reset_index(drop=True)
+
+
+def q97(root, sf):
+    import pandas as pd
+
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_month_seq"])
+    dd = dd[dd.d_month_seq.between(1200, 1211)]
+
+    def ci(fact, pre, fk):
+        fs = _read(root, sf, fact, [f"{pre}_sold_date_sk", fk, f"{pre}_item_sk"])
+        j = _merge(fs, dd, f"{pre}_sold_date_sk", "d_date_sk")
+        return j[[fk, f"{pre}_item_sk"]].drop_duplicates().rename(
+            columns={fk: "csk", f"{pre}_item_sk": "isk"})
+
+    ssci = ci("store_sales", "ss", "ss_customer_sk")
+    csci = ci("catalog_sales", "cs", "cs_bill_customer_sk")
+    j = ssci.dropna().merge(csci.dropna(), on=["csk", "isk"], how="outer",
+                            indicator=True)
+    # SQL full outer never matches null keys; null-key distinct pairs appear
+    # unmatched on their own side
+    null_ss = ssci[ssci.csk.isna() | ssci.isk.isna()]
+    null_cs = csci[csci.csk.isna() | csci.isk.isna()]
+    store_only = int((j._merge == "left_only").sum()) + len(null_ss[null_ss.csk.notna()])
+    catalog_only = int((j._merge == "right_only").sum()) + len(null_cs[null_cs.csk.notna()])
+    both = int((j._merge == "both").sum())
+    return pd.DataFrame({"store_only": [store_only], "catalog_only": [catalog_only],
+                         "store_and_catalog": [both]})
+
+
+ORACLES.update({"q21": q21, "q28": q28, "q35": q35, "q56": q56, "q59": q59,
+                "q71": q71, "q84": q84, "q86": q86, "q97": q97})

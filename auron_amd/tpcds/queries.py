@@ -2128,3 +2128,308 @@ def q74(cat, s):
 
 QUERIES.update({"q4": q4, "q13": q13, "q27": q27, "q30": q30, "q36": q36,
                 "q74": q74, "q76": q76, "q81": q81})
+
+
+# ------------------------------- batch 8
+def q21(cat, s):
+    from ..exprs import CaseWhen
+
+    pivot = _days(2000, 3, 11)
+    inv = cat.scan("inventory", ["inv_item_sk", "inv_warehouse_sk", "inv_date_sk",
+                                 "inv_quantity_on_hand"])
+    wh = cat.scan("warehouse", ["w_warehouse_sk", "w_warehouse_name"])
+    it = P.Filter(cat.scan("item", ["i_item_sk", "i_item_id", "i_current_price"]),
+                  col("i_current_price").between(0.99, 1.49))
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_date"]),
+                  col("d_date").cast(dtypes.int32).between(pivot - 30, pivot + 30))
+    j = bhj(inv, wh, ["inv_warehouse_sk"], ["w_warehouse_sk"])
+    j = bhj(j, it, ["inv_item_sk"], ["i_item_sk"])
+    j = bhj(j, dd, ["inv_date_sk"], ["d_date_sk"])
+    before = CaseWhen([(col("d_date").cast(dtypes.int32) < pivot,
+                        col("inv_quantity_on_hand"))], lit(0))
+    after = CaseWhen([(col("d_date").cast(dtypes.int32) >= pivot,
+                       col("inv_quantity_on_hand"))], lit(0))
+    pre = P.Project(j, [_a(col("w_warehouse_name"), "w_warehouse_name"),
+                        _a(col("i_item_id"), "i_item_id"),
+                        _a(before, "b"), _a(after, "a")])
+    agg = agg2(pre, ["w_warehouse_name", "i_item_id"],
+               [AggFunc("sum", col("b"), name="inv_before"),
+                AggFunc("sum", col("a"), name="inv_after")])
+    ratio = col("inv_after").cast(dtypes.float64) / col("inv_before").cast(dtypes.float64)
+    f = P.Filter(agg, (col("inv_before") > 0) & (ratio >= 2.0 / 3.0) & (ratio <= 1.5))
+    return topk(f, [(col("w_warehouse_name"), True), (col("i_item_id"), True)], 100)
+
+
+def q28(cat, s):
+    buckets = [
+        ("b1", 0, 5, 8.0, 459.0, 57.0), ("b2", 6, 10, 90.0, 2323.0, 31.0),
+        ("b3", 11, 15, 142.0, 12214.0, 79.0), ("b4", 16, 20, 135.0, 6071.0, 38.0),
+        ("b5", 21, 25, 122.0, 836.0, 17.0), ("b6", 26, 30, 154.0, 7326.0, 7.0),
+    ]
+    vals = {}
+    for name, qlo, qhi, lp, cp, wc in buckets:
+        ss = cat.scan("store_sales", ["ss_quantity", "ss_list_price",
+                                      "ss_coupon_amt", "ss_wholesale_cost"])
+        f = P.Filter(ss, col("ss_quantity").between(qlo, qhi)
+                     & (col("ss_list_price").between(lp, lp + 10)
+                        | col("ss_coupon_amt").between(cp, cp + 1000)
+                        | col("ss_wholesale_cost").between(wc, wc + 20)))
+        a = _global_agg(f, [AggFunc("avg", col("ss_list_price"), name="lp"),
+                            AggFunc("count", col("ss_list_price"), name="cnt"),
+                            AggFunc("count_distinct", col("ss_list_price"), name="cntd")])
+        d = s.collect_all(a).to_pydict()
+        vals[name] = (d["lp"][0] if d["lp"] else None,
+                      d["cnt"][0] if d["cnt"] else 0,
+                      d["cntd"][0] if d["cntd"] else 0)
+    from ..column import RecordBatch
+
+    n = 1 if s.rank == 0 else 0
+    data = {}
+    types = {}
+    for name, _, _, _, _, _ in buckets:
+        lp, cnt, cntd = vals[name]
+        data[f"{name}_lp"] = [lp] * max(n, 1)
+        data[f"{name}_cnt"] = [cnt] * max(n, 1)
+        data[f"{name}_cntd"] = [cntd] * max(n, 1)
+        types[f"{name}_lp"] = dtypes.float64
+        types[f"{name}_cnt"] = dtypes.int64
+        types[f"{name}_cntd"] = dtypes.int64
+    b = RecordBatch.from_pydict(data, types)
+    if n == 0:
+        b = b.slice(0, 0)
+    return P.MemoryScan([b])
+
+
+def q35(cat, s):
+    cust = cat.scan("customer", ["c_customer_sk", "c_current_addr_sk", "c_current_cdemo_sk"])
+    ca = cat.scan("customer_address", ["ca_address_sk", "ca_state"])
+    cd = cat.scan("customer_demographics",
+                  ["cd_demo_sk", "cd_gender", "cd_marital_status", "cd_dep_count",
+                   "cd_dep_employed_count", "cd_dep_college_count"])
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_year", "d_qoy"]),
+                  (col("d_year") == 2002) & (col("d_qoy") < 4))
+
+    def channel_cust(fact, pre, fk):
+        fs = cat.scan(fact, [f"{pre}_sold_date_sk", fk])
+        jj = bhj(fs, dd, [f"{pre}_sold_date_sk"], ["d_date_sk"])
+        return P.HashAgg(P.Exchange(P.Project(jj, [_a(col(fk), "xck")]), "hash", [col("xck")]),
+                         [_a(col("xck"), "xck")], [], mode="complete")
+
+    wb = P.Union([channel_cust("web_sales", "ws", "ws_bill_customer_sk"),
+                  channel_cust("catalog_sales", "cs", "cs_ship_customer_sk")])
+    j = bhj(cust, ca, ["c_current_addr_sk"], ["ca_address_sk"])
+    j = P.Exchange(j, "hash", [col("c_customer_sk")])
+    j = P.HashJoin(j, channel_cust("store_sales", "ss", "ss_customer_sk"),
+                   [col("c_customer_sk")], [col("xck")], how="semi", build_side="right")
+    j = P.HashJoin(j, P.Exchange(wb, "hash", [col("xck")]),
+                   [col("c_customer_sk")], [col("xck")], how="semi", build_side="right")
+    j = bhj(j, cd, ["c_current_cdemo_sk"], ["cd_demo_sk"])
+    keys = ["ca_state", "cd_gender", "cd_marital_status", "cd_dep_count",
+            "cd_dep_employed_count", "cd_dep_college_count"]
+    a = agg2(j, keys, [AggFunc("count_star", None, name="cnt1")])
+    proj = P.Project(a, [_a(col("ca_state"), "ca_state"), _a(col("cd_gender"), "cd_gender"),
+                         _a(col("cd_marital_status"), "cd_marital_status"),
+                         _a(col("cnt1"), "cnt1"),
+                         _a(col("cd_dep_count"), "cd_dep_count"),
+                         _a(col("cd_dep_employed_count"), "cd_dep_employed_count"),
+                         _a(col("cnt1"), "cnt2"),
+                         _a(col("cd_dep_college_count"), "cd_dep_college_count"),
+                         _a(col("cnt1"), "cnt3")])
+    return topk(proj, [(col(k), True) for k in keys], 100)
+
+
+def q56(cat, s):
+    # q60-shape with i_item_id restricted by item color subset
+    def chan(fact, pre, addr_fk):
+        fs = cat.scan(fact, [f"{pre}_item_sk", f"{pre}_sold_date_sk", addr_fk,
+                             f"{pre}_ext_sales_price"])
+        dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_year", "d_moy"]),
+                      (col("d_year") == 2001) & (col("d_moy") == 2))
+        ca = P.Filter(cat.scan("customer_address", ["ca_address_sk", "ca_gmt_offset"]),
+                      col("ca_gmt_offset") == -5.0)
+        it = cat.scan("item", ["i_item_sk", "i_item_id"])
+        sub = P.Filter(cat.scan("item", ["i_item_id", "i_color"]),
+                       col("i_color").isin(["slate", "blanched", "burnished"] +
+                                           ["red", "blue", "green"]))
+        it_f = P.HashJoin(it, P.Project(sub, [_a(col("i_item_id"), "_sub")]),
+                          [col("i_item_id")], [col("_sub")], how="semi",
+                          build_side="right", broadcast=True)
+        j = bhj(fs, dd, [f"{pre}_sold_date_sk"], ["d_date_sk"])
+        j = bhj(j, ca, [addr_fk], ["ca_address_sk"])
+        j = bhj(j, it_f, [f"{pre}_item_sk"], ["i_item_sk"])
+        return agg2(j, ["i_item_id"], [AggFunc("sum", col(f"{pre}_ext_sales_price"),
+                                               name="total_sales")])
+
+    u = P.Union([chan("store_sales", "ss", "ss_addr_sk"),
+                 chan("catalog_sales", "cs", "cs_bill_addr_sk"),
+                 chan("web_sales", "ws", "ws_bill_addr_sk")])
+    a = agg2(u, ["i_item_id"], [AggFunc("sum", col("total_sales"), name="total_sales")])
+    return topk(a, [(col("total_sales"), True), (col("i_item_id"), True)], 100)
+
+
+def q59(cat, s):
+    from ..exprs import CaseWhen, Literal
+
+    ss = cat.scan("store_sales", ["ss_sold_date_sk", "ss_store_sk", "ss_sales_price"])
+    dd = cat.scan("date_dim", ["d_date_sk", "d_week_seq", "d_day_name"])
+    j = bhj(ss, dd, ["ss_sold_date_sk"], ["d_date_sk"])
+    days = [("sun", "Sunday"), ("mon", "Monday"), ("tue", "Tuesday"),
+            ("wed", "Wednesday"), ("thu", "Thursday"), ("fri", "Friday"),
+            ("sat", "Saturday")]
+    aggs = [AggFunc("sum", CaseWhen([(col("d_day_name") == lit(day), col("ss_sales_price"))],
+                                    Literal(None, dtypes.float64)), name=f"{t}_sales")
+            for t, day in days]
+    wss = agg2(j, ["d_week_seq", "ss_store_sk"], aggs)
+    wss_b = s.execute(wss)
+
+    dweeks = cat.scan("date_dim", ["d_week_seq", "d_month_seq"])
+
+    def year_block(mlo, suffix, cols):
+        wk = P.HashAgg(P.Broadcast(P.Filter(dweeks, col("d_month_seq").between(mlo, mlo + 11))),
+                       [_a(col("d_week_seq"), "wk")], [], mode="complete")
+        jj = P.HashJoin(P.MemoryScan(wss_b), wk, [col("d_week_seq")], [col("wk")],
+                        how="semi", build_side="right")
+        st = cat.scan("store", ["s_store_sk", "s_store_name", "s_store_id"])
+        jj = bhj(jj, st, ["ss_store_sk"], ["s_store_sk"])
+        ren = [_a(col("d_week_seq"), f"d_week_seq{suffix}"),
+               _a(col("s_store_id"), f"s_store_id{suffix}")]
+        if suffix == "1":
+            ren.append(_a(col("s_store_name"), "s_store_name1"))
+        ren += [_a(col(f"{t}_sales"), f"{t}_sales{suffix}") for t, _ in days]
+        return P.Project(jj, ren)
+
+    y = year_block(1212, "1", None)
+    x = year_block(1224, "2", None)
+    x2 = P.Project(x, [_a(col("d_week_seq2") - lit(52), "wk_join"),
+                       _a(col("s_store_id2"), "s_store_id2")]
+                   + [_a(col(f"{t}_sales2"), f"{t}_sales2") for t, _ in days])
+    j2 = shj(y, x2, ["s_store_id1", "d_week_seq1"], ["s_store_id2", "wk_join"])
+    proj = P.Project(j2, [_a(col("s_store_name1"), "s_store_name1"),
+                          _a(col("s_store_id1"), "s_store_id1"),
+                          _a(col("d_week_seq1"), "d_week_seq1")]
+                     + [_a(col(f"{t}_sales1") / col(f"{t}_sales2"), f"{t}_ratio")
+                        for t, _ in days])
+    return topk(proj, [(col("s_store_name1"), True), (col("s_store_id1"), True),
+                       (col("d_week_seq1"), True)], 100)
+
+
+def q71(cat, s):
+    def chan(fact, pre):
+        fs = cat.scan(fact, [f"{pre}_ext_sales_price", f"{pre}_sold_date_sk",
+                             f"{pre}_item_sk", f"{pre}_sold_time_sk"])
+        dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_moy", "d_year"]),
+                      (col("d_moy") == 11) & (col("d_year") == 1999))
+        j = bhj(fs, dd, [f"{pre}_sold_date_sk"], ["d_date_sk"])
+        return P.Project(j, [_a(col(f"{pre}_ext_sales_price"), "ext_price"),
+                             _a(col(f"{pre}_item_sk"), "sold_item_sk"),
+                             _a(col(f"{pre}_sold_time_sk"), "time_sk")])
+
+    u = P.Union([chan("web_sales", "ws"), chan("catalog_sales", "cs"),
+                 chan("store_sales", "ss")])
+    it = P.Filter(cat.scan("item", ["i_item_sk", "i_brand_id", "i_brand", "i_manager_id"]),
+                  col("i_manager_id") == 1)
+    td = P.Filter(cat.scan("time_dim", ["t_time_sk", "t_hour", "t_minute", "t_meal_time"]),
+                  (col("t_meal_time") == lit("breakfast")) | (col("t_meal_time") == lit("dinner")))
+    j = bhj(u, it, ["sold_item_sk"], ["i_item_sk"])
+    j = bhj(j, td, ["time_sk"], ["t_time_sk"])
+    a = agg2(j, ["brand_id", "brand", "t_hour", "t_minute"],
+             [AggFunc("sum", col("ext_price"), name="ext_price")],
+             key_exprs=[col("i_brand_id"), col("i_brand"), col("t_hour"), col("t_minute")])
+    return topk(a, [(col("ext_price"), False), (col("brand_id"), True)], 100000)
+
+
+def q84(cat, s):
+    from ..exprs import ConcatStr
+
+    cust = cat.scan("customer", ["c_customer_sk", "c_customer_id", "c_first_name",
+                                 "c_last_name", "c_current_addr_sk", "c_current_cdemo_sk",
+                                 "c_current_hdemo_sk"])
+    # adapted literal: 'Edgewood' is not in the synthetic city list -> 'Fairview'
+    ca = P.Filter(cat.scan("customer_address", ["ca_address_sk", "ca_city"]),
+                  col("ca_city") == lit("Fairview"))
+    ib = P.Filter(cat.scan("income_band", ["ib_income_band_sk", "ib_lower_bound",
+                                           "ib_upper_bound"]),
+                  (col("ib_lower_bound") >= 38128) & (col("ib_upper_bound") <= 88128))
+    hd = cat.scan("household_demographics", ["hd_demo_sk", "hd_income_band_sk"])
+    sr = cat.scan("store_returns", ["sr_cdemo_sk"])
+    j = bhj(cust, ca, ["c_current_addr_sk"], ["ca_address_sk"])
+    j = bhj(j, hd, ["c_current_hdemo_sk"], ["hd_demo_sk"])
+    j = bhj(j, ib, ["hd_income_band_sk"], ["ib_income_band_sk"])
+    j = P.HashJoin(P.Exchange(j, "hash", [col("c_current_cdemo_sk")]),
+                   P.Exchange(sr, "hash", [col("sr_cdemo_sk")]),
+                   [col("c_current_cdemo_sk")], [col("sr_cdemo_sk")],
+                   how="inner", build_side="right")
+    proj = P.Project(j, [_a(col("c_customer_id"), "customer_id"),
+                         _a(ConcatStr([col("c_last_name"), lit(", "), col("c_first_name")]),
+                            "customername")])
+    return topk(proj, [(col("customer_id"), True)], 100)
+
+
+def q86(cat, s):
+    from ..exprs import CaseWhen, Literal, WindowFunc
+
+    ws = cat.scan("web_sales", ["ws_sold_date_sk", "ws_item_sk", "ws_net_paid"])
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_month_seq"]),
+                  col("d_month_seq").between(1200, 1211))
+    it = cat.scan("item", ["i_item_sk", "i_category", "i_class"])
+    j = bhj(ws, dd, ["ws_sold_date_sk"], ["d_date_sk"])
+    j = bhj(j, it, ["ws_item_sk"], ["i_item_sk"])
+    pre = P.Project(j, [_a(col("i_category"), "i_category"), _a(col("i_class"), "i_class"),
+                        _a(col("ws_net_paid"), "np")])
+    ex = rollup_expand(pre, ["i_category", "i_class"], [dtypes.string, dtypes.string],
+                       ["np"])
+    a = agg2(ex, ["i_category", "i_class", "_lochier"],
+             [AggFunc("sum", col("np"), name="total_sum")])
+    m = P.Project(a, [_a(col("total_sum"), "total_sum"), _a(col("i_category"), "i_category"),
+                      _a(col("i_class"), "i_class"), _a(col("_lochier"), "lochierarchy"),
+                      _a(CaseWhen([(col("_lochier") == 0, col("i_category"))],
+                                  Literal(None, dtypes.string)), "_pcat")])
+    # rank on cent-truncated total to keep tie-breaks comparable
+    rank_key = ((col("total_sum") * lit(100.0)) + lit(0.5)).cast(dtypes.int64)
+    w = P.Window(P.Exchange(m, "hash", [col("lochierarchy")]),
+                 [col("lochierarchy"), col("_pcat")], [(rank_key, False)],
+                 [_a(WindowFunc("rank"), "rank_within_parent")])
+    proj = P.Project(w, [_a(col("total_sum"), "total_sum"),
+                         _a(col("i_category"), "i_category"),
+                         _a(col("i_class"), "i_class"),
+                         _a(col("lochierarchy"), "lochierarchy"),
+                         _a(col("rank_within_parent"), "rank_within_parent"),
+                         _a(col("_pcat"), "_pcat")])
+    out = topk(proj, [(col("lochierarchy"), False), (col("_pcat"), True),
+                      (col("rank_within_parent"), True)], 100)
+    return P.Project(out, [_a(col(c), c) for c in
+                           ["total_sum", "i_category", "i_class", "lochierarchy",
+                            "rank_within_parent"]])
+
+
+def q97(cat, s):
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_month_seq"]),
+                  col("d_month_seq").between(1200, 1211))
+
+    def ci(fact, pre, fk):
+        fs = cat.scan(fact, [f"{pre}_sold_date_sk", fk, f"{pre}_item_sk"])
+        j = bhj(fs, dd, [f"{pre}_sold_date_sk"], ["d_date_sk"])
+        return agg2(j, [f"{pre}_csk", f"{pre}_isk"], [],
+                    key_exprs=[col(fk), col(f"{pre}_item_sk")])
+
+    ssci = ci("store_sales", "ss", "ss_customer_sk")
+    csci = ci("catalog_sales", "cs", "cs_bill_customer_sk")
+    j = shj(ssci, csci, ["ss_csk", "ss_isk"], ["cs_csk", "cs_isk"], how="full")
+    from ..exprs import CaseWhen, IsNull, Not
+
+    store_only = CaseWhen([(Not(IsNull(col("ss_csk"))) & IsNull(col("cs_csk")), lit(1))], lit(0))
+    cat_only = CaseWhen([(IsNull(col("ss_csk")) & Not(IsNull(col("cs_csk"))), lit(1))], lit(0))
+    both = CaseWhen([(Not(IsNull(col("ss_csk"))) & Not(IsNull(col("cs_csk"))), lit(1))], lit(0))
+    pre = P.Project(j, [_a(store_only, "so"), _a(cat_only, "co"), _a(both, "bo")])
+    partial = P.HashAgg(pre, [], [AggFunc("sum", col("so"), name="store_only"),
+                                  AggFunc("sum", col("co"), name="catalog_only"),
+                                  AggFunc("sum", col("bo"), name="store_and_catalog")],
+                        mode="partial")
+    return P.HashAgg(P.Exchange(partial, "single"), [],
+                     [AggFunc("sum", col("so"), name="store_only"),
+                      AggFunc("sum", col("co"), name="catalog_only"),
+                      AggFunc("sum", col("bo"), name="store_and_catalog")], mode="final")
+
+
+QUERIES.update({"q21": q21, "q28": q28, "q35": q35, "q56": q56, "q59": q59,
+                "q71": q71, "q84": q84, "q86": q86, "q97": q97})

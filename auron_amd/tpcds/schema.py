@@ -20,12 +20,12 @@ SCHEMAS = {
         "d_dom": i32, "d_qoy": i32, "d_day_name": s, "d_month_seq": i32,
         "d_week_seq": i32, "d_dow": i32,
     },
-    "time_dim": {"t_time_sk": i64, "t_hour": i32, "t_minute": i32},
+    "time_dim": {"t_time_sk": i64, "t_hour": i32, "t_minute": i32, "t_meal_time": s},
     "item": {
         "i_item_sk": i64, "i_item_id": s, "i_item_desc": s, "i_category": s,
         "i_category_id": i32, "i_brand": s, "i_brand_id": i32,
         "i_manufact_id": i32, "i_class": s, "i_class_id": i32,
-        "i_current_price": f64, "i_manager_id": i32, "i_product_name": s,
+        "i_current_price": f64, "i_manager_id": i32, "i_product_name": s, "i_color": s,
     },
     "customer": {
         "c_customer_sk": i64, "c_customer_id": s, "c_first_name": s,
@@ -68,7 +68,7 @@ SCHEMAS = {
         "cc_call_center_sk": i64, "cc_call_center_id": s, "cc_county": s,
         "cc_name": s, "cc_manager": s,
     },
-    "web_site": {"web_site_sk": i64, "web_site_id": s, "web_name": s},
+    "web_site": {"web_site_sk": i64, "web_site_id": s, "web_name": s, "web_company_name": s},
     "web_page": {"wp_web_page_sk": i64, "wp_char_count": i32},
     "catalog_page": {"cp_catalog_page_sk": i64, "cp_catalog_page_id": s},
     "store_sales": {
