@@ -880,7 +880,7 @@ def q16(root, sf):
 def q94(root, sf):
     return _ship_oracle(root, sf, "web_sales", "ws", (1999, 2, 1), "IL",
                         ("ws_web_site_sk", "web_site", "web_site_sk",
-                         "web_name", "site_1"), "web_returns", "wr")
+                         "web_company_name", "pri"), "web_returns", "wr")
 
 
 def _discount_oracle(root, sf, fact, pre, mfg, date0):
@@ -2070,3 +2070,317 @@ def q97(root, sf):
 
 ORACLES.update({"q21": q21, "q28": q28, "q35": q35, "q56": q56, "q59": q59,
                 "q71": q71, "q84": q84, "q86": q86, "q97": q97})
+
+
+# ------------------------------- batch 9 oracles
+def q2(root, sf):
+    import numpy as np
+    import pandas as pd
+
+    ws = _read(root, sf, "web_sales", ["ws_sold_date_sk", "ws_ext_sales_price"])
+    cs = _read(root, sf, "catalog_sales", ["cs_sold_date_sk", "cs_ext_sales_price"])
+    u = pd.concat([ws.rename(columns={"ws_sold_date_sk": "sold_date_sk",
+                                      "ws_ext_sales_price": "sales_price"}),
+                   cs.rename(columns={"cs_sold_date_sk": "sold_date_sk",
+                                      "cs_ext_sales_price": "sales_price"})],
+                  ignore_index=True)
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_week_seq", "d_day_name", "d_year"])
+    j = _merge(u, dd, "sold_date_sk", "d_date_sk")
+    days = [("sun", "Sunday"), ("mon", "Monday"), ("tue", "Tuesday"),
+            ("wed", "Wednesday"), ("thu", "Thursday"), ("fri", "Friday"),
+            ("sat", "Saturday")]
+    for t, day in days:
+        j[f"{t}_sales"] = j.sales_price.where(j.d_day_name == day)
+    wswscs = j.groupby("d_week_seq", dropna=False).agg(
+        **{f"{t}_sales": (f"{t}_sales", lambda x: x.sum(min_count=1)) for t, _ in days}) \
+        .reset_index()
+    weeks = dd[["d_week_seq", "d_year"]].drop_duplicates()
+    y = wswscs[wswscs.d_week_seq.isin(set(weeks[weeks.d_year == 2001].d_week_seq))]
+    z = wswscs[wswscs.d_week_seq.isin(set(weeks[weeks.d_year == 2002].d_week_seq))].copy()
+    z["wk_join"] = z.d_week_seq - 53
+    m = y.merge(z, left_on="d_week_seq", right_on="wk_join", suffixes=("1", "2"))
+    out = {"d_week_seq1": m.d_week_seq1}
+    for t, _ in days:
+        r = m[f"{t}_sales1"] / m[f"{t}_sales2"]
+        out[f"{t}_r"] = np.floor(r * 100.0 + 0.5) / 100.0
+    o = pd.DataFrame(out).sort_values("d_week_seq1")
+    return o.reset_index(drop=True)
+
+
+def q9(root, sf):
+    import pandas as pd
+
+    ss = _read(root, sf, "store_sales", ["ss_quantity", "ss_ext_discount_amt", "ss_net_paid"])
+    buckets = [(1, 20, 62316685), (21, 40, 19045798), (41, 60, 365541424),
+               (61, 80, 216357808), (81, 100, 184483884)]
+    data = {}
+    for i, (lo, hi, th) in enumerate(buckets):
+        f = ss[ss.ss_quantity.between(lo, hi).fillna(False)]
+        c = len(f)
+        v = f.ss_ext_discount_amt.mean() if c > th else f.ss_net_paid.mean()
+        data[f"bucket{i + 1}"] = [None if pd.isna(v) else v]
+    return pd.DataFrame(data)
+
+
+def q10(root, sf):
+    cust = _read(root, sf, "customer", ["c_customer_sk", "c_current_addr_sk",
+                                        "c_current_cdemo_sk"])
+    ca = _read(root, sf, "customer_address", ["ca_address_sk", "ca_county"])
+    ca = ca[ca.ca_county.isin(["Midway County", "Fairview County", "Oak Grove County",
+                               "Salem County", "Liberty County"])]
+    cd = _read(root, sf, "customer_demographics")
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year", "d_moy"])
+    dd = dd[(dd.d_year == 2002) & dd.d_moy.between(1, 4)]
+
+    def custs(fact, pre, fk):
+        fs = _read(root, sf, fact, [f"{pre}_sold_date_sk", fk])
+        jj = _merge(fs, dd, f"{pre}_sold_date_sk", "d_date_sk")
+        return set(jj[fk].dropna())
+
+    sset = custs("store_sales", "ss", "ss_customer_sk")
+    wset = custs("web_sales", "ws", "ws_bill_customer_sk")
+    cset = custs("catalog_sales", "cs", "cs_ship_customer_sk")
+    j = _merge(cust, ca, "c_current_addr_sk", "ca_address_sk")
+    j = j[j.c_customer_sk.isin(sset) & (j.c_customer_sk.isin(wset) | j.c_customer_sk.isin(cset))]
+    j = _merge(j, cd, "c_current_cdemo_sk", "cd_demo_sk")
+    keys = ["cd_gender", "cd_marital_status", "cd_education_status",
+            "cd_purchase_estimate", "cd_credit_rating", "cd_dep_count",
+            "cd_dep_employed_count", "cd_dep_college_count"]
+    g = j.groupby(keys, dropna=False).size().reset_index(name="cnt1")
+    for i in range(2, 7):
+        g[f"cnt{i}"] = g.cnt1
+    g = g.sort_values(keys, na_position="first").head(100)
+    cols = ["cd_gender", "cd_marital_status", "cd_education_status", "cnt1",
+            "cd_purchase_estimate", "cnt2", "cd_credit_rating", "cnt3",
+            "cd_dep_count", "cnt4", "cd_dep_employed_count", "cnt5",
+            "cd_dep_college_count", "cnt6"]
+    return g[cols].reset_index(drop=True)
+
+
+def q17(root, sf):
+    ss = _read(root, sf, "store_sales", ["ss_sold_date_sk", "ss_item_sk", "ss_store_sk",
+                                         "ss_customer_sk", "ss_ticket_number", "ss_quantity"])
+    sr = _read(root, sf, "store_returns", ["sr_returned_date_sk", "sr_item_sk",
+                                           "sr_customer_sk", "sr_ticket_number",
+                                           "sr_return_quantity"])
+    cs = _read(root, sf, "catalog_sales", ["cs_sold_date_sk", "cs_bill_customer_sk",
+                                           "cs_item_sk", "cs_quantity"])
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year", "d_qoy"])
+    d1 = dd[(dd.d_year == 2001) & (dd.d_qoy == 1)]
+    d23 = dd[(dd.d_year == 2001) & dd.d_qoy.isin([1, 2, 3])]
+    j_ss = _merge(ss, d1[["d_date_sk"]], "ss_sold_date_sk", "d_date_sk")
+    j_sr = _merge(sr, d23[["d_date_sk"]].rename(columns={"d_date_sk": "d2"}),
+                  "sr_returned_date_sk", "d2")
+    j_cs = _merge(cs, d23[["d_date_sk"]].rename(columns={"d_date_sk": "d3"}),
+                  "cs_sold_date_sk", "d3")
+    j1 = j_ss.dropna(subset=["ss_customer_sk", "ss_item_sk", "ss_ticket_number"]).merge(
+        j_sr.dropna(subset=["sr_customer_sk", "sr_item_sk", "sr_ticket_number"]),
+        left_on=["ss_customer_sk", "ss_item_sk", "ss_ticket_number"],
+        right_on=["sr_customer_sk", "sr_item_sk", "sr_ticket_number"])
+    j2 = j1.merge(j_cs.dropna(subset=["cs_bill_customer_sk", "cs_item_sk"]),
+                  left_on=["sr_customer_sk", "sr_item_sk"],
+                  right_on=["cs_bill_customer_sk", "cs_item_sk"])
+    st = _read(root, sf, "store", ["s_store_sk", "s_state"])
+    it = _read(root, sf, "item", ["i_item_sk", "i_item_id", "i_item_desc"])
+    j3 = _merge(j2, st, "ss_store_sk", "s_store_sk")
+    j4 = _merge(j3, it, "ss_item_sk", "i_item_sk")
+
+    def stats(g, c, tag):
+        n = g[c].count()
+        s_ = g[c].sum(min_count=1)
+        ssq = (g[c].astype(float) ** 2).sum(min_count=1)
+        return {f"{tag}_n": n, f"{tag}_s": s_, f"{tag}_ss": ssq}
+
+    rows = []
+    for key, g in j4.groupby(["i_item_id", "i_item_desc", "s_state"], dropna=False):
+        r = dict(zip(["i_item_id", "i_item_desc", "s_state"], key))
+        r.update(stats(g, "ss_quantity", "q"))
+        r.update(stats(g, "sr_return_quantity", "r"))
+        r.update(stats(g, "cs_quantity", "c"))
+        rows.append(r)
+    import pandas as pd
+
+    cols17 = (["i_item_id", "i_item_desc", "s_state"]
+              + [f"{t}_{x}" for t in "qrc" for x in ("n", "s", "ss")])
+    a = pd.DataFrame(rows, columns=cols17) if rows else pd.DataFrame(columns=cols17)
+
+    def var(tag):
+        n = a[f"{tag}_n"].astype(float)
+        m = a[f"{tag}_s"] / n
+        return (a[f"{tag}_ss"] - n * m * m) / (n - 1.0)
+
+    out = pd.DataFrame({
+        "i_item_id": a.i_item_id, "i_item_desc": a.i_item_desc, "s_state": a.s_state,
+        "store_sales_quantitycount": a.q_n,
+        "store_sales_quantityave": a.q_s / a.q_n.astype(float),
+        "store_sales_quantityvar": var("q"),
+        "store_returns_quantitycount": a.r_n,
+        "store_returns_quantityave": a.r_s / a.r_n.astype(float),
+        "store_returns_quantityvar": var("r"),
+        "catalog_sales_quantitycount": a.c_n,
+        "catalog_sales_quantityave": a.c_s / a.c_n.astype(float),
+        "catalog_sales_quantityvar": var("c")})
+    out = out.sort_values(["i_item_id", "i_item_desc", "s_state"],
+                          na_position="first").head(100)
+    return out.reset_index(drop=True)
+
+
+def q18(root, sf):
+    import pandas as pd
+
+    cs = _read(root, sf, "catalog_sales", ["cs_sold_date_sk", "cs_item_sk", "cs_bill_cdemo_sk",
+                                           "cs_bill_customer_sk", "cs_quantity", "cs_list_price",
+                                           "cs_coupon_amt", "cs_sales_price", "cs_net_profit"])
+    cd1 = _read(root, sf, "customer_demographics",
+                ["cd_demo_sk", "cd_gender", "cd_education_status", "cd_dep_count"])
+    cd1 = cd1[(cd1.cd_gender == "F") & (cd1.cd_education_status == "Unknown")]
+    cd2 = _read(root, sf, "customer_demographics", ["cd_demo_sk"]).rename(
+        columns={"cd_demo_sk": "cd2_sk"})
+    cust = _read(root, sf, "customer", ["c_customer_sk", "c_current_cdemo_sk",
+                                        "c_current_addr_sk", "c_birth_month", "c_birth_year"])
+    cust = cust[cust.c_birth_month.isin([1, 6, 8, 9, 12, 2])]
+    ca = _read(root, sf, "customer_address",
+               ["ca_address_sk", "ca_country", "ca_state", "ca_county"])
+    ca = ca[ca.ca_state.isin(["MS", "IN", "ND", "OK", "NM", "VA"])]
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year"])
+    dd = dd[dd.d_year == 1998]
+    it = _read(root, sf, "item", ["i_item_sk", "i_item_id"])
+    j = _merge(cs, dd, "cs_sold_date_sk", "d_date_sk")
+    j = _merge(j, it, "cs_item_sk", "i_item_sk")
+    j = _merge(j, cd1, "cs_bill_cdemo_sk", "cd_demo_sk")
+    j = _merge(j, cust, "cs_bill_customer_sk", "c_customer_sk")
+    j = _merge(j, cd2, "c_current_cdemo_sk", "cd2_sk")
+    j = _merge(j, ca, "c_current_addr_sk", "ca_address_sk")
+    carry = {"agg1": "cs_quantity", "agg2": "cs_list_price", "agg3": "cs_coupon_amt",
+             "agg4": "cs_sales_price", "agg5": "cs_net_profit", "agg6": "c_birth_year",
+             "agg7": "cd_dep_count"}
+    keys = ["i_item_id", "ca_country", "ca_state", "ca_county"]
+    frames = []
+    for depth in (4, 3, 2, 1, 0):
+        t = j.copy()
+        for i, k in enumerate(keys):
+            if i >= depth:
+                t[k] = None
+        g = t.groupby(keys, dropna=False).agg(
+            **{n: (c, "mean") for n, c in carry.items()}).reset_index()
+        frames.append(g)
+    out = pd.concat(frames, ignore_index=True)
+    out = out.sort_values(["ca_country", "ca_state", "ca_county", "i_item_id"],
+                          ascending=[False, False, False, True],
+                          na_position="first").head(100)
+    return out[keys + list(carry.keys())].reset_index(drop=True)
+
+
+def q44(root, sf):
+    import pandas as pd
+
+    ss = _read(root, sf, "store_sales", ["ss_item_sk", "ss_store_sk", "ss_addr_sk",
+                                         "ss_net_profit"])
+    f4 = ss[ss.ss_store_sk == 4]
+    base = f4.groupby("ss_item_sk", dropna=False).ss_net_profit.mean() \
+             .reset_index(name="rank_col")
+    nullf = f4[f4.ss_addr_sk.isna()]
+    th = nullf.ss_net_profit.mean()
+    if pd.isna(th):
+        hav = base.iloc[0:0]
+    else:
+        hav = base[base.rank_col > 0.9 * th]
+    asc = hav.sort_values("rank_col").reset_index(drop=True)
+    asc["rnk"] = asc.rank_col.rank(method="min").astype(int)
+    desc = hav.sort_values("rank_col", ascending=False).reset_index(drop=True)
+    desc["rnk"] = desc.rank_col.rank(method="min", ascending=False).astype(int)
+    a = asc[asc.rnk < 11][["ss_item_sk", "rnk"]].rename(columns={"ss_item_sk": "asc_item"})
+    d = desc[desc.rnk < 11][["ss_item_sk", "rnk"]].rename(columns={"ss_item_sk": "desc_item"})
+    m = a.merge(d, on="rnk")
+    it = _read(root, sf, "item", ["i_item_sk", "i_product_name"])
+    m = _merge(m, it.rename(columns={"i_item_sk": "i1", "i_product_name": "best_performing"}),
+               "asc_item", "i1")
+    m = _merge(m, it.rename(columns={"i_item_sk": "i2", "i_product_name": "worst_performing"}),
+               "desc_item", "i2")
+    out = m[["rnk", "best_performing", "worst_performing"]].sort_values("rnk").head(100)
+    return out.reset_index(drop=True)
+
+
+def q49(root, sf):
+    import pandas as pd
+
+    def chan(fact, pre, rets, rpre, tag):
+        amt = f"{rpre}_return_{'amount' if rpre == 'cr' else 'amt'}"
+        okey = f"{pre}_{'ticket_number' if pre == 'ss' else 'order_number'}"
+        rkey = f"{rpre}_{'ticket_number' if rpre == 'sr' else 'order_number'}"
+        fs = _read(root, sf, fact, [okey, f"{pre}_item_sk", f"{pre}_quantity",
+                                    f"{pre}_net_paid", f"{pre}_net_profit",
+                                    f"{pre}_sold_date_sk"])
+        rt = _read(root, sf, rets, [rkey, f"{rpre}_item_sk",
+                                    f"{rpre}_return_quantity", amt])
+        j = fs.merge(rt.dropna(subset=[rkey, f"{rpre}_item_sk"]),
+                     left_on=[okey, f"{pre}_item_sk"],
+                     right_on=[rkey, f"{rpre}_item_sk"], how="left")
+        dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year", "d_moy"])
+        dd = dd[(dd.d_year == 2001) & (dd.d_moy == 12)]
+        j = _merge(j, dd, f"{pre}_sold_date_sk", "d_date_sk")
+        f = j[((j[amt] > 10000.0) & (j[f"{pre}_net_profit"] > 1.0)
+               & (j[f"{pre}_net_paid"] > 0.0) & (j[f"{pre}_quantity"] > 0)).fillna(False)]
+        g = f.assign(rq=f[f"{rpre}_return_quantity"].fillna(0).astype(float),
+                     sq=f[f"{pre}_quantity"].astype(float),
+                     ra=f[amt].fillna(0.0), npaid=f[f"{pre}_net_paid"]) \
+            .groupby(f"{pre}_item_sk", dropna=False).agg(
+            srq=("rq", "sum"), ssq=("sq", "sum"), sra=("ra", "sum"),
+            snp=("npaid", "sum")).reset_index()
+        g["return_ratio"] = g.srq / g.ssq
+        g["currency_ratio"] = g.sra / g.snp
+        g["return_rank"] = g.return_ratio.rank(method="min").astype(int)
+        g["currency_rank"] = g.currency_ratio.rank(method="min").astype(int)
+        g = g[(g.return_rank <= 10) | (g.currency_rank <= 10)]
+        g["channel"] = tag
+        g = g.rename(columns={f"{pre}_item_sk": "item"})
+        return g[["channel", "item", "return_ratio", "return_rank", "currency_rank"]]
+
+    u = pd.concat([chan("web_sales", "ws", "web_returns", "wr", "web"),
+                   chan("catalog_sales", "cs", "catalog_returns", "cr", "catalog"),
+                   chan("store_sales", "ss", "store_returns", "sr", "store")],
+                  ignore_index=True).drop_duplicates()
+    u = u.sort_values(["channel", "return_rank", "currency_rank", "item"],
+                      na_position="first").head(100)
+    return u.reset_index(drop=True)
+
+
+def q58(root, sf):
+    dd_all = _read(root, sf, "date_dim", ["d_date_sk", "d_date", "d_week_seq"])
+    di = _date_i(dd_all)
+    wk = dd_all[di == _days(2000, 1, 3)].d_week_seq.iloc[0]
+    dd = dd_all[dd_all.d_week_seq == wk]
+
+    def items(fact, pre, rev):
+        fs = _read(root, sf, fact, [f"{pre}_item_sk", f"{pre}_sold_date_sk",
+                                    f"{pre}_ext_sales_price"])
+        it = _read(root, sf, "item", ["i_item_sk", "i_item_id"])
+        j = _merge(fs, dd, f"{pre}_sold_date_sk", "d_date_sk")
+        j = _merge(j, it, f"{pre}_item_sk", "i_item_sk")
+        return j.groupby("i_item_id", dropna=False)[f"{pre}_ext_sales_price"] \
+                .sum(min_count=1).reset_index(name=rev)
+
+    ssi = items("store_sales", "ss", "ss_item_rev")
+    csi = items("catalog_sales", "cs", "cs_item_rev")
+    wsi = items("web_sales", "ws", "ws_item_rev")
+    j = ssi.merge(csi, on="i_item_id").merge(wsi, on="i_item_id")
+    c = ((j.ss_item_rev >= 0.9 * j.cs_item_rev) & (j.ss_item_rev <= 1.1 * j.cs_item_rev)
+         & (j.ss_item_rev >= 0.9 * j.ws_item_rev) & (j.ss_item_rev <= 1.1 * j.ws_item_rev)
+         & (j.cs_item_rev >= 0.9 * j.ss_item_rev) & (j.cs_item_rev <= 1.1 * j.ss_item_rev)
+         & (j.cs_item_rev >= 0.9 * j.ws_item_rev) & (j.cs_item_rev <= 1.1 * j.ws_item_rev)
+         & (j.ws_item_rev >= 0.9 * j.ss_item_rev) & (j.ws_item_rev <= 1.1 * j.ss_item_rev)
+         & (j.ws_item_rev >= 0.9 * j.cs_item_rev) & (j.ws_item_rev <= 1.1 * j.cs_item_rev))
+    f = j[c.fillna(False)].copy()
+    tot = f.ss_item_rev + f.cs_item_rev + f.ws_item_rev
+    f["ss_dev"] = f.ss_item_rev / tot / 3.0 * 100.0
+    f["cs_dev"] = f.cs_item_rev / tot / 3.0 * 100.0
+    f["ws_dev"] = f.ws_item_rev / tot / 3.0 * 100.0
+    f["average"] = tot / 3.0
+    out = f[["i_item_id", "ss_item_rev", "ss_dev", "cs_item_rev", "cs_dev",
+             "ws_item_rev", "ws_dev", "average"]].rename(columns={"i_item_id": "item_id"})
+    out = out.sort_values(["item_id", "ss_item_rev"], na_position="first").head(100)
+    return out.reset_index(drop=True)
+
+
+ORACLES.update({"q2": q2, "q9": q9, "q10": q10, "q17": q17, "q18": q18,
+                "q44": q44, "q49": q49, "q58": q58})
