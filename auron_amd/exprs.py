@@ -493,6 +493,40 @@ class Length(Expr):
 
 
 @dataclass(eq=False)
+class Sqrt(Expr):
+    child: Expr
+
+    def eval(self, batch: RecordBatch) -> Column:
+        c = _cast_col(self.child.eval(batch), dtypes.float64)
+        return Column(dtypes.float64, torch.sqrt(c.data), c.validity)
+
+
+@dataclass(eq=False)
+class Abs(Expr):
+    child: Expr
+
+    def eval(self, batch: RecordBatch) -> Column:
+        c = self.child.eval(batch)
+        return Column(c.dtype, torch.abs(c.data), c.validity)
+
+
+@dataclass(eq=False)
+class Upper(Expr):
+    child: Expr
+
+    def eval(self, batch: RecordBatch) -> Column:
+        return strings.upper(self.child.eval(batch))
+
+
+@dataclass(eq=False)
+class Lower(Expr):
+    child: Expr
+
+    def eval(self, batch: RecordBatch) -> Column:
+        return strings.lower(self.child.eval(batch))
+
+
+@dataclass(eq=False)
 class ConcatStr(Expr):
     args: List[Expr]
 

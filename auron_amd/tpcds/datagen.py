@@ -151,8 +151,15 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
         put("i_product_name", [f"product{int(s)}" for s in sks])
         put("i_item_desc", [f"the quite famous item number {int(s)} description" for s in sks])
         _colors = ["red", "blue", "green", "yellow", "black", "white", "purple",
-                   "orange", "pink", "brown", "gray", "cyan"]
-        put("i_color", [_colors[int(v)] for v in rng.integers(0, 12, n)])
+                   "orange", "pink", "brown", "gray", "cyan", "pale", "powder",
+                   "khaki", "midnight", "snow", "forest", "ghost", "floral"]
+        put("i_color", [_colors[int(v)] for v in rng.integers(0, len(_colors), n)])
+        _units = ["Ounce", "Oz", "Bunch", "Ton", "N/A", "Dozen", "Box", "Pound",
+                  "Pallet", "Gross", "Cup", "Dram", "Each", "Tbl", "Lb", "Bundle"]
+        put("i_units", [_units[int(v)] for v in rng.integers(0, len(_units), n)])
+        put("i_size", [["small", "medium", "large", "extra large", "petite", "N/A"][int(v)]
+                       for v in rng.integers(0, 6, n)])
+        put("i_manufact", [f"manufact{int(m)}" for m in mfg])
     elif name == "customer":
         put("c_customer_sk", sks)
         put("c_customer_id", _id_str("AAAAAAAA", sks))
@@ -164,7 +171,8 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
         put("c_current_cdemo_sk", cd, cdv)
         hd, hdv = fk(n_hdemo, 0.02)
         put("c_current_hdemo_sk", hd, hdv)
-        put("c_birth_country", [_COUNTRIES[int(v)] for v in rng.integers(0, len(_COUNTRIES), n)])
+        # dsdgen stores birth country uppercased (q24 matches upper(ca_country))
+        put("c_birth_country", [_COUNTRIES[int(v)].upper() for v in rng.integers(0, len(_COUNTRIES), n)])
         put("c_birth_year", rng.integers(1924, 1993, n).astype(np.int32))
         put("c_birth_month", rng.integers(1, 13, n).astype(np.int32))
         put("c_email_address", [f"c{int(s)}@example.com" for s in sks])
@@ -220,6 +228,7 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
         put("s_number_employees", rng.integers(200, 301, n).astype(np.int32))
         put("s_gmt_offset", np.full(n, -5.0))
         put("s_company_id", np.ones(n, dtype=np.int32))
+        put("s_market_id", ((sks - 1) % 10 + 1).astype(np.int32))
         put("s_street_name", [f"{_LAST[int(s) % len(_LAST)]} Blvd" for s in sks])
     elif name == "warehouse":
         put("w_warehouse_sk", sks)
@@ -473,7 +482,7 @@ def _nparts_for(table: str, sf: float) -> int:
     return 1
 
 
-DATAGEN_VERSION = 5
+DATAGEN_VERSION = 6
 
 
 def dataset_root(root: str, sf: float) -> str:

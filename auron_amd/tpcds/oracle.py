@@ -2209,17 +2209,25 @@ def q17(root, sf):
         m = a[f"{tag}_s"] / n
         return (a[f"{tag}_ss"] - n * m * m) / (n - 1.0)
 
+    import numpy as np
+
+    def sd(tag):
+        return np.sqrt(var(tag))
+
     out = pd.DataFrame({
         "i_item_id": a.i_item_id, "i_item_desc": a.i_item_desc, "s_state": a.s_state,
         "store_sales_quantitycount": a.q_n,
         "store_sales_quantityave": a.q_s / a.q_n.astype(float),
-        "store_sales_quantityvar": var("q"),
+        "store_sales_quantitystdev": sd("q"),
+        "store_sales_quantitycov": sd("q") / (a.q_s / a.q_n.astype(float)),
         "store_returns_quantitycount": a.r_n,
         "store_returns_quantityave": a.r_s / a.r_n.astype(float),
-        "store_returns_quantityvar": var("r"),
+        "store_returns_quantitystdev": sd("r"),
+        "store_returns_quantitycov": sd("r") / (a.r_s / a.r_n.astype(float)),
         "catalog_sales_quantitycount": a.c_n,
         "catalog_sales_quantityave": a.c_s / a.c_n.astype(float),
-        "catalog_sales_quantityvar": var("c")})
+        "catalog_sales_quantitystdev": sd("c") / (a.c_s / a.c_n.astype(float)),
+        "catalog_sales_quantitycov": sd("c") / (a.c_s / a.c_n.astype(float))})
     out = out.sort_values(["i_item_id", "i_item_desc", "s_state"],
                           na_position="first").head(100)
     return out.reset_index(drop=True)
@@ -2384,3 +2392,156 @@ def q58(root, sf):
 
 ORACLES.update({"q2": q2, "q9": q9, "q10": q10, "q17": q17, "q18": q18,
                 "q44": q44, "q49": q49, "q58": q58})
+
+
+# ------------------------------- batch 10 oracles
+def q8(root, sf):
+    from .queries import _Q8_ZIPS
+
+    ca = _read(root, sf, "customer_address", ["ca_address_sk", "ca_zip"])
+    cust = _read(root, sf, "customer", ["c_current_addr_sk", "c_preferred_cust_flag"])
+    cust = cust[cust.c_preferred_cust_flag == "Y"]
+    j = _merge(ca, cust, "ca_address_sk", "c_current_addr_sk")
+    j["zip5"] = j.ca_zip.str[:5]
+    a1 = j.groupby("zip5", dropna=False).size().reset_index(name="cnt")
+    a1 = set(a1[a1.cnt > 10].zip5)
+    lit_set = set(ca.ca_zip.str[:5]) & set(_Q8_ZIPS)
+    v1 = a1 & lit_set
+    zip2 = {z[:2] for z in v1 if z is not None}
+    ss = _read(root, sf, "store_sales", ["ss_store_sk", "ss_sold_date_sk", "ss_net_profit"])
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_qoy", "d_year"])
+    dd = dd[(dd.d_qoy == 2) & (dd.d_year == 1998)]
+    st = _read(root, sf, "store", ["s_store_sk", "s_store_name", "s_zip"])
+    j2 = _merge(ss, dd, "ss_sold_date_sk", "d_date_sk")
+    j2 = _merge(j2, st, "ss_store_sk", "s_store_sk")
+    j2 = j2[j2.s_zip.str[:2].isin(zip2)]
+    g = j2.groupby("s_store_name", dropna=False).ss_net_profit.sum(min_count=1) \
+          .reset_index(name="profit")
+    g = g.sort_values("s_store_name", na_position="first").head(100)
+    return g.reset_index(drop=True)
+
+
+def q24(root, sf):
+    ss = _read(root, sf, "store_sales", ["ss_ticket_number", "ss_item_sk", "ss_customer_sk",
+                                         "ss_store_sk", "ss_net_paid"])
+    sr = _read(root, sf, "store_returns", ["sr_ticket_number", "sr_item_sk"])
+    st = _read(root, sf, "store", ["s_store_sk", "s_store_name", "s_state", "s_zip",
+                                   "s_market_id"])
+    st = st[st.s_market_id == 8]
+    it = _read(root, sf, "item", ["i_item_sk", "i_color", "i_current_price",
+                                  "i_manager_id", "i_units", "i_size"])
+    cust = _read(root, sf, "customer", ["c_customer_sk", "c_last_name", "c_first_name",
+                                        "c_birth_country"])
+    ca = _read(root, sf, "customer_address", ["ca_address_sk", "ca_state", "ca_country",
+                                              "ca_zip"])
+    j = ss.dropna(subset=["ss_ticket_number", "ss_item_sk"]).merge(
+        sr.dropna(subset=["sr_ticket_number", "sr_item_sk"]),
+        left_on=["ss_ticket_number", "ss_item_sk"],
+        right_on=["sr_ticket_number", "sr_item_sk"])
+    j = _merge(j, st, "ss_store_sk", "s_store_sk")
+    j = _merge(j, it, "ss_item_sk", "i_item_sk")
+    j = _merge(j, cust, "ss_customer_sk", "c_customer_sk")
+    ca = ca.assign(ca_ucountry=ca.ca_country.str.upper())
+    j = j.dropna(subset=["c_birth_country", "s_zip"]).merge(
+        ca.dropna(subset=["ca_ucountry", "ca_zip"]),
+        left_on=["c_birth_country", "s_zip"], right_on=["ca_ucountry", "ca_zip"])
+    keys = ["c_last_name", "c_first_name", "s_store_name", "ca_state", "s_state",
+            "i_color", "i_current_price", "i_manager_id", "i_units", "i_size"]
+    ssales = j.groupby(keys, dropna=False).ss_net_paid.sum(min_count=1) \
+              .reset_index(name="netpaid")
+    th = ssales.netpaid.mean()
+    f = ssales[ssales.i_color == "pale"]
+    g = f.groupby(["c_last_name", "c_first_name", "s_store_name"], dropna=False) \
+         .netpaid.sum(min_count=1).reset_index(name="paid")
+    g = g[g.paid > 0.05 * (th if th == th else 0.0)]
+    g = g.sort_values(["c_last_name", "c_first_name", "s_store_name"],
+                      na_position="first")
+    return g.reset_index(drop=True)
+
+
+def q39(root, sf):
+    import numpy as np
+
+    inv = _read(root, sf, "inventory")
+    wh = _read(root, sf, "warehouse", ["w_warehouse_sk", "w_warehouse_name"])
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year", "d_moy"])
+    dd = dd[dd.d_year == 2001]
+    j = _merge(inv, wh, "inv_warehouse_sk", "w_warehouse_sk")
+    j = _merge(j, dd, "inv_date_sk", "d_date_sk")
+    g = j.groupby(["w_warehouse_name", "w_warehouse_sk", "inv_item_sk", "d_moy"],
+                  dropna=False).agg(
+        n=("inv_quantity_on_hand", "count"),
+        s=("inv_quantity_on_hand", lambda x: x.astype(float).sum(min_count=1)),
+        ssq=("inv_quantity_on_hand", lambda x: (x.astype(float) ** 2).sum(min_count=1))) \
+        .reset_index()
+    n = g.n.astype(float)
+    mean = g.s / n
+    var = (g.ssq - n * mean * mean) / (n - 1.0)
+    cov = np.sqrt(var) / mean
+    g = g.assign(mean=mean, cov=cov)
+    f = g[(np.where(mean == 0, 0.0, cov) > 1.0)]
+    i1 = f[f.d_moy == 1]
+    i2 = f[f.d_moy == 2]
+    m = i1.merge(i2, on=["w_warehouse_sk", "inv_item_sk"], suffixes=("1", "2"))
+    out = m[["w_warehouse_sk", "inv_item_sk", "d_moy1", "mean1", "cov1",
+             "w_warehouse_sk", "inv_item_sk", "d_moy2", "mean2", "cov2"]]
+    out.columns = ["w_warehouse_sk1", "i_item_sk1", "d_moy1", "mean1", "cov1",
+                   "w_warehouse_sk2", "i_item_sk2", "d_moy2", "mean2", "cov2"]
+    out = out.sort_values(["w_warehouse_sk1", "i_item_sk1", "d_moy1", "mean1",
+                           "cov1", "d_moy2", "mean2", "cov2"], na_position="first")
+    return out.reset_index(drop=True)
+
+
+def q41(root, sf):
+    it = _read(root, sf, "item")
+    i1 = it[it.i_manufact_id.between(738, 778)]
+
+    def block(cat_, colors, units, sizes):
+        return ((it.i_category == cat_) & it.i_color.isin(colors)
+                & it.i_units.isin(units) & it.i_size.isin(sizes))
+
+    pred = (block("Women", ["powder", "khaki"], ["Ounce", "Oz"], ["medium", "extra large"])
+            | block("Women", ["brown", "honeydew"], ["Bunch", "Ton"], ["N/A", "small"])
+            | block("Men", ["floral", "deep"], ["N/A", "Dozen"], ["petite", "large"])
+            | block("Men", ["light", "cornflower"], ["Box", "Pound"], ["medium", "extra large"])
+            | block("Women", ["midnight", "snow"], ["Pallet", "Gross"], ["medium", "extra large"])
+            | block("Women", ["cyan", "papaya"], ["Cup", "Dram"], ["N/A", "small"])
+            | block("Men", ["orange", "frosted"], ["Each", "Tbl"], ["petite", "large"])
+            | block("Men", ["forest", "ghost"], ["Lb", "Bundle"], ["medium", "extra large"]))
+    mset = set(it[pred.fillna(False)].i_manufact.dropna())
+    f = i1[i1.i_manufact.isin(mset)]
+    out = f[["i_product_name"]].drop_duplicates().sort_values("i_product_name").head(100)
+    return out.reset_index(drop=True)
+
+
+def q95(root, sf):
+    import pandas as pd
+
+    lo = _days(1999, 2, 1)
+    ws = _read(root, sf, "web_sales", ["ws_ship_date_sk", "ws_ship_addr_sk",
+                                       "ws_order_number", "ws_warehouse_sk",
+                                       "ws_ext_ship_cost", "ws_net_profit",
+                                       "ws_web_site_sk"])
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_date"])
+    di = _date_i(dd)
+    dd = dd[(di >= lo) & (di <= lo + 60)]
+    ca = _read(root, sf, "customer_address", ["ca_address_sk", "ca_state"])
+    ca = ca[ca.ca_state == "IL"]
+    site = _read(root, sf, "web_site", ["web_site_sk", "web_company_name"])
+    site = site[site.web_company_name == "pri"]
+    j = _merge(ws, dd, "ws_ship_date_sk", "d_date_sk")
+    j = _merge(j, ca, "ws_ship_addr_sk", "ca_address_sk")
+    j = _merge(j, site, "ws_web_site_sk", "web_site_sk")
+    allf = _read(root, sf, "web_sales", ["ws_order_number", "ws_warehouse_sk"])
+    per = allf.dropna().drop_duplicates().groupby("ws_order_number").size()
+    multi = set(per[per > 1].index)
+    wr = _read(root, sf, "web_returns", ["wr_order_number"])
+    wr_multi = set(wr.wr_order_number.dropna()) & multi
+    j = j[j.ws_order_number.isin(multi) & j.ws_order_number.isin(wr_multi)]
+    return pd.DataFrame({
+        "order_count": [j.ws_order_number.nunique()],
+        "total_shipping_cost": [j.ws_ext_ship_cost.sum(min_count=1)],
+        "total_net_profit": [j.ws_net_profit.sum(min_count=1)]})
+
+
+ORACLES.update({"q8": q8, "q24": q24, "q39": q39, "q41": q41, "q95": q95})

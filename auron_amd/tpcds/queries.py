@@ -2587,24 +2587,31 @@ def q17(cat, s):
         + q_stats("cs_quantity", "c")
     a = agg2(j4, ["i_item_id", "i_item_desc", "s_state"], aggs)
 
+    from ..exprs import Sqrt
+
     def stdev(tag):
         n = col(f"{tag}_n").cast(dtypes.float64)
         m = col(f"{tag}_s") / n
-        var = (col(f"{tag}_ss") - n * m * m) / (n - lit(1.0))
-        return var  # sqrt applied host-side is unavailable; report variance-based cov
-    # expose count/avg/stdev^2 (variance); documented deviation: no sqrt expr yet
+        return Sqrt((col(f"{tag}_ss") - n * m * m) / (n - lit(1.0)))
+
+    def ave(tag):
+        return col(f"{tag}_s") / col(f"{tag}_n").cast(dtypes.float64)
+
     proj = P.Project(a, [
         _a(col("i_item_id"), "i_item_id"), _a(col("i_item_desc"), "i_item_desc"),
         _a(col("s_state"), "s_state"),
         _a(col("q_n"), "store_sales_quantitycount"),
-        _a(col("q_s") / col("q_n").cast(dtypes.float64), "store_sales_quantityave"),
-        _a(stdev("q"), "store_sales_quantityvar"),
+        _a(ave("q"), "store_sales_quantityave"),
+        _a(stdev("q"), "store_sales_quantitystdev"),
+        _a(stdev("q") / ave("q"), "store_sales_quantitycov"),
         _a(col("r_n"), "store_returns_quantitycount"),
-        _a(col("r_s") / col("r_n").cast(dtypes.float64), "store_returns_quantityave"),
-        _a(stdev("r"), "store_returns_quantityvar"),
+        _a(ave("r"), "store_returns_quantityave"),
+        _a(stdev("r"), "store_returns_quantitystdev"),
+        _a(stdev("r") / ave("r"), "store_returns_quantitycov"),
         _a(col("c_n"), "catalog_sales_quantitycount"),
-        _a(col("c_s") / col("c_n").cast(dtypes.float64), "catalog_sales_quantityave"),
-        _a(stdev("c"), "catalog_sales_quantityvar")])
+        _a(ave("c"), "catalog_sales_quantityave"),
+        _a(stdev("c") / ave("c"), "catalog_sales_quantitystdev"),
+        _a(stdev("c") / ave("c"), "catalog_sales_quantitycov")])
     return topk(proj, [(col("i_item_id"), True), (col("i_item_desc"), True),
                        (col("s_state"), True)], 100)
 
@@ -2783,3 +2790,221 @@ def q58(cat, s):
 
 QUERIES.update({"q2": q2, "q9": q9, "q10": q10, "q17": q17, "q18": q18,
                 "q44": q44, "q49": q49, "q58": q58})
+
+
+# ------------------------------- batch 10
+_Q8_ZIPS = ['24128', '76232', '65084', '87816', '83926', '77556', '20548',
+            '26231', '43848', '15126', '91137', '61265', '98294', '25782',
+            '17920', '18426', '98235', '40081', '84093', '28577', '55565',
+            '17183', '54601', '67897', '22752', '86284', '18376', '38607',
+            '45200', '21756', '29741', '96765', '23932', '89360', '29839',
+            '25989', '28898', '91068', '72550', '10390', '18845', '47770',
+            '82636', '41367', '76638', '86198', '81312', '37126', '39192',
+            '88424', '72175', '81426', '53672', '10445', '42666', '66864',
+            '66708', '41248', '48583', '82276', '18842', '78890', '49448',
+            '14089', '38122', '34425', '79077', '19849', '43285', '39861',
+            '66162', '77610', '13695', '99543', '83444', '83041', '12305',
+            '57665', '68341', '25003', '57834', '62878', '49130', '81096',
+            '18840', '27700', '23470', '50412', '21195', '16021', '76107',
+            '71954', '68309', '18119', '98359', '64544', '10336', '86379',
+            '27068', '39736', '98569', '28915', '24206', '56529', '57647']
+
+
+def q8(cat, s):
+    from ..exprs import Substr
+
+    # A1: 5-digit zip prefixes of preferred customers, count > 10
+    ca = cat.scan("customer_address", ["ca_address_sk", "ca_zip"])
+    cust = P.Filter(cat.scan("customer", ["c_current_addr_sk", "c_preferred_cust_flag"]),
+                    col("c_preferred_cust_flag") == lit("Y"))
+    j = shj(ca, cust, ["ca_address_sk"], ["c_current_addr_sk"])
+    a1 = P.Filter(agg2(j, ["zip5"], [AggFunc("count_star", None, name="cnt")],
+                       key_exprs=[Substr(col("ca_zip"), 1, 5)]),
+                  col("cnt") > 10)
+    # INTERSECT with the literal zip list
+    lit_zips = P.Filter(P.Project(cat.scan("customer_address", ["ca_zip"]),
+                                  [_a(Substr(col("ca_zip"), 1, 5), "zip5")]),
+                        col("zip5").isin(_Q8_ZIPS))
+    lit_zips_d = agg2(lit_zips, ["zip5"], [])
+    v1 = P.HashJoin(a1, P.Exchange(P.Project(lit_zips_d, [_a(col("zip5"), "lz")]),
+                                   "hash", [col("lz")]),
+                    [col("zip5")], [col("lz")], how="semi", build_side="right")
+    zip2 = P.Broadcast(P.HashAgg(P.Exchange(
+        P.Project(v1, [_a(Substr(col("zip5"), 1, 2), "zip2")]), "single"),
+        [_a(col("zip2"), "zip2")], [], mode="complete"))
+
+    ss = cat.scan("store_sales", ["ss_store_sk", "ss_sold_date_sk", "ss_net_profit"])
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_qoy", "d_year"]),
+                  (col("d_qoy") == 2) & (col("d_year") == 1998))
+    st = cat.scan("store", ["s_store_sk", "s_store_name", "s_zip"])
+    j2 = bhj(ss, dd, ["ss_sold_date_sk"], ["d_date_sk"])
+    j2 = bhj(j2, st, ["ss_store_sk"], ["s_store_sk"])
+    j2 = P.HashJoin(j2, zip2, [Substr(col("s_zip"), 1, 2)], [col("zip2")],
+                    how="semi", build_side="right")
+    a = agg2(j2, ["s_store_name"], [AggFunc("sum", col("ss_net_profit"), name="profit")])
+    return topk(a, [(col("s_store_name"), True)], 100)
+
+
+def _q24(cat, s, color):
+    from ..exprs import Upper
+
+    ss = cat.scan("store_sales", ["ss_ticket_number", "ss_item_sk", "ss_customer_sk",
+                                  "ss_store_sk", "ss_net_paid"])
+    sr = cat.scan("store_returns", ["sr_ticket_number", "sr_item_sk"])
+    st = P.Filter(cat.scan("store", ["s_store_sk", "s_store_name", "s_state", "s_zip",
+                                     "s_market_id"]),
+                  col("s_market_id") == 8)
+    it = cat.scan("item", ["i_item_sk", "i_color", "i_current_price", "i_manager_id",
+                           "i_units", "i_size"])
+    cust = cat.scan("customer", ["c_customer_sk", "c_last_name", "c_first_name",
+                                 "c_birth_country"])
+    ca = cat.scan("customer_address", ["ca_address_sk", "ca_state", "ca_country", "ca_zip"])
+    j = shj(ss, sr, ["ss_ticket_number", "ss_item_sk"], ["sr_ticket_number", "sr_item_sk"])
+    j = bhj(j, st, ["ss_store_sk"], ["s_store_sk"])
+    j = bhj(j, it, ["ss_item_sk"], ["i_item_sk"])
+    j = bhj(j, cust, ["ss_customer_sk"], ["c_customer_sk"])
+    # c_birth_country = upper(ca_country) AND s_zip = ca_zip
+    j = P.HashJoin(j, P.Broadcast(P.Project(ca, [_a(Upper(col("ca_country")), "ca_ucountry"),
+                                                 _a(col("ca_zip"), "ca_zip2"),
+                                                 _a(col("ca_state"), "ca_state")])),
+                   [col("c_birth_country"), col("s_zip")],
+                   [col("ca_ucountry"), col("ca_zip2")],
+                   how="inner", build_side="right")
+    keys = ["c_last_name", "c_first_name", "s_store_name", "ca_state", "s_state",
+            "i_color", "i_current_price", "i_manager_id", "i_units", "i_size"]
+    ssales = agg2(j, keys, [AggFunc("sum", col("ss_net_paid"), name="netpaid")])
+    ssales_b = s.execute(ssales)
+    thresh = scalar(s, _global_agg(P.MemoryScan(ssales_b),
+                                   [AggFunc("avg", col("netpaid"), name="a")]))
+    f = P.Filter(P.MemoryScan(ssales_b), col("i_color") == lit(color))
+    a = agg2(f, ["c_last_name", "c_first_name", "s_store_name"],
+             [AggFunc("sum", col("netpaid"), name="paid")])
+    h = P.Filter(a, col("paid") > lit(0.05 * (thresh or 0.0)))
+    return topk(h, [(col("c_last_name"), True), (col("c_first_name"), True),
+                    (col("s_store_name"), True)], 100000)
+
+
+def q24(cat, s):
+    return _q24(cat, s, "pale")
+
+
+def _q39(cat, s, cov_filter):
+    inv = cat.scan("inventory", ["inv_item_sk", "inv_warehouse_sk", "inv_date_sk",
+                                 "inv_quantity_on_hand"])
+    it = cat.scan("item", ["i_item_sk"])
+    wh = cat.scan("warehouse", ["w_warehouse_sk", "w_warehouse_name"])
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_year", "d_moy"]),
+                  col("d_year") == 2001)
+    j = bhj(inv, it, ["inv_item_sk"], ["i_item_sk"])
+    j = bhj(j, wh, ["inv_warehouse_sk"], ["w_warehouse_sk"])
+    j = bhj(j, dd, ["inv_date_sk"], ["d_date_sk"])
+    q = col("inv_quantity_on_hand").cast(dtypes.float64)
+    a = agg2(j, ["w_warehouse_name", "w_warehouse_sk", "i_item_sk", "d_moy"],
+             [AggFunc("count", col("inv_quantity_on_hand"), name="n"),
+              AggFunc("sum", q, name="s"),
+              AggFunc("sum", q * q, name="ssq")])
+    n = col("n").cast(dtypes.float64)
+    mean = col("s") / n
+    var = (col("ssq") - n * mean * mean) / (n - lit(1.0))
+    # cov uses stddev = sqrt(var); compare cov > c as var > c^2 * mean^2
+    from ..exprs import CaseWhen, Literal
+
+    from ..exprs import Sqrt
+
+    base = P.Project(a, [_a(col("w_warehouse_name"), "w_name"),
+                         _a(col("w_warehouse_sk"), "w_sk"),
+                         _a(col("i_item_sk"), "i_sk"), _a(col("d_moy"), "d_moy"),
+                         _a(mean, "mean"), _a(Sqrt(var) / mean, "cov")])
+    f = P.Filter(base, CaseWhen([(col("mean") == 0.0, lit(0.0))], col("cov")) > 1.0)
+    if cov_filter:
+        f = P.Filter(f, col("cov") > 1.5)
+    b = s.execute(f)
+    inv1 = P.Filter(P.MemoryScan(b), col("d_moy") == 1)
+    inv2 = P.Project(P.Filter(P.MemoryScan(b), col("d_moy") == 2),
+                     [_a(col("w_sk"), "w_sk2"), _a(col("i_sk"), "i_sk2"),
+                      _a(col("d_moy"), "d_moy2"), _a(col("mean"), "mean2"),
+                      _a(col("cov"), "cov2")])
+    j2 = shj(inv1, inv2, ["w_sk", "i_sk"], ["w_sk2", "i_sk2"])
+    proj = P.Project(j2, [_a(col("w_sk"), "w_warehouse_sk1"), _a(col("i_sk"), "i_item_sk1"),
+                          _a(col("d_moy"), "d_moy1"), _a(col("mean"), "mean1"),
+                          _a(col("cov"), "cov1"),
+                          _a(col("w_sk2"), "w_warehouse_sk2"), _a(col("i_sk2"), "i_item_sk2"),
+                          _a(col("d_moy2"), "d_moy2"), _a(col("mean2"), "mean2"),
+                          _a(col("cov2"), "cov2")])
+    return topk(proj, [(col("w_warehouse_sk1"), True), (col("i_item_sk1"), True),
+                       (col("d_moy1"), True), (col("mean1"), True),
+                       (col("cov1"), True), (col("d_moy2"), True),
+                       (col("mean2"), True), (col("cov2"), True)], 100000)
+
+
+def q39(cat, s):
+    return _q39(cat, s, cov_filter=False)
+
+
+def q41(cat, s):
+    i1 = P.Filter(cat.scan("item", ["i_product_name", "i_manufact_id", "i_manufact"]),
+                  col("i_manufact_id").between(738, 778))
+    it = cat.scan("item", ["i_manufact", "i_category", "i_color", "i_units", "i_size"])
+
+    def block(cat_, colors, units, sizes):
+        return (col("i_category") == lit(cat_)) & col("i_color").isin(colors) \
+            & col("i_units").isin(units) & col("i_size").isin(sizes)
+
+    pred = (block("Women", ["powder", "khaki"], ["Ounce", "Oz"], ["medium", "extra large"])
+            | block("Women", ["brown", "honeydew"], ["Bunch", "Ton"], ["N/A", "small"])
+            | block("Men", ["floral", "deep"], ["N/A", "Dozen"], ["petite", "large"])
+            | block("Men", ["light", "cornflower"], ["Box", "Pound"], ["medium", "extra large"])
+            | block("Women", ["midnight", "snow"], ["Pallet", "Gross"], ["medium", "extra large"])
+            | block("Women", ["cyan", "papaya"], ["Cup", "Dram"], ["N/A", "small"])
+            | block("Men", ["orange", "frosted"], ["Each", "Tbl"], ["petite", "large"])
+            | block("Men", ["forest", "ghost"], ["Lb", "Bundle"], ["medium", "extra large"]))
+    sub = P.Filter(it, pred)
+    mset = P.HashAgg(P.Broadcast(P.Project(sub, [_a(col("i_manufact"), "m")])),
+                     [_a(col("m"), "m")], [], mode="complete")
+    j = P.HashJoin(i1, mset, [col("i_manufact")], [col("m")], how="semi",
+                   build_side="right", broadcast=True)
+    d = agg2(j, ["i_product_name"], [])
+    return topk(d, [(col("i_product_name"), True)], 100)
+
+
+def q95(cat, s):
+    lo = _days(1999, 2, 1)
+    ws = cat.scan("web_sales", ["ws_ship_date_sk", "ws_ship_addr_sk", "ws_order_number",
+                                "ws_warehouse_sk", "ws_ext_ship_cost", "ws_net_profit",
+                                "ws_web_site_sk"])
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_date"]),
+                  col("d_date").cast(dtypes.int32).between(lo, lo + 60))
+    ca = P.Filter(cat.scan("customer_address", ["ca_address_sk", "ca_state"]),
+                  col("ca_state") == lit("IL"))
+    site = P.Filter(cat.scan("web_site", ["web_site_sk", "web_company_name"]),
+                    col("web_company_name") == lit("pri"))
+    j = bhj(ws, dd, ["ws_ship_date_sk"], ["d_date_sk"])
+    j = bhj(j, ca, ["ws_ship_addr_sk"], ["ca_address_sk"])
+    j = bhj(j, site, ["ws_web_site_sk"], ["web_site_sk"])
+    # ws_wh: orders shipped from more than one warehouse
+    all_orders = cat.scan("web_sales", ["ws_order_number", "ws_warehouse_sk"])
+    ord_wh = agg2(P.Project(all_orders, [_a(col("ws_order_number"), "o"),
+                                         _a(col("ws_warehouse_sk"), "w")]), ["o", "w"], [])
+    multi = P.Filter(P.HashAgg(ord_wh, [_a(col("o"), "o")],
+                               [AggFunc("count_star", None, name="nwh")], mode="complete"),
+                     col("nwh") > 1)
+    multi_b = s.execute(multi)
+    j = P.HashJoin(P.Exchange(j, "hash", [col("ws_order_number")]),
+                   P.Exchange(P.MemoryScan(multi_b), "hash", [col("o")]),
+                   [col("ws_order_number")], [col("o")], how="semi", build_side="right")
+    # returned orders that are also multi-warehouse orders
+    wr = cat.scan("web_returns", ["wr_order_number"])
+    wr_multi = P.HashJoin(P.Exchange(wr, "hash", [col("wr_order_number")]),
+                          P.Exchange(P.MemoryScan(multi_b), "hash", [col("o")]),
+                          [col("wr_order_number")], [col("o")], how="semi",
+                          build_side="right")
+    j = P.HashJoin(j, P.Exchange(wr_multi, "hash", [col("wr_order_number")]),
+                   [col("ws_order_number")], [col("wr_order_number")],
+                   how="semi", build_side="right")
+    return _global_agg(j, [
+        AggFunc("count_distinct", col("ws_order_number"), name="order_count"),
+        AggFunc("sum", col("ws_ext_ship_cost"), name="total_shipping_cost"),
+        AggFunc("sum", col("ws_net_profit"), name="total_net_profit")])
+
+
+QUERIES.update({"q8": q8, "q24": q24, "q39": q39, "q41": q41, "q95": q95})

@@ -25,7 +25,7 @@ SCHEMAS = {
         "i_item_sk": i64, "i_item_id": s, "i_item_desc": s, "i_category": s,
         "i_category_id": i32, "i_brand": s, "i_brand_id": i32,
         "i_manufact_id": i32, "i_class": s, "i_class_id": i32,
-        "i_current_price": f64, "i_manager_id": i32, "i_product_name": s, "i_color": s,
+        "i_current_price": f64, "i_manager_id": i32, "i_product_name": s, "i_color": s, "i_units": s, "i_size": s, "i_manufact": s,
     },
     "customer": {
         "c_customer_sk": i64, "c_customer_id": s, "c_first_name": s,
@@ -52,7 +52,7 @@ SCHEMAS = {
     "store": {
         "s_store_sk": i64, "s_store_id": s, "s_store_name": s, "s_state": s,
         "s_county": s, "s_zip": s, "s_city": s, "s_number_employees": i32,
-        "s_gmt_offset": f64, "s_company_id": i32, "s_street_name": s,
+        "s_gmt_offset": f64, "s_company_id": i32, "s_street_name": s, "s_market_id": i32,
     },
     "warehouse": {
         "w_warehouse_sk": i64, "w_warehouse_name": s, "w_warehouse_sq_ft": i32,
