@@ -414,6 +414,15 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
         put(f"{pre}_returning_customer_sk", c, cv)
         ra, rav = fk(n_addr)
         put(f"{pre}_returning_addr_sk", ra, rav)
+        if pre == "wr":
+            rc1, rc1v = fk(n_cdemo)
+            put("wr_refunded_cdemo_sk", rc1, rc1v)
+            rc2, rc2v = fk(n_cdemo)
+            put("wr_returning_cdemo_sk", rc2, rc2v)
+            ra2, ra2v = fk(n_addr)
+            put("wr_refunded_addr_sk", ra2, ra2v)
+            rr, rrv = fk(BASE_ROWS["reason"], 0.02)
+            put("wr_reason_sk", rr, rrv)
         if pre == "cr":
             cp, cpv = fk(BASE_ROWS["catalog_page"], 0.02)
             put("cr_catalog_page_sk", cp, cpv)
@@ -482,7 +491,7 @@ def _nparts_for(table: str, sf: float) -> int:
     return 1
 
 
-DATAGEN_VERSION = 6
+DATAGEN_VERSION = 7
 
 
 def dataset_root(root: str, sf: float) -> str:

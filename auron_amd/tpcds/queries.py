@@ -3008,3 +3008,216 @@ def q95(cat, s):
 
 
 QUERIES.update({"q8": q8, "q24": q24, "q39": q39, "q41": q41, "q95": q95})
+
+
+# ------------------------------- batch 11
+def q83(cat, s):
+    # the three dates' week sets
+    dates = [_days(2000, 6, 30), _days(2000, 9, 27), _days(2000, 11, 17)]
+    wk = P.HashAgg(P.Broadcast(P.Filter(
+        cat.scan("date_dim", ["d_date", "d_week_seq"]),
+        col("d_date").cast(dtypes.int32).isin(dates))),
+        [_a(col("d_week_seq"), "wk")], [], mode="complete")
+    dd = cat.scan("date_dim", ["d_date_sk", "d_week_seq"])
+    dd_in = P.HashJoin(dd, wk, [col("d_week_seq")], [col("wk")], how="semi",
+                       build_side="right", broadcast=True)
+
+    def items(rets, rpre, rev):
+        rt = cat.scan(rets, [f"{rpre}_item_sk", f"{rpre}_returned_date_sk",
+                             f"{rpre}_return_quantity"])
+        it = cat.scan("item", ["i_item_sk", "i_item_id"])
+        j = bhj(rt, dd_in, [f"{rpre}_returned_date_sk"], ["d_date_sk"])
+        j = bhj(j, it, [f"{rpre}_item_sk"], ["i_item_sk"])
+        return agg2(j, ["item_id"],
+                    [AggFunc("sum", col(f"{rpre}_return_quantity").cast(dtypes.int64), name=rev)],
+                    key_exprs=[col("i_item_id")])
+
+    sri = items("store_returns", "sr", "sr_item_qty")
+    cri = P.Project(items("catalog_returns", "cr", "cr_item_qty"),
+                    [_a(col("item_id"), "cr_id"), _a(col("cr_item_qty"), "cr_item_qty")])
+    wri = P.Project(items("web_returns", "wr", "wr_item_qty"),
+                    [_a(col("item_id"), "wr_id"), _a(col("wr_item_qty"), "wr_item_qty")])
+    j = shj(sri, cri, ["item_id"], ["cr_id"])
+    j = P.HashJoin(j, P.Exchange(wri, "hash", [col("wr_id")]), [col("item_id")],
+                   [col("wr_id")], how="inner", build_side="right")
+    srq = col("sr_item_qty").cast(dtypes.float64)
+    crq = col("cr_item_qty").cast(dtypes.float64)
+    wrq = col("wr_item_qty").cast(dtypes.float64)
+    tot = srq + crq + wrq
+    proj = P.Project(j, [_a(col("item_id"), "item_id"), _a(col("sr_item_qty"), "sr_item_qty"),
+                         _a(srq / tot / lit(3.0) * lit(100.0), "sr_dev"),
+                         _a(col("cr_item_qty"), "cr_item_qty"),
+                         _a(crq / tot / lit(3.0) * lit(100.0), "cr_dev"),
+                         _a(col("wr_item_qty"), "wr_item_qty"),
+                         _a(wrq / tot / lit(3.0) * lit(100.0), "wr_dev"),
+                         _a(tot / lit(3.0), "average")])
+    return topk(proj, [(col("item_id"), True), (col("sr_item_qty"), True)], 100)
+
+
+def q85(cat, s):
+    from ..exprs import Substr
+
+    ws = cat.scan("web_sales", ["ws_web_page_sk", "ws_item_sk", "ws_order_number",
+                                "ws_sold_date_sk", "ws_quantity", "ws_sales_price",
+                                "ws_net_profit"])
+    wr = cat.scan("web_returns", ["wr_item_sk", "wr_order_number", "wr_refunded_cdemo_sk",
+                                  "wr_returning_cdemo_sk", "wr_refunded_addr_sk",
+                                  "wr_reason_sk", "wr_refunded_cash", "wr_fee"])
+    j = shj(ws, wr, ["ws_item_sk", "ws_order_number"], ["wr_item_sk", "wr_order_number"])
+    wp = cat.scan("web_page", ["wp_web_page_sk"])
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_year"]), col("d_year") == 2000)
+    cd1 = P.Project(cat.scan("customer_demographics",
+                             ["cd_demo_sk", "cd_marital_status", "cd_education_status"]),
+                    [_a(col("cd_demo_sk"), "cd1_sk"), _a(col("cd_marital_status"), "cd1_ms"),
+                     _a(col("cd_education_status"), "cd1_es")])
+    cd2 = P.Project(cat.scan("customer_demographics",
+                             ["cd_demo_sk", "cd_marital_status", "cd_education_status"]),
+                    [_a(col("cd_demo_sk"), "cd2_sk"), _a(col("cd_marital_status"), "cd2_ms"),
+                     _a(col("cd_education_status"), "cd2_es")])
+    ca = P.Filter(cat.scan("customer_address", ["ca_address_sk", "ca_country", "ca_state"]),
+                  col("ca_country") == lit("United States"))
+    re = cat.scan("reason", ["r_reason_sk", "r_reason_desc"])
+    j = bhj(j, wp, ["ws_web_page_sk"], ["wp_web_page_sk"])
+    j = bhj(j, dd, ["ws_sold_date_sk"], ["d_date_sk"])
+    j = bhj(j, cd1, ["wr_refunded_cdemo_sk"], ["cd1_sk"])
+    j = bhj(j, cd2, ["wr_returning_cdemo_sk"], ["cd2_sk"])
+    j = bhj(j, ca, ["wr_refunded_addr_sk"], ["ca_address_sk"])
+    j = bhj(j, re, ["wr_reason_sk"], ["r_reason_sk"])
+    sp = col("ws_sales_price")
+    np_ = col("ws_net_profit")
+    msm = (col("cd1_ms") == col("cd2_ms")) & (col("cd1_es") == col("cd2_es"))
+    c1 = (((col("cd1_ms") == lit("M")) & (col("cd1_es") == lit("Advanced Degree"))
+           & msm & sp.between(100.0, 150.0))
+          | ((col("cd1_ms") == lit("S")) & (col("cd1_es") == lit("College"))
+             & msm & sp.between(50.0, 100.0))
+          | ((col("cd1_ms") == lit("W")) & (col("cd1_es") == lit("2 yr Degree"))
+             & msm & sp.between(150.0, 200.0)))
+    c2 = ((col("ca_state").isin(["IN", "OH", "NJ"]) & np_.between(100.0, 200.0))
+          | (col("ca_state").isin(["WI", "CT", "KY"]) & np_.between(150.0, 300.0))
+          | (col("ca_state").isin(["LA", "IA", "AR"]) & np_.between(50.0, 250.0)))
+    f = P.Filter(j, c1 & c2)
+    pre = P.Project(f, [_a(Substr(col("r_reason_desc"), 1, 20), "reason20"),
+                        _a(col("ws_quantity"), "q"), _a(col("wr_refunded_cash"), "rc"),
+                        _a(col("wr_fee"), "fee")])
+    a = agg2(pre, ["reason20"], [AggFunc("avg", col("q"), name="avg_q"),
+                                 AggFunc("avg", col("rc"), name="avg_rc"),
+                                 AggFunc("avg", col("fee"), name="avg_fee")])
+    return topk(a, [(col("reason20"), True)], 100)
+
+
+def q70(cat, s):
+    from ..exprs import CaseWhen, Literal, WindowFunc
+
+    ss = cat.scan("store_sales", ["ss_sold_date_sk", "ss_store_sk", "ss_net_profit"])
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_month_seq"]),
+                  col("d_month_seq").between(1200, 1211))
+    st = cat.scan("store", ["s_store_sk", "s_state", "s_county"])
+    # subquery: states whose per-state rank <= 5 (partition-by-state rank is
+    # always 1 — the published query's quirk — so this is "states with sales")
+    sub = bhj(ss, dd, ["ss_sold_date_sk"], ["d_date_sk"])
+    sub = bhj(sub, st, ["ss_store_sk"], ["s_store_sk"])
+    states = P.HashAgg(P.Exchange(P.Project(sub, [_a(col("s_state"), "st")]),
+                                  "hash", [col("st")]),
+                       [_a(col("st"), "st")], [], mode="complete")
+    j = bhj(ss, dd, ["ss_sold_date_sk"], ["d_date_sk"])
+    j = bhj(j, st, ["ss_store_sk"], ["s_store_sk"])
+    j = P.HashJoin(j, P.Broadcast(states), [col("s_state")], [col("st")],
+                   how="semi", build_side="right")
+    pre = P.Project(j, [_a(col("s_state"), "s_state"), _a(col("s_county"), "s_county"),
+                        _a(col("ss_net_profit"), "np")])
+    ex = rollup_expand(pre, ["s_state", "s_county"], [dtypes.string, dtypes.string],
+                       ["np"])
+    a = agg2(ex, ["s_state", "s_county", "_lochier"],
+             [AggFunc("sum", col("np"), name="total_sum")])
+    m = P.Project(a, [_a(col("total_sum"), "total_sum"), _a(col("s_state"), "s_state"),
+                      _a(col("s_county"), "s_county"), _a(col("_lochier"), "lochierarchy"),
+                      _a(CaseWhen([(col("_lochier") == 0, col("s_state"))],
+                                  Literal(None, dtypes.string)), "_pst")])
+    rank_key = ((col("total_sum") * lit(100.0)) + lit(0.5)).cast(dtypes.int64)
+    w = P.Window(P.Exchange(m, "hash", [col("lochierarchy")]),
+                 [col("lochierarchy"), col("_pst")], [(rank_key, False)],
+                 [_a(WindowFunc("rank"), "rank_within_parent")])
+    proj = P.Project(w, [_a(col("total_sum"), "total_sum"), _a(col("s_state"), "s_state"),
+                         _a(col("s_county"), "s_county"),
+                         _a(col("lochierarchy"), "lochierarchy"),
+                         _a(col("rank_within_parent"), "rank_within_parent"),
+                         _a(col("_pst"), "_pst")])
+    out = topk(proj, [(col("lochierarchy"), False), (col("_pst"), True),
+                      (col("rank_within_parent"), True)], 100)
+    return P.Project(out, [_a(col(c), c) for c in
+                           ["total_sum", "s_state", "s_county", "lochierarchy",
+                            "rank_within_parent"]])
+
+
+def q66(cat, s):
+    from ..exprs import CaseWhen
+
+    dd = cat.scan("date_dim", ["d_date_sk", "d_year", "d_moy"])
+    dd_y = P.Filter(dd, col("d_year") == 2001)
+    td = P.Filter(cat.scan("time_dim", ["t_time_sk"]),
+                  (col("t_time_sk") >= 30838) & (col("t_time_sk") <= 59838))
+    sm = P.Filter(cat.scan("ship_mode", ["sm_ship_mode_sk", "sm_carrier"]),
+                  col("sm_carrier").isin(["UPS", "FEDEX"]))  # adapted: DHL/BARIAN absent
+    wh = cat.scan("warehouse", ["w_warehouse_sk", "w_warehouse_name", "w_warehouse_sq_ft",
+                                "w_city", "w_county", "w_state", "w_country"])
+    months = ["jan", "feb", "mar", "apr", "may", "jun",
+              "jul", "aug", "sep", "oct", "nov", "dec"]
+
+    def chan(fact, pre, sales_expr, net_expr):
+        fs = cat.scan(fact, [f"{pre}_warehouse_sk", f"{pre}_sold_date_sk",
+                             f"{pre}_sold_time_sk", f"{pre}_ship_mode_sk",
+                             f"{pre}_ext_sales_price", f"{pre}_quantity",
+                             f"{pre}_net_paid" if pre == "ws" else f"{pre}_net_paid_inc_tax",
+                             ])
+        j = bhj(fs, dd_y, [f"{pre}_sold_date_sk"], ["d_date_sk"])
+        j = bhj(j, td, [f"{pre}_sold_time_sk"], ["t_time_sk"])
+        j = bhj(j, sm, [f"{pre}_ship_mode_sk"], ["sm_ship_mode_sk"])
+        j = bhj(j, wh, [f"{pre}_warehouse_sk"], ["w_warehouse_sk"])
+        aggs = []
+        pre_cols = [_a(col(c), c) for c in ["w_warehouse_name", "w_warehouse_sq_ft",
+                                            "w_city", "w_county", "w_state", "w_country"]]
+        pre_cols.append(_a(col("d_year"), "year"))
+        for m_i, m in enumerate(months):
+            pre_cols.append(_a(CaseWhen([(col("d_moy") == m_i + 1, sales_expr(pre))],
+                                        lit(0.0)), f"{m}_sales_v"))
+            pre_cols.append(_a(CaseWhen([(col("d_moy") == m_i + 1, net_expr(pre))],
+                                        lit(0.0)), f"{m}_net_v"))
+        prej = P.Project(j, pre_cols)
+        keys = ["w_warehouse_name", "w_warehouse_sq_ft", "w_city", "w_county",
+                "w_state", "w_country", "year"]
+        for m in months:
+            aggs.append(AggFunc("sum", col(f"{m}_sales_v"), name=f"{m}_sales"))
+            aggs.append(AggFunc("sum", col(f"{m}_net_v"), name=f"{m}_net"))
+        return agg2(j if False else prej, keys, aggs)
+
+    wsr = chan("web_sales", "ws",
+               lambda p: col("ws_ext_sales_price") * col("ws_quantity").cast(dtypes.float64),
+               lambda p: col("ws_net_paid") * col("ws_quantity").cast(dtypes.float64))
+    csr = chan("catalog_sales", "cs",
+               lambda p: col("cs_ext_sales_price") * col("cs_quantity").cast(dtypes.float64),
+               lambda p: col("cs_net_paid_inc_tax") * col("cs_quantity").cast(dtypes.float64))
+    u = P.Union([wsr, csr])
+    keys = ["w_warehouse_name", "w_warehouse_sq_ft", "w_city", "w_county",
+            "w_state", "w_country", "year"]
+    aggs = []
+    for m in months:
+        aggs.append(AggFunc("sum", col(f"{m}_sales"), name=f"{m}_sales"))
+        aggs.append(AggFunc("sum", col(f"{m}_sales") / col("w_warehouse_sq_ft").cast(dtypes.float64),
+                            name=f"{m}_spsf"))
+        aggs.append(AggFunc("sum", col(f"{m}_net"), name=f"{m}_net"))
+    a = agg2(u, keys, aggs)
+    # add ship_carriers constant column
+    proj_cols = [_a(col(k), k) for k in keys[:6]]
+    proj_cols.append(_a(lit("UPS,FEDEX"), "ship_carriers"))
+    proj_cols.append(_a(col("year"), "year"))
+    for m in months:
+        proj_cols.append(_a(col(f"{m}_sales"), f"{m}_sales"))
+    for m in months:
+        proj_cols.append(_a(col(f"{m}_spsf"), f"{m}_sales_per_sq_foot"))
+    for m in months:
+        proj_cols.append(_a(col(f"{m}_net"), f"{m}_net"))
+    proj = P.Project(a, proj_cols)
+    return topk(proj, [(col("w_warehouse_name"), True)], 100)
+
+
+QUERIES.update({"q66": q66, "q70": q70, "q83": q83, "q85": q85})

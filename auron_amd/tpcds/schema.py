@@ -124,7 +124,7 @@ SCHEMAS = {
     },
     "web_returns": {
         "wr_returned_date_sk": i64, "wr_item_sk": i64, "wr_order_number": i64,
-        "wr_returning_customer_sk": i64, "wr_returning_addr_sk": i64, "wr_return_quantity": i32,
+        "wr_returning_customer_sk": i64, "wr_returning_addr_sk": i64, "wr_refunded_cdemo_sk": i64, "wr_returning_cdemo_sk": i64, "wr_refunded_addr_sk": i64, "wr_reason_sk": i64, "wr_return_quantity": i32,
         "wr_return_amt": f64, "wr_net_loss": f64, "wr_fee": f64,
         "wr_refunded_cash": f64, "wr_reversed_charge": f64, "wr_account_credit": f64,
     },
