@@ -665,9 +665,14 @@ class Executor:
                     out = Column(dtypes.int64, torch.ones(n, dtype=torch.int64, device=device))
             elif wf.fn in ("sum", "avg", "count", "min", "max"):
                 val = wf.arg.eval(sb)
-                acc, cnt = ops.agg_scatter(seg, max(nseg, 1), val, wf.fn if wf.fn != "count" else "count")
-                fin = self._finalize_agg(AggFunc(wf.fn, None, name=al.name), val.dtype, acc, cnt)
-                out = fin.gather(seg)
+                if node.order_by and n:
+                    # Spark's default frame with ORDER BY: rows unbounded
+                    # preceding .. current row (running aggregate)
+                    out = self._running_window(wf.fn, val, seg, seg_start)
+                else:
+                    acc, cnt = ops.agg_scatter(seg, max(nseg, 1), val, wf.fn if wf.fn != "count" else "count")
+                    fin = self._finalize_agg(AggFunc(wf.fn, None, name=al.name), val.dtype, acc, cnt)
+                    out = fin.gather(seg)
             elif wf.fn in ("lead", "lag"):
                 val = wf.arg.eval(sb)
                 k = wf.offset if wf.offset else 1
@@ -683,6 +688,59 @@ class Executor:
             names.append(al.name)
             cols.append(out)
         return [RecordBatch(names, cols)]
+
+    def _running_window(self, fn: str, val: Column, seg: torch.Tensor,
+                        seg_start: torch.Tensor) -> Column:
+        """Cumulative frame within segments (nulls skipped)."""
+        device = seg.device
+        n = seg.numel()
+        v = val.data
+        if v.dtype in (torch.int8, torch.int16, torch.int32):
+            v = v.to(torch.int64)
+        elif v.dtype == torch.float32:
+            v = v.to(torch.float64)
+        valid = val.validity if val.validity is not None else torch.ones(n, dtype=torch.bool, device=device)
+        cum_valid = torch.cumsum(valid.to(torch.int64), 0)
+        base_valid = cum_valid[seg_start][seg] - valid[seg_start][seg].to(torch.int64)
+        run_count = cum_valid - base_valid
+        if fn == "count":
+            return Column(dtypes.int64, run_count)
+        if fn in ("sum", "avg"):
+            z = torch.where(valid, v, torch.zeros_like(v))
+            cs = torch.cumsum(z, 0)
+            base = cs[seg_start][seg] - z[seg_start][seg]
+            run = cs - base
+            validity = run_count > 0
+            if bool(validity.all()):
+                validity = None
+            if fn == "avg":
+                return Column(dtypes.float64,
+                              run.to(torch.float64) / run_count.clamp(min=1).to(torch.float64),
+                              validity)
+            dt = dtypes.float64 if run.dtype == torch.float64 else dtypes.int64
+            if val.dtype.code == dtypes.DECIMAL64:
+                dt = val.dtype
+            return Column(dt, run, validity)
+        # running min/max: segment-reset via a cummax over segment-shifted
+        # values (each segment's band is disjoint, so cummax cannot carry
+        # across segment boundaries; invalid rows get a never-wins filler)
+        neg = fn == "min"
+        x = v.to(torch.float64)
+        if neg:
+            x = -x
+        big = (x[valid].abs().max().item() + 1.0) if bool(valid.any()) else 1.0
+        x = torch.where(valid, x, torch.full_like(x, -2.0 * big))
+        shift = seg.to(torch.float64) * (8.0 * big)
+        run = torch.cummax(x + shift, 0).values - shift
+        if neg:
+            run = -run
+        validity = run_count > 0
+        if bool(validity.all()):
+            validity = None
+        out = run
+        if val.dtype.is_integer:
+            return Column(dtypes.int64, out.to(torch.int64), validity)
+        return Column(dtypes.float64, out, validity)
 
     def _col_eq_adjacent(self, c: Column) -> torch.Tensor:
         """eq mask between row i and i-1, for rows 1..n-1 (null==null)."""

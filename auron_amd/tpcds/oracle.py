@@ -2724,3 +2724,149 @@ def q66(root, sf):
 
 
 ORACLES.update({"q66": q66, "q70": q70, "q83": q83, "q85": q85})
+
+
+# ------------------------------- batch 12 oracles
+def q51(root, sf):
+    import numpy as np
+    import pandas as pd
+
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_date", "d_month_seq"])
+    dd = dd[dd.d_month_seq.between(1200, 1211)].copy()
+    dd["d_i"] = _date_i(dd)
+
+    def v1(fact, pre):
+        fs = _read(root, sf, fact, [f"{pre}_item_sk", f"{pre}_sold_date_sk",
+                                    f"{pre}_sales_price"])
+        fs = fs[fs[f"{pre}_item_sk"].notna()]
+        j = _merge(fs, dd, f"{pre}_sold_date_sk", "d_date_sk")
+        a = j.groupby([f"{pre}_item_sk", "d_i"], dropna=False)[f"{pre}_sales_price"] \
+             .sum(min_count=1).reset_index(name="s")
+        a = a.sort_values([f"{pre}_item_sk", "d_i"])
+        a["cume_sales"] = a.groupby(f"{pre}_item_sk").s.cumsum()
+        return a.rename(columns={f"{pre}_item_sk": "item_sk", "d_i": "d_date"})
+
+    web = v1("web_sales", "ws")[["item_sk", "d_date", "cume_sales"]] \
+        .rename(columns={"cume_sales": "web_sales"})
+    store = v1("store_sales", "ss")[["item_sk", "d_date", "cume_sales"]] \
+        .rename(columns={"cume_sales": "store_sales"})
+    j = web.merge(store, on=["item_sk", "d_date"], how="outer")
+    j = j.sort_values(["item_sk", "d_date"])
+    # SQL running max ignores NULL rows (carries the prior max forward);
+    # pandas cummax leaves NaN at NaN inputs, so ffill within the partition
+    j["web_cumulative"] = j.groupby("item_sk").web_sales.cummax()
+    j["web_cumulative"] = j.groupby("item_sk").web_cumulative.ffill()
+    j["store_cumulative"] = j.groupby("item_sk").store_sales.cummax()
+    j["store_cumulative"] = j.groupby("item_sk").store_cumulative.ffill()
+    f = j[(j.web_cumulative > j.store_cumulative).fillna(False)]
+    out = f[["item_sk", "d_date", "web_sales", "store_sales",
+             "web_cumulative", "store_cumulative"]].copy()
+    out["item_sk"] = out.item_sk.astype("Int64")
+    out = out.sort_values(["item_sk", "d_date"], na_position="first").head(100)
+    return out.reset_index(drop=True)
+
+
+def q75(root, sf):
+    import pandas as pd
+
+    it = _read(root, sf, "item", ["i_item_sk", "i_brand_id", "i_class_id",
+                                  "i_category_id", "i_manufact_id", "i_category"])
+    it = it[it.i_category == "Books"]
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year"])
+
+    def chan(fact, pre, rets, rpre, k2, rk2):
+        amt = f"{rpre}_return_{'amount' if rpre == 'cr' else 'amt'}"
+        fs = _read(root, sf, fact, [f"{pre}_item_sk", f"{pre}_sold_date_sk", k2,
+                                    f"{pre}_quantity", f"{pre}_ext_sales_price"])
+        rt = _read(root, sf, rets, [rk2, f"{rpre}_item_sk", f"{rpre}_return_quantity", amt])
+        j = fs.merge(rt.dropna(subset=[rk2, f"{rpre}_item_sk"]),
+                     left_on=[k2, f"{pre}_item_sk"], right_on=[rk2, f"{rpre}_item_sk"],
+                     how="left")
+        j = _merge(j, it, f"{pre}_item_sk", "i_item_sk")
+        j = _merge(j, dd, f"{pre}_sold_date_sk", "d_date_sk")
+        cnt = j[f"{pre}_quantity"] - j[f"{rpre}_return_quantity"].fillna(0)
+        amt_e = j[f"{pre}_ext_sales_price"] - j[amt].fillna(0.0)
+        return pd.DataFrame({"d_year": j.d_year, "i_brand_id": j.i_brand_id,
+                             "i_class_id": j.i_class_id, "i_category_id": j.i_category_id,
+                             "i_manufact_id": j.i_manufact_id,
+                             "sales_cnt": cnt, "sales_amt": amt_e})
+
+    u = pd.concat([
+        chan("catalog_sales", "cs", "catalog_returns", "cr", "cs_order_number", "cr_order_number"),
+        chan("store_sales", "ss", "store_returns", "sr", "ss_ticket_number", "sr_ticket_number"),
+        chan("web_sales", "ws", "web_returns", "wr", "ws_order_number", "wr_order_number"),
+    ], ignore_index=True).drop_duplicates()
+    g = u.groupby(["d_year", "i_brand_id", "i_class_id", "i_category_id",
+                   "i_manufact_id"], dropna=False).agg(
+        sales_cnt=("sales_cnt", lambda x: x.sum(min_count=1)),
+        sales_amt=("sales_amt", lambda x: x.sum(min_count=1))).reset_index()
+    curr = g[g.d_year == 2002]
+    prev = g[g.d_year == 2001]
+    keys = ["i_brand_id", "i_class_id", "i_category_id", "i_manufact_id"]
+    j = curr.dropna(subset=keys).merge(prev.dropna(subset=keys), on=keys,
+                                       suffixes=("", "_p"))
+    f = j[(j.sales_cnt / j.sales_cnt_p) < 0.9]
+    out = pd.DataFrame({"prev_year": f.d_year_p, "year": f.d_year,
+                        "i_brand_id": f.i_brand_id, "i_class_id": f.i_class_id,
+                        "i_category_id": f.i_category_id,
+                        "i_manufact_id": f.i_manufact_id,
+                        "prev_yr_cnt": f.sales_cnt_p.astype("Int64"),
+                        "curr_yr_cnt": f.sales_cnt.astype("Int64"),
+                        "sales_cnt_diff": (f.sales_cnt - f.sales_cnt_p).astype("Int64"),
+                        "sales_amt_diff": f.sales_amt - f.sales_amt_p})
+    out = out.sort_values(["sales_cnt_diff", "sales_amt_diff"], na_position="first").head(100)
+    return out.reset_index(drop=True)
+
+
+def q78(root, sf):
+    import numpy as np
+    import pandas as pd
+
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year"])
+
+    def cte(fact, pre, rets, rpre, cust_fk, k2, rk2):
+        fs = _read(root, sf, fact, [f"{pre}_item_sk", cust_fk, f"{pre}_sold_date_sk", k2,
+                                    f"{pre}_quantity", f"{pre}_wholesale_cost",
+                                    f"{pre}_sales_price"])
+        rt = _read(root, sf, rets, [rk2, f"{rpre}_item_sk"]).dropna()
+        j = fs.merge(rt, left_on=[k2, f"{pre}_item_sk"],
+                     right_on=[rk2, f"{rpre}_item_sk"], how="left")
+        j = j[j[rk2].isna()]
+        j = _merge(j, dd, f"{pre}_sold_date_sk", "d_date_sk")
+        return j.groupby(["d_year", f"{pre}_item_sk", cust_fk], dropna=False).agg(
+            qty=(f"{pre}_quantity", lambda x: x.sum(min_count=1)),
+            wc=(f"{pre}_wholesale_cost", lambda x: x.sum(min_count=1)),
+            sp=(f"{pre}_sales_price", lambda x: x.sum(min_count=1))).reset_index()
+
+    ssc = cte("store_sales", "ss", "store_returns", "sr", "ss_customer_sk",
+              "ss_ticket_number", "sr_ticket_number")
+    wsc = cte("web_sales", "ws", "web_returns", "wr", "ws_bill_customer_sk",
+              "ws_order_number", "wr_order_number")
+    csc = cte("catalog_sales", "cs", "catalog_returns", "cr", "cs_bill_customer_sk",
+              "cs_order_number", "cr_order_number")
+    ssc = ssc[ssc.d_year == 2000]
+    j = ssc.merge(wsc.rename(columns={"ws_item_sk": "ss_item_sk",
+                                      "ws_bill_customer_sk": "ss_customer_sk"}),
+                  on=["d_year", "ss_item_sk", "ss_customer_sk"], how="left",
+                  suffixes=("", "_w"))
+    j = j.merge(csc.rename(columns={"cs_item_sk": "ss_item_sk",
+                                    "cs_bill_customer_sk": "ss_customer_sk"}),
+                on=["d_year", "ss_item_sk", "ss_customer_sk"], how="left",
+                suffixes=("", "_c"))
+    f = j[(j.qty_w.fillna(0) > 0) & (j.qty_c.fillna(0) > 0)]
+    denom = (f.qty_w + f.qty_c).fillna(1)
+    ratio = np.floor(f.qty / denom * 100.0 + 0.5) / 100.0
+    out = pd.DataFrame({"ratio": ratio, "store_qty": f.qty.astype("Int64"),
+                        "store_wholesale_cost": f.wc, "store_sales_price": f.sp,
+                        "other_chan_qty": (f.qty_w.fillna(0) + f.qty_c.fillna(0)).astype("Int64"),
+                        "other_chan_wholesale_cost": f.wc_w.fillna(0) + f.wc_c.fillna(0),
+                        "other_chan_sales_price": f.sp_w.fillna(0) + f.sp_c.fillna(0)})
+    out = out.sort_values(["ratio", "store_qty", "store_wholesale_cost",
+                           "store_sales_price", "other_chan_qty",
+                           "other_chan_wholesale_cost", "other_chan_sales_price"],
+                          ascending=[True, False, False, False, True, True, True],
+                          na_position="first").head(100)
+    return out.reset_index(drop=True)
+
+
+ORACLES.update({"q51": q51, "q75": q75, "q78": q78})

@@ -3221,3 +3221,167 @@ def q66(cat, s):
 
 
 QUERIES.update({"q66": q66, "q70": q70, "q83": q83, "q85": q85})
+
+
+# ------------------------------- batch 12
+def q51(cat, s):
+    from ..exprs import CaseWhen, IsNull, Not, WindowFunc
+
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_date", "d_month_seq"]),
+                  col("d_month_seq").between(1200, 1211))
+
+    def v1(fact, pre):
+        fs = P.Filter(cat.scan(fact, [f"{pre}_item_sk", f"{pre}_sold_date_sk",
+                                      f"{pre}_sales_price"]),
+                      col(f"{pre}_item_sk").is_not_null())
+        j = bhj(fs, dd, [f"{pre}_sold_date_sk"], ["d_date_sk"])
+        a = agg2(j, ["item_sk", "d_date"],
+                 [AggFunc("sum", col(f"{pre}_sales_price"), name="s")],
+                 key_exprs=[col(f"{pre}_item_sk"), col("d_date")])
+        ex = P.Exchange(a, "hash", [col("item_sk")])
+        return P.Window(ex, [col("item_sk")], [(col("d_date"), True)],
+                        [_a(WindowFunc("sum", col("s")), "cume_sales")])
+
+    web = P.Project(v1("web_sales", "ws"),
+                    [_a(col("item_sk"), "w_item"), _a(col("d_date"), "w_date"),
+                     _a(col("cume_sales"), "web_sales")])
+    store = P.Project(v1("store_sales", "ss"),
+                      [_a(col("item_sk"), "s_item"), _a(col("d_date"), "s_date"),
+                       _a(col("cume_sales"), "store_sales")])
+    j = shj(web, store, ["w_item", "w_date"], ["s_item", "s_date"], how="full")
+    item = CaseWhen([(Not(IsNull(col("w_item"))), col("w_item"))], col("s_item"))
+    ddate = CaseWhen([(Not(IsNull(col("w_date"))), col("w_date").cast(dtypes.int32))],
+                     col("s_date").cast(dtypes.int32))
+    x = P.Project(j, [_a(item, "item_sk"), _a(ddate, "d_date"),
+                      _a(col("web_sales"), "web_sales"),
+                      _a(col("store_sales"), "store_sales")])
+    w = P.Window(P.Exchange(x, "hash", [col("item_sk")]),
+                 [col("item_sk")], [(col("d_date"), True)],
+                 [_a(WindowFunc("max", col("web_sales")), "web_cumulative"),
+                  _a(WindowFunc("max", col("store_sales")), "store_cumulative")])
+    f = P.Filter(w, col("web_cumulative") > col("store_cumulative"))
+    proj = P.Project(f, [_a(col(c), c) for c in
+                         ["item_sk", "d_date", "web_sales", "store_sales",
+                          "web_cumulative", "store_cumulative"]])
+    return topk(proj, [(col("item_sk"), True), (col("d_date"), True)], 100)
+
+
+def q75(cat, s):
+    from ..exprs import Coalesce
+
+    it = P.Filter(cat.scan("item", ["i_item_sk", "i_brand_id", "i_class_id",
+                                    "i_category_id", "i_manufact_id", "i_category"]),
+                  col("i_category") == lit("Books"))
+    dd = cat.scan("date_dim", ["d_date_sk", "d_year"])
+
+    def chan(fact, pre, rets, rpre, k2, rk2):
+        fs = cat.scan(fact, [f"{pre}_item_sk", f"{pre}_sold_date_sk", k2,
+                             f"{pre}_quantity", f"{pre}_ext_sales_price"])
+        amt = f"{rpre}_return_{'amount' if rpre == 'cr' else 'amt'}"
+        rt = cat.scan(rets, [rk2, f"{rpre}_item_sk", f"{rpre}_return_quantity", amt])
+        j = shj(fs, rt, [k2, f"{pre}_item_sk"], [rk2, f"{rpre}_item_sk"], how="left")
+        j = bhj(j, it, [f"{pre}_item_sk"], ["i_item_sk"])
+        j = bhj(j, dd, [f"{pre}_sold_date_sk"], ["d_date_sk"])
+        cnt = col(f"{pre}_quantity") - Coalesce([col(f"{rpre}_return_quantity"), lit(0)])
+        amt_e = col(f"{pre}_ext_sales_price") - Coalesce([col(amt), lit(0.0)])
+        return P.Project(j, [_a(col("d_year"), "d_year"),
+                             _a(col("i_brand_id"), "i_brand_id"),
+                             _a(col("i_class_id"), "i_class_id"),
+                             _a(col("i_category_id"), "i_category_id"),
+                             _a(col("i_manufact_id"), "i_manufact_id"),
+                             _a(cnt.cast(dtypes.int64), "sales_cnt"),
+                             _a(amt_e, "sales_amt")])
+
+    u = P.Union([
+        chan("catalog_sales", "cs", "catalog_returns", "cr", "cs_order_number", "cr_order_number"),
+        chan("store_sales", "ss", "store_returns", "sr", "ss_ticket_number", "sr_ticket_number"),
+        chan("web_sales", "ws", "web_returns", "wr", "ws_order_number", "wr_order_number")])
+    dedup = agg2(u, ["d_year", "i_brand_id", "i_class_id", "i_category_id",
+                     "i_manufact_id", "sales_cnt", "sales_amt"], [])
+    allsales = agg2(dedup, ["d_year", "i_brand_id", "i_class_id", "i_category_id",
+                            "i_manufact_id"],
+                    [AggFunc("sum", col("sales_cnt"), name="sales_cnt"),
+                     AggFunc("sum", col("sales_amt"), name="sales_amt")])
+    ab = s.execute(allsales)
+    curr = P.Filter(P.MemoryScan(ab), col("d_year") == 2002)
+    prev = P.Project(P.Filter(P.MemoryScan(ab), col("d_year") == 2001),
+                     [_a(col("i_brand_id"), "pb"), _a(col("i_class_id"), "pc"),
+                      _a(col("i_category_id"), "pg"), _a(col("i_manufact_id"), "pm"),
+                      _a(col("d_year"), "prev_year"),
+                      _a(col("sales_cnt"), "prev_yr_cnt"),
+                      _a(col("sales_amt"), "prev_amt")])
+    j = shj(curr, prev, ["i_brand_id", "i_class_id", "i_category_id", "i_manufact_id"],
+            ["pb", "pc", "pg", "pm"])
+    f = P.Filter(j, col("sales_cnt").cast(dtypes.float64)
+                 / col("prev_yr_cnt").cast(dtypes.float64) < 0.9)
+    proj = P.Project(f, [_a(col("prev_year"), "prev_year"), _a(col("d_year"), "year"),
+                         _a(col("i_brand_id"), "i_brand_id"),
+                         _a(col("i_class_id"), "i_class_id"),
+                         _a(col("i_category_id"), "i_category_id"),
+                         _a(col("i_manufact_id"), "i_manufact_id"),
+                         _a(col("prev_yr_cnt"), "prev_yr_cnt"),
+                         _a(col("sales_cnt"), "curr_yr_cnt"),
+                         _a(col("sales_cnt") - col("prev_yr_cnt"), "sales_cnt_diff"),
+                         _a(col("sales_amt") - col("prev_amt"), "sales_amt_diff")])
+    return topk(proj, [(col("sales_cnt_diff"), True), (col("sales_amt_diff"), True)], 100)
+
+
+def q78(cat, s):
+    from ..exprs import Coalesce, IsNull
+
+    dd = cat.scan("date_dim", ["d_date_sk", "d_year"])
+
+    def cte(fact, pre, rets, rpre, cust_fk, k2, rk2):
+        fs = cat.scan(fact, [f"{pre}_item_sk", cust_fk, f"{pre}_sold_date_sk", k2,
+                             f"{pre}_quantity", f"{pre}_wholesale_cost",
+                             f"{pre}_sales_price"])
+        rt = cat.scan(rets, [rk2, f"{rpre}_item_sk"])
+        j = shj(fs, rt, [k2, f"{pre}_item_sk"], [rk2, f"{rpre}_item_sk"], how="left")
+        j = P.Filter(j, IsNull(col(rk2)))
+        j = bhj(j, dd, [f"{pre}_sold_date_sk"], ["d_date_sk"])
+        return agg2(j, ["sold_year", "item", "customer"],
+                    [AggFunc("sum", col(f"{pre}_quantity"), name=f"{pre}_qty"),
+                     AggFunc("sum", col(f"{pre}_wholesale_cost"), name=f"{pre}_wc"),
+                     AggFunc("sum", col(f"{pre}_sales_price"), name=f"{pre}_sp")],
+                    key_exprs=[col("d_year"), col(f"{pre}_item_sk"), col(cust_fk)])
+
+    ssc = cte("store_sales", "ss", "store_returns", "sr", "ss_customer_sk",
+              "ss_ticket_number", "sr_ticket_number")
+    wsc = P.Project(cte("web_sales", "ws", "web_returns", "wr", "ws_bill_customer_sk",
+                        "ws_order_number", "wr_order_number"),
+                    [_a(col("sold_year"), "wy"), _a(col("item"), "wi"),
+                     _a(col("customer"), "wcst"), _a(col("ws_qty"), "ws_qty"),
+                     _a(col("ws_wc"), "ws_wc"), _a(col("ws_sp"), "ws_sp")])
+    csc = P.Project(cte("catalog_sales", "cs", "catalog_returns", "cr",
+                        "cs_bill_customer_sk", "cs_order_number", "cr_order_number"),
+                    [_a(col("sold_year"), "cy"), _a(col("item"), "ci"),
+                     _a(col("customer"), "ccst"), _a(col("cs_qty"), "cs_qty"),
+                     _a(col("cs_wc"), "cs_wc"), _a(col("cs_sp"), "cs_sp")])
+    j = shj(P.Filter(ssc, col("sold_year") == 2000), wsc,
+            ["sold_year", "item", "customer"], ["wy", "wi", "wcst"], how="left")
+    j = P.HashJoin(j, P.Exchange(csc, "hash", [col("cy"), col("ci"), col("ccst")]),
+                   [col("sold_year"), col("item"), col("customer")],
+                   [col("cy"), col("ci"), col("ccst")], how="left", build_side="right")
+    f = P.Filter(j, (Coalesce([col("ws_qty"), lit(0)]) > 0)
+                 & (Coalesce([col("cs_qty"), lit(0)]) > 0))
+    other_qty = Coalesce([col("ws_qty"), lit(0)]) + Coalesce([col("cs_qty"), lit(0)])
+    denom = Coalesce([col("ws_qty") + col("cs_qty"), lit(1)])
+    ratio_raw = col("ss_qty").cast(dtypes.float64) / denom.cast(dtypes.float64)
+    r2 = ((ratio_raw * lit(100.0) + lit(0.5)).cast(dtypes.int64)).cast(dtypes.float64) / lit(100.0)
+    proj = P.Project(f, [_a(r2, "ratio"), _a(col("ss_qty"), "store_qty"),
+                         _a(col("ss_wc"), "store_wholesale_cost"),
+                         _a(col("ss_sp"), "store_sales_price"),
+                         _a(other_qty, "other_chan_qty"),
+                         _a(Coalesce([col("ws_wc"), lit(0.0)]) + Coalesce([col("cs_wc"), lit(0.0)]),
+                            "other_chan_wholesale_cost"),
+                         _a(Coalesce([col("ws_sp"), lit(0.0)]) + Coalesce([col("cs_sp"), lit(0.0)]),
+                            "other_chan_sales_price")])
+    return topk(proj, [(col("ratio"), True), (col("store_qty"), False),
+                       (col("store_wholesale_cost"), False),
+                       (col("store_sales_price"), False),
+                       (col("other_chan_qty"), True),
+                       (col("other_chan_wholesale_cost"), True),
+                       (col("other_chan_sales_price"), True)], 100)
+
+
+QUERIES.update({"q51": q51, "q75": q75, "q78": q78})
