@@ -2743,7 +2743,10 @@ def q51(root, sf):
         a = j.groupby([f"{pre}_item_sk", "d_i"], dropna=False)[f"{pre}_sales_price"] \
              .sum(min_count=1).reset_index(name="s")
         a = a.sort_values([f"{pre}_item_sk", "d_i"])
+        # SQL running sum ignores NULL inputs; pandas cumsum leaves NaN at
+        # the NaN row, so carry the prior cumulative forward
         a["cume_sales"] = a.groupby(f"{pre}_item_sk").s.cumsum()
+        a["cume_sales"] = a.groupby(f"{pre}_item_sk").cume_sales.ffill()
         return a.rename(columns={f"{pre}_item_sk": "item_sk", "d_i": "d_date"})
 
     web = v1("web_sales", "ws")[["item_sk", "d_date", "cume_sales"]] \
