@@ -222,7 +222,11 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
         put("s_store_name", [["ought", "able", "pri", "ese", "anti", "cally"][int(s) % 6] for s in sks])
         # dsdgen: SF<=1000 stores are mostly TN
         put("s_state", ["TN" if int(s) % 4 != 0 else _STATES[int(s) % len(_STATES)] for s in sks])
-        put("s_county", ["Williamson County"] * n)
+        # half the stores share county names with the address universe so
+        # county-equi joins (q54) are non-trivial
+        put("s_county", ["Williamson County" if int(s2) % 2 == 0
+                         else f"{_CITIES[int(s2) % len(_CITIES)]} County"
+                         for s2 in sks])
         put("s_zip", [f"{int(v):05d}" for v in rng.integers(30000, 40000, n)])
         put("s_city", [_CITIES[int(s) % len(_CITIES)] for s in sks])
         put("s_number_employees", rng.integers(200, 301, n).astype(np.int32))
@@ -491,7 +495,7 @@ def _nparts_for(table: str, sf: float) -> int:
     return 1
 
 
-DATAGEN_VERSION = 7
+DATAGEN_VERSION = 8
 
 
 def dataset_root(root: str, sf: float) -> str:

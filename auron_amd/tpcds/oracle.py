@@ -2873,3 +2873,87 @@ def q78(root, sf):
 
 
 ORACLES.update({"q51": q51, "q75": q75, "q78": q78})
+
+
+# ------------------------------- batch 13 oracles
+def q54(root, sf):
+    import numpy as np
+    import pandas as pd
+
+    dd_all = _read(root, sf, "date_dim", ["d_date_sk", "d_moy", "d_year", "d_month_seq"])
+    ms0 = dd_all[(dd_all.d_year == 1998) & (dd_all.d_moy == 12)].d_month_seq.iloc[0]
+    it = _read(root, sf, "item", ["i_item_sk", "i_category", "i_class"])
+    it = it[(it.i_category == "Women") & (it.i_class == "class1")]
+    dd = dd_all[(dd_all.d_moy == 12) & (dd_all.d_year == 1998)]
+
+    def chan(fact, pre, fk):
+        fs = _read(root, sf, fact, [f"{pre}_sold_date_sk", fk, f"{pre}_item_sk"])
+        return fs.rename(columns={f"{pre}_sold_date_sk": "sold_date_sk",
+                                  fk: "customer_sk", f"{pre}_item_sk": "item_sk"})
+
+    u = pd.concat([chan("catalog_sales", "cs", "cs_bill_customer_sk"),
+                   chan("web_sales", "ws", "ws_bill_customer_sk")], ignore_index=True)
+    j = _merge(u, dd, "sold_date_sk", "d_date_sk")
+    j = _merge(j, it, "item_sk", "i_item_sk")
+    cust = _read(root, sf, "customer", ["c_customer_sk", "c_current_addr_sk"])
+    j = _merge(j, cust, "customer_sk", "c_customer_sk")
+    mc = j[["c_customer_sk", "c_current_addr_sk"]].drop_duplicates()
+    ca = _read(root, sf, "customer_address", ["ca_address_sk", "ca_county", "ca_state"])
+    st = _read(root, sf, "store", ["s_county", "s_state"]).drop_duplicates()
+    mc = _merge(mc, ca, "c_current_addr_sk", "ca_address_sk")
+    import numpy as np
+
+    pairs = set(zip(st.s_county, st.s_state))
+    mask = np.array([(c, s_) in pairs for c, s_ in zip(mc.ca_county, mc.ca_state)],
+                    dtype=bool)
+    mc = mc[mask] if len(mc) else mc
+    ss = _read(root, sf, "store_sales", ["ss_sold_date_sk", "ss_customer_sk",
+                                         "ss_ext_sales_price"])
+    dd2 = dd_all[dd_all.d_month_seq.between(ms0 + 1, ms0 + 3)]
+    sj = _merge(ss, dd2, "ss_sold_date_sk", "d_date_sk")
+    sj = _merge(sj, mc, "ss_customer_sk", "c_customer_sk")
+    rev = sj.groupby("c_customer_sk", dropna=False).ss_ext_sales_price \
+            .sum(min_count=1).reset_index(name="revenue")
+    seg = (rev.revenue / 50.0).astype(np.int64)
+    g = pd.Series(seg).value_counts().reset_index()
+    g.columns = ["segment", "num_customers"]
+    g["segment_base"] = g.segment * 50
+    g = g.sort_values(["segment", "num_customers"]).head(100)
+    return g.reset_index(drop=True)
+
+
+def q67(root, sf):
+    import numpy as np
+    import pandas as pd
+
+    ss = _read(root, sf, "store_sales", ["ss_sold_date_sk", "ss_item_sk", "ss_store_sk",
+                                         "ss_sales_price", "ss_quantity"])
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_month_seq", "d_year", "d_qoy", "d_moy"])
+    dd = dd[dd.d_month_seq.between(1200, 1211)]
+    st = _read(root, sf, "store", ["s_store_sk", "s_store_id"])
+    it = _read(root, sf, "item", ["i_item_sk", "i_category", "i_class", "i_brand",
+                                  "i_product_name"])
+    j = _merge(ss, dd, "ss_sold_date_sk", "d_date_sk")
+    j = _merge(j, st, "ss_store_sk", "s_store_sk")
+    j = _merge(j, it, "ss_item_sk", "i_item_sk")
+    j["v"] = (j.ss_sales_price * j.ss_quantity).fillna(0.0)
+    keys = ["i_category", "i_class", "i_brand", "i_product_name", "d_year",
+            "d_qoy", "d_moy", "s_store_id"]
+    frames = []
+    for depth in range(len(keys), -1, -1):
+        t = j.copy()
+        for i, k in enumerate(keys):
+            if i >= depth:
+                t[k] = None
+        g = t.groupby(keys, dropna=False).v.sum(min_count=1).reset_index(name="sumsales")
+        frames.append(g)
+    out = pd.concat(frames, ignore_index=True)
+    out["_key"] = np.trunc(out.sumsales * 100.0 + 0.5)
+    out["rk"] = out.groupby("i_category", dropna=False)._key \
+        .rank(method="min", ascending=False).astype(int)
+    f = out[out.rk <= 100]
+    f = f.sort_values(keys + ["sumsales", "rk"], na_position="first").head(100)
+    return f[keys + ["sumsales", "rk"]].reset_index(drop=True)
+
+
+ORACLES.update({"q54": q54, "q67": q67})

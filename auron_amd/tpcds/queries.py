@@ -3385,3 +3385,96 @@ def q78(cat, s):
 
 
 QUERIES.update({"q51": q51, "q75": q75, "q78": q78})
+
+
+# ------------------------------- batch 13
+def q54(cat, s):
+    month_seq0 = scalar(s, P.Limit(P.Exchange(P.Project(
+        P.Filter(cat.scan("date_dim", ["d_month_seq", "d_year", "d_moy"]),
+                 (col("d_year") == 1998) & (col("d_moy") == 12)),
+        [_a(col("d_month_seq"), "ms")]), "single"), 1))
+    it = P.Filter(cat.scan("item", ["i_item_sk", "i_category", "i_class"]),
+                  (col("i_category") == lit("Women")) & (col("i_class") == lit("class1")))
+    # adapted literal: 'maternity' class -> synthetic class list
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_moy", "d_year"]),
+                  (col("d_moy") == 12) & (col("d_year") == 1998))
+
+    def chan(fact, pre, fk):
+        fs = cat.scan(fact, [f"{pre}_sold_date_sk", fk, f"{pre}_item_sk"])
+        return P.Project(fs, [_a(col(f"{pre}_sold_date_sk"), "sold_date_sk"),
+                              _a(col(fk), "customer_sk"),
+                              _a(col(f"{pre}_item_sk"), "item_sk")])
+
+    u = P.Union([chan("catalog_sales", "cs", "cs_bill_customer_sk"),
+                 chan("web_sales", "ws", "ws_bill_customer_sk")])
+    j = bhj(u, dd, ["sold_date_sk"], ["d_date_sk"])
+    j = bhj(j, it, ["item_sk"], ["i_item_sk"])
+    cust = cat.scan("customer", ["c_customer_sk", "c_current_addr_sk"])
+    j = P.HashJoin(P.Exchange(j, "hash", [col("customer_sk")]),
+                   P.Exchange(cust, "hash", [col("c_customer_sk")]),
+                   [col("customer_sk")], [col("c_customer_sk")],
+                   how="inner", build_side="right")
+    my_customers = agg2(j, ["c_customer_sk", "c_current_addr_sk"], [])
+
+    ca = cat.scan("customer_address", ["ca_address_sk", "ca_county", "ca_state"])
+    st = cat.scan("store", ["s_county", "s_state"])
+    mc = bhj(my_customers, ca, ["c_current_addr_sk"], ["ca_address_sk"])
+    mc = P.HashJoin(mc, P.Broadcast(P.HashAgg(
+        P.Broadcast(st), [_a(col("s_county"), "s_county"), _a(col("s_state"), "s_state")],
+        [], mode="complete")),
+        [col("ca_county"), col("ca_state")], [col("s_county"), col("s_state")],
+        how="semi", build_side="right")
+    ss = cat.scan("store_sales", ["ss_sold_date_sk", "ss_customer_sk", "ss_ext_sales_price"])
+    dd2 = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_month_seq"]),
+                   col("d_month_seq").between((month_seq0 or 0) + 1, (month_seq0 or 0) + 3))
+    sj = bhj(ss, dd2, ["ss_sold_date_sk"], ["d_date_sk"])
+    sj = P.HashJoin(P.Exchange(sj, "hash", [col("ss_customer_sk")]),
+                    P.Exchange(mc, "hash", [col("c_customer_sk")]),
+                    [col("ss_customer_sk")], [col("c_customer_sk")],
+                    how="inner", build_side="right")
+    rev = agg2(sj, ["c_customer_sk2"],
+               [AggFunc("sum", col("ss_ext_sales_price"), name="revenue")],
+               key_exprs=[col("c_customer_sk")])
+    seg = P.Project(rev, [_a((col("revenue") / lit(50.0)).cast(dtypes.int32), "segment")])
+    a = agg2(seg, ["segment"], [AggFunc("count_star", None, name="num_customers")])
+    proj = P.Project(a, [_a(col("segment"), "segment"),
+                         _a(col("num_customers"), "num_customers"),
+                         _a(col("segment") * lit(50), "segment_base")])
+    return topk(proj, [(col("segment"), True), (col("num_customers"), True)], 100)
+
+
+def q67(cat, s):
+    from ..exprs import CaseWhen, Coalesce, Literal, WindowFunc
+
+    ss = cat.scan("store_sales", ["ss_sold_date_sk", "ss_item_sk", "ss_store_sk",
+                                  "ss_sales_price", "ss_quantity"])
+    dd = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_month_seq", "d_year",
+                                        "d_qoy", "d_moy"]),
+                  col("d_month_seq").between(1200, 1211))
+    st = cat.scan("store", ["s_store_sk", "s_store_id"])
+    it = cat.scan("item", ["i_item_sk", "i_category", "i_class", "i_brand",
+                           "i_product_name"])
+    j = bhj(ss, dd, ["ss_sold_date_sk"], ["d_date_sk"])
+    j = bhj(j, st, ["ss_store_sk"], ["s_store_sk"])
+    j = bhj(j, it, ["ss_item_sk"], ["i_item_sk"])
+    val = Coalesce([col("ss_sales_price") * col("ss_quantity").cast(dtypes.float64),
+                    lit(0.0)])
+    keys = ["i_category", "i_class", "i_brand", "i_product_name", "d_year",
+            "d_qoy", "d_moy", "s_store_id"]
+    kdts = [dtypes.string, dtypes.string, dtypes.string, dtypes.string,
+            dtypes.int32, dtypes.int32, dtypes.int32, dtypes.string]
+    pre = P.Project(j, [_a(col(k), k) for k in keys] + [_a(val, "v")])
+    ex = rollup_expand(pre, keys, kdts, ["v"])
+    a = agg2(ex, keys + ["_lochier"], [AggFunc("sum", col("v"), name="sumsales")])
+    rank_key = ((col("sumsales") * lit(100.0)) + lit(0.5)).cast(dtypes.int64)
+    w = P.Window(P.Exchange(a, "hash", [col("i_category")]),
+                 [col("i_category")], [(rank_key, False)],
+                 [_a(WindowFunc("rank"), "rk")])
+    f = P.Filter(w, col("rk") <= 100)
+    proj = P.Project(f, [_a(col(k), k) for k in keys]
+                     + [_a(col("sumsales"), "sumsales"), _a(col("rk"), "rk")])
+    return topk(proj, [(col(k), True) for k in keys]
+                + [(col("sumsales"), True), (col("rk"), True)], 100)
+
+
+QUERIES.update({"q54": q54, "q67": q67})
