@@ -2957,3 +2957,68 @@ def q67(root, sf):
 
 
 ORACLES.update({"q54": q54, "q67": q67})
+
+
+def q14(root, sf):
+    import pandas as pd
+
+    years = [1999, 2000, 2001]
+    it = _read(root, sf, "item", ["i_item_sk", "i_brand_id", "i_class_id", "i_category_id"])
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year", "d_moy"])
+    dd_y = dd[dd.d_year.isin(years)]
+
+    def triples(fact, pre):
+        fs = _read(root, sf, fact, [f"{pre}_item_sk", f"{pre}_sold_date_sk"])
+        j = _merge(fs, dd_y, f"{pre}_sold_date_sk", "d_date_sk")
+        j = _merge(j, it, f"{pre}_item_sk", "i_item_sk")
+        return set(zip(j.i_brand_id, j.i_class_id, j.i_category_id))
+
+    inter = triples("store_sales", "ss") & triples("catalog_sales", "cs") \
+        & triples("web_sales", "ws")
+    ci = set(it[[t in inter for t in
+                 zip(it.i_brand_id, it.i_class_id, it.i_category_id)]].i_item_sk)
+
+    vals = []
+    for fact, pre in (("store_sales", "ss"), ("catalog_sales", "cs"), ("web_sales", "ws")):
+        fs = _read(root, sf, fact, [f"{pre}_sold_date_sk", f"{pre}_quantity",
+                                    f"{pre}_list_price"])
+        j = _merge(fs, dd_y, f"{pre}_sold_date_sk", "d_date_sk")
+        vals.append(j[f"{pre}_quantity"] * j[f"{pre}_list_price"])
+    average_sales = pd.concat(vals, ignore_index=True).mean()
+
+    dd_m = dd[(dd.d_year == 2001) & (dd.d_moy == 11)]
+    frames = []
+    for fact, pre, tag in (("store_sales", "ss", "store"),
+                           ("catalog_sales", "cs", "catalog"),
+                           ("web_sales", "ws", "web")):
+        fs = _read(root, sf, fact, [f"{pre}_item_sk", f"{pre}_sold_date_sk",
+                                    f"{pre}_quantity", f"{pre}_list_price"])
+        j = _merge(fs, dd_m, f"{pre}_sold_date_sk", "d_date_sk")
+        j = j[j[f"{pre}_item_sk"].isin(ci)]
+        j = _merge(j, it, f"{pre}_item_sk", "i_item_sk")
+        j["v"] = j[f"{pre}_quantity"] * j[f"{pre}_list_price"]
+        g = j.groupby(["i_brand_id", "i_class_id", "i_category_id"], dropna=False).agg(
+            sales=("v", lambda x: x.sum(min_count=1)),
+            number_sales=("v", "size")).reset_index()
+        g = g[g.sales > average_sales]
+        g["channel"] = tag
+        frames.append(g)
+    y = pd.concat(frames, ignore_index=True)
+    keys = ["channel", "i_brand_id", "i_class_id", "i_category_id"]
+    out_frames = []
+    for depth in range(len(keys), -1, -1):
+        t = y.copy()
+        for i, k in enumerate(keys):
+            if i >= depth:
+                t[k] = None
+        g = t.groupby(keys, dropna=False).agg(
+            sum_sales=("sales", lambda x: x.sum(min_count=1)),
+            sum_number_sales=("number_sales", lambda x: x.sum(min_count=1))).reset_index()
+        out_frames.append(g)
+    out = pd.concat(out_frames, ignore_index=True)
+    out["sum_number_sales"] = out.sum_number_sales.astype("Int64")
+    out = out.sort_values(keys, na_position="first").head(100)
+    return out[keys + ["sum_sales", "sum_number_sales"]].reset_index(drop=True)
+
+
+ORACLES.update({"q14": q14})

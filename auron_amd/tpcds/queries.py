@@ -3478,3 +3478,92 @@ def q67(cat, s):
 
 
 QUERIES.update({"q54": q54, "q67": q67})
+
+
+# ------------------------------- batch 14
+def q14(cat, s):
+    from ..exprs import Literal
+
+    years = [1999, 2000, 2001]
+    it_full = cat.scan("item", ["i_item_sk", "i_brand_id", "i_class_id", "i_category_id"])
+    dd_y = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_year"]),
+                    col("d_year").isin(years))
+
+    def triples(fact, pre):
+        fs = cat.scan(fact, [f"{pre}_item_sk", f"{pre}_sold_date_sk"])
+        j = bhj(fs, dd_y, [f"{pre}_sold_date_sk"], ["d_date_sk"])
+        j = bhj(j, it_full, [f"{pre}_item_sk"], ["i_item_sk"])
+        return agg2(j, ["b", "c", "g"], [],
+                    key_exprs=[col("i_brand_id"), col("i_class_id"), col("i_category_id")])
+
+    tss = triples("store_sales", "ss")
+    tcs = P.Project(triples("catalog_sales", "cs"),
+                    [_a(col("b"), "cb"), _a(col("c"), "cc"), _a(col("g"), "cg")])
+    tws = P.Project(triples("web_sales", "ws"),
+                    [_a(col("b"), "wb"), _a(col("c"), "wc"), _a(col("g"), "wg")])
+    inter = P.HashJoin(P.Exchange(tss, "hash", [col("b"), col("c"), col("g")]),
+                       P.Exchange(tcs, "hash", [col("cb"), col("cc"), col("cg")]),
+                       [col("b"), col("c"), col("g")], [col("cb"), col("cc"), col("cg")],
+                       how="semi", build_side="right")
+    inter = P.HashJoin(inter, P.Exchange(tws, "hash", [col("wb"), col("wc"), col("wg")]),
+                       [col("b"), col("c"), col("g")], [col("wb"), col("wc"), col("wg")],
+                       how="semi", build_side="right")
+    cross_items = P.HashJoin(it_full, P.Broadcast(inter),
+                             [col("i_brand_id"), col("i_class_id"), col("i_category_id")],
+                             [col("b"), col("c"), col("g")], how="semi",
+                             build_side="right")
+    ci_b = s.execute(P.Project(cross_items, [_a(col("i_item_sk"), "ci_sk")]))
+
+    def avg_chan(fact, pre):
+        fs = cat.scan(fact, [f"{pre}_sold_date_sk", f"{pre}_quantity", f"{pre}_list_price"])
+        j = bhj(fs, dd_y, [f"{pre}_sold_date_sk"], ["d_date_sk"])
+        return P.Project(j, [_a(col(f"{pre}_quantity").cast(dtypes.float64)
+                                * col(f"{pre}_list_price"), "v")])
+
+    u = P.Union([avg_chan("store_sales", "ss"), avg_chan("catalog_sales", "cs"),
+                 avg_chan("web_sales", "ws")])
+    average_sales = scalar(s, _global_agg(u, [AggFunc("avg", col("v"), name="a")])) or 0.0
+
+    dd_m = P.Filter(cat.scan("date_dim", ["d_date_sk", "d_year", "d_moy"]),
+                    (col("d_year") == 2001) & (col("d_moy") == 11))
+
+    def chan(fact, pre, tag):
+        fs = cat.scan(fact, [f"{pre}_item_sk", f"{pre}_sold_date_sk",
+                             f"{pre}_quantity", f"{pre}_list_price"])
+        j = bhj(fs, dd_m, [f"{pre}_sold_date_sk"], ["d_date_sk"])
+        j = P.HashJoin(P.Exchange(j, "hash", [col(f"{pre}_item_sk")]),
+                       P.Exchange(P.MemoryScan(ci_b), "hash", [col("ci_sk")]),
+                       [col(f"{pre}_item_sk")], [col("ci_sk")], how="semi",
+                       build_side="right")
+        j = bhj(j, it_full, [f"{pre}_item_sk"], ["i_item_sk"])
+        pre_rows = P.Project(j, [_a(col("i_brand_id"), "i_brand_id"),
+                                 _a(col("i_class_id"), "i_class_id"),
+                                 _a(col("i_category_id"), "i_category_id"),
+                                 _a(col(f"{pre}_quantity").cast(dtypes.float64)
+                                    * col(f"{pre}_list_price"), "v")])
+        a = agg2(pre_rows, ["i_brand_id", "i_class_id", "i_category_id"],
+                 [AggFunc("sum", col("v"), name="sales"),
+                  AggFunc("count_star", None, name="number_sales")])
+        h = P.Filter(a, col("sales") > lit(average_sales))
+        return P.Project(h, [_a(lit(tag), "channel"), _a(col("i_brand_id"), "i_brand_id"),
+                             _a(col("i_class_id"), "i_class_id"),
+                             _a(col("i_category_id"), "i_category_id"),
+                             _a(col("sales"), "sales"),
+                             _a(col("number_sales"), "number_sales")])
+
+    y = P.Union([chan("store_sales", "ss", "store"),
+                 chan("catalog_sales", "cs", "catalog"),
+                 chan("web_sales", "ws", "web")])
+    keys = ["channel", "i_brand_id", "i_class_id", "i_category_id"]
+    kdts = [dtypes.string, dtypes.int32, dtypes.int32, dtypes.int32]
+    ex = rollup_expand(y, keys, kdts, ["sales", "number_sales"])
+    a = agg2(ex, keys + ["_lochier"],
+             [AggFunc("sum", col("sales"), name="sum_sales"),
+              AggFunc("sum", col("number_sales"), name="sum_number_sales")])
+    proj = P.Project(a, [_a(col(k), k) for k in keys]
+                     + [_a(col("sum_sales"), "sum_sales"),
+                        _a(col("sum_number_sales"), "sum_number_sales")])
+    return topk(proj, [(col(k), True) for k in keys], 100)
+
+
+QUERIES.update({"q14": q14})
