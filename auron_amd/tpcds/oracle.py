@@ -3022,3 +3022,106 @@ def q14(root, sf):
 
 
 ORACLES.update({"q14": q14})
+
+
+def q64(root, sf):
+    import pandas as pd
+
+    cs = _read(root, sf, "catalog_sales", ["cs_item_sk", "cs_order_number",
+                                           "cs_ext_list_price"])
+    cr = _read(root, sf, "catalog_returns", ["cr_item_sk", "cr_order_number",
+                                             "cr_refunded_cash", "cr_reversed_charge",
+                                             "cr_store_credit"])
+    jcr = cs.dropna(subset=["cs_item_sk", "cs_order_number"]).merge(
+        cr.dropna(subset=["cr_item_sk", "cr_order_number"]),
+        left_on=["cs_item_sk", "cs_order_number"],
+        right_on=["cr_item_sk", "cr_order_number"])
+    jcr["refund_v"] = jcr.cr_refunded_cash + jcr.cr_reversed_charge + jcr.cr_store_credit
+    ui = jcr.groupby("cs_item_sk", dropna=False).agg(
+        sale=("cs_ext_list_price", lambda x: x.sum(min_count=1)),
+        refund=("refund_v", lambda x: x.sum(min_count=1))).reset_index()
+    ui = ui[ui.sale > 2.0 * ui.refund]
+
+    ss = _read(root, sf, "store_sales")
+    sr = _read(root, sf, "store_returns", ["sr_item_sk", "sr_ticket_number"])
+    j = ss.dropna(subset=["ss_item_sk", "ss_ticket_number"]).merge(
+        sr.dropna(subset=["sr_item_sk", "sr_ticket_number"]),
+        left_on=["ss_item_sk", "ss_ticket_number"],
+        right_on=["sr_item_sk", "sr_ticket_number"])
+    j = j[j.ss_item_sk.isin(set(ui.cs_item_sk))]
+    st = _read(root, sf, "store", ["s_store_sk", "s_store_name", "s_zip"])
+    dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year"])
+    cust = _read(root, sf, "customer", ["c_customer_sk", "c_current_cdemo_sk",
+                                        "c_current_hdemo_sk", "c_current_addr_sk",
+                                        "c_first_sales_date_sk", "c_first_shipto_date_sk"])
+    cd = _read(root, sf, "customer_demographics", ["cd_demo_sk", "cd_marital_status"])
+    hd = _read(root, sf, "household_demographics", ["hd_demo_sk", "hd_income_band_sk"])
+    ad = _read(root, sf, "customer_address", ["ca_address_sk", "ca_street_number",
+                                              "ca_street_name", "ca_city", "ca_zip"])
+    ib = _read(root, sf, "income_band", ["ib_income_band_sk"])
+    it = _read(root, sf, "item", ["i_item_sk", "i_product_name", "i_color",
+                                  "i_current_price"])
+    it = it[it.i_color.isin(["purple", "burlywood", "indian", "spring", "floral",
+                             "medium"])
+            & it.i_current_price.between(64, 74) & it.i_current_price.between(65, 79)]
+    j = _merge(j, st, "ss_store_sk", "s_store_sk")
+    j = _merge(j, dd.rename(columns={"d_date_sk": "d1sk", "d_year": "syear"}),
+               "ss_sold_date_sk", "d1sk")
+    j = _merge(j, cust, "ss_customer_sk", "c_customer_sk")
+    j = _merge(j, cd.rename(columns={"cd_demo_sk": "cd1sk", "cd_marital_status": "cd1_ms"}),
+               "ss_cdemo_sk", "cd1sk")
+    j = _merge(j, hd.rename(columns={"hd_demo_sk": "hd1sk", "hd_income_band_sk": "ib1fk"}),
+               "ss_hdemo_sk", "hd1sk")
+    j = _merge(j, ad.rename(columns={"ca_address_sk": "ad1sk",
+                                     "ca_street_number": "b_street_number",
+                                     "ca_street_name": "b_streen_name",
+                                     "ca_city": "b_city", "ca_zip": "b_zip"}),
+               "ss_addr_sk", "ad1sk")
+    j = _merge(j, it, "ss_item_sk", "i_item_sk")
+    j = _merge(j, cd.rename(columns={"cd_demo_sk": "cd2sk", "cd_marital_status": "cd2_ms"}),
+               "c_current_cdemo_sk", "cd2sk")
+    j = _merge(j, hd.rename(columns={"hd_demo_sk": "hd2sk", "hd_income_band_sk": "ib2fk"}),
+               "c_current_hdemo_sk", "hd2sk")
+    j = _merge(j, ad.rename(columns={"ca_address_sk": "ad2sk",
+                                     "ca_street_number": "c_street_number",
+                                     "ca_street_name": "c_street_name",
+                                     "ca_city": "c_city", "ca_zip": "c_zip"}),
+               "c_current_addr_sk", "ad2sk")
+    j = _merge(j, dd.rename(columns={"d_date_sk": "d2sk", "d_year": "fsyear"}),
+               "c_first_sales_date_sk", "d2sk")
+    j = _merge(j, dd.rename(columns={"d_date_sk": "d3sk", "d_year": "s2year"}),
+               "c_first_shipto_date_sk", "d3sk")
+    pr = _read(root, sf, "promotion", ["p_promo_sk"])
+    j = _merge(j, pr, "ss_promo_sk", "p_promo_sk")
+    j = _merge(j, ib.rename(columns={"ib_income_band_sk": "ib1sk"}), "ib1fk", "ib1sk")
+    j = _merge(j, ib.rename(columns={"ib_income_band_sk": "ib2sk"}), "ib2fk", "ib2sk")
+    j = j[(j.cd1_ms != j.cd2_ms) & j.cd1_ms.notna() & j.cd2_ms.notna()]
+    keys = ["i_product_name", "ss_item_sk", "s_store_name", "s_zip",
+            "b_street_number", "b_streen_name", "b_city", "b_zip",
+            "c_street_number", "c_street_name", "c_city", "c_zip",
+            "syear", "fsyear", "s2year"]
+    g = j.groupby(keys, dropna=False).agg(
+        cnt=("ss_item_sk", "size"),
+        s1=("ss_wholesale_cost", lambda x: x.sum(min_count=1)),
+        s2=("ss_list_price", lambda x: x.sum(min_count=1)),
+        s3=("ss_coupon_amt", lambda x: x.sum(min_count=1))).reset_index()
+    cs1 = g[g.syear == 1999]
+    cs2 = g[g.syear == 2000]
+    m = cs1.merge(cs2, on=["ss_item_sk", "s_store_name", "s_zip"],
+                  suffixes=("", "_2"))
+    m = m[m.cnt_2 <= m.cnt]
+    out = pd.DataFrame({
+        "i_product_name": m.i_product_name, "s_store_name": m.s_store_name,
+        "s_zip": m.s_zip, "b_street_number": m.b_street_number,
+        "b_streen_name": m.b_streen_name, "b_city": m.b_city, "b_zip": m.b_zip,
+        "c_street_number": m.c_street_number, "c_street_name": m.c_street_name,
+        "c_city": m.c_city, "c_zip": m.c_zip, "syear": m.syear,
+        "cnt": m.cnt.astype("Int64"), "s1": m.s1, "s2": m.s2, "s3": m.s3,
+        "s1_2": m.s1_2, "s2_2": m.s2_2, "s3_2": m.s3_2, "syear2": m.syear_2,
+        "cnt2": m.cnt_2.astype("Int64")})
+    out = out.sort_values(["i_product_name", "s_store_name", "cnt2", "s1", "s1_2"],
+                          na_position="first")
+    return out.reset_index(drop=True)
+
+
+ORACLES.update({"q64": q64})

@@ -3567,3 +3567,118 @@ def q14(cat, s):
 
 
 QUERIES.update({"q14": q14})
+
+
+def q64(cat, s):
+    cs = cat.scan("catalog_sales", ["cs_item_sk", "cs_order_number", "cs_ext_list_price"])
+    cr = cat.scan("catalog_returns", ["cr_item_sk", "cr_order_number", "cr_refunded_cash",
+                                      "cr_reversed_charge", "cr_store_credit"])
+    jcr = shj(cs, cr, ["cs_item_sk", "cs_order_number"], ["cr_item_sk", "cr_order_number"])
+    refund_e = col("cr_refunded_cash") + col("cr_reversed_charge") + col("cr_store_credit")
+    pre = P.Project(jcr, [_a(col("cs_item_sk"), "ui_item"), _a(col("cs_ext_list_price"), "sale_v"),
+                          _a(refund_e, "refund_v")])
+    cs_ui = P.Filter(agg2(pre, ["ui_item"],
+                          [AggFunc("sum", col("sale_v"), name="sale"),
+                           AggFunc("sum", col("refund_v"), name="refund")]),
+                     col("sale") > lit(2.0) * col("refund"))
+
+    ss = cat.scan("store_sales", ["ss_item_sk", "ss_ticket_number", "ss_store_sk",
+                                  "ss_sold_date_sk", "ss_customer_sk", "ss_cdemo_sk",
+                                  "ss_hdemo_sk", "ss_addr_sk", "ss_promo_sk",
+                                  "ss_wholesale_cost", "ss_list_price", "ss_coupon_amt"])
+    sr = cat.scan("store_returns", ["sr_item_sk", "sr_ticket_number"])
+    j = shj(ss, sr, ["ss_item_sk", "ss_ticket_number"], ["sr_item_sk", "sr_ticket_number"])
+    j = P.HashJoin(P.Exchange(j, "hash", [col("ss_item_sk")]),
+                   P.Exchange(cs_ui, "hash", [col("ui_item")]),
+                   [col("ss_item_sk")], [col("ui_item")], how="inner", build_side="right")
+    st = cat.scan("store", ["s_store_sk", "s_store_name", "s_zip"])
+    d1 = P.Project(cat.scan("date_dim", ["d_date_sk", "d_year"]),
+                   [_a(col("d_date_sk"), "d1_sk"), _a(col("d_year"), "syear")])
+    d2 = P.Project(cat.scan("date_dim", ["d_date_sk", "d_year"]),
+                   [_a(col("d_date_sk"), "d2_sk"), _a(col("d_year"), "fsyear")])
+    d3 = P.Project(cat.scan("date_dim", ["d_date_sk", "d_year"]),
+                   [_a(col("d_date_sk"), "d3_sk"), _a(col("d_year"), "s2year")])
+    cust = cat.scan("customer", ["c_customer_sk", "c_current_cdemo_sk", "c_current_hdemo_sk",
+                                 "c_current_addr_sk", "c_first_sales_date_sk",
+                                 "c_first_shipto_date_sk"])
+    cd1 = P.Project(cat.scan("customer_demographics", ["cd_demo_sk", "cd_marital_status"]),
+                    [_a(col("cd_demo_sk"), "cd1_sk"), _a(col("cd_marital_status"), "cd1_ms")])
+    cd2 = P.Project(cat.scan("customer_demographics", ["cd_demo_sk", "cd_marital_status"]),
+                    [_a(col("cd_demo_sk"), "cd2_sk"), _a(col("cd_marital_status"), "cd2_ms")])
+    pr = cat.scan("promotion", ["p_promo_sk"])
+    hd1 = P.Project(cat.scan("household_demographics", ["hd_demo_sk", "hd_income_band_sk"]),
+                    [_a(col("hd_demo_sk"), "hd1_sk"), _a(col("hd_income_band_sk"), "ib1_fk")])
+    hd2 = P.Project(cat.scan("household_demographics", ["hd_demo_sk", "hd_income_band_sk"]),
+                    [_a(col("hd_demo_sk"), "hd2_sk"), _a(col("hd_income_band_sk"), "ib2_fk")])
+    ad1 = P.Project(cat.scan("customer_address",
+                             ["ca_address_sk", "ca_street_number", "ca_street_name",
+                              "ca_city", "ca_zip"]),
+                    [_a(col("ca_address_sk"), "ad1_sk"),
+                     _a(col("ca_street_number"), "b_street_number"),
+                     _a(col("ca_street_name"), "b_streen_name"),
+                     _a(col("ca_city"), "b_city"), _a(col("ca_zip"), "b_zip")])
+    ad2 = P.Project(cat.scan("customer_address",
+                             ["ca_address_sk", "ca_street_number", "ca_street_name",
+                              "ca_city", "ca_zip"]),
+                    [_a(col("ca_address_sk"), "ad2_sk"),
+                     _a(col("ca_street_number"), "c_street_number"),
+                     _a(col("ca_street_name"), "c_street_name"),
+                     _a(col("ca_city"), "c_city"), _a(col("ca_zip"), "c_zip")])
+    ib1 = P.Project(cat.scan("income_band", ["ib_income_band_sk"]),
+                    [_a(col("ib_income_band_sk"), "ib1_sk")])
+    ib2 = P.Project(cat.scan("income_band", ["ib_income_band_sk"]),
+                    [_a(col("ib_income_band_sk"), "ib2_sk")])
+    it = P.Filter(cat.scan("item", ["i_item_sk", "i_product_name", "i_color",
+                                    "i_current_price"]),
+                  col("i_color").isin(["purple", "burlywood", "indian", "spring",
+                                       "floral", "medium"])
+                  & col("i_current_price").between(64.0, 74.0)
+                  & col("i_current_price").between(65.0, 79.0))
+    j = bhj(j, st, ["ss_store_sk"], ["s_store_sk"])
+    j = bhj(j, d1, ["ss_sold_date_sk"], ["d1_sk"])
+    j = bhj(j, cust, ["ss_customer_sk"], ["c_customer_sk"])
+    j = bhj(j, cd1, ["ss_cdemo_sk"], ["cd1_sk"])
+    j = bhj(j, hd1, ["ss_hdemo_sk"], ["hd1_sk"])
+    j = bhj(j, ad1, ["ss_addr_sk"], ["ad1_sk"])
+    j = bhj(j, it, ["ss_item_sk"], ["i_item_sk"])
+    j = bhj(j, cd2, ["c_current_cdemo_sk"], ["cd2_sk"])
+    j = bhj(j, hd2, ["c_current_hdemo_sk"], ["hd2_sk"])
+    j = bhj(j, ad2, ["c_current_addr_sk"], ["ad2_sk"])
+    j = bhj(j, d2, ["c_first_sales_date_sk"], ["d2_sk"])
+    j = bhj(j, d3, ["c_first_shipto_date_sk"], ["d3_sk"])
+    j = bhj(j, pr, ["ss_promo_sk"], ["p_promo_sk"])
+    j = bhj(j, ib1, ["ib1_fk"], ["ib1_sk"])
+    j = bhj(j, ib2, ["ib2_fk"], ["ib2_sk"])
+    j = P.Filter(j, ~(col("cd1_ms") == col("cd2_ms")))
+    keys = ["i_product_name", "ss_item_sk", "s_store_name", "s_zip",
+            "b_street_number", "b_streen_name", "b_city", "b_zip",
+            "c_street_number", "c_street_name", "c_city", "c_zip",
+            "syear", "fsyear", "s2year"]
+    cross_sales = agg2(j, keys,
+                       [AggFunc("count_star", None, name="cnt"),
+                        AggFunc("sum", col("ss_wholesale_cost"), name="s1"),
+                        AggFunc("sum", col("ss_list_price"), name="s2"),
+                        AggFunc("sum", col("ss_coupon_amt"), name="s3")])
+    cs_b = s.execute(cross_sales)
+    cs1 = P.Filter(P.MemoryScan(cs_b), col("syear") == 1999)
+    cs2 = P.Project(P.Filter(P.MemoryScan(cs_b), col("syear") == 2000),
+                    [_a(col("ss_item_sk"), "item2"), _a(col("s_store_name"), "sn2"),
+                     _a(col("s_zip"), "sz2"), _a(col("cnt"), "cnt2"),
+                     _a(col("s1"), "s1_2"), _a(col("s2"), "s2_2"),
+                     _a(col("s3"), "s3_2"), _a(col("syear"), "syear2")])
+    j2 = shj(cs1, cs2, ["ss_item_sk", "s_store_name", "s_zip"],
+             ["item2", "sn2", "sz2"])
+    f = P.Filter(j2, col("cnt2") <= col("cnt"))
+    out_cols = ["i_product_name", "s_store_name", "s_zip", "b_street_number",
+                "b_streen_name", "b_city", "b_zip", "c_street_number",
+                "c_street_name", "c_city", "c_zip", "syear", "cnt", "s1", "s2", "s3"]
+    proj = P.Project(f, [_a(col(c), c) for c in out_cols]
+                     + [_a(col("s1_2"), "s1_2"), _a(col("s2_2"), "s2_2"),
+                        _a(col("s3_2"), "s3_2"), _a(col("syear2"), "syear2"),
+                        _a(col("cnt2"), "cnt2")])
+    return topk(proj, [(col("i_product_name"), True), (col("s_store_name"), True),
+                       (col("cnt2"), True), (col("s1"), True), (col("s1_2"), True)],
+                100000)
+
+
+QUERIES.update({"q64": q64})
