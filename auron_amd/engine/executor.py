@@ -327,16 +327,10 @@ class Executor:
         for expr, asc in reversed(list(keys)):
             c = expr.eval(batch).gather(perm)
             if c.dtype.is_string:
-                import numpy as np
-
-                vals = np.array(c.to_pylist(), dtype=object)
-                keyed = np.array(["" if v is None else v for v in vals], dtype=object)
-                order = np.argsort(keyed, kind="stable")
-                if not asc:
-                    order = order[::-1].copy()
-                    # reverse breaks stability; re-stabilize via argsort of ranks
-                ot = torch.from_numpy(order.astype("int64")).to(batch.device)
-                null_rank = torch.tensor([v is None for v in vals], dtype=torch.int8)[ot.cpu()].to(batch.device)
+                ranks = ops.string_sort_ranks(c)
+                ot = torch.argsort(ranks, stable=True, descending=not asc)
+                nulls = (~c.validity) if c.validity is not None else torch.zeros(n, dtype=torch.bool, device=batch.device)
+                null_rank = nulls[ot].to(torch.int8)
             else:
                 v = c.data
                 if v.dtype == torch.bool:

@@ -333,6 +333,24 @@ def join_counts(build_cols: List[Column], probe_cols: List[Column]) -> torch.Ten
     return counts
 
 
+def string_sort_ranks(c: Column) -> torch.Tensor:
+    """Order-preserving dense ranks for a string column (device int64).
+
+    Unique strings (group-table reps) are sorted on the host — uniques are
+    small — and each row maps to its rank through the device group ids.
+    Used by sort/window so string order-by keys never fall back to a host
+    argsort over the full column."""
+    gids, reps = group_ids([c])
+    rep_vals = c.gather(reps).to_pylist()
+    order = sorted(range(len(rep_vals)),
+                   key=lambda i: (rep_vals[i] is None, rep_vals[i] or ""))
+    rank = [0] * len(rep_vals)
+    for r, i in enumerate(order):
+        rank[i] = r
+    rank_t = torch.tensor(rank, dtype=torch.int64, device=c.device)
+    return rank_t[gids]
+
+
 # ================================================================ partition
 def partition_ids(cols: List[Column], nparts: int) -> torch.Tensor:
     """Spark HashPartitioning: pmod(murmur3(keys, seed=42), nparts)."""
