@@ -1,0 +1,26 @@
+"""Observability HTTP service (reference http-service parity)."""
+import json
+import urllib.request
+
+from auron_amd import AuronSession, col, dtypes
+from auron_amd import http_service
+from auron_amd.column import RecordBatch
+from auron_amd.plan import nodes as P
+
+
+def test_metrics_endpoints():
+    s = AuronSession()
+    b = RecordBatch.from_pydict({"x": [1, 2, 3]}, {"x": dtypes.int64})
+    s.collect(P.Filter(P.MemoryScan([b]), col("x") > 1))
+    srv = http_service.start(s)
+    try:
+        base = f"http://127.0.0.1:{srv.port}"
+        health = json.load(urllib.request.urlopen(f"{base}/healthz"))
+        assert health == {"ok": True}
+        metrics = json.load(urllib.request.urlopen(f"{base}/metrics"))
+        assert "op.Filter" in metrics["operators"]
+        assert metrics["memory"]["budget_bytes"] > 0
+        conf = json.load(urllib.request.urlopen(f"{base}/config"))
+        assert "spark.auron.batchSize" in conf
+    finally:
+        srv.stop()
