@@ -68,6 +68,21 @@ class ExecContext:
         return _T()
 
 
+def _normalize_join_keys(lkeys, rkeys):
+    """Promote corresponding key pairs to one dtype: the native kernels
+    hash and compare by the build side's dtype, so int32-vs-int64 pairs
+    would silently mismatch (and murmur3 differs between widths)."""
+    from ..exprs import _promote
+
+    lo, ro = [], []
+    for l, r in zip(lkeys, rkeys):
+        if l.dtype.code != r.dtype.code:
+            l, r, _ = _promote(l, r)
+        lo.append(l)
+        ro.append(r)
+    return lo, ro
+
+
 def _concat(batches: List[RecordBatch]) -> RecordBatch:
     assert batches, "executor invariant violated: empty batch list"
     if len(batches) == 1:
@@ -564,6 +579,7 @@ class Executor:
             left = _concat(left_batches())
         lkeys = [k.eval(left) for k in node.left_keys]
         rkeys = [k.eval(right) for k in node.right_keys]
+        lkeys, rkeys = _normalize_join_keys(lkeys, rkeys)
         how = node.how
         device = self.ctx.device
 

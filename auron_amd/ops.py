@@ -352,8 +352,23 @@ def string_sort_ranks(c: Column) -> torch.Tensor:
 
 
 # ================================================================ partition
+def _normalize_hash_col(c: Column) -> Column:
+    """Promote key columns to a canonical width before hashing: murmur3
+    hashes int32 and int64 differently, so `week_seq` (int32) vs
+    `week_seq - 52` (promoted int64) would land on different ranks. Our
+    internal exchange only needs consistency, not Spark shuffle-file
+    layout, so ints hash as int64 and floats as float64."""
+    if c.dtype.code in (dtypes.BOOL, dtypes.INT8, dtypes.INT16, dtypes.INT32,
+                        dtypes.DATE32):
+        return Column(dtypes.int64, c.data.to(torch.int64), c.validity)
+    if c.dtype.code == dtypes.FLOAT32:
+        return Column(dtypes.float64, c.data.to(torch.float64), c.validity)
+    return c
+
+
 def partition_ids(cols: List[Column], nparts: int) -> torch.Tensor:
-    """Spark HashPartitioning: pmod(murmur3(keys, seed=42), nparts)."""
+    """Internal hash partitioning: pmod(murmur3(normalized keys, 42), n)."""
+    cols = [_normalize_hash_col(c) for c in cols]
     device = cols[0].device
     h = murmur3(cols)
     if not _use_native(device):

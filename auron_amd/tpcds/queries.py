@@ -118,8 +118,11 @@ def q6(cat: Catalog, s):
                         [_a(col("i_category"), "avg_cat")],
                         [AggFunc("avg", col("i_current_price"), name="cat_avg_price")],
                         mode="complete")
-    it = bhj(cat.scan("item", ["i_item_sk", "i_category", "i_current_price"]),
-             cat_avg, ["i_category"], ["avg_cat"])
+    # cat_avg is already replicated on every rank (Broadcast input) — join
+    # locally; re-broadcasting would duplicate the build side
+    it = P.HashJoin(cat.scan("item", ["i_item_sk", "i_category", "i_current_price"]),
+                    cat_avg, [col("i_category")], [col("avg_cat")],
+                    how="inner", build_side="right", broadcast=False)
     it_f = P.Filter(it, col("i_current_price") > col("cat_avg_price") * lit(1.2))
 
     ss = cat.scan("store_sales", ["ss_sold_date_sk", "ss_customer_sk", "ss_item_sk"])
@@ -1060,8 +1063,10 @@ def _ship_q(cat, s, fact, pre, date0, state, site_join, returns_table, ret_pre):
     ord_wh = agg2(P.Project(all_orders, [_a(col(f"{pre}_order_number"), "o"),
                                          _a(col(f"{pre}_warehouse_sk"), "w")]),
                   ["o", "w"], [])
+    # distinct (o,w) pairs are partitioned by (o,w): regroup by o through an
+    # exchange before counting warehouses per order
     multi = P.Filter(
-        P.HashAgg(ord_wh, [_a(col("o"), "o")],
+        P.HashAgg(P.Exchange(ord_wh, "hash", [col("o")]), [_a(col("o"), "o")],
                   [AggFunc("count_star", None, name="nwh")], mode="complete"),
         col("nwh") > 1)
     j = P.HashJoin(P.Exchange(j, "hash", [col(f"{pre}_order_number")]),
@@ -1367,8 +1372,9 @@ def q31(cat, s):
 
     def inst(batches, q, name):
         f = P.Filter(P.MemoryScan(batches), col("d_qoy") == q)
-        return P.Broadcast(P.Project(f, [_a(col("ca_county"), f"{name}_county"),
-                                         _a(col("v"), name)]))
+        return P.Exchange(P.Project(f, [_a(col("ca_county"), f"{name}_county"),
+                                        _a(col("v"), name)]),
+                          "hash", [col(f"{name}_county")])
 
     base = inst(ss_b, 1, "ss1")
     j = P.HashJoin(base, inst(ss_b, 2, "ss2"), [col("ss1_county")], [col("ss2_county")],
@@ -2985,7 +2991,8 @@ def q95(cat, s):
     all_orders = cat.scan("web_sales", ["ws_order_number", "ws_warehouse_sk"])
     ord_wh = agg2(P.Project(all_orders, [_a(col("ws_order_number"), "o"),
                                          _a(col("ws_warehouse_sk"), "w")]), ["o", "w"], [])
-    multi = P.Filter(P.HashAgg(ord_wh, [_a(col("o"), "o")],
+    multi = P.Filter(P.HashAgg(P.Exchange(ord_wh, "hash", [col("o")]),
+                               [_a(col("o"), "o")],
                                [AggFunc("count_star", None, name="nwh")], mode="complete"),
                      col("nwh") > 1)
     multi_b = s.execute(multi)

@@ -9,7 +9,8 @@ import pytest
 import torch.distributed as dist
 import torch.multiprocessing as mp
 
-QUERIES_TO_CHECK = ["q3", "q23", "q72", "q38", "q47", "q5", "q1", "q88"]
+QUERIES_TO_CHECK = ["q3", "q23", "q72", "q38", "q47", "q5", "q1", "q88",
+                    "q2", "q6", "q16", "q31", "q59", "q95", "q51", "q14"]
 SF = 0.01
 ROOT = os.path.join(os.path.dirname(__file__), "..", ".tpcds_cache")
 
