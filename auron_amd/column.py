@@ -171,6 +171,9 @@ class Column:
     def concat(cols: List["Column"]) -> "Column":
         assert cols
         dt = cols[0].dtype
+        for c in cols[1:]:
+            assert c.dtype.code == dt.code and c.dtype.scale == dt.scale, \
+                f"concat dtype mismatch: {dt.name} vs {c.dtype.name}"
         device = cols[0].device
         any_null = any(c.validity is not None for c in cols)
         validity = None
@@ -226,8 +229,16 @@ class Column:
                 offsets = torch.from_numpy(off_np) if start == 0 else torch.from_numpy((off_np - start))
             col = Column(dt, data, validity, offsets)
         elif dt.code == dtypes.DECIMAL64:
-            np_vals = arr.cast(pa.decimal128(dt.precision, dt.scale)).to_numpy(zero_copy_only=False)
-            scaled = np.round(np_vals.astype(np.float64) * (10 ** dt.scale)).astype(np.int64)
+            d128 = arr.cast(pa.decimal128(dt.precision, dt.scale))
+            d128 = d128.combine_chunks() if isinstance(d128, pa.ChunkedArray) else d128
+            buf = d128.buffers()[1]
+            if buf is not None:
+                # exact unscaled ints: decimal128 little-endian, low 8 bytes
+                raw = np.frombuffer(buf, dtype=np.int64,
+                                    count=2 * (len(d128) + d128.offset))
+                scaled = raw[2 * d128.offset::2].copy()
+            else:
+                scaled = np.zeros(len(d128), dtype=np.int64)
             col = Column(dt, torch.from_numpy(scaled), validity, None)
         else:
             if dt.code == dtypes.DATE32:

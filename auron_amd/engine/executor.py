@@ -748,6 +748,9 @@ class Executor:
         if bool(validity.all()):
             validity = None
         out = run
+        if val.dtype.code == dtypes.DECIMAL64:
+            # decimal running min/max stays in scaled-int64 space
+            return Column(val.dtype, out.to(torch.int64), validity)
         if val.dtype.is_integer:
             return Column(dtypes.int64, out.to(torch.int64), validity)
         return Column(dtypes.float64, out, validity)

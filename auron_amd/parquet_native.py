@@ -325,9 +325,14 @@ def _build_meta(path: str, columns: List[str]) -> Optional[_FileMeta]:
     for cname in columns:
         ci = nf.names.index(cname)
         sc = nf.schema.column(ci)
-        logical = str(sc.logical_type).lower()
+        logical = str(sc.logical_type)
         phys = sc.physical_type
-        if "date" in logical:
+        if logical.startswith("Decimal"):
+            import re
+
+            m = re.search(r"precision=(\d+), scale=(\d+)", logical)
+            dt = dtypes.decimal64(int(m.group(1)), int(m.group(2)))
+        elif "date" in logical.lower():
             dt = dtypes.date32
         elif phys == "INT32":
             dt = dtypes.int32
@@ -420,6 +425,8 @@ def read_columns_native(path: str, columns: List[str], device,
                 data_np, valid_np = decode_chunk_np(buf, pages, nvals, cm.phys)
                 data_t = torch.from_numpy(data_np)
                 valid_t = torch.from_numpy(valid_np).to(torch.bool) if valid_np is not None else None
+            if cm.dtype.code == dtypes.DECIMAL64 and data_t.dtype == torch.int32:
+                data_t = data_t.to(torch.int64)  # widen INT32-backed decimals
             parts_data.append(data_t)
             if valid_t is not None:
                 parts_valid.append(valid_t)

@@ -462,6 +462,22 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
 
     from .. import dtypes as dt
 
+    def decimal_array(values, mask, precision, scale):
+        """Exact decimal128 from float money values (2dp): unscaled int
+        cents packed into 128-bit little-endian buffers."""
+        v = np.asarray(values, dtype=np.float64)
+        cents = np.round(v * (10 ** scale)).astype(np.int64)
+        n2 = len(cents)
+        packed = np.zeros(2 * n2, dtype=np.int64)
+        packed[0::2] = cents
+        packed[1::2] = np.where(cents < 0, -1, 0)  # sign extension
+        if mask is not None:
+            validity = np.packbits(~mask, bitorder="little")
+            bufs = [pa.py_buffer(validity.tobytes()), pa.py_buffer(packed.tobytes())]
+        else:
+            bufs = [None, pa.py_buffer(packed.tobytes())]
+        return pa.Array.from_buffers(pa.decimal128(precision, scale), n2, bufs)
+
     schema = SCHEMAS[name]
     arrays = []
     for colname, dtype in schema.items():
@@ -469,6 +485,9 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
         mask = None
         if colname in valids:
             mask = ~valids[colname]
+        if dtype.code == dt.DECIMAL64:
+            arrays.append(decimal_array(arr, mask, dtype.precision, dtype.scale))
+            continue
         atype = {
             dt.INT32: pa.int32(), dt.INT64: pa.int64(), dt.FLOAT64: pa.float64(),
             dt.STRING: pa.string(), dt.DATE32: pa.date32(),
@@ -495,7 +514,7 @@ def _nparts_for(table: str, sf: float) -> int:
     return 1
 
 
-DATAGEN_VERSION = 8
+DATAGEN_VERSION = 9
 
 
 def dataset_root(root: str, sf: float) -> str:
@@ -535,6 +554,7 @@ def write_dataset(root: str, sf: float, tables: Optional[List[str]] = None,
                     pq.write_table(tbl, path, compression="NONE",
                                    use_dictionary=False,
                                    data_page_version="1.0",
+                                   store_decimal_as_integer=True,
                                    row_group_size=1 << 20)
             job += 1
     return base

@@ -16,6 +16,10 @@ def _read(root, sf, table, columns=None):
     import pyarrow as pa
 
     t = pa.concat_tables([pq.read_table(p, columns=columns) for p in paths])
+    # decimals compare as float64 in the oracle (tolerance-based checks)
+    fields = [(i, f) for i, f in enumerate(t.schema) if pa.types.is_decimal(f.type)]
+    for i, f in fields:
+        t = t.set_column(i, f.name, t.column(i).cast(pa.float64()))
     return t.to_pandas()
 
 

@@ -1510,9 +1510,10 @@ def q5(cat, s):
     z = lit(0.0)
 
     def sel(scan, fk_col, date_col, sp, pr, ra, nl):
+        f64 = dtypes.float64
         return P.Project(scan, [_a(col(fk_col), "fk"), _a(col(date_col), "date_sk"),
-                                _a(sp, "sales_price"), _a(pr, "profit"),
-                                _a(ra, "return_amt"), _a(nl, "net_loss")])
+                                _a(sp.cast(f64), "sales_price"), _a(pr.cast(f64), "profit"),
+                                _a(ra.cast(f64), "return_amt"), _a(nl.cast(f64), "net_loss")])
 
     ss_rows = sel(cat.scan("store_sales", ["ss_store_sk", "ss_sold_date_sk",
                                            "ss_ext_sales_price", "ss_net_profit"]),
@@ -1549,8 +1550,8 @@ def q5(cat, s):
     wr_rows = P.Project(wrj, [_a(col("ws_web_site_sk"), "fk"),
                               _a(col("wr_returned_date_sk"), "date_sk"),
                               _a(z, "sales_price"), _a(z, "profit"),
-                              _a(col("wr_return_amt"), "return_amt"),
-                              _a(col("wr_net_loss"), "net_loss")])
+                              _a(col("wr_return_amt").cast(dtypes.float64), "return_amt"),
+                              _a(col("wr_net_loss").cast(dtypes.float64), "net_loss")])
     web = cat.scan("web_site", ["web_site_sk", "web_site_id"])
     wsr = part(ws_rows, wr_rows, web, "web_site_sk", "fk", "web_site_id", "web site")
 
@@ -1581,11 +1582,13 @@ def q77(cat, s):
              "sr_return_amt", "sr_net_loss", ["rsk", "returns", "profit_loss"])
     store_part = P.HashJoin(ss, P.Broadcast(sr), [col("sk")], [col("rsk")],
                             how="left", build_side="right")
+    f64 = dtypes.float64
     store_rows = P.Project(store_part, [
         _a(lit("store channel"), "channel"), _a(col("sk").cast(dtypes.string), "id"),
-        _a(col("sales"), "sales"),
-        _a(Coalesce([col("returns"), lit(0.0)]), "returns"),
-        _a(col("profit") - Coalesce([col("profit_loss"), lit(0.0)]), "profit")])
+        _a(col("sales").cast(f64), "sales"),
+        _a(Coalesce([col("returns").cast(f64), lit(0.0)]), "returns"),
+        _a(col("profit").cast(f64) - Coalesce([col("profit_loss").cast(f64), lit(0.0)]),
+           "profit")])
 
     cs = cte("catalog_sales", "cs", "cs_sold_date_sk", "cs_call_center_sk",
              "cs_ext_sales_price", "cs_net_profit", ["csk", "sales", "profit"])
@@ -1599,8 +1602,8 @@ def q77(cat, s):
     cr_loss = cr_b["profit_loss"][0] if cr_b["profit_loss"] else 0.0
     catalog_rows = P.Project(cs, [
         _a(lit("catalog channel"), "channel"), _a(col("csk").cast(dtypes.string), "id"),
-        _a(col("sales"), "sales"), _a(lit(cr_ret or 0.0), "returns"),
-        _a(col("profit") - lit(cr_loss or 0.0), "profit")])
+        _a(col("sales").cast(f64), "sales"), _a(lit(float(cr_ret or 0.0)), "returns"),
+        _a(col("profit").cast(f64) - lit(float(cr_loss or 0.0)), "profit")])
 
     ws = cte("web_sales", "ws", "ws_sold_date_sk", "ws_web_page_sk",
              "ws_ext_sales_price", "ws_net_profit", ["wsk", "sales", "profit"])
@@ -1620,9 +1623,10 @@ def q77(cat, s):
                           how="left", build_side="right")
     web_rows = P.Project(web_part, [
         _a(lit("web channel"), "channel"), _a(col("wsk").cast(dtypes.string), "id"),
-        _a(col("sales"), "sales"),
-        _a(Coalesce([col("returns"), lit(0.0)]), "returns"),
-        _a(col("profit") - Coalesce([col("profit_loss"), lit(0.0)]), "profit")])
+        _a(col("sales").cast(f64), "sales"),
+        _a(Coalesce([col("returns").cast(f64), lit(0.0)]), "returns"),
+        _a(col("profit").cast(f64) - Coalesce([col("profit_loss").cast(f64), lit(0.0)]),
+           "profit")])
 
     u = P.Union([store_rows, catalog_rows, web_rows])
     r = rollup2(u, "channel", "id", None)

@@ -9,6 +9,7 @@ from __future__ import annotations
 from .. import dtypes as dt
 
 f64 = dt.float64
+d72 = dt.decimal64(7, 2)
 i32 = dt.int32
 i64 = dt.int64
 s = dt.string
@@ -25,7 +26,7 @@ SCHEMAS = {
         "i_item_sk": i64, "i_item_id": s, "i_item_desc": s, "i_category": s,
         "i_category_id": i32, "i_brand": s, "i_brand_id": i32,
         "i_manufact_id": i32, "i_class": s, "i_class_id": i32,
-        "i_current_price": f64, "i_manager_id": i32, "i_product_name": s, "i_color": s, "i_units": s, "i_size": s, "i_manufact": s,
+        "i_current_price": d72, "i_manager_id": i32, "i_product_name": s, "i_color": s, "i_units": s, "i_size": s, "i_manufact": s,
     },
     "customer": {
         "c_customer_sk": i64, "c_customer_id": s, "c_first_name": s,
@@ -75,20 +76,20 @@ SCHEMAS = {
         "ss_sold_date_sk": i64, "ss_sold_time_sk": i64, "ss_item_sk": i64,
         "ss_customer_sk": i64, "ss_cdemo_sk": i64, "ss_hdemo_sk": i64,
         "ss_addr_sk": i64, "ss_store_sk": i64, "ss_promo_sk": i64,
-        "ss_ticket_number": i64, "ss_quantity": i32, "ss_wholesale_cost": f64,
-        "ss_list_price": f64, "ss_sales_price": f64, "ss_ext_discount_amt": f64,
-        "ss_ext_sales_price": f64, "ss_ext_wholesale_cost": f64,
-        "ss_ext_list_price": f64, "ss_ext_tax": f64, "ss_coupon_amt": f64,
-        "ss_net_paid": f64, "ss_net_paid_inc_tax": f64, "ss_net_profit": f64,
+        "ss_ticket_number": i64, "ss_quantity": i32, "ss_wholesale_cost": d72,
+        "ss_list_price": d72, "ss_sales_price": d72, "ss_ext_discount_amt": d72,
+        "ss_ext_sales_price": d72, "ss_ext_wholesale_cost": d72,
+        "ss_ext_list_price": d72, "ss_ext_tax": d72, "ss_coupon_amt": d72,
+        "ss_net_paid": d72, "ss_net_paid_inc_tax": d72, "ss_net_profit": d72,
     },
     "store_returns": {
         "sr_returned_date_sk": i64, "sr_return_time_sk": i64, "sr_item_sk": i64,
         "sr_customer_sk": i64, "sr_cdemo_sk": i64, "sr_hdemo_sk": i64,
         "sr_addr_sk": i64, "sr_store_sk": i64, "sr_reason_sk": i64,
-        "sr_ticket_number": i64, "sr_return_quantity": i32, "sr_return_amt": f64,
-        "sr_return_tax": f64, "sr_return_amt_inc_tax": f64, "sr_fee": f64,
-        "sr_return_ship_cost": f64, "sr_refunded_cash": f64,
-        "sr_reversed_charge": f64, "sr_store_credit": f64, "sr_net_loss": f64,
+        "sr_ticket_number": i64, "sr_return_quantity": i32, "sr_return_amt": d72,
+        "sr_return_tax": d72, "sr_return_amt_inc_tax": d72, "sr_fee": d72,
+        "sr_return_ship_cost": d72, "sr_refunded_cash": d72,
+        "sr_reversed_charge": d72, "sr_store_credit": d72, "sr_net_loss": d72,
     },
     "catalog_sales": {
         "cs_sold_date_sk": i64, "cs_sold_time_sk": i64, "cs_ship_date_sk": i64,
@@ -96,19 +97,19 @@ SCHEMAS = {
         "cs_bill_addr_sk": i64, "cs_ship_customer_sk": i64, "cs_ship_addr_sk": i64,
         "cs_call_center_sk": i64, "cs_catalog_page_sk": i64, "cs_ship_mode_sk": i64,
         "cs_warehouse_sk": i64, "cs_item_sk": i64, "cs_promo_sk": i64,
-        "cs_order_number": i64, "cs_quantity": i32, "cs_wholesale_cost": f64,
-        "cs_list_price": f64, "cs_sales_price": f64, "cs_ext_discount_amt": f64,
-        "cs_ext_sales_price": f64, "cs_ext_wholesale_cost": f64,
-        "cs_ext_list_price": f64, "cs_ext_tax": f64, "cs_coupon_amt": f64,
-        "cs_ext_ship_cost": f64, "cs_net_paid": f64, "cs_net_paid_inc_tax": f64,
-        "cs_net_paid_inc_ship": f64, "cs_net_paid_inc_ship_tax": f64, "cs_net_profit": f64,
+        "cs_order_number": i64, "cs_quantity": i32, "cs_wholesale_cost": d72,
+        "cs_list_price": d72, "cs_sales_price": d72, "cs_ext_discount_amt": d72,
+        "cs_ext_sales_price": d72, "cs_ext_wholesale_cost": d72,
+        "cs_ext_list_price": d72, "cs_ext_tax": d72, "cs_coupon_amt": d72,
+        "cs_ext_ship_cost": d72, "cs_net_paid": d72, "cs_net_paid_inc_tax": d72,
+        "cs_net_paid_inc_ship": f64, "cs_net_paid_inc_ship_tax": d72, "cs_net_profit": d72,
     },
     "catalog_returns": {
         "cr_returned_date_sk": i64, "cr_item_sk": i64, "cr_order_number": i64,
         "cr_returning_customer_sk": i64, "cr_returning_addr_sk": i64, "cr_catalog_page_sk": i64,
-        "cr_return_quantity": i32, "cr_return_amount": f64, "cr_return_tax": f64, "cr_return_amt_inc_tax": f64,
-        "cr_net_loss": f64, "cr_refunded_cash": f64, "cr_reversed_charge": f64,
-        "cr_store_credit": f64, "cr_call_center_sk": i64,
+        "cr_return_quantity": i32, "cr_return_amount": d72, "cr_return_tax": d72, "cr_return_amt_inc_tax": d72,
+        "cr_net_loss": d72, "cr_refunded_cash": d72, "cr_reversed_charge": d72,
+        "cr_store_credit": d72, "cr_call_center_sk": i64,
     },
     "web_sales": {
         "ws_sold_date_sk": i64, "ws_sold_time_sk": i64, "ws_ship_date_sk": i64,
@@ -116,17 +117,17 @@ SCHEMAS = {
         "ws_bill_hdemo_sk": i64, "ws_bill_addr_sk": i64, "ws_ship_customer_sk": i64, "ws_ship_addr_sk": i64,
         "ws_web_page_sk": i64, "ws_web_site_sk": i64, "ws_ship_mode_sk": i64,
         "ws_warehouse_sk": i64, "ws_promo_sk": i64, "ws_order_number": i64,
-        "ws_quantity": i32, "ws_wholesale_cost": f64, "ws_list_price": f64,
-        "ws_sales_price": f64, "ws_ext_discount_amt": f64, "ws_ext_sales_price": f64,
-        "ws_ext_wholesale_cost": f64, "ws_ext_list_price": f64, "ws_ext_tax": f64,
-        "ws_coupon_amt": f64, "ws_ext_ship_cost": f64, "ws_net_paid": f64,
-        "ws_net_paid_inc_tax": f64, "ws_net_profit": f64,
+        "ws_quantity": i32, "ws_wholesale_cost": d72, "ws_list_price": d72,
+        "ws_sales_price": d72, "ws_ext_discount_amt": d72, "ws_ext_sales_price": d72,
+        "ws_ext_wholesale_cost": d72, "ws_ext_list_price": d72, "ws_ext_tax": d72,
+        "ws_coupon_amt": d72, "ws_ext_ship_cost": d72, "ws_net_paid": d72,
+        "ws_net_paid_inc_tax": d72, "ws_net_profit": d72,
     },
     "web_returns": {
         "wr_returned_date_sk": i64, "wr_item_sk": i64, "wr_order_number": i64,
         "wr_returning_customer_sk": i64, "wr_returning_addr_sk": i64, "wr_refunded_cdemo_sk": i64, "wr_returning_cdemo_sk": i64, "wr_refunded_addr_sk": i64, "wr_reason_sk": i64, "wr_return_quantity": i32,
-        "wr_return_amt": f64, "wr_net_loss": f64, "wr_fee": f64,
-        "wr_refunded_cash": f64, "wr_reversed_charge": f64, "wr_account_credit": f64,
+        "wr_return_amt": d72, "wr_net_loss": d72, "wr_fee": d72,
+        "wr_refunded_cash": d72, "wr_reversed_charge": d72, "wr_account_credit": d72,
     },
     "inventory": {
         "inv_date_sk": i64, "inv_item_sk": i64, "inv_warehouse_sk": i64,
