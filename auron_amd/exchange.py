@@ -30,6 +30,10 @@ def _pad8(x: int) -> int:
 
 
 def _byte_view(t: torch.Tensor) -> torch.Tensor:
+    if t.numel() == 0:
+        # 0-row tensors may carry stride 0 (e.g. empty numpy slices), which
+        # .contiguous() keeps and .view() rejects
+        return torch.empty(0, dtype=torch.uint8, device=t.device)
     t = t.contiguous()
     if t.dtype == torch.bool:
         t = t.view(torch.uint8)
