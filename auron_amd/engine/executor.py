@@ -116,6 +116,10 @@ class Executor:
         return node
 
     def execute(self, node: P.PlanNode) -> List[RecordBatch]:
+        from .. import functions as F
+        ctx = F.EVAL_CONTEXT.get()
+        if ctx.get("partition_id") != self.ctx.rank:
+            F.EVAL_CONTEXT.set({**ctx, "partition_id": self.ctx.rank})
         node = self._rewrite(node)
         name = type(node).__name__
         fn = getattr(self, f"_exec_{name}", None)

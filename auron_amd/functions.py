@@ -1,0 +1,529 @@
+"""Spark-semantics scalar functions (ext-functions parity).
+
+Role parity: datafusion-ext-functions (spark_strings.rs, spark_dates.rs,
+spark_math) and ext-exprs (SparkPartitionId, MonotonicallyIncreasingId,
+RowNum). Functions are torch-vectorized; a few cold string paths bounce
+through the host (documented inline).
+"""
+from __future__ import annotations
+
+import contextvars
+from dataclasses import dataclass, field
+from typing import List, Optional
+
+import torch
+
+from . import dtypes, strings
+from .column import Column, RecordBatch
+from .dtypes import DataType
+from .exprs import (Cast, Expr, Literal, _all_valid, _cast_col,
+                    _civil_from_days, combine_validity)
+
+# executor-injected evaluation context (partition id etc.)
+EVAL_CONTEXT: contextvars.ContextVar = contextvars.ContextVar(
+    "auron_eval_context", default={"partition_id": 0, "batch_ordinal": 0})
+
+
+def _f64(e, batch) -> Column:
+    return _cast_col(e.eval(batch), dtypes.float64)
+
+
+@dataclass(eq=False)
+class Round(Expr):
+    """Spark round: half-up away from zero."""
+    child: Expr
+    ndigits: int = 0
+
+    def eval(self, batch):
+        c = _f64(self.child, batch)
+        f = 10.0 ** self.ndigits
+        x = c.data * f
+        data = torch.where(x >= 0, torch.floor(x + 0.5), torch.ceil(x - 0.5)) / f
+        return Column(dtypes.float64, data, c.validity)
+
+
+@dataclass(eq=False)
+class Floor(Expr):
+    child: Expr
+
+    def eval(self, batch):
+        c = _f64(self.child, batch)
+        return Column(dtypes.int64, torch.floor(c.data).to(torch.int64), c.validity)
+
+
+@dataclass(eq=False)
+class Ceil(Expr):
+    child: Expr
+
+    def eval(self, batch):
+        c = _f64(self.child, batch)
+        return Column(dtypes.int64, torch.ceil(c.data).to(torch.int64), c.validity)
+
+
+@dataclass(eq=False)
+class Exp(Expr):
+    child: Expr
+
+    def eval(self, batch):
+        c = _f64(self.child, batch)
+        return Column(dtypes.float64, torch.exp(c.data), c.validity)
+
+
+@dataclass(eq=False)
+class Ln(Expr):
+    child: Expr
+
+    def eval(self, batch):
+        c = _f64(self.child, batch)
+        ok = c.data > 0
+        v = combine_validity(c)
+        v = ok if v is None else (v & ok)
+        return Column(dtypes.float64, torch.log(c.data.clamp(min=1e-300)), v)
+
+
+@dataclass(eq=False)
+class Log10(Expr):
+    child: Expr
+
+    def eval(self, batch):
+        c = _f64(self.child, batch)
+        ok = c.data > 0
+        v = combine_validity(c)
+        v = ok if v is None else (v & ok)
+        return Column(dtypes.float64, torch.log10(c.data.clamp(min=1e-300)), v)
+
+
+@dataclass(eq=False)
+class Pow(Expr):
+    base: Expr
+    exponent: Expr
+
+    def eval(self, batch):
+        b = _f64(self.base, batch)
+        e = _f64(self.exponent, batch)
+        return Column(dtypes.float64, torch.pow(b.data, e.data),
+                      combine_validity(b, e))
+
+
+@dataclass(eq=False)
+class Sign(Expr):
+    child: Expr
+
+    def eval(self, batch):
+        c = _f64(self.child, batch)
+        return Column(dtypes.float64, torch.sign(c.data), c.validity)
+
+
+@dataclass(eq=False)
+class Greatest(Expr):
+    args: List[Expr]
+
+    def eval(self, batch):
+        cols = [_f64(a, batch) for a in self.args]
+        data = cols[0].data
+        for c in cols[1:]:
+            data = torch.maximum(data, c.data)
+        # Spark greatest skips nulls; null only when ALL null
+        n = batch.num_rows
+        device = batch.device
+        any_valid = torch.zeros(n, dtype=torch.bool, device=device)
+        acc = torch.full((n,), float("-inf"), dtype=torch.float64, device=device)
+        for c in cols:
+            v = c.validity if c.validity is not None else _all_valid(n, device)
+            acc = torch.where(v, torch.maximum(acc, c.data), acc)
+            any_valid = any_valid | v
+        return Column(dtypes.float64, acc, None if bool(any_valid.all()) else any_valid)
+
+
+@dataclass(eq=False)
+class Least(Expr):
+    args: List[Expr]
+
+    def eval(self, batch):
+        cols = [_f64(a, batch) for a in self.args]
+        n = batch.num_rows
+        device = batch.device
+        any_valid = torch.zeros(n, dtype=torch.bool, device=device)
+        acc = torch.full((n,), float("inf"), dtype=torch.float64, device=device)
+        for c in cols:
+            v = c.validity if c.validity is not None else _all_valid(n, device)
+            acc = torch.where(v, torch.minimum(acc, c.data), acc)
+            any_valid = any_valid | v
+        return Column(dtypes.float64, acc, None if bool(any_valid.all()) else any_valid)
+
+
+@dataclass(eq=False)
+class NullIf(Expr):
+    left: Expr
+    right: Expr
+
+    def eval(self, batch):
+        from .exprs import Cmp
+
+        l = self.left.eval(batch)
+        eq = Cmp("==", self.left, self.right).eval(batch)
+        hit = eq.data.bool()
+        if eq.validity is not None:
+            hit = hit & eq.validity
+        v = l.validity if l.validity is not None else _all_valid(len(l), l.device)
+        return Column(l.dtype, l.data, v & ~hit, l.offsets)
+
+
+@dataclass(eq=False)
+class Nvl2(Expr):
+    check: Expr
+    if_not_null: Expr
+    if_null: Expr
+
+    def eval(self, batch):
+        from .exprs import CaseWhen, IsNull, Not
+
+        return CaseWhen([(Not(IsNull(self.check)), self.if_not_null)],
+                        self.if_null).eval(batch)
+
+
+@dataclass(eq=False)
+class If(Expr):
+    cond: Expr
+    then: Expr
+    otherwise: Expr
+
+    def eval(self, batch):
+        from .exprs import CaseWhen
+
+        return CaseWhen([(self.cond, self.then)], self.otherwise).eval(batch)
+
+
+# ------------------------------------------------------------------ dates
+def _days_in_month(y, m):
+    dim = torch.tensor([31, 28, 31, 30, 31, 30, 31, 31, 30, 31, 30, 31],
+                       dtype=torch.int64, device=y.device)
+    d = dim[(m - 1).clamp(0, 11)]
+    leap = ((y % 4 == 0) & (y % 100 != 0)) | (y % 400 == 0)
+    return torch.where((m == 2) & leap, d + 1, d)
+
+
+def _days_from_civil(y, m, d):
+    """Hinnant's days_from_civil, vectorized (inverse of _civil_from_days)."""
+    y = y - (m <= 2).to(torch.int64)
+    era = torch.div(torch.where(y >= 0, y, y - 399), 400, rounding_mode="floor")
+    yoe = y - era * 400
+    mp = torch.where(m > 2, m - 3, m + 9)
+    doy = torch.div(153 * mp + 2, 5, rounding_mode="floor") + d - 1
+    doe = yoe * 365 + torch.div(yoe, 4, rounding_mode="floor") \
+        - torch.div(yoe, 100, rounding_mode="floor") + doy
+    return era * 146097 + doe - 719468
+
+
+@dataclass(eq=False)
+class DateAdd(Expr):
+    child: Expr
+    days: Expr
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        d = self.days.eval(batch)
+        data = (c.data.to(torch.int64) + d.data.to(torch.int64)).to(torch.int32)
+        return Column(dtypes.date32, data, combine_validity(c, d))
+
+
+@dataclass(eq=False)
+class DateSub(Expr):
+    child: Expr
+    days: Expr
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        d = self.days.eval(batch)
+        data = (c.data.to(torch.int64) - d.data.to(torch.int64)).to(torch.int32)
+        return Column(dtypes.date32, data, combine_validity(c, d))
+
+
+@dataclass(eq=False)
+class DateDiff(Expr):
+    end: Expr
+    start: Expr
+
+    def eval(self, batch):
+        e = self.end.eval(batch)
+        s = self.start.eval(batch)
+        data = (e.data.to(torch.int64) - s.data.to(torch.int64)).to(torch.int32)
+        return Column(dtypes.int32, data, combine_validity(e, s))
+
+
+@dataclass(eq=False)
+class AddMonths(Expr):
+    child: Expr
+    months: int
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        y, m, d = _civil_from_days(c.data)
+        t = y * 12 + (m - 1) + self.months
+        ny = torch.div(t, 12, rounding_mode="floor")
+        nm = t - ny * 12 + 1
+        nd = torch.minimum(d, _days_in_month(ny, nm))
+        return Column(dtypes.date32, _days_from_civil(ny, nm, nd).to(torch.int32),
+                      c.validity)
+
+
+@dataclass(eq=False)
+class LastDay(Expr):
+    child: Expr
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        y, m, d = _civil_from_days(c.data)
+        nd = _days_in_month(y, m)
+        return Column(dtypes.date32, _days_from_civil(y, m, nd).to(torch.int32),
+                      c.validity)
+
+
+@dataclass(eq=False)
+class Quarter(Expr):
+    child: Expr
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        _, m, _ = _civil_from_days(c.data)
+        return Column(dtypes.int32, (torch.div(m - 1, 3, rounding_mode="floor") + 1)
+                      .to(torch.int32), c.validity)
+
+
+@dataclass(eq=False)
+class DayOfWeek(Expr):
+    """Spark: 1 = Sunday .. 7 = Saturday."""
+    child: Expr
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        # 1970-01-01 was a Thursday (dow 5 in Spark numbering)
+        dow = torch.remainder(c.data.to(torch.int64) + 4, 7) + 1
+        return Column(dtypes.int32, dow.to(torch.int32), c.validity)
+
+
+@dataclass(eq=False)
+class WeekOfYear(Expr):
+    """ISO week number."""
+    child: Expr
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        days = c.data.to(torch.int64)
+        # ISO: week of the Thursday of this week
+        dow_mon0 = torch.remainder(days + 3, 7)  # 0 = Monday
+        thursday = days - dow_mon0 + 3
+        y, _, _ = _civil_from_days(thursday)
+        jan1 = _days_from_civil(y, torch.ones_like(y), torch.ones_like(y))
+        week = torch.div(thursday - jan1, 7, rounding_mode="floor") + 1
+        return Column(dtypes.int32, week.to(torch.int32), c.validity)
+
+
+# ----------------------------------------------------------------- strings
+@dataclass(eq=False)
+class Trim(Expr):
+    child: Expr
+    mode: str = "both"  # both | leading | trailing
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        n = len(c)
+        if n == 0:
+            return c
+        off = c.offsets.to(torch.int64)
+        lens = off[1:] - off[:-1]
+        W = int(lens.max().item()) if n else 0
+        if W == 0:
+            return c
+        padded = strings.to_padded(c, W)
+        pos = torch.arange(W, device=c.device).unsqueeze(0)
+        in_str = pos < lens.unsqueeze(1)
+        is_space = (padded == 32) & in_str
+        nonspace = in_str & ~is_space
+        any_ns = nonspace.any(dim=1)
+        first_ns = torch.argmax(nonspace.to(torch.int8), dim=1)
+        last_ns = W - 1 - torch.argmax(nonspace.flip(1).to(torch.int8), dim=1)
+        start = first_ns if self.mode in ("both", "leading") else torch.zeros_like(first_ns)
+        end = (last_ns + 1) if self.mode in ("both", "trailing") else lens
+        start = torch.where(any_ns, start, torch.zeros_like(start))
+        end = torch.where(any_ns, end, torch.zeros_like(end))
+        new_lens = (end - start).clamp(min=0)
+        new_off = torch.zeros(n + 1, dtype=torch.int64, device=c.device)
+        torch.cumsum(new_lens, 0, out=new_off[1:])
+        total = int(new_off[-1].item())
+        if total == 0:
+            data = torch.empty(0, dtype=torch.uint8, device=c.device)
+        else:
+            row = torch.repeat_interleave(new_lens)
+            p = torch.arange(total, dtype=torch.int64, device=c.device)
+            within = p - new_off[:-1][row]
+            src = off[:-1][row] + start[row] + within
+            data = c.data[src]
+        return Column(dtypes.string, data, c.validity, new_off.to(torch.int32))
+
+
+@dataclass(eq=False)
+class Left(Expr):
+    child: Expr
+    n: int
+
+    def eval(self, batch):
+        from .exprs import Substr
+
+        return Substr(self.child, 1, self.n).eval(batch)
+
+
+@dataclass(eq=False)
+class Right(Expr):
+    child: Expr
+    n: int
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        vals = c.to_pylist()  # cold path
+        out = [None if v is None else v[-self.n:] if self.n else "" for v in vals]
+        return Column.from_pylist(out, dtypes.string, str(c.device))
+
+
+@dataclass(eq=False)
+class Replace(Expr):
+    child: Expr
+    search: str
+    replacement: str = ""
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        vals = c.to_pylist()  # cold path (regexless literal replace)
+        out = [None if v is None else v.replace(self.search, self.replacement)
+               for v in vals]
+        return Column.from_pylist(out, dtypes.string, str(c.device))
+
+
+@dataclass(eq=False)
+class LPad(Expr):
+    child: Expr
+    length: int
+    pad: str = " "
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        vals = c.to_pylist()
+        out = []
+        for v in vals:
+            if v is None:
+                out.append(None)
+            elif len(v) >= self.length:
+                out.append(v[:self.length])
+            else:
+                need = self.length - len(v)
+                p = (self.pad * need)[:need]
+                out.append(p + v)
+        return Column.from_pylist(out, dtypes.string, str(c.device))
+
+
+@dataclass(eq=False)
+class RPad(Expr):
+    child: Expr
+    length: int
+    pad: str = " "
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        vals = c.to_pylist()
+        out = []
+        for v in vals:
+            if v is None:
+                out.append(None)
+            elif len(v) >= self.length:
+                out.append(v[:self.length])
+            else:
+                need = self.length - len(v)
+                out.append(v + (self.pad * need)[:need])
+        return Column.from_pylist(out, dtypes.string, str(c.device))
+
+
+@dataclass(eq=False)
+class SplitPart(Expr):
+    child: Expr
+    delimiter: str
+    part: int  # 1-based
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        vals = c.to_pylist()
+        out = []
+        for v in vals:
+            if v is None:
+                out.append(None)
+            else:
+                parts = v.split(self.delimiter)
+                out.append(parts[self.part - 1] if 0 < self.part <= len(parts) else "")
+        return Column.from_pylist(out, dtypes.string, str(c.device))
+
+
+@dataclass(eq=False)
+class Instr(Expr):
+    """1-based position of substr, 0 if absent (Spark instr)."""
+    child: Expr
+    sub: str
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        n = len(c)
+        pat = torch.tensor(list(self.sub.encode()), dtype=torch.uint8, device=c.device)
+        m = pat.numel()
+        lens = strings.lengths(c)
+        if m == 0:
+            return Column(dtypes.int32, torch.ones(n, dtype=torch.int32, device=c.device),
+                          c.validity)
+        W = int(lens.max().item()) if n else 0
+        if W < m:
+            return Column(dtypes.int32, torch.zeros(n, dtype=torch.int32, device=c.device),
+                          c.validity)
+        A = strings.to_padded(c, W)
+        win = A.unfold(1, m, 1)
+        hit = (win == pat).all(dim=2)
+        starts = torch.arange(hit.shape[1], device=c.device).unsqueeze(0)
+        valid_win = starts + m <= lens.unsqueeze(1)
+        hit = hit & valid_win
+        any_hit = hit.any(dim=1)
+        first = torch.argmax(hit.to(torch.int8), dim=1) + 1
+        data = torch.where(any_hit, first, torch.zeros_like(first)).to(torch.int32)
+        return Column(dtypes.int32, data, c.validity)
+
+
+@dataclass(eq=False)
+class Ascii(Expr):
+    child: Expr
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        lens = strings.lengths(c)
+        off = c.offsets.to(torch.int64)
+        first = c.data[off[:-1].clamp(max=max(int(c.data.numel()) - 1, 0))] \
+            if c.data.numel() else torch.zeros(len(c), dtype=torch.uint8, device=c.device)
+        data = torch.where(lens > 0, first.to(torch.int32),
+                           torch.zeros(len(c), dtype=torch.int32, device=c.device))
+        return Column(dtypes.int32, data, c.validity)
+
+
+# ------------------------------------------------------------- engine ids
+@dataclass(eq=False)
+class SparkPartitionId(Expr):
+    def eval(self, batch):
+        pid = EVAL_CONTEXT.get()["partition_id"]
+        return Column(dtypes.int32,
+                      torch.full((batch.num_rows,), pid, dtype=torch.int32,
+                                 device=batch.device))
+
+
+@dataclass(eq=False)
+class MonotonicallyIncreasingId(Expr):
+    """Spark layout: partition_id << 33 | row index within partition."""
+
+    def eval(self, batch):
+        ctx = EVAL_CONTEXT.get()
+        base = (ctx["partition_id"] << 33) + ctx.get("row_base", 0)
+        n = batch.num_rows
+        return Column(dtypes.int64,
+                      base + torch.arange(n, dtype=torch.int64, device=batch.device))

@@ -1,0 +1,151 @@
+"""Scalar function library tests (ext-functions parity).
+
+Each function is checked against plain Python semantics matching Spark
+(round half-away-from-zero, null-skipping greatest/least, ISO weekofyear,
+Spark dayofweek numbering, 1-based instr, etc.).
+"""
+import datetime
+
+import pytest
+import torch
+
+from auron_amd import dtypes, functions as F
+from auron_amd.column import Column, RecordBatch
+from auron_amd.exprs import col, lit
+
+
+def _b(**cols):
+    names, cs = [], []
+    for k, (vals, dt) in cols.items():
+        names.append(k)
+        cs.append(Column.from_pylist(vals, dt))
+    return RecordBatch(names, cs)
+
+
+def ev(expr, batch):
+    return expr.eval(batch).to_pylist()
+
+
+def test_round_half_away():
+    b = _b(x=([2.5, -2.5, 1.44, 1.45, None], dtypes.float64))
+    assert ev(F.Round(col("x")), b) == [3.0, -3.0, 1.0, 1.0, None]
+    r = ev(F.Round(col("x"), 1), b)
+    assert r[2] == 1.4 and abs(r[3] - 1.5) < 1e-9
+
+
+def test_floor_ceil_sign():
+    b = _b(x=([1.5, -1.5, 0.0, None], dtypes.float64))
+    assert ev(F.Floor(col("x")), b) == [1, -2, 0, None]
+    assert ev(F.Ceil(col("x")), b) == [2, -1, 0, None]
+    assert ev(F.Sign(col("x")), b) == [1.0, -1.0, 0.0, None]
+
+
+def test_exp_ln_log10_pow():
+    b = _b(x=([1.0, 0.0, -1.0, None], dtypes.float64))
+    r = ev(F.Exp(col("x")), b)
+    assert abs(r[0] - 2.718281828) < 1e-6
+    ln = ev(F.Ln(col("x")), b)
+    assert abs(ln[0]) < 1e-12 and ln[1] is None and ln[2] is None and ln[3] is None
+    lg = ev(F.Log10(lit(100.0)), _b(x=([0.0], dtypes.float64)))
+    assert abs(lg[0] - 2.0) < 1e-12
+    p = ev(F.Pow(col("x"), lit(2.0)), b)
+    assert p[:3] == [1.0, 0.0, 1.0] and p[3] is None
+
+
+def test_greatest_least_null_skipping():
+    b = _b(x=([1.0, None, None], dtypes.float64),
+           y=([3.0, 2.0, None], dtypes.float64))
+    assert ev(F.Greatest([col("x"), col("y")]), b) == [3.0, 2.0, None]
+    assert ev(F.Least([col("x"), col("y")]), b) == [1.0, 2.0, None]
+
+
+def test_nullif_nvl2_if():
+    b = _b(x=([1, 2, None], dtypes.int64), y=([1, 9, 9], dtypes.int64))
+    assert ev(F.NullIf(col("x"), col("y")), b) == [None, 2, None]
+    assert ev(F.Nvl2(col("x"), lit(10), lit(20)), b) == [10, 10, 20]
+    from auron_amd.exprs import Cmp
+    assert ev(F.If(Cmp(">", col("y"), lit(5)), lit(1), lit(0)), b) == [0, 1, 1]
+
+
+def _date(s):
+    return (datetime.date.fromisoformat(s) - datetime.date(1970, 1, 1)).days
+
+
+def _dcol(*isodates):
+    return ([None if s is None else _date(s) for s in isodates], dtypes.date32)
+
+
+def test_date_add_sub_diff():
+    b = _b(d=_dcol("2001-01-30", None), n=([3, 3], dtypes.int32))
+    assert ev(F.DateAdd(col("d"), col("n")), b) == [_date("2001-02-02"), None]
+    assert ev(F.DateSub(col("d"), col("n")), b) == [_date("2001-01-27"), None]
+    b2 = _b(a=_dcol("2001-03-01"), c=_dcol("2001-02-01"))
+    assert ev(F.DateDiff(col("a"), col("c")), b2) == [28]
+
+
+def test_add_months_clamps_day():
+    b = _b(d=_dcol("2001-01-31", "2000-01-31", "2001-11-15"))
+    r = ev(F.AddMonths(col("d"), 1), b)
+    assert r == [_date("2001-02-28"), _date("2000-02-29"), _date("2001-12-15")]
+    r2 = ev(F.AddMonths(col("d"), 14), b)
+    assert r2[0] == _date("2002-03-31")
+
+
+def test_last_day_quarter():
+    b = _b(d=_dcol("2000-02-10", "2001-02-10", "2001-12-01"))
+    assert ev(F.LastDay(col("d")), b) == [
+        _date("2000-02-29"), _date("2001-02-28"), _date("2001-12-31")]
+    assert ev(F.Quarter(col("d")), b) == [1, 1, 4]
+
+
+def test_dayofweek_weekofyear():
+    # 1970-01-01 Thursday → Spark dayofweek 5; 2001-01-01 Monday → 2
+    b = _b(d=_dcol("1970-01-01", "2001-01-01", "2001-01-07"))
+    assert ev(F.DayOfWeek(col("d")), b) == [5, 2, 1]
+    # ISO weeks: 2001-01-01 is week 1; 2000-01-01 (Saturday) is ISO week 52/1999;
+    # 2004-12-31 is week 53
+    b2 = _b(d=_dcol("2001-01-01", "2000-01-01", "2004-12-31"))
+    exp = [datetime.date.fromisoformat(s).isocalendar()[1]
+           for s in ("2001-01-01", "2000-01-01", "2004-12-31")]
+    assert ev(F.WeekOfYear(col("d")), b2) == exp
+
+
+def test_trim_family():
+    b = _b(s=(["  hi  ", "a", "   ", "", None, " x y "], dtypes.string))
+    assert ev(F.Trim(col("s")), b) == ["hi", "a", "", "", None, "x y"]
+    assert ev(F.Trim(col("s"), "leading"), b) == ["hi  ", "a", "", "", None, "x y "]
+    assert ev(F.Trim(col("s"), "trailing"), b) == ["  hi", "a", "", "", None, " x y"]
+
+
+def test_left_right_pad():
+    b = _b(s=(["hello", "ab", None], dtypes.string))
+    assert ev(F.Left(col("s"), 3), b) == ["hel", "ab", None]
+    assert ev(F.Right(col("s"), 3), b) == ["llo", "ab", None]
+    assert ev(F.LPad(col("s"), 7, "*"), b) == ["**hello", "*****ab", None]
+    assert ev(F.RPad(col("s"), 7, "xy"), b) == ["helloxy", "abxyxyx", None]
+    assert ev(F.LPad(col("s"), 3), b) == ["hel", " ab", None]
+
+
+def test_replace_split_part():
+    b = _b(s=(["a,b,c", "nope", None], dtypes.string))
+    assert ev(F.Replace(col("s"), ",", "-"), b) == ["a-b-c", "nope", None]
+    assert ev(F.SplitPart(col("s"), ",", 2), b) == ["b", "", None]
+    assert ev(F.SplitPart(col("s"), ",", 9), b) == ["", "", None]
+
+
+def test_instr_ascii():
+    b = _b(s=(["hello", "xyz", "", None, "ohell"], dtypes.string))
+    assert ev(F.Instr(col("s"), "ell"), b) == [2, 0, 0, None, 3]
+    assert ev(F.Instr(col("s"), ""), b) == [1, 1, 1, None, 1]
+    assert ev(F.Ascii(col("s")), b) == [104, 120, 0, None, 111]
+
+
+def test_partition_ids():
+    b = _b(x=([1, 2, 3], dtypes.int64))
+    tok = F.EVAL_CONTEXT.set({"partition_id": 5, "row_base": 10})
+    try:
+        assert ev(F.SparkPartitionId(), b) == [5, 5, 5]
+        base = (5 << 33) + 10
+        assert ev(F.MonotonicallyIncreasingId(), b) == [base, base + 1, base + 2]
+    finally:
+        F.EVAL_CONTEXT.reset(tok)
