@@ -507,6 +507,32 @@ class Ascii(Expr):
         return Column(dtypes.int32, data, c.validity)
 
 
+# ------------------------------------------------------------- sketches
+@dataclass(eq=False)
+class XxHash64(Expr):
+    """Spark xxhash64(...) — seed-chained over args, nulls skipped."""
+    args: List[Expr]
+    seed: int = 42
+
+    def eval(self, batch):
+        from . import sketch
+
+        cols = [a.eval(batch) for a in self.args]
+        return Column(dtypes.int64, sketch.xxhash64(cols, self.seed))
+
+
+@dataclass(eq=False)
+class BloomFilterMightContain(Expr):
+    """Runtime-filter probe: true iff the long key may be in the filter."""
+    bloom: object  # sketch.BloomFilter
+    child: Expr
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        mask = self.bloom.might_contain_longs(c.data.to(torch.int64))
+        return Column(dtypes.bool_, mask, c.validity)
+
+
 # ------------------------------------------------------------- engine ids
 @dataclass(eq=False)
 class SparkPartitionId(Expr):
