@@ -371,10 +371,121 @@ class Executor:
         return [b.gather(perm)]
 
     def _exec_SortMergeJoin(self, node: P.SortMergeJoin) -> List[RecordBatch]:
-        # lowered to hash join until the GPU merge-path kernel lands
-        hj = P.HashJoin(node.left, node.right, node.left_keys, node.right_keys,
-                        how=node.how, build_side="right", broadcast=False)
-        return self._exec_HashJoin(hj)
+        """Order-based equi-join (sort_merge_join_exec.rs analogue).
+
+        Both sides' valid-key rows are sorted TOGETHER (side as the last
+        tie-breaker, so within each equal-key run left rows precede right
+        rows); runs of equal keys are detected by adjacent comparison and
+        each run's left block is cross-producted with its right rows via
+        cumsum/repeat_interleave — no hash table anywhere. Null keys never
+        match (SQL semantics) and bypass the merge entirely."""
+        from ..exprs import col as _col
+
+        left = _concat(self.execute(node.left))
+        right = _concat(self.execute(node.right))
+        lkeys = [k.eval(left) for k in node.left_keys]
+        rkeys = [k.eval(right) for k in node.right_keys]
+        lkeys, rkeys = _normalize_join_keys(lkeys, rkeys)
+        device = left.device
+        nl_rows, nr_rows = left.num_rows, right.num_rows
+        how = node.how
+
+        def _valid(keys, n):
+            v = torch.ones(n, dtype=torch.bool, device=device)
+            for c in keys:
+                if c.validity is not None:
+                    v &= c.validity
+            return v
+
+        lval = _valid(lkeys, nl_rows)
+        rval = _valid(rkeys, nr_rows)
+        lidx = torch.nonzero(lval, as_tuple=False).flatten()
+        ridx = torch.nonzero(rval, as_tuple=False).flatten()
+
+        knames = [f"__k{i}" for i in range(len(lkeys))]
+        comb = RecordBatch(
+            knames + ["__side", "__row"],
+            [Column.concat([lk.gather(lidx), rk.gather(ridx)])
+             for lk, rk in zip(lkeys, rkeys)]
+            + [Column(dtypes.int8, torch.cat([
+                torch.zeros(lidx.numel(), dtype=torch.int8, device=device),
+                torch.ones(ridx.numel(), dtype=torch.int8, device=device)])),
+               Column(dtypes.int64, torch.cat([lidx, ridx]))])
+        n = comb.num_rows
+        if n:
+            skeys = [(_col(k), True) for k in knames] + [(_col("__side"), True)]
+            perm = self._sort_permutation(comb, skeys)
+            sc = comb.gather(perm)
+            side = sc.columns[-2].data
+            row = sc.columns[-1].data
+            first = torch.ones(n, dtype=torch.bool, device=device)
+            if n > 1:
+                same = torch.ones(n - 1, dtype=torch.bool, device=device)
+                for i in range(len(knames)):
+                    same &= self._col_eq_adjacent(sc.columns[i])
+                first[1:] = ~same
+            seg = torch.cumsum(first.to(torch.int64), 0) - 1
+            nseg = int(seg[-1].item()) + 1
+            seg_start = torch.nonzero(first, as_tuple=False).flatten()
+            is_l = side == 0
+            is_r = ~is_l
+            nl_seg = torch.bincount(seg[is_l], minlength=nseg)
+            nr_seg = torch.bincount(seg[is_r], minlength=nseg)
+        else:
+            row = side = seg = seg_start = None
+            nl_seg = nr_seg = torch.zeros(0, dtype=torch.int64, device=device)
+
+        if how in ("semi", "anti", "existence"):
+            lmatched = torch.zeros(nl_rows, dtype=torch.bool, device=device)
+            if n:
+                lpos = torch.nonzero(is_l, as_tuple=False).flatten()
+                lmatched[row[lpos]] = nr_seg[seg[lpos]] > 0
+            if how == "semi":
+                return [left.filter(lmatched)]
+            if how == "anti":
+                return [left.filter(~lmatched)]
+            exists = Column(dtypes.bool_, lmatched)
+            return [RecordBatch(left.names + [node.existence_col],
+                                left.columns + [exists])]
+
+        if n:
+            rpos = torch.nonzero(is_r, as_tuple=False).flatten()
+            cnt = nl_seg[seg[rpos]]
+            total = int(cnt.sum().item())
+            ri = torch.repeat_interleave(row[rpos], cnt)
+            off = torch.cumsum(cnt, 0) - cnt
+            within = torch.arange(total, dtype=torch.int64, device=device) \
+                - torch.repeat_interleave(off, cnt)
+            # left rows of a run sit at seg_start .. seg_start+nl_seg-1
+            lsortpos = torch.repeat_interleave(seg_start[seg[rpos]], cnt) + within
+            li = row[lsortpos]
+        else:
+            li = ri = torch.zeros(0, dtype=torch.int64, device=device)
+
+        preserved = {"inner": set(), "left": {"left"}, "right": {"right"},
+                     "full": {"left", "right"}}[how]
+        if "left" in preserved:
+            lmatched = torch.zeros(nl_rows, dtype=torch.bool, device=device)
+            if n:
+                lpos = torch.nonzero(is_l, as_tuple=False).flatten()
+                lmatched[row[lpos]] = nr_seg[seg[lpos]] > 0
+            un = torch.nonzero(~lmatched, as_tuple=False).flatten()
+            li = torch.cat([li, un])
+            ri = torch.cat([ri, torch.full((un.numel(),), -1, dtype=torch.int64,
+                                           device=device)])
+        if "right" in preserved:
+            rmatched = torch.zeros(nr_rows, dtype=torch.bool, device=device)
+            if n:
+                rpos2 = torch.nonzero(is_r, as_tuple=False).flatten()
+                rmatched[row[rpos2]] = nl_seg[seg[rpos2]] > 0
+            un = torch.nonzero(~rmatched, as_tuple=False).flatten()
+            ri = torch.cat([ri, un])
+            li = torch.cat([li, torch.full((un.numel(),), -1, dtype=torch.int64,
+                                           device=device)])
+        out_left = left.gather(li)
+        out_right = right.gather(ri)
+        return [RecordBatch(out_left.names + out_right.names,
+                            out_left.columns + out_right.columns)]
 
     # ------------------------------------------------------------- exchange
     def _exec_Exchange(self, node: P.Exchange) -> List[RecordBatch]:

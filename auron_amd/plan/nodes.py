@@ -94,13 +94,15 @@ class HashJoin(PlanNode):
 
 @dataclass(eq=False)
 class SortMergeJoin(PlanNode):
-    """Declared for plan parity; lowered to HashJoin until the merge-path
-    kernel lands (sort_merge_join_exec.rs analogue)."""
+    """Order-based equi-join (sort_merge_join_exec.rs analogue): both
+    sides sorted together, equal-key runs cross-producted — no hash
+    table. Same output contract as HashJoin."""
     left: PlanNode
     right: PlanNode
     left_keys: List[Expr]
     right_keys: List[Expr]
     how: str = "inner"
+    existence_col: str = "exists"
 
     def children(self):
         return [self.left, self.right]
