@@ -791,9 +791,26 @@ class Executor:
             elif wf.fn in ("sum", "avg", "count", "min", "max"):
                 val = wf.arg.eval(sb)
                 if node.order_by and n:
-                    # Spark's default frame with ORDER BY: rows unbounded
+                    # Spark's default frame with ORDER BY: unbounded
                     # preceding .. current row (running aggregate)
                     out = self._running_window(wf.fn, val, seg, seg_start)
+                    if node.frame == "range":
+                        # RANGE frame: peer rows (equal order keys within the
+                        # partition) share the value at the LAST peer row
+                        okeys = [k.eval(sb) for k, _ in node.order_by]
+                        peer_first = torch.ones(n, dtype=torch.bool, device=device)
+                        if n > 1:
+                            sp = torch.ones(n - 1, dtype=torch.bool, device=device)
+                            for c in okeys:
+                                sp = sp & self._col_eq_adjacent(c)
+                            peer_first[1:] = ~sp
+                        peer_first[seg_start] = True
+                        pg = torch.cumsum(peer_first.to(torch.int64), 0) - 1
+                        pstart = torch.nonzero(peer_first, as_tuple=False).flatten()
+                        pend = torch.cat([pstart[1:] - 1,
+                                          torch.tensor([n - 1], dtype=torch.int64,
+                                                       device=device)])
+                        out = out.gather(pend[pg])
                 else:
                     acc, cnt = ops.agg_scatter(seg, max(nseg, 1), val, wf.fn if wf.fn != "count" else "count")
                     fin = self._finalize_agg(AggFunc(wf.fn, None, name=al.name), val.dtype, acc, cnt)

@@ -173,11 +173,16 @@ class Expand(PlanNode):
 
 @dataclass(eq=False)
 class Window(PlanNode):
-    """Window functions over (partition_by, order_by) (window_exec.rs)."""
+    """Window functions over (partition_by, order_by) (window_exec.rs).
+
+    `frame` applies to running aggregates when order_by is present:
+    "range" (Spark's default — UNBOUNDED PRECEDING..CURRENT ROW with
+    peer rows sharing the frame end) or "rows"."""
     child: PlanNode
     partition_by: List[Expr]
     order_by: List[Tuple[Expr, bool]]
     functions: List[Aliased]  # WindowFunc exprs aliased to output names
+    frame: str = "range"
 
     def children(self):
         return [self.child]

@@ -328,3 +328,21 @@ def test_sort_merge_join_matches_hash_join_random(how):
     smj = s.collect(P.SortMergeJoin(lscan, rscan, keys_l, keys_r, how=how)).to_pydict()
     hj = s.collect(P.HashJoin(lscan, rscan, keys_l, keys_r, how=how)).to_pydict()
     assert_rows_equal(smj, hj)
+
+
+def test_window_range_vs_rows_frame_ties():
+    """Default RANGE frame: peer rows (tied order keys) share the running
+    value at the last peer; ROWS frame counts each row individually."""
+    from auron_amd.exprs import WindowFunc, Aliased
+
+    data = {"p": [1, 1, 1, 1], "o": [10, 20, 20, 30], "v": [1.0, 2.0, 3.0, 4.0]}
+    t = {"p": dtypes.int64, "o": dtypes.int64, "v": dtypes.float64}
+    scan_n = P.MemoryScan([RecordBatch.from_pydict(data, t)])
+    fns = [Aliased(WindowFunc("sum", col("v")), "rs")]
+    s = AuronSession()
+    rng = s.collect(P.Window(scan_n, [col("p")], [(col("o"), True)], fns,
+                             frame="range")).to_pydict()
+    assert rng["rs"] == [1.0, 6.0, 6.0, 10.0]
+    rows = s.collect(P.Window(scan_n, [col("p")], [(col("o"), True)], fns,
+                              frame="rows")).to_pydict()
+    assert rows["rs"] == [1.0, 3.0, 6.0, 10.0]
