@@ -154,8 +154,11 @@ class Executor:
             if native_cols:
                 got = parquet_native.read_columns_native(path, native_cols, self.ctx.device)
                 if got is not None:
-                    if host_cols:
-                        t = pq.read_table(path, columns=host_cols)
+                    # host path covers statically-unsupported columns AND any
+                    # that failed page-level parse (per-column fallback)
+                    missing = [c for c in columns if c not in got]
+                    if missing:
+                        t = pq.read_table(path, columns=missing)
                         hb = RecordBatch.from_arrow(t, self.ctx.device)
                         for n, c in zip(hb.names, hb.columns):
                             got[n] = c
@@ -733,14 +736,21 @@ class Executor:
         b = _concat(self.execute(node.child))
         device = b.device
         n = b.num_rows
-        sort_keys = [(k, True) for k in node.partition_by] + list(node.order_by)
-        perm = self._sort_permutation(b, sort_keys) if sort_keys else torch.arange(n, dtype=torch.int64, device=device)
-        sb = b.gather(perm)
         if node.partition_by:
-            pcols = [k.eval(sb) for k in node.partition_by]
-            gids, reps = ops.group_ids(pcols)
-            # group ids are order-of-first-appearance per sorted order on CPU,
-            # arbitrary on GPU; normalize to segment ids over the sorted rows
+            # partition membership only needs GROUPING, not lexicographic
+            # order: map partition keys (often several strings) to hash
+            # group ids once and sort by the integer id — avoids per-pass
+            # host string-rank sorts that dominated q47/q57/q67
+            pcols = [k.eval(b) for k in node.partition_by]
+            gids0, _ = ops.group_ids(pcols)
+            tmp = RecordBatch(list(b.names) + ["__wgid"],
+                              list(b.columns) + [Column(dtypes.int64, gids0)])
+            skeys = [(Col("__wgid"), True)] + list(node.order_by)
+            perm = self._sort_permutation(tmp, skeys)
+            sb = b.gather(perm)
+            gids = gids0[perm]
+            # normalize arbitrary hash ids to dense segment ids over the
+            # sorted rows
             if n:
                 changed = torch.zeros(n, dtype=torch.int64, device=device)
                 changed[1:] = (gids[1:] != gids[:-1]).to(torch.int64)
@@ -749,6 +759,10 @@ class Executor:
                 seg = gids
             nseg = int(seg[-1].item()) + 1 if n else 0
         else:
+            perm = (self._sort_permutation(b, list(node.order_by))
+                    if node.order_by else
+                    torch.arange(n, dtype=torch.int64, device=device))
+            sb = b.gather(perm)
             seg = torch.zeros(n, dtype=torch.int64, device=device)
             nseg = 1 if n else 0
         seg_start = torch.zeros(max(nseg, 1), dtype=torch.int64, device=device)

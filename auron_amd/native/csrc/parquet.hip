@@ -125,6 +125,105 @@ AU_EXPORT int au_pq_rle1(const void* pages_dev, int npages, const void* buf,
   return (int)hipGetLastError();
 }
 
+// ------------------------------------------------------------------ RLE idx
+// Generalized RLE/bit-packed hybrid decode for DICTIONARY indices
+// (bit width 1..32). One workgroup per page; same two-phase structure as
+// k_pq_rle1. Descriptor reuses PqPage with repurposed fields:
+//   def_off=idx_off  def_len=idx_len  values_off=out_base (compact)
+//   n_values=nvalid  row_start=unused pad=bit_width
+// The staged buffer is padded by 8 bytes so the unaligned 8-byte loads at
+// a literal run's tail never read out of bounds.
+__global__ void k_pq_rle_idx(const PqPage* pages, const uint8_t* buf, int32_t* out) {
+  const PqPage p = pages[blockIdx.x];
+  const uint8_t* src = buf + p.def_off;
+  const int64_t src_len = p.def_len;
+  int32_t* dst = out + p.values_off;
+  const int64_t n = p.n_values;
+  const int bw = (int)p.pad;
+  const uint64_t mask = bw >= 64 ? ~0ull : ((1ull << bw) - 1);
+  const int vbytes = (bw + 7) >> 3;
+
+  __shared__ Run runs[RLE_WINDOW];
+  __shared__ int nruns;
+  __shared__ int64_t s_pos, s_i;
+
+  if (threadIdx.x == 0) {
+    s_pos = 0;
+    s_i = 0;
+  }
+  __syncthreads();
+
+  while (true) {
+    if (threadIdx.x == 0) {
+      int64_t pos = s_pos;
+      int64_t i = s_i;
+      int r = 0;
+      while (r < RLE_WINDOW && i < n && pos < src_len) {
+        uint64_t header = 0;
+        int shift = 0;
+        while (true) {
+          uint8_t b = src[pos++];
+          header |= (uint64_t)(b & 0x7F) << shift;
+          if (!(b & 0x80)) break;
+          shift += 7;
+        }
+        if (header & 1) {  // literal: (header>>1) groups of 8, bw bits each
+          int64_t ngroups = header >> 1;
+          int64_t nvals = ngroups * 8;
+          if (nvals > n - i) nvals = n - i;
+          runs[r].out_start = (int32_t)i;
+          runs[r].count = (int32_t)nvals;
+          runs[r].src_off = (int32_t)pos;
+          runs[r].rep_val = -1;
+          pos += ngroups * bw;  // bw bytes per group of 8
+          i += nvals;
+        } else {
+          int64_t cnt = header >> 1;
+          if (cnt > n - i) cnt = n - i;
+          uint32_t v = 0;
+          for (int b2 = 0; b2 < vbytes; b2++) v |= (uint32_t)src[pos + b2] << (8 * b2);
+          runs[r].out_start = (int32_t)i;
+          runs[r].count = (int32_t)cnt;
+          runs[r].rep_val = (int32_t)(v & mask);
+          pos += vbytes;
+          i += cnt;
+        }
+        r++;
+      }
+      nruns = r;
+      s_pos = pos;
+      s_i = i;
+    }
+    __syncthreads();
+    int count = nruns;
+    if (count == 0) break;
+    for (int r = threadIdx.x / 64; r < count; r += blockDim.x / 64) {
+      const Run run = runs[r];
+      int lane = threadIdx.x & 63;
+      if (run.rep_val >= 0) {
+        for (int j = lane; j < run.count; j += 64) dst[run.out_start + j] = run.rep_val;
+      } else {
+        for (int j = lane; j < run.count; j += 64) {
+          int64_t bitpos = (int64_t)j * bw;
+          uint64_t w;
+          __builtin_memcpy(&w, src + run.src_off + (bitpos >> 3), 8);
+          dst[run.out_start + j] = (int32_t)((w >> (bitpos & 7)) & mask);
+        }
+      }
+    }
+    __syncthreads();
+    if (count < RLE_WINDOW) break;
+  }
+}
+
+AU_EXPORT int au_pq_rle_idx(const void* pages_dev, int npages, const void* buf,
+                            int32_t* out, void* stream) {
+  if (npages == 0) return 0;
+  hipLaunchKernelGGL(k_pq_rle_idx, dim3(npages), dim3(256), 0, (hipStream_t)stream,
+                     (const PqPage*)pages_dev, (const uint8_t*)buf, out);
+  return (int)hipGetLastError();
+}
+
 // ----------------------------------------------------------------- scatter
 // out[row] = valid(row) ? values[prefix[row]-1 - page_base] : 0
 // prefix = inclusive cumsum of validity over the whole chunk (int64).

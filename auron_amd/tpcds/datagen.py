@@ -43,7 +43,11 @@ _DAY_NAMES = ["Monday", "Tuesday", "Wednesday", "Thursday", "Friday",
 
 
 def _seed(table: str, part: int) -> int:
-    return abs(hash((table, part, "auron-tpcds-v1"))) % (2 ** 31)
+    # stable across processes (builtin hash() is randomized per process,
+    # which would make regenerated datasets irreproducible)
+    import zlib
+
+    return zlib.crc32(f"auron-tpcds-v1:{table}:{part}".encode()) % (2 ** 31)
 
 
 def _rng(table: str, part: int) -> np.random.Generator:
@@ -514,7 +518,7 @@ def _nparts_for(table: str, sf: float) -> int:
     return 1
 
 
-DATAGEN_VERSION = 9
+DATAGEN_VERSION = 10
 
 
 def dataset_root(root: str, sf: float) -> str:
@@ -547,14 +551,18 @@ def write_dataset(root: str, sf: float, tables: Optional[List[str]] = None,
             if job % world == rank:
                 if force or not os.path.exists(path):
                     tbl = generate_table(t, sf, p, nparts)
-                    # uncompressed PLAIN pages: the GPU parquet decoder
-                    # (parquet_native.py + csrc/parquet.hip) consumes numeric
-                    # page bytes directly in HBM — no host decode at all;
-                    # string columns fall back to the pyarrow host path
+                    # uncompressed pages: the GPU parquet decoder
+                    # (parquet_native.py + csrc/parquet.hip) consumes page
+                    # bytes directly in HBM — numerics as PLAIN, strings as
+                    # RLE_DICTIONARY (indices decoded + bytes gathered on
+                    # device); no host decode at all
+                    str_cols = [f.name for f in tbl.schema
+                                if f.type == __import__("pyarrow").string()]
                     pq.write_table(tbl, path, compression="NONE",
-                                   use_dictionary=False,
+                                   use_dictionary=str_cols,
                                    data_page_version="1.0",
                                    store_decimal_as_integer=True,
+                                   dictionary_pagesize_limit=1 << 26,
                                    row_group_size=1 << 20)
             job += 1
     return base

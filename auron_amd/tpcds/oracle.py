@@ -749,14 +749,16 @@ def q62(root, sf):
     j = _merge(j, web, "ws_web_site_sk", "web_site_sk")
     j["wname20"] = j.w_warehouse_name.str[:20]
     lag = j.ws_ship_date_sk - j.ws_sold_date_sk
-    j["d30"] = ((lag <= 30)).astype("float").where(lag.notna())
-    j["d31_60"] = ((lag > 30) & (lag <= 60)).astype("float").where(lag.notna())
-    j["d61_90"] = ((lag > 60) & (lag <= 90)).astype("float").where(lag.notna())
-    j["d91_120"] = ((lag > 90) & (lag <= 120)).astype("float").where(lag.notna())
-    j["d120p"] = ((lag > 120)).astype("float").where(lag.notna())
+    # CASE WHEN <null cond> THEN 1 ELSE 0 yields 0 in SQL, so null-lag
+    # rows contribute 0 (NaN comparisons are already False in pandas)
+    j["d30"] = (lag <= 30).astype("float")
+    j["d31_60"] = ((lag > 30) & (lag <= 60)).astype("float")
+    j["d61_90"] = ((lag > 60) & (lag <= 90)).astype("float")
+    j["d91_120"] = ((lag > 90) & (lag <= 120)).astype("float")
+    j["d120p"] = (lag > 120).astype("float")
     cols = ["d30", "d31_60", "d61_90", "d91_120", "d120p"]
     g = j.groupby(["wname20", "sm_type", "web_name"], dropna=False)[cols] \
-         .sum(min_count=1).reset_index()
+         .sum().reset_index()
     for c in cols:
         g[c] = g[c].astype("Int64")
     g = g.sort_values(["wname20", "sm_type", "web_name"], na_position="first").head(100)
@@ -1953,7 +1955,10 @@ def q59(root, sf):
     out = {"s_store_name1": m.s_store_name1, "s_store_id1": m.s_store_id,
            "d_week_seq1": m.d_week_seq1}
     for t, _ in days:
-        out[f"{t}_ratio"] = m[f"{t}_sales1"] / m[f"{t}_sales2"]
+        # decimal division by zero is NULL in Spark (non-ANSI), not inf
+        r = m[f"{t}_sales1"] / m[f"{t}_sales2"]
+        import numpy as _np
+        out[f"{t}_ratio"] = r.replace([_np.inf, -_np.inf], _np.nan)
     import pandas as pd
 
     # full (pre-LIMIT) result: the ORDER BY keys tie across store_sk pairs
@@ -2852,6 +2857,10 @@ def q78(root, sf):
     csc = cte("catalog_sales", "cs", "catalog_returns", "cr", "cs_bill_customer_sk",
               "cs_order_number", "cr_order_number")
     ssc = ssc[ssc.d_year == 2000]
+    # SQL equality never matches NULL customer keys; pandas merge matches
+    # NaN==NaN, so strip null-key rows from the right sides
+    wsc = wsc[wsc["ws_bill_customer_sk"].notna()]
+    csc = csc[csc["cs_bill_customer_sk"].notna()]
     j = ssc.merge(wsc.rename(columns={"ws_item_sk": "ss_item_sk",
                                       "ws_bill_customer_sk": "ss_customer_sk"}),
                   on=["d_year", "ss_item_sk", "ss_customer_sk"], how="left",

@@ -96,3 +96,74 @@ def test_gpu_decode_large_random(tmp_path):
             assert gv is None and rv is None
         else:
             assert torch.equal(gv, rv)
+
+
+@pytest.fixture(scope="module")
+def dict_file(tmp_path_factory):
+    """Dictionary-encoded file: strings + low-cardinality numerics,
+    small pages so chunks have many RLE index pages."""
+    rng = np.random.default_rng(3)
+    n = 80_000
+    words = ["alpha", "beta", "gamma", "", "a-much-longer-dictionary-entry",
+             "x" * 40] + [f"w{i}" for i in range(500)]
+    d = {
+        "s": pa.array([words[i] for i in rng.integers(0, len(words), n)],
+                      mask=rng.random(n) < 0.04),
+        "snn": pa.array([words[i] for i in rng.integers(0, len(words), n)]),
+        "di32": pa.array(rng.integers(0, 300, n).astype(np.int32),
+                         mask=rng.random(n) < 0.05),
+        "di64": pa.array(rng.integers(0, 1000, n) * 10**9,
+                         mask=rng.random(n) < 0.03),
+        "df64": pa.array(np.round(rng.normal(0, 10, n), 1)),
+    }
+    t = pa.table(d)
+    p = tmp_path_factory.mktemp("pqd") / "dict.parquet"
+    pq.write_table(t, str(p), compression="NONE", use_dictionary=True,
+                   data_page_version="1.0", row_group_size=16384,
+                   data_page_size=4096, dictionary_pagesize_limit=1 << 24)
+    return str(p), t
+
+
+def test_np_dict_decoder(dict_file):
+    path, t = dict_file
+    cols = parquet_native.read_columns_native(path, t.schema.names, "cpu",
+                                              _np_only=True)
+    assert cols is not None and set(cols) == set(t.schema.names), \
+        "dictionary fast path rejected"
+    _check(cols, t)
+
+
+def test_plain_byte_array_falls_back(tmp_path):
+    t = pa.table({"s": pa.array(["aa", "bb"] * 100)})
+    p = tmp_path / "pba.parquet"
+    pq.write_table(t, str(p), compression="NONE", use_dictionary=False,
+                   data_page_version="1.0")
+    assert parquet_native.read_columns_native(str(p), ["s"], "cpu",
+                                              _np_only=True) is None
+
+
+def test_partial_column_fallback(tmp_path):
+    """One dict column decodes natively while a PLAIN string column is
+    dropped for the host path — no wholesale failure."""
+    n = 5000
+    rng = np.random.default_rng(5)
+    t = pa.table({
+        "num": pa.array(rng.integers(0, 50, n)),
+        "ps": pa.array([f"unique-{i}" for i in range(n)]),
+    })
+    p = str(tmp_path / "mix.parquet")
+    pq.write_table(t, p, compression="NONE", data_page_version="1.0",
+                   use_dictionary=["num"])
+    cols = parquet_native.read_columns_native(p, ["num", "ps"], "cpu",
+                                              _np_only=True)
+    assert cols is not None and "num" in cols and "ps" not in cols
+    _check({"num": cols["num"]}, t.select(["num"]))
+
+
+@pytest.mark.gpu
+def test_gpu_dict_decode(dict_file):
+    path, t = dict_file
+    cols = parquet_native.read_columns_native(path, t.schema.names, "cuda:0")
+    assert cols is not None and set(cols) == set(t.schema.names)
+    assert all(c.data.is_cuda for c in cols.values())
+    _check({k: v.to("cpu") for k, v in cols.items()}, t)

@@ -22,11 +22,13 @@ def dataset():
 
 
 def rows_of(df):
+    import pandas as pd
+
     out = []
     for _, r in df.iterrows():
         row = []
         for v in r:
-            if isinstance(v, float) and math.isnan(v):
+            if v is pd.NA or (isinstance(v, float) and math.isnan(v)):
                 row.append(None)
             elif isinstance(v, (np.integer,)):
                 row.append(int(v))
