@@ -38,6 +38,7 @@ class ExecContext:
     batch_rows: int = 1 << 22
     metrics: Dict[str, float] = field(default_factory=dict)
     memmgr: object = None  # auron_amd.memory.MemManager
+    shuffle_seq: int = 0  # deterministic stage ids for persistent shuffles
 
     def __post_init__(self):
         if self.memmgr is None:
@@ -492,9 +493,12 @@ class Executor:
 
     # ------------------------------------------------------------- exchange
     def _exec_Exchange(self, node: P.Exchange) -> List[RecordBatch]:
+        from .. import shuffle
+
         bs = self.execute(node.child)
         W = self.ctx.world_size
-        if W == 1:
+        persist = shuffle.persist_enabled(node.persist)
+        if W == 1 and not persist:
             return bs
         b = _concat(bs)
         device = self.ctx.device
@@ -518,7 +522,17 @@ class Executor:
                 for d in range(W):
                     dest.append(reordered.slice(pos, int(cl[d])))
                     pos += int(cl[d])
-        received = all_to_all(dest, device, self.ctx.group)
+        if persist:
+            sid = str(self.ctx.shuffle_seq)
+            self.ctx.shuffle_seq += 1
+            w = shuffle.ShuffleWriter(shuffle.shuffle_root(), sid, self.ctx.rank)
+            w.write([d.to("cpu") for d in dest])
+            if W > 1 and dist.is_initialized():
+                dist.barrier(group=self.ctx.group)
+            received = shuffle.ShuffleReader(
+                shuffle.shuffle_root(), sid).read_partition(self.ctx.rank, device)
+        else:
+            received = all_to_all(dest, device, self.ctx.group)
         return [_concat(received)] if received else [_empty_like(b)]
 
     def _exec_Broadcast(self, node: P.Broadcast) -> List[RecordBatch]:

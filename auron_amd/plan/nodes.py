@@ -134,10 +134,14 @@ class Exchange(PlanNode):
     all-to-all), single (gather everything to rank 0), roundrobin.
 
     Reference analogue: shuffle_writer_exec.rs + ipc_reader_exec.rs; here
-    the write+read pair collapses into one in-flight collective."""
+    the write+read pair collapses into one in-flight collective.
+    `persist=True` (or AURON_SHUFFLE_PERSIST=1) routes the exchange
+    through durable shuffle files instead (auron_amd.shuffle), giving
+    the reference's stage-retry contract."""
     child: PlanNode
     kind: str = "hash"  # hash | single | roundrobin
     keys: List[Expr] = field(default_factory=list)
+    persist: bool = False
 
     def children(self):
         return [self.child]
