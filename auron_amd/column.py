@@ -277,7 +277,8 @@ class Column:
 
 
 class RecordBatch:
-    __slots__ = ("names", "columns")
+    # _eval_memo: optional CSE scope opened by exprs.eval_scope
+    __slots__ = ("names", "columns", "_eval_memo")
 
     def __init__(self, names: List[str], columns: List[Column]):
         assert len(names) == len(columns)

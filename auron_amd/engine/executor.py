@@ -258,9 +258,12 @@ class Executor:
 
     # ------------------------------------------------------- row operators
     def _exec_Filter(self, node: P.Filter) -> List[RecordBatch]:
+        from ..exprs import eval_scope
+
         out = []
         for b in self.execute(node.child):
-            c = node.predicate.eval(b)
+            with eval_scope(b):
+                c = node.predicate.eval(b)
             mask = c.data.bool()
             if c.validity is not None:
                 mask = mask & c.validity
@@ -268,9 +271,12 @@ class Executor:
         return out
 
     def _exec_Project(self, node: P.Project) -> List[RecordBatch]:
+        from ..exprs import eval_scope
+
         out = []
         for b in self.execute(node.child):
-            cols = [a.expr.eval(b) for a in node.exprs]
+            with eval_scope(b):
+                cols = [a.expr.eval(b) for a in node.exprs]
             out.append(RecordBatch([a.name for a in node.exprs], cols))
         return out
 
@@ -563,7 +569,13 @@ class Executor:
 
     # -------------------------------------------------------------- hash agg
     def _exec_HashAgg(self, node: P.HashAgg) -> List[RecordBatch]:
+        from ..exprs import eval_scope
+
         b = _concat(self.execute(node.child))
+        with eval_scope(b):
+            return self._hash_agg_body(node, b)
+
+    def _hash_agg_body(self, node: P.HashAgg, b: RecordBatch) -> List[RecordBatch]:
         device = b.device
         n = b.num_rows
         if node.mode == "final":
@@ -801,6 +813,7 @@ class Executor:
             seg_start = torch.nonzero(first_mask, as_tuple=False).flatten()
         pos_in_seg = torch.arange(n, dtype=torch.int64, device=device) - seg_start[seg] if n else torch.zeros(0, dtype=torch.int64, device=device)
 
+        sb._eval_memo = {}  # CSE scope; sb is operator-local, dies with it
         names = list(sb.names)
         cols = list(sb.columns)
         for al in node.functions:

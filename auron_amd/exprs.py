@@ -27,6 +27,35 @@ from .dtypes import DataType
 
 
 class Expr:
+    """Base expression. Subclasses implement eval(batch) -> Column.
+
+    CSE: every subclass's eval is wrapped with an identity-keyed memo
+    (CachedExprsEvaluator analogue, datafusion-ext-commons cached_exprs_
+    evaluator): when an operator opens a memo scope on the batch
+    (`eval_scope`), shared subtree OBJECTS evaluate once per batch."""
+
+    def __init_subclass__(cls, **kw):
+        super().__init_subclass__(**kw)
+        inner = cls.__dict__.get("eval")
+        if inner is None or getattr(inner, "_cse_wrapped", False):
+            return
+
+        def eval(self, batch, _inner=inner):
+            memo = getattr(batch, "_eval_memo", None)
+            if memo is None:
+                return _inner(self, batch)
+            key = id(self)
+            hit = memo.get(key)
+            # the memo pins the expr object so its id can't be recycled by
+            # a new expr allocated during the same scope
+            if hit is None or hit[0] is not self:
+                hit = (self, _inner(self, batch))
+                memo[key] = hit
+            return hit[1]
+
+        eval._cse_wrapped = True
+        cls.eval = eval
+
     def eval(self, batch: RecordBatch) -> Column:
         raise NotImplementedError
 
@@ -96,6 +125,22 @@ class Expr:
 
     def substr(self, start: int, length: int):
         return Substr(self, start, length)
+
+
+import contextlib
+
+
+@contextlib.contextmanager
+def eval_scope(batch):
+    """Open a CSE memo on `batch`: shared Expr objects evaluate once."""
+    had = hasattr(batch, "_eval_memo")
+    if not had:
+        batch._eval_memo = {}
+    try:
+        yield batch
+    finally:
+        if not had:
+            del batch._eval_memo
 
 
 @dataclass(eq=False)

@@ -346,3 +346,27 @@ def test_window_range_vs_rows_frame_ties():
     rows = s.collect(P.Window(scan_n, [col("p")], [(col("o"), True)], fns,
                               frame="rows")).to_pydict()
     assert rows["rs"] == [1.0, 3.0, 6.0, 10.0]
+
+
+def test_cse_shared_subtree_evaluated_once():
+    """A subtree OBJECT shared between project exprs runs once per batch
+    (CachedExprsEvaluator parity)."""
+    from auron_amd import exprs as E
+
+    calls = {"n": 0}
+
+    class Probe(E.Expr):
+        def eval(self, batch):
+            calls["n"] += 1
+            return col("x").eval(batch)
+
+    p = Probe()
+    shared = p + lit(1)
+    plan = P.Project(scan(), [exprs.Aliased(shared + lit(1), "a"),
+                              exprs.Aliased(shared + lit(2), "b"),
+                              exprs.Aliased(p, "c")])
+    s = AuronSession()
+    out = s.collect(plan).to_pydict()
+    assert calls["n"] == 1
+    assert out["a"] == [None if v is None else v + 2 for v in out["c"]]
+    assert out["b"] == [None if v is None else v + 3 for v in out["c"]]
