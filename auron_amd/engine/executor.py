@@ -104,6 +104,9 @@ class Executor:
         # same (paths, columns) many times for independent subaggregates;
         # decode once per query (cleared at each top-level execute)
         self._scan_cache: Dict[tuple, List[RecordBatch]] = {}
+        # MetricNode tree of the last top-level execute (SQLMetrics parity)
+        self._metric_stack: List[list] = [[]]
+        self.last_metric_tree: Optional[dict] = None
 
     # ------------------------------------------------------------- dispatch
     def _rewrite(self, node: P.PlanNode) -> P.PlanNode:
@@ -125,19 +128,34 @@ class Executor:
         ctx = F.EVAL_CONTEXT.get()
         if ctx.get("partition_id") != self.ctx.rank:
             F.EVAL_CONTEXT.set({**ctx, "partition_id": self.ctx.rank})
-        if len(self._child_time) == 1:  # top-level call = one query
+        top = len(self._child_time) == 1
+        if top:  # top-level call = one query
             self._scan_cache.clear()
         node = self._rewrite(node)
         name = type(node).__name__
         fn = getattr(self, f"_exec_{name}", None)
         if fn is None:
             raise NotImplementedError(f"operator {name}")
+        # MetricNode tree (SQLMetrics analogue): one record per executed
+        # operator, child records nested in execution order
+        rec = {"op": name, "time_s": 0.0, "rows": 0, "batches": 0, "children": []}
+        self._metric_stack.append(rec["children"])
         t0 = time.perf_counter()
         self._child_time.append(0.0)
-        out = fn(node)
-        dt = time.perf_counter() - t0
-        child = self._child_time.pop()
-        self._child_time[-1] += dt
+        try:
+            out = fn(node)
+        finally:
+            dt = time.perf_counter() - t0
+            child = self._child_time.pop()
+            self._child_time[-1] += dt
+            self._metric_stack.pop()
+        rec["time_s"] = dt - child
+        rec["rows"] = sum(b.num_rows for b in out)
+        rec["batches"] = len(out)
+        if top:
+            self.last_metric_tree = rec
+        else:
+            self._metric_stack[-1].append(rec)
         m = self.ctx.metrics
         m[f"op.{name}"] = m.get(f"op.{name}", 0.0) + (dt - child)
         assert out, f"{name} returned no batches"

@@ -62,6 +62,29 @@ class AuronSession:
     def metrics(self):
         return dict(self.ctx.metrics)
 
+    def metric_tree(self) -> Optional[dict]:
+        """MetricNode tree of the last collect/execute (SQLMetrics
+        analogue): {op, time_s, rows, batches, children}."""
+        return self.executor.last_metric_tree
+
+    def explain_metrics(self) -> str:
+        """Render the last metric tree like the reference's MetricNode
+        display (operator, exclusive wall, output rows/batches)."""
+        tree = self.metric_tree()
+        if tree is None:
+            return "<no query executed>"
+        lines = []
+
+        def walk(rec, depth):
+            lines.append(f"{'  ' * depth}{rec['op']}: "
+                         f"{rec['time_s'] * 1e3:.2f} ms, "
+                         f"rows={rec['rows']}, batches={rec['batches']}")
+            for c in rec["children"]:
+                walk(c, depth + 1)
+
+        walk(tree, 0)
+        return "\n".join(lines)
+
 
 def init_distributed(backend: Optional[str] = None) -> ExecContext:
     """Initialize torch.distributed from torchrun env (RCCL on GPU)."""

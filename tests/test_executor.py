@@ -370,3 +370,16 @@ def test_cse_shared_subtree_evaluated_once():
     assert calls["n"] == 1
     assert out["a"] == [None if v is None else v + 2 for v in out["c"]]
     assert out["b"] == [None if v is None else v + 3 for v in out["c"]]
+
+
+def test_metric_tree():
+    """MetricNode tree parity: nested per-operator records with rows."""
+    s = AuronSession()
+    plan = P.Limit(P.Filter(scan(), col("x") > 10), 3)
+    out = s.collect(plan)
+    t = s.metric_tree()
+    assert t["op"] == "Limit" and t["rows"] == out.num_rows
+    assert t["children"][0]["op"] == "Filter"
+    assert t["children"][0]["children"][0]["op"] == "MemoryScan"
+    assert t["children"][0]["children"][0]["rows"] == len(DATA["x"])
+    assert "Limit" in s.explain_metrics()
