@@ -16,13 +16,18 @@ exchange, so a reader can be re-run (stage retry) without the writers.
 
 File format:
   data file:  concat of per-dest segments, each = pack_batch payload
-  index file: msgpack {"version", "world", "offsets": [w+1], "metas": [w]}
-              (offsets[d]..offsets[d+1] is dest d's byte range; metas[d]
-              is the pack_batch meta dict or None for an empty dest)
+              (zlib-compressed when codec="zlib" — the ipc_compression.rs
+              role; in-flight xGMI exchange stays raw on purpose, disk is
+              where a codec pays)
+  index file: msgpack {"version", "world", "codec", "offsets": [w+1],
+              "metas": [w]} (offsets[d]..offsets[d+1] is dest d's byte
+              range; metas[d] is the pack_batch meta dict or None for an
+              empty dest)
 """
 from __future__ import annotations
 
 import os
+import zlib
 from typing import List, Optional
 
 import msgpack
@@ -32,6 +37,10 @@ from .column import RecordBatch
 from .exchange import pack_batch, unpack_batch
 
 _VERSION = 1
+
+
+def _codec() -> str:
+    return os.environ.get("AURON_SHUFFLE_CODEC", "none")
 
 
 def _stage_dir(root: str, stage_id: str) -> str:
@@ -49,6 +58,7 @@ class ShuffleWriter:
         self._index_path = os.path.join(self.dir, f"map-{map_rank}.index")
 
     def write(self, batches_by_dest: List[Optional[RecordBatch]]) -> None:
+        codec = _codec()
         offsets = [0]
         metas = []
         tmp_data = self._data_path + ".tmp"
@@ -60,11 +70,13 @@ class ShuffleWriter:
                     continue
                 meta, buf = pack_batch(b, "cpu")
                 raw = buf.numpy().tobytes()
+                if codec == "zlib":
+                    raw = zlib.compress(raw, 1)
                 f.write(raw)
                 metas.append(meta)
                 offsets.append(offsets[-1] + len(raw))
         index = {"version": _VERSION, "world": len(batches_by_dest),
-                 "offsets": offsets, "metas": metas}
+                 "codec": codec, "offsets": offsets, "metas": metas}
         tmp_idx = self._index_path + ".tmp"
         with open(tmp_idx, "wb") as f:
             f.write(msgpack.packb(index, use_bin_type=True))
@@ -100,6 +112,8 @@ class ShuffleReader:
             with open(os.path.join(self.dir, f"map-{m}.data"), "rb") as f:
                 f.seek(lo)
                 raw = f.read(hi - lo)
+            if index.get("codec", "none") == "zlib":
+                raw = zlib.decompress(raw)
             buf = torch.frombuffer(bytearray(raw), dtype=torch.uint8)
             b = unpack_batch(meta, buf)
             out.append(b.to(device) if str(device) != "cpu" else b)

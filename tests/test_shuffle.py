@@ -105,3 +105,17 @@ def test_exchange_persist_world2(tmp_path):
     for p in procs:
         p.join(timeout=30)
     assert merged == {1: 4.0, 2: 12.0, 4: 54.0, None: 40.0}
+
+
+def test_shuffle_zlib_codec(tmp_path, monkeypatch):
+    monkeypatch.setenv("AURON_SHUFFLE_CODEC", "zlib")
+    b0 = _batch(([1, 2, None] * 100, ["x", None, "zz"] * 100))
+    ShuffleWriter(str(tmp_path), "c1", 0).write([b0, b0])
+    r = ShuffleReader(str(tmp_path), "c1")
+    got = r.read_partition(1)
+    assert len(got) == 1 and got[0].to_pydict() == b0.to_pydict()
+    raw = os.path.getsize(os.path.join(str(tmp_path), "stage-c1", "map-0.data"))
+    monkeypatch.delenv("AURON_SHUFFLE_CODEC")
+    ShuffleWriter(str(tmp_path), "c2", 0).write([b0, b0])
+    plain = os.path.getsize(os.path.join(str(tmp_path), "stage-c2", "map-0.data"))
+    assert raw < plain  # codec actually engaged
