@@ -121,9 +121,11 @@ class BatchHolder:
 
 class MemManager:
     """Tracks resident bytes of registered holders against a budget and
-    spills least-recently-touched holders on pressure (the reference's
-    fair-share policy simplified to LRU-victim; lib.rs:365 divides the
-    budget by spillable count — with whole-holder spilling LRU dominates)."""
+    spills on pressure with the reference's fair-share policy
+    (lib.rs:365 divides the budget by spillable count): holders above
+    their fair share spill first, largest overage first, so one huge
+    consumer cannot evict many small recently-used ones; remaining
+    pressure falls back to LRU victims."""
 
     def __init__(self, budget_bytes: Optional[int] = None, fraction: float = 0.8,
                  spill_dir: Optional[str] = None):
@@ -157,15 +159,18 @@ class MemManager:
                 self._holders.remove(h)
 
     def reserve(self, nbytes: int, exclude: Optional[BatchHolder] = None):
-        """Ensure `nbytes` headroom, spilling LRU holders if needed."""
+        """Ensure `nbytes` headroom: fair-share victims first, then LRU."""
         with self._lock:
             used = self.resident_bytes()
             if used + nbytes <= self.budget:
                 return
-            victims = sorted((h for h in self._holders
-                              if h.resident and h is not exclude),
-                             key=lambda h: h.last_touch)
-            for v in victims:
+            cands = [h for h in self._holders if h.resident and h is not exclude]
+            fair = self.budget / max(len(cands), 1)
+            over = sorted((h for h in cands if h.bytes > fair),
+                          key=lambda h: h.bytes, reverse=True)
+            lru = sorted((h for h in cands if h.bytes <= fair),
+                         key=lambda h: h.last_touch)
+            for v in over + lru:
                 used -= v.spill()
                 self.metrics["spill_count"] = self.metrics.get("spill_count", 0) + 1
                 if used + nbytes <= self.budget:

@@ -61,3 +61,25 @@ def test_join_correct_under_tiny_budget(tmp_path):
     want = s2.collect(join)
     assert out.num_rows == want.num_rows
     assert mgr.metrics.get("spill_count", 0) >= 1
+
+
+def test_fair_share_spills_largest_over_share_first():
+    """One oversized consumer spills before small recently-idle ones."""
+    import torch
+    from auron_amd import dtypes
+    from auron_amd.column import Column, RecordBatch
+    from auron_amd.memory import MemManager
+
+    def mk(n):
+        return [RecordBatch(["x"], [Column(dtypes.int64, torch.arange(n))])]
+
+    mgr = MemManager(budget_bytes=100 * 8)
+    small1 = mgr.register("small1", mk(10))
+    small2 = mgr.register("small2", mk(10))
+    big = mgr.register("big", mk(80))  # exactly at budget together
+    # touch order makes the smalls the LRU victims; fair share (~33 rows)
+    # must still pick `big` when pressure arrives
+    small1.batches(); small2.batches(); big.batches()
+    mgr.register("new", mk(10))
+    assert not big.resident, "fair-share should spill the oversized holder"
+    assert small1.resident and small2.resident
