@@ -853,13 +853,8 @@ class Executor:
                         r = torch.cumsum(incr, 0)
                         out = Column(dtypes.int64, r - r[seg_start][seg] + 1)
                     else:
-                        # rank = (index of last order-key change) - seg_start + 1;
-                        # seg starts count as changes, so the cummax of change
-                        # positions never leaks across partitions
-                        gpos = torch.arange(n, dtype=torch.int64, device=device)
-                        change_pos = torch.where(same_as_prev, torch.full_like(gpos, -1), gpos)
-                        lcp = torch.cummax(change_pos, 0).values
-                        out = Column(dtypes.int64, lcp - seg_start[seg] + 1)
+                        out = Column(dtypes.int64, self._window_rank(
+                            node, sb, seg, seg_start, n, device))
                 else:
                     out = Column(dtypes.int64, torch.ones(n, dtype=torch.int64, device=device))
             elif wf.fn in ("sum", "avg", "count", "min", "max"):
@@ -871,20 +866,8 @@ class Executor:
                     if node.frame == "range":
                         # RANGE frame: peer rows (equal order keys within the
                         # partition) share the value at the LAST peer row
-                        okeys = [k.eval(sb) for k, _ in node.order_by]
-                        peer_first = torch.ones(n, dtype=torch.bool, device=device)
-                        if n > 1:
-                            sp = torch.ones(n - 1, dtype=torch.bool, device=device)
-                            for c in okeys:
-                                sp = sp & self._col_eq_adjacent(c)
-                            peer_first[1:] = ~sp
-                        peer_first[seg_start] = True
-                        pg = torch.cumsum(peer_first.to(torch.int64), 0) - 1
-                        pstart = torch.nonzero(peer_first, as_tuple=False).flatten()
-                        pend = torch.cat([pstart[1:] - 1,
-                                          torch.tensor([n - 1], dtype=torch.int64,
-                                                       device=device)])
-                        out = out.gather(pend[pg])
+                        out = out.gather(self._peer_end(node, sb, seg_start, n,
+                                                        device))
                 else:
                     acc, cnt = ops.agg_scatter(seg, max(nseg, 1), val, wf.fn if wf.fn != "count" else "count")
                     fin = self._finalize_agg(AggFunc(wf.fn, None, name=al.name), val.dtype, acc, cnt)
@@ -899,11 +882,87 @@ class Executor:
                 ok = ok & (seg[idx_c] == seg) if n else ok
                 gi = torch.where(ok, idx_c, torch.full_like(idx_c, -1))
                 out = val.gather(gi)
+            elif wf.fn in ("percent_rank", "cume_dist", "ntile"):
+                seg_last = (torch.cat([seg_start[1:] - 1,
+                                       torch.tensor([n - 1], dtype=torch.int64,
+                                                    device=device)])
+                            if n else seg_start)
+                seg_len = seg_last[seg] - seg_start[seg] + 1 if n else seg
+                if wf.fn == "percent_rank":
+                    rank = self._window_rank(node, sb, seg, seg_start, n, device)
+                    denom = (seg_len - 1).clamp(min=1).to(torch.float64)
+                    out = Column(dtypes.float64,
+                                 (rank - 1).to(torch.float64) / denom)
+                elif wf.fn == "cume_dist":
+                    pend = self._peer_end(node, sb, seg_start, n, device)
+                    out = Column(dtypes.float64,
+                                 (pend - seg_start[seg] + 1).to(torch.float64)
+                                 / seg_len.to(torch.float64))
+                else:  # ntile(k): first (len%k) buckets get one extra row
+                    k = wf.offset if wf.offset else 1
+                    base = torch.div(seg_len, k, rounding_mode="floor")
+                    rem = seg_len - base * k
+                    cut = rem * (base + 1)
+                    lo = torch.div(pos_in_seg, (base + 1).clamp(min=1),
+                                   rounding_mode="floor") + 1
+                    hi = rem + torch.div((pos_in_seg - cut), base.clamp(min=1),
+                                         rounding_mode="floor") + 1
+                    out = Column(dtypes.int64,
+                                 torch.where(pos_in_seg < cut, lo, hi))
+            elif wf.fn in ("first_value", "last_value", "nth_value"):
+                val = wf.arg.eval(sb)
+                if wf.fn == "first_value":
+                    out = val.gather(seg_start[seg]) if n else val
+                elif wf.fn == "last_value":
+                    # frame unbounded preceding..current: ROWS -> current row;
+                    # RANGE -> last peer row
+                    if node.frame == "range" and node.order_by and n:
+                        out = val.gather(self._peer_end(node, sb, seg_start, n,
+                                                        device))
+                    else:
+                        out = val
+                else:
+                    k = wf.offset if wf.offset else 1
+                    tgt = seg_start[seg] + (k - 1) if n else seg
+                    ok = pos_in_seg >= (k - 1)
+                    out = val.gather(torch.where(ok, tgt, torch.full_like(tgt, -1)))
             else:
                 raise NotImplementedError(f"window fn {wf.fn}")
             names.append(al.name)
             cols.append(out)
         return [RecordBatch(names, cols)]
+
+    def _window_rank(self, node, sb, seg, seg_start, n, device):
+        """SQL rank(): 1 + index distance from the last order-key change."""
+        if not (node.order_by and n):
+            return torch.ones(n, dtype=torch.int64, device=device)
+        okeys = [k.eval(sb) for k, _ in node.order_by]
+        same_as_prev = torch.zeros(n, dtype=torch.bool, device=device)
+        if n > 1:
+            sp = torch.ones(n - 1, dtype=torch.bool, device=device)
+            for c in okeys:
+                sp = sp & self._col_eq_adjacent(c)
+            same_as_prev[1:] = sp
+        same_as_prev[seg_start] = False
+        gpos = torch.arange(n, dtype=torch.int64, device=device)
+        change_pos = torch.where(same_as_prev, torch.full_like(gpos, -1), gpos)
+        lcp = torch.cummax(change_pos, 0).values
+        return lcp - seg_start[seg] + 1
+
+    def _peer_end(self, node, sb, seg_start, n, device):
+        """Per row: index of its last peer (equal order keys in partition)."""
+        peer_first = torch.ones(n, dtype=torch.bool, device=device)
+        if node.order_by and n > 1:
+            sp = torch.ones(n - 1, dtype=torch.bool, device=device)
+            for k, _ in node.order_by:
+                sp = sp & self._col_eq_adjacent(k.eval(sb))
+            peer_first[1:] = ~sp
+        peer_first[seg_start] = True
+        pg = torch.cumsum(peer_first.to(torch.int64), 0) - 1
+        pstart = torch.nonzero(peer_first, as_tuple=False).flatten()
+        pend = torch.cat([pstart[1:] - 1,
+                          torch.tensor([n - 1], dtype=torch.int64, device=device)])
+        return pend[pg]
 
     def _running_window(self, fn: str, val: Column, seg: torch.Tensor,
                         seg_start: torch.Tensor) -> Column:

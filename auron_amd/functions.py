@@ -507,6 +507,125 @@ class Ascii(Expr):
         return Column(dtypes.int32, data, c.validity)
 
 
+@dataclass(eq=False)
+class Bround(Expr):
+    """Spark bround: half-even (banker's) rounding."""
+    child: Expr
+    ndigits: int = 0
+
+    def eval(self, batch):
+        c = _f64(self.child, batch)
+        f = 10.0 ** self.ndigits
+        # torch.round is round-half-to-even
+        return Column(dtypes.float64, torch.round(c.data * f) / f, c.validity)
+
+
+@dataclass(eq=False)
+class IsNan(Expr):
+    child: Expr
+
+    def eval(self, batch):
+        c = _f64(self.child, batch)
+        nan = torch.isnan(c.data)
+        if c.validity is not None:
+            nan = nan & c.validity  # null is not NaN
+        return Column(dtypes.bool_, nan)
+
+
+@dataclass(eq=False)
+class NormalizeNanAndZero(Expr):
+    """Spark NormalizeNaNAndZero: -0.0 -> 0.0, all NaNs -> one NaN."""
+    child: Expr
+
+    def eval(self, batch):
+        c = _f64(self.child, batch)
+        d = torch.where(c.data == 0, torch.zeros_like(c.data), c.data)
+        d = torch.where(torch.isnan(d), torch.full_like(d, float("nan")), d)
+        return Column(dtypes.float64, d, c.validity)
+
+
+@dataclass(eq=False)
+class InitCap(Expr):
+    """First letter of each word upper-cased, rest lowered (host path)."""
+    child: Expr
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        out = [None if v is None else " ".join(
+            w[:1].upper() + w[1:].lower() if w else w for w in v.split(" "))
+            for v in c.to_pylist()]
+        return Column.from_pylist(out, dtypes.string, str(c.device))
+
+
+@dataclass(eq=False)
+class CryptoHash(Expr):
+    """md5 / sha1 / sha2-N hex digests (host path, hashlib)."""
+    child: Expr
+    algo: str = "md5"  # md5 | sha1 | sha224 | sha256 | sha384 | sha512
+
+    def eval(self, batch):
+        import hashlib
+
+        c = self.child.eval(batch)
+        fn = getattr(hashlib, self.algo)
+        out = [None if v is None else fn(v.encode()).hexdigest()
+               for v in c.to_pylist()]
+        return Column.from_pylist(out, dtypes.string, str(c.device))
+
+
+def _json_path_get(obj, path: str):
+    """Spark get_json_object path subset: $.a.b[0].c"""
+    if not path.startswith("$"):
+        return None
+    import re as _re
+
+    cur = obj
+    for tok in _re.findall(r"\.([^.\[\]]+)|\[(\d+)\]", path[1:]):
+        name, idx = tok
+        if name:
+            if not isinstance(cur, dict) or name not in cur:
+                return None
+            cur = cur[name]
+        else:
+            i = int(idx)
+            if not isinstance(cur, list) or i >= len(cur):
+                return None
+            cur = cur[i]
+    return cur
+
+
+@dataclass(eq=False)
+class GetJsonObject(Expr):
+    """Spark get_json_object (host path; spark_get_json_object.rs role)."""
+    child: Expr
+    path: str
+
+    def eval(self, batch):
+        import json
+
+        c = self.child.eval(batch)
+        out = []
+        for v in c.to_pylist():
+            if v is None:
+                out.append(None)
+                continue
+            try:
+                r = _json_path_get(json.loads(v), self.path)
+            except (ValueError, TypeError):
+                r = None
+            if r is None:
+                out.append(None)
+            elif isinstance(r, str):
+                out.append(r)
+            elif isinstance(r, bool):
+                out.append("true" if r else "false")
+            elif isinstance(r, (dict, list)):
+                out.append(json.dumps(r, separators=(",", ":")))
+            else:
+                out.append(str(r))
+        return Column.from_pylist(out, dtypes.string, str(c.device))
+
+
 # ------------------------------------------------------------- sketches
 @dataclass(eq=False)
 class XxHash64(Expr):

@@ -54,7 +54,7 @@ def _try_load():
     lib.au_join_fill.argtypes = [c, c, ctypes.c_int, i64, c, i64, c, c, c, c, c, c, ctypes.c_int, c]
     lib.au_pmod.argtypes = [c, i64, i32, c, c]
     lib.au_agg_scatter.argtypes = [c, i64, c, c, ctypes.c_int, ctypes.c_int,
-                                   ctypes.c_int, c, c, c]
+                                   ctypes.c_int, c, c, i64, c]
     lib.au_agg_count.argtypes = [c, i64, c, c, c]
     for f in ("au_agg_scatter", "au_agg_count"):
         getattr(lib, f).restype = ctypes.c_int

@@ -90,6 +90,99 @@ __global__ void k_agg_scatter(const int64_t* gids, int64_t n, const uint8_t* val
   }
 }
 
+// ------------------------------------------------------- LDS-staged path
+// For small group counts the global-atomic scatter serializes on hot
+// accumulator addresses (every lane of every CU hits the same few
+// cachelines). Stage per-workgroup partial accumulators in LDS (shared
+// atomics are per-CU, no cross-XCD traffic), then flush once per group
+// per block. Selected by the host when ngroups fits in 64 KB of LDS and
+// n is large enough to amortize the flush.
+__device__ __forceinline__ void lds_min_f64(unsigned long long* a, double v) {
+  unsigned long long old = *a, assumed;
+  do {
+    assumed = old;
+    if (v >= __longlong_as_double((long long)assumed)) return;
+    old = atomicCAS(a, assumed, (unsigned long long)__double_as_longlong(v));
+  } while (old != assumed);
+}
+
+__device__ __forceinline__ void lds_max_f64(unsigned long long* a, double v) {
+  unsigned long long old = *a, assumed;
+  do {
+    assumed = old;
+    if (v <= __longlong_as_double((long long)assumed)) return;
+    old = atomicCAS(a, assumed, (unsigned long long)__double_as_longlong(v));
+  } while (old != assumed);
+}
+
+__device__ __forceinline__ void lds_min_i64(unsigned long long* a, int64_t v) {
+  unsigned long long old = *a, assumed;
+  do {
+    assumed = old;
+    if (v >= (int64_t)assumed) return;
+    old = atomicCAS(a, assumed, (unsigned long long)v);
+  } while (old != assumed);
+}
+
+__device__ __forceinline__ void lds_max_i64(unsigned long long* a, int64_t v) {
+  unsigned long long old = *a, assumed;
+  do {
+    assumed = old;
+    if (v <= (int64_t)assumed) return;
+    old = atomicCAS(a, assumed, (unsigned long long)v);
+  } while (old != assumed);
+}
+
+template <typename VT, bool ACC_F64>
+__global__ void k_agg_scatter_lds(const int64_t* gids, int64_t n,
+                                  const uint8_t* validity, const VT* values,
+                                  int op, void* acc, int64_t* counts,
+                                  int ngroups, unsigned long long init_bits) {
+  extern __shared__ unsigned long long smem[];
+  unsigned long long* sacc = smem;
+  unsigned long long* scnt = counts ? smem + ngroups : nullptr;
+  for (int i = threadIdx.x; i < ngroups; i += blockDim.x) {
+    sacc[i] = init_bits;
+    if (scnt) scnt[i] = 0;
+  }
+  __syncthreads();
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    if (validity && !validity[i]) continue;
+    int g = (int)gids[i];
+    if (scnt) atomicAdd(&scnt[g], 1ull);
+    if (ACC_F64) {
+      double v = (double)values[i];
+      if (op == 0) atomicAdd((double*)&sacc[g], v);
+      else if (op == 1) lds_min_f64(&sacc[g], v);
+      else lds_max_f64(&sacc[g], v);
+    } else {
+      int64_t v = (int64_t)values[i];
+      if (op == 0) atomicAdd(&sacc[g], (unsigned long long)v);
+      else if (op == 1) lds_min_i64(&sacc[g], v);
+      else lds_max_i64(&sacc[g], v);
+    }
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < ngroups; i += blockDim.x) {
+    if (scnt && scnt[i]) atomicAdd((unsigned long long*)&counts[i], scnt[i]);
+    if (sacc[i] == init_bits) continue;  // untouched (or identity: no-op)
+    if (ACC_F64) {
+      double v = __longlong_as_double((long long)sacc[i]);
+      double* a = (double*)acc;
+      if (op == 0) atomicAdd(&a[i], v);
+      else if (op == 1) atomic_min_f64(&a[i], v);
+      else atomic_max_f64(&a[i], v);
+    } else {
+      int64_t v = (int64_t)sacc[i];
+      int64_t* a = (int64_t*)acc;
+      if (op == 0) atomicAdd((unsigned long long*)&a[i], (unsigned long long)v);
+      else if (op == 1) atomic_min_i64(&a[i], v);
+      else atomic_max_i64(&a[i], v);
+    }
+  }
+}
+
 // count-only (count_star / count of a validity-masked column)
 __global__ void k_agg_count(const int64_t* gids, int64_t n, const uint8_t* validity,
                             int64_t* counts) {
@@ -102,9 +195,49 @@ __global__ void k_agg_count(const int64_t* gids, int64_t n, const uint8_t* valid
 
 AU_EXPORT int au_agg_scatter(const int64_t* gids, int64_t n, const uint8_t* validity,
                              const void* values, int vtype, int op, int acc_f64,
-                             void* acc, int64_t* counts, void* stream) {
+                             void* acc, int64_t* counts, int64_t ngroups,
+                             void* stream) {
   if (n == 0) return 0;
   hipStream_t s = (hipStream_t)stream;
+  // LDS staging: accumulators (+counts) must fit in 64 KB and the row
+  // count must amortize the per-block flush (grid*ngroups global atomics)
+  int slots = counts ? 2 : 1;
+  bool use_lds = ngroups > 0 && ngroups * slots * 8 <= (64 << 10) &&
+                 n >= 16 * ngroups;
+  if (use_lds) {
+    unsigned long long init;
+    if (op == 0) init = 0ull;  // sum identity (0.0 and 0 share the bit pattern)
+    else if (acc_f64) {
+      double d = op == 1 ? INFINITY : -INFINITY;
+      unsigned long long bits;
+      __builtin_memcpy(&bits, &d, 8);  // host-side double_as_longlong
+      init = bits;
+    } else {
+      init = (unsigned long long)(op == 1 ? INT64_MAX : INT64_MIN);
+    }
+    int64_t g64 = (n + 255) / 256;
+    int64_t cap = n / (16 * ngroups) + 1;  // keep flush << accumulate
+    if (g64 > cap) g64 = cap;
+    if (g64 > 2048) g64 = 2048;
+    if (g64 < 1) g64 = 1;
+    dim3 g((int)g64), b(256);
+    size_t smem = (size_t)ngroups * slots * 8;
+#define CASE_LDS(VT)                                                              \
+    if (acc_f64)                                                                  \
+      hipLaunchKernelGGL((k_agg_scatter_lds<VT, true>), g, b, smem, s, gids, n,   \
+                         validity, (const VT*)values, op, acc, counts,            \
+                         (int)ngroups, init);                                     \
+    else                                                                          \
+      hipLaunchKernelGGL((k_agg_scatter_lds<VT, false>), g, b, smem, s, gids, n,  \
+                         validity, (const VT*)values, op, acc, counts,            \
+                         (int)ngroups, init);
+    if (vtype == 0) { CASE_LDS(double) }
+    else if (vtype == 1) { CASE_LDS(int64_t) }
+    else if (vtype == 2) { CASE_LDS(int32_t) }
+    else { CASE_LDS(float) }
+#undef CASE_LDS
+    return (int)hipGetLastError();
+  }
   dim3 g(agg_grid(n)), b(256);
 #define CASE(VT)                                                                  \
   if (acc_f64)                                                                    \

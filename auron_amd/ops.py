@@ -439,7 +439,8 @@ def _agg_scatter_native(gids: torch.Tensor, num_groups: int, values: Column, fn:
     acc = torch.full((num_groups,), init,
                      dtype=torch.float64 if acc_f64 else torch.int64, device=device)
     rc = lib.au_agg_scatter(g.data_ptr(), n, vptr, v.data_ptr(), vtype, op,
-                            1 if acc_f64 else 0, acc.data_ptr(), counts.data_ptr(), sp)
+                            1 if acc_f64 else 0, acc.data_ptr(), counts.data_ptr(),
+                            num_groups, sp)
     native.check(rc, "au_agg_scatter")
     del keep
     if fn in ("min", "max"):

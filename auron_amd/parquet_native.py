@@ -601,10 +601,11 @@ def read_columns_native(path: str, columns: List[str], device,
             if use_gpu:
                 if ck.is_dict:
                     data_t, valid_t = _decode_chunk_gpu_dict(
-                        dbuf, buf, ck, nvals, cm.phys, device)
+                        dbuf, buf, ck, nvals, cm.phys, device, chunk_nulls)
                 else:
                     data_t, valid_t = _decode_chunk_gpu(dbuf, ck.pages, nvals,
-                                                        cm.phys, device)
+                                                        cm.phys, device,
+                                                        chunk_nulls)
             else:
                 if ck.is_dict:
                     data_np, valid_np, _ = decode_chunk_np_dict(buf, ck, nvals,
@@ -637,12 +638,14 @@ def read_columns_native(path: str, columns: List[str], device,
     return out
 
 
-def _gpu_validity_prefix(dbuf, pages, num_values, device):
-    """Decode def levels on device; returns (validity uint8 | None, prefix)."""
+def _gpu_validity_prefix(dbuf, pages, num_values, device, chunk_nulls=True):
+    """Decode def levels on device; returns (validity uint8 | None, prefix).
+
+    Null-free chunks (stats null_count == 0) skip the decode."""
     from . import native
 
     lib = native.require()
-    has_def = pages[0].def_off >= 0
+    has_def = pages[0].def_off >= 0 and chunk_nulls
     if not has_def:
         return None, None
     npages = len(pages)
@@ -697,9 +700,10 @@ def _gpu_dict_indices(dbuf, buf, ck: ChunkPages, prefix, num_values, device):
 
 
 def _decode_chunk_gpu_dict(dbuf, buf, ck: ChunkPages, num_values: int,
-                           phys: str, device):
+                           phys: str, device, chunk_nulls=True):
     """Dictionary-encoded fixed-width chunk on device."""
-    validity, prefix = _gpu_validity_prefix(dbuf, ck.pages, num_values, device)
+    validity, prefix = _gpu_validity_prefix(dbuf, ck.pages, num_values, device,
+                                            chunk_nulls)
     idx = _gpu_dict_indices(dbuf, buf, ck, prefix, num_values, device)
     tdt = {"INT32": torch.int32, "INT64": torch.int64,
            "FLOAT": torch.float32, "DOUBLE": torch.float64}[phys]
@@ -725,7 +729,8 @@ def _decode_chunk_strings(buf, dbuf, ck: ChunkPages, num_values: int, device,
             vt = torch.from_numpy(validity.astype(bool))
         return Column(dtypes.string, torch.from_numpy(data), vt,
                       torch.from_numpy(offsets))
-    validity, prefix = _gpu_validity_prefix(dbuf, ck.pages, num_values, device)
+    validity, prefix = _gpu_validity_prefix(dbuf, ck.pages, num_values, device,
+                                            chunk_nulls)
     idx = _gpu_dict_indices(dbuf, buf, ck, prefix, num_values, device)
     parse_dict_strings(buf, ck)
     dict_offs = torch.from_numpy(ck.dict_str_offs).to(device)
@@ -756,7 +761,7 @@ def _decode_chunk_strings(buf, dbuf, ck: ChunkPages, num_values: int, device,
 
 
 def _decode_chunk_gpu(dbuf: torch.Tensor, pages: List[PageDesc], num_values: int,
-                      phys: str, device):
+                      phys: str, device, chunk_nulls: bool = True):
     """HIP kernel path: RLE def-levels + PLAIN value scatter on device."""
     from . import native
 
@@ -764,7 +769,11 @@ def _decode_chunk_gpu(dbuf: torch.Tensor, pages: List[PageDesc], num_values: int
     import ctypes
 
     esize = _PHYS_NP[phys]().itemsize
-    has_def = pages[0].def_off >= 0
+    has_def = pages[0].def_off >= 0 and chunk_nulls
+    # null-free chunks (column statistics say null_count == 0) skip the
+    # def-level decode entirely: all values present -> dense PLAIN copy.
+    # This removes the single hottest kernel of the suite (k_pq_rle1 was
+    # 41% of GPU time; see profiles/).
     npages = len(pages)
     # descriptor layout (int64 x 6): def_off, def_len, values_off, n, row_start, pad
     arr = np.zeros((npages, 6), dtype=np.int64)

@@ -518,7 +518,7 @@ def _nparts_for(table: str, sf: float) -> int:
     return 1
 
 
-DATAGEN_VERSION = 10
+DATAGEN_VERSION = 11
 
 
 def dataset_root(root: str, sf: float) -> str:
@@ -563,6 +563,7 @@ def write_dataset(root: str, sf: float, tables: Optional[List[str]] = None,
                                    data_page_version="1.0",
                                    store_decimal_as_integer=True,
                                    dictionary_pagesize_limit=1 << 26,
+                                   data_page_size=64 << 10,
                                    row_group_size=1 << 20)
             job += 1
     return base

@@ -116,7 +116,7 @@ def main():
             "higher_is_better": False,
             "scaling": "strong",
             "vs_baseline": None,
-            "dtype": "fp64",
+            "dtype": "decimal64/fp64",
             "data": "synthetic",
             "config": {
                 "model": f"tpcds-{len(qnames)}q",

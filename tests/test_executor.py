@@ -383,3 +383,45 @@ def test_metric_tree():
     assert t["children"][0]["children"][0]["op"] == "MemoryScan"
     assert t["children"][0]["children"][0]["rows"] == len(DATA["x"])
     assert "Limit" in s.explain_metrics()
+
+
+def test_window_percent_rank_cume_dist_ntile():
+    from auron_amd.exprs import Aliased, WindowFunc
+
+    data = {"p": [1, 1, 1, 1, 2, 2], "o": [10, 20, 20, 30, 5, 5],
+            "v": [1.0, 2.0, 3.0, 4.0, 5.0, 6.0]}
+    t = {"p": dtypes.int64, "o": dtypes.int64, "v": dtypes.float64}
+    sc = P.MemoryScan([RecordBatch.from_pydict(data, t)])
+    fns = [Aliased(WindowFunc("percent_rank", None), "pr"),
+           Aliased(WindowFunc("cume_dist", None), "cd"),
+           Aliased(WindowFunc("ntile", None, 2), "nt"),
+           Aliased(WindowFunc("rank", None), "rk")]
+    s = AuronSession()
+    out = s.collect(P.Window(sc, [col("p")], [(col("o"), True)], fns)).to_pydict()
+    rows = sorted(zip(out["p"], out["o"], out["pr"], out["cd"], out["nt"], out["rk"]))
+    # partition 1: ranks 1,2,2,4 over 4 rows
+    assert rows[0][2:] == (0.0, 0.25, 1, 1)
+    assert rows[1][2] == pytest.approx(1 / 3) and rows[1][3] == 0.75 and rows[1][5] == 2
+    assert rows[2][2] == pytest.approx(1 / 3) and rows[2][3] == 0.75 and rows[2][4] == 2
+    assert rows[3][2:] == (1.0, 1.0, 2, 4)
+    # partition 2: two tied rows
+    assert rows[4][2:] == (0.0, 1.0, 1, 1)
+    assert rows[5][2:] == (0.0, 1.0, 2, 1)
+
+
+def test_window_first_last_nth_value():
+    from auron_amd.exprs import Aliased, WindowFunc
+
+    data = {"p": [1, 1, 1, 2], "o": [1, 2, 3, 1], "v": [10.0, 20.0, 30.0, 40.0]}
+    t = {"p": dtypes.int64, "o": dtypes.int64, "v": dtypes.float64}
+    sc = P.MemoryScan([RecordBatch.from_pydict(data, t)])
+    fns = [Aliased(WindowFunc("first_value", col("v")), "fv"),
+           Aliased(WindowFunc("last_value", col("v")), "lv"),
+           Aliased(WindowFunc("nth_value", col("v"), 2), "nv")]
+    s = AuronSession()
+    out = s.collect(P.Window(sc, [col("p")], [(col("o"), True)], fns)).to_pydict()
+    rows = sorted(zip(out["p"], out["o"], out["fv"], out["lv"], out["nv"]))
+    assert rows[0][2:] == (10.0, 10.0, None)   # p1 o1
+    assert rows[1][2:] == (10.0, 20.0, 20.0)   # p1 o2
+    assert rows[2][2:] == (10.0, 30.0, 20.0)   # p1 o3
+    assert rows[3][2:] == (40.0, 40.0, None)   # p2 o1

@@ -149,3 +149,43 @@ def test_partition_ids():
         assert ev(F.MonotonicallyIncreasingId(), b) == [base, base + 1, base + 2]
     finally:
         F.EVAL_CONTEXT.reset(tok)
+
+
+def test_bround_half_even():
+    b = _b(x=([2.5, 3.5, -2.5, 1.25, None], dtypes.float64))
+    assert ev(F.Bround(col("x")), b) == [2.0, 4.0, -2.0, 1.0, None]
+    assert ev(F.Bround(col("x"), 1), b) == [2.5, 3.5, -2.5, 1.2, None]
+
+
+def test_isnan_normalize():
+    b = _b(x=([float("nan"), 1.0, None], dtypes.float64))
+    assert ev(F.IsNan(col("x")), b) == [True, False, False]
+    import torch
+    from auron_amd.column import Column
+    neg = Column(dtypes.float64, torch.tensor([-0.0], dtype=torch.float64))
+    nb = RecordBatch(["x"], [neg])
+    r = F.NormalizeNanAndZero(col("x")).eval(nb)
+    import math
+    assert math.copysign(1.0, r.data[0].item()) == 1.0
+
+
+def test_initcap_crypto():
+    b = _b(s=(["hello world", "aBC dEF", "", None], dtypes.string))
+    assert ev(F.InitCap(col("s")), b) == ["Hello World", "Abc Def", "", None]
+    import hashlib
+    got = ev(F.CryptoHash(col("s"), "md5"), b)
+    assert got[0] == hashlib.md5(b"hello world").hexdigest() and got[3] is None
+    got = ev(F.CryptoHash(col("s"), "sha256"), b)
+    assert got[1] == hashlib.sha256(b"aBC dEF").hexdigest()
+
+
+def test_get_json_object():
+    docs = ['{"a": {"b": [1, 2, {"c": "x"}]}, "n": null, "t": true}',
+            'not json', None, '{"a": 1.5}']
+    b = _b(j=(docs, dtypes.string))
+    assert ev(F.GetJsonObject(col("j"), "$.a.b[2].c"), b) == ["x", None, None, None]
+    assert ev(F.GetJsonObject(col("j"), "$.a"), b) == [
+        '{"b":[1,2,{"c":"x"}]}', None, None, "1.5"]
+    assert ev(F.GetJsonObject(col("j"), "$.t"), b) == ["true", None, None, None]
+    assert ev(F.GetJsonObject(col("j"), "$.n"), b) == [None, None, None, None]
+    assert ev(F.GetJsonObject(col("j"), "$.missing"), b) == [None, None, None, None]
