@@ -314,6 +314,10 @@ class Executor:
         """explode/posexplode (generate_exec.rs). Without a list dtype the
         exploded source is a delimited string column: generator
         'explode_split' / 'posexplode_split' with args [expr, lit(delim)]."""
+        if node.generator == "json_tuple":
+            return self._generate_json_tuple(node)
+        if node.generator == "udtf":
+            return self._generate_udtf(node)
         if node.generator not in ("explode_split", "posexplode_split"):
             raise NotImplementedError(f"generator {node.generator}")
         delim = node.args[1].value if len(node.args) > 1 else ","
@@ -339,6 +343,65 @@ class Executor:
                                    torch.tensor(poss, dtype=torch.int32, device=b.device)))
             names.append("col")
             cols.append(Column.from_pylist(toks, dtypes.string, str(b.device)))
+            out.append(RecordBatch(names, cols))
+        return out
+
+    def _generate_json_tuple(self, node: P.Generate) -> List[RecordBatch]:
+        """json_tuple(json, k1, k2, ...) -> one row, one column per key
+        (generate/json_tuple.rs analogue; host JSON parse)."""
+        import json
+
+        keys = [a.value for a in node.args[1:]]
+        out = []
+        for b in self.execute(node.child):
+            vals = node.args[0].eval(b).to("cpu").to_pylist()
+            cols_out: List[list] = [[] for _ in keys]
+            for v in vals:
+                doc = None
+                if v is not None:
+                    try:
+                        doc = json.loads(v)
+                    except ValueError:
+                        doc = None
+                for j, k in enumerate(keys):
+                    r = doc.get(k) if isinstance(doc, dict) else None
+                    if r is None:
+                        cols_out[j].append(None)
+                    elif isinstance(r, str):
+                        cols_out[j].append(r)
+                    elif isinstance(r, bool):
+                        cols_out[j].append("true" if r else "false")
+                    elif isinstance(r, (dict, list)):
+                        cols_out[j].append(json.dumps(r, separators=(",", ":")))
+                    else:
+                        cols_out[j].append(str(r))
+            names = list(b.names) + [f"c{j}" for j in range(len(keys))]
+            cols = list(b.columns) + [
+                Column.from_pylist(c, dtypes.string, str(b.device))
+                for c in cols_out]
+            out.append(RecordBatch(names, cols))
+        return out
+
+    def _generate_udtf(self, node: P.Generate) -> List[RecordBatch]:
+        """Python UDTF (spark_udtf_wrapper.rs analogue): node.udtf maps one
+        input row (tuple of evaluated args) to an iterable of output rows;
+        each output row is a tuple matching node.udtf_schema."""
+        out = []
+        for b in self.execute(node.child):
+            args = [a.eval(b).to("cpu").to_pylist() for a in node.args]
+            rows, gen_rows = [], []
+            for i, tup in enumerate(zip(*args) if args else []):
+                for r in node.udtf(*tup):
+                    rows.append(i)
+                    gen_rows.append(r)
+            idx = torch.tensor(rows, dtype=torch.int64, device=b.device)
+            base = b.gather(idx)
+            names = list(base.names)
+            cols = list(base.columns)
+            for j, (cname, cdt) in enumerate(node.udtf_schema):
+                vals = [r[j] for r in gen_rows]
+                cols.append(Column.from_pylist(vals, cdt, str(b.device)))
+                names.append(cname)
             out.append(RecordBatch(names, cols))
         return out
 
@@ -789,6 +852,37 @@ class Executor:
         out_right = right.gather(ri)
         return [RecordBatch(out_left.names + out_right.names,
                             out_left.columns + out_right.columns)]
+
+    def _exec_PyUdaf(self, node: P.PyUdaf) -> List[RecordBatch]:
+        """Python UDAF: device grouping, host per-group callback."""
+        b = _concat(self.execute(node.child))
+        device = b.device
+        n = b.num_rows
+        key_cols = [a.expr.eval(b) for a in node.keys]
+        if key_cols:
+            gids, reps = ops.group_ids(key_cols)
+            ngroups = int(reps.numel())
+            out_keys = [c.gather(reps) for c in key_cols]
+        else:
+            gids = torch.zeros(n, dtype=torch.int64, device=device)
+            ngroups = 1
+            out_keys = []
+        in_vals = [e.eval(b).to("cpu").to_pylist() for e in node.inputs]
+        gl = gids.to("cpu").tolist()
+        buckets: List[List[int]] = [[] for _ in range(ngroups)]
+        for i, g in enumerate(gl):
+            buckets[g].append(i)
+        results = []
+        for rows in buckets:
+            args = [[vals[i] for i in rows] for vals in in_vals]
+            results.append(node.fn(*args))
+        names = [a.name for a in node.keys]
+        cols = list(out_keys)
+        for j, (cname, cdt) in enumerate(node.out_schema):
+            vals = [r[j] if isinstance(r, tuple) else r for r in results]
+            cols.append(Column.from_pylist(vals, cdt, str(device)))
+            names.append(cname)
+        return [RecordBatch(names, cols)]
 
     # --------------------------------------------------------------- window
     def _exec_Window(self, node: P.Window) -> List[RecordBatch]:

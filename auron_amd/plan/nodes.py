@@ -194,11 +194,31 @@ class Window(PlanNode):
 
 @dataclass(eq=False)
 class Generate(PlanNode):
-    """explode/posexplode (generate_exec.rs). Round-1: explode of
-    literal lists is unsupported; placeholder for API parity."""
+    """Generators (generate_exec.rs): explode_split / posexplode_split
+    (delimited-string explode), json_tuple(json, k1..kn), and udtf
+    (Python callback mapping an input row to 0..n output rows, the
+    spark_udtf_wrapper analogue; udtf_schema = [(name, dtype), ...])."""
     child: PlanNode
     generator: str
     args: List[Expr] = field(default_factory=list)
+    udtf: object = None  # callable for generator="udtf" (not serialized)
+    udtf_schema: list = field(default_factory=list)
+
+    def children(self):
+        return [self.child]
+
+
+@dataclass(eq=False)
+class PyUdaf(PlanNode):
+    """Python UDAF (agg/spark_udaf_wrapper.rs analogue): group rows by
+    `keys`, call `fn(*value_lists)` once per group with that group's
+    input column values as Python lists, emit one row per group with
+    the scalar(s) it returns (tuple matching out_schema)."""
+    child: PlanNode
+    keys: List  # List[Aliased]
+    inputs: List[Expr] = field(default_factory=list)
+    fn: object = None  # callable (not serialized)
+    out_schema: list = field(default_factory=list)  # [(name, dtype)]
 
     def children(self):
         return [self.child]

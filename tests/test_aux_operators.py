@@ -54,3 +54,58 @@ def test_generate_explode():
     assert out["col"] == ["a", "b", "c", "x", "p", "q"]
     assert out["pos"] == [0, 1, 2, 0, 0, 1]
     assert out["v"] == [1, 1, 1, 2, 4, 4]
+
+
+def test_generate_json_tuple():
+    from auron_amd import AuronSession, col, dtypes, lit
+    from auron_amd.column import RecordBatch
+    from auron_amd.plan import nodes as P
+
+    docs = ['{"a": 1, "b": "x"}', '{"a": null}', "bad", None]
+    sc = P.MemoryScan([RecordBatch.from_pydict(
+        {"id": [1, 2, 3, 4], "j": docs}, {"id": dtypes.int64, "j": dtypes.string})])
+    g = P.Generate(sc, "json_tuple", [col("j"), lit("a"), lit("b")])
+    out = AuronSession().collect(g).to_pydict()
+    assert out["c0"] == ["1", None, None, None]
+    assert out["c1"] == ["x", None, None, None]
+    assert out["id"] == [1, 2, 3, 4]
+
+
+def test_generate_udtf():
+    from auron_amd import AuronSession, col, dtypes
+    from auron_amd.column import RecordBatch
+    from auron_amd.plan import nodes as P
+
+    def dup(n):
+        # emits n rows of (i, i*10) for input n
+        for i in range(int(n)):
+            yield (i, i * 10)
+
+    sc = P.MemoryScan([RecordBatch.from_pydict(
+        {"n": [2, 0, 1]}, {"n": dtypes.int64})])
+    g = P.Generate(sc, "udtf", [col("n")], udtf=dup,
+                   udtf_schema=[("i", dtypes.int64), ("tens", dtypes.int64)])
+    out = AuronSession().collect(g).to_pydict()
+    assert out["n"] == [2, 2, 1]
+    assert out["i"] == [0, 1, 0]
+    assert out["tens"] == [0, 10, 0]
+
+
+def test_py_udaf():
+    from auron_amd import AuronSession, col, dtypes
+    from auron_amd.column import RecordBatch
+    from auron_amd.exprs import Aliased
+    from auron_amd.plan import nodes as P
+
+    def geo_mean_ish(vals):
+        xs = [v for v in vals if v is not None]
+        return (sum(xs) / len(xs), len(xs)) if xs else (None, 0)
+
+    sc = P.MemoryScan([RecordBatch.from_pydict(
+        {"k": ["a", "b", "a", "a"], "v": [1.0, 10.0, 3.0, None]},
+        {"k": dtypes.string, "v": dtypes.float64})])
+    g = P.PyUdaf(sc, [Aliased(col("k"), "k")], [col("v")], geo_mean_ish,
+                 [("m", dtypes.float64), ("cnt", dtypes.int64)])
+    out = AuronSession().collect(g).to_pydict()
+    m = {k: (a, c) for k, a, c in zip(out["k"], out["m"], out["cnt"])}
+    assert m == {"a": (2.0, 2), "b": (10.0, 1)}
