@@ -403,6 +403,42 @@ _ARROW_TO_AURON = {
     "double": dtypes.float64, "date32[day]": dtypes.date32,
 }
 
+class _PinnedPool:
+    """Reusable pinned staging buffers. hipHostMalloc costs ~0.1-1 ms per
+    call; a TPC-DS query stages ~30 column-chunk buffers, so per-scan
+    allocation was a measurable share of host time. Buffers are returned
+    with a recorded event and only reused once their H2D copy finished."""
+
+    MAX_ENTRIES = 32
+
+    def __init__(self):
+        import threading
+
+        self._free: list = []  # (capacity, base_tensor, event|None)
+        self._lock = threading.Lock()
+
+    def acquire(self, nbytes: int):
+        with self._lock:
+            for i, (cap, t, ev) in enumerate(self._free):
+                if cap >= nbytes and (ev is None or ev.query()):
+                    self._free.pop(i)
+                    return t, t[:nbytes]
+        cap = 1 << max(12, (nbytes - 1).bit_length())
+        t = torch.empty(cap, dtype=torch.uint8, pin_memory=True)
+        return t, t[:nbytes]
+
+    def release(self, base: torch.Tensor, device) -> None:
+        ev = None
+        if torch.device(device).type == "cuda":
+            ev = torch.cuda.Event()
+            ev.record(torch.cuda.current_stream(device))
+        with self._lock:
+            if len(self._free) < self.MAX_ENTRIES:
+                self._free.append((base.numel(), base, ev))
+
+
+_PINNED = _PinnedPool()
+
 # footer + page-header metadata cache, keyed by (path, mtime, columns).
 # Metadata only (never data): the Spark-side analogue is the parquet
 # footer cache; pages are re-read and re-decoded on every scan.
@@ -553,90 +589,97 @@ def read_columns_native(path: str, columns: List[str], device,
 
     mm = np.memmap(path, dtype=np.uint8, mode="r")
     # +8 pad: the dict-index kernel's tail does unaligned 8-byte loads
-    buf_t = torch.empty(meta.total + 8, dtype=torch.uint8, pin_memory=use_gpu)
-    buf = buf_t.numpy()
-    for (src, clen, dst) in meta.ranges:
-        buf[dst:dst + clen] = mm[src:src + clen]
+    if use_gpu:
+        pin_base, buf_t = _PINNED.acquire(meta.total + 8)
+    else:
+        pin_base = None
+        buf_t = torch.empty(meta.total + 8, dtype=torch.uint8)
+    try:
+        buf = buf_t.numpy()
+        for (src, clen, dst) in meta.ranges:
+            buf[dst:dst + clen] = mm[src:src + clen]
 
-    if not meta.parsed:
-        ok = []
+        if not meta.parsed:
+            ok = []
+            for cm in meta.cols:
+                cm.pages = []
+                good = True
+                for (new_off, clen, nvals, _an) in cm.chunks:
+                    ck = parse_pages(buf, new_off, clen, nvals, cm.has_def)
+                    if ck is None or (cm.phys == "BYTE_ARRAY" and not ck.is_dict):
+                        good = False  # e.g. PLAIN byte-array pages -> host path
+                        break
+                    cm.pages.append(ck)
+                if good:
+                    ok.append(cm)
+            # per-column fallback: drop unparseable columns; their chunk ranges
+            # stay in the staging layout (small waste, correctness unaffected)
+            meta.cols = ok
+            meta.parsed = True
+            if not ok:
+                return None
+
+        dbuf = buf_t.to(device, non_blocking=True) if use_gpu else None
+
+        out: Dict[str, Column] = {}
         for cm in meta.cols:
-            cm.pages = []
-            good = True
-            for (new_off, clen, nvals, _an) in cm.chunks:
-                ck = parse_pages(buf, new_off, clen, nvals, cm.has_def)
-                if ck is None or (cm.phys == "BYTE_ARRAY" and not ck.is_dict):
-                    good = False  # e.g. PLAIN byte-array pages -> host path
-                    break
-                cm.pages.append(ck)
-            if good:
-                ok.append(cm)
-        # per-column fallback: drop unparseable columns; their chunk ranges
-        # stay in the staging layout (small waste, correctness unaffected)
-        meta.cols = ok
-        meta.parsed = True
-        if not ok:
-            return None
-
-    dbuf = buf_t.to(device, non_blocking=True) if use_gpu else None
-
-    out: Dict[str, Column] = {}
-    for cm in meta.cols:
-        if cm.phys == "BYTE_ARRAY":
-            parts = []
+            if cm.phys == "BYTE_ARRAY":
+                parts = []
+                for (chunk, ck) in zip(cm.chunks, cm.pages):
+                    (_off, _clen, nvals, chunk_nulls) = chunk
+                    parts.append(_decode_chunk_strings(
+                        buf, dbuf, ck, nvals, device, use_gpu, chunk_nulls))
+                col = parts[0] if len(parts) == 1 else Column.concat(parts)
+                if not use_gpu:
+                    col = col.to(device)
+                out[cm.name] = col
+                continue
+            parts_data = []
+            parts_valid = []
+            any_nulls = False
             for (chunk, ck) in zip(cm.chunks, cm.pages):
                 (_off, _clen, nvals, chunk_nulls) = chunk
-                parts.append(_decode_chunk_strings(
-                    buf, dbuf, ck, nvals, device, use_gpu, chunk_nulls))
-            col = parts[0] if len(parts) == 1 else Column.concat(parts)
+                if use_gpu:
+                    if ck.is_dict:
+                        data_t, valid_t = _decode_chunk_gpu_dict(
+                            dbuf, buf, ck, nvals, cm.phys, device, chunk_nulls)
+                    else:
+                        data_t, valid_t = _decode_chunk_gpu(dbuf, ck.pages, nvals,
+                                                            cm.phys, device,
+                                                            chunk_nulls)
+                else:
+                    if ck.is_dict:
+                        data_np, valid_np, _ = decode_chunk_np_dict(buf, ck, nvals,
+                                                                    cm.phys)
+                    else:
+                        data_np, valid_np = decode_chunk_np(buf, ck.pages, nvals,
+                                                            cm.phys)
+                    data_t = torch.from_numpy(data_np)
+                    valid_t = torch.from_numpy(valid_np).to(torch.bool) if valid_np is not None else None
+                if cm.dtype.code == dtypes.DECIMAL64 and data_t.dtype == torch.int32:
+                    data_t = data_t.to(torch.int64)  # widen INT32-backed decimals
+                parts_data.append(data_t)
+                if valid_t is not None:
+                    parts_valid.append(valid_t)
+                    if chunk_nulls:
+                        any_nulls = True
+            data = torch.cat(parts_data) if len(parts_data) > 1 else parts_data[0]
+            validity = None
+            if parts_valid and len(parts_valid) == len(parts_data):
+                validity = torch.cat(parts_valid) if len(parts_valid) > 1 else parts_valid[0]
+                if not any_nulls:
+                    validity = None
+                elif not use_gpu and bool(validity.all()):
+                    validity = None
             if not use_gpu:
-                col = col.to(device)
-            out[cm.name] = col
-            continue
-        parts_data = []
-        parts_valid = []
-        any_nulls = False
-        for (chunk, ck) in zip(cm.chunks, cm.pages):
-            (_off, _clen, nvals, chunk_nulls) = chunk
-            if use_gpu:
-                if ck.is_dict:
-                    data_t, valid_t = _decode_chunk_gpu_dict(
-                        dbuf, buf, ck, nvals, cm.phys, device, chunk_nulls)
-                else:
-                    data_t, valid_t = _decode_chunk_gpu(dbuf, ck.pages, nvals,
-                                                        cm.phys, device,
-                                                        chunk_nulls)
-            else:
-                if ck.is_dict:
-                    data_np, valid_np, _ = decode_chunk_np_dict(buf, ck, nvals,
-                                                                cm.phys)
-                else:
-                    data_np, valid_np = decode_chunk_np(buf, ck.pages, nvals,
-                                                        cm.phys)
-                data_t = torch.from_numpy(data_np)
-                valid_t = torch.from_numpy(valid_np).to(torch.bool) if valid_np is not None else None
-            if cm.dtype.code == dtypes.DECIMAL64 and data_t.dtype == torch.int32:
-                data_t = data_t.to(torch.int64)  # widen INT32-backed decimals
-            parts_data.append(data_t)
-            if valid_t is not None:
-                parts_valid.append(valid_t)
-                if chunk_nulls:
-                    any_nulls = True
-        data = torch.cat(parts_data) if len(parts_data) > 1 else parts_data[0]
-        validity = None
-        if parts_valid and len(parts_valid) == len(parts_data):
-            validity = torch.cat(parts_valid) if len(parts_valid) > 1 else parts_valid[0]
-            if not any_nulls:
-                validity = None
-            elif not use_gpu and bool(validity.all()):
-                validity = None
-        if not use_gpu:
-            data = data.to(device)
-            if validity is not None:
-                validity = validity.to(device)
-        out[cm.name] = Column(cm.dtype, data, validity)
-    return out
-
+                data = data.to(device)
+                if validity is not None:
+                    validity = validity.to(device)
+            out[cm.name] = Column(cm.dtype, data, validity)
+        return out
+    finally:
+        if pin_base is not None:
+            _PINNED.release(pin_base, device)
 
 def _gpu_validity_prefix(dbuf, pages, num_values, device, chunk_nulls=True):
     """Decode def levels on device; returns (validity uint8 | None, prefix).
