@@ -59,7 +59,7 @@ def _try_load():
     for f in ("au_agg_scatter", "au_agg_count"):
         getattr(lib, f).restype = ctypes.c_int
     lib.au_pq_rle1.argtypes = [c, ctypes.c_int, c, c, c]
-    lib.au_pq_rle_idx.argtypes = [c, ctypes.c_int, c, c, c]
+    lib.au_pq_rle_idx.argtypes = [c, ctypes.c_int, c, c, c, c]
     lib.au_pq_scatter.argtypes = [c, ctypes.c_int, c, c, c, c, ctypes.c_int, i64, c]
     lib.au_pq_copy_plain.argtypes = [c, ctypes.c_int, c, c, ctypes.c_int, i64, c]
     for f in ("au_pq_rle1", "au_pq_rle_idx", "au_pq_scatter", "au_pq_copy_plain"):

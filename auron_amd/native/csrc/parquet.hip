@@ -129,16 +129,27 @@ AU_EXPORT int au_pq_rle1(const void* pages_dev, int npages, const void* buf,
 // Generalized RLE/bit-packed hybrid decode for DICTIONARY indices
 // (bit width 1..32). One workgroup per page; same two-phase structure as
 // k_pq_rle1. Descriptor reuses PqPage with repurposed fields:
-//   def_off=idx_off  def_len=idx_len  values_off=out_base (compact)
-//   n_values=nvalid  row_start=unused pad=bit_width
+//   def_off=idx_off  def_len=idx_len  values_off=unused
+//   n_values=page rows  row_start=chunk-relative row  pad=bit_width
+// `prefix` is the chunk-wide inclusive cumsum of validity (or null for
+// no-null chunks): the kernel derives each page's valid count and its
+// compact output base on-device, so the host never syncs.
 // The staged buffer is padded by 8 bytes so the unaligned 8-byte loads at
 // a literal run's tail never read out of bounds.
-__global__ void k_pq_rle_idx(const PqPage* pages, const uint8_t* buf, int32_t* out) {
+__global__ void k_pq_rle_idx(const PqPage* pages, const uint8_t* buf, int32_t* out,
+                             const int64_t* prefix) {
   const PqPage p = pages[blockIdx.x];
   const uint8_t* src = buf + p.def_off;
   const int64_t src_len = p.def_len;
-  int32_t* dst = out + p.values_off;
-  const int64_t n = p.n_values;
+  int64_t base, n;
+  if (prefix) {
+    base = p.row_start == 0 ? 0 : prefix[p.row_start - 1];
+    n = prefix[p.row_start + p.n_values - 1] - base;
+  } else {
+    base = p.row_start;
+    n = p.n_values;
+  }
+  int32_t* dst = out + base;
   const int bw = (int)p.pad;
   const uint64_t mask = bw >= 64 ? ~0ull : ((1ull << bw) - 1);
   const int vbytes = (bw + 7) >> 3;
@@ -217,10 +228,10 @@ __global__ void k_pq_rle_idx(const PqPage* pages, const uint8_t* buf, int32_t* o
 }
 
 AU_EXPORT int au_pq_rle_idx(const void* pages_dev, int npages, const void* buf,
-                            int32_t* out, void* stream) {
+                            int32_t* out, const int64_t* prefix, void* stream) {
   if (npages == 0) return 0;
   hipLaunchKernelGGL(k_pq_rle_idx, dim3(npages), dim3(256), 0, (hipStream_t)stream,
-                     (const PqPage*)pages_dev, (const uint8_t*)buf, out);
+                     (const PqPage*)pages_dev, (const uint8_t*)buf, out, prefix);
   return (int)hipGetLastError();
 }
 

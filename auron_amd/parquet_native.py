@@ -706,40 +706,29 @@ def _gpu_validity_prefix(dbuf, pages, num_values, device, chunk_nulls=True):
 
 
 def _gpu_dict_indices(dbuf, buf, ck: ChunkPages, prefix, num_values, device):
-    """Decode all RLE dictionary-index pages of a chunk -> compact int32."""
+    """Decode all RLE dictionary-index pages of a chunk into one int32
+    array: compact (one entry per VALID row) when `prefix` is given,
+    dense otherwise. Per-page valid counts and output bases come from
+    the prefix sum ON DEVICE — no host sync."""
     from . import native
 
     lib = native.require()
     pages = ck.pages
     npages = len(pages)
-    if prefix is not None:
-        # per-page valid counts from the prefix sum (small D2H)
-        bounds = []
-        for p in pages:
-            bounds.append(p.row_start - 1)
-            bounds.append(p.row_start + p.n_values - 1)
-        bt = torch.tensor(bounds, dtype=torch.int64, device=device).clamp(min=0)
-        bvals = prefix[bt].cpu()
-        nvalid = []
-        for i, p in enumerate(pages):
-            lo = 0 if p.row_start == 0 else int(bvals[2 * i].item())
-            hi = int(bvals[2 * i + 1].item())
-            nvalid.append(hi - lo)
-    else:
-        nvalid = [p.n_values for p in pages]
     arr = np.zeros((npages, 6), dtype=np.int64)
-    out_base = 0
     for i, p in enumerate(pages):
         bw = int(buf[p.values_off])
-        arr[i] = (p.values_off + 1, p.values_len - 1, out_base, nvalid[i], 0, bw)
-        out_base += nvalid[i]
+        arr[i] = (p.values_off + 1, p.values_len - 1, 0, p.n_values,
+                  p.row_start, bw)
     darr = torch.from_numpy(arr.reshape(-1)).to(device)
     sp = native.stream_ptr(device)
-    idx = torch.empty(max(out_base, 1), dtype=torch.int32, device=device)
+    idx = torch.empty(max(num_values, 1), dtype=torch.int32, device=device)
     rc = lib.au_pq_rle_idx(darr.data_ptr(), npages, dbuf.data_ptr(),
-                           idx.data_ptr(), sp)
+                           idx.data_ptr(),
+                           prefix.data_ptr() if prefix is not None else None,
+                           sp)
     native.check(rc, "au_pq_rle_idx")
-    return idx[:out_base]
+    return idx
 
 
 def _decode_chunk_gpu_dict(dbuf, buf, ck: ChunkPages, num_values: int,
@@ -752,12 +741,15 @@ def _decode_chunk_gpu_dict(dbuf, buf, ck: ChunkPages, num_values: int,
            "FLOAT": torch.float32, "DOUBLE": torch.float64}[phys]
     esize = _PHYS_NP[phys]().itemsize
     dvals = dbuf[ck.dict_off:ck.dict_off + ck.dict_nvals * esize].clone().view(tdt)
-    gathered = dvals[idx.to(torch.int64)]
     if validity is None:
-        return gathered, None
-    out = torch.zeros(num_values, dtype=tdt, device=device)
+        return dvals[idx[:num_values].to(torch.int64)], None
+    # idx is compact over valid rows; route each row to its entry via the
+    # prefix sum (invalid rows read a clamped slot and are masked out)
     vmask = validity.to(torch.bool)
-    out[vmask] = gathered
+    pos = (prefix - 1).clamp(min=0)
+    idx_row = idx[pos].to(torch.int64).clamp_(0, max(ck.dict_nvals - 1, 0))
+    out = torch.where(vmask, dvals[idx_row], torch.zeros((), dtype=tdt,
+                                                         device=device))
     return out, vmask
 
 
@@ -780,13 +772,13 @@ def _decode_chunk_strings(buf, dbuf, ck: ChunkPages, num_values: int, device,
     dict_lens = torch.from_numpy(ck.dict_str_lens).to(device)
     if validity is not None:
         vmask = validity.to(torch.bool)
-        row_idx = torch.zeros(num_values, dtype=torch.int64, device=device)
-        row_idx[vmask] = idx.to(torch.int64)
+        pos = (prefix - 1).clamp(min=0)
+        row_idx = idx[pos].to(torch.int64).clamp_(0, max(ck.dict_nvals - 1, 0))
         lens = torch.where(vmask, dict_lens[row_idx],
                            torch.zeros_like(dict_lens[row_idx]))
     else:
         vmask = None
-        row_idx = idx.to(torch.int64)
+        row_idx = idx[:num_values].to(torch.int64)
         lens = dict_lens[row_idx]
     offsets = torch.zeros(num_values + 1, dtype=torch.int64, device=device)
     torch.cumsum(lens, 0, out=offsets[1:])
