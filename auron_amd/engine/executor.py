@@ -84,6 +84,12 @@ def _normalize_join_keys(lkeys, rkeys):
     return lo, ro
 
 
+def _cast_for_range(c: Column) -> Column:
+    if c.dtype.is_string:
+        raise NotImplementedError("range partitioning on string keys")
+    return c
+
+
 def _concat(batches: List[RecordBatch]) -> RecordBatch:
     assert batches, "executor invariant violated: empty batch list"
     if len(batches) == 1:
@@ -609,6 +615,31 @@ class Executor:
         elif node.kind == "roundrobin":
             n = b.num_rows
             part = torch.arange(n, dtype=torch.int64, device=device) % W
+            dest = [b.filter(part == d) for d in range(W)]
+        elif node.kind == "range":
+            # range repartition (Partitioning::RangePartitioning): sample
+            # the first key on every rank, agree on W-1 global bounds,
+            # route rows by searchsorted. Rank d holds keys in
+            # (bounds[d-1], bounds[d]] — a global sort order across ranks.
+            key = _cast_for_range(node.keys[0].eval(b))
+            n = b.num_rows
+            take = min(n, 4096)
+            sample = key.data[torch.randperm(n, device=device)[:take]] if n \
+                else key.data
+            gathered = [None] * W
+            if W > 1 and dist.is_initialized():
+                dist.all_gather_object(gathered, sample.cpu(),
+                                       group=self.ctx.group)
+            else:
+                gathered = [sample.cpu()]
+            allsamp = torch.cat([g for g in gathered if g is not None and g.numel()]) \
+                if any(g is not None and g.numel() for g in gathered) else torch.zeros(1, dtype=key.data.dtype)
+            qs = torch.quantile(allsamp.to(torch.float64),
+                                torch.linspace(0, 1, W + 1, dtype=torch.float64)[1:-1])
+            bounds = qs.to(device)
+            part = torch.searchsorted(bounds, key.data.to(torch.float64))
+            if key.validity is not None:  # nulls first (rank 0)
+                part = torch.where(key.validity, part, torch.zeros_like(part))
             dest = [b.filter(part == d) for d in range(W)]
         else:  # hash
             key_cols = [k.eval(b) for k in node.keys]

@@ -143,3 +143,28 @@ def test_single_exchange_gathers_to_rank0():
     results = _spawn("body_single_exchange")
     assert results[0] == 8
     assert results[1] == 0
+
+
+def _range_exchange(rank, world):
+    import numpy as np
+
+    rng = np.random.default_rng(rank)
+    data = {"k": rng.integers(0, 1000, 200).tolist(),
+            "v": list(range(200))}
+    t = {"k": dtypes.int64, "v": dtypes.int64}
+    s = AuronSession()
+    plan = P.Exchange(
+        P.MemoryScan([RecordBatch.from_pydict(data, t)]), "range", [col("k")])
+    got = s.collect(plan).to_pydict()
+    return {"rank": rank, "keys": got["k"], "n": len(got["k"])}
+
+
+def test_range_exchange_global_order():
+    """Range repartition: every key on rank 0 <= every key on rank 1,
+    and no rows lost."""
+    res = _spawn("_range_exchange")
+    assert res[0]["n"] + res[1]["n"] == 400
+    if res[0]["keys"] and res[1]["keys"]:
+        assert max(res[0]["keys"]) <= min(res[1]["keys"])
+    # rough balance (quantile bounds from 400 samples of uniform keys)
+    assert 100 < res[0]["n"] < 300
