@@ -287,6 +287,18 @@ class Cast(Expr):
 
 
 @dataclass(eq=False)
+class TryCast(Expr):
+    """try_cast (ext-exprs cast.rs): unparseable inputs yield null.
+    String parsing in _cast_string already nulls on failure; this node
+    declares the intent at the plan surface."""
+    child: Expr
+    to: DataType
+
+    def eval(self, batch: RecordBatch) -> Column:
+        return _cast_col(self.child.eval(batch), self.to)
+
+
+@dataclass(eq=False)
 class Arith(Expr):
     op: str
     left: Expr

@@ -663,6 +663,18 @@ class SparkPartitionId(Expr):
 
 
 @dataclass(eq=False)
+class RowNum(Expr):
+    """ext-exprs RowNum: 1-based row number within this task's stream."""
+
+    def eval(self, batch):
+        base = EVAL_CONTEXT.get().get("row_base", 0)
+        n = batch.num_rows
+        return Column(dtypes.int64,
+                      base + 1 + torch.arange(n, dtype=torch.int64,
+                                              device=batch.device))
+
+
+@dataclass(eq=False)
 class MonotonicallyIncreasingId(Expr):
     """Spark layout: partition_id << 33 | row index within partition."""
 

@@ -189,3 +189,11 @@ def test_get_json_object():
     assert ev(F.GetJsonObject(col("j"), "$.t"), b) == ["true", None, None, None]
     assert ev(F.GetJsonObject(col("j"), "$.n"), b) == [None, None, None, None]
     assert ev(F.GetJsonObject(col("j"), "$.missing"), b) == [None, None, None, None]
+
+
+def test_trycast_and_rownum():
+    from auron_amd.exprs import TryCast
+
+    b = _b(s=(["12", "x", None, "3.5"], dtypes.string))
+    assert ev(TryCast(col("s"), dtypes.int64), b) == [12, None, None, 3]
+    assert ev(F.RowNum(), b) == [1, 2, 3, 4]
