@@ -124,8 +124,9 @@ def pack_descs(cols, device) -> Tuple[torch.Tensor, list]:
             arr[i]["validity"] = v.data_ptr()
         arr[i]["dtype"] = col.dtype.code
         arr[i]["scale"] = col.dtype.scale
-    host = torch.from_numpy(arr.view(np.uint8).reshape(-1).copy())
-    dev = host.to(device)
+    from ..pinned import to_device
+
+    dev = to_device(arr.view(np.uint8).reshape(-1), device)
     keep.append(dev)
     return dev, keep
 

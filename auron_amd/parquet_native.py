@@ -403,41 +403,8 @@ _ARROW_TO_AURON = {
     "double": dtypes.float64, "date32[day]": dtypes.date32,
 }
 
-class _PinnedPool:
-    """Reusable pinned staging buffers. hipHostMalloc costs ~0.1-1 ms per
-    call; a TPC-DS query stages ~30 column-chunk buffers, so per-scan
-    allocation was a measurable share of host time. Buffers are returned
-    with a recorded event and only reused once their H2D copy finished."""
-
-    MAX_ENTRIES = 32
-
-    def __init__(self):
-        import threading
-
-        self._free: list = []  # (capacity, base_tensor, event|None)
-        self._lock = threading.Lock()
-
-    def acquire(self, nbytes: int):
-        with self._lock:
-            for i, (cap, t, ev) in enumerate(self._free):
-                if cap >= nbytes and (ev is None or ev.query()):
-                    self._free.pop(i)
-                    return t, t[:nbytes]
-        cap = 1 << max(12, (nbytes - 1).bit_length())
-        t = torch.empty(cap, dtype=torch.uint8, pin_memory=True)
-        return t, t[:nbytes]
-
-    def release(self, base: torch.Tensor, device) -> None:
-        ev = None
-        if torch.device(device).type == "cuda":
-            ev = torch.cuda.Event()
-            ev.record(torch.cuda.current_stream(device))
-        with self._lock:
-            if len(self._free) < self.MAX_ENTRIES:
-                self._free.append((base.numel(), base, ev))
-
-
-_PINNED = _PinnedPool()
+from .pinned import POOL as _PINNED
+from .pinned import to_device as _pin_to_device
 
 # footer + page-header metadata cache, keyed by (path, mtime, columns).
 # Metadata only (never data): the Spark-side analogue is the parquet
@@ -695,7 +662,7 @@ def _gpu_validity_prefix(dbuf, pages, num_values, device, chunk_nulls=True):
     arr = np.zeros((npages, 6), dtype=np.int64)
     for i, p in enumerate(pages):
         arr[i] = (p.def_off, p.def_len, p.values_off, p.n_values, p.row_start, 0)
-    darr = torch.from_numpy(arr.reshape(-1)).to(device)
+    darr = _pin_to_device(arr.reshape(-1), device)
     sp = native.stream_ptr(device)
     validity = torch.empty(num_values, dtype=torch.uint8, device=device)
     rc = lib.au_pq_rle1(darr.data_ptr(), npages, dbuf.data_ptr(),
@@ -720,7 +687,7 @@ def _gpu_dict_indices(dbuf, buf, ck: ChunkPages, prefix, num_values, device):
         bw = int(buf[p.values_off])
         arr[i] = (p.values_off + 1, p.values_len - 1, 0, p.n_values,
                   p.row_start, bw)
-    darr = torch.from_numpy(arr.reshape(-1)).to(device)
+    darr = _pin_to_device(arr.reshape(-1), device)
     sp = native.stream_ptr(device)
     idx = torch.empty(max(num_values, 1), dtype=torch.int32, device=device)
     rc = lib.au_pq_rle_idx(darr.data_ptr(), npages, dbuf.data_ptr(),
@@ -814,7 +781,7 @@ def _decode_chunk_gpu(dbuf: torch.Tensor, pages: List[PageDesc], num_values: int
     arr = np.zeros((npages, 6), dtype=np.int64)
     for i, p in enumerate(pages):
         arr[i] = (p.def_off, p.def_len, p.values_off, p.n_values, p.row_start, 0)
-    darr = torch.from_numpy(arr.reshape(-1)).to(device)
+    darr = _pin_to_device(arr.reshape(-1), device)
     sp = native.stream_ptr(device)
 
     tdt = {"INT32": torch.int32, "INT64": torch.int64,
