@@ -7,9 +7,10 @@ exposes engine metrics and memory-manager state as JSON; profiling hooks
 point at rocprofv3 (out-of-process on ROCm).
 
 Endpoints:
-  /healthz   liveness
-  /metrics   per-operator exclusive wall times + memmgr counters
-  /config    resolved configuration
+  /healthz     liveness
+  /metrics     per-operator exclusive wall times + memmgr counters
+  /last_query  MetricNode tree of the most recent execute
+  /config      resolved configuration
 """
 from __future__ import annotations
 
@@ -49,6 +50,11 @@ class _Handler(BaseHTTPRequestHandler):
                 "rank": s.rank,
                 "world_size": s.world_size,
             })
+        elif self.path == "/last_query":
+            # MetricNode tree of the most recent execute (spark-ui's
+            # per-query native-metrics surface)
+            self._send(200, {"tree": s.metric_tree(),
+                             "rendered": s.explain_metrics()})
         elif self.path == "/config":
             from .config import AuronConf
 
