@@ -20,6 +20,7 @@ class PinnedPool:
     def __init__(self):
         self._free: list = []  # (capacity, base_tensor, event|None)
         self._lock = threading.Lock()
+        self._events: dict = {}  # id(base) -> reusable hip event
 
     def acquire(self, nbytes: int):
         with self._lock:
@@ -34,21 +35,28 @@ class PinnedPool:
     def release(self, base: torch.Tensor, device) -> None:
         ev = None
         if torch.device(device).type == "cuda":
-            ev = torch.cuda.Event()
+            ev = self._events.get(id(base))
+            if ev is None:
+                ev = torch.cuda.Event()
+                self._events[id(base)] = ev
             ev.record(torch.cuda.current_stream(device))
         with self._lock:
             if len(self._free) < self.MAX_ENTRIES:
                 self._free.append((base.numel(), base, ev))
+            else:
+                self._events.pop(id(base), None)
 
 
 POOL = PinnedPool()
 
 
 def to_device(arr: np.ndarray, device) -> torch.Tensor:
-    """Upload a small host array through a pooled pinned slab (async)."""
+    """Upload a host array; pooled pinned slab for payloads big enough to
+    beat torch's own small-copy staging (tiny descriptor arrays go the
+    plain route — pool bookkeeping costs more than it saves there)."""
     flat = np.ascontiguousarray(arr).view(np.uint8).reshape(-1)
     dev = torch.device(device)
-    if dev.type != "cuda":
+    if dev.type != "cuda" or flat.nbytes < (8 << 10):
         return torch.from_numpy(flat.copy()).to(dev)
     base, view = POOL.acquire(flat.nbytes)
     view.numpy()[:] = flat
