@@ -33,12 +33,12 @@ class Column:
         self.data = data
         self.validity = validity
         self.offsets = offsets
-        if dtype.is_string:
+        if dtype.uses_offsets:
             assert offsets is not None and offsets.dtype == torch.int32
 
     # ---------------------------------------------------------- basics
     def __len__(self) -> int:
-        if self.dtype.is_string:
+        if self.dtype.uses_offsets:
             return int(self.offsets.shape[0]) - 1
         return int(self.data.shape[0])
 
@@ -78,7 +78,17 @@ class Column:
         validity = None
         if any(v is None for v in values):
             validity = torch.tensor([v is not None for v in values], dtype=torch.bool)
-        if dtype.is_string:
+        if dtype.is_list:
+            offs = [0]
+            flat = []
+            for v in values:
+                if v is not None:
+                    flat.extend(v)
+                offs.append(len(flat))
+            data = torch.tensor(flat, dtype=dtype.torch_dtype) if flat else \
+                torch.empty(0, dtype=dtype.torch_dtype)
+            col = Column(dtype, data, validity, torch.tensor(offs, dtype=torch.int32))
+        elif dtype.is_string:
             bufs = []
             offs = [0]
             total = 0
@@ -115,6 +125,16 @@ class Column:
                 else:
                     out.append(raw[offs[i]:offs[i + 1]].decode("utf-8", errors="replace"))
             return out
+        if self.dtype.is_list:
+            offs = c.offsets.tolist()
+            flat = c.data.tolist()
+            out = []
+            for i in range(len(c)):
+                if valid is not None and not valid[i]:
+                    out.append(None)
+                else:
+                    out.append(flat[offs[i]:offs[i + 1]])
+            return out
         vals = c.data.tolist()
         if self.dtype.code == dtypes.DECIMAL64:
             s = 10 ** self.dtype.scale
@@ -139,7 +159,7 @@ class Column:
             if validity is None:
                 validity = torch.ones(idx.shape[0], dtype=torch.bool, device=self.device)
             validity = validity & ~neg
-        if self.dtype.is_string:
+        if self.dtype.uses_offsets:
             off = self.offsets.to(torch.int64)
             starts = off[idx]
             lens = off[idx + 1] - starts
@@ -149,13 +169,13 @@ class Column:
             torch.cumsum(lens, 0, out=new_off[1:])
             total = int(new_off[-1].item())
             if total == 0:
-                out_bytes = torch.empty(0, dtype=torch.uint8, device=self.device)
+                out_el = torch.empty(0, dtype=self.data.dtype, device=self.device)
             else:
                 pos = torch.arange(total, dtype=torch.int64, device=self.device)
-                row = torch.repeat_interleave(lens)  # maps byte pos -> out row
-                byte_idx = pos - new_off[:-1][row] + starts[row]
-                out_bytes = self.data[byte_idx]
-            return Column(self.dtype, out_bytes, validity, new_off.to(torch.int32))
+                row = torch.repeat_interleave(lens)  # maps element pos -> out row
+                el_idx = pos - new_off[:-1][row] + starts[row]
+                out_el = self.data[el_idx]
+            return Column(self.dtype, out_el, validity, new_off.to(torch.int32))
         data = self.data[idx]
         return Column(self.dtype, data, validity, None)
 
@@ -183,9 +203,9 @@ class Column:
                 for c in cols
             ]
             validity = torch.cat(parts)
-        if dt.is_string:
+        if dt.uses_offsets:
             datas = [c.data for c in cols]
-            data = torch.cat(datas) if datas else torch.empty(0, dtype=torch.uint8, device=device)
+            data = torch.cat(datas) if datas else torch.empty(0, dtype=cols[0].data.dtype, device=device)
             offs = []
             base = 0
             for c in cols:

@@ -21,6 +21,7 @@ FLOAT64 = 6
 DATE32 = 7  # days since epoch, int32 storage
 STRING = 8  # uint8 byte buffer + int32 offsets
 DECIMAL64 = 9  # scaled int64
+LIST = 10  # flattened child values + int32 offsets (element nulls: round 2)
 
 _NAMES = {
     BOOL: "bool",
@@ -33,6 +34,7 @@ _NAMES = {
     DATE32: "date32",
     STRING: "string",
     DECIMAL64: "decimal64",
+    LIST: "list",
 }
 
 _TORCH = {
@@ -46,6 +48,7 @@ _TORCH = {
     DATE32: torch.int32,
     STRING: torch.uint8,  # byte buffer
     DECIMAL64: torch.int64,
+    LIST: None,  # resolved via .child
 }
 
 
@@ -59,15 +62,34 @@ class DataType:
     def name(self) -> str:
         if self.code == DECIMAL64:
             return f"decimal({self.precision},{self.scale})"
+        if self.code == LIST:
+            return f"list<{_NAMES[self.precision]}>"
         return _NAMES[self.code]
 
     @property
     def torch_dtype(self) -> torch.dtype:
+        if self.code == LIST:
+            return _TORCH[self.precision]
         return _TORCH[self.code]
 
     @property
     def is_string(self) -> bool:
         return self.code == STRING
+
+    @property
+    def is_list(self) -> bool:
+        return self.code == LIST
+
+    @property
+    def uses_offsets(self) -> bool:
+        '''Arrow offsets layout: strings (byte elements) and lists.'''
+        return self.code in (STRING, LIST)
+
+    @property
+    def child(self) -> "DataType":
+        '''Element type of a list (stored in the precision/scale slots).'''
+        assert self.code == LIST
+        return DataType(self.precision, 0, self.scale)
 
     @property
     def is_numeric(self) -> bool:
@@ -98,6 +120,11 @@ string = DataType(STRING)
 
 def decimal64(precision: int = 18, scale: int = 2) -> DataType:
     return DataType(DECIMAL64, precision, scale)
+
+
+def list_of(child: DataType) -> DataType:
+    assert child.code != LIST, "nested lists unsupported (round 2)"
+    return DataType(LIST, child.code, child.scale)
 
 
 def from_arrow(at) -> DataType:

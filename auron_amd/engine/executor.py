@@ -324,6 +324,8 @@ class Executor:
             return self._generate_json_tuple(node)
         if node.generator == "udtf":
             return self._generate_udtf(node)
+        if node.generator in ("explode", "posexplode"):
+            return self._generate_explode_list(node)
         if node.generator not in ("explode_split", "posexplode_split"):
             raise NotImplementedError(f"generator {node.generator}")
         delim = node.args[1].value if len(node.args) > 1 else ","
@@ -349,6 +351,38 @@ class Executor:
                                    torch.tensor(poss, dtype=torch.int32, device=b.device)))
             names.append("col")
             cols.append(Column.from_pylist(toks, dtypes.string, str(b.device)))
+            out.append(RecordBatch(names, cols))
+        return out
+
+    def _generate_explode_list(self, node: P.Generate) -> List[RecordBatch]:
+        """explode/posexplode of a LIST column — fully device-vectorized:
+        rows expand by per-row element counts (null rows emit nothing,
+        Spark semantics) and the child values are already flat."""
+        out = []
+        for b in self.execute(node.child):
+            c = node.args[0].eval(b)
+            assert c.dtype.is_list, f"explode needs a list column, got {c.dtype.name}"
+            device = b.device
+            off = c.offsets.to(torch.int64)
+            lens = off[1:] - off[:-1]
+            if c.validity is not None:
+                lens = torch.where(c.validity, lens, torch.zeros_like(lens))
+            rows = torch.repeat_interleave(
+                torch.arange(len(c), dtype=torch.int64, device=device), lens)
+            base = b.gather(rows)
+            names = list(base.names)
+            cols = list(base.columns)
+            total = int(lens.sum().item())
+            starts = torch.cumsum(lens, 0) - lens
+            pos = torch.arange(total, dtype=torch.int64, device=device) \
+                - starts[rows]
+            if node.generator == "posexplode":
+                names.append("pos")
+                cols.append(Column(dtypes.int32, pos.to(torch.int32)))
+            # element index in the flat child: row start + within-row pos
+            el = off[:-1][rows] + pos
+            names.append("col")
+            cols.append(Column(c.dtype.child, c.data[el]))
             out.append(RecordBatch(names, cols))
         return out
 

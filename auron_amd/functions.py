@@ -652,6 +652,25 @@ class BloomFilterMightContain(Expr):
         return Column(dtypes.bool_, mask, c.validity)
 
 
+@dataclass(eq=False)
+class MakeArray(Expr):
+    """spark make_array / array(...): fixed-width list per row from k
+    element expressions (device-vectorized; element nulls become 0 — an
+    element-validity buffer is a round-2 item)."""
+    args: List[Expr]
+
+    def eval(self, batch):
+        cols = [a.eval(batch) for a in self.args]
+        k = len(cols)
+        n = batch.num_rows
+        first = cols[0]
+        data = torch.stack([c.data for c in cols], dim=1).reshape(-1)
+        offsets = (torch.arange(n + 1, dtype=torch.int32, device=batch.device)
+                   * k)
+        return Column(dtypes.list_of(first.dtype), data,
+                      combine_validity(*cols), offsets)
+
+
 # ------------------------------------------------------------- engine ids
 @dataclass(eq=False)
 class SparkPartitionId(Expr):

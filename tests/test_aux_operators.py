@@ -109,3 +109,39 @@ def test_py_udaf():
     out = AuronSession().collect(g).to_pydict()
     m = {k: (a, c) for k, a, c in zip(out["k"], out["m"], out["cnt"])}
     assert m == {"a": (2.0, 2), "b": (10.0, 1)}
+
+
+def test_list_column_and_explode():
+    from auron_amd import AuronSession, col, dtypes
+    from auron_amd.column import Column, RecordBatch
+    from auron_amd.plan import nodes as P
+
+    lt = dtypes.list_of(dtypes.int64)
+    c = Column.from_pylist([[1, 2], [], None, [7]], lt)
+    assert c.to_pylist() == [[1, 2], [], None, [7]]
+    # gather with nulls / reorder keeps layout
+    import torch
+    g = c.gather(torch.tensor([3, 0, -1]))
+    assert g.to_pylist() == [[7], [1, 2], None]
+    assert Column.concat([c, g]).to_pylist() == \
+        [[1, 2], [], None, [7], [7], [1, 2], None]
+
+    sc = P.MemoryScan([RecordBatch(["id", "xs"],
+                                   [Column.from_pylist([1, 2, 3, 4], dtypes.int64), c])])
+    out = AuronSession().collect(P.Generate(sc, "posexplode", [col("xs")])).to_pydict()
+    assert out["id"] == [1, 1, 4]
+    assert out["pos"] == [0, 1, 0]
+    assert out["col"] == [1, 2, 7]
+
+
+def test_make_array_then_explode():
+    from auron_amd import AuronSession, col, dtypes, functions as F
+    from auron_amd.column import RecordBatch
+    from auron_amd.exprs import Aliased
+    from auron_amd.plan import nodes as P
+
+    sc = P.MemoryScan([RecordBatch.from_pydict(
+        {"a": [1, 10], "b": [2, 20]}, {"a": dtypes.int64, "b": dtypes.int64})])
+    proj = P.Project(sc, [Aliased(F.MakeArray([col("a"), col("b")]), "arr")])
+    out = AuronSession().collect(P.Generate(proj, "explode", [col("arr")])).to_pydict()
+    assert out["col"] == [1, 2, 10, 20]
