@@ -66,26 +66,83 @@ __device__ __forceinline__ void atomic_max_i64(int64_t* addr, int64_t v) {
   } while (old != assumed);
 }
 
+// Wave-segmented scatter: lanes load 64 consecutive rows (coalesced),
+// then a shfl-based segmented reduction combines lanes sharing a group
+// id BEFORE touching global atomics. Group-by outputs are usually
+// correlated with input order (rollups over sorted-ish facts produce
+// long equal-gid runs), so this collapses most atomics; for random gids
+// it degrades gracefully to one atomic per lane plus 6 shfl rounds.
 template <typename VT, bool ACC_F64>
 __global__ void k_agg_scatter(const int64_t* gids, int64_t n, const uint8_t* validity,
                               const VT* values, int op, void* acc, int64_t* counts) {
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += (int64_t)gridDim.x * blockDim.x) {
-    if (validity && !validity[i]) continue;
-    int64_t g = gids[i];
-    if (counts) atomicAdd((unsigned long long*)&counts[g], 1ull);
+  const int lane = threadIdx.x & 63;
+  const int64_t wave_global = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) >> 6;
+  for (int64_t base = wave_global * 64; base < n; base += nwaves * 64) {
+    int64_t i = base + lane;
+    bool live = i < n && (!validity || validity[i]);
+    int64_t g = (i < n) ? gids[i] : (int64_t)-1;
     if (ACC_F64) {
-      double v = (double)values[i];
-      double* a = (double*)acc;
-      if (op == 0) atomicAdd(&a[g], v);
-      else if (op == 1) atomic_min_f64(&a[g], v);
-      else atomic_max_f64(&a[g], v);
+      double ident = op == 0 ? 0.0 : (op == 1 ? INFINITY : -INFINITY);
+      double v = live ? (double)values[i] : ident;
+      int64_t c = live ? 1 : 0;
+      // head-flag segmented inclusive scan (a gid may recur in separate
+      // runs within the wave; plain gid-equality folding would double
+      // count across the intervening segment)
+      int64_t pg1 = __shfl_up(g, 1, 64);
+      int flag = (lane == 0) || (pg1 != g);
+      for (int d = 1; d < 64; d <<= 1) {
+        double pv = __shfl_up(v, d, 64);
+        int64_t pc = __shfl_up(c, d, 64);
+        int pf = __shfl_up(flag, d, 64);
+        if (lane >= d) {
+          if (!flag) {
+            if (op == 0) v += pv;
+            else if (op == 1) v = v < pv ? v : pv;
+            else v = v > pv ? v : pv;
+            c += pc;
+          }
+          flag = flag || pf;
+        }
+      }
+      int64_t ng = __shfl_down(g, 1, 64);
+      bool seg_last = (lane == 63) || (ng != g);
+      if (seg_last && g >= 0 && c > 0) {
+        double* a = (double*)acc;
+        if (counts) atomicAdd((unsigned long long*)&counts[g], (unsigned long long)c);
+        if (op == 0) atomicAdd(&a[g], v);
+        else if (op == 1) atomic_min_f64(&a[g], v);
+        else atomic_max_f64(&a[g], v);
+      }
     } else {
-      int64_t v = (int64_t)values[i];
-      int64_t* a = (int64_t*)acc;
-      if (op == 0) atomicAdd((unsigned long long*)&a[g], (unsigned long long)v);
-      else if (op == 1) atomic_min_i64(&a[g], v);
-      else atomic_max_i64(&a[g], v);
+      int64_t ident = op == 0 ? 0 : (op == 1 ? INT64_MAX : INT64_MIN);
+      int64_t v = live ? (int64_t)values[i] : ident;
+      int64_t c = live ? 1 : 0;
+      int64_t pg1 = __shfl_up(g, 1, 64);
+      int flag = (lane == 0) || (pg1 != g);
+      for (int d = 1; d < 64; d <<= 1) {
+        int64_t pv = __shfl_up(v, d, 64);
+        int64_t pc = __shfl_up(c, d, 64);
+        int pf = __shfl_up(flag, d, 64);
+        if (lane >= d) {
+          if (!flag) {
+            if (op == 0) v += pv;
+            else if (op == 1) v = v < pv ? v : pv;
+            else v = v > pv ? v : pv;
+            c += pc;
+          }
+          flag = flag || pf;
+        }
+      }
+      int64_t ng = __shfl_down(g, 1, 64);
+      bool seg_last = (lane == 63) || (ng != g);
+      if (seg_last && g >= 0 && c > 0) {
+        int64_t* a = (int64_t*)acc;
+        if (counts) atomicAdd((unsigned long long*)&counts[g], (unsigned long long)c);
+        if (op == 0) atomicAdd((unsigned long long*)&a[g], (unsigned long long)v);
+        else if (op == 1) atomic_min_i64(&a[g], v);
+        else atomic_max_i64(&a[g], v);
+      }
     }
   }
 }

@@ -174,3 +174,44 @@ def test_agg_scatter_native_vs_torch():
         a_ref, c_ref = ops.agg_scatter(gids, 1000, icol_cpu, fn)
         a_gpu, c_gpu = ops.agg_scatter(gids.to(DEV), 1000, icol_gpu, fn)
         assert torch.equal(a_ref, a_gpu.cpu()), fn
+
+
+def test_agg_scatter_segmented_path_patterns():
+    """Large-ngroups path (wave-segmented scan + global atomics): runs,
+    recurring gids inside one 64-lane wave ([A,A,B,A]-style), random,
+    and all-invalid stretches must all match the CPU reference."""
+    rng = np.random.default_rng(5)
+    NG = 50_000  # > LDS limit -> segmented global path
+    pats = []
+    # long runs
+    pats.append(np.repeat(rng.integers(0, NG, 2000), rng.integers(1, 400, 2000)))
+    # recurring gid within a wave: A A B A C A ...
+    base = rng.integers(0, NG, 30_000)
+    recur = np.empty(60_000, dtype=np.int64)
+    recur[0::2] = 7  # gid 7 every other lane
+    recur[1::2] = base
+    pats.append(recur)
+    # pure random
+    pats.append(rng.integers(0, NG, 100_000))
+    for pat in pats:
+        n = len(pat)
+        gids = torch.from_numpy(np.ascontiguousarray(pat)).to(torch.int64)
+        vals = torch.from_numpy(rng.normal(10, 5, n))
+        validity = torch.from_numpy(rng.random(n) > 0.1)
+        ccpu = Column(dtypes.float64, vals, validity)
+        cgpu = ccpu.to(DEV)
+        for fn in ("sum", "min", "max"):
+            a_ref, c_ref = ops.agg_scatter(gids, NG, ccpu, fn)
+            a_gpu, c_gpu = ops.agg_scatter(gids.to(DEV), NG, cgpu, fn)
+            assert torch.equal(c_ref, c_gpu.cpu()), (fn, n)
+            if fn == "sum":
+                assert torch.allclose(a_ref, a_gpu.cpu(), rtol=1e-9, atol=1e-6)
+            else:
+                assert torch.equal(a_ref, a_gpu.cpu()), (fn, n)
+        icpu = Column(dtypes.int64,
+                      torch.from_numpy(rng.integers(-10**7, 10**7, n)), validity)
+        igpu = icpu.to(DEV)
+        for fn in ("sum", "min", "max"):
+            a_ref, _ = ops.agg_scatter(gids, NG, icpu, fn)
+            a_gpu, _ = ops.agg_scatter(gids.to(DEV), NG, igpu, fn)
+            assert torch.equal(a_ref, a_gpu.cpu()), (fn, n)
