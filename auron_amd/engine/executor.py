@@ -765,8 +765,7 @@ class Executor:
             else:
                 val = agg.expr.eval(b) if agg.expr is not None else None
                 if agg.fn == "count_star":
-                    cnt = torch.zeros(ngroups, dtype=torch.int64, device=device)
-                    cnt.scatter_add_(0, gids, torch.ones(n, dtype=torch.int64, device=device))
+                    cnt = ops.agg_count_star(gids, ngroups)
                     acc, vcnt, vdt = cnt, cnt, dtypes.int64
                 elif agg.fn == "count_distinct":
                     assert node.mode == "complete", "count_distinct needs complete mode (pre-exchanged)"

@@ -240,13 +240,32 @@ __global__ void k_agg_scatter_lds(const int64_t* gids, int64_t n,
   }
 }
 
-// count-only (count_star / count of a validity-masked column)
+// count-only (count_star / count of a validity-masked column);
+// wave-segmented like k_agg_scatter
 __global__ void k_agg_count(const int64_t* gids, int64_t n, const uint8_t* validity,
                             int64_t* counts) {
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += (int64_t)gridDim.x * blockDim.x) {
-    if (validity && !validity[i]) continue;
-    atomicAdd((unsigned long long*)&counts[gids[i]], 1ull);
+  const int lane = threadIdx.x & 63;
+  const int64_t wave_global = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) >> 6;
+  for (int64_t base = wave_global * 64; base < n; base += nwaves * 64) {
+    int64_t i = base + lane;
+    bool live = i < n && (!validity || validity[i]);
+    int64_t g = (i < n) ? gids[i] : (int64_t)-1;
+    int64_t c = live ? 1 : 0;
+    int64_t pg1 = __shfl_up(g, 1, 64);
+    int flag = (lane == 0) || (pg1 != g);
+    for (int d = 1; d < 64; d <<= 1) {
+      int64_t pc = __shfl_up(c, d, 64);
+      int pf = __shfl_up(flag, d, 64);
+      if (lane >= d) {
+        if (!flag) c += pc;
+        flag = flag || pf;
+      }
+    }
+    int64_t ng = __shfl_down(g, 1, 64);
+    bool seg_last = (lane == 63) || (ng != g);
+    if (seg_last && g >= 0 && c > 0)
+      atomicAdd((unsigned long long*)&counts[g], (unsigned long long)c);
   }
 }
 

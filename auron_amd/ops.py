@@ -453,6 +453,25 @@ def _agg_scatter_native(gids: torch.Tensor, num_groups: int, values: Column, fn:
     return acc, counts
 
 
+def agg_count_star(gids: torch.Tensor, num_groups: int) -> torch.Tensor:
+    """Per-group row counts (no values column): native wave-segmented
+    count kernel on GPU, torch scatter_add_ on CPU."""
+    device = gids.device
+    if num_groups > 0 and _use_native(device):
+        lib = native.lib()
+        sp = native.stream_ptr(device)
+        g = gids.contiguous()
+        counts = torch.zeros(num_groups, dtype=torch.int64, device=device)
+        rc = lib.au_agg_count(g.data_ptr(), g.numel(), None,
+                              counts.data_ptr(), sp)
+        native.check(rc, "au_agg_count")
+        return counts
+    cnt = torch.zeros(num_groups, dtype=torch.int64, device=device)
+    if gids.numel():
+        cnt.scatter_add_(0, gids, torch.ones_like(gids, dtype=torch.int64))
+    return cnt
+
+
 def agg_scatter(gids: torch.Tensor, num_groups: int, values: Column, fn: str):
     """Scatter-accumulate values into per-group accumulators.
 
