@@ -45,6 +45,9 @@ SMJ_FALLBACK_ROWS = _opt("spark.auron.smjfallback.rows.threshold", 10_000_000, i
 PARTIAL_AGG_SKIPPING_RATIO = _opt("spark.auron.partialAggSkipping.ratio", 0.999, float)
 UDF_FALLBACK = _opt("spark.auron.udf.hostFallback.enable", True, bool)
 LOG_LEVEL = _opt("spark.auron.native.log.level", "WARN", str, env="AURON_LOG_LEVEL")
+IGNORE_CORRUPTED_FILES = _opt("spark.auron.ignoreCorruptedFiles", False, bool,
+                              "skip unreadable input files instead of failing the task",
+                              env="AURON_IGNORE_CORRUPTED_FILES")
 
 
 class AuronConf:

@@ -84,6 +84,10 @@ def _normalize_join_keys(lkeys, rkeys):
     return lo, ro
 
 
+class TaskCancelled(RuntimeError):
+    """Raised inside execute() after Executor.cancel()."""
+
+
 def _cast_for_range(c: Column) -> Column:
     if c.dtype.is_string:
         raise NotImplementedError("range partitioning on string keys")
@@ -113,6 +117,18 @@ class Executor:
         # MetricNode tree of the last top-level execute (SQLMetrics parity)
         self._metric_stack: List[list] = [[]]
         self.last_metric_tree: Optional[dict] = None
+        # cooperative cancellation (rt.rs is_task_running/cancel_all_tasks
+        # parity): checked at every operator dispatch
+        import threading
+
+        self._cancelled = threading.Event()
+
+    def cancel(self):
+        """Request cancellation; the plan walk aborts at the next operator."""
+        self._cancelled.set()
+
+    def reset_cancel(self):
+        self._cancelled.clear()
 
     # ------------------------------------------------------------- dispatch
     def _rewrite(self, node: P.PlanNode) -> P.PlanNode:
@@ -134,6 +150,8 @@ class Executor:
         ctx = F.EVAL_CONTEXT.get()
         if ctx.get("partition_id") != self.ctx.rank:
             F.EVAL_CONTEXT.set({**ctx, "partition_id": self.ctx.rank})
+        if self._cancelled.is_set():
+            raise TaskCancelled("task cancelled")
         top = len(self._child_time) == 1
         if top:  # top-level call = one query
             self._scan_cache.clear()
@@ -215,12 +233,27 @@ class Executor:
             # keep the >=1 batch invariant: 0-row batch with the file schema
             t = pq.read_table(node.paths[0], columns=node.columns).slice(0, 0)
             return [RecordBatch.from_arrow(t, self.ctx.device)]
+        def read_one(f):
+            try:
+                return self._read_parquet(f, node.columns, node.filters)
+            except Exception:
+                from ..config import IGNORE_CORRUPTED_FILES, AuronConf
+
+                if AuronConf().get(IGNORE_CORRUPTED_FILES):
+                    return None  # conf.rs IGNORE_CORRUPTED_FILES semantics
+                raise
+
         if len(my_files) == 1:
-            return [self._read_parquet(my_files[0], node.columns, node.filters)]
-        # overlap host page reads / chunk staging across files
-        with ThreadPoolExecutor(max_workers=min(8, len(my_files))) as pool:
-            return list(pool.map(
-                lambda f: self._read_parquet(f, node.columns, node.filters), my_files))
+            out = [read_one(my_files[0])]
+        else:
+            # overlap host page reads / chunk staging across files
+            with ThreadPoolExecutor(max_workers=min(8, len(my_files))) as pool:
+                out = list(pool.map(read_one, my_files))
+        out = [b for b in out if b is not None]
+        if not out:
+            t = pq.read_table(node.paths[0], columns=node.columns).slice(0, 0)
+            return [RecordBatch.from_arrow(t, self.ctx.device)]
+        return out
 
     def _exec_OrcScan(self, node: P.OrcScan) -> List[RecordBatch]:
         import pyarrow.orc as orc

@@ -19,6 +19,19 @@ from .engine.executor import ExecContext, Executor
 from .plan import nodes as P
 
 
+class AuronTaskError(RuntimeError):
+    """Task failure with (task_id, stage, partition) context."""
+
+    def __init__(self, task_id, stage_id, partition, cause):
+        super().__init__(
+            f"task {task_id} stage {stage_id} partition {partition} failed: "
+            f"{type(cause).__name__}: {cause}")
+        self.task_id = task_id
+        self.stage_id = stage_id
+        self.partition = partition
+        self.cause = cause
+
+
 class AuronSession:
     def __init__(self, conf: Optional[AuronConf] = None, device=None):
         self.conf = conf or AuronConf()
@@ -39,11 +52,16 @@ class AuronSession:
 
     def execute_serialized(self, task_bytes: bytes):
         """JniBridge.callNative analogue: run a serialized TaskDefinition
-        (plan/serde.py) and return this rank's batches."""
+        (plan/serde.py) and return this rank's batches. Failures surface
+        as AuronTaskError carrying (task, stage, partition) — the
+        reference's setError-upcall contract (rt.rs:315)."""
         from .plan import serde
 
         task_id, stage_id, partition, plan = serde.deserialize_task(task_bytes)
-        return self.executor.execute(plan)
+        try:
+            return self.executor.execute(plan)
+        except Exception as e:
+            raise AuronTaskError(task_id, stage_id, partition, e) from e
 
     def collect(self, plan: P.PlanNode) -> RecordBatch:
         """Run and concat this rank's result (driver-side rows analogue)."""

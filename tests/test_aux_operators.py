@@ -145,3 +145,53 @@ def test_make_array_then_explode():
     proj = P.Project(sc, [Aliased(F.MakeArray([col("a"), col("b")]), "arr")])
     out = AuronSession().collect(P.Generate(proj, "explode", [col("arr")])).to_pydict()
     assert out["col"] == [1, 2, 10, 20]
+
+
+def test_cancel_and_task_error(tmp_path):
+    import pytest as _pytest
+
+    from auron_amd import AuronSession, col, dtypes
+    from auron_amd.column import RecordBatch
+    from auron_amd.engine.executor import TaskCancelled
+    from auron_amd.plan import nodes as P
+    from auron_amd.plan.serde import serialize_task
+    from auron_amd.session import AuronTaskError
+
+    s = AuronSession()
+    plan = P.Filter(P.MemoryScan([RecordBatch.from_pydict(
+        {"x": [1, 2]}, {"x": dtypes.int64})]), col("x") > 0)
+    s.executor.cancel()
+    with _pytest.raises(TaskCancelled):
+        s.collect(plan)
+    s.executor.reset_cancel()
+    assert s.collect(plan).num_rows == 2
+    # serialized task failure carries task context
+    bad = P.ParquetScan([str(tmp_path / "nope.parquet")], columns=["x"])
+    blob = serialize_task("t-1", 3, 7, bad)
+    with _pytest.raises(AuronTaskError) as ei:
+        s.execute_serialized(blob)
+    assert ei.value.stage_id == 3 and ei.value.partition == 7
+
+
+def test_ignore_corrupted_files(tmp_path, monkeypatch):
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    from auron_amd import AuronSession, dtypes
+    from auron_amd.plan import nodes as P
+
+    good = str(tmp_path / "good.parquet")
+    pq.write_table(pa.table({"x": pa.array([1, 2, 3])}), good,
+                   compression="NONE", use_dictionary=False,
+                   data_page_version="1.0")
+    bad = str(tmp_path / "bad.parquet")
+    with open(bad, "wb") as f:
+        f.write(b"not a parquet file")
+    s = AuronSession()
+    plan = P.ParquetScan([good, bad], columns=["x"])
+    import pytest as _pytest
+    with _pytest.raises(Exception):
+        s.collect(plan)
+    monkeypatch.setenv("AURON_IGNORE_CORRUPTED_FILES", "1")
+    out = AuronSession().collect(P.ParquetScan([good, bad], columns=["x"]))
+    assert sorted(out.to_pydict()["x"]) == [1, 2, 3]
