@@ -16,7 +16,7 @@ from __future__ import annotations
 import os
 import time
 from dataclasses import dataclass, field
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Optional
 
 import torch
 import torch.distributed as dist
@@ -25,7 +25,7 @@ from .. import dtypes, ops
 from ..column import Column, RecordBatch
 from ..dtypes import DataType
 from ..exchange import all_gather_batch, all_to_all
-from ..exprs import AggFunc, Aliased, Col, Expr, WindowFunc
+from ..exprs import AggFunc, Col, WindowFunc
 from ..plan import nodes as P
 
 

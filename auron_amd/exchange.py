@@ -12,8 +12,7 @@ is exercised by world_size>1 CPU tests.
 """
 from __future__ import annotations
 
-import math
-from typing import Dict, List, Optional, Tuple
+from typing import List, Optional, Tuple
 
 import torch
 import torch.distributed as dist

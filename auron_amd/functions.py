@@ -8,16 +8,15 @@ through the host (documented inline).
 from __future__ import annotations
 
 import contextvars
-from dataclasses import dataclass, field
-from typing import List, Optional
+from dataclasses import dataclass
+from typing import List
 
 import torch
 
 from . import dtypes, strings
-from .column import Column, RecordBatch
-from .dtypes import DataType
-from .exprs import (Cast, Expr, Literal, _all_valid, _cast_col,
-                    _civil_from_days, combine_validity)
+from .column import Column
+from .exprs import (Expr, _all_valid, _cast_col, _civil_from_days,
+                    combine_validity)
 
 # executor-injected evaluation context (partition id etc.)
 EVAL_CONTEXT: contextvars.ContextVar = contextvars.ContextVar(

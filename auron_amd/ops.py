@@ -9,9 +9,8 @@ auron_amd.native.require().
 """
 from __future__ import annotations
 
-import ctypes
 import os
-from typing import List, Optional, Tuple
+from typing import List, Tuple
 
 import torch
 

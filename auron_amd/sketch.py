@@ -11,7 +11,7 @@ from __future__ import annotations
 
 import math
 import struct
-from typing import List, Optional
+from typing import List
 
 import torch
 
