@@ -120,3 +120,64 @@ def test_murmur3_ref_chunking_invariance(vals):
     parts = ops.murmur3_ref([a]).tolist() + ops.murmur3_ref([b2]).tolist() \
         if vals else []
     assert whole == parts
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.lists(st.tuples(st.integers(0, 3), ints), max_size=60))
+def test_window_row_number_rank_matches_python(rows):
+    ps = [r[0] for r in rows]
+    os_ = [r[1] for r in rows]
+    b = _batch({"p": ps, "o": os_}, {"p": dtypes.int64, "o": dtypes.int64})
+    from auron_amd.exprs import WindowFunc
+
+    fns = [Aliased(WindowFunc("row_number", None), "rn"),
+           Aliased(WindowFunc("rank", None), "rk"),
+           Aliased(WindowFunc("dense_rank", None), "dr")]
+    got = AuronSession().collect(
+        P.Window(P.MemoryScan([b]), [col("p")], [(col("o"), True)], fns)).to_pydict()
+    # python oracle per partition (Spark: asc nulls first)
+    import collections
+
+    parts = collections.defaultdict(list)
+    for p, o in zip(ps, os_):
+        parts[p].append(o)
+    okey = lambda o: (0, 0) if o is None else (1, o)
+    for p, o, rn, rk, dr in zip(got["p"], got["o"], got["rn"], got["rk"], got["dr"]):
+        ordered = sorted(parts[p], key=okey)
+        # rank = 1 + #strictly-smaller; dense_rank = 1 + #distinct smaller
+        smaller = sum(1 for x in ordered if okey(x) < okey(o))
+        distinct_smaller = len({okey(x) for x in ordered if okey(x) < okey(o)})
+        assert rk == smaller + 1, (p, o)
+        assert dr == distinct_smaller + 1, (p, o)
+        assert 1 <= rn <= len(ordered)
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.lists(strs, max_size=40), st.integers(1, 6), st.integers(1, 6))
+def test_substr_upper_length_matches_python(vals, start, ln):
+    b = _batch({"s": vals}, {"s": dtypes.string})
+    from auron_amd.exprs import Length, Substr, Upper
+
+    plan = P.Project(P.MemoryScan([b]), [
+        Aliased(Substr(col("s"), start, ln), "sub"),
+        Aliased(Upper(col("s")), "up"),
+        Aliased(Length(col("s")), "len")])
+    got = AuronSession().collect(plan).to_pydict()
+    for v, sub, up, l in zip(vals, got["sub"], got["up"], got["len"]):
+        if v is None:
+            assert sub is None and up is None and l is None
+        else:
+            enc = v.encode("utf-8")  # SQL substr is 1-based, byte-wise here
+            assert sub == enc[start - 1:start - 1 + ln].decode("utf-8", "replace")
+            assert l == len(enc)
+
+
+@settings(max_examples=30, deadline=None)
+@given(st.lists(ints, max_size=50))
+def test_distinct_union_matches_python(vals):
+    b = _batch({"x": vals}, {"x": dtypes.int64})
+    plan = P.HashAgg(P.Union([P.MemoryScan([b]), P.MemoryScan([b])]),
+                     [Aliased(col("x"), "x")], [], mode="complete")
+    got = AuronSession().collect(plan).to_pydict()["x"]
+    assert sorted(got, key=lambda v: (v is None, v)) == \
+        sorted(set(vals), key=lambda v: (v is None, v))
