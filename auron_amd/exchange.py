@@ -92,7 +92,8 @@ def unpack_batch(meta: dict, buf: torch.Tensor) -> RecordBatch:
     for cm in meta["cols"]:
         code, prec, scale = cm["dtype"]
         dt = DataType(code, prec, scale)
-        td = _TD[code]
+        # LIST stores its element code in `prec`; torch_dtype resolves it
+        td = dt.torch_dtype if code == dtypes.LIST else _TD[code]
 
         def take(nbytes):
             nonlocal pos

@@ -181,3 +181,34 @@ def test_distinct_union_matches_python(vals):
     got = AuronSession().collect(plan).to_pydict()["x"]
     assert sorted(got, key=lambda v: (v is None, v)) == \
         sorted(set(vals), key=lambda v: (v is None, v))
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.lists(ints, max_size=30), st.lists(strs, max_size=30),
+       st.lists(floats, max_size=30))
+def test_exchange_pack_unpack_roundtrip(a, b, c):
+    """pack_batch/unpack_batch must be lossless for every column layout
+    (ints with nulls, strings, floats, decimals, lists, empties)."""
+    from auron_amd.exchange import pack_batch, unpack_batch
+
+    n = min(len(a), len(b), len(c))
+    cols = {
+        "i": (a[:n], dtypes.int64),
+        "s": (b[:n], dtypes.string),
+        "f": (c[:n], dtypes.float64),
+        "d": ([None if v is None else round(v, 2) % 10**5 for v in c[:n]],
+              dtypes.decimal64(9, 2)),
+        "ls": ([None if v is None else [v % 7, v % 11] for v in a[:n]],
+               dtypes.list_of(dtypes.int64)),
+    }
+    rb = RecordBatch([k for k in cols],
+                     [Column.from_pylist(v, t) for k, (v, t) in cols.items()])
+    meta, buf = pack_batch(rb, "cpu")
+    back = unpack_batch(meta, buf)
+    for k in cols:
+        want = rb.column(k).to_pylist()
+        got = back.column(k).to_pylist()
+        if cols[k][1].code == dtypes.DECIMAL64:
+            want = [None if v is None else round(v, 2) for v in want]
+            got = [None if v is None else round(v, 2) for v in got]
+        assert got == want, k

@@ -56,3 +56,25 @@ def test_tpcds_plans_serializable():
     blob = serde.serialize_plan(plan)
     p2 = serde.deserialize_plan(blob)
     assert s.collect(p2).to_pydict() == s.collect(plan).to_pydict()
+
+
+def test_all_99_plans_roundtrip():
+    """Every TPC-DS plan must survive serialize->deserialize with an
+    identical structural fingerprint (planner/serde full-surface check)."""
+    import os
+
+    from auron_amd import AuronSession
+    from auron_amd.plan.serde import deserialize_task, serialize_task
+    from auron_amd.tpcds import datagen
+    from auron_amd.tpcds.queries import QUERIES, Catalog
+    from tests.test_plan_stability import _shape
+
+    root = os.path.join(os.path.dirname(__file__), "..", ".tpcds_cache")
+    datagen.write_dataset(root, 0.01)
+    s = AuronSession()
+    cat = Catalog(root, 0.01)
+    for qn in sorted(QUERIES):
+        plan = QUERIES[qn](cat, s)
+        blob = serialize_task("t", 0, 0, plan)
+        _, _, _, back = deserialize_task(blob)
+        assert _shape(back) == _shape(plan), qn
