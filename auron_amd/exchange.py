@@ -140,15 +140,13 @@ def all_to_all(batches_by_dest: List[Optional[RecordBatch]], device,
             m, buf = pack_batch(b, device)
             metas.append(m)
             bufs.append(buf)
-    # meta exchange: all_gather_object of my per-dest meta list
-    all_metas: List[List[Optional[dict]]] = [None] * world  # type: ignore
-    dist.all_gather_object(all_metas, metas, group=group)
-    recv_metas = [all_metas[s][rank] for s in range(world)]
-
-    sizes = torch.tensor([b.numel() for b in bufs], dtype=torch.int64)
-    all_sizes = [torch.zeros(world, dtype=torch.int64) for _ in range(world)]
-    dist.all_gather_object(all_sizes, sizes, group=group)
-    recv_sizes = [int(all_sizes[s][rank].item()) for s in range(world)]
+    # ONE object collective carries both per-dest metas and buffer sizes
+    # (object all-gathers are host round-trips; halving them matters at N=8)
+    payload = (metas, [b.numel() for b in bufs])
+    gathered: List[Optional[tuple]] = [None] * world  # type: ignore
+    dist.all_gather_object(gathered, payload, group=group)
+    recv_metas = [gathered[s][0][rank] for s in range(world)]
+    recv_sizes = [int(gathered[s][1][rank]) for s in range(world)]
 
     backend = dist.get_backend(group)
     out_batches: List[RecordBatch] = []
@@ -200,10 +198,10 @@ def all_gather_batch(batch: Optional[RecordBatch], device, group=None) -> List[R
         meta, buf = pack_batch(batch, device)
     else:
         meta, buf = None, torch.zeros(0, dtype=torch.uint8, device=device)
-    all_meta: List[Optional[dict]] = [None] * world  # type: ignore
-    dist.all_gather_object(all_meta, meta, group=group)
-    sizes: List[Optional[int]] = [None] * world  # type: ignore
-    dist.all_gather_object(sizes, buf.numel(), group=group)
+    gathered: List[Optional[tuple]] = [None] * world  # type: ignore
+    dist.all_gather_object(gathered, (meta, buf.numel()), group=group)
+    all_meta = [g[0] for g in gathered]
+    sizes = [g[1] for g in gathered]
     maxsz = max(sizes) if sizes else 0
     padded = torch.zeros(max(maxsz, 1), dtype=torch.uint8, device=device)
     if buf.numel():
