@@ -218,3 +218,22 @@ def test_agg_scatter_segmented_path_patterns():
             a_gpu, _ = ops.agg_scatter(gids.to(DEV), NG, igpu, fn)
             nz = c_ref > 0
             assert torch.equal(a_ref[nz], a_gpu.cpu()[nz]), (fn, n)
+
+
+def test_memmgr_spill_restore_on_device():
+    """HBM -> host spill and restore round-trip with a tiny budget."""
+    from auron_amd.memory import MemManager
+
+    def mk(tag, rows):
+        data = torch.arange(rows, dtype=torch.int64, device=DEV)
+        return [RecordBatch(["x"], [Column(dtypes.int64, data)])]
+
+    mgr = MemManager(budget_bytes=1 << 20)  # 1 MiB
+    h1 = mgr.register("h1", mk("a", 100_000))  # 800 KB
+    h2 = mgr.register("h2", mk("b", 100_000))  # forces h1 to spill
+    assert not h1.resident and h2.resident
+    assert mgr.metrics.get("spill_d2h_bytes", 0) > 0
+    back = h1.batches()  # restore (spills h2 in turn)
+    assert back[0].columns[0].data.is_cuda
+    assert int(back[0].columns[0].data[12345].item()) == 12345
+    h1.release(); h2.release()
