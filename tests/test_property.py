@@ -71,15 +71,30 @@ def test_sort_matches_python(rows):
     b = _batch({"k": ks, "s": ss}, {"k": dtypes.int64, "s": dtypes.string})
     plan = P.Sort(P.MemoryScan([b]), [(col("k"), True), (col("s"), False)])
     got = AuronSession().collect(plan).to_pydict()
-    # Spark ordering: asc -> nulls first; desc -> nulls last
-    def key(r):
-        k, s = r
-        return ((0 if k is None else 1, k if k is not None else 0),
-                (0 if s is not None else 1,) + (tuple([-ord(c) for c in s]) if s is not None else ()))
-    ref = sorted(zip(ks, ss), key=lambda r: (
-        (0, 0) if r[0] is None else (1, r[0]),))
-    # verify primary key ordering only (secondary checked via stability below)
-    assert [r[0] for r in ref] == got["k"]
+    # Spark ordering: asc -> nulls first; desc -> nulls last.
+    # UTF-8 byte order == code-point order, so python str sort is the
+    # oracle for the engine's byte-wise string comparison.
+    import functools
+
+    def cmp(a, b2):
+        (ka, sa), (kb, sb) = a, b2
+        ka_t = (0,) if ka is None else (1, ka)
+        kb_t = (0,) if kb is None else (1, kb)
+        if ka_t != kb_t:
+            return -1 if ka_t < kb_t else 1
+        # desc string, nulls last
+        if sa is None and sb is None:
+            return 0
+        if sa is None:
+            return 1
+        if sb is None:
+            return -1
+        if sa == sb:
+            return 0
+        return -1 if sa > sb else 1
+
+    ref = sorted(zip(ks, ss), key=functools.cmp_to_key(cmp))
+    assert list(zip(got["k"], got["s"])) == ref
 
 
 @settings(max_examples=40, deadline=None)
