@@ -788,6 +788,20 @@ class Executor:
             out_keys = []
         names = [a.name for a in node.keys]
         cols = list(out_keys)
+        # fuse plain sum/avg/min/max aggregates into ONE kernel pass
+        # (k_agg_multi): gids are read once, head flags computed once
+        fused: Dict[int, tuple] = {}
+        if node.mode in ("partial", "complete") and ngroups > 0:
+            fuse_items = []
+            fuse_idx = []
+            for i, agg in enumerate(node.aggs):
+                if agg.fn in ("sum", "avg", "min", "max") and agg.expr is not None:
+                    fuse_items.append((agg.expr.eval(b), agg.fn))
+                    fuse_idx.append(i)
+            if len(fuse_items) > 1:
+                for i, res in zip(fuse_idx,
+                                  ops.agg_scatter_multi(gids, ngroups, fuse_items)):
+                    fused[i] = (fuse_items[fuse_idx.index(i)][0], res)
         for i, agg in enumerate(node.aggs):
             s0, s1 = f"__agg{i}_0", f"__agg{i}_1"
             if node.mode == "final":
@@ -824,6 +838,9 @@ class Executor:
                 elif agg.fn == "first":
                     acc_col, cnt = self._agg_first(gids, ngroups, val)
                     acc, vcnt, vdt = acc_col, cnt, val.dtype
+                elif i in fused:
+                    fval, (facc, fcnt) = fused[i]
+                    acc, vcnt, vdt = facc, fcnt, fval.dtype
                 else:
                     fn = {"sum": "sum", "avg": "sum", "min": "min", "max": "max", "count": "count"}[agg.fn]
                     acc, vcnt, vdt = *ops.agg_scatter(gids, ngroups, val, fn), val.dtype

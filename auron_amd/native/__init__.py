@@ -56,7 +56,8 @@ def _try_load():
     lib.au_agg_scatter.argtypes = [c, i64, c, c, ctypes.c_int, ctypes.c_int,
                                    ctypes.c_int, c, c, i64, c]
     lib.au_agg_count.argtypes = [c, i64, c, c, c]
-    for f in ("au_agg_scatter", "au_agg_count"):
+    lib.au_agg_multi.argtypes = [c, i64, c, ctypes.c_int, c]
+    for f in ("au_agg_scatter", "au_agg_count", "au_agg_multi"):
         getattr(lib, f).restype = ctypes.c_int
     lib.au_pq_rle1.argtypes = [c, ctypes.c_int, c, c, c]
     lib.au_pq_rle_idx.argtypes = [c, ctypes.c_int, c, c, c, c]

@@ -240,6 +240,115 @@ __global__ void k_agg_scatter_lds(const int64_t* gids, int64_t n,
   }
 }
 
+// ----------------------------------------------------------- fused multi
+// All aggregates of one HashAgg in a single pass: gids are read once and
+// the wave head-flags are computed once; each aggregate folds through
+// the same segmented scan and flushes its own accumulator.
+// AggDesc layout (int64 x 6, packed host-side):
+//   [0]=values ptr [1]=validity ptr(0=none) [2]=vtype(0 f64,1 i64,2 i32,3 f32)
+//   [3]=op(0 sum,1 min,2 max) [4]=acc ptr [5]=counts ptr(0=none)
+struct AggDesc {
+  const void* values;
+  const uint8_t* validity;
+  int64_t vtype;
+  int64_t op;
+  void* acc;
+  int64_t* counts;
+};
+
+__device__ __forceinline__ double agg_load_f64(const void* v, int64_t vt, int64_t i) {
+  switch (vt) {
+    case 0: return ((const double*)v)[i];
+    case 3: return (double)((const float*)v)[i];
+    case 2: return (double)((const int32_t*)v)[i];
+    default: return (double)((const int64_t*)v)[i];
+  }
+}
+
+__device__ __forceinline__ int64_t agg_load_i64(const void* v, int64_t vt, int64_t i) {
+  return vt == 2 ? (int64_t)((const int32_t*)v)[i] : ((const int64_t*)v)[i];
+}
+
+__global__ void k_agg_multi(const int64_t* gids, int64_t n, const AggDesc* descs,
+                            int ndescs) {
+  const int lane = threadIdx.x & 63;
+  const int64_t wave_global = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) >> 6;
+  for (int64_t base = wave_global * 64; base < n; base += nwaves * 64) {
+    int64_t i = base + lane;
+    int64_t g = (i < n) ? gids[i] : (int64_t)-1;
+    int64_t pg1 = __shfl_up(g, 1, 64);
+    int head = (lane == 0) || (pg1 != g);
+    int64_t ng = __shfl_down(g, 1, 64);
+    bool seg_last = (lane == 63) || (ng != g);
+    for (int a = 0; a < ndescs; a++) {
+      const AggDesc d = descs[a];
+      bool live = i < n && (!d.validity || d.validity[i]);
+      bool f64p = d.vtype == 0 || d.vtype == 3;
+      int64_t c = live ? 1 : 0;
+      int flag = head;
+      if (f64p) {
+        double ident = d.op == 0 ? 0.0 : (d.op == 1 ? INFINITY : -INFINITY);
+        double v = live ? agg_load_f64(d.values, d.vtype, i) : ident;
+        for (int dd = 1; dd < 64; dd <<= 1) {
+          double pv = __shfl_up(v, dd, 64);
+          int64_t pc = __shfl_up(c, dd, 64);
+          int pf = __shfl_up(flag, dd, 64);
+          if (lane >= dd) {
+            if (!flag) {
+              if (d.op == 0) v += pv;
+              else if (d.op == 1) v = v < pv ? v : pv;
+              else v = v > pv ? v : pv;
+              c += pc;
+            }
+            flag = flag || pf;
+          }
+        }
+        if (seg_last && g >= 0 && c > 0) {
+          double* acc = (double*)d.acc;
+          if (d.counts) atomicAdd((unsigned long long*)&d.counts[g], (unsigned long long)c);
+          if (d.op == 0) atomicAdd(&acc[g], v);
+          else if (d.op == 1) atomic_min_f64(&acc[g], v);
+          else atomic_max_f64(&acc[g], v);
+        }
+      } else {
+        int64_t ident = d.op == 0 ? 0 : (d.op == 1 ? INT64_MAX : INT64_MIN);
+        int64_t v = live ? agg_load_i64(d.values, d.vtype, i) : ident;
+        for (int dd = 1; dd < 64; dd <<= 1) {
+          int64_t pv = __shfl_up(v, dd, 64);
+          int64_t pc = __shfl_up(c, dd, 64);
+          int pf = __shfl_up(flag, dd, 64);
+          if (lane >= dd) {
+            if (!flag) {
+              if (d.op == 0) v += pv;
+              else if (d.op == 1) v = v < pv ? v : pv;
+              else v = v > pv ? v : pv;
+              c += pc;
+            }
+            flag = flag || pf;
+          }
+        }
+        if (seg_last && g >= 0 && c > 0) {
+          int64_t* acc = (int64_t*)d.acc;
+          if (d.counts) atomicAdd((unsigned long long*)&d.counts[g], (unsigned long long)c);
+          if (d.op == 0) atomicAdd((unsigned long long*)&acc[g], (unsigned long long)v);
+          else if (d.op == 1) atomic_min_i64(&acc[g], v);
+          else atomic_max_i64(&acc[g], v);
+        }
+      }
+    }
+  }
+}
+
+AU_EXPORT int au_agg_multi(const int64_t* gids, int64_t n, const void* descs_dev,
+                           int ndescs, void* stream) {
+  if (n == 0 || ndescs == 0) return 0;
+  hipLaunchKernelGGL(k_agg_multi, dim3(agg_grid(n)), dim3(256), 0,
+                     (hipStream_t)stream, gids, n, (const AggDesc*)descs_dev,
+                     ndescs);
+  return (int)hipGetLastError();
+}
+
 // count-only (count_star / count of a validity-masked column);
 // wave-segmented like k_agg_scatter
 __global__ void k_agg_count(const int64_t* gids, int64_t n, const uint8_t* validity,

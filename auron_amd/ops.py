@@ -452,6 +452,70 @@ def _agg_scatter_native(gids: torch.Tensor, num_groups: int, values: Column, fn:
     return acc, counts
 
 
+def agg_scatter_multi(gids: torch.Tensor, num_groups: int, items):
+    """Fused accumulate of several (values Column, fn) pairs in ONE kernel
+    pass over the rows (gids read once, wave head-flags computed once).
+    fn in sum/avg/min/max. -> [(acc, counts), ...] like agg_scatter."""
+    device = gids.device
+    if not items:
+        return []
+    if num_groups == 0 or not _use_native(device) or len(items) == 1:
+        return [agg_scatter(gids, num_groups, c, f) for c, f in items]
+    import numpy as np
+
+    lib = native.lib()
+    sp = native.stream_ptr(device)
+    g = gids.contiguous()
+    n = g.numel()
+    descs = np.zeros((len(items), 6), dtype=np.int64)
+    outs = []
+    keep = [g]
+    for j, (values, fn) in enumerate(items):
+        v = values.data
+        if v.dtype in (torch.int8, torch.int16, torch.bool):
+            v = v.to(torch.int32)
+        v = v.contiguous()
+        keep.append(v)
+        vtype = {torch.float64: 0, torch.int64: 1, torch.int32: 2,
+                 torch.float32: 3}[v.dtype]
+        acc_f64 = v.dtype in (torch.float32, torch.float64)
+        op = {"sum": 0, "avg": 0, "min": 1, "max": 2}[fn]
+        if op == 0:
+            init = 0.0 if acc_f64 else 0
+        elif op == 1:
+            init = float("inf") if acc_f64 else torch.iinfo(torch.int64).max
+        else:
+            init = float("-inf") if acc_f64 else torch.iinfo(torch.int64).min
+        acc = torch.full((num_groups,), init,
+                         dtype=torch.float64 if acc_f64 else torch.int64,
+                         device=device)
+        counts = torch.zeros(num_groups, dtype=torch.int64, device=device)
+        vptr = 0
+        if values.validity is not None:
+            val = values.validity.contiguous()
+            keep.append(val)
+            vptr = val.data_ptr()
+        descs[j] = (v.data_ptr(), vptr, vtype, op, acc.data_ptr(),
+                    counts.data_ptr())
+        outs.append((values, fn, acc, counts))
+    from .pinned import to_device
+
+    darr = to_device(descs.reshape(-1), device)
+    keep.append(darr)
+    rc = lib.au_agg_multi(g.data_ptr(), n, darr.data_ptr(), len(items), sp)
+    native.check(rc, "au_agg_multi")
+    del keep
+    res = []
+    for values, fn, acc, counts in outs:
+        if fn in ("min", "max"):
+            tgt = values.data.dtype if values.data.dtype not in \
+                (torch.int8, torch.int16, torch.bool) else torch.int64
+            if acc.dtype != tgt:
+                acc = acc.to(tgt)
+        res.append((acc, counts))
+    return res
+
+
 def agg_count_star(gids: torch.Tensor, num_groups: int) -> torch.Tensor:
     """Per-group row counts (no values column): native wave-segmented
     count kernel on GPU, torch scatter_add_ on CPU."""

@@ -237,3 +237,35 @@ def test_memmgr_spill_restore_on_device():
     assert back[0].columns[0].data.is_cuda
     assert int(back[0].columns[0].data[12345].item()) == 12345
     h1.release(); h2.release()
+
+
+def test_agg_multi_fused_matches_single():
+    """k_agg_multi must equal per-agg scatters for mixed dtypes/ops."""
+    rng = np.random.default_rng(13)
+    n = 300_000
+    gids = torch.from_numpy(np.repeat(rng.integers(0, 20_000, 6000),
+                                      rng.integers(1, 100, 6000))[:n]).to(torch.int64)
+    n = gids.numel()
+    items = []
+    for dt, tdt in ((dtypes.float64, None), (dtypes.int64, None),
+                    (dtypes.float32, np.float32), (dtypes.int32, np.int32)):
+        raw = rng.normal(50, 20, n)
+        if dt in (dtypes.int64, dtypes.int32):
+            data = torch.from_numpy(raw.astype(np.int64 if dt == dtypes.int64 else np.int32))
+        else:
+            data = torch.from_numpy(raw.astype(np.float64 if dt == dtypes.float64 else np.float32))
+        validity = torch.from_numpy(rng.random(n) > 0.07)
+        items.append(Column(dt, data, validity))
+    NG = 20_000
+    fns = ["sum", "min", "max", "avg"]
+    pairs = [(c.to(DEV), f) for c, f in zip(items, fns)]
+    fused = ops.agg_scatter_multi(gids.to(DEV), NG, pairs)
+    for (c, f), (facc, fcnt) in zip(pairs, fused):
+        sacc, scnt = ops.agg_scatter(gids.to(DEV), NG, c, f)
+        assert torch.equal(fcnt, scnt), f
+        nz = (scnt > 0).cpu()
+        if f in ("sum", "avg"):
+            assert torch.allclose(facc.cpu()[nz], sacc.cpu()[nz],
+                                  rtol=1e-9, atol=1e-6), f
+        else:
+            assert torch.equal(facc.cpu()[nz], sacc.cpu()[nz]), f
