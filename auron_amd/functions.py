@@ -318,6 +318,68 @@ class WeekOfYear(Expr):
         return Column(dtypes.int32, week.to(torch.int32), c.validity)
 
 
+@dataclass(eq=False)
+class MonthsBetween(Expr):
+    """Spark months_between(end, start): whole months + day-fraction/31;
+    both-on-last-day and same-day-of-month give integral results."""
+    end: Expr
+    start: Expr
+
+    def eval(self, batch):
+        e = self.end.eval(batch)
+        s = self.start.eval(batch)
+        ye, me, de = _civil_from_days(e.data)
+        ys, ms, ds = _civil_from_days(s.data)
+        months = (ye - ys) * 12 + (me - ms)
+        last_e = de == _days_in_month(ye, me)
+        last_s = ds == _days_in_month(ys, ms)
+        frac = (de - ds).to(torch.float64) / 31.0
+        res = months.to(torch.float64) + torch.where(
+            (de == ds) | (last_e & last_s), torch.zeros_like(frac), frac)
+        return Column(dtypes.float64, res, combine_validity(e, s))
+
+
+@dataclass(eq=False)
+class NextDay(Expr):
+    """next_day(date, dow): first date AFTER `date` falling on weekday
+    `dow` ('monday'..'sunday', Spark accepts 2-letter+ prefixes)."""
+    child: Expr
+    dow: str
+
+    def eval(self, batch):
+        names = ["monday", "tuesday", "wednesday", "thursday", "friday",
+                 "saturday", "sunday"]
+        want = next(i for i, n in enumerate(names)
+                    if n.startswith(self.dow.lower()[:2]))
+        c = self.child.eval(batch)
+        days = c.data.to(torch.int64)
+        cur = torch.remainder(days + 3, 7)  # 0 = Monday
+        delta = torch.remainder(want - cur + 7 - 1, 7) + 1
+        return Column(dtypes.date32, (days + delta).to(torch.int32), c.validity)
+
+
+@dataclass(eq=False)
+class TruncDate(Expr):
+    """trunc(date, 'year'|'month'|'week')."""
+    child: Expr
+    unit: str = "month"
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        days = c.data.to(torch.int64)
+        if self.unit.lower() in ("year", "yy", "yyyy"):
+            y, m, d = _civil_from_days(days)
+            out = _days_from_civil(y, torch.ones_like(y), torch.ones_like(y))
+        elif self.unit.lower() in ("month", "mm", "mon"):
+            y, m, d = _civil_from_days(days)
+            out = _days_from_civil(y, m, torch.ones_like(y))
+        elif self.unit.lower() == "week":  # Monday of this week
+            out = days - torch.remainder(days + 3, 7)
+        else:
+            raise ValueError(f"trunc unit {self.unit}")
+        return Column(dtypes.date32, out.to(torch.int32), c.validity)
+
+
 # ----------------------------------------------------------------- strings
 @dataclass(eq=False)
 class Trim(Expr):
@@ -439,6 +501,93 @@ class RPad(Expr):
                 need = self.length - len(v)
                 out.append(v + (self.pad * need)[:need])
         return Column.from_pylist(out, dtypes.string, str(c.device))
+
+
+@dataclass(eq=False)
+class ConcatWs(Expr):
+    """concat_ws(sep, args...): null args are SKIPPED (Spark), unlike
+    concat which nulls the whole row."""
+    sep: str
+    args: List[Expr]
+
+    def eval(self, batch):
+        cols = [a.eval(batch) for a in self.args]
+        lists = [c.to_pylist() for c in cols]
+        out = [self.sep.join(v for v in row if v is not None)
+               for row in zip(*lists)] if lists else [""] * batch.num_rows
+        return Column.from_pylist(out, dtypes.string, str(batch.device))
+
+
+@dataclass(eq=False)
+class Reverse(Expr):
+    child: Expr
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        out = [None if v is None else v[::-1] for v in c.to_pylist()]
+        return Column.from_pylist(out, dtypes.string, str(c.device))
+
+
+@dataclass(eq=False)
+class Repeat(Expr):
+    child: Expr
+    n: int
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        k = max(self.n, 0)
+        out = [None if v is None else v * k for v in c.to_pylist()]
+        return Column.from_pylist(out, dtypes.string, str(c.device))
+
+
+@dataclass(eq=False)
+class Space(Expr):
+    n: Expr
+
+    def eval(self, batch):
+        c = self.n.eval(batch)
+        out = [None if v is None else " " * max(int(v), 0)
+               for v in c.to_pylist()]
+        return Column.from_pylist(out, dtypes.string, str(batch.device))
+
+
+@dataclass(eq=False)
+class Translate(Expr):
+    """translate(s, from, to): per-character mapping; chars beyond `to`'s
+    length are deleted (SQL semantics)."""
+    child: Expr
+    src: str
+    dst: str
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        table = {}
+        for i, ch in enumerate(self.src):
+            table[ord(ch)] = self.dst[i] if i < len(self.dst) else None
+        out = [None if v is None else v.translate(table) for v in c.to_pylist()]
+        return Column.from_pylist(out, dtypes.string, str(c.device))
+
+
+@dataclass(eq=False)
+class FindInSet(Expr):
+    """find_in_set(s, csv): 1-based index of s in comma-separated csv
+    column; 0 if absent or s contains a comma."""
+    child: Expr
+    csv: Expr
+
+    def eval(self, batch):
+        s = self.child.eval(batch)
+        l = self.csv.eval(batch)
+        out = []
+        for v, lst in zip(s.to_pylist(), l.to_pylist()):
+            if v is None or lst is None:
+                out.append(None)
+            elif "," in v:
+                out.append(0)
+            else:
+                parts = lst.split(",")
+                out.append(parts.index(v) + 1 if v in parts else 0)
+        return Column.from_pylist(out, dtypes.int32, str(batch.device))
 
 
 @dataclass(eq=False)

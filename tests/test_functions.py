@@ -197,3 +197,36 @@ def test_trycast_and_rownum():
     b = _b(s=(["12", "x", None, "3.5"], dtypes.string))
     assert ev(TryCast(col("s"), dtypes.int64), b) == [12, None, None, 3]
     assert ev(F.RowNum(), b) == [1, 2, 3, 4]
+
+
+def test_concat_ws_reverse_repeat_space():
+    b = _b(a=(["x", None, "p"], dtypes.string), c=(["y", "z", None], dtypes.string))
+    assert ev(F.ConcatWs("-", [col("a"), col("c")]), b) == ["x-y", "z", "p"]
+    assert ev(F.Reverse(col("a")), b) == ["x", None, "p"]
+    b2 = _b(s=(["ab", None], dtypes.string))
+    assert ev(F.Repeat(col("s"), 3), b2) == ["ababab", None]
+    b3 = _b(n=([2, 0, None], dtypes.int64))
+    assert ev(F.Space(col("n")), b3) == ["  ", "", None]
+
+
+def test_translate_find_in_set():
+    b = _b(s=(["abcba", None], dtypes.string))
+    assert ev(F.Translate(col("s"), "abc", "AB"), b) == ["ABBA", None]
+    b2 = _b(s=(["b", "d", "a,b", None], dtypes.string),
+            l=(["a,b,c", "a,b,c", "a,b,c", "a"], dtypes.string))
+    assert ev(F.FindInSet(col("s"), col("l")), b2) == [2, 0, 0, None]
+
+
+def test_months_between_next_day_trunc():
+    b = _b(a=_dcol("2001-03-31", "2001-03-15", "2001-03-10"),
+           c=_dcol("2001-02-28", "2001-01-15", "2001-01-20"))
+    r = ev(F.MonthsBetween(col("a"), col("c")), b)
+    assert r[0] == 1.0 and r[1] == 2.0
+    assert abs(r[2] - (2 + (10 - 20) / 31.0)) < 1e-9
+    b2 = _b(d=_dcol("2001-01-01"))  # a Monday
+    assert ev(F.NextDay(col("d"), "monday"), b2) == [_date("2001-01-08")]
+    assert ev(F.NextDay(col("d"), "tu"), b2) == [_date("2001-01-02")]
+    b3 = _b(d=_dcol("2001-03-15"))
+    assert ev(F.TruncDate(col("d"), "year"), b3) == [_date("2001-01-01")]
+    assert ev(F.TruncDate(col("d"), "month"), b3) == [_date("2001-03-01")]
+    assert ev(F.TruncDate(col("d"), "week"), b3) == [_date("2001-03-12")]
