@@ -869,6 +869,9 @@ def _ship_oracle(root, sf, fact, pre, date0, state, site, rets, rpre):
     allf = _read(root, sf, fact, [f"{pre}_order_number", f"{pre}_warehouse_sk"])
     wh_per_order = allf.dropna().drop_duplicates().groupby(f"{pre}_order_number").size()
     multi = set(wh_per_order[wh_per_order > 1].index)
+    # `wh1 <> wh2` is never true for a NULL wh1: the row itself must have a
+    # non-null warehouse to satisfy the EXISTS
+    j = j[j[f"{pre}_warehouse_sk"].notna()]
     j = j[j[f"{pre}_order_number"].isin(multi)]
     ret = _read(root, sf, rets, [f"{rpre}_order_number"])
     j = j[~j[f"{pre}_order_number"].isin(set(ret[f"{rpre}_order_number"].dropna()))]

@@ -1058,8 +1058,14 @@ def _ship_q(cat, s, fact, pre, date0, state, site_join, returns_table, ret_pre):
         fk, table, tkey, pred = site_join
         site = P.Filter(cat.scan(table, [tkey] + pred[0]), pred[1])
         j = bhj(j, site, [fk], [tkey])
-    # EXISTS (same order, different warehouse): orders with >1 distinct wh
-    all_orders = cat.scan(fact, [f"{pre}_order_number", f"{pre}_warehouse_sk"])
+    # EXISTS (same order, DIFFERENT warehouse): `wh1 <> wh2` is never true
+    # when either side is NULL, so (a) a NULL-warehouse row can never
+    # satisfy the EXISTS and (b) NULL warehouses don't count toward the
+    # distinct-warehouse tally. Survivors = rows with a NON-NULL warehouse
+    # whose order has >= 2 distinct non-null warehouses.
+    all_orders = P.Filter(
+        cat.scan(fact, [f"{pre}_order_number", f"{pre}_warehouse_sk"]),
+        col(f"{pre}_warehouse_sk").is_not_null())
     ord_wh = agg2(P.Project(all_orders, [_a(col(f"{pre}_order_number"), "o"),
                                          _a(col(f"{pre}_warehouse_sk"), "w")]),
                   ["o", "w"], [])
@@ -1069,6 +1075,7 @@ def _ship_q(cat, s, fact, pre, date0, state, site_join, returns_table, ret_pre):
         P.HashAgg(P.Exchange(ord_wh, "hash", [col("o")]), [_a(col("o"), "o")],
                   [AggFunc("count_star", None, name="nwh")], mode="complete"),
         col("nwh") > 1)
+    j = P.Filter(j, col(f"{pre}_warehouse_sk").is_not_null())
     j = P.HashJoin(P.Exchange(j, "hash", [col(f"{pre}_order_number")]),
                    P.Exchange(multi, "hash", [col("o")]),
                    [col(f"{pre}_order_number")], [col("o")], how="semi", build_side="right")

@@ -11,7 +11,7 @@ from auron_amd.tpcds import datagen
 from auron_amd.tpcds.oracle import ORACLES
 from auron_amd.tpcds.queries import QUERIES, Catalog
 
-SF = 0.01
+SF = float(os.environ.get("AURON_TEST_SF", "0.01"))
 ROOT = os.path.join(os.path.dirname(__file__), "..", ".tpcds_cache")
 
 
