@@ -16,7 +16,7 @@ QUERIES_TO_CHECK = ["q3", "q23", "q72", "q38", "q47", "q5", "q1", "q88",
                     "q4", "q9", "q11", "q18", "q22", "q27", "q33", "q36",
                     "q44", "q49", "q54", "q57", "q64", "q67", "q70", "q75",
                     "q77", "q78", "q80", "q87", "q93", "q97"]
-SF = 0.01
+SF = float(os.environ.get("AURON_TEST_SF", "0.01"))
 ROOT = os.path.join(os.path.dirname(__file__), "..", ".tpcds_cache")
 
 
