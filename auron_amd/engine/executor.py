@@ -835,6 +835,14 @@ class Executor:
                     if og.numel():
                         cnt.scatter_add_(0, og, torch.ones(og.numel(), dtype=torch.int64, device=device))
                     acc, vcnt, vdt = cnt, cnt, dtypes.int64
+                elif agg.fn in ("collect_list", "collect_set"):
+                    assert node.mode == "complete", \
+                        "collect_* needs complete mode (pre-exchanged)"
+                    acc_col = self._agg_collect(gids, ngroups, val,
+                                                dedupe=agg.fn == "collect_set",
+                                                key_cols=key_cols)
+                    cnt = ops.agg_scatter(gids, ngroups, val, "count")[1]
+                    acc, vcnt, vdt = acc_col, cnt, acc_col.dtype
                 elif agg.fn == "first":
                     acc_col, cnt = self._agg_first(gids, ngroups, val)
                     acc, vcnt, vdt = acc_col, cnt, val.dtype
@@ -896,7 +904,36 @@ class Executor:
             return dtypes.float64
         return vdt  # min/max/first keep input type
 
+    def _agg_collect(self, gids, ngroups, val: Column, dedupe: bool,
+                     key_cols) -> Column:
+        """collect_list / collect_set -> LIST column, one row per group
+        (agg/collect.rs analogue). Null inputs excluded; a group with no
+        collected values yields an EMPTY list (Spark semantics). Order
+        within a list is unspecified (as in Spark)."""
+        device = gids.device
+        assert not val.dtype.uses_offsets, \
+            "collect over string/list children: round 2"
+        keep = val.validity if val.validity is not None else \
+            torch.ones(len(val), dtype=torch.bool, device=device)
+        if dedupe:
+            base = key_cols if key_cols else [Column(dtypes.int64, gids)]
+            _, d_reps = ops.group_ids(base + [val])
+            sel = torch.zeros(len(val), dtype=torch.bool, device=device)
+            sel[d_reps] = True
+            keep = keep & sel
+        idx = torch.nonzero(keep, as_tuple=False).flatten()
+        g = gids[idx]
+        order = torch.argsort(g, stable=True)
+        rows = idx[order]
+        lens = torch.bincount(g[order], minlength=ngroups)
+        offsets = torch.zeros(ngroups + 1, dtype=torch.int64, device=device)
+        torch.cumsum(lens, 0, out=offsets[1:])
+        return Column(dtypes.list_of(val.dtype), val.data[rows], None,
+                      offsets.to(torch.int32))
+
     def _finalize_agg(self, agg: AggFunc, vdt: DataType, acc, cnt: torch.Tensor) -> Column:
+        if agg.fn in ("collect_list", "collect_set"):
+            return acc  # empty groups stay empty lists, not null
         validity = cnt > 0
         if bool(validity.all()):
             validity = None

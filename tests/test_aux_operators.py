@@ -195,3 +195,25 @@ def test_ignore_corrupted_files(tmp_path, monkeypatch):
     monkeypatch.setenv("AURON_IGNORE_CORRUPTED_FILES", "1")
     out = AuronSession().collect(P.ParquetScan([good, bad], columns=["x"]))
     assert sorted(out.to_pydict()["x"]) == [1, 2, 3]
+
+
+def test_collect_list_and_set():
+    from auron_amd import AggFunc, AuronSession, col, dtypes
+    from auron_amd.column import RecordBatch
+    from auron_amd.exprs import Aliased
+    from auron_amd.plan import nodes as P
+
+    data = {"k": ["a", "a", "b", "a", "b", "c"],
+            "v": [1, 2, 5, 2, 5, None]}
+    t = {"k": dtypes.string, "v": dtypes.int64}
+    sc = P.MemoryScan([RecordBatch.from_pydict(data, t)])
+    plan = P.HashAgg(sc, [Aliased(col("k"), "k")],
+                     [AggFunc("collect_list", col("v"), name="cl"),
+                      AggFunc("collect_set", col("v"), name="cs")],
+                     mode="complete")
+    out = AuronSession().collect(plan).to_pydict()
+    m = {k: (sorted(cl), sorted(cs)) for k, cl, cs in
+         zip(out["k"], out["cl"], out["cs"])}
+    assert m["a"] == ([1, 2, 2], [1, 2])
+    assert m["b"] == ([5, 5], [5])
+    assert m["c"] == ([], [])  # only-null group -> empty list, not null
