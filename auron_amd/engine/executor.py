@@ -812,7 +812,8 @@ class Executor:
                     cols.append(Column(dtypes.int64, merged_cnt))
                     names.append(agg.name)
                     continue
-                comb = {"sum": "sum", "avg": "sum", "min": "min", "max": "max", "first": "first"}[agg.fn]
+                comb = {"sum": "sum", "avg": "sum", "min": "min", "max": "max",
+                        "first": "first", "first_ignores_null": "first"}[agg.fn]
                 if comb == "first":
                     acc, cnt = self._agg_first(gids, ngroups, sv)
                 else:
@@ -843,7 +844,7 @@ class Executor:
                                                 key_cols=key_cols)
                     cnt = ops.agg_scatter(gids, ngroups, val, "count")[1]
                     acc, vcnt, vdt = acc_col, cnt, acc_col.dtype
-                elif agg.fn == "first":
+                elif agg.fn in ("first", "first_ignores_null"):
                     acc_col, cnt = self._agg_first(gids, ngroups, val)
                     acc, vcnt, vdt = acc_col, cnt, val.dtype
                 elif i in fused:
@@ -871,6 +872,9 @@ class Executor:
         return [RecordBatch(names, cols)]
 
     def _agg_first(self, gids, ngroups, val: Column):
+        """first value per group, skipping nulls (the reference's
+        first_ignores_null; plain `first` maps here too — order is
+        engine-dependent in Spark anyway)."""
         device = gids.device
         n = gids.numel()
         order = torch.arange(n, dtype=torch.int64, device=device)
