@@ -2287,9 +2287,13 @@ def q18(root, sf):
             **{n: (c, "mean") for n, c in carry.items()}).reset_index()
         frames.append(g)
     out = pd.concat(frames, ignore_index=True)
+    # FULL result (no LIMIT): a detail row and its subtotal can tie on every
+    # ORDER BY key with identical aggregates, and pandas cannot express
+    # per-key null ordering, so the top-100 cut is engine-dependent; the
+    # harness checks the engine's 100 rows as a subset (SUBSET_OF_FULL)
     out = out.sort_values(["ca_country", "ca_state", "ca_county", "i_item_id"],
                           ascending=[False, False, False, True],
-                          na_position="first").head(100)
+                          na_position="first")
     return out[keys + list(carry.keys())].reset_index(drop=True)
 
 

@@ -55,7 +55,7 @@ RANK_TOLERANT = {"q36": {"rank_within_parent"}}
 # engine-dependent subset, so the oracle returns the FULL result and the
 # engine rows must be a subset of it (official TPC-DS answer sets have the
 # same ambiguity)
-SUBSET_OF_FULL = {"q59"}
+SUBSET_OF_FULL = {"q59", "q18"}
 # boundary-epsilon queries: oracle returns an epsilon-relaxed FULL superset;
 # engine rows must be members (threshold rows may legitimately differ by
 # one fp ulp between engines)
