@@ -179,3 +179,39 @@ def test_expr_xxhash64_and_bloom_might_contain():
     bf.put_longs(torch.tensor([10], dtype=torch.int64))
     r = F.BloomFilterMightContain(bf, col("k")).eval(b)
     assert r.to_pylist() == [True, False, None]
+
+
+def test_runtime_bloom_filter_join_pruning():
+    """End-to-end runtime-filter pattern (the reference's bloom-filter
+    join pruning): build a bloom from the dimension keys, pre-filter the
+    fact side with BloomFilterMightContain, and verify the join result
+    is unchanged while the probe input shrinks."""
+    from auron_amd import AuronSession, functions as F
+    from auron_amd.column import RecordBatch
+    from auron_amd.exprs import col
+    from auron_amd.plan import nodes as P
+
+    dim_keys = list(range(0, 100, 7))  # 15 keys
+    fact_keys = [i % 400 for i in range(5000)]
+    dim = P.MemoryScan([RecordBatch.from_pydict(
+        {"dk": dim_keys, "dv": [k * 10 for k in dim_keys]},
+        {"dk": dtypes.int64, "dv": dtypes.int64})])
+    fact = P.MemoryScan([RecordBatch.from_pydict(
+        {"fk": fact_keys, "fv": list(range(5000))},
+        {"fk": dtypes.int64, "fv": dtypes.int64})])
+    s = AuronSession()
+
+    # driver-side build: collect dim keys, build the Spark-layout filter
+    dk = s.collect(dim).column("dk")
+    bf = sketch.BloomFilter.create(len(dim_keys), 0.01)
+    bf.put_column(dk)
+
+    plain = s.collect(P.HashJoin(fact, dim, [col("fk")], [col("dk")],
+                                 how="inner")).to_pydict()
+    pruned_scan = P.Filter(fact, F.BloomFilterMightContain(bf, col("fk")))
+    pruned_rows = s.collect(pruned_scan).num_rows
+    pruned = s.collect(P.HashJoin(pruned_scan, dim, [col("fk")], [col("dk")],
+                                  how="inner")).to_pydict()
+    key = lambda d: sorted(zip(d["fv"], d["dk"]))
+    assert key(pruned) == key(plain)
+    assert pruned_rows < 5000 * 0.2  # ~15/400 keys + fpp survive
