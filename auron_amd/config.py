@@ -40,8 +40,8 @@ REQUIRE_NATIVE = _opt("spark.auron.requireNativeKernels", True, bool,
 SPILL_COMPRESSION = _opt("spark.auron.spill.compression.codec", "lz4", str)
 SHUFFLE_COMPRESSION = _opt("spark.auron.shuffle.compression.codec", "none", str,
                            "xGMI is fast enough that intra-node shuffle ships raw")
-SMJ_FALLBACK_ENABLE = _opt("spark.auron.smjfallback.enable", False, bool)
-SMJ_FALLBACK_ROWS = _opt("spark.auron.smjfallback.rows.threshold", 10_000_000, int)
+SMJ_FALLBACK_ENABLE = _opt("spark.auron.smjfallback.enable", False, bool, env="AURON_SMJ_FALLBACK")
+SMJ_FALLBACK_ROWS = _opt("spark.auron.smjfallback.rows.threshold", 10_000_000, int, env="AURON_SMJ_FALLBACK_ROWS")
 PARTIAL_AGG_SKIPPING_RATIO = _opt("spark.auron.partialAggSkipping.ratio", 0.999, float)
 UDF_FALLBACK = _opt("spark.auron.udf.hostFallback.enable", True, bool)
 LOG_LEVEL = _opt("spark.auron.native.log.level", "WARN", str, env="AURON_LOG_LEVEL")

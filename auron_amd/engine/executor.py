@@ -982,6 +982,23 @@ class Executor:
         else:
             right = _concat(self.execute(node.right))
             left = _concat(left_batches())
+        # SMJ fallback (conf.rs:55-57): an oversized build side makes the
+        # chained hash table the wrong tool; lower to the order-based join
+        if not node.broadcast:
+            from ..config import (SMJ_FALLBACK_ENABLE, SMJ_FALLBACK_ROWS,
+                                  AuronConf)
+
+            conf = AuronConf()
+            build = right if node.build_side == "right" else left
+            if (conf.get(SMJ_FALLBACK_ENABLE)
+                    and build.num_rows > conf.get(SMJ_FALLBACK_ROWS)
+                    and node.how in ("inner", "left", "right", "full",
+                                     "semi", "anti", "existence")):
+                smj = P.SortMergeJoin(P.MemoryScan([left]), P.MemoryScan([right]),
+                                      node.left_keys, node.right_keys,
+                                      how=node.how,
+                                      existence_col=node.existence_col)
+                return self._exec_SortMergeJoin(smj)
         lkeys = [k.eval(left) for k in node.left_keys]
         rkeys = [k.eval(right) for k in node.right_keys]
         lkeys, rkeys = _normalize_join_keys(lkeys, rkeys)

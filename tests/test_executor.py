@@ -425,3 +425,19 @@ def test_window_first_last_nth_value():
     assert rows[1][2:] == (10.0, 20.0, 20.0)   # p1 o2
     assert rows[2][2:] == (10.0, 30.0, 20.0)   # p1 o3
     assert rows[3][2:] == (40.0, 40.0, None)   # p2 o1
+
+
+def test_smj_fallback_for_large_builds(monkeypatch):
+    """spark.auron.smjfallback: hash joins with oversized build sides
+    lower to the sort-merge join and produce identical results."""
+    monkeypatch.setenv("AURON_SMJ_FALLBACK", "1")
+    monkeypatch.setenv("AURON_SMJ_FALLBACK_ROWS", "2")  # tiny threshold
+    s = AuronSession()
+    for how in ("inner", "left", "full", "semi", "anti"):
+        out = s.collect(_join_plan(how)).to_pydict()
+        if how == "semi":
+            assert sorted(out["lv"]) == ["b", "c"]
+        elif how == "anti":
+            assert sorted(out["lv"]) == ["a", "d", "e"]
+        else:
+            assert_rows_equal(out, _sql_join_oracle(how))
