@@ -42,7 +42,7 @@ SHUFFLE_COMPRESSION = _opt("spark.auron.shuffle.compression.codec", "none", str,
                            "xGMI is fast enough that intra-node shuffle ships raw")
 SMJ_FALLBACK_ENABLE = _opt("spark.auron.smjfallback.enable", False, bool, env="AURON_SMJ_FALLBACK")
 SMJ_FALLBACK_ROWS = _opt("spark.auron.smjfallback.rows.threshold", 10_000_000, int, env="AURON_SMJ_FALLBACK_ROWS")
-PARTIAL_AGG_SKIPPING_RATIO = _opt("spark.auron.partialAggSkipping.ratio", 0.999, float)
+PARTIAL_AGG_SKIPPING_RATIO = _opt("spark.auron.partialAggSkipping.ratio", 0.999, float, env="AURON_PARTIAL_SKIP_RATIO")
 UDF_FALLBACK = _opt("spark.auron.udf.hostFallback.enable", True, bool)
 LOG_LEVEL = _opt("spark.auron.native.log.level", "WARN", str, env="AURON_LOG_LEVEL")
 IGNORE_CORRUPTED_FILES = _opt("spark.auron.ignoreCorruptedFiles", False, bool,

@@ -772,6 +772,8 @@ class Executor:
         else:
             key_cols = [a.expr.eval(b) for a in node.keys]
         if node.keys:
+            if node.mode == "partial" and self._partial_skip(key_cols, n):
+                return [self._partial_passthrough(node, b, key_cols)]
             gids, reps = ops.group_ids(key_cols)
             ngroups = int(reps.numel())
             out_keys = [c.gather(reps) for c in key_cols]
@@ -870,6 +872,60 @@ class Executor:
                         cols.append(self._finalize_agg(agg, vdt, a_data if not isinstance(acc, Column) else acc, vcnt))
                     names.append(agg.name)
         return [RecordBatch(names, cols)]
+
+    def _partial_skip(self, key_cols, n: int) -> bool:
+        """Partial-agg skipping (conf.rs:39-42): sample the reduction
+        ratio; when grouping barely reduces rows, skip the hash table and
+        emit singleton states — the post-exchange final agg regroups."""
+        if n < (1 << 14):
+            return False
+        from ..config import PARTIAL_AGG_SKIPPING_RATIO, AuronConf
+
+        ratio = AuronConf().get(PARTIAL_AGG_SKIPPING_RATIO)
+        if ratio >= 1.0:
+            return False
+        sn = min(n, 1 << 16)
+        idx = torch.arange(sn, dtype=torch.int64, device=key_cols[0].device)
+        sample = [c.gather(idx) for c in key_cols]
+        _, reps = ops.group_ids(sample)
+        return int(reps.numel()) > ratio * sn
+
+    def _partial_passthrough(self, node: P.HashAgg, b: RecordBatch,
+                             key_cols) -> RecordBatch:
+        """Each row becomes its own group: states carry the raw value and
+        a 0/1 count. Schema-identical to the grouped partial output."""
+        n = b.num_rows
+        device = b.device
+        names = [a.name for a in node.keys]
+        cols = list(key_cols)
+        for i, agg in enumerate(node.aggs):
+            s0, s1 = f"__agg{i}_0", f"__agg{i}_1"
+            val = agg.expr.eval(b) if agg.expr is not None else None
+            if agg.fn == "count_star":
+                ones = torch.ones(n, dtype=torch.int64, device=device)
+                cols.append(Column(dtypes.int64, ones))
+                names.append(s0)
+                cols.append(Column(dtypes.int64, ones))
+                names.append(s1)
+                continue
+            valid = val.validity if val.validity is not None else \
+                torch.ones(n, dtype=torch.bool, device=device)
+            cnt = valid.to(torch.int64)
+            if agg.fn == "count":
+                cols.append(Column(dtypes.int64, cnt))
+                names.append(s0)
+                cols.append(Column(dtypes.int64, cnt))
+                names.append(s1)
+                continue
+            state_dt = self._state_dtype(agg, val.dtype)
+            data = val.data
+            if data.dtype != state_dt.torch_dtype and not state_dt.uses_offsets:
+                data = data.to(state_dt.torch_dtype)
+            cols.append(Column(state_dt, data, val.validity, val.offsets))
+            names.append(s0)
+            cols.append(Column(dtypes.int64, cnt))
+            names.append(s1)
+        return RecordBatch(names, cols)
 
     def _agg_first(self, gids, ngroups, val: Column):
         """first value per group, skipping nulls (the reference's

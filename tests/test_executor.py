@@ -441,3 +441,50 @@ def test_smj_fallback_for_large_builds(monkeypatch):
             assert sorted(out["lv"]) == ["a", "d", "e"]
         else:
             assert_rows_equal(out, _sql_join_oracle(how))
+
+
+def test_partial_agg_skipping(monkeypatch):
+    """With a forced 0.0 ratio, high-cardinality partial aggs emit
+    singleton states; the final agg still produces exact results."""
+    monkeypatch.setenv("AURON_PARTIAL_SKIP_RATIO", "0.0")
+    import numpy as np
+
+    n = 20_000  # above the sampling floor
+    rng = np.random.default_rng(3)
+    data = {"k": rng.integers(0, 15_000, n).tolist(),
+            "v": rng.normal(0, 10, n).tolist()}
+    t = {"k": dtypes.int64, "v": dtypes.float64}
+    scan_n = P.MemoryScan([RecordBatch.from_pydict(data, t)])
+    tower = P.HashAgg(
+        P.Exchange(P.HashAgg(scan_n, [exprs.Aliased(col("k"), "k")],
+                             [AggFunc("sum", col("v"), name="s"),
+                              AggFunc("count", col("v"), name="c"),
+                              AggFunc("max", col("v"), name="m")],
+                             mode="partial"), "hash", [col("k")]),
+        [exprs.Aliased(col("k"), "k")],
+        [AggFunc("sum", col("v"), name="s"), AggFunc("count", col("v"), name="c"),
+         AggFunc("max", col("v"), name="m")], mode="final")
+    s = AuronSession()
+    # world_size==1 rewrite would collapse the tower; execute WITHOUT rewrite
+    # by running the partial stage separately first
+    part = s.execute(P.HashAgg(scan_n, [exprs.Aliased(col("k"), "k")],
+                               [AggFunc("sum", col("v"), name="s"),
+                                AggFunc("count", col("v"), name="c"),
+                                AggFunc("max", col("v"), name="m")],
+                               mode="partial"))
+    assert part[0].num_rows == n  # pass-through engaged
+    fin = s.collect(P.HashAgg(P.MemoryScan(part), [exprs.Aliased(col("k"), "k")],
+                              [AggFunc("sum", col("v"), name="s"),
+                               AggFunc("count", col("v"), name="c"),
+                               AggFunc("max", col("v"), name="m")], mode="final"))
+    got = {k: (s_, c_, m_) for k, s_, c_, m_ in
+           zip(*[fin.to_pydict()[x] for x in ("k", "s", "c", "m")])}
+    import collections
+    ref = collections.defaultdict(list)
+    for k, v in zip(data["k"], data["v"]):
+        ref[k].append(v)
+    assert len(got) == len(ref)
+    for k, vals in ref.items():
+        s_, c_, m_ = got[k]
+        assert c_ == len(vals) and abs(s_ - sum(vals)) < 1e-6
+        assert abs(m_ - max(vals)) < 1e-12
