@@ -947,7 +947,11 @@ def q40(root, sf):
     j = _merge(j, dd, "cs_sold_date_sk", "d_date_sk")
     net = j.cs_sales_price - j.cr_refunded_cash.fillna(0.0)
     j = j.assign(b=np.where(j.d_i < pivot, net, 0.0), a=np.where(j.d_i >= pivot, net, 0.0))
-    j.loc[j.cs_sales_price.isna(), ["b", "a"]] = np.nan
+    # CASE WHEN cond THEN net ELSE 0: the result is NULL only when the
+    # TAKEN branch is the null `net`; else-branch rows contribute 0 even
+    # with a null price
+    j.loc[j.cs_sales_price.isna() & (j.d_i < pivot), "b"] = np.nan
+    j.loc[j.cs_sales_price.isna() & (j.d_i >= pivot), "a"] = np.nan
     g = j.groupby(["w_state", "i_item_id"], dropna=False).agg(
         sales_before=("b", lambda x: x.sum(min_count=1)),
         sales_after=("a", lambda x: x.sum(min_count=1))).reset_index()
