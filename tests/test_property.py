@@ -227,3 +227,59 @@ def test_exchange_pack_unpack_roundtrip(a, b, c):
             want = [None if v is None else round(v, 2) for v in want]
             got = [None if v is None else round(v, 2) for v in got]
         assert got == want, k
+
+
+@settings(max_examples=50, deadline=None)
+@given(st.lists(st.one_of(st.none(),
+                          st.decimals(min_value=-99999, max_value=99999,
+                                      places=2)), max_size=60))
+def test_decimal_sum_exact(vals):
+    """decimal(7,2) sums are EXACT (scaled int64), matching python
+    Decimal arithmetic bit-for-bit."""
+    from decimal import Decimal
+
+    pyvals = [None if v is None else float(v) for v in vals]
+    b = _batch({"k": [1] * len(vals), "v": pyvals},
+               {"k": dtypes.int64, "v": dtypes.decimal64(7, 2)})
+    plan = P.HashAgg(P.MemoryScan([b]), [Aliased(col("k"), "k")],
+                     [AggFunc("sum", col("v"), name="s"),
+                      AggFunc("min", col("v"), name="mn"),
+                      AggFunc("max", col("v"), name="mx")], mode="complete")
+    got = AuronSession().collect(plan).to_pydict()
+    nn = [Decimal(str(v)) for v in vals if v is not None]
+    if not vals:
+        return
+    if not nn:
+        assert got["s"] == [None] and got["mn"] == [None]
+        return
+    assert Decimal(str(got["s"][0])) == sum(nn)
+    assert Decimal(str(got["mn"][0])) == min(nn)
+    assert Decimal(str(got["mx"][0])) == max(nn)
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.lists(st.decimals(min_value=0, max_value=999, places=2),
+                min_size=1, max_size=40))
+def test_decimal_roundtrip_through_parquet(vals):
+    """decimal column -> parquet (store_decimal_as_integer) -> native np
+    decode keeps exact cents."""
+    import tempfile
+
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    from auron_amd import parquet_native
+
+    arr = pa.array([v for v in vals], type=pa.decimal128(7, 2))
+    with tempfile.TemporaryDirectory() as td:
+        p = f"{td}/d.parquet"
+        pq.write_table(pa.table({"d": arr}), p, compression="NONE",
+                       use_dictionary=False, data_page_version="1.0",
+                       store_decimal_as_integer=True)
+        cols = parquet_native.read_columns_native(p, ["d"], "cpu", _np_only=True)
+        assert cols is not None
+        got = cols["d"].to_pylist()
+    from decimal import Decimal
+
+    for g, w in zip(got, vals):
+        assert Decimal(str(g)) == w
