@@ -1214,6 +1214,23 @@ class Executor:
                 ok = ok & (seg[idx_c] == seg) if n else ok
                 gi = torch.where(ok, idx_c, torch.full_like(idx_c, -1))
                 out = val.gather(gi)
+                if wf.default is not None and out.validity is not None:
+                    from ..exprs import Literal, eval_scope
+
+                    dcol = Literal(wf.default, val.dtype).eval(sb)
+                    miss = ~out.validity
+                    # only frame-escapes get the default; genuine nulls stay
+                    src_valid = val.validity if val.validity is not None else None
+                    if src_valid is not None:
+                        esc = miss & ~torch.where(gi >= 0, ~src_valid[gi.clamp(min=0)],
+                                                  torch.zeros_like(miss))
+                    else:
+                        esc = miss
+                    data = torch.where(esc, dcol.data, out.data)
+                    validity = out.validity | esc
+                    if bool(validity.all()):
+                        validity = None
+                    out = Column(out.dtype, data, validity)
             elif wf.fn in ("percent_rank", "cume_dist", "ntile"):
                 seg_last = (torch.cat([seg_start[1:] - 1,
                                        torch.tensor([n - 1], dtype=torch.int64,

@@ -626,9 +626,10 @@ class DatePart(Expr):
 class WindowFunc(Expr):
     """Window function marker expr (window/mod.rs WindowFunctionProcessor
     analogue). Evaluated by the Window operator, not by Expr.eval."""
-    fn: str  # row_number | rank | dense_rank | sum | avg | count | min | max | lead | lag
+    fn: str  # row_number | rank | dense_rank | sum | avg | count | min | max | lead | lag | ...
     arg: Optional[Expr] = None
-    offset: int = 1  # lead/lag
+    offset: int = 1  # lead/lag/ntile/nth_value parameter
+    default: object = None  # lead/lag third argument
 
     def eval(self, batch):  # pragma: no cover
         raise RuntimeError("WindowFunc must appear under a Window operator")

@@ -488,3 +488,20 @@ def test_partial_agg_skipping(monkeypatch):
         s_, c_, m_ = got[k]
         assert c_ == len(vals) and abs(s_ - sum(vals)) < 1e-6
         assert abs(m_ - max(vals)) < 1e-12
+
+
+def test_lead_lag_default_value():
+    from auron_amd.exprs import Aliased, WindowFunc
+
+    data = {"p": [1, 1, 1], "o": [1, 2, 3], "v": [10.0, None, 30.0]}
+    t = {"p": dtypes.int64, "o": dtypes.int64, "v": dtypes.float64}
+    sc = P.MemoryScan([RecordBatch.from_pydict(data, t)])
+    fns = [Aliased(WindowFunc("lag", col("v"), 1, default=-1.0), "lg"),
+           Aliased(WindowFunc("lead", col("v"), 1, default=-1.0), "ld")]
+    out = AuronSession().collect(
+        P.Window(sc, [col("p")], [(col("o"), True)], fns)).to_pydict()
+    rows = sorted(zip(out["o"], out["lg"], out["ld"]))
+    # frame escapes get the default; a genuine NULL neighbor stays NULL
+    assert rows[0][1:] == (-1.0, None)   # o=1: no lag row; lead = null v
+    assert rows[1][1:] == (10.0, 30.0)   # o=2
+    assert rows[2][1:] == (None, -1.0)   # o=3: lag = null v; no lead row
