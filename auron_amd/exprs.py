@@ -256,24 +256,37 @@ def _cast_col(c: Column, dt: DataType) -> Column:
 
 
 def _cast_string(c: Column, dt: DataType) -> Column:
+    import datetime as _dt_mod
+
+    _EPOCH = _dt_mod.date(1970, 1, 1)
     if dt.is_string and c.dtype.is_string:
         return c
     if c.dtype.is_string:
-        # string -> numeric: host round-trip (cold path in TPC-DS)
+        # string -> numeric/date: host round-trip (cold path in TPC-DS);
+        # unparseable values yield null (try_cast/non-ANSI cast semantics)
         vals = c.to_pylist()
         out = []
         for v in vals:
             if v is None:
                 out.append(None)
+            elif dt.code == dtypes.DATE32:
+                try:
+                    out.append((_dt_mod.date.fromisoformat(v.strip()) - _EPOCH).days)
+                except ValueError:
+                    out.append(None)
             else:
                 try:
                     out.append(float(v) if dt.is_float or dt.code == dtypes.DECIMAL64 else int(float(v)))
                 except ValueError:
                     out.append(None)
         return Column.from_pylist(out, dt, c.device)
-    # numeric -> string: host round-trip
+    # numeric/date -> string: host round-trip
     vals = c.to_pylist()
-    out = [None if v is None else (str(int(v)) if c.dtype.is_integer else str(v)) for v in vals]
+    if c.dtype.code == dtypes.DATE32:
+        out = [None if v is None else (_EPOCH + _dt_mod.timedelta(days=int(v))).isoformat()
+               for v in vals]
+    else:
+        out = [None if v is None else (str(int(v)) if c.dtype.is_integer else str(v)) for v in vals]
     return Column.from_pylist(out, dt, c.device)
 
 

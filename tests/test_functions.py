@@ -230,3 +230,13 @@ def test_months_between_next_day_trunc():
     assert ev(F.TruncDate(col("d"), "year"), b3) == [_date("2001-01-01")]
     assert ev(F.TruncDate(col("d"), "month"), b3) == [_date("2001-03-01")]
     assert ev(F.TruncDate(col("d"), "week"), b3) == [_date("2001-03-12")]
+
+
+def test_string_date_casts():
+    from auron_amd.exprs import Cast
+
+    b = _b(s=(["2001-03-15", " 1999-12-31", "bogus", None], dtypes.string))
+    got = ev(Cast(col("s"), dtypes.date32), b)
+    assert got == [_date("2001-03-15"), _date("1999-12-31"), None, None]
+    b2 = _b(d=_dcol("2001-03-15", None))
+    assert ev(Cast(col("d"), dtypes.string), b2) == ["2001-03-15", None]
