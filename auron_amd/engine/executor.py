@@ -693,6 +693,34 @@ class Executor:
             n = b.num_rows
             part = torch.arange(n, dtype=torch.int64, device=device) % W
             dest = [b.filter(part == d) for d in range(W)]
+        elif node.kind == "range" and node.keys[0].eval(b).dtype.is_string:
+            # string range partition: gather sampled strings, agree on
+            # W-1 bound strings, rank rows against the bounds with the
+            # byte-order comparator (dictionary ranks)
+            key = node.keys[0].eval(b)
+            n = b.num_rows
+            take = min(n, 2048)
+            samp_idx = torch.randperm(n, device=device)[:take] if n else \
+                torch.zeros(0, dtype=torch.int64, device=device)
+            sample = key.gather(samp_idx).to("cpu").to_pylist()
+            gathered = [None] * W
+            if W > 1 and dist.is_initialized():
+                dist.all_gather_object(gathered, sample, group=self.ctx.group)
+            else:
+                gathered = [sample]
+            allsamp = sorted(v for g in gathered if g for v in g if v is not None)
+            bounds = [allsamp[len(allsamp) * (d + 1) // W]
+                      for d in range(W - 1)] if allsamp else [""] * (W - 1)
+            from .. import strings as S
+
+            part = torch.zeros(n, dtype=torch.int64, device=device)
+            for bd in bounds:
+                gt = ~S.compare(key, Column.from_pylist([bd] * n, dtypes.string,
+                                                        str(device)), "<=")
+                part += gt.to(torch.int64)
+            if key.validity is not None:
+                part = torch.where(key.validity, part, torch.zeros_like(part))
+            dest = [b.filter(part == d) for d in range(W)]
         elif node.kind == "range":
             # range repartition (Partitioning::RangePartitioning): sample
             # the first key on every rank, agree on W-1 global bounds,

@@ -168,3 +168,23 @@ def test_range_exchange_global_order():
         assert max(res[0]["keys"]) <= min(res[1]["keys"])
     # rough balance (quantile bounds from 400 samples of uniform keys)
     assert 100 < res[0]["n"] < 300
+
+
+def _range_exchange_strings(rank, world):
+    import numpy as np
+
+    rng = np.random.default_rng(100 + rank)
+    words = [f"w{int(v):04d}" for v in rng.integers(0, 1000, 150)]
+    s = AuronSession()
+    plan = P.Exchange(
+        P.MemoryScan([RecordBatch.from_pydict(
+            {"k": words}, {"k": dtypes.string})]), "range", [col("k")])
+    got = s.collect(plan).to_pydict()
+    return {"keys": got["k"], "n": len(got["k"])}
+
+
+def test_range_exchange_strings_global_order():
+    res = _spawn("_range_exchange_strings")
+    assert res[0]["n"] + res[1]["n"] == 300
+    if res[0]["keys"] and res[1]["keys"]:
+        assert max(res[0]["keys"]) <= min(res[1]["keys"])
