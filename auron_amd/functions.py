@@ -801,6 +801,69 @@ class BloomFilterMightContain(Expr):
 
 
 @dataclass(eq=False)
+class GetArrayItem(Expr):
+    """ext-exprs GetIndexedField analogue: 0-based element of a LIST
+    column; out-of-bounds -> null."""
+    child: Expr
+    index: int
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        assert c.dtype.is_list
+        off = c.offsets.to(torch.int64)
+        lens = off[1:] - off[:-1]
+        ok = torch.tensor(self.index, device=c.device) < lens
+        ok = ok & (self.index >= 0)
+        if c.validity is not None:
+            ok = ok & c.validity
+        pos = (off[:-1] + self.index).clamp(0, max(int(c.data.numel()) - 1, 0))
+        data = c.data[pos] if c.data.numel() else \
+            torch.zeros(len(c), dtype=c.dtype.torch_dtype, device=c.device)
+        return Column(c.dtype.child, data, ok if not bool(ok.all()) else None)
+
+
+@dataclass(eq=False)
+class ElementAt(Expr):
+    """Spark element_at(list, i): 1-based; negative i counts from the
+    end; out-of-bounds -> null."""
+    child: Expr
+    index: int
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        assert c.dtype.is_list and self.index != 0
+        off = c.offsets.to(torch.int64)
+        lens = off[1:] - off[:-1]
+        if self.index > 0:
+            k = torch.full_like(lens, self.index - 1)
+        else:
+            k = lens + self.index
+        ok = (k >= 0) & (k < lens)
+        if c.validity is not None:
+            ok = ok & c.validity
+        pos = (off[:-1] + k).clamp(0, max(int(c.data.numel()) - 1, 0))
+        data = c.data[pos] if c.data.numel() else \
+            torch.zeros(len(c), dtype=c.dtype.torch_dtype, device=c.device)
+        return Column(c.dtype.child, data, ok if not bool(ok.all()) else None)
+
+
+@dataclass(eq=False)
+class ArraySize(Expr):
+    """size(list): element count; null list -> -1 (Spark legacy default)."""
+    child: Expr
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        assert c.dtype.is_list
+        off = c.offsets.to(torch.int64)
+        lens = (off[1:] - off[:-1]).to(torch.int32)
+        if c.validity is not None:
+            lens = torch.where(c.validity, lens,
+                               torch.full_like(lens, -1))
+        return Column(dtypes.int32, lens)
+
+
+@dataclass(eq=False)
 class MakeArray(Expr):
     """spark make_array / array(...): fixed-width list per row from k
     element expressions (device-vectorized; element nulls become 0 — an

@@ -240,3 +240,15 @@ def test_string_date_casts():
     assert got == [_date("2001-03-15"), _date("1999-12-31"), None, None]
     b2 = _b(d=_dcol("2001-03-15", None))
     assert ev(Cast(col("d"), dtypes.string), b2) == ["2001-03-15", None]
+
+
+def test_array_item_element_at_size():
+    lt = dtypes.list_of(dtypes.int64)
+    c = Column.from_pylist([[1, 2, 3], [], None, [7]], lt)
+    b = RecordBatch(["xs"], [c])
+    assert ev(F.GetArrayItem(col("xs"), 0), b) == [1, None, None, 7]
+    assert ev(F.GetArrayItem(col("xs"), 2), b) == [3, None, None, None]
+    assert ev(F.ElementAt(col("xs"), 1), b) == [1, None, None, 7]
+    assert ev(F.ElementAt(col("xs"), -1), b) == [3, None, None, 7]
+    assert ev(F.ElementAt(col("xs"), 5), b) == [None, None, None, None]
+    assert ev(F.ArraySize(col("xs")), b) == [3, 0, -1, 1]
