@@ -693,11 +693,11 @@ class Executor:
             n = b.num_rows
             part = torch.arange(n, dtype=torch.int64, device=device) % W
             dest = [b.filter(part == d) for d in range(W)]
-        elif node.kind == "range" and node.keys[0].eval(b).dtype.is_string:
+        elif node.kind == "range" and (_rk := node.keys[0].eval(b)).dtype.is_string:
             # string range partition: gather sampled strings, agree on
             # W-1 bound strings, rank rows against the bounds with the
             # byte-order comparator (dictionary ranks)
-            key = node.keys[0].eval(b)
+            key = _rk
             n = b.num_rows
             take = min(n, 2048)
             samp_idx = torch.randperm(n, device=device)[:take] if n else \
@@ -726,7 +726,7 @@ class Executor:
             # the first key on every rank, agree on W-1 global bounds,
             # route rows by searchsorted. Rank d holds keys in
             # (bounds[d-1], bounds[d]] — a global sort order across ranks.
-            key = _cast_for_range(node.keys[0].eval(b))
+            key = _cast_for_range(_rk)
             n = b.num_rows
             take = min(n, 4096)
             sample = key.data[torch.randperm(n, device=device)[:take]] if n \
