@@ -160,12 +160,7 @@ def all_to_all(batches_by_dest: List[Optional[RecordBatch]], device,
             input_split_sizes=[b.numel() for b in bufs],
             group=group,
         )
-        pos = 0
-        for s in range(world):
-            nb = recv_sizes[s]
-            if recv_metas[s] is not None:
-                out_batches.append(unpack_batch(recv_metas[s], recv_flat[pos:pos + nb]))
-            pos += nb
+        out_batches = _split_recv_flat(recv_flat, recv_sizes, recv_metas)
     else:
         # pairwise deterministic schedule (gloo CPU CI path)
         recv_bufs: List[Optional[torch.Tensor]] = [None] * world
@@ -189,6 +184,22 @@ def all_to_all(batches_by_dest: List[Optional[RecordBatch]], device,
             if recv_metas[s] is not None:
                 out_batches.append(unpack_batch(recv_metas[s], recv_bufs[s]))
     return out_batches
+
+
+def _split_recv_flat(recv_flat: torch.Tensor, recv_sizes, recv_metas):
+    """Slice the all_to_all_single output back into per-source batches.
+
+    Factored out of the nccl branch so the slicing math is unit-tested on
+    CPU (gloo cannot run all_to_all_single; this logic otherwise first
+    executes on the multi-GPU box)."""
+    out = []
+    pos = 0
+    for s in range(len(recv_sizes)):
+        nb = recv_sizes[s]
+        if recv_metas[s] is not None:
+            out.append(unpack_batch(recv_metas[s], recv_flat[pos:pos + nb]))
+        pos += nb
+    return out
 
 
 def all_gather_batch(batch: Optional[RecordBatch], device, group=None) -> List[RecordBatch]:

@@ -119,3 +119,30 @@ def test_shuffle_zlib_codec(tmp_path, monkeypatch):
     ShuffleWriter(str(tmp_path), "c2", 0).write([b0, b0])
     plain = os.path.getsize(os.path.join(str(tmp_path), "stage-c2", "map-0.data"))
     assert raw < plain  # codec actually engaged
+
+
+def test_nccl_branch_slicing_math():
+    """Simulate all_to_all_single's output contract for W=3 and verify the
+    nccl-branch reassembly (`_split_recv_flat`) recovers every batch —
+    including empty and None destinations."""
+    import torch
+
+    from auron_amd.exchange import _split_recv_flat, pack_batch
+
+    b1 = _batch(([1, None, 3], ["a", "bb", None]))
+    b2 = _batch(([9], ["z"]))
+    metas, bufs = [], []
+    for src in (b1, None, b2):
+        if src is None:
+            metas.append(None)
+            bufs.append(torch.zeros(0, dtype=torch.uint8))
+        else:
+            m, buf = pack_batch(src, "cpu")
+            metas.append(m)
+            bufs.append(buf)
+    recv_sizes = [b.numel() for b in bufs]
+    recv_flat = torch.cat(bufs) if sum(recv_sizes) else torch.zeros(0, dtype=torch.uint8)
+    out = _split_recv_flat(recv_flat, recv_sizes, metas)
+    assert len(out) == 2
+    assert out[0].to_pydict() == b1.to_pydict()
+    assert out[1].to_pydict() == b2.to_pydict()
