@@ -18,3 +18,12 @@ def pytest_collection_modifyitems(config, items):
 @pytest.fixture
 def device():
     return torch.device("cuda:0") if torch.cuda.is_available() else torch.device("cpu")
+
+
+try:  # deeper fuzzing on demand: pytest --hypothesis-profile=thorough
+    from hypothesis import settings as _hyp_settings
+
+    _hyp_settings.register_profile("thorough", max_examples=300,
+                                   deadline=None)
+except ImportError:  # pragma: no cover
+    pass
