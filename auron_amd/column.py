@@ -286,9 +286,16 @@ class Column:
             vals = c.to_pylist()
             return pa.array(vals, type=pa.string())
         if self.dtype.code == dtypes.DECIMAL64:
-            s = 10 ** self.dtype.scale
-            vals = [None if (mask is not None and mask[i]) else c.data[i].item() / s for i in range(len(c))]
-            return pa.array(vals, type=pa.float64())
+            # exact scaled-int -> arrow decimal128 (schema + exactness
+            # survive a sink/scan round-trip; advisor finding r1)
+            from decimal import Decimal
+
+            sc = self.dtype.scale
+            vals = [None if (mask is not None and mask[i])
+                    else Decimal(int(c.data[i].item())).scaleb(-sc)
+                    for i in range(len(c))]
+            p = max(self.dtype.precision, sc + 1, 1)
+            return pa.array(vals, type=pa.decimal128(p, sc))
         np_arr = c.data.numpy()
         return pa.array(np_arr, from_pandas=False, mask=mask)
 

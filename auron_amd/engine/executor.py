@@ -739,10 +739,20 @@ class Executor:
                 gathered = [sample.cpu()]
             allsamp = torch.cat([g for g in gathered if g is not None and g.numel()]) \
                 if any(g is not None and g.numel() for g in gathered) else torch.zeros(1, dtype=key.data.dtype)
-            qs = torch.quantile(allsamp.to(torch.float64),
-                                torch.linspace(0, 1, W + 1, dtype=torch.float64)[1:-1])
-            bounds = qs.to(device)
-            part = torch.searchsorted(bounds, key.data.to(torch.float64))
+            if allsamp.dtype in (torch.int8, torch.int16, torch.int32, torch.int64):
+                # integer keys: pick bounds in the integer domain (a float64
+                # quantile would truncate keys beyond 2^53 and silently
+                # mis-partition the global sort order)
+                ss = allsamp.to(torch.int64).sort().values
+                qi = (torch.linspace(0, 1, W + 1, dtype=torch.float64)[1:-1]
+                      * (ss.numel() - 1)).round().to(torch.int64)
+                bounds = ss[qi].to(device)
+                part = torch.searchsorted(bounds, key.data.to(torch.int64))
+            else:
+                qs = torch.quantile(allsamp.to(torch.float64),
+                                    torch.linspace(0, 1, W + 1, dtype=torch.float64)[1:-1])
+                bounds = qs.to(device)
+                part = torch.searchsorted(bounds, key.data.to(torch.float64))
             if key.validity is not None:  # nulls first (rank 0)
                 part = torch.where(key.validity, part, torch.zeros_like(part))
             dest = [b.filter(part == d) for d in range(W)]
@@ -826,7 +836,10 @@ class Executor:
             fuse_idx = []
             for i, agg in enumerate(node.aggs):
                 if agg.fn in ("sum", "avg", "min", "max") and agg.expr is not None:
-                    fuse_items.append((agg.expr.eval(b), agg.fn))
+                    v = agg.expr.eval(b)
+                    if agg.fn in ("sum", "avg") and self._decimal_sum_unsafe(v, n):
+                        continue  # exact split path below
+                    fuse_items.append((v, agg.fn))
                     fuse_idx.append(i)
             if len(fuse_items) > 1:
                 for i, res in zip(fuse_idx,
@@ -846,6 +859,8 @@ class Executor:
                         "first": "first", "first_ignores_null": "first"}[agg.fn]
                 if comb == "first":
                     acc, cnt = self._agg_first(gids, ngroups, sv)
+                elif comb == "sum" and self._decimal_sum_unsafe(sv, n):
+                    acc, cnt = self._sum_split_exact(gids, ngroups, sv)
                 else:
                     acc, cnt = ops.agg_scatter(gids, ngroups, sv, comb)
                 cols.append(self._finalize_agg(agg, sv.dtype, acc, merged_cnt))
@@ -880,6 +895,9 @@ class Executor:
                 elif i in fused:
                     fval, (facc, fcnt) = fused[i]
                     acc, vcnt, vdt = facc, fcnt, fval.dtype
+                elif agg.fn in ("sum", "avg") and self._decimal_sum_unsafe(val, n):
+                    acc, vcnt = self._sum_split_exact(gids, ngroups, val)
+                    vdt = val.dtype
                 else:
                     fn = {"sum": "sum", "avg": "sum", "min": "min", "max": "max", "count": "count"}[agg.fn]
                     acc, vcnt, vdt = *ops.agg_scatter(gids, ngroups, val, fn), val.dtype
@@ -900,6 +918,47 @@ class Executor:
                         cols.append(self._finalize_agg(agg, vdt, a_data if not isinstance(acc, Column) else acc, vcnt))
                     names.append(agg.name)
         return [RecordBatch(names, cols)]
+
+    @staticmethod
+    def _decimal_sum_unsafe(val: Column, nrows: int) -> bool:
+        """True when a plain int64 scatter-sum of this decimal column could
+        wrap (advisor finding r1: n*maxabs must fit int64)."""
+        if val.dtype.code != dtypes.DECIMAL64 or nrows == 0:
+            return False
+        m = int(val.data.abs().max().item())
+        return m > 0 and nrows * m >= (1 << 62)
+
+    def _sum_split_exact(self, gids, ngroups, val: Column):
+        """Exact decimal sum via hi/lo 32-bit split accumulators: each
+        int64 addend is split into (v>>32, v&0xffffffff); both partial
+        sums stay far from int64 range for <2^31 rows, and the recombine
+        detects true overflow of the mathematical result instead of
+        silently wrapping (reference promotes sum(decimal) to 128-bit)."""
+        device = gids.device
+        n = gids.numel()
+        assert n < (1 << 31), "batch too large for split accumulation"
+        valid = val.validity if val.validity is not None else \
+            torch.ones(n, dtype=torch.bool, device=device)
+        x = torch.where(valid, val.data, torch.zeros_like(val.data))
+        lo = x & 0xFFFFFFFF
+        hi = x >> 32
+        sum_lo = torch.zeros(ngroups, dtype=torch.int64, device=device)
+        sum_hi = torch.zeros(ngroups, dtype=torch.int64, device=device)
+        cnt = torch.zeros(ngroups, dtype=torch.int64, device=device)
+        if n:
+            sum_lo.scatter_add_(0, gids, lo)
+            sum_hi.scatter_add_(0, gids, hi)
+            cnt.scatter_add_(0, gids, valid.to(torch.int64))
+        carry = sum_lo >> 32
+        rem = sum_lo & 0xFFFFFFFF
+        h = sum_hi + carry
+        if bool(((h < -(1 << 31)) | (h > (1 << 31) - 1)).any()):
+            from ..session import AuronTaskError
+
+            raise AuronTaskError(
+                "decimal sum overflow: group total exceeds 18 significant "
+                "digits (decimal64 backing); rescale or cast to double")
+        return h * (1 << 32) + rem, cnt
 
     def _partial_skip(self, key_cols, n: int) -> bool:
         """Partial-agg skipping (conf.rs:39-42): sample the reduction
@@ -1373,29 +1432,38 @@ class Executor:
             if val.dtype.code == dtypes.DECIMAL64:
                 dt = val.dtype
             return Column(dt, run, validity)
-        # running min/max: segment-reset via a cummax over segment-shifted
-        # values (each segment's band is disjoint, so cummax cannot carry
-        # across segment boundaries; invalid rows get a never-wins filler)
-        neg = fn == "min"
-        x = v.to(torch.float64)
-        if neg:
-            x = -x
-        big = (x[valid].abs().max().item() + 1.0) if bool(valid.any()) else 1.0
-        x = torch.where(valid, x, torch.full_like(x, -2.0 * big))
-        shift = seg.to(torch.float64) * (8.0 * big)
-        run = torch.cummax(x + shift, 0).values - shift
-        if neg:
-            run = -run
+        # running min/max: exact segmented scan in the VALUE domain (int64
+        # for int/decimal, float64 for floats) via log2(maxseglen) doubling
+        # steps — never routed through shifted floats, so int64/decimal
+        # values beyond 2^53 stay exact (advisor finding r1).
+        is_int = v.dtype == torch.int64
+        work = v if is_int else v.to(torch.float64)
+        if fn == "min":
+            fill = torch.iinfo(torch.int64).max if is_int else float("inf")
+            combine = torch.minimum
+        else:
+            fill = torch.iinfo(torch.int64).min if is_int else float("-inf")
+            combine = torch.maximum
+        run = torch.where(valid, work, torch.full_like(work, fill))
+        start = seg_start[seg]
+        pos = torch.arange(n, dtype=torch.int64, device=device)
+        maxlen = int((seg_start[1:] - seg_start[:-1]).max().item()) if seg_start.numel() > 1 else n
+        maxlen = max(maxlen, n - int(seg_start[-1].item()) if seg_start.numel() else n)
+        off = 1
+        while off < maxlen:
+            cand = torch.empty_like(run)
+            cand[off:] = run[:-off]
+            ok = (pos - off) >= start
+            run = torch.where(ok, combine(run, cand), run)
+            off <<= 1
         validity = run_count > 0
         if bool(validity.all()):
             validity = None
-        out = run
         if val.dtype.code == dtypes.DECIMAL64:
-            # decimal running min/max stays in scaled-int64 space
-            return Column(val.dtype, out.to(torch.int64), validity)
+            return Column(val.dtype, run, validity)
         if val.dtype.is_integer:
-            return Column(dtypes.int64, out.to(torch.int64), validity)
-        return Column(dtypes.float64, out, validity)
+            return Column(dtypes.int64, run, validity)
+        return Column(dtypes.float64, run, validity)
 
     def _col_eq_adjacent(self, c: Column) -> torch.Tensor:
         """eq mask between row i and i-1, for rows 1..n-1 (null==null)."""

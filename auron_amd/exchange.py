@@ -119,10 +119,6 @@ def unpack_batch(meta: dict, buf: torch.Tensor) -> RecordBatch:
     return RecordBatch(names, cols)
 
 
-def _empty_batch_like(names_types) -> RecordBatch:
-    raise NotImplementedError
-
-
 def all_to_all(batches_by_dest: List[Optional[RecordBatch]], device,
                group=None) -> List[RecordBatch]:
     """Send batches_by_dest[d] to rank d; return batches received (one per

@@ -50,6 +50,10 @@ class BatchHolder:
     def resident(self) -> bool:
         return self._batches is not None
 
+    @property
+    def host_resident(self) -> bool:
+        return self._spilled_host is not None
+
     def spill(self) -> int:
         """Move payload one tier down. Returns bytes released."""
         if self._batches is None:
@@ -62,6 +66,16 @@ class BatchHolder:
             # CPU mode: straight to disk so CI exercises the file tier
             self._spill_to_disk(self._batches)
         self._batches = None
+        self.mgr._host_pressure()
+        return self.bytes
+
+    def spill_host_to_disk(self) -> int:
+        """Tier 2: push a host-resident payload to a disk file (keeps host
+        spill bytes bounded under mgr.host_budget — advisor finding r1)."""
+        if self._spilled_host is None:
+            return 0
+        self._spill_to_disk(self._spilled_host)
+        self._spilled_host = None
         return self.bytes
 
     def _spill_to_disk(self, batches: List[RecordBatch]):
@@ -128,7 +142,7 @@ class MemManager:
     pressure falls back to LRU victims."""
 
     def __init__(self, budget_bytes: Optional[int] = None, fraction: float = 0.8,
-                 spill_dir: Optional[str] = None):
+                 spill_dir: Optional[str] = None, host_budget_bytes: Optional[int] = None):
         if budget_bytes is None:
             if torch.cuda.is_available():
                 free, total = torch.cuda.mem_get_info()
@@ -136,10 +150,28 @@ class MemManager:
             else:
                 budget_bytes = 16 << 30
         self.budget = budget_bytes
+        if host_budget_bytes is None:
+            host_budget_bytes = int(os.environ.get("AURON_HOST_SPILL_BUDGET",
+                                                   str(64 << 30)))
+        self.host_budget = host_budget_bytes
         self.spill_dir = spill_dir or tempfile.gettempdir()
         self._holders: List[BatchHolder] = []
         self._lock = threading.Lock()
         self.metrics: Dict[str, int] = {}
+
+    def _host_pressure(self):
+        """Enforce the host-DRAM spill budget: LRU host-resident holders go
+        down to the disk tier (reference: memmgr/spill.rs two-tier Spill)."""
+        hosts = [h for h in self._holders if h.host_resident]
+        used = sum(h.bytes for h in hosts)
+        if used <= self.host_budget:
+            return
+        for h in sorted(hosts, key=lambda h: h.last_touch):
+            used -= h.spill_host_to_disk()
+            self.metrics["spill_host2disk_count"] = \
+                self.metrics.get("spill_host2disk_count", 0) + 1
+            if used <= self.host_budget:
+                return
 
     def resident_bytes(self) -> int:
         return sum(h.bytes for h in self._holders if h.resident)
