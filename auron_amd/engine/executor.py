@@ -624,7 +624,7 @@ class Executor:
             row = side = seg = seg_start = None
             nl_seg = nr_seg = torch.zeros(0, dtype=torch.int64, device=device)
 
-        if how in ("semi", "anti", "existence"):
+        if how in ("semi", "anti", "existence") and node.residual is None:
             lmatched = torch.zeros(nl_rows, dtype=torch.bool, device=device)
             if n:
                 lpos = torch.nonzero(is_l, as_tuple=False).flatten()
@@ -650,6 +650,10 @@ class Executor:
             li = row[lsortpos]
         else:
             li = ri = torch.zeros(0, dtype=torch.int64, device=device)
+
+        if node.residual is not None:
+            return self._finish_join_pairs(left, right, li, ri, how,
+                                           node.residual, node.existence_col)
 
         preserved = {"inner": set(), "left": {"left"}, "right": {"right"},
                      "full": {"left", "right"}}[how]
@@ -1140,13 +1144,22 @@ class Executor:
                 smj = P.SortMergeJoin(P.MemoryScan([left]), P.MemoryScan([right]),
                                       node.left_keys, node.right_keys,
                                       how=node.how,
-                                      existence_col=node.existence_col)
+                                      existence_col=node.existence_col,
+                                      residual=node.residual)
                 return self._exec_SortMergeJoin(smj)
         lkeys = [k.eval(left) for k in node.left_keys]
         rkeys = [k.eval(right) for k in node.right_keys]
         lkeys, rkeys = _normalize_join_keys(lkeys, rkeys)
         how = node.how
         device = self.ctx.device
+
+        if node.residual is not None:
+            # equi-pairs first, then the non-equi condition per pair
+            bi, pi, _ = ops.hash_join(rkeys, lkeys,
+                                      emit_unmatched_probe=False,
+                                      need_build_matched=False)
+            return self._finish_join_pairs(left, right, pi, bi, how,
+                                           node.residual, node.existence_col)
 
         if how in ("semi", "anti", "existence"):
             counts = ops.join_counts(rkeys, lkeys)  # build=right, probe=left
@@ -1173,6 +1186,58 @@ class Executor:
                 bi = torch.cat([bi, un])
                 pi = torch.cat([pi, torch.full((un.numel(),), -1, dtype=torch.int64, device=bi.device)])
         li, ri = (pi, bi) if build_side == "right" else (bi, pi)
+        out_left = left.gather(li)
+        out_right = right.gather(ri)
+        return [RecordBatch(out_left.names + out_right.names,
+                            out_left.columns + out_right.columns)]
+
+    def _finish_join_pairs(self, left: RecordBatch, right: RecordBatch,
+                           li: torch.Tensor, ri: torch.Tensor, how: str,
+                           residual, existence_col: str) -> List[RecordBatch]:
+        """Finish a join from raw equi-matched (left,right) row pairs after
+        applying a residual (non-equi) condition per pair. Null residual
+        results count as non-matches (SQL three-valued logic)."""
+        device = self.ctx.device
+        if residual is not None and li.numel():
+            pair = RecordBatch(left.names + right.names,
+                               left.gather(li).columns + right.gather(ri).columns)
+            m = residual.eval(pair)
+            keep = m.data
+            if m.validity is not None:
+                keep = keep & m.validity
+            li, ri = li[keep], ri[keep]
+        nl, nr = left.num_rows, right.num_rows
+        if how in ("semi", "anti", "existence"):
+            lmatched = torch.zeros(nl, dtype=torch.bool, device=device)
+            if li.numel():
+                lmatched[li] = True
+            if how == "semi":
+                return [left.filter(lmatched)]
+            if how == "anti":
+                return [left.filter(~lmatched)]
+            return [RecordBatch(left.names + [existence_col],
+                                left.columns + [Column(dtypes.bool_, lmatched)])]
+        preserved = {"inner": set(), "left": {"left"}, "right": {"right"},
+                     "full": {"left", "right"}}[how]
+        if "left" in preserved:
+            lm = torch.zeros(nl, dtype=torch.bool, device=device)
+            if li.numel():
+                lm[li] = True
+            un = torch.nonzero(~lm, as_tuple=False).flatten()
+            if un.numel():
+                li = torch.cat([li, un])
+                ri = torch.cat([ri, torch.full((un.numel(),), -1,
+                                               dtype=torch.int64, device=device)])
+        if "right" in preserved:
+            rm = torch.zeros(nr, dtype=torch.bool, device=device)
+            matched_r = ri[ri >= 0]
+            if matched_r.numel():
+                rm[matched_r] = True
+            un = torch.nonzero(~rm, as_tuple=False).flatten()
+            if un.numel():
+                ri = torch.cat([ri, un])
+                li = torch.cat([li, torch.full((un.numel(),), -1,
+                                               dtype=torch.int64, device=device)])
         out_left = left.gather(li)
         out_right = right.gather(ri)
         return [RecordBatch(out_left.names + out_right.names,

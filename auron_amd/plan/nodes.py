@@ -87,6 +87,10 @@ class HashJoin(PlanNode):
     build_side: str = "right"  # right|left
     broadcast: bool = False  # build side is broadcast (BHJ) vs co-partitioned (SHJ)
     existence_col: str = "exists"
+    # extra non-equi join condition evaluated per candidate pair (the
+    # reference SMJ inequality-join / Catalyst ExtraCondition analogue);
+    # pairs where it is false/null do not count as matches
+    residual: Optional[Expr] = None
 
     def children(self):
         return [self.left, self.right]
@@ -103,6 +107,7 @@ class SortMergeJoin(PlanNode):
     right_keys: List[Expr]
     how: str = "inner"
     existence_col: str = "exists"
+    residual: Optional[Expr] = None
 
     def children(self):
         return [self.left, self.right]

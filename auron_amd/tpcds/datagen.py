@@ -127,9 +127,12 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
         put("d_month_seq", ((years.astype(np.int64) - 1900) * 12 + moy - 1).astype(np.int32))
         put("d_week_seq", (idx // 7).astype(np.int32) + 1)
         put("d_dow", (idx % 7).astype(np.int32))
+        qoy = ((moy - 1) // 3 + 1)
+        put("d_quarter_name", [f"{int(y)}Q{int(q)}" for y, q in zip(years, qoy)])
     elif name == "time_dim":
         t = np.arange(n, dtype=np.int64) + lo
         put("t_time_sk", t)
+        put("t_time", t)
         hour = (t // 3600).astype(np.int32)
         put("t_hour", hour)
         put("t_minute", ((t % 3600) // 60).astype(np.int32))
@@ -151,6 +154,7 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
         put("i_class", [f"class{int(v)}" for v in rng.integers(1, 17, n)])
         put("i_class_id", rng.integers(1, 17, n).astype(np.int32))
         put("i_current_price", _money(rng, n, 0.09, 99.0), _with_nulls(rng, sks, 0.01)[1])
+        put("i_wholesale_cost", _money(rng, n, 0.02, 88.0), _with_nulls(rng, sks, 0.01)[1])
         put("i_manager_id", rng.integers(1, 101, n).astype(np.int32))
         put("i_product_name", [f"product{int(s)}" for s in sks])
         put("i_item_desc", [f"the quite famous item number {int(s)} description" for s in sks])
@@ -178,13 +182,17 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
         # dsdgen stores birth country uppercased (q24 matches upper(ca_country))
         put("c_birth_country", [_COUNTRIES[int(v)].upper() for v in rng.integers(0, len(_COUNTRIES), n)])
         put("c_birth_year", rng.integers(1924, 1993, n).astype(np.int32))
+        put("c_birth_day", rng.integers(1, 29, n).astype(np.int32))
         put("c_birth_month", rng.integers(1, 13, n).astype(np.int32))
+        put("c_login", ["" for _ in range(n)])
         put("c_email_address", [f"c{int(s)}@example.com" for s in sks])
         d, dv = date_fk(0.02)
         put("c_first_sales_date_sk", d, dv)
         d2, d2v = date_fk(0.02)
         put("c_first_shipto_date_sk", d2, d2v)
         put("c_preferred_cust_flag", ["Y" if v else "N" for v in rng.random(n) < 0.5])
+        d3, d3v = date_fk(0.02)
+        put("c_last_review_date_sk", d3, d3v)
         put("c_salutation", [["Mr.", "Mrs.", "Ms.", "Dr."][int(v)] for v in rng.integers(0, 4, n)])
     elif name == "customer_address":
         put("ca_address_sk", sks)
@@ -236,6 +244,10 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
         put("s_number_employees", rng.integers(200, 301, n).astype(np.int32))
         put("s_gmt_offset", np.full(n, -5.0))
         put("s_company_id", np.ones(n, dtype=np.int32))
+        put("s_company_name", ["Unknown" for _ in range(n)])
+        put("s_street_number", [str(int(v)) for v in rng.integers(1, 1000, n)])
+        put("s_street_type", [["Street", "Avenue", "Blvd", "Court", "Lane"][int(s) % 5] for s in sks])
+        put("s_suite_number", [f"Suite {int(s) % 100}" for s in sks])
         put("s_market_id", ((sks - 1) % 10 + 1).astype(np.int32))
         put("s_street_name", [f"{_LAST[int(s) % len(_LAST)]} Blvd" for s in sks])
     elif name == "warehouse":
@@ -376,6 +388,8 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
         else:
             wp, wpv = fk(BASE_ROWS["web_page"], 0.02)
             put("ws_web_page_sk", wp, wpv)
+            sh, shv = fk(n_hdemo)
+            put("ws_ship_hdemo_sk", sh, shv)
             wsi, wsiv = fk(BASE_ROWS["web_site"], 0.02)
             put("ws_web_site_sk", wsi, wsiv)
         sm, smv = fk(BASE_ROWS["ship_mode"], 0.02)
@@ -431,6 +445,8 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
             put("wr_refunded_addr_sk", ra2, ra2v)
             rr, rrv = fk(BASE_ROWS["reason"], 0.02)
             put("wr_reason_sk", rr, rrv)
+            wwp, wwpv = fk(BASE_ROWS["web_page"], 0.02)
+            put("wr_web_page_sk", wwp, wwpv)
         if pre == "cr":
             cp, cpv = fk(BASE_ROWS["catalog_page"], 0.02)
             put("cr_catalog_page_sk", cp, cpv)

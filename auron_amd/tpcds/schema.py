@@ -19,20 +19,20 @@ SCHEMAS = {
     "date_dim": {
         "d_date_sk": i64, "d_date": d32, "d_year": i32, "d_moy": i32,
         "d_dom": i32, "d_qoy": i32, "d_day_name": s, "d_month_seq": i32,
-        "d_week_seq": i32, "d_dow": i32,
+        "d_week_seq": i32, "d_dow": i32, "d_quarter_name": s,
     },
-    "time_dim": {"t_time_sk": i64, "t_hour": i32, "t_minute": i32, "t_meal_time": s},
+    "time_dim": {"t_time_sk": i64, "t_time": i64, "t_hour": i32, "t_minute": i32, "t_meal_time": s},
     "item": {
         "i_item_sk": i64, "i_item_id": s, "i_item_desc": s, "i_category": s,
         "i_category_id": i32, "i_brand": s, "i_brand_id": i32,
         "i_manufact_id": i32, "i_class": s, "i_class_id": i32,
-        "i_current_price": d72, "i_manager_id": i32, "i_product_name": s, "i_color": s, "i_units": s, "i_size": s, "i_manufact": s,
+        "i_current_price": d72, "i_wholesale_cost": d72, "i_manager_id": i32, "i_product_name": s, "i_color": s, "i_units": s, "i_size": s, "i_manufact": s,
     },
     "customer": {
         "c_customer_sk": i64, "c_customer_id": s, "c_first_name": s,
         "c_last_name": s, "c_current_addr_sk": i64, "c_current_cdemo_sk": i64,
-        "c_current_hdemo_sk": i64, "c_birth_country": s, "c_birth_year": i32,
-        "c_birth_month": i32, "c_email_address": s, "c_first_sales_date_sk": i64,
+        "c_current_hdemo_sk": i64, "c_birth_country": s, "c_birth_year": i32, "c_birth_day": i32,
+        "c_birth_month": i32, "c_login": s, "c_last_review_date_sk": i64, "c_email_address": s, "c_first_sales_date_sk": i64,
         "c_first_shipto_date_sk": i64, "c_preferred_cust_flag": s, "c_salutation": s,
     },
     "customer_address": {
@@ -53,7 +53,8 @@ SCHEMAS = {
     "store": {
         "s_store_sk": i64, "s_store_id": s, "s_store_name": s, "s_state": s,
         "s_county": s, "s_zip": s, "s_city": s, "s_number_employees": i32,
-        "s_gmt_offset": f64, "s_company_id": i32, "s_street_name": s, "s_market_id": i32,
+        "s_gmt_offset": f64, "s_company_id": i32, "s_company_name": s, "s_street_name": s,
+        "s_street_number": s, "s_street_type": s, "s_suite_number": s, "s_market_id": i32,
     },
     "warehouse": {
         "w_warehouse_sk": i64, "w_warehouse_name": s, "w_warehouse_sq_ft": i32,
@@ -115,7 +116,7 @@ SCHEMAS = {
         "ws_sold_date_sk": i64, "ws_sold_time_sk": i64, "ws_ship_date_sk": i64,
         "ws_item_sk": i64, "ws_bill_customer_sk": i64, "ws_bill_cdemo_sk": i64,
         "ws_bill_hdemo_sk": i64, "ws_bill_addr_sk": i64, "ws_ship_customer_sk": i64, "ws_ship_addr_sk": i64,
-        "ws_web_page_sk": i64, "ws_web_site_sk": i64, "ws_ship_mode_sk": i64,
+        "ws_web_page_sk": i64, "ws_web_site_sk": i64, "ws_ship_mode_sk": i64, "ws_ship_hdemo_sk": i64,
         "ws_warehouse_sk": i64, "ws_promo_sk": i64, "ws_order_number": i64,
         "ws_quantity": i32, "ws_wholesale_cost": d72, "ws_list_price": d72,
         "ws_sales_price": d72, "ws_ext_discount_amt": d72, "ws_ext_sales_price": d72,
@@ -125,7 +126,7 @@ SCHEMAS = {
     },
     "web_returns": {
         "wr_returned_date_sk": i64, "wr_item_sk": i64, "wr_order_number": i64,
-        "wr_returning_customer_sk": i64, "wr_returning_addr_sk": i64, "wr_refunded_cdemo_sk": i64, "wr_returning_cdemo_sk": i64, "wr_refunded_addr_sk": i64, "wr_reason_sk": i64, "wr_return_quantity": i32,
+        "wr_returning_customer_sk": i64, "wr_returning_addr_sk": i64, "wr_refunded_cdemo_sk": i64, "wr_returning_cdemo_sk": i64, "wr_refunded_addr_sk": i64, "wr_reason_sk": i64, "wr_web_page_sk": i64, "wr_return_quantity": i32,
         "wr_return_amt": d72, "wr_net_loss": d72, "wr_fee": d72,
         "wr_refunded_cash": d72, "wr_reversed_charge": d72, "wr_account_credit": d72,
     },
