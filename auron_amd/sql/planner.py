@@ -374,6 +374,10 @@ class Planner:
         self.ctes: Dict[str, object] = {}  # name -> MatCTE | ("inline", AST)
         self.mat_ctes: List[MatCTE] = []
         self.sc_n = 0
+        # columns synthesized by subquery joins (__scval/__exists): they
+        # exist only AFTER the sub-join attach, so predicates referencing
+        # them must ride the residual path, never a pre-attach unit filter
+        self._synth_cols: Set[str] = set()
 
     # ------------------------------------------------------------- entry
     def plan(self, q: A.Query) -> LQuery:
@@ -727,9 +731,12 @@ class Planner:
             if refs - set(unit_of):
                 residuals.append((uset, cnd_expr))
                 continue
-            if len(uset) <= 1:
+            if len(uset) <= 1 and not (refs & self._synth_cols):
                 tgt = next(iter(uset)) if uset else 0
                 filters[tgt].append(cnd_expr)
+                continue
+            if refs & self._synth_cols:
+                residuals.append((uset, cnd_expr))
                 continue
             pair = _equi_pair_units(cnd, scope, self, unit_of)
             if pair is not None:
@@ -896,15 +903,19 @@ class Planner:
                                    sub_joins, aliases):
         if isinstance(node, A.Exists):
             exname = self.alloc.fresh("__exists", None)
-            self._plan_exists(node.query, "existence", None, scope, unit_of,
-                              units, sub_joins, exname)
+            anchor = self._plan_exists(node.query, "existence", None, scope,
+                                       unit_of, units, sub_joins, exname)
+            unit_of[exname] = anchor
+            self._synth_cols.add(exname)
             aliases[exname] = col(exname)
             e = A.Ident([exname])
             return A.UnOp("not", e) if node.negated else e
         if isinstance(node, A.InSubquery):
             exname = self.alloc.fresh("__exists", None)
-            self._plan_exists(node.query, "existence", node.operand, scope,
-                              unit_of, units, sub_joins, exname)
+            anchor = self._plan_exists(node.query, "existence", node.operand,
+                                       scope, unit_of, units, sub_joins, exname)
+            unit_of[exname] = anchor
+            self._synth_cols.add(exname)
             aliases[exname] = col(exname)
             e = A.Ident([exname])
             return A.UnOp("not", e) if node.negated else e
@@ -936,8 +947,11 @@ class Planner:
             if len(anchors) != 1:
                 raise SqlError("correlated scalar subquery spans "
                                f"{len(anchors)} relations")
-            sub_joins.append((next(iter(anchors)), rel, "inner", okeys,
+            anchor = next(iter(anchors))
+            sub_joins.append((anchor, rel, "inner", okeys,
                               [col(n) for n in ikey_names], None, ""))
+            unit_of[vn] = anchor
+            self._synth_cols.add(vn)
             aliases[vn] = col(vn)
             return A.Ident([vn])
         if dataclasses.is_dataclass(node):
@@ -1029,8 +1043,10 @@ class Planner:
         anchors = {unit_of[r] for r in outer_refs if r in unit_of}
         if len(anchors) != 1:
             raise SqlError(f"predicate subquery spans {len(anchors)} relations")
-        sub_joins.append((next(iter(anchors)), sub_rel, kind, lkeys, rkeys,
+        anchor = next(iter(anchors))
+        sub_joins.append((anchor, sub_rel, kind, lkeys, rkeys,
                           residual, exname))
+        return anchor
 
     # ------------------------------------------------------- aggregation
     def plan_aggregate(self, sel: A.Select, rel: Rel, scope: Scope, ctx: AggCtx):
@@ -1343,9 +1359,10 @@ class Planner:
             for c, v in node.whens:
                 branches.append((conv(c), conv(v)))
         otherwise = conv(node.else_) if node.else_ is not None else None
-        # retype NULL literals from sibling branches
-        dts = [infer_dtype(v, self.alloc.types) for _, v in branches]
-        if otherwise is not None:
+        # retype NULL literals from (non-null) sibling branches
+        dts = [infer_dtype(v, self.alloc.types) for _, v in branches
+               if not _is_null_lit(v)]
+        if otherwise is not None and not _is_null_lit(otherwise):
             dts.append(infer_dtype(otherwise, self.alloc.types))
         dt = next((d for d in dts if d is not None), None)
         if dt is not None:
@@ -1419,7 +1436,8 @@ class Planner:
         if name == "coalesce":
             args = [conv(a) for a in fc.args]
             dt = next((infer_dtype(a, self.alloc.types) for a in args
-                       if infer_dtype(a, self.alloc.types) is not None), None)
+                       if not _is_null_lit(a)
+                       and infer_dtype(a, self.alloc.types) is not None), None)
             if dt is not None:
                 args = [Literal(None, dt) if _is_null_lit(a) else a for a in args]
             return Coalesce(args)

@@ -534,7 +534,7 @@ def _nparts_for(table: str, sf: float) -> int:
     return 1
 
 
-DATAGEN_VERSION = 11
+DATAGEN_VERSION = 12
 
 
 def dataset_root(root: str, sf: float) -> str:
