@@ -343,7 +343,9 @@ class Arith(Expr):
                 validity = v2 if validity is None else (validity & v2)
             if self.op == "/":
                 if dt.is_integer:
-                    data = torch.div(a, b, rounding_mode="trunc")
+                    # Spark SQL `/` is double division for any input type
+                    # (integer division is a separate `div` operator)
+                    data = a.to(torch.float64) / b.to(torch.float64)
                 else:
                     data = a / b
             else:
@@ -354,7 +356,7 @@ class Arith(Expr):
         else:
             raise ValueError(self.op)
         if self.op == "/" and dt.is_integer:
-            return Column(DataType(dt.code), data, validity)
+            return Column(dtypes.float64, data, validity)
         return Column(dt, data, validity)
 
 

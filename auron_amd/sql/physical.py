@@ -90,7 +90,16 @@ class Lowering:
                 plan = P.Limit(plan, lq.limit)
         elif lq.limit is not None:
             plan = P.Limit(P.Exchange(plan, "single"), lq.limit)
-        items = [Aliased(self.subst_scalars(e), n) for e, n in lq.outputs]
+        # duplicate output names (legal SQL, e.g. q39's self-join selecting
+        # both sides' w_warehouse_sk) get positional suffixes so the
+        # columnar batch stays addressable by name
+        seen: Dict[str, int] = {}
+        items = []
+        for e, n in lq.outputs:
+            k = seen.get(n, 0)
+            seen[n] = k + 1
+            items.append(Aliased(self.subst_scalars(e),
+                                 n if k == 0 else f"{n}__{k + 1}"))
         return P.Project(plan, items)
 
     # ---------------------------------------------------------- dispatch

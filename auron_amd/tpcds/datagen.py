@@ -30,7 +30,15 @@ _STATES = ["TN", "GA", "AL", "SC", "NC", "KY", "VA", "FL", "MS", "LA",
 _COUNTRIES = ["United States"] * 9 + ["Canada"]
 _CITIES = ["Midway", "Fairview", "Oak Grove", "Five Points", "Centerville",
            "Liberty", "Pleasant Hill", "Mount Zion", "Salem", "Union",
-           "Riverside", "Greenfield", "Oakland", "Springdale", "Shiloh"]
+           "Riverside", "Greenfield", "Oakland", "Springdale", "Shiloh",
+           "Edgewood"]
+# county universe = the literals the benchmark queries filter on plus
+# city-derived names (q54 joins ca_county = s_county, so both tables draw
+# from this same pool)
+_COUNTIES = ["Williamson County", "Rush County", "Toole County",
+             "Jefferson County", "Dona Ana County", "La Porte County",
+             "Bronx County", "Franklin Parish", "Orange County"] + \
+    [f"{c} County" for c in _CITIES]
 _FIRST = ["James", "Mary", "John", "Patricia", "Robert", "Jennifer", "Michael",
           "Linda", "William", "Barbara", "David", "Susan", "Richard", "Jessica"]
 _LAST = ["Smith", "Johnson", "Williams", "Brown", "Jones", "Garcia", "Miller",
@@ -38,6 +46,40 @@ _LAST = ["Smith", "Johnson", "Williams", "Brown", "Jones", "Garcia", "Miller",
 _BUY_POTENTIAL = [">10000", "5001-10000", "1001-5000", "501-1000", "0-500", "Unknown"]
 _EDUCATION = ["Primary", "Secondary", "College", "2 yr Degree", "4 yr Degree",
               "Advanced Degree", "Unknown"]
+
+# dsdgen item domains: per-category class lists (tpcds.dst classes) and the
+# brand-name vocabulary; the named brands are the exact literals the
+# benchmark queries filter on, kept dense enough to select rows at small SF
+_CLASSES = [
+    ["arts", "business", "computers", "cooking", "entertainments", "fiction",
+     "history", "home repair", "mystery", "parenting", "reference", "romance",
+     "science", "self-help", "sports", "travel"],  # Books
+    ["infants", "newborn", "school-uniforms", "toddlers"],  # Children
+    ["audio", "automotive", "cameras", "camcorders", "dvd/vcr players",
+     "karoke", "memory", "monitors", "musical", "personal", "portable",
+     "scanners", "stereo", "televisions", "wireless", "disk drives"],  # Electronics
+    ["accent", "bathroom", "bedding", "blinds/shades", "curtains/drapes",
+     "decor", "flatware", "furniture", "glassware", "kids", "lighting",
+     "mattresses", "paint", "rugs", "tables", "wallpaper"],  # Home
+    ["birdal", "costume", "diamonds", "estate", "gold", "jewelry boxes",
+     "loose stones", "mens watch", "pendants", "rings", "semi-precious",
+     "womens watch"],  # Jewelry
+    ["accessories", "pants", "shirts", "sportswear"],  # Men
+    ["classical", "country", "pop", "rock"],  # Music
+    ["athletic", "kids", "mens", "womens"],  # Shoes
+    ["archery", "athletic shoes", "baseball", "basketball", "camping",
+     "fishing", "fitness", "football", "golf", "guns", "hockey", "optics",
+     "outdoor", "pools", "sailing", "tennis"],  # Sports
+    ["dresses", "fragrances", "maternity", "swimwear"],  # Women
+]
+_NAMED_BRANDS = [
+    "amalgimporto #1", "edu packscholar #1", "exportiimporto #1",
+    "exportiunivamalg #9", "importoamalg #1", "scholaramalgamalg #14",
+    "scholaramalgamalg #7", "scholaramalgamalg #9",
+]
+_BRAND_SYL = ["amalg", "importo", "edu pack", "scholar", "brand", "corp",
+              "maxi", "univ", "exporti", "nameless"]
+
 _DAY_NAMES = ["Monday", "Tuesday", "Wednesday", "Thursday", "Friday",
               "Saturday", "Sunday"]  # 1900-01-01 was a Monday
 
@@ -150,9 +192,21 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
         put("i_manufact_id", mfg, _with_nulls(rng, mfg, 0.01)[1])
         brand_id = (cat_id.astype(np.int64) * 1000000 + mfg * 100 + rng.integers(1, 10, n)).astype(np.int32)
         put("i_brand_id", brand_id)
-        put("i_brand", [f"brand#{b}" for b in brand_id])
-        put("i_class", [f"class{int(v)}" for v in rng.integers(1, 17, n)])
-        put("i_class_id", rng.integers(1, 17, n).astype(np.int32))
+        # dsdgen-style brand names: ~30% of items carry one of the named
+        # brands the benchmark queries filter on; the rest get random
+        # syllable-pair brands (mk_word emulation)
+        named = rng.random(n) < 0.30
+        bname_named = rng.integers(0, len(_NAMED_BRANDS), n)
+        s1 = rng.integers(0, len(_BRAND_SYL), n)
+        s2 = rng.integers(0, len(_BRAND_SYL), n)
+        bnum = rng.integers(1, 17, n)
+        put("i_brand", [(_NAMED_BRANDS[int(bn)] if nm else
+                         f"{_BRAND_SYL[int(a)]}{_BRAND_SYL[int(b)]} #{int(k)}")
+                        for nm, bn, a, b, k in zip(named, bname_named, s1, s2, bnum)])
+        cls_idx = rng.integers(0, 64, n)
+        put("i_class", [_CLASSES[int(c - 1)][int(v) % len(_CLASSES[int(c - 1)])]
+                        for c, v in zip(cat_id, cls_idx)])
+        put("i_class_id", (cls_idx % 16 + 1).astype(np.int32))
         put("i_current_price", _money(rng, n, 0.09, 99.0), _with_nulls(rng, sks, 0.01)[1])
         put("i_wholesale_cost", _money(rng, n, 0.02, 88.0), _with_nulls(rng, sks, 0.01)[1])
         put("i_manager_id", rng.integers(1, 101, n).astype(np.int32))
@@ -160,7 +214,10 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
         put("i_item_desc", [f"the quite famous item number {int(s)} description" for s in sks])
         _colors = ["red", "blue", "green", "yellow", "black", "white", "purple",
                    "orange", "pink", "brown", "gray", "cyan", "pale", "powder",
-                   "khaki", "midnight", "snow", "forest", "ghost", "floral"]
+                   "khaki", "midnight", "snow", "forest", "ghost", "floral",
+                   "blanched", "burlywood", "burnished", "chiffon",
+                   "cornflower", "deep", "frosted", "honeydew", "indian",
+                   "light", "medium", "papaya", "slate", "spring"]
         put("i_color", [_colors[int(v)] for v in rng.integers(0, len(_colors), n)])
         _units = ["Ounce", "Oz", "Bunch", "Ton", "N/A", "Dozen", "Box", "Pound",
                   "Pallet", "Gross", "Cup", "Dram", "Each", "Tbl", "Lb", "Bundle"]
@@ -201,7 +258,7 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
         put("ca_zip", [f"{int(v):05d}" for v in rng.integers(10000, 99999, n)])
         put("ca_country", [_COUNTRIES[int(v)] for v in rng.integers(0, len(_COUNTRIES), n)])
         put("ca_city", [_CITIES[int(v)] for v in rng.integers(0, len(_CITIES), n)])
-        put("ca_county", [f"{_CITIES[int(v)]} County" for v in rng.integers(0, len(_CITIES), n)])
+        put("ca_county", [_COUNTIES[int(v)] for v in rng.integers(0, len(_COUNTIES), n)])
         put("ca_gmt_offset", rng.choice([-5.0, -6.0, -7.0, -8.0], n))
         put("ca_street_name", [f"{_LAST[int(v)]} St" for v in rng.integers(0, len(_LAST), n)])
         put("ca_street_type", [["Street", "Avenue", "Blvd", "Court", "Lane"][int(v)] for v in rng.integers(0, 5, n)])
@@ -237,7 +294,7 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
         # half the stores share county names with the address universe so
         # county-equi joins (q54) are non-trivial
         put("s_county", ["Williamson County" if int(s2) % 2 == 0
-                         else f"{_CITIES[int(s2) % len(_CITIES)]} County"
+                         else _COUNTIES[int(s2) % len(_COUNTIES)]
                          for s2 in sks])
         put("s_zip", [f"{int(v):05d}" for v in rng.integers(30000, 40000, n)])
         put("s_city", [_CITIES[int(s) % len(_CITIES)] for s in sks])
@@ -534,7 +591,7 @@ def _nparts_for(table: str, sf: float) -> int:
     return 1
 
 
-DATAGEN_VERSION = 12
+DATAGEN_VERSION = 13
 
 
 def dataset_root(root: str, sf: float) -> str:

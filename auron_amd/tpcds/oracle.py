@@ -79,7 +79,7 @@ def q3(root, sf):
 
 def q6(root, sf):
     dd_all = _read(root, sf, "date_dim", ["d_date_sk", "d_month_seq", "d_year", "d_moy"])
-    ms = dd_all[(dd_all.d_year == 2001) & (dd_all.d_moy == 1)].d_month_seq.iloc[0]
+    ms = dd_all[(dd_all.d_year == 2000) & (dd_all.d_moy == 1)].d_month_seq.iloc[0]
     it = _read(root, sf, "item", ["i_item_sk", "i_category", "i_current_price"])
     cat_avg = it.groupby("i_category", dropna=False).i_current_price.mean().reset_index()
     cat_avg.columns = ["i_category", "cat_avg_price"]
@@ -165,12 +165,15 @@ def q19(root, sf):
     j = _merge(j, ca, "c_current_addr_sk", "ca_address_sk")
     j = _merge(j, st, "ss_store_sk", "s_store_sk")
     j = j[j.ca_zip.str[:5] != j.s_zip.str[:5]]
-    g = j.groupby(["i_brand", "i_brand_id", "i_manufact_id"], dropna=False) \
+    g = j.groupby(["i_brand", "i_brand_id", "i_manufact_id", "i_manufact"],
+                  dropna=False) \
          .ss_ext_sales_price.sum(min_count=1).reset_index()
-    g.columns = ["i_brand", "i_brand_id", "i_manufact_id", "ext_price"]
-    g = g.sort_values(["ext_price", "i_brand", "i_brand_id", "i_manufact_id"],
-                      ascending=[False, True, True, True]).head(100)
-    return g.reset_index(drop=True)
+    g.columns = ["brand", "brand_id", "i_manufact_id", "i_manufact", "ext_price"]
+    g = g.sort_values(["ext_price", "brand", "brand_id", "i_manufact_id",
+                       "i_manufact"],
+                      ascending=[False, True, True, True, True]).head(100)
+    return g[["brand_id", "brand", "i_manufact_id", "i_manufact",
+              "ext_price"]].reset_index(drop=True)
 
 
 def q96(root, sf):
@@ -202,7 +205,8 @@ def q68(root, sf):
     j = _merge(j, st, "ss_store_sk", "s_store_sk")
     j = _merge(j, hd, "ss_hdemo_sk", "hd_demo_sk")
     j = _merge(j, ca, "ss_addr_sk", "ca_address_sk")
-    g = j.groupby(["ss_ticket_number", "ss_customer_sk", "ca_city"], dropna=False).agg(
+    g = j.groupby(["ss_ticket_number", "ss_customer_sk", "ss_addr_sk", "ca_city"],
+                  dropna=False).agg(
         extended_price=("ss_ext_sales_price", lambda x: x.sum(min_count=1)),
         list_price=("ss_ext_list_price", lambda x: x.sum(min_count=1)),
         extended_tax=("ss_ext_tax", lambda x: x.sum(min_count=1))).reset_index()
@@ -214,8 +218,9 @@ def q68(root, sf):
     j3 = _merge(j2, ca2, "c_current_addr_sk", "current_addr_sk")
     f = j3[j3.current_city != j3.bought_city].copy()
     f = f[~(f.current_city.isna() | f.bought_city.isna())]
-    out = f[["c_last_name", "c_first_name", "bought_city", "ss_ticket_number",
-             "extended_price", "extended_tax", "list_price"]]
+    f = f.rename(columns={"current_city": "ca_city"})
+    out = f[["c_last_name", "c_first_name", "ca_city", "bought_city",
+             "ss_ticket_number", "extended_price", "extended_tax", "list_price"]]
     out = out.sort_values(["c_last_name", "ss_ticket_number"]).head(100)
     return out.reset_index(drop=True)
 
@@ -361,10 +366,12 @@ def q65(root, sf):
     # boundary rows may flip; the test checks membership in this superset
     f = j2[j2.revenue <= 0.1 * j2.ave * (1 + 1e-9) + 1e-9]
     st = _read(root, sf, "store", ["s_store_sk", "s_store_name"])
-    it = _read(root, sf, "item", ["i_item_sk", "i_item_desc", "i_current_price", "i_brand"])
+    it = _read(root, sf, "item", ["i_item_sk", "i_item_desc", "i_current_price",
+                                  "i_wholesale_cost", "i_brand"])
     j3 = _merge(f, st, "ss_store_sk", "s_store_sk")
     j4 = _merge(j3, it, "ss_item_sk", "i_item_sk")
-    out = j4[["s_store_name", "i_item_desc", "revenue", "i_current_price", "i_brand"]]
+    out = j4[["s_store_name", "i_item_desc", "revenue", "i_current_price",
+              "i_wholesale_cost", "i_brand"]]
     out = out.sort_values(["s_store_name", "i_item_desc"], na_position="first")
     return out.reset_index(drop=True)
 
@@ -539,7 +546,7 @@ def _date_i(dd):
     return pd.to_datetime(dd.d_date).map(lambda x: x.toordinal() - 719163)
 
 
-def _ratio_window_oracle(root, sf, fact, pre, measure):
+def _ratio_window_oracle(root, sf, fact, pre, measure, limit=100):
     ss = _read(root, sf, fact, [f"{pre}_sold_date_sk", f"{pre}_item_sk", measure])
     it = _read(root, sf, "item", ["i_item_sk", "i_item_id", "i_item_desc",
                                   "i_category", "i_class", "i_current_price"])
@@ -556,7 +563,9 @@ def _ratio_window_oracle(root, sf, fact, pre, measure):
     g["_clsrev"] = g.groupby("i_class", dropna=False).itemrevenue.transform("sum")
     g["revenueratio"] = g.itemrevenue * 100.0 / g._clsrev
     g = g.sort_values(["i_category", "i_class", "i_item_id", "i_item_desc",
-                       "revenueratio"], na_position="first").head(100)
+                       "revenueratio"], na_position="first")
+    if limit is not None:
+        g = g.head(limit)
     return g[["i_item_desc", "i_category", "i_class", "i_current_price",
               "itemrevenue", "revenueratio"]].reset_index(drop=True)
 
@@ -570,7 +579,7 @@ def q20(root, sf):
 
 
 def q98(root, sf):
-    return _ratio_window_oracle(root, sf, "store_sales", "ss", "ss_ext_sales_price")
+    return _ratio_window_oracle(root, sf, "store_sales", "ss", "ss_ext_sales_price", limit=None)
 
 
 def q15(root, sf):
@@ -639,14 +648,18 @@ def q26(root, sf):
 
 
 def _monthly_window_oracle(root, sf, group_key, extra_group, out_cols, sort_cols,
-                           dd_pred):
+                           dd_pred, classes1):
     ss = _read(root, sf, "store_sales", ["ss_item_sk", "ss_sold_date_sk", "ss_store_sk",
                                          "ss_sales_price"])
     it = _read(root, sf, "item")
     m = ((it.i_category.isin(["Books", "Children", "Electronics"])
-          & it.i_class.isin(["class1", "class2", "class3", "class4"]))
+          & it.i_class.isin(classes1)
+          & it.i_brand.isin(["scholaramalgamalg #14", "scholaramalgamalg #7",
+                             "exportiunivamalg #9", "scholaramalgamalg #9"]))
          | (it.i_category.isin(["Women", "Music", "Men"])
-            & it.i_class.isin(["class5", "class6", "class7", "class8"])))
+            & it.i_class.isin(["accessories", "classical", "fragrances", "pants"])
+            & it.i_brand.isin(["amalgimporto #1", "edu packscholar #1",
+                               "exportiimporto #1", "importoamalg #1"])))
     it = it[m]
     dd = _read(root, sf, "date_dim")
     dd = dd[dd_pred(dd)]
@@ -667,43 +680,45 @@ def q53(root, sf):
     return _monthly_window_oracle(root, sf, "i_manufact_id", "d_qoy",
                                   ["i_manufact_id", "sum_sales", "avg_sales"],
                                   ["avg_sales", "sum_sales", "i_manufact_id"],
-                                  lambda d: d.d_month_seq.between(1200, 1211))
+                                  lambda d: d.d_month_seq.between(1200, 1211),
+                                  ["personal", "portable", "reference", "self-help"])
 
 
 def q63(root, sf):
     return _monthly_window_oracle(root, sf, "i_manager_id", "d_moy",
                                   ["i_manager_id", "sum_sales", "avg_sales"],
                                   ["i_manager_id", "avg_sales", "sum_sales"],
-                                  lambda d: d.d_month_seq.between(1200, 1211))
+                                  lambda d: d.d_month_seq.between(1200, 1211),
+                                  ["personal", "portable", "refernece", "self-help"])
 
 
 def q89(root, sf):
     ss = _read(root, sf, "store_sales", ["ss_item_sk", "ss_sold_date_sk", "ss_store_sk",
                                          "ss_sales_price"])
     it = _read(root, sf, "item")
-    m = ((it.i_category.isin(["Books", "Children", "Electronics"])
-          & it.i_class.isin(["class1", "class2", "class3", "class4"]))
-         | (it.i_category.isin(["Women", "Music", "Men"])
-            & it.i_class.isin(["class5", "class6", "class7", "class8"])))
+    m = ((it.i_category.isin(["Books", "Electronics", "Sports"])
+          & it.i_class.isin(["computers", "stereo", "football"]))
+         | (it.i_category.isin(["Men", "Jewelry", "Women"])
+            & it.i_class.isin(["shirts", "birdal", "dresses"])))
     it = it[m]
     dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year", "d_moy"])
     dd = dd[dd.d_year == 1999]
-    st = _read(root, sf, "store", ["s_store_sk", "s_store_name", "s_company_id"])
+    st = _read(root, sf, "store", ["s_store_sk", "s_store_name", "s_company_name"])
     j = _merge(ss, it, "ss_item_sk", "i_item_sk")
     j = _merge(j, dd, "ss_sold_date_sk", "d_date_sk")
     j = _merge(j, st, "ss_store_sk", "s_store_sk")
     g = j.groupby(["i_category", "i_class", "i_brand", "s_store_name",
-                   "s_company_id", "d_moy"], dropna=False) \
+                   "s_company_name", "d_moy"], dropna=False) \
          .ss_sales_price.sum(min_count=1).reset_index(name="sum_sales")
     g["avg_monthly_sales"] = g.groupby(
-        ["i_category", "i_brand", "s_store_name", "s_company_id"],
+        ["i_category", "i_brand", "s_store_name", "s_company_name"],
         dropna=False).sum_sales.transform("mean")
     cond = (g.avg_monthly_sales != 0) & \
         ((g.sum_sales - g.avg_monthly_sales).abs() / g.avg_monthly_sales > 0.1)
     f = g[cond.fillna(False)].copy()
     f["_d"] = f.sum_sales - f.avg_monthly_sales
     f = f.sort_values(["_d", "s_store_name"], na_position="first").head(100)
-    return f[["i_category", "i_class", "i_brand", "s_store_name", "s_company_id",
+    return f[["i_category", "i_class", "i_brand", "s_store_name", "s_company_name",
               "d_moy", "sum_sales", "avg_monthly_sales"]].reset_index(drop=True)
 
 
@@ -769,7 +784,7 @@ def q90(root, sf):
     import pandas as pd
 
     def cnt(h_lo, h_hi):
-        ws = _read(root, sf, "web_sales", ["ws_sold_time_sk", "ws_bill_hdemo_sk", "ws_web_page_sk"])
+        ws = _read(root, sf, "web_sales", ["ws_sold_time_sk", "ws_ship_hdemo_sk", "ws_web_page_sk"])
         td = _read(root, sf, "time_dim")
         td = td[td.t_hour.between(h_lo, h_hi)]
         hd = _read(root, sf, "household_demographics")
@@ -777,7 +792,7 @@ def q90(root, sf):
         wp = _read(root, sf, "web_page")
         wp = wp[wp.wp_char_count.between(5000, 5200)]
         j = _merge(ws, td, "ws_sold_time_sk", "t_time_sk")
-        j = _merge(j, hd, "ws_bill_hdemo_sk", "hd_demo_sk")
+        j = _merge(j, hd, "ws_ship_hdemo_sk", "hd_demo_sk")
         j = _merge(j, wp, "ws_web_page_sk", "wp_web_page_sk")
         return len(j)
 
@@ -995,7 +1010,8 @@ def q50(root, sf):
         srj.dropna(subset=["sr_ticket_number", "sr_item_sk", "sr_customer_sk"]),
         left_on=["ss_ticket_number", "ss_item_sk", "ss_customer_sk"],
         right_on=["sr_ticket_number", "sr_item_sk", "sr_customer_sk"])
-    st_cols = ["s_store_name", "s_company_id", "s_street_name", "s_city",
+    st_cols = ["s_store_name", "s_company_id", "s_street_number",
+               "s_street_name", "s_street_type", "s_suite_number", "s_city",
                "s_county", "s_state", "s_zip"]
     st = _read(root, sf, "store", ["s_store_sk"] + st_cols)
     j = _merge(j, st, "ss_store_sk", "s_store_sk")
@@ -1244,7 +1260,7 @@ def q5(root, sf):
     di = _date_i(dd)
     dd = dd[(di >= d0) & (di <= d0 + 14)]
 
-    def part(rows, dim, dim_key, id_col, tag):
+    def part(rows, dim, dim_key, id_col, tag, id_prefix):
         j = _merge(rows, dd, "date_sk", "d_date_sk")
         j = _merge(j, dim, "fk", dim_key)
         a = j.groupby(id_col, dropna=False).agg(
@@ -1253,7 +1269,7 @@ def q5(root, sf):
             p1=("profit", lambda x: x.sum(min_count=1)),
             p2=("net_loss", lambda x: x.sum(min_count=1))).reset_index()
         a["channel"] = f"{tag} channel"
-        a["id"] = tag.replace(" ", "_") + a[id_col].astype(str)
+        a["id"] = id_prefix + a[id_col].astype(str)
         a["profit"] = a.p1 - a.p2
         return a[["channel", "id", "sales", "returns", "profit"]]
 
@@ -1273,7 +1289,7 @@ def q5(root, sf):
     st = _read(root, sf, "store", ["s_store_sk", "s_store_id"])
     ssr = part(pd.concat([mk(ss, "ss_store_sk", "ss_sold_date_sk", sp="ss_ext_sales_price", pr="ss_net_profit"),
                           mk(sr, "sr_store_sk", "sr_returned_date_sk", ra="sr_return_amt", nl="sr_net_loss")],
-                         ignore_index=True), st, "s_store_sk", "s_store_id", "store")
+                         ignore_index=True), st, "s_store_sk", "s_store_id", "store", "store")
     cs = _read(root, sf, "catalog_sales", ["cs_catalog_page_sk", "cs_sold_date_sk",
                                            "cs_ext_sales_price", "cs_net_profit"])
     cr = _read(root, sf, "catalog_returns", ["cr_catalog_page_sk", "cr_returned_date_sk",
@@ -1281,7 +1297,7 @@ def q5(root, sf):
     cp = _read(root, sf, "catalog_page", ["cp_catalog_page_sk", "cp_catalog_page_id"])
     csr = part(pd.concat([mk(cs, "cs_catalog_page_sk", "cs_sold_date_sk", sp="cs_ext_sales_price", pr="cs_net_profit"),
                           mk(cr, "cr_catalog_page_sk", "cr_returned_date_sk", ra="cr_return_amount", nl="cr_net_loss")],
-                         ignore_index=True), cp, "cp_catalog_page_sk", "cp_catalog_page_id", "catalog page")
+                         ignore_index=True), cp, "cp_catalog_page_sk", "cp_catalog_page_id", "catalog", "catalog_page")
     ws = _read(root, sf, "web_sales", ["ws_web_site_sk", "ws_sold_date_sk",
                                        "ws_ext_sales_price", "ws_net_profit",
                                        "ws_item_sk", "ws_order_number"])
@@ -1294,7 +1310,7 @@ def q5(root, sf):
     web = _read(root, sf, "web_site", ["web_site_sk", "web_site_id"])
     wsr = part(pd.concat([mk(ws, "ws_web_site_sk", "ws_sold_date_sk", sp="ws_ext_sales_price", pr="ws_net_profit"),
                           mk(wrj, "ws_web_site_sk", "wr_returned_date_sk", ra="wr_return_amt", nl="wr_net_loss")],
-                         ignore_index=True), web, "web_site_sk", "web_site_id", "web site")
+                         ignore_index=True), web, "web_site_sk", "web_site_id", "web", "web_site")
     u = pd.concat([ssr, csr, wsr], ignore_index=True)
     out = _rollup2_oracle(u, "channel", "id")
     out = out.sort_values(["channel", "id"], na_position="first").head(100)
@@ -1318,6 +1334,7 @@ def q77(root, sf):
 
     ss = cte("store_sales", "ss_sold_date_sk", "ss_store_sk",
              "ss_ext_sales_price", "ss_net_profit")
+    ss = ss.dropna(subset=["ss_store_sk"])  # SQL joins store on the key
     srr = _read(root, sf, "store_returns", ["sr_returned_date_sk", "sr_store_sk",
                                             "sr_return_amt", "sr_net_loss"])
     srj = _merge(srr, dd, "sr_returned_date_sk", "d_date_sk")
@@ -1328,7 +1345,7 @@ def q77(root, sf):
                      right_on="sr_store_sk", how="left")
     store_rows = pd.DataFrame({
         "channel": "store channel",
-        "id": store.ss_store_sk.map(lambda v: None if pd.isna(v) else str(int(v))),
+        "id": store.ss_store_sk.astype("Int64"),
         "sales": store.sales,
         "returns": store.returns.fillna(0.0),
         "profit": store.profit - store.profit_loss.fillna(0.0)})
@@ -1342,26 +1359,25 @@ def q77(root, sf):
     cr_loss = crj.cr_net_loss.sum(min_count=1)
     catalog_rows = pd.DataFrame({
         "channel": "catalog channel",
-        "id": cs.cs_call_center_sk.map(lambda v: None if pd.isna(v) else str(int(v))),
+        "id": cs.cs_call_center_sk.astype("Int64"),
         "sales": cs.sales, "returns": float(cr_ret or 0.0),
         "profit": cs.profit - float(cr_loss or 0.0)})
 
     ws = cte("web_sales", "ws_sold_date_sk", "ws_web_page_sk",
              "ws_ext_sales_price", "ws_net_profit")
-    wrr = _read(root, sf, "web_returns", ["wr_item_sk", "wr_order_number",
+    ws = ws.dropna(subset=["ws_web_page_sk"])  # SQL joins web_page
+    wrr = _read(root, sf, "web_returns", ["wr_web_page_sk",
                                           "wr_returned_date_sk", "wr_return_amt", "wr_net_loss"])
     wrj = _merge(wrr, dd, "wr_returned_date_sk", "d_date_sk")
-    wss = _read(root, sf, "web_sales", ["ws_item_sk", "ws_order_number", "ws_web_page_sk"])
-    wrj = wrj.merge(wss.dropna(subset=["ws_item_sk", "ws_order_number"]),
-                    left_on=["wr_item_sk", "wr_order_number"],
-                    right_on=["ws_item_sk", "ws_order_number"], how="left")
+    wrj = wrj.dropna(subset=["wr_web_page_sk"])
+    wrj = wrj.rename(columns={"wr_web_page_sk": "ws_web_page_sk"})
     wra = wrj.groupby("ws_web_page_sk", dropna=False).agg(
         returns=("wr_return_amt", lambda x: x.sum(min_count=1)),
         profit_loss=("wr_net_loss", lambda x: x.sum(min_count=1))).reset_index()
     web = ws.merge(wra.dropna(subset=["ws_web_page_sk"]), on="ws_web_page_sk", how="left")
     web_rows = pd.DataFrame({
         "channel": "web channel",
-        "id": web.ws_web_page_sk.map(lambda v: None if pd.isna(v) else str(int(v))),
+        "id": web.ws_web_page_sk.astype("Int64"),
         "sales": web.sales, "returns": web.returns.fillna(0.0),
         "profit": web.profit - web.profit_loss.fillna(0.0)})
 
@@ -1402,7 +1418,7 @@ def q80(root, sf):
             sales=("sales_v", lambda x: x.sum(min_count=1)),
             returns=("ret_v", lambda x: x.sum(min_count=1)),
             profit=("prof_v", lambda x: x.sum(min_count=1))).reset_index()
-        a["channel"] = f"{tag} channel"
+        a["channel"] = f"{tag.split('_')[0]} channel"
         a["id"] = tag + a[id_col].astype(str)
         return a[["channel", "id", "sales", "returns", "profit"]]
 
@@ -1442,9 +1458,13 @@ def _v1_window_oracle(root, sf, fact, pre, fk, dim, dim_key, dim_cols, measure, 
     g = g.sort_values(part4 + ["d_year", "d_moy"])
     g["psum"] = g.groupby(part4, dropna=False).sum_sales.shift(1)
     g["nsum"] = g.groupby(part4, dropna=False).sum_sales.shift(-1)
+    # neighbor-ROW existence, not neighbor-sum non-null: the SQL self-join
+    # keeps rows whose lag/lead month exists even if its sum is NULL
+    g["_pex"] = g.groupby(part4, dropna=False).d_moy.shift(1).notna()
+    g["_nex"] = g.groupby(part4, dropna=False).d_moy.shift(-1).notna()
     cond = ((g.d_year == 1999) & (g.avg_monthly_sales > 0)
             & ((g.sum_sales - g.avg_monthly_sales).abs() / g.avg_monthly_sales > 0.1)
-            & g.psum.notna() & g.nsum.notna())
+            & g._pex & g._nex)
     f = g[cond.fillna(False)].copy()
     f["_d"] = f.sum_sales - f.avg_monthly_sales
     f = f.sort_values(["_d", order_out], na_position="first").head(100)
@@ -1454,9 +1474,9 @@ def _v1_window_oracle(root, sf, fact, pre, fk, dim, dim_key, dim_cols, measure, 
 
 def q47(root, sf):
     return _v1_window_oracle(root, sf, "store_sales", "ss", "ss_store_sk",
-                             "store", "s_store_sk", ["s_store_name", "s_company_id"],
+                             "store", "s_store_sk", ["s_store_name", "s_company_name"],
                              "ss_sales_price",
-                             ["i_category", "i_brand", "s_store_name", "s_company_id"],
+                             ["i_category", "i_brand", "s_store_name", "s_company_name"],
                              "s_store_name")
 
 
@@ -1689,7 +1709,8 @@ def q76(root, sf):
         j = _merge(fs, dd, f"{pre}_sold_date_sk", "d_date_sk")
         j = _merge(j, it, f"{pre}_item_sk", "i_item_sk")
         j["channel"] = tag
-        j["col_name"] = null_col
+        # SQL selects the (all-NULL by the filter) column VALUE, not its name
+        j["col_name"] = None
         j = j.rename(columns={f"{pre}_ext_sales_price": "ext_sales_price"})
         return j[["channel", "col_name", "d_year", "d_qoy", "i_category", "ext_sales_price"]]
 
@@ -1742,24 +1763,32 @@ def q81(root, sf):
 def q30(root, sf):
     return _ctr_state_oracle(root, sf, "web_returns", "wr", "wr_return_amt", 2002,
                              ["c_customer_id", "c_salutation", "c_first_name", "c_last_name",
-                              "c_preferred_cust_flag", "c_birth_month", "c_birth_year",
-                              "c_birth_country", "c_email_address"])
+                              "c_preferred_cust_flag", "c_birth_day", "c_birth_month",
+                              "c_birth_year", "c_birth_country", "c_login",
+                              "c_email_address", "c_last_review_date_sk"])
 
 
-def _yoy_oracle(root, sf, channels, first_year, out_expr, out_name):
+def _yoy_oracle(root, sf, channels, first_year, out_expr, out_name,
+                out_cols=None, mask_fn=None):
+    """out_cols: [(customer col, output name)] taken from the second-year
+    row of the first channel (t_s_secyear.* in the SQL); falls back to the
+    single out_expr/out_name pair."""
+    cust_attrs = ["c_customer_id", "c_preferred_cust_flag", "c_first_name",
+                  "c_last_name"]
+    if out_cols is not None:
+        for c, _ in out_cols:
+            if c not in cust_attrs:
+                cust_attrs.append(c)
     totals = {}
     for tag, (fact, pre, fk, cols, mfn) in channels.items():
         fs = _read(root, sf, fact, sorted({fk, f"{pre}_sold_date_sk"} | set(cols)))
         dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year"])
         dd = dd[dd.d_year.isin([first_year, first_year + 1])]
-        cust = _read(root, sf, "customer", ["c_customer_sk", "c_customer_id",
-                                            "c_preferred_cust_flag", "c_first_name",
-                                            "c_last_name"])
+        cust = _read(root, sf, "customer", ["c_customer_sk"] + cust_attrs)
         j = _merge(fs, dd, f"{pre}_sold_date_sk", "d_date_sk")
         j = _merge(j, cust, fk, "c_customer_sk")
         j["v"] = mfn(j)
-        totals[tag] = j.groupby(["c_customer_id", "c_preferred_cust_flag",
-                                 "c_first_name", "c_last_name", "d_year"], dropna=False) \
+        totals[tag] = j.groupby(cust_attrs + ["d_year"], dropna=False) \
                        .v.sum(min_count=1).reset_index(name="year_total")
 
     tags = list(channels.keys())
@@ -1778,10 +1807,23 @@ def _yoy_oracle(root, sf, channels, first_year, out_expr, out_name):
             .rename(columns={"year_total": f"x{i}s"})
         keep = keep.merge(t1, on="c_customer_id").merge(t2, on="c_customer_id")
     ratio_s = keep.year_total_ssec / keep.year_total_sf
-    for i, tag in enumerate(tags[1:]):
-        m = (keep[f"x{i}s"] / keep[f"x{i}f"]) > ratio_s
-        mask = m if mask is None else (mask & m)
+    if mask_fn is not None:
+        mask = mask_fn(keep, ratio_s)
+    else:
+        for i, tag in enumerate(tags[1:]):
+            m = (keep[f"x{i}s"] / keep[f"x{i}f"]) > ratio_s
+            mask = m if mask is None else (mask & m)
     f = keep[mask.fillna(False)]
+    if out_cols is not None:
+        sel = {}
+        for c, name in out_cols:
+            src_col = c if c in f.columns else f"{c}_ssec"
+            sel[name] = f[src_col]
+        import pandas as pd
+
+        out = pd.DataFrame(sel)
+        out = out.sort_values(list(sel.keys()), na_position="first").head(100)
+        return out.reset_index(drop=True)
     out = f[[out_expr]].rename(columns={out_expr: out_name})
     out = out.sort_values(out_name, na_position="first").head(100)
     return out.reset_index(drop=True)
@@ -1807,7 +1849,17 @@ def q4(root, sf):
                "ws_ext_sales_price"],
               half("ws_ext_list_price", "ws_ext_wholesale_cost", "ws_ext_discount_amt",
                    "ws_ext_sales_price")),
-    }, 2001, "c_preferred_cust_flag_ssec", "customer_preferred_cust_flag")
+    }, 2001, "c_preferred_cust_flag_ssec", "customer_preferred_cust_flag",
+        out_cols=[("c_customer_id", "customer_id"),
+                  ("c_first_name", "customer_first_name"),
+                  ("c_last_name", "customer_last_name"),
+                  ("c_preferred_cust_flag", "customer_preferred_cust_flag"),
+                  ("c_birth_country", "customer_birth_country"),
+                  ("c_login", "customer_login"),
+                  ("c_email_address", "customer_email_address")],
+        # SQL: ratio_c > ratio_s AND ratio_c > ratio_w
+        mask_fn=lambda k, rs: ((k.x0s / k.x0f) > rs)
+        & ((k.x0s / k.x0f) > (k.x1s / k.x1f)))
 
 
 def q74(root, sf):
@@ -1816,7 +1868,10 @@ def q74(root, sf):
               lambda j: j.ss_net_paid),
         "w": ("web_sales", "ws", "ws_bill_customer_sk", ["ws_net_paid"],
               lambda j: j.ws_net_paid),
-    }, 2001, "c_customer_id", "customer_id")
+    }, 2001, "c_customer_id", "customer_id",
+        out_cols=[("c_customer_id", "customer_id"),
+                  ("c_first_name", "customer_first_name"),
+                  ("c_last_name", "customer_last_name")])
 
 
 ORACLES.update({"q4": q4, "q13": q13, "q27": q27, "q30": q30, "q36": q36,
@@ -1899,18 +1954,26 @@ def q35(root, sf):
     g = j.groupby(keys, dropna=False).size().reset_index(name="cnt1")
     g["cnt2"] = g.cnt1
     g["cnt3"] = g.cnt1
+    # min/max/avg of a column that is itself a group key (SQL selects them)
+    for c, tag in [("cd_dep_count", "dc"), ("cd_dep_employed_count", "ec"),
+                   ("cd_dep_college_count", "cc")]:
+        g[f"min_{tag}"] = g[c]
+        g[f"max_{tag}"] = g[c]
+        g[f"avg_{tag}"] = g[c].astype(float)
     g = g.sort_values(keys, na_position="first").head(100)
-    return g[["ca_state", "cd_gender", "cd_marital_status", "cnt1", "cd_dep_count",
-              "cd_dep_employed_count", "cnt2", "cd_dep_college_count", "cnt3"]] \
-        .reset_index(drop=True)
+    return g[["ca_state", "cd_gender", "cd_marital_status", "cnt1",
+              "min_dc", "max_dc", "avg_dc",
+              "cd_dep_employed_count", "cnt2", "min_ec", "max_ec", "avg_ec",
+              "cd_dep_college_count", "cnt3", "min_cc", "max_cc",
+              "avg_cc"]].reset_index(drop=True)
 
 
 def q56(root, sf):
     import pandas as pd
 
     it_all = _read(root, sf, "item", ["i_item_sk", "i_item_id", "i_color"])
-    ids = set(it_all[it_all.i_color.isin(["slate", "blanched", "burnished",
-                                          "red", "blue", "green"])].i_item_id)
+    ids = set(it_all[it_all.i_color.isin(["slate", "blanched",
+                                          "burnished"])].i_item_id)
 
     def chan(fact, pre, addr_fk):
         fs = _read(root, sf, fact, [f"{pre}_item_sk", f"{pre}_sold_date_sk", addr_fk,
@@ -2112,9 +2175,10 @@ def q2(root, sf):
     wswscs = j.groupby("d_week_seq", dropna=False).agg(
         **{f"{t}_sales": (f"{t}_sales", lambda x: x.sum(min_count=1)) for t, _ in days}) \
         .reset_index()
-    weeks = dd[["d_week_seq", "d_year"]].drop_duplicates()
-    y = wswscs[wswscs.d_week_seq.isin(set(weeks[weeks.d_year == 2001].d_week_seq))]
-    z = wswscs[wswscs.d_week_seq.isin(set(weeks[weeks.d_year == 2002].d_week_seq))].copy()
+    # the SQL re-joins wswscs against date_dim ROWS (one per day), so each
+    # week contributes 7x multiplicity on each side — keep it
+    y = wswscs.merge(dd[dd.d_year == 2001][["d_week_seq"]], on="d_week_seq")
+    z = wswscs.merge(dd[dd.d_year == 2002][["d_week_seq"]], on="d_week_seq").copy()
     z["wk_join"] = z.d_week_seq - 53
     m = y.merge(z, left_on="d_week_seq", right_on="wk_join", suffixes=("1", "2"))
     out = {"d_week_seq1": m.d_week_seq1}
@@ -2144,8 +2208,8 @@ def q10(root, sf):
     cust = _read(root, sf, "customer", ["c_customer_sk", "c_current_addr_sk",
                                         "c_current_cdemo_sk"])
     ca = _read(root, sf, "customer_address", ["ca_address_sk", "ca_county"])
-    ca = ca[ca.ca_county.isin(["Midway County", "Fairview County", "Oak Grove County",
-                               "Salem County", "Liberty County"])]
+    ca = ca[ca.ca_county.isin(["Rush County", "Toole County", "Jefferson County",
+                               "Dona Ana County", "La Porte County"])]
     cd = _read(root, sf, "customer_demographics")
     dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year", "d_moy"])
     dd = dd[(dd.d_year == 2002) & dd.d_moy.between(1, 4)]
@@ -2695,20 +2759,20 @@ def q66(root, sf):
     dd = _read(root, sf, "date_dim", ["d_date_sk", "d_year", "d_moy"])
     dd = dd[dd.d_year == 2001]
     sm = _read(root, sf, "ship_mode", ["sm_ship_mode_sk", "sm_carrier"])
-    sm = sm[sm.sm_carrier.isin(["UPS", "FEDEX"])]
+    sm = sm[sm.sm_carrier.isin(["DHL", "BARIAN"])]
     wh = _read(root, sf, "warehouse")
     months = ["jan", "feb", "mar", "apr", "may", "jun",
               "jul", "aug", "sep", "oct", "nov", "dec"]
 
-    def chan(fact, pre, netcol):
+    def chan(fact, pre, netcol, salescol):
         fs = _read(root, sf, fact, [f"{pre}_warehouse_sk", f"{pre}_sold_date_sk",
                                     f"{pre}_sold_time_sk", f"{pre}_ship_mode_sk",
-                                    f"{pre}_ext_sales_price", f"{pre}_quantity", netcol])
-        fs = fs[(fs[f"{pre}_sold_time_sk"] >= 30838) & (fs[f"{pre}_sold_time_sk"] <= 59838)]
+                                    salescol, f"{pre}_quantity", netcol])
+        fs = fs[(fs[f"{pre}_sold_time_sk"] >= 30838) & (fs[f"{pre}_sold_time_sk"] <= 59638)]
         j = _merge(fs, dd, f"{pre}_sold_date_sk", "d_date_sk")
         j = _merge(j, sm, f"{pre}_ship_mode_sk", "sm_ship_mode_sk")
         j = _merge(j, wh, f"{pre}_warehouse_sk", "w_warehouse_sk")
-        sales = j[f"{pre}_ext_sales_price"] * j[f"{pre}_quantity"]
+        sales = j[salescol] * j[f"{pre}_quantity"]
         net = j[netcol] * j[f"{pre}_quantity"]
         for i, m in enumerate(months):
             j[f"{m}_sales"] = np.where(j.d_moy == i + 1, sales, 0.0)
@@ -2720,8 +2784,8 @@ def q66(root, sf):
         cols = [f"{m}_sales" for m in months] + [f"{m}_net" for m in months]
         return j.groupby(keys, dropna=False)[cols].sum(min_count=1).reset_index()
 
-    wsr = chan("web_sales", "ws", "ws_net_paid")
-    csr = chan("catalog_sales", "cs", "cs_net_paid_inc_tax")
+    wsr = chan("web_sales", "ws", "ws_net_paid", "ws_ext_sales_price")
+    csr = chan("catalog_sales", "cs", "cs_net_paid_inc_tax", "cs_sales_price")
     u = pd.concat([wsr, csr], ignore_index=True)
     keys = ["w_warehouse_name", "w_warehouse_sq_ft", "w_city", "w_county",
             "w_state", "w_country", "d_year"]
@@ -2731,7 +2795,7 @@ def q66(root, sf):
     cols = ([f"{m}_sales" for m in months] + [f"{m}_spsf" for m in months]
             + [f"{m}_net" for m in months])
     g = u.groupby(keys, dropna=False)[cols].sum(min_count=1).reset_index()
-    g["ship_carriers"] = "UPS,FEDEX"
+    g["ship_carriers"] = "DHL,BARIAN"
     g = g.rename(columns={"d_year": "year"})
     out_cols = (keys[:6] + ["ship_carriers", "year"]
                 + [f"{m}_sales" for m in months]
@@ -2907,7 +2971,7 @@ def q54(root, sf):
     dd_all = _read(root, sf, "date_dim", ["d_date_sk", "d_moy", "d_year", "d_month_seq"])
     ms0 = dd_all[(dd_all.d_year == 1998) & (dd_all.d_moy == 12)].d_month_seq.iloc[0]
     it = _read(root, sf, "item", ["i_item_sk", "i_category", "i_class"])
-    it = it[(it.i_category == "Women") & (it.i_class == "class1")]
+    it = it[(it.i_category == "Women") & (it.i_class == "maternity")]
     dd = dd_all[(dd_all.d_moy == 12) & (dd_all.d_year == 1998)]
 
     def chan(fact, pre, fk):
