@@ -527,8 +527,9 @@ class Planner:
             e = self.to_expr(it.expr, scope, ctx)
             name = it.alias
             if name is None:
-                name = it.expr.name if isinstance(it.expr, A.Ident) else f"_c{len(outputs)}"
-            outputs.append((e, name.lower() if not name.startswith("_c") else name))
+                name = (it.expr.name if isinstance(it.expr, A.Ident)
+                        else _auto_name(it.expr))
+            outputs.append((e, name.lower()))
         if sel.distinct:
             # materialize outputs, then group by all
             names = []
@@ -1652,6 +1653,48 @@ def _contains_subquery(n) -> bool:
                     if isinstance(x, A.ANode) and _contains_subquery(x):
                         return True
     return False
+
+
+def _auto_name(node: A.ANode) -> str:
+    """Spark-style auto-name for an unaliased select item: a compact
+    lower-cased rendering of the expression text (NativeConverters keeps
+    Catalyst's generated names; this mirrors that convention)."""
+    r = _render
+    return r(node)
+
+
+def _render(n) -> str:
+    if isinstance(n, A.Num):
+        return n.text
+    if isinstance(n, A.Str):
+        return n.value
+    if isinstance(n, A.Null):
+        return "null"
+    if isinstance(n, A.Ident):
+        return n.name.lower()
+    if isinstance(n, A.FuncCall):
+        if n.star:
+            return f"{n.name}(1)" if n.name == "count" else f"{n.name}(*)"
+        inner = ", ".join(_render(a) for a in n.args)
+        if n.distinct:
+            inner = f"distinct {inner}"
+        base = f"{n.name}({inner})"
+        if n.over is not None:
+            return base + " over (...)"
+        return base
+    if isinstance(n, A.BinOp):
+        return f"({_render(n.left)} {n.op} {_render(n.right)})"
+    if isinstance(n, A.UnOp):
+        return f"({n.op} {_render(n.operand)})"
+    if isinstance(n, A.Case):
+        return "case when ... end"
+    if isinstance(n, A.CastE):
+        return f"cast({_render(n.operand)} as {n.typename})"
+    if isinstance(n, A.ScalarSubquery):
+        return "scalarsubquery()"
+    if isinstance(n, A.Interval):
+        return f"interval {n.n} day"
+    return type(n).__name__.lower()
 
 
 def _fold_const(e: Expr) -> Expr:

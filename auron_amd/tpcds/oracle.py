@@ -3213,3 +3213,26 @@ def q64(root, sf):
 
 
 ORACLES.update({"q64": q64})
+
+
+# ---------------------------------------------------------------------------
+# Output-name fidelity: the SQL front-end names result columns exactly as the
+# reference query text does (aliases verbatim, Spark-style auto-names for
+# unaliased expressions). The oracle bodies above predate that; this table
+# renames their outputs positionally to the SQL-faithful names.
+ORACLE_OUT_NAMES = {'q2': ['d_week_seq1', 'round((sun_sales1 / sun_sales2), 2)', 'round((mon_sales1 / mon_sales2), 2)', 'round((tue_sales1 / tue_sales2), 2)', 'round((wed_sales1 / wed_sales2), 2)', 'round((thu_sales1 / thu_sales2), 2)', 'round((fri_sales1 / fri_sales2), 2)', 'round((sat_sales1 / sat_sales2), 2)'], 'q3': ['d_year', 'brand_id', 'brand', 'sum_agg'], 'q8': ['s_store_name', 'sum(ss_net_profit)'], 'q13': ['avg(ss_quantity)', 'avg(ss_ext_sales_price)', 'avg(ss_ext_wholesale_cost)', 'sum(ss_ext_wholesale_cost)'], 'q14': ['channel', 'i_brand_id', 'i_class_id', 'i_category_id', 'sum(sales)', 'sum(number_sales)'], 'q15': ['ca_zip', 'sum(cs_sales_price)'], 'q16': ['order count ', 'total shipping cost ', 'total net profit '], 'q17': ['i_item_id', 'i_item_desc', 's_state', 'store_sales_quantitycount', 'store_sales_quantityave', 'store_sales_quantitystdev', 'store_sales_quantitycov', 'as_store_returns_quantitycount', 'as_store_returns_quantityave', 'as_store_returns_quantitystdev', 'store_returns_quantitycov', 'catalog_sales_quantitycount', 'catalog_sales_quantityave', 'catalog_sales_quantitystdev', 'catalog_sales_quantitycov'], 'q23': ['sum(sales)'], 'q32': ['excess discount amount'], 'q35': ['ca_state', 'cd_gender', 'cd_marital_status', 'cnt1', 'min(cd_dep_count)', 'max(cd_dep_count)', 'avg(cd_dep_count)', 'cd_dep_employed_count', 'cnt2', 'min(cd_dep_employed_count)', 'max(cd_dep_employed_count)', 'avg(cd_dep_employed_count)', 'cd_dep_college_count', 'cnt3', 'min(cd_dep_college_count)', 'max(cd_dep_college_count)', 'avg(cd_dep_college_count)'], 'q38': ['count(1)'], 'q39': ['w_warehouse_sk', 'i_item_sk', 'd_moy', 'mean', 'cov', 'w_warehouse_sk__2', 'i_item_sk__2', 'd_moy__2', 'mean__2', 'cov__2'], 'q42': ['d_year', 'i_category_id', 'i_category', 'sum(ss_ext_sales_price)'], 'q45': ['ca_zip', 'ca_city', 'sum(ws_sales_price)'], 'q48': ['sum(ss_quantity)'], 'q50': ['s_store_name', 's_company_id', 's_street_number', 's_street_name', 's_street_type', 's_suite_number', 's_city', 's_county', 's_state', 's_zip', '30 days ', '31 - 60 days ', '61 - 90 days ', '91 - 120 days ', '>120 days '], 'q52': ['d_year', 'brand_id', 'brand', 'ext_price'], 'q53': ['i_manufact_id', 'sum_sales', 'avg_quarterly_sales'], 'q55': ['brand_id', 'brand', 'ext_price'], 'q59': ['s_store_name1', 's_store_id1', 'd_week_seq1', '(sun_sales1 / sun_sales2)', '(mon_sales1 / mon_sales2)', '(tue_sales1 / tue_sales2)', '(wed_sales1 / wed_sales2)', '(thu_sales1 / thu_sales2)', '(fri_sales1 / fri_sales2)', '(sat_sales1 / sat_sales2)'], 'q61': ['promotions', 'total', '((cast(promotions as decimal(15,4)) / cast(total as decimal(15,4))) * 100)'], 'q62': ['substr(w_warehouse_name, 1, 20)', 'sm_type', 'web_name', '30 days ', '31 - 60 days ', '61 - 90 days ', '91 - 120 days ', '>120 days '], 'q63': ['i_manager_id', 'sum_sales', 'avg_monthly_sales'], 'q64': ['product_name', 'store_name', 'store_zip', 'b_street_number', 'b_streen_name', 'b_city', 'b_zip', 'c_street_number', 'c_street_name', 'c_city', 'c_zip', 'syear', 'cnt', 's1', 's2', 's3', 's1__2', 's2__2', 's3__2', 'syear__2', 'cnt__2'], 'q72': ['i_item_desc', 'w_warehouse_name', 'd_week_seq', 'no_promo', 'promo', 'total_cnt'], 'q79': ['c_last_name', 'c_first_name', 'substr(s_city, 1, 30)', 'ss_ticket_number', 'amt', 'profit'], 'q81': ['c_customer_id', 'c_salutation', 'c_first_name', 'c_last_name', 'ca_street_number', 'ca_street_name', 'ca_street_type', 'ca_suite_number', 'ca_city', 'ca_county', 'ca_state', 'ca_zip', 'ca_country', 'ca_gmt_offset', 'ca_location_type', 'ctr_total_return'], 'q85': ['substr(r_reason_desc, 1, 20)', 'avg(ws_quantity)', 'avg(wr_refunded_cash)', 'avg(wr_fee)'], 'q87': ['count(1)'], 'q92': ['excess discount amount '], 'q94': ['order count ', 'total shipping cost ', 'total net profit '], 'q95': ['order count ', 'total shipping cost ', 'total net profit '], 'q96': ['count(1)'], 'q99': ['substr(w_warehouse_name, 1, 20)', 'sm_type', 'cc_name', '30 days ', '31 - 60 days ', '61 - 90 days ', '91 - 120 days ', '>120 days ']}
+
+
+def _with_sql_names(fn, names):
+    def wrapped(root, sf):
+        df = fn(root, sf)
+        assert len(df.columns) == len(names), (fn.__name__, list(df.columns), names)
+        df = df.copy()
+        df.columns = names
+        return df
+    wrapped.__name__ = fn.__name__
+    return wrapped
+
+
+for _qn, _names in ORACLE_OUT_NAMES.items():
+    ORACLES[_qn] = _with_sql_names(ORACLES[_qn], _names)
