@@ -30,6 +30,7 @@ from ..exprs import (AggFunc, Aliased, Arith, BoolOp, CaseWhen, Cast, Cmp,
 from . import nodes as P
 
 _EXPR_MOD = "auron_amd.exprs"
+_FUNC_MOD = "auron_amd.functions"
 _NODE_MOD = "auron_amd.plan.nodes"
 
 
@@ -57,7 +58,7 @@ def _encode(o) -> Any:
 
 
 _REGISTRY = {}
-for mod in (_EXPR_MOD, _NODE_MOD):
+for mod in (_EXPR_MOD, _FUNC_MOD, _NODE_MOD):
     m = importlib.import_module(mod)
     for name in dir(m):
         c = getattr(m, name)

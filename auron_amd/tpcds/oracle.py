@@ -2481,8 +2481,24 @@ ORACLES.update({"q2": q2, "q9": q9, "q10": q10, "q17": q17, "q18": q18,
 
 
 # ------------------------------- batch 10 oracles
+_Q8_ZIPS = ['24128', '76232', '65084', '87816', '83926', '77556', '20548',
+            '26231', '43848', '15126', '91137', '61265', '98294', '25782',
+            '17920', '18426', '98235', '40081', '84093', '28577', '55565',
+            '17183', '54601', '67897', '22752', '86284', '18376', '38607',
+            '45200', '21756', '29741', '96765', '23932', '89360', '29839',
+            '25989', '28898', '91068', '72550', '10390', '18845', '47770',
+            '82636', '41367', '76638', '86198', '81312', '37126', '39192',
+            '88424', '72175', '81426', '53672', '10445', '42666', '66864',
+            '66708', '41248', '48583', '82276', '18842', '78890', '49448',
+            '14089', '38122', '34425', '79077', '19849', '43285', '39861',
+            '66162', '77610', '13695', '99543', '83444', '83041', '12305',
+            '57665', '68341', '25003', '57834', '62878', '49130', '81096',
+            '18840', '27700', '23470', '50412', '21195', '16021', '76107',
+            '71954', '68309', '18119', '98359', '64544', '10336', '86379',
+            '27068', '39736', '98569', '28915', '24206', '56529', '57647']
+
+
 def q8(root, sf):
-    from .queries import _Q8_ZIPS
 
     ca = _read(root, sf, "customer_address", ["ca_address_sk", "ca_zip"])
     cust = _read(root, sf, "customer", ["c_current_addr_sk", "c_preferred_cust_flag"])
