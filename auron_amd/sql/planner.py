@@ -549,6 +549,11 @@ class Planner:
     def plan_from_where(self, sel: A.Select, outer: Optional[Scope]):
         units: List[Rel] = []
         conds: List[A.ANode] = []
+        # trailing outer joins of a left-deep FROM chain: their inner
+        # PREFIX flattens into the join graph (so WHERE equi conjuncts
+        # like q72's d1.d_week_seq = d2.d_week_seq become graph edges and
+        # filters push into scans); each outer join applies afterwards
+        pending_outer: List[tuple] = []
 
         def add_item(item):
             if isinstance(item, A.Join) and item.kind == "inner":
@@ -557,7 +562,9 @@ class Planner:
                 if item.on is not None:
                     conds.extend(_conjuncts(item.on))
             elif isinstance(item, A.Join):
-                units.append(self.plan_outer_join(item, outer))
+                add_item(item.left)
+                rrel = self._join_unit(item.right, outer)
+                pending_outer.append((item.kind, rrel, item.on))
             else:
                 units.append(self.plan_table(item, outer))
 
@@ -568,9 +575,67 @@ class Planner:
             raise SqlError("SELECT without FROM unsupported")
         if sel.where is not None:
             conds.extend(_conjuncts(sel.where))
-        scope = Scope(units, outer)
+        scope = Scope(units + [r for _, r, _ in pending_outer], outer)
+        post_conds: List[A.ANode] = []
+        if pending_outer:
+            outer_cols = {c.engine for _, r, _ in pending_outer for c in r.cols}
+            unsafe_all = any(k in ("full", "right") for k, _, _ in pending_outer)
+            graph_conds = []
+            for cnd in conds:
+                if unsafe_all:
+                    post_conds.append(cnd)
+                    continue
+                if _contains_subquery(cnd) or _contains_scalar_subquery(cnd):
+                    graph_conds.append(cnd)
+                    continue
+                conv = self._try_convert(cnd, scope)
+                if conv is None or (expr_cols(conv) & outer_cols):
+                    # references an outer-join right side: apply after it
+                    post_conds.append(cnd)
+                else:
+                    graph_conds.append(cnd)
+            conds = graph_conds
         rel = self.join_graph(units, conds, scope)
+        for kind, rrel, on in pending_outer:
+            rel = self._apply_outer(rel, kind, rrel, on, outer)
+        for cnd in post_conds:
+            pred = self.to_expr(cnd, Scope([rel], outer), None)
+            rel = Rel(LFilter(rel.node, pred), rel.cols,
+                      rel.est * _selectivity(pred), rel.base_dim_only)
         return rel, Scope([rel], outer)
+
+    def _try_convert(self, cnd, scope) -> Optional[Expr]:
+        try:
+            return self.to_expr(cnd, scope, None)
+        except SqlError:
+            return None
+
+    def _apply_outer(self, lrel: Rel, kind: str, rrel: Rel, on,
+                     outer: Optional[Scope]) -> Rel:
+        scope = Scope([lrel, rrel], outer)
+        lkeys, rkeys, residual, rfilters = [], [], None, []
+        lcols = {rc.engine for rc in lrel.cols}
+        rcols = {rc.engine for rc in rrel.cols}
+        for c in _conjuncts(on) if on is not None else []:
+            pair = _equi_pair(c, scope, self, lcols, rcols)
+            if pair is not None:
+                lkeys.append(pair[0])
+                rkeys.append(pair[1])
+                continue
+            conv = self.to_expr(c, scope, None)
+            refs = expr_cols(conv)
+            if refs <= rcols and kind == "left":
+                rfilters.append(conv)
+            else:
+                residual = conv if residual is None else (residual & conv)
+        rnode = rrel.node
+        for f in rfilters:
+            rnode = LFilter(rnode, f)
+        node = LJoin(lrel.node, rnode, kind, lkeys, rkeys, residual,
+                     l_est=lrel.est, r_est=rrel.est,
+                     r_base_dim=rrel.base_dim_only)
+        return Rel(node, lrel.cols + rrel.cols, max(lrel.est, rrel.est),
+                   lrel.base_dim_only and rrel.base_dim_only)
 
     def plan_table(self, item: A.ANode, outer: Optional[Scope]) -> Rel:
         if isinstance(item, A.Table):
@@ -798,27 +863,26 @@ class Planner:
 
         while len({root(i) for i in alive}) > 1:
             roots = sorted({root(i) for i in alive})
-            # pick the edge whose two sides are distinct roots; prefer the
-            # pair (big, small): start from the biggest component
-            best = None
-            big = max(roots, key=lambda r: units[r].est)
+            # choose the globally cheapest join pair: attaching a dimension
+            # barely grows the fact side, and a fact-fact join gets cheaper
+            # with every extra equi edge (q72: cs><inv must wait until the
+            # week-linking date dims are attached so BOTH keys apply)
+            pair_edges: Dict[Tuple[int, int], int] = {}
             for li, le, ri, re_ in edges:
                 lr, rr = root(li), root(ri)
                 if lr == rr:
                     continue
-                if lr != big and rr != big:
-                    continue
-                other = rr if lr == big else lr
-                score = units[other].est
+                key = (min(lr, rr), max(lr, rr))
+                pair_edges[key] = pair_edges.get(key, 0) + 1
+            best = None
+            for (lr, rr), n_edges in pair_edges.items():
+                a, b = units[lr], units[rr]
+                if a.base_dim_only or b.base_dim_only:
+                    score = max(a.est, b.est) * 0.8
+                else:
+                    score = max(a.est, b.est) * (2.0 / (1.0 + n_edges))
                 if best is None or score < best[0]:
                     best = (score, lr, rr)
-            if best is None:
-                # try any edge between two roots (disconnected from big)
-                for li, le, ri, re_ in edges:
-                    lr, rr = root(li), root(ri)
-                    if lr != rr:
-                        best = (units[rr].est, lr, rr)
-                        break
             if best is None:
                 # cartesian: join the two smallest on a constant key
                 small = sorted(roots, key=lambda r: units[r].est)
