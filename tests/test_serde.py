@@ -78,3 +78,50 @@ def test_all_99_plans_roundtrip():
         blob = serialize_task("t", 0, 0, plan)
         _, _, _, back = deserialize_task(blob)
         assert _shape(back) == _shape(plan), qn
+
+
+def test_all_99_plans_protobuf_roundtrip():
+    """The protobuf TaskDefinition contract (plan/auron.proto) carries
+    every TPC-DS plan: encode -> standard-wire decode -> same structure."""
+    import os
+
+    from auron_amd import AuronSession
+    from auron_amd.plan.proto import deserialize_task_pb, serialize_task_pb
+    from auron_amd.tpcds import datagen
+    from auron_amd.tpcds.queries import QUERIES, Catalog
+    from tests.test_plan_stability import _shape
+
+    root = os.path.join(os.path.dirname(__file__), "..", ".tpcds_cache")
+    datagen.write_dataset(root, 0.01)
+    s = AuronSession()
+    cat = Catalog(root, 0.01)
+    for qn in sorted(QUERIES):
+        plan = QUERIES[qn](cat, s)
+        blob = serialize_task_pb("t", 1, 2, plan)
+        tid, stage, part, back = deserialize_task_pb(blob)
+        assert (tid, stage, part) == ("t", 1, 2)
+        assert _shape(back) == _shape(plan), qn
+
+
+def test_protobuf_wire_vectors():
+    """Byte-level wire-format vectors (varint, tags, nested messages) so
+    the pure-python codec stays interchangeable with protoc bindings."""
+    from auron_amd.plan.pbwire import (decode_fields, write_int, write_len,
+                                       write_str)
+
+    out = bytearray()
+    write_int(out, 1, 150)
+    assert bytes(out) == b"\x08\x96\x01"  # canonical protobuf example
+    out = bytearray()
+    write_str(out, 2, "testing")
+    assert bytes(out) == b"\x12\x07testing"
+    out = bytearray()
+    write_int(out, 1, -2)  # negative int64: 10-byte twos complement
+    f = decode_fields(bytes(out))
+    assert f[1][-1] == -2
+    inner = bytearray()
+    write_int(inner, 1, 1)
+    out = bytearray()
+    write_len(out, 3, bytes(inner))
+    f = decode_fields(bytes(out))
+    assert decode_fields(f[3][-1])[1][-1] == 1
