@@ -802,9 +802,111 @@ class Executor:
     def _exec_HashAgg(self, node: P.HashAgg) -> List[RecordBatch]:
         from ..exprs import eval_scope
 
+        if node.mode == "partial":
+            # streaming partial agg (agg_table.rs analogue): inputs are
+            # consumed chunk-by-chunk, states re-merged when they shrink,
+            # and the accumulated state registers with the memmgr so it
+            # can spill under pressure — the whole input never has to be
+            # resident at once
+            return self._exec_hash_agg_partial_chunked(node)
         b = _concat(self.execute(node.child))
         with eval_scope(b):
             return self._hash_agg_body(node, b)
+
+    def _exec_hash_agg_partial_chunked(self, node: P.HashAgg) -> List[RecordBatch]:
+        from ..exprs import eval_scope
+
+        batches = self.execute(node.child)
+        limit = self.ctx.batch_rows
+        acc: List[RecordBatch] = []
+        acc_rows = 0
+        merge_worthwhile = True
+        holder = self.ctx.memmgr.register("agg-partial-state", acc)
+
+        def flush_merge():
+            nonlocal acc, acc_rows, merge_worthwhile, holder
+            cur = holder.batches()
+            merged = self._merge_states(node, _concat(cur)) if cur else None
+            holder.release()
+            if merged is not None:
+                if merged.num_rows > 0.8 * acc_rows:
+                    # low-reduction keys: stop paying for re-merges; the
+                    # post-exchange final agg regroups anyway
+                    merge_worthwhile = False
+                acc = [merged]
+                acc_rows = merged.num_rows
+            else:
+                acc = []
+                acc_rows = 0
+            holder = self.ctx.memmgr.register("agg-partial-state", acc)
+
+        for b in batches:
+            for lo in range(0, max(b.num_rows, 1), limit):
+                chunk = b if b.num_rows <= limit else \
+                    b.slice(lo, min(limit, b.num_rows - lo))
+                with eval_scope(chunk):
+                    out = self._hash_agg_body(node, chunk)
+                cur = holder.batches()
+                cur.extend(out)
+                acc_rows += sum(x.num_rows for x in out)
+                holder.refresh()
+                if b.num_rows <= limit:
+                    break
+            if merge_worthwhile and acc_rows > 2 * limit:
+                flush_merge()
+        result = holder.batches()
+        holder.release()
+        if not result:
+            # preserve the empty-input partial schema
+            with eval_scope(_concat(batches)):
+                return self._hash_agg_body(node, _concat(batches))
+        return result
+
+    def _merge_states(self, node: P.HashAgg, b: RecordBatch) -> RecordBatch:
+        """Combine partial-state rows with equal keys into one state row
+        (state schema in, state schema out — the pre-exchange self-merge
+        of the reference's in-memory agg table)."""
+        device = b.device
+        key_cols = [Col(a.name).eval(b) for a in node.keys]
+        if key_cols:
+            gids, reps = ops.group_ids(key_cols)
+            ngroups = int(reps.numel())
+            out_keys = [c.gather(reps) for c in key_cols]
+        else:
+            gids = torch.zeros(b.num_rows, dtype=torch.int64, device=device)
+            ngroups = 1 if b.num_rows else 0
+            out_keys = []
+        names = [a.name for a in node.keys]
+        cols = list(out_keys)
+        for i, agg in enumerate(node.aggs):
+            s0 = b.column(f"__agg{i}_0")
+            s1 = b.column(f"__agg{i}_1")
+            merged_cnt, _ = ops.agg_scatter(gids, ngroups, s1, "sum")
+            if agg.fn in ("count", "count_star"):
+                cols.append(Column(dtypes.int64, merged_cnt))
+                names.append(f"__agg{i}_0")
+                cols.append(Column(dtypes.int64, merged_cnt))
+                names.append(f"__agg{i}_1")
+                continue
+            comb = {"sum": "sum", "avg": "sum", "min": "min", "max": "max",
+                    "first": "first", "first_ignores_null": "first"}[agg.fn]
+            if comb == "first":
+                acc_col, _ = self._agg_first(gids, ngroups, s0)
+                cols.append(acc_col)
+            elif comb == "sum" and self._decimal_sum_unsafe(s0, b.num_rows):
+                data, cnt = self._sum_split_exact(gids, ngroups, s0)
+                cols.append(Column(s0.dtype, data, cnt > 0))
+            else:
+                data, cnt = ops.agg_scatter(gids, ngroups, s0, comb)
+                if isinstance(data, Column):
+                    cols.append(data)
+                else:
+                    cols.append(Column(s0.dtype, data, (cnt > 0)
+                                       if not bool((cnt > 0).all()) else None))
+            names.append(f"__agg{i}_0")
+            cols.append(Column(dtypes.int64, merged_cnt))
+            names.append(f"__agg{i}_1")
+        return RecordBatch(names, cols)
 
     def _hash_agg_body(self, node: P.HashAgg, b: RecordBatch) -> List[RecordBatch]:
         device = b.device

@@ -54,13 +54,24 @@ class BatchHolder:
     def host_resident(self) -> bool:
         return self._spilled_host is not None
 
+    def refresh(self):
+        """Re-account after the caller mutated the resident batch list
+        (update_mem_used analogue); may trigger spills elsewhere."""
+        if self._batches is not None:
+            self.bytes = sum(_batch_bytes(b) for b in self._batches)
+            self.last_touch = time.monotonic()
+            self.mgr.reserve(0)
+
     def spill(self) -> int:
         """Move payload one tier down. Returns bytes released."""
         if self._batches is None:
             return 0
         if self.device.type == "cuda":
-            # tier 1: device -> host (pageable-safe copies; see Column.to)
-            self._spilled_host = [b.to("cpu") for b in self._batches]
+            # tier 1: device -> host via the pinned bounce buffer on the
+            # dedicated copy stream (SURVEY §3.5 hipMemcpyAsync mapping)
+            from .pinned import batch_to_host
+
+            self._spilled_host = [batch_to_host(b) for b in self._batches]
             self.mgr.metrics["spill_d2h_bytes"] = self.mgr.metrics.get("spill_d2h_bytes", 0) + self.bytes
         else:
             # CPU mode: straight to disk so CI exercises the file tier
@@ -144,7 +155,10 @@ class MemManager:
     def __init__(self, budget_bytes: Optional[int] = None, fraction: float = 0.8,
                  spill_dir: Optional[str] = None, host_budget_bytes: Optional[int] = None):
         if budget_bytes is None:
-            if torch.cuda.is_available():
+            env = os.environ.get("AURON_MEM_BUDGET")
+            if env is not None:
+                budget_bytes = int(env)
+            elif torch.cuda.is_available():
                 free, total = torch.cuda.mem_get_info()
                 budget_bytes = int(total * fraction)
             else:

@@ -50,6 +50,59 @@ class PinnedPool:
 POOL = PinnedPool()
 
 
+_copy_stream = None
+_SPILL_BOUNCE = 128 << 20
+
+
+def copy_stream():
+    """Dedicated D2H copy stream (spill = hipMemcpyAsync on a
+    side stream, never the compute stream)."""
+    global _copy_stream
+    if _copy_stream is None:
+        _copy_stream = torch.cuda.Stream()
+    return _copy_stream
+
+
+def d2h_staged(t: torch.Tensor) -> torch.Tensor:
+    """Device -> pageable host through a pooled pinned bounce buffer on
+    the copy stream. Pageable d2h copies in HIP stage through a tiny
+    internal buffer; an explicit 128 MB pinned bounce keeps the transfer
+    at link speed without pinning the full payload."""
+    if t.device.type != "cuda":
+        return t
+    src = t.contiguous()
+    nbytes = src.numel() * src.element_size()
+    if nbytes == 0:
+        return torch.empty(t.shape, dtype=t.dtype)
+    flat = src.reshape(-1).view(torch.uint8)
+    dst = torch.empty(nbytes, dtype=torch.uint8)
+    s = copy_stream()
+    base, bounce = POOL.acquire(min(nbytes, _SPILL_BOUNCE))
+    cap = bounce.numel()
+    s.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(s):
+        for off in range(0, nbytes, cap):
+            n = min(cap, nbytes - off)
+            bounce[:n].copy_(flat[off:off + n], non_blocking=True)
+            s.synchronize()
+            dst[off:off + n].copy_(bounce[:n])
+    POOL.release(base, t.device)
+    return dst.view(t.dtype).reshape(t.shape)
+
+
+def batch_to_host(batch) -> "object":
+    """RecordBatch device->host via the staged path (spill tier 1)."""
+    from .column import Column, RecordBatch
+
+    cols = []
+    for c in batch.columns:
+        cols.append(Column(
+            c.dtype, d2h_staged(c.data),
+            None if c.validity is None else d2h_staged(c.validity),
+            None if c.offsets is None else d2h_staged(c.offsets)))
+    return RecordBatch(batch.names, cols)
+
+
 def to_device(arr: np.ndarray, device) -> torch.Tensor:
     """Upload a host array; pooled pinned slab for payloads big enough to
     beat torch's own small-copy staging (tiny descriptor arrays go the

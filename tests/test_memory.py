@@ -83,3 +83,44 @@ def test_fair_share_spills_largest_over_share_first():
     mgr.register("new", mk(10))
     assert not big.resident, "fair-share should spill the oversized holder"
     assert small1.resident and small2.resident
+
+
+def test_chunked_partial_agg_spills_and_matches():
+    """Streaming partial agg: chunk-wise consumption with a tiny memory
+    budget must spill its state (spill_count > 0) and still produce the
+    same final aggregate as the monolithic path."""
+    import torch
+
+    from auron_amd import AggFunc, col, dtypes
+    from auron_amd.column import Column, RecordBatch
+    from auron_amd.engine.executor import ExecContext, Executor
+    from auron_amd.exprs import Aliased
+    from auron_amd.memory import MemManager
+    from auron_amd.plan import nodes as P
+
+    torch.manual_seed(7)
+    batches = []
+    for i in range(6):
+        n = 5000
+        k = torch.randint(0, 400, (n,))
+        v = torch.rand(n, dtype=torch.float64)
+        batches.append(RecordBatch(["k", "v"], [
+            Column(dtypes.int64, k), Column(dtypes.float64, v)]))
+    plan = P.HashAgg(
+        P.HashAgg(P.MemoryScan(batches),
+                  [Aliased(col("k"), "k")],
+                  [AggFunc("sum", col("v"), name="s")], mode="partial"),
+        [Aliased(col("k"), "k")],
+        [AggFunc("sum", col("v"), name="s")], mode="final")
+
+    ctx = ExecContext(memmgr=MemManager(budget_bytes=64 << 10), batch_rows=2000)
+    out = Executor(ctx).collect(plan)
+    assert ctx.memmgr.metrics.get("spill_count", 0) > 0, ctx.memmgr.metrics
+
+    ctx2 = ExecContext(memmgr=MemManager(budget_bytes=16 << 30))
+    ref = Executor(ctx2).collect(plan)
+    got = dict(zip(out.to_pydict()["k"], out.to_pydict()["s"]))
+    want = dict(zip(ref.to_pydict()["k"], ref.to_pydict()["s"]))
+    assert set(got) == set(want)
+    for k in want:
+        assert abs(got[k] - want[k]) < 1e-9 * max(1.0, abs(want[k]))
