@@ -802,7 +802,7 @@ class Executor:
     def _exec_HashAgg(self, node: P.HashAgg) -> List[RecordBatch]:
         from ..exprs import eval_scope
 
-        if node.mode == "partial":
+        if node.mode == "partial" and os.environ.get("AURON_AGG_STREAMING", "1") != "0":
             # streaming partial agg (agg_table.rs analogue): inputs are
             # consumed chunk-by-chunk, states re-merged when they shrink,
             # and the accumulated state registers with the memmgr so it
@@ -1197,9 +1197,20 @@ class Executor:
             return Column(acc.dtype, acc.data, v, acc.offsets)
         if agg.fn == "avg":
             if vdt.code == dtypes.DECIMAL64:
-                data = (acc.to(torch.float64) / 10 ** vdt.scale) / cnt.clamp(min=1).to(torch.float64)
-            else:
-                data = acc.to(torch.float64) / cnt.clamp(min=1).to(torch.float64)
+                # Spark: avg(decimal(p,s)) -> decimal(p+4, s+4), HALF_UP;
+                # exact integer long division (advisor finding r1)
+                cnt_ = cnt.clamp(min=1)
+                sign = torch.sign(acc)
+                a = acc.abs()
+                q = torch.div(a, cnt_, rounding_mode="floor")
+                r = a - q * cnt_
+                frac = torch.div(r * 20000 + cnt_, cnt_ * 2,
+                                 rounding_mode="floor")
+                data = sign * (q * 10000 + frac)
+                out_dt = dtypes.decimal64(min(vdt.precision + 4, 18),
+                                          vdt.scale + 4)
+                return Column(out_dt, data, validity)
+            data = acc.to(torch.float64) / cnt.clamp(min=1).to(torch.float64)
             return Column(dtypes.float64, data, validity)
         out_dt = self._state_dtype(agg, vdt)
         return Column(out_dt, acc, validity)

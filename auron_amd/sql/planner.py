@@ -909,8 +909,10 @@ class Planner:
 
     def _merge(self, units, merged, lr, rr, lkeys, rkeys, edges):
         lu, ru = units[lr], units[rr]
-        # keep the big side as the probe (left); build = right
-        if ru.est > lu.est and not lu.base_dim_only:
+        # keep the big side as the probe (left); build = right — always
+        # (a constant-key cross join with the big side as build degrades
+        # to one multi-million-entry hash chain, q93)
+        if ru.est > lu.est:
             lu, ru = ru, lu
             lkeys, rkeys = rkeys, lkeys
             lr, rr = rr, lr

@@ -111,6 +111,15 @@ def _id_str(prefix, arr):
     return [f"{prefix}{int(v):012d}" for v in arr]
 
 
+
+def _row_item(j, n_item):
+    """Stateless pseudo-random item for fact row j: returns tables can
+    reconstruct the (line, item) pair of the sale they reference the way
+    dsdgen's returns reference actual sales lines."""
+    h = (j.astype(np.uint64) * np.uint64(0x9E3779B97F4A7C15)) >> np.uint64(17)
+    return (h % np.uint64(max(n_item, 1))).astype(np.int64) + 1
+
+
 def _calendar():
     n = BASE_ROWS["date_dim"]
     idx = np.arange(n, dtype=np.int64)
@@ -349,7 +358,7 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
         put("ss_sold_date_sk", d, dv)
         t, tv = fk(86400, 0.02)
         put("ss_sold_time_sk", t, tv)
-        put("ss_item_sk", rng.integers(1, n_item + 1, n))
+        put("ss_item_sk", _row_item(lo + np.arange(n, dtype=np.int64), n_item))
         c, cv = fk(n_cust)
         put("ss_customer_sk", c, cv)
         cd, cdv = fk(n_cdemo)
@@ -362,7 +371,7 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
         put("ss_store_sk", st, stv)
         p, pv = fk(n_promo, 0.02)
         put("ss_promo_sk", p, pv)
-        put("ss_ticket_number", lo + np.arange(n, dtype=np.int64) // 3 + 1)
+        put("ss_ticket_number", (lo + np.arange(n, dtype=np.int64)) // 3 + 1)
         qty = rng.integers(1, 101, n).astype(np.int32)
         put("ss_quantity", qty, _with_nulls(rng, qty, 0.02)[1])
         whole = _money(rng, n, 1, 100)
@@ -390,7 +399,8 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
         put("sr_returned_date_sk", d, dv)
         t, tv = fk(86400, 0.02)
         put("sr_return_time_sk", t, tv)
-        put("sr_item_sk", rng.integers(1, n_item + 1, n))
+        src_rows = rng.integers(0, max(row_count("store_sales", sf), 1), n)
+        put("sr_item_sk", _row_item(src_rows, n_item))
         c, cv = fk(n_cust)
         put("sr_customer_sk", c, cv)
         cd, cdv = fk(n_cdemo)
@@ -403,7 +413,7 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
         put("sr_store_sk", st, stv)
         r, rv = fk(BASE_ROWS["reason"], 0.02)
         put("sr_reason_sk", r, rv)
-        put("sr_ticket_number", rng.integers(1, max(row_count("store_sales", sf) // 3, 2), n))
+        put("sr_ticket_number", src_rows // 3 + 1)
         q = rng.integers(1, 50, n).astype(np.int32)
         put("sr_return_quantity", q, _with_nulls(rng, q, 0.02)[1])
         amt = _money(rng, n, 1, 2000)
@@ -424,7 +434,7 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
         put(f"{pre}_sold_time_sk", t, tv)
         sd, sdv = date_fk()
         put(f"{pre}_ship_date_sk", sd, sdv)
-        put(f"{pre}_item_sk", rng.integers(1, n_item + 1, n))
+        put(f"{pre}_item_sk", _row_item(lo + np.arange(n, dtype=np.int64), n_item))
         c, cv = fk(n_cust)
         put(f"{pre}_bill_customer_sk", c, cv)
         cd, cdv = fk(n_cdemo)
@@ -455,7 +465,7 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
         put(f"{pre}_warehouse_sk", w, wv)
         pp, ppv = fk(n_promo, 0.02)
         put(f"{pre}_promo_sk", pp, ppv)
-        put(f"{pre}_order_number", lo + np.arange(n, dtype=np.int64) // 4 + 1)
+        put(f"{pre}_order_number", (lo + np.arange(n, dtype=np.int64)) // 4 + 1)
         qty = rng.integers(1, 101, n).astype(np.int32)
         put(f"{pre}_quantity", qty, _with_nulls(rng, qty, 0.02)[1])
         whole = _money(rng, n, 1, 100)
@@ -486,9 +496,10 @@ def generate_table(name: str, sf: float, part: int = 0, nparts: int = 1):
         pre = "cr" if name == "catalog_returns" else "wr"
         d, dv = date_fk()
         put(f"{pre}_returned_date_sk", d, dv)
-        put(f"{pre}_item_sk", rng.integers(1, n_item + 1, n))
         src_fact = "catalog_sales" if pre == "cr" else "web_sales"
-        put(f"{pre}_order_number", rng.integers(1, max(row_count(src_fact, sf) // 4, 2), n))
+        src_rows = rng.integers(0, max(row_count(src_fact, sf), 1), n)
+        put(f"{pre}_item_sk", _row_item(src_rows, n_item))
+        put(f"{pre}_order_number", src_rows // 4 + 1)
         c, cv = fk(n_cust)
         put(f"{pre}_returning_customer_sk", c, cv)
         ra, rav = fk(n_addr)
@@ -591,7 +602,7 @@ def _nparts_for(table: str, sf: float) -> int:
     return 1
 
 
-DATAGEN_VERSION = 13
+DATAGEN_VERSION = 14
 
 
 def dataset_root(root: str, sf: float) -> str:
@@ -610,10 +621,9 @@ def write_dataset(root: str, sf: float, tables: Optional[List[str]] = None,
 
     Work is sharded across ranks by (table, part) index; call from every
     rank, then barrier before reading."""
-    import pyarrow.parquet as pq
-
     base = dataset_root(root, sf)
     tables = tables or list(SCHEMAS.keys())
+    todo = []
     job = 0
     for t in tables:
         nparts = _nparts_for(t, sf)
@@ -621,22 +631,39 @@ def write_dataset(root: str, sf: float, tables: Optional[List[str]] = None,
         os.makedirs(tdir, exist_ok=True)
         for p in range(nparts):
             path = os.path.join(tdir, f"part-{p:04d}.parquet")
-            if job % world == rank:
-                if force or not os.path.exists(path):
-                    tbl = generate_table(t, sf, p, nparts)
-                    # uncompressed pages: the GPU parquet decoder
-                    # (parquet_native.py + csrc/parquet.hip) consumes page
-                    # bytes directly in HBM — numerics as PLAIN, strings as
-                    # RLE_DICTIONARY (indices decoded + bytes gathered on
-                    # device); no host decode at all
-                    str_cols = [f.name for f in tbl.schema
-                                if f.type == __import__("pyarrow").string()]
-                    pq.write_table(tbl, path, compression="NONE",
-                                   use_dictionary=str_cols,
-                                   data_page_version="1.0",
-                                   store_decimal_as_integer=True,
-                                   dictionary_pagesize_limit=1 << 26,
-                                   data_page_size=64 << 10,
-                                   row_group_size=1 << 20)
+            if job % world == rank and (force or not os.path.exists(path)):
+                todo.append((t, sf, p, nparts, path))
             job += 1
+    workers = int(os.environ.get("AURON_DATAGEN_WORKERS",
+                                 str(max(1, (os.cpu_count() or 4) - 2))))
+    if len(todo) > 3 and workers > 1:
+        # dsdgen-style parallel generation: (table, part) jobs fan out
+        # across processes (numpy-only work, no device state involved)
+        from concurrent.futures import ProcessPoolExecutor
+
+        with ProcessPoolExecutor(max_workers=min(workers, len(todo))) as ex:
+            list(ex.map(_write_one, todo, chunksize=1))
+    else:
+        for t in todo:
+            _write_one(t)
     return base
+
+
+def _write_one(args):
+    import pyarrow.parquet as pq
+
+    t, sf, p, nparts, path = args
+    tbl = generate_table(t, sf, p, nparts)
+    # uncompressed pages: the GPU parquet decoder (parquet_native.py +
+    # csrc/parquet.hip) consumes page bytes directly in HBM — numerics as
+    # PLAIN, strings as RLE_DICTIONARY (indices decoded + bytes gathered
+    # on device); no host decode at all
+    str_cols = [f.name for f in tbl.schema
+                if f.type == __import__("pyarrow").string()]
+    pq.write_table(tbl, path, compression="NONE",
+                   use_dictionary=str_cols,
+                   data_page_version="1.0",
+                   store_decimal_as_integer=True,
+                   dictionary_pagesize_limit=1 << 26,
+                   data_page_size=64 << 10,
+                   row_group_size=1 << 20)
