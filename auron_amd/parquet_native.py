@@ -171,13 +171,46 @@ class ChunkPages:
         return self.dict_off >= 0
 
 
+class _ExtraAlloc:
+    """Cursor over the decompressed-page region appended after the raw
+    staging buffer, plus the job list that refills it on every read
+    (host page decompression -> device decode; parquet_exec.rs analogue
+    of arrow-rs's page decompression)."""
+
+    def __init__(self, base: int):
+        self.base = base
+        self.cursor = base
+        self.jobs: List[tuple] = []  # ("d", src, clen, dest, ulen, codec) | ("c", src, n, dest)
+
+    def decompress(self, src: int, clen: int, ulen: int, codec: str) -> int:
+        dest = self.cursor
+        self.jobs.append(("d", src, clen, dest, ulen, codec))
+        self.cursor += ulen
+        return dest
+
+    def copy(self, src: int, n: int) -> int:
+        dest = self.cursor
+        self.jobs.append(("c", src, n, dest))
+        self.cursor += n
+        return dest
+
+
+def _codec_decompress(codec: str, raw: bytes, ulen: int) -> bytes:
+    import pyarrow as pa
+
+    return pa.Codec(codec.lower()).decompress(raw, decompressed_size=ulen) \
+        .to_pybytes()
+
+
 def parse_pages(buf: np.ndarray, chunk_off: int, chunk_len: int,
-                num_values: int, has_def: bool) -> Optional[ChunkPages]:
-    """Parse v1 page headers of one uncompressed column chunk.
+                num_values: int, has_def: bool, codec: Optional[str] = None,
+                extra: Optional[_ExtraAlloc] = None) -> Optional[ChunkPages]:
+    """Parse page headers of one column chunk (v1 + v2 data pages).
 
     Supports PLAIN data pages and RLE_DICTIONARY/PLAIN_DICTIONARY data
-    pages with a PLAIN dictionary page. Chunks mixing dictionary and
-    plain data pages (dict-overflow spill) fall back to the host path."""
+    pages with a PLAIN dictionary page; snappy/zstd/gzip page payloads
+    are routed through the extra-region decompress jobs. Chunks mixing
+    dictionary and plain data pages fall back to the host path."""
     mv = memoryview(buf)
     pos = chunk_off
     end = chunk_off + chunk_len
@@ -187,9 +220,11 @@ def parse_pages(buf: np.ndarray, chunk_off: int, chunk_len: int,
         hdr, pos2 = _read_struct_fields(mv, pos, {
             1: "type", 2: "uncompressed_page_size", 3: "compressed_page_size",
             5: "data_page_header", 7: "dictionary_page_header",
+            8: "data_page_header_v2",
         })
         page_data = pos2
         page_len = hdr.get("compressed_page_size", 0)
+        ulen = hdr.get("uncompressed_page_size", page_len)
         ptype = hdr.get("type", -1)
         if ptype == 2:  # dictionary page (PLAIN values)
             dph_range = hdr.get("dictionary_page_header")
@@ -199,12 +234,52 @@ def parse_pages(buf: np.ndarray, chunk_off: int, chunk_len: int,
                 1: "num_values", 2: "encoding"})
             if dph.get("encoding", 0) not in (0, 2):  # PLAIN / PLAIN_DICTIONARY
                 return None
-            ck.dict_off = page_data
-            ck.dict_len = page_len
+            if codec:
+                ck.dict_off = extra.decompress(page_data, page_len, ulen, codec)
+                ck.dict_len = ulen
+            else:
+                ck.dict_off = page_data
+                ck.dict_len = page_len
             ck.dict_nvals = dph["num_values"]
             pos = page_data + page_len
             continue
-        if ptype != 0:  # v2 or index page
+        if ptype == 3:  # DATA_PAGE_V2
+            dph_range = hdr.get("data_page_header_v2")
+            if dph_range is None:
+                return None
+            dph, _ = _read_struct_fields(mv, dph_range[0], {
+                1: "num_values", 4: "encoding", 5: "def_len", 6: "rep_len",
+                7: "is_compressed"})
+            enc = dph.get("encoding", 0)
+            if enc in (2, 8):
+                enc = 8
+            elif enc != 0:
+                return None
+            nv = dph["num_values"]
+            dlen = dph.get("def_len", 0) if has_def else 0
+            rlen = dph.get("rep_len", 0)
+            if rlen:
+                return None  # repeated columns stay on the host path
+            vals_raw = page_data + dlen
+            vals_clen = page_len - dlen
+            vals_ulen = ulen - dlen
+            compressed = bool(dph.get("is_compressed", True)) and bool(codec)
+            if compressed:
+                # v2 keeps levels uncompressed; only values are coded
+                def_off = extra.copy(page_data, dlen) if dlen else -1
+                values_off = extra.decompress(vals_raw, vals_clen, vals_ulen,
+                                              codec)
+                values_len = vals_ulen
+            else:
+                def_off = page_data if dlen else -1
+                values_off = vals_raw
+                values_len = vals_clen
+            ck.pages.append(PageDesc(nv, def_off, dlen, values_off, values_len,
+                                     row, enc))
+            row += nv
+            pos = page_data + page_len
+            continue
+        if ptype != 0:  # index page etc.
             return None
         dph_range = hdr.get("data_page_header")
         if dph_range is None:
@@ -219,7 +294,18 @@ def parse_pages(buf: np.ndarray, chunk_off: int, chunk_len: int,
         elif enc != 0:
             return None
         nv = dph["num_values"]
-        if has_def:
+        if codec:
+            # v1 compresses the whole payload (def prefix included); the
+            # def length lives inside, so peek via one parse-time pass
+            payload = _codec_decompress(codec, bytes(mv[page_data:page_data + page_len]), ulen)
+            dest = extra.decompress(page_data, page_len, ulen, codec)
+            if has_def:
+                dlen = int.from_bytes(payload[:4], "little")
+                ck.pages.append(PageDesc(nv, dest + 4, dlen, dest + 4 + dlen,
+                                         ulen - 4 - dlen, row, enc))
+            else:
+                ck.pages.append(PageDesc(nv, -1, 0, dest, ulen, row, enc))
+        elif has_def:
             dlen = int.from_bytes(bytes(mv[page_data:page_data + 4]), "little")
             def_off = page_data + 4
             values_off = def_off + dlen
@@ -440,7 +526,7 @@ class NativeParquetFile:
                 return False
         for rg in range(self.md.num_row_groups):
             cm = self.md.row_group(rg).column(i)
-            if cm.compression != "UNCOMPRESSED":
+            if cm.compression not in ("UNCOMPRESSED", "SNAPPY", "ZSTD", "GZIP"):
                 return False
         return True
 
@@ -450,7 +536,9 @@ def _chunk_meta(md, rg: int, ci: int):
     off = cm.data_page_offset
     if cm.dictionary_page_offset is not None:
         off = min(off, cm.dictionary_page_offset)
-    return off, cm.total_compressed_size, cm.num_values, cm.physical_type, cm.statistics
+    codec = None if cm.compression == "UNCOMPRESSED" else cm.compression
+    return (off, cm.total_compressed_size, cm.num_values, cm.physical_type,
+            cm.statistics, codec)
 
 
 @dataclass
@@ -469,6 +557,8 @@ class _FileMeta:
     ranges: list  # [(src_off, clen, new_off)]
     cols: list  # [_ColMeta]
     parsed: bool = False
+    extra_total: int = 0  # decompressed-page region size (after raw+pad)
+    jobs: list = None  # decompress/copy jobs refilled per read
 
 
 def _build_meta(path: str, columns: List[str]) -> Optional[_FileMeta]:
@@ -504,10 +594,10 @@ def _build_meta(path: str, columns: List[str]) -> Optional[_FileMeta]:
             dt = dtypes.float64
         cm = _ColMeta(cname, dt, phys, sc.max_definition_level == 1, [], None)
         for rg in range(md.num_row_groups):
-            off, clen, nvals, _phys, stats = _chunk_meta(md, rg, ci)
+            off, clen, nvals, _phys, stats, codec = _chunk_meta(md, rg, ci)
             any_nulls = stats is None or not stats.has_null_count or stats.null_count > 0
             ranges.append((off, clen, pos))
-            cm.chunks.append((pos, clen, nvals, any_nulls))
+            cm.chunks.append((pos, clen, nvals, any_nulls, codec))
             pos += clen
         cols.append(cm)
     return _FileMeta(pos, ranges, cols)
@@ -555,36 +645,70 @@ def read_columns_native(path: str, columns: List[str], device,
     use_gpu = (not _np_only) and torch.device(device).type == "cuda"
 
     mm = np.memmap(path, dtype=np.uint8, mode="r")
-    # +8 pad: the dict-index kernel's tail does unaligned 8-byte loads
-    if use_gpu:
-        pin_base, buf_t = _PINNED.acquire(meta.total + 8)
-    else:
-        pin_base = None
-        buf_t = torch.empty(meta.total + 8, dtype=torch.uint8)
+
+    def alloc(nbytes):
+        if use_gpu:
+            return _PINNED.acquire(nbytes)
+        return None, torch.empty(nbytes, dtype=torch.uint8)
+
+    def run_jobs(buf, jobs):
+        for job in jobs:
+            if job[0] == "d":
+                _, s0, clen, dest, ulen, codec = job
+                buf[dest:dest + ulen] = np.frombuffer(
+                    _codec_decompress(codec, bytes(buf[s0:s0 + clen]), ulen),
+                    dtype=np.uint8)
+            else:
+                _, s0, n, dest = job
+                buf[dest:dest + n] = buf[s0:s0 + n]
+
+    # +8 pad: the dict-index kernel's tail does unaligned 8-byte loads;
+    # decompressed pages land after the pad (+8 tail pad again)
+    full = meta.total + 8 + meta.extra_total + (8 if meta.extra_total else 0)
+    pin_base, buf_t = alloc(full)
     try:
         buf = buf_t.numpy()
         for (src, clen, dst) in meta.ranges:
             buf[dst:dst + clen] = mm[src:src + clen]
 
         if not meta.parsed:
+            extra = _ExtraAlloc(meta.total + 8)
             ok = []
+            jobs = []
             for cm in meta.cols:
                 cm.pages = []
                 good = True
-                for (new_off, clen, nvals, _an) in cm.chunks:
-                    ck = parse_pages(buf, new_off, clen, nvals, cm.has_def)
+                j0 = len(extra.jobs)
+                for (new_off, clen, nvals, _an, codec) in cm.chunks:
+                    ck = parse_pages(buf, new_off, clen, nvals, cm.has_def,
+                                     codec, extra)
                     if ck is None or (cm.phys == "BYTE_ARRAY" and not ck.is_dict):
                         good = False  # e.g. PLAIN byte-array pages -> host path
                         break
                     cm.pages.append(ck)
                 if good:
                     ok.append(cm)
+                    jobs.extend(extra.jobs[j0:])
             # per-column fallback: drop unparseable columns; their chunk ranges
             # stay in the staging layout (small waste, correctness unaffected)
             meta.cols = ok
             meta.parsed = True
+            meta.extra_total = extra.cursor - extra.base
+            meta.jobs = jobs
             if not ok:
                 return None
+            if meta.extra_total:
+                # first read discovered compressed pages: grow the staging
+                # buffer to cover the decompressed region
+                if use_gpu and pin_base is not None:
+                    _PINNED.release(pin_base, "cpu")
+                pin_base, buf_t2 = alloc(meta.total + 8 + meta.extra_total + 8)
+                buf2 = buf_t2.numpy()
+                buf2[:meta.total] = buf[:meta.total]
+                buf_t, buf = buf_t2, buf2
+
+        if meta.jobs:
+            run_jobs(buf, meta.jobs)
 
         dbuf = buf_t.to(device, non_blocking=True) if use_gpu else None
 
@@ -593,7 +717,7 @@ def read_columns_native(path: str, columns: List[str], device,
             if cm.phys == "BYTE_ARRAY":
                 parts = []
                 for (chunk, ck) in zip(cm.chunks, cm.pages):
-                    (_off, _clen, nvals, chunk_nulls) = chunk
+                    (_off, _clen, nvals, chunk_nulls, _codec) = chunk
                     parts.append(_decode_chunk_strings(
                         buf, dbuf, ck, nvals, device, use_gpu, chunk_nulls))
                 col = parts[0] if len(parts) == 1 else Column.concat(parts)
@@ -605,7 +729,7 @@ def read_columns_native(path: str, columns: List[str], device,
             parts_valid = []
             any_nulls = False
             for (chunk, ck) in zip(cm.chunks, cm.pages):
-                (_off, _clen, nvals, chunk_nulls) = chunk
+                (_off, _clen, nvals, chunk_nulls, _codec) = chunk
                 if use_gpu:
                     if ck.is_dict:
                         data_t, valid_t = _decode_chunk_gpu_dict(

@@ -48,18 +48,38 @@ def test_np_reference_decoder(pq_file):
     _check(cols, t)
 
 
-def test_fallback_on_dictionary(tmp_path):
+def test_pyarrow_default_file_decodes_natively(tmp_path):
+    # pyarrow defaults = snappy + dictionary; since round 2 the native
+    # reader host-decompresses pages and decodes on device
     t = pa.table({"s": pa.array(["a", "b"] * 100)})
     p = tmp_path / "d.parquet"
     pq.write_table(t, str(p))
-    assert parquet_native.read_columns_native(str(p), ["s"], "cpu", _np_only=True) is None
+    out = parquet_native.read_columns_native(str(p), ["s"], "cpu", _np_only=True)
+    assert out is not None
+    assert out["s"].to_pylist() == ["a", "b"] * 100
 
 
-def test_fallback_on_compression(tmp_path):
-    t = pa.table({"x": pa.array(np.arange(1000))})
-    p = tmp_path / "z.parquet"
-    pq.write_table(t, str(p), compression="zstd", use_dictionary=False)
-    assert parquet_native.read_columns_native(str(p), ["x"], "cpu", _np_only=True) is None
+@pytest.mark.parametrize("codec", ["snappy", "zstd", "gzip"])
+@pytest.mark.parametrize("pv", ["1.0", "2.0"])
+def test_compressed_pages_decode_natively(tmp_path, codec, pv):
+    """Compressed v1/v2 data pages: host page decompression feeds the
+    same device decode path (parquet_exec.rs/arrow-rs parity)."""
+    rng = np.random.default_rng(11)
+    n = 30_000
+    t = pa.table({
+        "x": pa.array(rng.integers(-10**9, 10**9, n), mask=rng.random(n) < 0.04),
+        "f": pa.array(rng.normal(size=n)),
+        "s": pa.array([f"w{int(v)}" for v in rng.integers(0, 50, n)],
+                      mask=rng.random(n) < 0.03),
+    })
+    p = str(tmp_path / f"c-{codec}-{pv}.parquet")
+    pq.write_table(t, p, compression=codec, use_dictionary=["s"],
+                   data_page_version=pv, row_group_size=8192,
+                   data_page_size=4096)
+    cols = parquet_native.read_columns_native(p, ["x", "f", "s"], "cpu",
+                                              _np_only=True)
+    assert cols is not None and set(cols) == {"x", "f", "s"},         f"{codec}/{pv} rejected by the native path"
+    _check(cols, t)
 
 
 @pytest.mark.gpu
