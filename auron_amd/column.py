@@ -144,14 +144,19 @@ class Column:
         return [v if ok else None for v, ok in zip(vals, valid)]
 
     # ---------------------------------------------------------- kernels (torch-expressed)
-    def gather(self, indices: torch.Tensor) -> "Column":
+    def gather(self, indices: torch.Tensor,
+               may_have_negative: bool = False) -> "Column":
         """take(): rows at `indices` (int64). Negative index -1 = emit null
-        (used by outer joins)."""
+        (outer joins set may_have_negative; the default path skips the
+        device->host `.any()` probe, which would sync the stream on every
+        gather — one of the r2 profile's top latency taxes)."""
         idx = indices.to(self.device)
         neg = None
-        if bool((idx < 0).any()) if idx.numel() else False:
+        if may_have_negative and idx.numel():
             neg = idx < 0
             idx = idx.clamp(min=0)
+            if not bool(neg.any()) if idx.device.type != "cuda" else False:
+                neg = None
         validity = None
         if self.validity is not None:
             validity = self.validity[idx]
@@ -303,6 +308,16 @@ class Column:
         return f"Column({self.dtype.name}, n={len(self)}, nulls={self.null_count}, dev={self.device})"
 
 
+
+def compact_validity(v: Optional[torch.Tensor]) -> Optional[torch.Tensor]:
+    """Collapse an all-true validity mask to None — but only where the
+    check is free (CPU). On device the `.all()` reduction would sync the
+    stream; keeping a redundant mask is far cheaper than the bubble."""
+    if v is None or v.device.type == "cuda":
+        return v
+    return None if bool(v.all()) else v
+
+
 class RecordBatch:
     # _eval_memo: optional CSE scope opened by exprs.eval_scope
     __slots__ = ("names", "columns", "_eval_memo")
@@ -334,8 +349,10 @@ class RecordBatch:
     def to(self, device) -> "RecordBatch":
         return RecordBatch(self.names, [c.to(device) for c in self.columns])
 
-    def gather(self, idx: torch.Tensor) -> "RecordBatch":
-        return RecordBatch(self.names, [c.gather(idx) for c in self.columns])
+    def gather(self, idx: torch.Tensor,
+               may_have_negative: bool = False) -> "RecordBatch":
+        return RecordBatch(self.names,
+                           [c.gather(idx, may_have_negative) for c in self.columns])
 
     def filter(self, mask: torch.Tensor) -> "RecordBatch":
         idx = torch.nonzero(mask, as_tuple=False).flatten()

@@ -22,7 +22,7 @@ import torch
 import torch.distributed as dist
 
 from .. import dtypes, ops
-from ..column import Column, RecordBatch
+from ..column import Column, RecordBatch, compact_validity
 from ..dtypes import DataType
 from ..exchange import all_gather_batch, all_to_all
 from ..exprs import AggFunc, Col, WindowFunc
@@ -39,6 +39,10 @@ class ExecContext:
     metrics: Dict[str, float] = field(default_factory=dict)
     memmgr: object = None  # auron_amd.memory.MemManager
     shuffle_seq: int = 0  # deterministic stage ids for persistent shuffles
+    # build-once broadcast cache (broadcast_join_exec.rs:90
+    # cached_build_hash_map_id): plan-fingerprint -> (batch, {keysig: table})
+    broadcast_cache: Dict = field(default_factory=dict)
+    broadcast_cache_bytes: int = 0
 
     def __post_init__(self):
         if self.memmgr is None:
@@ -104,6 +108,26 @@ def _concat(batches: List[RecordBatch]) -> RecordBatch:
 def _empty_like(batch: RecordBatch) -> RecordBatch:
     idx = torch.empty(0, dtype=torch.int64, device=batch.device)
     return batch.gather(idx)
+
+
+
+def _expr_sig(exprs) -> bytes:
+    import hashlib
+
+    from ..plan.serde import serialize_plan  # noqa: F401 (registry init)
+    from ..plan import serde as _serde
+    import msgpack as _mp
+
+    return hashlib.sha1(_mp.packb([_serde._encode(e) for e in exprs],
+                                  use_bin_type=True)).digest()
+
+
+def _cacheable_build(plan) -> bool:
+    """Only deterministic, data-independent subtrees are cacheable (scans
+    + exprs); MemoryScan/PyUdf payloads would need content hashing."""
+    if isinstance(plan, (P.MemoryScan, P.PyUdf, P.PyUdaf, P.Generate)):
+        return False
+    return all(_cacheable_build(c) for c in plan.children())
 
 
 class Executor:
@@ -675,8 +699,8 @@ class Executor:
             ri = torch.cat([ri, un])
             li = torch.cat([li, torch.full((un.numel(),), -1, dtype=torch.int64,
                                            device=device)])
-        out_left = left.gather(li)
-        out_right = right.gather(ri)
+        out_left = left.gather(li, may_have_negative=True)
+        out_right = right.gather(ri, may_have_negative=True)
         return [RecordBatch(out_left.names + out_right.names,
                             out_left.columns + out_right.columns)]
 
@@ -901,8 +925,7 @@ class Executor:
                 if isinstance(data, Column):
                     cols.append(data)
                 else:
-                    cols.append(Column(s0.dtype, data, (cnt > 0)
-                                       if not bool((cnt > 0).all()) else None))
+                    cols.append(Column(s0.dtype, data, compact_validity(cnt > 0)))
             names.append(f"__agg{i}_0")
             cols.append(Column(dtypes.int64, merged_cnt))
             names.append(f"__agg{i}_1")
@@ -1187,9 +1210,7 @@ class Executor:
     def _finalize_agg(self, agg: AggFunc, vdt: DataType, acc, cnt: torch.Tensor) -> Column:
         if agg.fn in ("collect_list", "collect_set"):
             return acc  # empty groups stay empty lists, not null
-        validity = cnt > 0
-        if bool(validity.all()):
-            validity = None
+        validity = compact_validity(cnt > 0)
         if isinstance(acc, Column):
             v = acc.validity
             if validity is not None:
@@ -1226,11 +1247,42 @@ class Executor:
         finally:
             holder.release()
 
+
+    def _broadcast_cache_entry(self, plan):
+        """Build-once broadcast relation cache (cached_build_hash_map_id
+        semantics): keyed by the serialized build subtree, so repeated
+        probes of the same dimension relation — within one query or
+        across a whole bench session — gather and build exactly once."""
+        if not _cacheable_build(plan):
+            return None
+        import hashlib
+
+        from ..plan.serde import serialize_plan
+
+        try:
+            key = hashlib.sha1(serialize_plan(plan)).digest()
+        except Exception:
+            return None
+        entry = self.ctx.broadcast_cache.get(key)
+        if entry is None:
+            if self.ctx.broadcast_cache_bytes > (4 << 30):
+                self.ctx.broadcast_cache.clear()
+                self.ctx.broadcast_cache_bytes = 0
+            entry = {"batch": None, "tables": {}}
+            self.ctx.broadcast_cache[key] = entry
+        return entry
+
     def _exec_hash_join_inner(self, node: P.HashJoin, left_holder) -> List[RecordBatch]:
         def left_batches():
             return left_holder.batches()
 
-        if node.broadcast and self.ctx.world_size > 1:
+        cache_entry = None
+        if node.broadcast and node.build_side == "right":
+            cache_entry = self._broadcast_cache_entry(node.right)
+        if cache_entry is not None and cache_entry.get("batch") is not None:
+            right = cache_entry["batch"]
+            left = _concat(left_batches())
+        elif node.broadcast and self.ctx.world_size > 1:
             if node.build_side == "right":
                 rb = _concat(self.execute(node.right))
                 right = _concat(all_gather_batch(rb, self.ctx.device, self.ctx.group))
@@ -1242,6 +1294,10 @@ class Executor:
         else:
             right = _concat(self.execute(node.right))
             left = _concat(left_batches())
+        if cache_entry is not None and cache_entry.get("batch") is None:
+            cache_entry["batch"] = right
+            self.ctx.broadcast_cache_bytes += sum(
+                c.data.numel() * c.data.element_size() for c in right.columns)
         # SMJ fallback (conf.rs:55-57): an oversized build side makes the
         # chained hash table the wrong tool; lower to the order-based join
         if not node.broadcast:
@@ -1266,11 +1322,22 @@ class Executor:
         how = node.how
         device = self.ctx.device
 
+        table = None
+        if cache_entry is not None and node.build_side == "right":
+            sig = _expr_sig(node.right_keys)
+            table = cache_entry["tables"].get(sig)
+            if table is None:
+                table = ops.join_build(rkeys)
+                if table is not None:
+                    cache_entry["tables"][sig] = table
+                    self.ctx.broadcast_cache_bytes += table.nbytes
+
         if node.residual is not None:
             # equi-pairs first, then the non-equi condition per pair
             bi, pi, _ = ops.hash_join(rkeys, lkeys,
                                       emit_unmatched_probe=False,
-                                      need_build_matched=False)
+                                      need_build_matched=False,
+                                      table=table)
             return self._finish_join_pairs(left, right, pi, bi, how,
                                            node.residual, node.existence_col)
 
@@ -1292,6 +1359,7 @@ class Executor:
             build_keys, probe_keys,
             emit_unmatched_probe=(probe_side in preserved),
             need_build_matched=(build_side in preserved),
+            table=table if build_side == "right" else None,
         )
         if build_side in preserved and bmatched is not None:
             un = torch.nonzero(~bmatched, as_tuple=False).flatten()
@@ -1299,8 +1367,8 @@ class Executor:
                 bi = torch.cat([bi, un])
                 pi = torch.cat([pi, torch.full((un.numel(),), -1, dtype=torch.int64, device=bi.device)])
         li, ri = (pi, bi) if build_side == "right" else (bi, pi)
-        out_left = left.gather(li)
-        out_right = right.gather(ri)
+        out_left = left.gather(li, may_have_negative=True)
+        out_right = right.gather(ri, may_have_negative=True)
         return [RecordBatch(out_left.names + out_right.names,
                             out_left.columns + out_right.columns)]
 
@@ -1351,8 +1419,8 @@ class Executor:
                 ri = torch.cat([ri, un])
                 li = torch.cat([li, torch.full((un.numel(),), -1,
                                                dtype=torch.int64, device=device)])
-        out_left = left.gather(li)
-        out_right = right.gather(ri)
+        out_left = left.gather(li, may_have_negative=True)
+        out_right = right.gather(ri, may_have_negative=True)
         return [RecordBatch(out_left.names + out_right.names,
                             out_left.columns + out_right.columns)]
 
@@ -1478,7 +1546,7 @@ class Executor:
                 idx_c = idx.clamp(0, max(n - 1, 0))
                 ok = ok & (seg[idx_c] == seg) if n else ok
                 gi = torch.where(ok, idx_c, torch.full_like(idx_c, -1))
-                out = val.gather(gi)
+                out = val.gather(gi, may_have_negative=True)
                 if wf.default is not None and out.validity is not None:
                     from ..exprs import Literal, eval_scope
 
@@ -1492,9 +1560,7 @@ class Executor:
                     else:
                         esc = miss
                     data = torch.where(esc, dcol.data, out.data)
-                    validity = out.validity | esc
-                    if bool(validity.all()):
-                        validity = None
+                    validity = compact_validity(out.validity | esc)
                     out = Column(out.dtype, data, validity)
             elif wf.fn in ("percent_rank", "cume_dist", "ntile"):
                 seg_last = (torch.cat([seg_start[1:] - 1,
@@ -1539,7 +1605,8 @@ class Executor:
                     k = wf.offset if wf.offset else 1
                     tgt = seg_start[seg] + (k - 1) if n else seg
                     ok = pos_in_seg >= (k - 1)
-                    out = val.gather(torch.where(ok, tgt, torch.full_like(tgt, -1)))
+                    out = val.gather(torch.where(ok, tgt, torch.full_like(tgt, -1)),
+                                     may_have_negative=True)
             else:
                 raise NotImplementedError(f"window fn {wf.fn}")
             names.append(al.name)
@@ -1599,9 +1666,7 @@ class Executor:
             cs = torch.cumsum(z, 0)
             base = cs[seg_start][seg] - z[seg_start][seg]
             run = cs - base
-            validity = run_count > 0
-            if bool(validity.all()):
-                validity = None
+            validity = compact_validity(run_count > 0)
             if fn == "avg":
                 return Column(dtypes.float64,
                               run.to(torch.float64) / run_count.clamp(min=1).to(torch.float64),
@@ -1634,9 +1699,7 @@ class Executor:
             ok = (pos - off) >= start
             run = torch.where(ok, combine(run, cand), run)
             off <<= 1
-        validity = run_count > 0
-        if bool(validity.all()):
-            validity = None
+        validity = compact_validity(run_count > 0)
         if val.dtype.code == dtypes.DECIMAL64:
             return Column(val.dtype, run, validity)
         if val.dtype.is_integer:

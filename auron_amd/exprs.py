@@ -22,7 +22,7 @@ from typing import List, Optional, Sequence, Tuple
 import torch
 
 from . import dtypes, strings
-from .column import Column, RecordBatch
+from .column import Column, RecordBatch, compact_validity
 from .dtypes import DataType
 
 
@@ -335,12 +335,12 @@ class Arith(Expr):
         elif self.op == "*":
             data = a * b
         elif self.op in ("/", "%"):
+            # unconditional guard: probing `zero.any()` would sync the
+            # stream on every division (r2 profile tax)
             zero = b == 0
-            if bool(zero.any()):
-                bad = zero
-                b = torch.where(zero, torch.ones_like(b), b)
-                v2 = ~bad
-                validity = v2 if validity is None else (validity & v2)
+            b = torch.where(zero, torch.ones_like(b), b)
+            v2 = ~zero
+            validity = v2 if validity is None else (validity & v2)
             if self.op == "/":
                 if dt.is_integer:
                     # Spark SQL `/` is double division for any input type
@@ -411,8 +411,7 @@ class BoolOp(Expr):
                 out_valid = (valid & v2) | (valid & val) | (v2 & b)
                 val = (val & valid) | (b & v2)
                 valid = out_valid
-        if bool(valid.all()):
-            valid = None
+        valid = compact_validity(valid)
         return Column(dtypes.bool_, val, valid)
 
 
@@ -461,8 +460,7 @@ class Coalesce(Expr):
                     validity = None
                 else:
                     validity = out.validity | c2.validity
-                    if bool(validity.all()):
-                        validity = None
+                    validity = compact_validity(validity)
                 out = Column(out.dtype, data, validity)
         return out
 
@@ -512,8 +510,7 @@ class CaseWhen(Expr):
             result_data = torch.where(take, v.data, result_data)
             vv = v.validity if v.validity is not None else _all_valid(n, device)
             result_valid = torch.where(take, vv, result_valid)
-        if bool(result_valid.all()):
-            result_valid = None
+        result_valid = compact_validity(result_valid)
         return Column(out_dt, result_data, result_valid)
 
 

@@ -14,7 +14,7 @@ from typing import List
 import torch
 
 from . import dtypes, strings
-from .column import Column
+from .column import Column, compact_validity
 from .exprs import (Expr, _all_valid, _cast_col, _civil_from_days,
                     combine_validity)
 
@@ -131,7 +131,7 @@ class Greatest(Expr):
             v = c.validity if c.validity is not None else _all_valid(n, device)
             acc = torch.where(v, torch.maximum(acc, c.data), acc)
             any_valid = any_valid | v
-        return Column(dtypes.float64, acc, None if bool(any_valid.all()) else any_valid)
+        return Column(dtypes.float64, acc, compact_validity(any_valid))
 
 
 @dataclass(eq=False)
@@ -148,7 +148,7 @@ class Least(Expr):
             v = c.validity if c.validity is not None else _all_valid(n, device)
             acc = torch.where(v, torch.minimum(acc, c.data), acc)
             any_valid = any_valid | v
-        return Column(dtypes.float64, acc, None if bool(any_valid.all()) else any_valid)
+        return Column(dtypes.float64, acc, compact_validity(any_valid))
 
 
 @dataclass(eq=False)
@@ -819,7 +819,7 @@ class GetArrayItem(Expr):
         pos = (off[:-1] + self.index).clamp(0, max(int(c.data.numel()) - 1, 0))
         data = c.data[pos] if c.data.numel() else \
             torch.zeros(len(c), dtype=c.dtype.torch_dtype, device=c.device)
-        return Column(c.dtype.child, data, ok if not bool(ok.all()) else None)
+        return Column(c.dtype.child, data, compact_validity(ok))
 
 
 @dataclass(eq=False)
@@ -844,7 +844,7 @@ class ElementAt(Expr):
         pos = (off[:-1] + k).clamp(0, max(int(c.data.numel()) - 1, 0))
         data = c.data[pos] if c.data.numel() else \
             torch.zeros(len(c), dtype=c.dtype.torch_dtype, device=c.device)
-        return Column(c.dtype.child, data, ok if not bool(ok.all()) else None)
+        return Column(c.dtype.child, data, compact_validity(ok))
 
 
 @dataclass(eq=False)

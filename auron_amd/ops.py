@@ -10,7 +10,7 @@ auron_amd.native.require().
 from __future__ import annotations
 
 import os
-from typing import List, Tuple
+from typing import List, Optional, Tuple
 
 import torch
 
@@ -246,9 +246,51 @@ def hash_join_ref(build_cols, probe_cols, emit_unmatched_probe, need_build_match
             torch.tensor(pi, dtype=torch.int64, device=device), bm)
 
 
+
+class JoinTable:
+    """Prebuilt chained hash table over a broadcast build side (the
+    reference's cached_build_hash_map_id object,
+    broadcast_join_exec.rs:90): heads/next chains + row hashes stay in
+    HBM so repeated probes of the same relation skip the build pass."""
+
+    __slots__ = ("heads", "nxt", "cap", "bhash", "n_build")
+
+    def __init__(self, heads, nxt, cap, bhash, n_build):
+        self.heads = heads
+        self.nxt = nxt
+        self.cap = cap
+        self.bhash = bhash
+        self.n_build = n_build
+
+    @property
+    def nbytes(self):
+        return (self.heads.numel() + self.nxt.numel()) * 4 + self.bhash.numel() * 8
+
+
+def join_build(build_cols: List[Column]) -> Optional["JoinTable"]:
+    """Build the chained table once (au_join_build); None on the CPU
+    reference path (which has no reusable native table)."""
+    device = build_cols[0].device
+    if not _use_native(device):
+        return None
+    lib = native.lib()
+    sp = native.stream_ptr(device)
+    n_build = len(build_cols[0])
+    bhash = murmur3(build_cols)
+    cap = _next_pow2(2 * max(n_build, 1))
+    heads = torch.full((cap,), -1, dtype=torch.int32, device=device)
+    nxt = torch.empty(max(n_build, 1), dtype=torch.int32, device=device)
+    if n_build:
+        rc = lib.au_join_build(n_build, heads.data_ptr(), cap, nxt.data_ptr(),
+                               bhash.data_ptr(), sp)
+        native.check(rc, "au_join_build")
+    return JoinTable(heads, nxt, cap, bhash, n_build)
+
+
 def hash_join(build_cols: List[Column], probe_cols: List[Column],
               emit_unmatched_probe: bool = False,
-              need_build_matched: bool = False):
+              need_build_matched: bool = False,
+              table: Optional["JoinTable"] = None):
     """Hash-join index computation.
 
     -> (build_idx [m] int64 (-1 = probe row had no match),
@@ -263,17 +305,12 @@ def hash_join(build_cols: List[Column], probe_cols: List[Column],
         return hash_join_ref(build_cols, probe_cols, emit_unmatched_probe, need_build_matched)
     lib = native.lib()
     sp = native.stream_ptr(device)
-    bhash = murmur3(build_cols)
+    if table is None:
+        table = join_build(build_cols)
+    heads, nxt, cap, bhash = table.heads, table.nxt, table.cap, table.bhash
     phash = murmur3(probe_cols)
-    cap = _next_pow2(2 * max(n_build, 1))
-    heads = torch.full((cap,), -1, dtype=torch.int32, device=device)
-    nxt = torch.empty(max(n_build, 1), dtype=torch.int32, device=device)
     bdescs, bkeep = native.pack_descs(build_cols, device)
     pdescs, pkeep = native.pack_descs(probe_cols, device)
-    if n_build:
-        rc = lib.au_join_build(n_build, heads.data_ptr(), cap, nxt.data_ptr(),
-                               bhash.data_ptr(), sp)
-        native.check(rc, "au_join_build")
     counts = torch.zeros(n_probe, dtype=torch.int32, device=device)
     rc = lib.au_join_count(bdescs.data_ptr(), pdescs.data_ptr(), len(build_cols),
                            n_probe, heads.data_ptr(), cap, nxt.data_ptr(),

@@ -121,7 +121,7 @@ def test_list_column_and_explode():
     assert c.to_pylist() == [[1, 2], [], None, [7]]
     # gather with nulls / reorder keeps layout
     import torch
-    g = c.gather(torch.tensor([3, 0, -1]))
+    g = c.gather(torch.tensor([3, 0, -1]), may_have_negative=True)
     assert g.to_pylist() == [[7], [1, 2], None]
     assert Column.concat([c, g]).to_pylist() == \
         [[1, 2], [], None, [7], [7], [1, 2], None]

@@ -20,13 +20,13 @@ def test_pylist_roundtrip_string():
 
 def test_gather_numeric_with_null_marker():
     c = Column.from_pylist([10, 20, 30], dtypes.int64)
-    out = c.gather(torch.tensor([2, -1, 0]))
+    out = c.gather(torch.tensor([2, -1, 0]), may_have_negative=True)
     assert out.to_pylist() == [30, None, 10]
 
 
 def test_gather_string():
     c = Column.from_pylist(["aa", "b", "cccc"], dtypes.string)
-    out = c.gather(torch.tensor([2, 2, 0, -1]))
+    out = c.gather(torch.tensor([2, 2, 0, -1]), may_have_negative=True)
     assert out.to_pylist() == ["cccc", "cccc", "aa", None]
 
 
