@@ -40,12 +40,52 @@ struct Run {
 
 #define RLE_WINDOW 512
 
+// Split factor: each page gets RLE1_SPLIT workgroups. Writers (pyarrow,
+// parquet-mr) emit definition levels as ONE hybrid run per page in the
+// common case; that run is sliced across the page's workgroups so the
+// whole chip fills even for a handful of pages. Multi-run pages fall back
+// to the windowed serial-parse path in workgroup 0.
+#define RLE1_SPLIT 16
+
 __global__ void k_pq_rle1(const PqPage* pages, const uint8_t* buf, uint8_t* out) {
-  const PqPage p = pages[blockIdx.x];
+  const PqPage p = pages[blockIdx.x / RLE1_SPLIT];
+  const int slice = blockIdx.x % RLE1_SPLIT;
   const uint8_t* src = buf + p.def_off;
   int64_t src_len = p.def_len;
   uint8_t* dst = out + p.row_start;
   const int64_t n = p.n_values;
+
+  // ---- fast path: single run spans the page ----
+  {
+    int64_t pos = 0;
+    uint64_t header = 0;
+    int shift = 0;
+    while (pos < src_len) {
+      uint8_t b = src[pos++];
+      header |= (uint64_t)(b & 0x7F) << shift;
+      if (!(b & 0x80)) break;
+      shift += 7;
+    }
+    int64_t covered = (header & 1) ? (int64_t)(header >> 1) * 8
+                                   : (int64_t)(header >> 1);
+    if (covered >= n) {
+      const int64_t per = (n + RLE1_SPLIT - 1) / RLE1_SPLIT;
+      const int64_t lo = slice * per;
+      const int64_t hi = min(lo + per, n);
+      if (lo >= hi) return;
+      if (header & 1) {
+        const uint8_t* bits = src + pos;
+        for (int64_t j = lo + threadIdx.x; j < hi; j += blockDim.x)
+          dst[j] = (bits[j >> 3] >> (j & 7)) & 1;
+      } else {
+        const uint8_t v = src[pos] & 1;
+        for (int64_t j = lo + threadIdx.x; j < hi; j += blockDim.x)
+          dst[j] = v;
+      }
+      return;
+    }
+  }
+  if (slice != 0) return;  // general path: one workgroup per page
 
   __shared__ Run runs[RLE_WINDOW];
   __shared__ int nruns;
@@ -120,7 +160,8 @@ __global__ void k_pq_rle1(const PqPage* pages, const uint8_t* buf, uint8_t* out)
 AU_EXPORT int au_pq_rle1(const void* pages_dev, int npages, const void* buf,
                          uint8_t* out, void* stream) {
   if (npages == 0) return 0;
-  hipLaunchKernelGGL(k_pq_rle1, dim3(npages), dim3(256), 0, (hipStream_t)stream,
+  hipLaunchKernelGGL(k_pq_rle1, dim3(npages * RLE1_SPLIT), dim3(256), 0,
+                     (hipStream_t)stream,
                      (const PqPage*)pages_dev, (const uint8_t*)buf, out);
   return (int)hipGetLastError();
 }
