@@ -48,6 +48,41 @@ LOG_LEVEL = _opt("spark.auron.native.log.level", "WARN", str, env="AURON_LOG_LEV
 IGNORE_CORRUPTED_FILES = _opt("spark.auron.ignoreCorruptedFiles", False, bool,
                               "skip unreadable input files instead of failing the task",
                               env="AURON_IGNORE_CORRUPTED_FILES")
+BROADCAST_MAX_ROWS = _opt("spark.auron.broadcast.maxRows", 20_000_000, int,
+                          "estimated build-side rows above which the planner "
+                          "shuffles both join sides instead of broadcasting",
+                          env="AURON_BROADCAST_MAX_ROWS")
+BROADCAST_CACHE_BYTES = _opt("spark.auron.broadcast.cache.maxBytes", 4 << 30, int,
+                             "device bytes of build-once broadcast relations/"
+                             "hash tables kept across queries",
+                             env="AURON_BROADCAST_CACHE_BYTES")
+FORCE_SHUFFLED_HASH_JOIN = _opt("spark.auron.forceShuffledHashJoin", False, bool,
+                                "never broadcast: hash-exchange both sides",
+                                env="AURON_FORCE_SHJ")
+AGG_STREAMING = _opt("spark.auron.agg.streaming.enable", True, bool,
+                     "chunked spill-capable partial aggregation",
+                     env="AURON_AGG_STREAMING")
+HOST_SPILL_BUDGET = _opt("spark.auron.memory.hostSpillBudget", 64 << 30, int,
+                         "host-DRAM bytes of spilled holders before the disk "
+                         "tier engages", env="AURON_HOST_SPILL_BUDGET")
+SHUFFLE_PERSIST = _opt("spark.auron.shuffle.persist", False, bool,
+                       "route exchanges through durable .data/.index files "
+                       "(stage-retry contract)", env="AURON_SHUFFLE_PERSIST")
+PARQUET_NATIVE = _opt("spark.auron.parquet.native.enable", True, bool,
+                      "device parquet decode (else pyarrow host read)",
+                      env="AURON_PARQUET_NATIVE")
+PARQUET_PAGE_DECOMPRESS = _opt("spark.auron.parquet.pageDecompression.enable",
+                               True, bool,
+                               "host-decompress snappy/zstd/gzip pages for "
+                               "the device decode path")
+PARTIAL_AGG_SKIPPING_MINROWS = _opt("spark.auron.partialAggSkipping.minRows",
+                                    1 << 14, int,
+                                    "below this many rows partial-agg "
+                                    "skipping is never sampled")
+SORT_LIMIT_TOPK = _opt("spark.auron.sort.topK.enable", True, bool,
+                       "ORDER BY+LIMIT keeps only the top K rows per batch")
+EXCHANGE_SAMPLE_ROWS = _opt("spark.auron.rangePartition.sampleRows", 4096, int,
+                            "per-rank sample size for range-partition bounds")
 
 
 class AuronConf:

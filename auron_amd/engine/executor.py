@@ -27,6 +27,7 @@ from ..dtypes import DataType
 from ..exchange import all_gather_batch, all_to_all
 from ..exprs import AggFunc, Col, WindowFunc
 from ..plan import nodes as P
+from ..config import AGG_STREAMING, BROADCAST_CACHE_BYTES, AuronConf
 
 
 @dataclass
@@ -826,7 +827,7 @@ class Executor:
     def _exec_HashAgg(self, node: P.HashAgg) -> List[RecordBatch]:
         from ..exprs import eval_scope
 
-        if node.mode == "partial" and os.environ.get("AURON_AGG_STREAMING", "1") != "0":
+        if node.mode == "partial" and AuronConf().get(AGG_STREAMING):
             # streaming partial agg (agg_table.rs analogue): inputs are
             # consumed chunk-by-chunk, states re-merged when they shrink,
             # and the accumulated state registers with the memmgr so it
@@ -1265,7 +1266,7 @@ class Executor:
             return None
         entry = self.ctx.broadcast_cache.get(key)
         if entry is None:
-            if self.ctx.broadcast_cache_bytes > (4 << 30):
+            if self.ctx.broadcast_cache_bytes > AuronConf().get(BROADCAST_CACHE_BYTES):
                 self.ctx.broadcast_cache.clear()
                 self.ctx.broadcast_cache_bytes = 0
             entry = {"batch": None, "tables": {}}

@@ -165,8 +165,9 @@ class MemManager:
                 budget_bytes = 16 << 30
         self.budget = budget_bytes
         if host_budget_bytes is None:
-            host_budget_bytes = int(os.environ.get("AURON_HOST_SPILL_BUDGET",
-                                                   str(64 << 30)))
+            from .config import HOST_SPILL_BUDGET, AuronConf
+
+            host_budget_bytes = int(AuronConf().get(HOST_SPILL_BUDGET))
         self.host_budget = host_budget_bytes
         self.spill_dir = spill_dir or tempfile.gettempdir()
         self._holders: List[BatchHolder] = []

@@ -189,9 +189,13 @@ class Lowering:
         lkeys = [self.subst_scalars(k) for k in node.lkeys]
         rkeys = [self.subst_scalars(k) for k in node.rkeys]
         residual = self.subst_scalars(node.residual) if node.residual is not None else None
+        from ..config import FORCE_SHUFFLED_HASH_JOIN, AuronConf
+
         const_keys = all(isinstance(k, Literal) for k in rkeys)
         can_broadcast = node.kind in ("inner", "left", "semi", "anti",
                                       "existence")
+        if AuronConf().get(FORCE_SHUFFLED_HASH_JOIN):
+            can_broadcast = const_keys  # cross joins still broadcast the 1-row side
         small = node.r_base_dim or node.r_est <= BROADCAST_ROWS or const_keys
         if can_broadcast and small:
             plan = P.HashJoin(lplan, rplan, lkeys, rkeys, how=node.kind,

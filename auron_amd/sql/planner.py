@@ -211,8 +211,13 @@ FACT_TABLES = {
     "store_returns", "catalog_returns", "web_returns",
 }
 
-BROADCAST_ROWS = float(int(__import__("os").environ.get(
-    "AURON_BROADCAST_MAX_ROWS", str(20_000_000))))
+def _broadcast_rows() -> float:
+    from ..config import BROADCAST_MAX_ROWS, AuronConf
+
+    return float(AuronConf().get(BROADCAST_MAX_ROWS))
+
+
+BROADCAST_ROWS = _broadcast_rows()
 
 
 def expr_key(e) -> tuple:
