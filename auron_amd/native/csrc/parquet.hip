@@ -47,6 +47,8 @@ struct Run {
 // to the windowed serial-parse path in workgroup 0.
 #define RLE1_SPLIT 16
 
+#define RLE1_LDS_BYTES 16384
+
 __global__ void k_pq_rle1(const PqPage* pages, const uint8_t* buf, uint8_t* out) {
   const PqPage p = pages[blockIdx.x / RLE1_SPLIT];
   const int slice = blockIdx.x % RLE1_SPLIT;
@@ -54,6 +56,18 @@ __global__ void k_pq_rle1(const PqPage* pages, const uint8_t* buf, uint8_t* out)
   int64_t src_len = p.def_len;
   uint8_t* dst = out + p.row_start;
   const int64_t n = p.n_values;
+
+  // Stage the page's level bytes through LDS: the run-header parse is a
+  // serial chain of dependent byte loads, and HBM-latency per byte is
+  // what made this kernel the r2 profile's #1 entry. A bit-width-1 page
+  // of 8k values is ~1-3 KB, far under the 160 KB LDS per CU.
+  __shared__ uint8_t sbuf[RLE1_LDS_BYTES];
+  if (src_len <= RLE1_LDS_BYTES) {
+    for (int64_t i = threadIdx.x; i < src_len; i += blockDim.x)
+      sbuf[i] = src[i];
+    __syncthreads();
+    src = sbuf;
+  }
 
   // ---- fast path: single run spans the page ----
   {
