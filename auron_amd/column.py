@@ -175,6 +175,17 @@ class Column:
             total = int(new_off[-1].item())
             if total == 0:
                 out_el = torch.empty(0, dtype=self.data.dtype, device=self.device)
+            elif self.device.type == "cuda" and self.data.dtype == torch.uint8:
+                # one kernel instead of the arange/repeat_interleave/index
+                # chain (k_bytes_gather, see kernels.hip)
+                from . import native
+
+                out_el = torch.empty(total, dtype=torch.uint8, device=self.device)
+                rc = native.lib().au_bytes_gather(
+                    self.data.data_ptr(), starts.contiguous().data_ptr(),
+                    new_off.data_ptr(), out_el.data_ptr(), idx.shape[0],
+                    native.stream_ptr(self.device))
+                native.check(rc, "au_bytes_gather")
             else:
                 pos = torch.arange(total, dtype=torch.int64, device=self.device)
                 row = torch.repeat_interleave(lens)  # maps element pos -> out row

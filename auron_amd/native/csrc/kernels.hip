@@ -444,5 +444,38 @@ AU_EXPORT int au_part_scatter(const int32_t* part_ids, int64_t n,
   return (int)hipGetLastError();
 }
 
+// ------------------------------------------------------------- byte gather
+// Variable-length row gather for string columns (arrow/selection.rs take()
+// analogue): one wavefront per output row copies its bytes, replacing the
+// arange/repeat_interleave/fancy-index chain (5 torch kernels) the r2
+// profile showed on every string gather.
+__global__ void k_bytes_gather(const uint8_t* __restrict__ src,
+                               const int64_t* __restrict__ src_start,
+                               const int64_t* __restrict__ out_off,
+                               uint8_t* __restrict__ out, int64_t nrows) {
+  const int64_t wave = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const int lane = threadIdx.x & 63;
+  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) >> 6;
+  for (int64_t row = wave; row < nrows; row += nwaves) {
+    const int64_t dst0 = out_off[row];
+    const int64_t len = out_off[row + 1] - dst0;
+    const int64_t s0 = src_start[row];
+    for (int64_t j = lane; j < len; j += 64) out[dst0 + j] = src[s0 + j];
+  }
+}
+
+AU_EXPORT int au_bytes_gather(const uint8_t* src, const int64_t* src_start,
+                              const int64_t* out_off, uint8_t* out,
+                              int64_t nrows, void* stream) {
+  if (nrows == 0) return 0;
+  int64_t waves_needed = nrows;
+  int64_t blocks = (waves_needed * 64 + 255) / 256;
+  if (blocks > 8192) blocks = 8192;
+  if (blocks < 1) blocks = 1;
+  hipLaunchKernelGGL(k_bytes_gather, dim3((uint32_t)blocks), dim3(256), 0,
+                     (hipStream_t)stream, src, src_start, out_off, out, nrows);
+  return (int)hipGetLastError();
+}
+
 // ------------------------------------------------------------------ version
 AU_EXPORT int au_abi_version() { return 1; }

@@ -269,3 +269,21 @@ def test_agg_multi_fused_matches_single():
                                   rtol=1e-9, atol=1e-6), f
         else:
             assert torch.equal(facc.cpu()[nz], sacc.cpu()[nz]), f
+
+
+@pytest.mark.gpu
+def test_bytes_gather_kernel_matches_cpu():
+    import numpy as np
+
+    rng = np.random.default_rng(9)
+    words = ["", "a", "bb", "long-string-value-" * 3] + [f"w{i}" for i in range(200)]
+    vals = [words[i] for i in rng.integers(0, len(words), 100_000)]
+    c = Column.from_pylist(vals, dtypes.string, "cuda:0")
+    idx = torch.as_tensor(rng.integers(0, len(vals), 250_000), dtype=torch.int64)
+    got = c.gather(idx.to("cuda:0")).to("cpu")
+    want = c.to("cpu").gather(idx)
+    assert got.to_pylist() == want.to_pylist()
+    # negative markers emit nulls
+    idx2 = torch.tensor([0, -1, 5, -1], dtype=torch.int64, device="cuda:0")
+    g2 = c.gather(idx2, may_have_negative=True).to("cpu")
+    assert g2.to_pylist() == [vals[0], None, vals[5], None]
