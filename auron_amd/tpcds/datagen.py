@@ -595,14 +595,15 @@ def _nparts_for(table: str, sf: float) -> int:
     rows = row_count(table, sf)
     if table in ("store_sales", "catalog_sales", "web_sales", "inventory",
                  "store_returns", "catalog_returns", "web_returns"):
-        # >=8 parts so an 8-GPU node shards fact scans evenly
-        return max(8, min(256, (rows + 2_000_000 - 1) // 2_000_000))
+        # >=8 parts so an 8-GPU node shards fact scans evenly; large parts
+        # keep decode launches few and grids full (8M rows per part)
+        return max(8, min(256, (rows + 8_000_000 - 1) // 8_000_000))
     if table == "customer_demographics":
         return max(1, min(16, (rows + 2_000_000 - 1) // 2_000_000))
     return 1
 
 
-DATAGEN_VERSION = 14
+DATAGEN_VERSION = 15
 
 
 def dataset_root(root: str, sf: float) -> str:
