@@ -61,6 +61,10 @@ def _try_load():
     for f in ("au_agg_scatter", "au_agg_count", "au_agg_multi"):
         getattr(lib, f).restype = ctypes.c_int
     lib.au_pq_rle1.argtypes = [c, ctypes.c_int, c, c, c]
+    lib.au_rle1_expand.argtypes = [c, i64, c, c, c]
+    lib.au_rle1_expand.restype = ctypes.c_int
+    lib.au_host_rle1_parse.argtypes = [c, i64, i64, i64, i64, c, i64]
+    lib.au_host_rle1_parse.restype = i64
     lib.au_pq_rle_idx.argtypes = [c, ctypes.c_int, c, c, c, c]
     lib.au_pq_scatter.argtypes = [c, ctypes.c_int, c, c, c, c, ctypes.c_int, i64, c]
     lib.au_pq_copy_plain.argtypes = [c, ctypes.c_int, c, c, ctypes.c_int, i64, c]
@@ -90,6 +94,12 @@ def require():
 
 
 def lib():
+    return require()
+
+
+def host_lib():
+    """Host-callable entry points (e.g. au_host_rle1_parse) work without a
+    GPU; same .so, same loader."""
     return require()
 
 

@@ -171,6 +171,87 @@ __global__ void k_pq_rle1(const PqPage* pages, const uint8_t* buf, uint8_t* out)
   }
 }
 
+// --------------------------------------------------- host-parsed RLE1 path
+// The run-header walk is inherently serial PER PAGE and, at ~2-3k runs per
+// page for a few-percent-null column, dominates decode time if done on one
+// device thread. Pages are static per file, so the HOST parses headers once
+// (this plain-C routine, cached in the reader's file metadata) and the
+// device only runs the embarrassingly-parallel expansion kernel below.
+// Run encoding: out_start/count are chunk-absolute rows; src_off is an
+// absolute byte offset into the staged buffer; rep_val -1 = literal run.
+AU_EXPORT int64_t au_host_rle1_parse(const uint8_t* src, int64_t src_len,
+                                     int64_t n, int64_t out_base,
+                                     int64_t src_base, int32_t* runs,
+                                     int64_t cap) {
+  int64_t pos = 0, i = 0, r = 0;
+  while (i < n && pos < src_len) {
+    uint64_t header = 0;
+    int shift = 0;
+    while (pos < src_len) {
+      uint8_t b = src[pos++];
+      header |= (uint64_t)(b & 0x7F) << shift;
+      if (!(b & 0x80)) break;
+      shift += 7;
+    }
+    if (r >= cap) return -1;
+    if (header & 1) {
+      int64_t ngroups = (int64_t)(header >> 1);
+      int64_t nvals = ngroups * 8;
+      if (nvals > n - i) nvals = n - i;
+      runs[4 * r + 0] = (int32_t)(out_base + i);
+      runs[4 * r + 1] = (int32_t)nvals;
+      runs[4 * r + 2] = (int32_t)(src_base + pos);
+      runs[4 * r + 3] = -1;
+      pos += ngroups;
+      i += nvals;
+    } else {
+      int64_t cnt = (int64_t)(header >> 1);
+      if (cnt > n - i) cnt = n - i;
+      runs[4 * r + 0] = (int32_t)(out_base + i);
+      runs[4 * r + 1] = (int32_t)cnt;
+      runs[4 * r + 2] = 0;
+      runs[4 * r + 3] = src[pos] & 1;
+      pos += 1;
+      i += cnt;
+    }
+    r++;
+  }
+  return i == n ? r : -1;
+}
+
+__global__ void k_rle1_expand(const int32_t* __restrict__ runs, int64_t nruns,
+                              const uint8_t* __restrict__ buf,
+                              uint8_t* __restrict__ out) {
+  const int64_t wave = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const int lane = threadIdx.x & 63;
+  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) >> 6;
+  for (int64_t r = wave; r < nruns; r += nwaves) {
+    const int32_t out_start = runs[4 * r + 0];
+    const int32_t count = runs[4 * r + 1];
+    const int32_t src_off = runs[4 * r + 2];
+    const int32_t rep = runs[4 * r + 3];
+    if (rep >= 0) {
+      const uint8_t v = (uint8_t)rep;
+      for (int32_t j = lane; j < count; j += 64) out[out_start + j] = v;
+    } else {
+      const uint8_t* bits = buf + src_off;
+      for (int32_t j = lane; j < count; j += 64)
+        out[out_start + j] = (bits[j >> 3] >> (j & 7)) & 1;
+    }
+  }
+}
+
+AU_EXPORT int au_rle1_expand(const int32_t* runs_dev, int64_t nruns,
+                             const void* buf, uint8_t* out, void* stream) {
+  if (nruns == 0) return 0;
+  int64_t blocks = (nruns * 64 + 255) / 256;
+  if (blocks > 4096) blocks = 4096;
+  hipLaunchKernelGGL(k_rle1_expand, dim3((uint32_t)blocks), dim3(256), 0,
+                     (hipStream_t)stream, runs_dev, nruns,
+                     (const uint8_t*)buf, out);
+  return (int)hipGetLastError();
+}
+
 AU_EXPORT int au_pq_rle1(const void* pages_dev, int npages, const void* buf,
                          uint8_t* out, void* stream) {
   if (npages == 0) return 0;

@@ -165,6 +165,9 @@ class ChunkPages:
     # the staged buffer, entry byte lengths)
     dict_str_offs: Optional[np.ndarray] = None
     dict_str_lens: Optional[np.ndarray] = None
+    # host-parsed def-level runs (au_host_rle1_parse): [N,4] int32 of
+    # (abs out row, count, abs src byte, rep|-1); device expansion only
+    runs_np: Optional[np.ndarray] = None
 
     @property
     def is_dict(self) -> bool:
@@ -559,6 +562,7 @@ class _FileMeta:
     parsed: bool = False
     extra_total: int = 0  # decompressed-page region size (after raw+pad)
     jobs: list = None  # decompress/copy jobs refilled per read
+    runs_parsed: bool = False  # def-level run headers host-parsed
 
 
 def _build_meta(path: str, columns: List[str]) -> Optional[_FileMeta]:
@@ -710,6 +714,16 @@ def read_columns_native(path: str, columns: List[str], device,
         if meta.jobs:
             run_jobs(buf, meta.jobs)
 
+        if use_gpu and not meta.runs_parsed:
+            # one-time host parse of def-level run headers (see
+            # _parse_rle1_runs); must run after decompression jobs so
+            # compressed chunks' level bytes are in place
+            for cm in meta.cols:
+                for ck in cm.pages:
+                    if ck.pages and ck.pages[0].def_off >= 0:
+                        ck.runs_np = _parse_rle1_runs(buf, ck.pages)
+            meta.runs_parsed = True
+
         dbuf = buf_t.to(device, non_blocking=True) if use_gpu else None
 
         out: Dict[str, Column] = {}
@@ -737,7 +751,8 @@ def read_columns_native(path: str, columns: List[str], device,
                     else:
                         data_t, valid_t = _decode_chunk_gpu(dbuf, ck.pages, nvals,
                                                             cm.phys, device,
-                                                            chunk_nulls)
+                                                            chunk_nulls,
+                                                            ck.runs_np)
                 else:
                     if ck.is_dict:
                         data_np, valid_np, _ = decode_chunk_np_dict(buf, ck, nvals,
@@ -772,7 +787,40 @@ def read_columns_native(path: str, columns: List[str], device,
         if pin_base is not None:
             _PINNED.release(pin_base, device)
 
-def _gpu_validity_prefix(dbuf, pages, num_values, device, chunk_nulls=True):
+
+def _parse_rle1_runs(buf: np.ndarray, pages: List[PageDesc]) -> Optional[np.ndarray]:
+    """Host-parse all def-level run headers of a chunk once (cached in the
+    file meta): the serial walk that made k_pq_rle1 the #1 kernel moves to
+    the CPU where it costs microseconds, leaving the device a flat
+    expansion (k_rle1_expand)."""
+    import ctypes
+
+    from . import native
+
+    if not native.available():
+        return None
+    lib = native.host_lib()
+    cap = sum(p.def_len + 2 for p in pages)
+    if cap <= 0:
+        return None
+    runs = np.empty((cap, 4), dtype=np.int32)
+    base = buf.ctypes.data
+    n = 0
+    for p in pages:
+        r = lib.au_host_rle1_parse(
+            ctypes.c_void_p(base + p.def_off), ctypes.c_int64(p.def_len),
+            ctypes.c_int64(p.n_values), ctypes.c_int64(p.row_start),
+            ctypes.c_int64(p.def_off),
+            ctypes.c_void_p(runs.ctypes.data + n * 16),
+            ctypes.c_int64(cap - n))
+        if r < 0:
+            return None
+        n += r
+    return np.ascontiguousarray(runs[:n])
+
+
+def _gpu_validity_prefix(dbuf, pages, num_values, device, chunk_nulls=True,
+                         runs_np=None):
     """Decode def levels on device; returns (validity uint8 | None, prefix).
 
     Null-free chunks (stats null_count == 0) skip the decode."""
@@ -782,16 +830,22 @@ def _gpu_validity_prefix(dbuf, pages, num_values, device, chunk_nulls=True):
     has_def = pages[0].def_off >= 0 and chunk_nulls
     if not has_def:
         return None, None
-    npages = len(pages)
-    arr = np.zeros((npages, 6), dtype=np.int64)
-    for i, p in enumerate(pages):
-        arr[i] = (p.def_off, p.def_len, p.values_off, p.n_values, p.row_start, 0)
-    darr = _pin_to_device(arr.reshape(-1), device)
     sp = native.stream_ptr(device)
     validity = torch.empty(num_values, dtype=torch.uint8, device=device)
-    rc = lib.au_pq_rle1(darr.data_ptr(), npages, dbuf.data_ptr(),
-                        validity.data_ptr(), sp)
-    native.check(rc, "au_pq_rle1")
+    if runs_np is not None:
+        druns = _pin_to_device(runs_np.reshape(-1), device)
+        rc = lib.au_rle1_expand(druns.data_ptr(), runs_np.shape[0],
+                                dbuf.data_ptr(), validity.data_ptr(), sp)
+        native.check(rc, "au_rle1_expand")
+    else:
+        npages = len(pages)
+        arr = np.zeros((npages, 6), dtype=np.int64)
+        for i, p in enumerate(pages):
+            arr[i] = (p.def_off, p.def_len, p.values_off, p.n_values, p.row_start, 0)
+        darr = _pin_to_device(arr.reshape(-1), device)
+        rc = lib.au_pq_rle1(darr.data_ptr(), npages, dbuf.data_ptr(),
+                            validity.data_ptr(), sp)
+        native.check(rc, "au_pq_rle1")
     prefix = torch.cumsum(validity.to(torch.int64), 0)
     return validity, prefix
 
@@ -826,7 +880,7 @@ def _decode_chunk_gpu_dict(dbuf, buf, ck: ChunkPages, num_values: int,
                            phys: str, device, chunk_nulls=True):
     """Dictionary-encoded fixed-width chunk on device."""
     validity, prefix = _gpu_validity_prefix(dbuf, ck.pages, num_values, device,
-                                            chunk_nulls)
+                                            chunk_nulls, ck.runs_np)
     idx = _gpu_dict_indices(dbuf, buf, ck, prefix, num_values, device)
     tdt = {"INT32": torch.int32, "INT64": torch.int64,
            "FLOAT": torch.float32, "DOUBLE": torch.float64}[phys]
@@ -856,7 +910,7 @@ def _decode_chunk_strings(buf, dbuf, ck: ChunkPages, num_values: int, device,
         return Column(dtypes.string, torch.from_numpy(data), vt,
                       torch.from_numpy(offsets))
     validity, prefix = _gpu_validity_prefix(dbuf, ck.pages, num_values, device,
-                                            chunk_nulls)
+                                            chunk_nulls, ck.runs_np)
     idx = _gpu_dict_indices(dbuf, buf, ck, prefix, num_values, device)
     parse_dict_strings(buf, ck)
     dict_offs = torch.from_numpy(ck.dict_str_offs).to(device)
@@ -887,7 +941,8 @@ def _decode_chunk_strings(buf, dbuf, ck: ChunkPages, num_values: int, device,
 
 
 def _decode_chunk_gpu(dbuf: torch.Tensor, pages: List[PageDesc], num_values: int,
-                      phys: str, device, chunk_nulls: bool = True):
+                      phys: str, device, chunk_nulls: bool = True,
+                      runs_np=None):
     """HIP kernel path: RLE def-levels + PLAIN value scatter on device."""
     from . import native
 
@@ -916,11 +971,8 @@ def _decode_chunk_gpu(dbuf: torch.Tensor, pages: List[PageDesc], num_values: int
                                   out.data_ptr(), esize, num_values, sp)
         native.check(rc, "au_pq_copy_plain")
         return out, None
-    validity = torch.empty(num_values, dtype=torch.uint8, device=device)
-    rc = lib.au_pq_rle1(darr.data_ptr(), npages, dbuf.data_ptr(),
-                        validity.data_ptr(), sp)
-    native.check(rc, "au_pq_rle1")
-    prefix = torch.cumsum(validity.to(torch.int64), 0)
+    validity, prefix = _gpu_validity_prefix(dbuf, pages, num_values, device,
+                                            True, runs_np)
     rc = lib.au_pq_scatter(darr.data_ptr(), npages, dbuf.data_ptr(),
                            validity.data_ptr(), prefix.data_ptr(),
                            out.data_ptr(), esize, num_values, sp)
