@@ -30,8 +30,12 @@ def _run(rank, world, port, q):
 
         s = AuronSession()
         cat = Catalog(ROOT, SF)
+        qlist = QUERIES_TO_CHECK
+        env_q = os.environ.get("AURON_DIST_QUERIES")
+        if env_q:
+            qlist = env_q.split(",")
         results = {}
-        for qn in QUERIES_TO_CHECK:
+        for qn in qlist:
             plan = QUERIES[qn](cat, s)
             results[qn] = s.collect_all(plan).to_pydict()
         if rank == 0:
@@ -92,6 +96,59 @@ def test_world2_matches_single_rank():
             def fmt(r):
                 return tuple(f"{v:.6e}" if isinstance(v, float) else v
                              for v in r)
+
+            assert {fmt(r) for r in a} == {fmt(r) for r in b}, qn
+            continue
+        for ra, rb in zip(a, b):
+            for x, y in zip(ra, rb):
+                if isinstance(x, float) and isinstance(y, float):
+                    assert math.isclose(x, y, rel_tol=1e-6, abs_tol=1e-9), (qn, ra, rb)
+                else:
+                    assert x == y, (qn, ra, rb)
+
+
+def test_world4_matches_single_rank():
+    """World=4 over gloo on representative exchange-heavy queries: the
+    collective schedule the 8-GPU driver bench will execute (all-to-all
+    hash exchanges, broadcast gathers, single-rank collects) at a world
+    size with non-trivial partitioning."""
+    from auron_amd import AuronSession
+    from auron_amd.tpcds import datagen
+    from auron_amd.tpcds.queries import QUERIES, Catalog
+
+    qs = ["q3", "q23", "q72", "q59", "q95", "q14", "q75", "q87"]
+    datagen.write_dataset(ROOT, SF)
+    s = AuronSession()
+    cat = Catalog(ROOT, SF)
+    single = {qn: s.collect(QUERIES[qn](cat, s)).to_pydict() for qn in qs}
+
+    import random
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = random.randint(20000, 40000)
+    old = os.environ.get("AURON_DIST_QUERIES")
+    os.environ["AURON_DIST_QUERIES"] = ",".join(qs)
+    try:
+        procs = [ctx.Process(target=_run, args=(r, 4, port, q)) for r in range(4)]
+        for p in procs:
+            p.start()
+        status, dist_results = q.get(timeout=900)
+        for p in procs:
+            p.join(timeout=60)
+    finally:
+        if old is None:
+            os.environ.pop("AURON_DIST_QUERIES", None)
+        else:
+            os.environ["AURON_DIST_QUERIES"] = old
+    assert status == "ok", dist_results
+    for qn in qs:
+        a = _rows(single[qn])
+        b = _rows(dist_results[qn])
+        assert len(a) == len(b), f"{qn}: {len(a)} vs {len(b)} rows"
+        if qn in ("q59",):
+            def fmt(r):
+                return tuple(f"{v:.6e}" if isinstance(v, float) else v for v in r)
 
             assert {fmt(r) for r in a} == {fmt(r) for r in b}, qn
             continue
