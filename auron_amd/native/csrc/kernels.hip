@@ -453,14 +453,21 @@ __global__ void k_bytes_gather(const uint8_t* __restrict__ src,
                                const int64_t* __restrict__ src_start,
                                const int64_t* __restrict__ out_off,
                                uint8_t* __restrict__ out, int64_t nrows) {
-  const int64_t wave = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
-  const int lane = threadIdx.x & 63;
-  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) >> 6;
-  for (int64_t row = wave; row < nrows; row += nwaves) {
+  // thread-per-row: TPC-DS strings average 10-30 bytes, so row-level
+  // parallelism beats wavefront-per-row lane masking; long rows chunk
+  // through an 8-byte loop
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       row < nrows; row += (int64_t)gridDim.x * blockDim.x) {
     const int64_t dst0 = out_off[row];
     const int64_t len = out_off[row + 1] - dst0;
     const int64_t s0 = src_start[row];
-    for (int64_t j = lane; j < len; j += 64) out[dst0 + j] = src[s0 + j];
+    int64_t j = 0;
+    for (; j + 8 <= len; j += 8) {
+      uint64_t w;
+      __builtin_memcpy(&w, src + s0 + j, 8);
+      __builtin_memcpy(out + dst0 + j, &w, 8);
+    }
+    for (; j < len; j++) out[dst0 + j] = src[s0 + j];
   }
 }
 
@@ -468,9 +475,8 @@ AU_EXPORT int au_bytes_gather(const uint8_t* src, const int64_t* src_start,
                               const int64_t* out_off, uint8_t* out,
                               int64_t nrows, void* stream) {
   if (nrows == 0) return 0;
-  int64_t waves_needed = nrows;
-  int64_t blocks = (waves_needed * 64 + 255) / 256;
-  if (blocks > 8192) blocks = 8192;
+  int64_t blocks = (nrows + 255) / 256;
+  if (blocks > 16384) blocks = 16384;
   if (blocks < 1) blocks = 1;
   hipLaunchKernelGGL(k_bytes_gather, dim3((uint32_t)blocks), dim3(256), 0,
                      (hipStream_t)stream, src, src_start, out_off, out, nrows);
