@@ -32,7 +32,7 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=3)
     ap.add_argument("--warmup", type=int, default=1)
-    ap.add_argument("--sf", type=float, default=float(os.environ.get("AURON_BENCH_SF", "1")))
+    ap.add_argument("--sf", type=float, default=float(os.environ.get("AURON_BENCH_SF", "10")))
     ap.add_argument("--queries", type=str, default="all")
     ap.add_argument("--data-root", type=str,
                     default=os.environ.get("AURON_DATA_ROOT",
@@ -115,7 +115,10 @@ def main():
             "ms_per_step": suite_seconds * 1000.0,
             "higher_is_better": False,
             "scaling": "strong",
-            "vs_baseline": None,
+            # Auron total TPC-DS **1TB** = ~1519 s on the
+            # reference's (unspecified CPU) hardware; only an SF=1000 run
+            # here is the same config, so smaller SFs report null
+            "vs_baseline": (suite_seconds / 1519.0) if args.sf >= 1000 else None,
             "dtype": "decimal64/fp64",
             "data": "synthetic",
             "config": {
