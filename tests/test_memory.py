@@ -124,3 +124,47 @@ def test_chunked_partial_agg_spills_and_matches():
     assert set(got) == set(want)
     for k in want:
         assert abs(got[k] - want[k]) < 1e-9 * max(1.0, abs(want[k]))
+
+
+def test_external_sort_buckets_and_spills():
+    """Range-split external sort: bounded-memory path produces globally
+    sorted output and spills idle buckets under a tiny budget."""
+    import torch
+
+    from auron_amd import col, dtypes
+    from auron_amd.column import Column, RecordBatch
+    from auron_amd.engine.executor import ExecContext, Executor
+    from auron_amd.memory import MemManager
+    from auron_amd.plan import nodes as P
+
+    torch.manual_seed(3)
+    batches = []
+    for i in range(8):
+        n = 4000
+        k = torch.randint(-10**9, 10**9, (n,))
+        v = torch.arange(n) + i * n
+        val = torch.where(torch.rand(n) < 0.03, torch.zeros(n, dtype=torch.bool), torch.ones(n, dtype=torch.bool))
+        batches.append(RecordBatch(["k", "v"], [
+            Column(dtypes.int64, k, val), Column(dtypes.int64, v)]))
+    plan = P.Sort(P.MemoryScan(batches), [(col("k"), True)])
+    ctx = ExecContext(memmgr=MemManager(budget_bytes=96 << 10), batch_rows=2000)
+    ex = Executor(ctx)
+    out = ex.collect(plan)
+    assert ctx.metrics.get("sort.external_buckets", 0) > 0
+    assert ctx.memmgr.metrics.get("spill_count", 0) > 0, ctx.memmgr.metrics
+    ks = out.column("k")
+    vals = ks.to_pylist()
+    n_null = sum(1 for v in vals if v is None)
+    assert all(vals[i] is None for i in range(n_null))  # nulls first (asc)
+    nonnull = vals[n_null:]
+    assert nonnull == sorted(nonnull)
+    assert out.num_rows == 8 * 4000
+
+    # descending: nulls last
+    pland = P.Sort(P.MemoryScan(batches), [(col("k"), False)])
+    outd = Executor(ExecContext(memmgr=MemManager(budget_bytes=96 << 10),
+                                batch_rows=2000)).collect(pland)
+    vd = outd.column("k").to_pylist()
+    nn = [v for v in vd if v is not None]
+    assert nn == sorted(nn, reverse=True)
+    assert all(v is None for v in vd[len(nn):])
