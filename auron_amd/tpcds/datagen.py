@@ -635,8 +635,10 @@ def write_dataset(root: str, sf: float, tables: Optional[List[str]] = None,
             if job % world == rank and (force or not os.path.exists(path)):
                 todo.append((t, sf, p, nparts, path))
             job += 1
+    # cap default workers: each in-flight (table, part) job materializes
+    # a multi-GB arrow table, so 256-core boxes must not run 254 at once
     workers = int(os.environ.get("AURON_DATAGEN_WORKERS",
-                                 str(max(1, (os.cpu_count() or 4) - 2))))
+                                 str(max(1, min(32, (os.cpu_count() or 4) - 2)))))
     if len(todo) > 3 and workers > 1:
         # dsdgen-style parallel generation: (table, part) jobs fan out
         # across processes (numpy-only work, no device state involved)
