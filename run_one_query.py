@@ -7,7 +7,8 @@ from auron_amd.tpcds.queries import QUERIES, Catalog
 
 qname = sys.argv[1] if len(sys.argv) > 1 else "q3"
 sf = float(sys.argv[2]) if len(sys.argv) > 2 else 1.0
-root = "data"
+import os
+root = os.environ.get("AURON_DATA_ROOT", "data")
 datagen.write_dataset(root, sf)
 s = AuronSession()
 cat = Catalog(root, sf)
