@@ -34,7 +34,7 @@ class Column:
         self.validity = validity
         self.offsets = offsets
         if dtype.uses_offsets:
-            assert offsets is not None and offsets.dtype == torch.int32
+            assert offsets is not None and offsets.dtype == torch.int64
 
     # ---------------------------------------------------------- basics
     def __len__(self) -> int:
@@ -87,7 +87,7 @@ class Column:
                 offs.append(len(flat))
             data = torch.tensor(flat, dtype=dtype.torch_dtype) if flat else \
                 torch.empty(0, dtype=dtype.torch_dtype)
-            col = Column(dtype, data, validity, torch.tensor(offs, dtype=torch.int32))
+            col = Column(dtype, data, validity, torch.tensor(offs, dtype=torch.int64))
         elif dtype.is_string:
             bufs = []
             offs = [0]
@@ -98,7 +98,7 @@ class Column:
                 total += len(b)
                 offs.append(total)
             data = torch.frombuffer(bytearray(b"".join(bufs)), dtype=torch.uint8) if total else torch.empty(0, dtype=torch.uint8)
-            offsets = torch.tensor(offs, dtype=torch.int32)
+            offsets = torch.tensor(offs, dtype=torch.int64)
             col = Column(dtype, data, validity, offsets)
         else:
             fill = 0
@@ -191,7 +191,7 @@ class Column:
                 row = torch.repeat_interleave(lens)  # maps element pos -> out row
                 el_idx = pos - new_off[:-1][row] + starts[row]
                 out_el = self.data[el_idx]
-            return Column(self.dtype, out_el, validity, new_off.to(torch.int32))
+            return Column(self.dtype, out_el, validity, new_off)
         data = self.data[idx]
         return Column(self.dtype, data, validity, None)
 
@@ -229,7 +229,7 @@ class Column:
                 offs.append(o[:-1] + base if len(offs) else o[:-1] + base)
                 base += int(o[-1].item())
             offs.append(torch.tensor([base], dtype=torch.int64, device=device))
-            offsets = torch.cat(offs).to(torch.int32)
+            offsets = torch.cat(offs).to(torch.int64)
             return Column(dt, data, validity, offsets)
         return Column(dt, torch.cat([c.data for c in cols]), validity, None)
 
@@ -253,16 +253,17 @@ class Column:
             # the arrow table is kept alive by the returned tensors' base)
             arr = arr.combine_chunks() if isinstance(arr, pa.ChunkedArray) else arr
             off_np = np.frombuffer(arr.buffers()[1], dtype=np.int32,
-                                   count=len(arr) + 1, offset=arr.offset * 4)
+                                   count=len(arr) + 1,
+                                   offset=arr.offset * 4).astype(np.int64)
             start = int(off_np[0])
             end = int(off_np[-1])
             buf = arr.buffers()[2]
             if buf is None or end == start:
                 data = torch.empty(0, dtype=torch.uint8)
-                offsets = torch.zeros(len(arr) + 1, dtype=torch.int32)
+                offsets = torch.zeros(len(arr) + 1, dtype=torch.int64)
             else:
                 data = torch.from_numpy(np.frombuffer(buf, dtype=np.uint8, count=end - start, offset=start))
-                offsets = torch.from_numpy(off_np) if start == 0 else torch.from_numpy((off_np - start))
+                offsets = torch.from_numpy(off_np if start == 0 else off_np - start)
             col = Column(dt, data, validity, offsets)
         elif dt.code == dtypes.DECIMAL64:
             d128 = arr.cast(pa.decimal128(dt.precision, dt.scale))

@@ -1279,7 +1279,7 @@ class Executor:
         offsets = torch.zeros(ngroups + 1, dtype=torch.int64, device=device)
         torch.cumsum(lens, 0, out=offsets[1:])
         return Column(dtypes.list_of(val.dtype), val.data[rows], None,
-                      offsets.to(torch.int32))
+                      offsets.to(torch.int64))
 
     def _finalize_agg(self, agg: AggFunc, vdt: DataType, acc, cnt: torch.Tensor) -> Column:
         if agg.fn in ("collect_list", "collect_set"):

@@ -113,7 +113,7 @@ def unpack_batch(meta: dict, buf: torch.Tensor) -> RecordBatch:
             validity = take(cm["val"]).to(torch.bool)
         offsets = None
         if cm["off"] >= 0:
-            offsets = take(cm["off"]).view(torch.int32)
+            offsets = take(cm["off"]).view(torch.int64)
         names.append(cm["name"])
         cols.append(Column(dt, data, validity, offsets))
     return RecordBatch(names, cols)

@@ -194,12 +194,12 @@ class Literal(Expr):
             validity = torch.zeros(n, dtype=torch.bool, device=device)
             if dt.is_string:
                 return Column(dt, torch.empty(0, dtype=torch.uint8, device=device), validity,
-                              torch.zeros(n + 1, dtype=torch.int32, device=device))
+                              torch.zeros(n + 1, dtype=torch.int64, device=device))
             return Column(dt, torch.zeros(n, dtype=dt.torch_dtype, device=device), validity)
         if dt.is_string:
             b = self.value.encode("utf-8")
             data = torch.tensor(list(b), dtype=torch.uint8, device=device).repeat(n) if b else torch.empty(0, dtype=torch.uint8, device=device)
-            offsets = (torch.arange(n + 1, dtype=torch.int32, device=device) * len(b))
+            offsets = (torch.arange(n + 1, dtype=torch.int64, device=device) * len(b))
             return Column(dt, data, None, offsets)
         if dt.code == dtypes.DECIMAL64:
             v = int(round(float(self.value) * 10 ** dt.scale))

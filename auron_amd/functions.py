@@ -420,7 +420,7 @@ class Trim(Expr):
             within = p - new_off[:-1][row]
             src = off[:-1][row] + start[row] + within
             data = c.data[src]
-        return Column(dtypes.string, data, c.validity, new_off.to(torch.int32))
+        return Column(dtypes.string, data, c.validity, new_off.to(torch.int64))
 
 
 @dataclass(eq=False)
@@ -876,7 +876,7 @@ class MakeArray(Expr):
         n = batch.num_rows
         first = cols[0]
         data = torch.stack([c.data for c in cols], dim=1).reshape(-1)
-        offsets = (torch.arange(n + 1, dtype=torch.int32, device=batch.device)
+        offsets = (torch.arange(n + 1, dtype=torch.int64, device=batch.device)
                    * k)
         return Column(dtypes.list_of(first.dtype), data,
                       combine_validity(*cols), offsets)

@@ -28,7 +28,7 @@ def _batch_bytes(b: RecordBatch) -> int:
         if c.validity is not None:
             total += c.validity.numel()
         if c.offsets is not None:
-            total += c.offsets.numel() * 4
+            total += c.offsets.numel() * 8
     return total
 
 

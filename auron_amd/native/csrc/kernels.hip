@@ -36,7 +36,7 @@ enum AuDType : int32_t {
 
 struct AuColDesc {
   const void* data;        // typed buffer; for STRING: uint8 bytes
-  const int32_t* offsets;  // STRING only, n+1
+  const int64_t* offsets;  // STRING only, n+1 (LargeString: 64-bit)
   const uint8_t* validity; // bool bytes (1 = valid) or nullptr
   int32_t dtype;
   int32_t scale;
@@ -128,9 +128,9 @@ __device__ uint32_t hash_one(const AuColDesc& c, int64_t i, uint32_t seed) {
       return hash_long((uint64_t)__double_as_longlong(d), seed);
     }
     case AU_STRING: {
-      int32_t s = c.offsets[i];
-      int32_t e = c.offsets[i + 1];
-      return hash_bytes((const uint8_t*)c.data + s, e - s, seed);
+      int64_t s = c.offsets[i];
+      int64_t e = c.offsets[i + 1];
+      return hash_bytes((const uint8_t*)c.data + s, (int32_t)(e - s), seed);
     }
   }
   return seed;
@@ -191,12 +191,12 @@ __device__ bool keys_equal(const AuColDesc* a_cols, int64_t ai,
         if (((const double*)a.data)[ai] != ((const double*)b.data)[bi]) return false;
         break;
       case AU_STRING: {
-        int32_t as = a.offsets[ai], ae = a.offsets[ai + 1];
-        int32_t bs = b.offsets[bi], be = b.offsets[bi + 1];
+        int64_t as = a.offsets[ai], ae = a.offsets[ai + 1];
+        int64_t bs = b.offsets[bi], be = b.offsets[bi + 1];
         if (ae - as != be - bs) return false;
         const uint8_t* ap = (const uint8_t*)a.data + as;
         const uint8_t* bp = (const uint8_t*)b.data + bs;
-        int32_t len = ae - as;
+        int32_t len = (int32_t)(ae - as);
         int32_t k = 0;
         for (; k + 8 <= len; k += 8) {
           uint64_t wa, wb;

@@ -440,7 +440,7 @@ def decode_chunk_np_dict(buf: np.ndarray, ck: ChunkPages, num_values: int,
         vmask = validity.astype(bool) if validity is not None else np.ones(num_values, dtype=bool)
         row_idx[vmask] = idx
         lens = np.where(vmask, ck.dict_str_lens[row_idx], 0)
-        offsets = np.zeros(num_values + 1, dtype=np.int32)
+        offsets = np.zeros(num_values + 1, dtype=np.int64)
         np.cumsum(lens, out=offsets[1:])
         total = int(offsets[-1])
         data = np.empty(total, dtype=np.uint8)
@@ -1002,7 +1002,7 @@ def _decode_chunk_strings(buf, dbuf, ck: ChunkPages, num_values: int, device,
     else:
         data = torch.empty(0, dtype=torch.uint8, device=device)
     vt = vmask if (vmask is not None and chunk_nulls) else None
-    return Column(dtypes.string, data, vt, offsets.to(torch.int32))
+    return Column(dtypes.string, data, vt, offsets.to(torch.int64))
 
 
 def _decode_chunk_gpu(dbuf: torch.Tensor, pages: List[PageDesc], num_values: int,

@@ -194,7 +194,7 @@ def substr(c: Column, start: int, length: int) -> Column:
         pos = torch.arange(total, dtype=torch.int64, device=c.device)
         byte_idx = pos - new_off[:-1][row] + off[:-1][row] + s0
         data = c.data[byte_idx]
-    return Column(dtypes.string, data, c.validity, new_off.to(torch.int32))
+    return Column(dtypes.string, data, c.validity, new_off.to(torch.int64))
 
 
 def concat(cols: List[Column]) -> Column:
@@ -226,7 +226,7 @@ def concat(cols: List[Column]) -> Column:
     for c in cols:
         if c.validity is not None:
             validity = c.validity if validity is None else (validity & c.validity)
-    return Column(dtypes.string, data, validity, new_off.to(torch.int32))
+    return Column(dtypes.string, data, validity, new_off.to(torch.int64))
 
 
 def upper(c: Column) -> Column:
