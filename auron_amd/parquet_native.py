@@ -515,7 +515,7 @@ def _dbuf_cache_cap() -> int:
         return int(env)
     if torch.cuda.is_available():
         _, total = torch.cuda.mem_get_info()
-        return int(total * 0.45)
+        return int(total * 0.30)
     return 0
 
 
@@ -524,6 +524,12 @@ def _dbuf_cache_get(key):
     if ent is not None:
         _DBUF_CACHE.move_to_end(key)
     return ent
+
+
+def dbuf_cache_clear():
+    """Drop all cached staged bytes (OOM-pressure escape hatch)."""
+    _DBUF_CACHE.clear()
+    _DBUF_BYTES[0] = 0
 
 
 def _dbuf_cache_put(key, dbuf):

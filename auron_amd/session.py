@@ -8,9 +8,9 @@ runs physical plans on it.
 """
 from __future__ import annotations
 
+import torch
 from typing import Optional
 
-import torch
 import torch.distributed as dist
 
 from .column import RecordBatch
@@ -64,8 +64,24 @@ class AuronSession:
             raise AuronTaskError(task_id, stage_id, partition, e) from e
 
     def collect(self, plan: P.PlanNode) -> RecordBatch:
-        """Run and concat this rank's result (driver-side rows analogue)."""
-        return self.executor.collect(plan)
+        """Run and concat this rank's result (driver-side rows analogue).
+
+        On device OOM the query retries once with every cross-query cache
+        dropped (staged-bytes HBM cache, broadcast relations) — the
+        memmgr's spill tiers handle registered holders, these caches are
+        the unregistered residents."""
+        try:
+            return self.executor.collect(plan)
+        except torch.cuda.OutOfMemoryError:
+            if self.ctx.device.type != "cuda":
+                raise
+            from . import parquet_native
+
+            parquet_native.dbuf_cache_clear()
+            self.ctx.broadcast_cache.clear()
+            self.ctx.broadcast_cache_bytes = 0
+            torch.cuda.empty_cache()
+            return self.executor.collect(plan)
 
     def collect_all(self, plan: P.PlanNode) -> RecordBatch:
         """Gather the full result on every rank (for result checking)."""

@@ -19,6 +19,10 @@ import os
 import sys
 import time
 
+import os as _os
+
+_os.environ.setdefault("PYTORCH_ALLOC_CONF", "expandable_segments:True")
+
 import torch
 import torch.distributed as dist
 
