@@ -74,6 +74,13 @@ class ExecContext:
         return _T()
 
 
+# agg fns whose state is mergeable across chunks (everything _merge_states
+# can combine); count_distinct / collect_* need the whole group resident
+_SPLITTABLE_AGGS = frozenset(
+    {"sum", "avg", "min", "max", "count", "count_star",
+     "first", "first_ignores_null"})
+
+
 def _normalize_join_keys(lkeys, rkeys):
     """Promote corresponding key pairs to one dtype: the native kernels
     hash and compare by the build side's dtype, so int32-vs-int64 pairs
@@ -225,10 +232,172 @@ class Executor:
     def collect(self, node: P.PlanNode) -> RecordBatch:
         return _concat(self.execute(node))
 
+    # ------------------------------------------------------ streaming pull
+    def execute_iter(self, node: P.PlanNode):
+        """Lazy batch stream for streaming consumers (the chunked agg).
+
+        Operators with an _iter_ form yield batches as they are produced
+        (per scan file, per probe batch, per Expand projection), so a huge
+        intermediate — a fact-scan join output or an Expand'ed rollup
+        input, billions of rows at SF>=100 — is never resident all at
+        once; each batch is dropped as soon as the consumer folds it into
+        its aggregation state. Operators without an _iter_ form fall back
+        to the materializing execute() (which records metrics normally).
+        """
+        node = self._rewrite(node)
+        fn = getattr(self, f"_iter_{type(node).__name__}", None)
+        if fn is None:
+            yield from self.execute(node)
+        else:
+            yield from fn(node)
+
+    def _iter_Filter(self, node: P.Filter):
+        from ..exprs import eval_scope
+
+        for b in self.execute_iter(node.child):
+            with eval_scope(b):
+                c = node.predicate.eval(b)
+            mask = c.data.bool()
+            if c.validity is not None:
+                mask = mask & c.validity
+            yield b.filter(mask)
+
+    def _iter_Project(self, node: P.Project):
+        from ..exprs import eval_scope
+
+        for b in self.execute_iter(node.child):
+            with eval_scope(b):
+                cols = [a.expr.eval(b) for a in node.exprs]
+            yield RecordBatch([a.name for a in node.exprs], cols)
+
+    def _iter_RenameColumns(self, node: P.RenameColumns):
+        for b in self.execute_iter(node.child):
+            yield RecordBatch(node.names, b.columns)
+
+    def _iter_Expand(self, node: P.Expand):
+        # one output batch per (input batch x projection): the G-way row
+        # multiplication of ROLLUP never materializes in full
+        for b in self.execute_iter(node.child):
+            for proj in node.projections:
+                cols = [a.expr.eval(b) for a in proj]
+                yield RecordBatch([a.name for a in proj], cols)
+
+    def _iter_Union(self, node: P.Union):
+        names = None
+        for ch in node.inputs:
+            for b in self.execute_iter(ch):
+                if names is None:
+                    names = b.names
+                yield RecordBatch(names, b.columns)
+
+    def _iter_ParquetScan(self, node: P.ParquetScan):
+        key = (tuple(node.paths), tuple(node.columns or ()), repr(node.filters))
+        hit = self._scan_cache.get(key)
+        if hit is not None:
+            yield from hit
+            return
+        my_files = node.paths[self.ctx.rank::self.ctx.world_size]
+        if len(my_files) <= 1:
+            # single-file scans keep the cached exec path (dimension
+            # tables are re-read by many joins within one query)
+            yield from self._exec_ParquetScan(node)
+            return
+        # bounded prefetch: the pool decodes file i+1 while the consumer
+        # folds file i; never caches (a streamed fact scan at SF>=100 is
+        # exactly what must not be retained)
+        from collections import deque
+
+        pool = self._scan_pool()
+        pending: deque = deque()
+        idx = 0
+        got = 0
+        while idx < len(my_files) or pending:
+            while idx < len(my_files) and len(pending) < 2:
+                pending.append(pool.submit(
+                    self._read_parquet_tolerant, my_files[idx],
+                    node.columns, node.filters))
+                idx += 1
+            b = pending.popleft().result()
+            if b is not None:
+                got += 1
+                yield b
+        if not got:
+            yield from self._exec_ParquetScan(node)
+
+    def _iter_HashJoin(self, node: P.HashJoin):
+        """Stream the probe side of a build-right join: the build relation
+        and its hash table are constructed once (reusing the broadcast
+        cache), then each probe batch joins and is yielded immediately.
+        Valid when every probe row is decided independently — inner/left
+        (+ residual); other shapes fall back to the materializing path."""
+        if node.build_side != "right" or node.how not in ("inner", "left"):
+            yield from self.execute(node)
+            return
+        cache_entry = None
+        if node.broadcast:
+            cache_entry = self._broadcast_cache_entry(node.right)
+        if cache_entry is not None and cache_entry.get("batch") is not None:
+            right = cache_entry["batch"]
+        else:
+            rb = _concat(self.execute(node.right))
+            if node.broadcast and self.ctx.world_size > 1:
+                right = _concat(all_gather_batch(rb, self.ctx.device, self.ctx.group))
+            else:
+                right = rb
+            if cache_entry is not None:
+                cache_entry["batch"] = right
+                self.ctx.broadcast_cache_bytes += sum(
+                    c.data.numel() * c.data.element_size() for c in right.columns)
+        rkeys = [k.eval(right) for k in node.right_keys]
+        table = None
+        sig = None
+        if cache_entry is not None:
+            sig = _expr_sig(node.right_keys)
+            table = cache_entry["tables"].get(sig)
+        normalized = False
+        for left in self.execute_iter(node.left):
+            lkeys = [k.eval(left) for k in node.left_keys]
+            lkeys, rkeys = _normalize_join_keys(lkeys, rkeys)
+            if not normalized:
+                normalized = True
+                if table is None:
+                    table = ops.join_build(rkeys)
+                    if table is not None and cache_entry is not None:
+                        cache_entry["tables"][sig] = table
+                        self.ctx.broadcast_cache_bytes += table.nbytes
+            if node.residual is not None:
+                bi, pi, _ = ops.hash_join(rkeys, lkeys,
+                                          emit_unmatched_probe=False,
+                                          need_build_matched=False,
+                                          table=table)
+                yield from self._finish_join_pairs(left, right, pi, bi,
+                                                   node.how, node.residual,
+                                                   node.existence_col)
+            else:
+                bi, pi, _ = ops.hash_join(
+                    rkeys, lkeys,
+                    emit_unmatched_probe=(node.how == "left"),
+                    need_build_matched=False,
+                    table=table)
+                out_left = left.gather(pi, may_have_negative=True)
+                out_right = right.gather(bi, may_have_negative=True)
+                yield RecordBatch(out_left.names + out_right.names,
+                                  out_left.columns + out_right.columns)
+
     # ---------------------------------------------------------------- scans
     def _exec_MemoryScan(self, node: P.MemoryScan) -> List[RecordBatch]:
         assert node.batches
         return [b.to(self.ctx.device) for b in node.batches]
+
+    def _read_parquet_tolerant(self, path: str, columns, filters):
+        try:
+            return self._read_parquet(path, columns, filters)
+        except Exception:
+            from ..config import IGNORE_CORRUPTED_FILES, AuronConf
+
+            if AuronConf().get(IGNORE_CORRUPTED_FILES):
+                return None  # conf.rs IGNORE_CORRUPTED_FILES semantics
+            raise
 
     def _read_parquet(self, path: str, columns, filters) -> RecordBatch:
         import pyarrow.parquet as pq
@@ -269,15 +438,9 @@ class Executor:
             # keep the >=1 batch invariant: 0-row batch with the file schema
             t = pq.read_table(node.paths[0], columns=node.columns).slice(0, 0)
             return [RecordBatch.from_arrow(t, self.ctx.device)]
-        def read_one(f):
-            try:
-                return self._read_parquet(f, node.columns, node.filters)
-            except Exception:
-                from ..config import IGNORE_CORRUPTED_FILES, AuronConf
 
-                if AuronConf().get(IGNORE_CORRUPTED_FILES):
-                    return None  # conf.rs IGNORE_CORRUPTED_FILES semantics
-                raise
+        def read_one(f):
+            return self._read_parquet_tolerant(f, node.columns, node.filters)
 
         if len(my_files) == 1:
             out = [read_one(my_files[0])]
@@ -907,6 +1070,21 @@ class Executor:
             # can spill under pressure — the whole input never has to be
             # resident at once
             return self._exec_hash_agg_partial_chunked(node)
+        if (node.mode == "complete" and AuronConf().get(AGG_STREAMING)
+                and all(a.fn in _SPLITTABLE_AGGS for a in node.aggs)
+                and (node.keys or self.ctx.world_size == 1)):
+            # (keyless complete at W>1 keeps the one-shot path: its
+            # emit-null-row-only-on-rank-0 rule lives in _hash_agg_body)
+            # W=1 collapses partial+exchange+final towers to one complete
+            # agg (see _rewrite) — stream it anyway: chunk the child into
+            # partial states, then run ONE final pass over the merged
+            # states. Same composition the distributed path uses, so a
+            # rollup over a SF>=100 fact join never materializes.
+            part = P.HashAgg(node.child, node.keys, node.aggs, mode="partial")
+            states = _concat(self._exec_hash_agg_partial_chunked(part))
+            fin = P.HashAgg(node.child, node.keys, node.aggs, mode="final")
+            with eval_scope(states):
+                return self._hash_agg_body(fin, states)
         b = _concat(self.execute(node.child))
         with eval_scope(b):
             return self._hash_agg_body(node, b)
@@ -914,7 +1092,8 @@ class Executor:
     def _exec_hash_agg_partial_chunked(self, node: P.HashAgg) -> List[RecordBatch]:
         from ..exprs import eval_scope
 
-        batches = self.execute(node.child)
+        batches = self.execute_iter(node.child)
+        schema_batch: Optional[RecordBatch] = None
         limit = self.ctx.batch_rows
         acc: List[RecordBatch] = []
         acc_rows = 0
@@ -939,6 +1118,8 @@ class Executor:
             holder = self.ctx.memmgr.register("agg-partial-state", acc)
 
         for b in batches:
+            if schema_batch is None:
+                schema_batch = b.slice(0, 0)
             for lo in range(0, max(b.num_rows, 1), limit):
                 chunk = b if b.num_rows <= limit else \
                     b.slice(lo, min(limit, b.num_rows - lo))
@@ -955,9 +1136,11 @@ class Executor:
         result = holder.batches()
         holder.release()
         if not result:
-            # preserve the empty-input partial schema
-            with eval_scope(_concat(batches)):
-                return self._hash_agg_body(node, _concat(batches))
+            # preserve the empty-input partial schema (the child stream is
+            # consumed; a 0-row slice of its first batch carries the schema)
+            assert schema_batch is not None, "child stream yielded no batches"
+            with eval_scope(schema_batch):
+                return self._hash_agg_body(node, schema_batch)
         return result
 
     def _merge_states(self, node: P.HashAgg, b: RecordBatch) -> RecordBatch:

@@ -539,6 +539,19 @@ def _dbuf_cache_put(key, dbuf):
     nbytes = dbuf.numel()
     if nbytes > cap:
         return
+    # allocation-pressure guard: the cache must never be what pushes a
+    # query over the edge — evict (and refuse to insert) whenever free
+    # HBM would drop below 25% of the device, so big working sets
+    # (SF>=100 joins/aggs) reclaim the cache before the allocator OOMs
+    if torch.cuda.is_available():
+        free, total = torch.cuda.mem_get_info()
+        reserve = int(total * 0.25)
+        while free - nbytes < reserve and _DBUF_CACHE:
+            _, old = _DBUF_CACHE.popitem(last=False)
+            _DBUF_BYTES[0] -= old.numel()
+            free += old.numel()
+        if free - nbytes < reserve:
+            return
     _DBUF_CACHE[key] = dbuf
     _DBUF_BYTES[0] += nbytes
     while _DBUF_BYTES[0] > cap and _DBUF_CACHE:

@@ -84,15 +84,21 @@ def main():
     session = AuronSession(device=device)
     cat = Catalog(args.data_root, args.sf)
 
+    trace = os.environ.get("AURON_BENCH_TRACE", "") == "1"
+
     def run_suite():
         per_q = {}
         for qn in qnames:
+            if trace:
+                log(f"start {qn}")
             tq = time.perf_counter()
             plan = QUERIES[qn](cat, session)
             res = session.collect(plan)
             if have_gpu:
                 torch.cuda.synchronize()
             per_q[qn] = time.perf_counter() - tq
+            if trace:
+                log(f"done  {qn} {per_q[qn]:.2f}s")
         return per_q
 
     def barrier_sync():
