@@ -59,6 +59,10 @@ BROADCAST_CACHE_BYTES = _opt("spark.auron.broadcast.cache.maxBytes", 4 << 30, in
 FORCE_SHUFFLED_HASH_JOIN = _opt("spark.auron.forceShuffledHashJoin", False, bool,
                                 "never broadcast: hash-exchange both sides",
                                 env="AURON_FORCE_SHJ")
+EXPR_FUSION = _opt("spark.auron.expr.fusion.enable", True, bool,
+                   "compile project/filter scalar expression trees into one "
+                   "fused interpreter kernel launch",
+                   env="AURON_EXPR_FUSION")
 AGG_STREAMING = _opt("spark.auron.agg.streaming.enable", True, bool,
                      "chunked spill-capable partial aggregation",
                      env="AURON_AGG_STREAMING")

@@ -146,6 +146,11 @@ class Executor:
         # same (paths, columns) many times for independent subaggregates;
         # decode once per query (cleared at each top-level execute)
         self._scan_cache: Dict[tuple, List[RecordBatch]] = {}
+        # compiled fused-expression programs per plan node (expr_fusion):
+        # keyed by id(node) + input schema signature, cleared per query so
+        # recycled ids can never resolve to a stale program
+        self._fused_progs: Dict[int, tuple] = {}
+        self._scan_cache_bytes = 0
         # MetricNode tree of the last top-level execute (SQLMetrics parity)
         self._metric_stack: List[list] = [[]]
         self.last_metric_tree: Optional[dict] = None
@@ -199,6 +204,8 @@ class Executor:
         top = len(self._child_time) == 1
         if top:  # top-level call = one query
             self._scan_cache.clear()
+            self._scan_cache_bytes = 0
+            self._fused_progs.clear()
         node = self._rewrite(node)
         name = type(node).__name__
         fn = getattr(self, f"_exec_{name}", None)
@@ -252,23 +259,13 @@ class Executor:
             yield from fn(node)
 
     def _iter_Filter(self, node: P.Filter):
-        from ..exprs import eval_scope
-
         for b in self.execute_iter(node.child):
-            with eval_scope(b):
-                c = node.predicate.eval(b)
-            mask = c.data.bool()
-            if c.validity is not None:
-                mask = mask & c.validity
-            yield b.filter(mask)
+            yield b.filter(self._eval_predicate(node, b))
 
     def _iter_Project(self, node: P.Project):
-        from ..exprs import eval_scope
-
         for b in self.execute_iter(node.child):
-            with eval_scope(b):
-                cols = [a.expr.eval(b) for a in node.exprs]
-            yield RecordBatch([a.name for a in node.exprs], cols)
+            yield RecordBatch([a.name for a in node.exprs],
+                              self._project_cols(node, b))
 
     def _iter_RenameColumns(self, node: P.RenameColumns):
         for b in self.execute_iter(node.child):
@@ -303,14 +300,21 @@ class Executor:
             yield from self._exec_ParquetScan(node)
             return
         # bounded prefetch: the pool decodes file i+1 while the consumer
-        # folds file i; never caches (a streamed fact scan at SF>=100 is
-        # exactly what must not be retained)
+        # folds file i. Batches are retained for the per-query scan cache
+        # only while they fit the cache budget — re-scanned tables (q9's
+        # 15 subaggregates) decode once, while an over-budget fact scan at
+        # SF>=100 streams through without being retained.
         from collections import deque
+
+        from ..memory import _batch_bytes
 
         pool = self._scan_pool()
         pending: deque = deque()
         idx = 0
         got = 0
+        retain: Optional[List[RecordBatch]] = []
+        retain_bytes = 0
+        budget = self._scan_cache_budget() - self._scan_cache_bytes
         while idx < len(my_files) or pending:
             while idx < len(my_files) and len(pending) < 2:
                 pending.append(pool.submit(
@@ -320,9 +324,18 @@ class Executor:
             b = pending.popleft().result()
             if b is not None:
                 got += 1
+                if retain is not None:
+                    retain_bytes += _batch_bytes(b)
+                    if retain_bytes <= budget:
+                        retain.append(b)
+                    else:
+                        retain = None
                 yield b
         if not got:
             yield from self._exec_ParquetScan(node)
+        elif retain is not None:
+            self._scan_cache[key] = retain
+            self._scan_cache_bytes += retain_bytes
 
     def _iter_HashJoin(self, node: P.HashJoin):
         """Stream the probe side of a build-right join: the build relation
@@ -421,13 +434,30 @@ class Executor:
         t = pq.read_table(path, columns=columns, filters=filters)
         return RecordBatch.from_arrow(t, self.ctx.device)
 
+    def _scan_cache_budget(self) -> int:
+        env = os.environ.get("AURON_SCAN_CACHE_BYTES")
+        if env is not None:
+            return int(env)
+        if self.ctx.device.type == "cuda":
+            _, total = torch.cuda.mem_get_info()
+            return int(total * 0.15)
+        return 4 << 30
+
+    def _scan_cache_put(self, key, batches: List[RecordBatch]):
+        from ..memory import _batch_bytes
+
+        nb = sum(_batch_bytes(b) for b in batches)
+        if self._scan_cache_bytes + nb <= self._scan_cache_budget():
+            self._scan_cache[key] = batches
+            self._scan_cache_bytes += nb
+
     def _exec_ParquetScan(self, node: P.ParquetScan) -> List[RecordBatch]:
         key = (tuple(node.paths), tuple(node.columns or ()), repr(node.filters))
         hit = self._scan_cache.get(key)
         if hit is not None:
             return hit
         out = self._exec_parquet_scan_uncached(node)
-        self._scan_cache[key] = out
+        self._scan_cache_put(key, out)
         return out
 
     def _exec_parquet_scan_uncached(self, node: P.ParquetScan) -> List[RecordBatch]:
@@ -512,27 +542,75 @@ class Executor:
         return out
 
     # ------------------------------------------------------- row operators
-    def _exec_Filter(self, node: P.Filter) -> List[RecordBatch]:
+    def _fused_prog(self, node, exprs, batch):
+        """Compiled fused-expression program for this node against this
+        batch schema (None = not compilable / fusion off / CPU)."""
+        if batch.device.type != "cuda":
+            return None
+        from ..config import EXPR_FUSION
+        from .. import native
+
+        if not AuronConf().get(EXPR_FUSION) or not native.available():
+            return None
+        sig = tuple((nm, c.dtype.code, c.dtype.scale)
+                    for nm, c in zip(batch.names, batch.columns))
+        ent = self._fused_progs.get(id(node))
+        if ent is not None and ent[0] == sig:
+            return ent[1]
+        from .. import fused
+
+        schema = {nm: c.dtype for nm, c in zip(batch.names, batch.columns)}
+        prog = fused.compile_exprs(exprs, schema)
+        self._fused_progs[id(node)] = (sig, prog)
+        return prog
+
+    def _eval_predicate(self, node, b):
         from ..exprs import eval_scope
 
-        out = []
-        for b in self.execute(node.child):
+        prog = self._fused_prog(node, [node.predicate], b)
+        if prog is not None:
+            from .. import fused
+
+            c = fused.run(prog, b)[0]
+        else:
             with eval_scope(b):
                 c = node.predicate.eval(b)
-            mask = c.data.bool()
-            if c.validity is not None:
-                mask = mask & c.validity
-            out.append(b.filter(mask))
+        mask = c.data.bool()
+        if c.validity is not None:
+            mask = mask & c.validity
+        return mask
+
+    def _project_cols(self, node, b):
+        from ..exprs import eval_scope
+
+        prog = self._fused_prog(node, [a.expr for a in node.exprs], b)
+        if prog is None:
+            with eval_scope(b):
+                return [a.expr.eval(b) for a in node.exprs]
+        from .. import fused
+
+        cols: List[Optional[Column]] = [None] * len(node.exprs)
+        fused_cols = fused.run(prog, b)
+        for k, i in enumerate(prog.expr_idx):
+            cols[i] = fused_cols[k]
+        if any(c is None for c in cols):
+            with eval_scope(b):
+                for i, a in enumerate(node.exprs):
+                    if cols[i] is None:
+                        cols[i] = a.expr.eval(b)
+        return cols
+
+    def _exec_Filter(self, node: P.Filter) -> List[RecordBatch]:
+        out = []
+        for b in self.execute(node.child):
+            out.append(b.filter(self._eval_predicate(node, b)))
         return out
 
     def _exec_Project(self, node: P.Project) -> List[RecordBatch]:
-        from ..exprs import eval_scope
-
         out = []
         for b in self.execute(node.child):
-            with eval_scope(b):
-                cols = [a.expr.eval(b) for a in node.exprs]
-            out.append(RecordBatch([a.name for a in node.exprs], cols))
+            out.append(RecordBatch([a.name for a in node.exprs],
+                                   self._project_cols(node, b)))
         return out
 
     def _exec_RenameColumns(self, node: P.RenameColumns) -> List[RecordBatch]:
@@ -1076,12 +1154,28 @@ class Executor:
             # (keyless complete at W>1 keeps the one-shot path: its
             # emit-null-row-only-on-rank-0 rule lives in _hash_agg_body)
             # W=1 collapses partial+exchange+final towers to one complete
-            # agg (see _rewrite) — stream it anyway: chunk the child into
-            # partial states, then run ONE final pass over the merged
-            # states. Same composition the distributed path uses, so a
-            # rollup over a SF>=100 fact join never materializes.
+            # agg (see _rewrite) — stream it when the input is large:
+            # chunk the child into partial states, then run ONE final pass
+            # over the merged states (the distributed composition), so a
+            # rollup over a SF>=100 fact join never materializes. Small
+            # inputs (peeked below one chunk) keep the one-pass grouping.
+            import itertools
+
+            it = self.execute_iter(node.child)
+            buf: List[RecordBatch] = []
+            rows = 0
+            for b in it:
+                buf.append(b)
+                rows += b.num_rows
+                if rows > self.ctx.batch_rows:
+                    break
+            else:
+                bb = _concat(buf)
+                with eval_scope(bb):
+                    return self._hash_agg_body(node, bb)
             part = P.HashAgg(node.child, node.keys, node.aggs, mode="partial")
-            states = _concat(self._exec_hash_agg_partial_chunked(part))
+            states = _concat(self._exec_hash_agg_partial_chunked(
+                part, batches=itertools.chain(buf, it)))
             fin = P.HashAgg(node.child, node.keys, node.aggs, mode="final")
             with eval_scope(states):
                 return self._hash_agg_body(fin, states)
@@ -1089,10 +1183,12 @@ class Executor:
         with eval_scope(b):
             return self._hash_agg_body(node, b)
 
-    def _exec_hash_agg_partial_chunked(self, node: P.HashAgg) -> List[RecordBatch]:
+    def _exec_hash_agg_partial_chunked(self, node: P.HashAgg,
+                                       batches=None) -> List[RecordBatch]:
         from ..exprs import eval_scope
 
-        batches = self.execute_iter(node.child)
+        if batches is None:
+            batches = self.execute_iter(node.child)
         schema_batch: Optional[RecordBatch] = None
         limit = self.ctx.batch_rows
         acc: List[RecordBatch] = []

@@ -13,7 +13,8 @@ HERE = os.path.dirname(os.path.abspath(__file__))
 CSRC = os.path.join(HERE, "csrc")
 OUT = os.path.join(HERE, "libauron_hip.so")
 SOURCES = [os.path.join(CSRC, "kernels.hip"), os.path.join(CSRC, "parquet.hip"),
-           os.path.join(CSRC, "agg.hip")]
+           os.path.join(CSRC, "agg.hip"),
+           os.path.join(CSRC, "fused.hip")]
 HIPCC = os.environ.get("HIPCC", "hipcc")
 ARCH = os.environ.get("AURON_OFFLOAD_ARCH", "gfx950")
 
