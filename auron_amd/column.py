@@ -280,6 +280,9 @@ class Column:
         else:
             if dt.code == dtypes.DATE32:
                 arr = arr.cast(pa.int32())
+            if dt.code == dtypes.TIMESTAMP:
+                # normalize any unit to int64 microseconds
+                arr = arr.cast(pa.timestamp("us")).cast(pa.int64())
             if arr.null_count:
                 arr = arr.fill_null(0)  # validity carries the null mask
             np_arr = arr.to_numpy(zero_copy_only=arr.type != pa.bool_())
@@ -314,6 +317,9 @@ class Column:
             p = max(self.dtype.precision, sc + 1, 1)
             return pa.array(vals, type=pa.decimal128(p, sc))
         np_arr = c.data.numpy()
+        if self.dtype.code == dtypes.TIMESTAMP:
+            return pa.array(np_arr, from_pandas=False, mask=mask).cast(
+                pa.timestamp("us"))
         return pa.array(np_arr, from_pandas=False, mask=mask)
 
     def __repr__(self) -> str:  # pragma: no cover

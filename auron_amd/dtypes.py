@@ -22,6 +22,7 @@ DATE32 = 7  # days since epoch, int32 storage
 STRING = 8  # uint8 byte buffer + int32 offsets
 DECIMAL64 = 9  # scaled int64
 LIST = 10  # flattened child values + int32 offsets (element nulls: round 2)
+TIMESTAMP = 11  # microseconds since epoch UTC, int64 storage
 
 _NAMES = {
     BOOL: "bool",
@@ -35,6 +36,7 @@ _NAMES = {
     STRING: "string",
     DECIMAL64: "decimal64",
     LIST: "list",
+    TIMESTAMP: "timestamp",
 }
 
 _TORCH = {
@@ -49,6 +51,7 @@ _TORCH = {
     STRING: torch.uint8,  # byte buffer
     DECIMAL64: torch.int64,
     LIST: None,  # resolved via .child
+    TIMESTAMP: torch.int64,
 }
 
 
@@ -116,6 +119,7 @@ float32 = DataType(FLOAT32)
 float64 = DataType(FLOAT64)
 date32 = DataType(DATE32)
 string = DataType(STRING)
+timestamp = DataType(TIMESTAMP)
 
 
 def decimal64(precision: int = 18, scale: int = 2) -> DataType:
@@ -149,6 +153,8 @@ def from_arrow(at) -> DataType:
         return date32
     if pa.types.is_string(at) or pa.types.is_large_string(at):
         return string
+    if pa.types.is_timestamp(at):
+        return timestamp
     if pa.types.is_decimal(at):
         return decimal64(at.precision, at.scale)
     if pa.types.is_dictionary(at):
@@ -172,4 +178,6 @@ def to_arrow(dt: DataType):
     }
     if dt.code == DECIMAL64:
         return pa.decimal128(dt.precision, dt.scale)
+    if dt.code == TIMESTAMP:
+        return pa.timestamp("us")
     return m[dt.code]

@@ -343,7 +343,13 @@ class Executor:
         cache), then each probe batch joins and is yielded immediately.
         Valid when every probe row is decided independently — inner/left
         (+ residual); other shapes fall back to the materializing path."""
-        if node.build_side != "right" or node.how not in ("inner", "left"):
+        if (node.build_side != "right" or node.how not in ("inner", "left")
+                or (self.ctx.device.type != "cuda" and not os.environ.get(
+                    "AURON_FORCE_STREAM_JOIN"))):
+            # on CPU, hash_join runs the host reference impl, which would
+            # rebuild the build-side table per probe batch; materialize
+            # instead (AURON_FORCE_STREAM_JOIN=1 keeps the streaming path
+            # testable on CPU)
             yield from self.execute(node)
             return
         cache_entry = None
@@ -1162,20 +1168,14 @@ class Executor:
             import itertools
 
             it = self.execute_iter(node.child)
-            buf: List[RecordBatch] = []
-            rows = 0
-            for b in it:
-                buf.append(b)
-                rows += b.num_rows
-                if rows > self.ctx.batch_rows:
-                    break
-            else:
+            buf, rest = self._peek_stream(it)
+            if rest is None:
                 bb = _concat(buf)
                 with eval_scope(bb):
                     return self._hash_agg_body(node, bb)
             part = P.HashAgg(node.child, node.keys, node.aggs, mode="partial")
             states = _concat(self._exec_hash_agg_partial_chunked(
-                part, batches=itertools.chain(buf, it)))
+                part, batches=itertools.chain(buf, rest)))
             fin = P.HashAgg(node.child, node.keys, node.aggs, mode="final")
             with eval_scope(states):
                 return self._hash_agg_body(fin, states)
@@ -1183,12 +1183,36 @@ class Executor:
         with eval_scope(b):
             return self._hash_agg_body(node, b)
 
+    def _peek_stream(self, it, threshold: Optional[int] = None):
+        """Pull from a batch iterator until `threshold` rows are buffered.
+        Returns (buffered, rest_iterator) — rest is None when the stream
+        was exhausted under the threshold (small input: the caller should
+        take the cheaper one-shot path; the chunk/merge machinery only
+        pays for itself when the input is genuinely large)."""
+        if threshold is None:
+            threshold = 8 * self.ctx.batch_rows
+        buf: List[RecordBatch] = []
+        rows = 0
+        for b in it:
+            buf.append(b)
+            rows += b.num_rows
+            if rows > threshold:
+                return buf, it
+        return buf, None
+
     def _exec_hash_agg_partial_chunked(self, node: P.HashAgg,
                                        batches=None) -> List[RecordBatch]:
+        import itertools
+
         from ..exprs import eval_scope
 
         if batches is None:
-            batches = self.execute_iter(node.child)
+            buf, rest = self._peek_stream(self.execute_iter(node.child))
+            if rest is None:
+                bb = _concat(buf)
+                with eval_scope(bb):
+                    return self._hash_agg_body(node, bb)
+            batches = itertools.chain(buf, rest)
         schema_batch: Optional[RecordBatch] = None
         limit = self.ctx.batch_rows
         acc: List[RecordBatch] = []

@@ -80,6 +80,7 @@ _TD = {
     dtypes.INT32: torch.int32, dtypes.INT64: torch.int64,
     dtypes.FLOAT32: torch.float32, dtypes.FLOAT64: torch.float64,
     dtypes.DATE32: torch.int32, dtypes.STRING: torch.uint8, dtypes.DECIMAL64: torch.int64,
+    dtypes.TIMESTAMP: torch.int64,
 }
 _ESIZE = {torch.bool: 1, torch.int8: 1, torch.int16: 2, torch.int32: 4,
           torch.int64: 8, torch.float32: 4, torch.float64: 8, torch.uint8: 1}

@@ -232,11 +232,17 @@ def _promote(l: Column, r: Column) -> Tuple[Column, Column, DataType]:
     return _cast_col(l, target), _cast_col(r, target), target
 
 
+_US_PER_DAY = 86_400_000_000
+_US_PER_SEC = 1_000_000
+
+
 def _cast_col(c: Column, dt: DataType) -> Column:
     if c.dtype.code == dt.code and c.dtype.scale == dt.scale:
         return c
     if c.dtype.is_string or dt.is_string:
         return _cast_string(c, dt)
+    if c.dtype.code == dtypes.TIMESTAMP or dt.code == dtypes.TIMESTAMP:
+        return _cast_timestamp(c, dt)
     if c.dtype.code == dtypes.DECIMAL64 and dt.code == dtypes.DECIMAL64:
         diff = dt.scale - c.dtype.scale
         data = c.data * (10 ** diff) if diff >= 0 else torch.div(c.data, 10 ** (-diff), rounding_mode="trunc")
@@ -253,6 +259,62 @@ def _cast_col(c: Column, dt: DataType) -> Column:
         # Spark cast double->int truncates toward zero
         return Column(dt, c.data.trunc().to(dt.torch_dtype), c.validity)
     return Column(dt, c.data.to(dt.torch_dtype), c.validity)
+
+
+def _cast_timestamp(c: Column, dt: DataType) -> Column:
+    """Spark timestamp cast matrix (micros-since-epoch storage):
+    ts<->date via whole days (floor, so pre-epoch rounds down),
+    ts<->integral via SECONDS, ts<->float via fractional seconds."""
+    if c.dtype.code == dtypes.TIMESTAMP:
+        us = c.data
+        if dt.code == dtypes.DATE32:
+            days = torch.div(us, _US_PER_DAY, rounding_mode="floor")
+            return Column(dt, days.to(torch.int32), c.validity)
+        if dt.is_integer:
+            secs = torch.div(us, _US_PER_SEC, rounding_mode="floor")
+            return Column(dt, secs.to(dt.torch_dtype), c.validity)
+        if dt.is_float:
+            return Column(dt, (us.to(torch.float64) / _US_PER_SEC).to(dt.torch_dtype),
+                          c.validity)
+        raise TypeError(f"cast timestamp -> {dt}")
+    # -> timestamp
+    if c.dtype.code == dtypes.DATE32:
+        return Column(dt, c.data.to(torch.int64) * _US_PER_DAY, c.validity)
+    if c.dtype.is_integer:
+        return Column(dt, c.data.to(torch.int64) * _US_PER_SEC, c.validity)
+    if c.dtype.is_float:
+        return Column(dt, (c.data.to(torch.float64) * _US_PER_SEC).to(torch.int64),
+                      c.validity)
+    raise TypeError(f"cast {c.dtype} -> timestamp")
+
+
+def _parse_ts_micros(v: str):
+    """'YYYY-MM-DD[ HH:MM:SS[.ffffff]]' -> micros since epoch (UTC),
+    None on parse failure (non-ANSI cast)."""
+    import datetime as _dt_mod
+
+    t = v.strip().replace("T", " ")
+    try:
+        if " " in t:
+            d = _dt_mod.datetime.fromisoformat(t)
+        else:
+            d = _dt_mod.datetime.combine(_dt_mod.date.fromisoformat(t),
+                                         _dt_mod.time())
+    except ValueError:
+        return None
+    epoch = _dt_mod.datetime(1970, 1, 1)
+    delta = d.replace(tzinfo=None) - epoch
+    return delta.days * 86_400_000_000 + delta.seconds * 1_000_000 + delta.microseconds
+
+
+def _format_ts_micros(us: int) -> str:
+    import datetime as _dt_mod
+
+    d = _dt_mod.datetime(1970, 1, 1) + _dt_mod.timedelta(microseconds=us)
+    base = d.strftime("%Y-%m-%d %H:%M:%S")
+    if d.microsecond:
+        base += f".{d.microsecond:06d}".rstrip("0")
+    return base
 
 
 def _cast_string(c: Column, dt: DataType) -> Column:
@@ -274,6 +336,8 @@ def _cast_string(c: Column, dt: DataType) -> Column:
                     out.append((_dt_mod.date.fromisoformat(v.strip()) - _EPOCH).days)
                 except ValueError:
                     out.append(None)
+            elif dt.code == dtypes.TIMESTAMP:
+                out.append(_parse_ts_micros(v))
             else:
                 try:
                     out.append(float(v) if dt.is_float or dt.code == dtypes.DECIMAL64 else int(float(v)))
@@ -285,6 +349,8 @@ def _cast_string(c: Column, dt: DataType) -> Column:
     if c.dtype.code == dtypes.DATE32:
         out = [None if v is None else (_EPOCH + _dt_mod.timedelta(days=int(v))).isoformat()
                for v in vals]
+    elif c.dtype.code == dtypes.TIMESTAMP:
+        out = [None if v is None else _format_ts_micros(int(v)) for v in vals]
     else:
         out = [None if v is None else (str(int(v)) if c.dtype.is_integer else str(v)) for v in vals]
     return Column.from_pylist(out, dt, c.device)

@@ -27,7 +27,7 @@
 enum AuDType : int32_t {
   AU_BOOL = 0, AU_INT8 = 1, AU_INT16 = 2, AU_INT32 = 3, AU_INT64 = 4,
   AU_FLOAT32 = 5, AU_FLOAT64 = 6, AU_DATE32 = 7, AU_STRING = 8,
-  AU_DECIMAL64 = 9,
+  AU_DECIMAL64 = 9, AU_LIST = 10, AU_TIMESTAMP = 11,
 };
 
 struct AuColDesc {
@@ -94,7 +94,8 @@ __global__ void __launch_bounds__(EX_BLOCK) k_expr_exec(
             case AU_INT32:
             case AU_DATE32: v.i = ((const int32_t*)c.data)[row]; break;
             case AU_INT64:
-            case AU_DECIMAL64: v.i = ((const int64_t*)c.data)[row]; break;
+            case AU_DECIMAL64:
+            case AU_TIMESTAMP: v.i = ((const int64_t*)c.data)[row]; break;
             case AU_FLOAT32: v.d = ((const float*)c.data)[row]; break;
             case AU_FLOAT64: v.d = ((const double*)c.data)[row]; break;
             default: break;
@@ -224,7 +225,8 @@ __global__ void __launch_bounds__(EX_BLOCK) k_expr_exec(
             case AU_INT32:
             case AU_DATE32: ((int32_t*)o.data)[row] = (int32_t)v.i; break;
             case AU_INT64:
-            case AU_DECIMAL64: ((int64_t*)o.data)[row] = v.i; break;
+            case AU_DECIMAL64:
+            case AU_TIMESTAMP: ((int64_t*)o.data)[row] = v.i; break;
             case AU_FLOAT32: ((float*)o.data)[row] = (float)v.d; break;
             case AU_FLOAT64: ((double*)o.data)[row] = v.d; break;
             default: break;

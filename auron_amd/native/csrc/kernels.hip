@@ -32,6 +32,8 @@ enum AuDType : int32_t {
   AU_DATE32 = 7,
   AU_STRING = 8,
   AU_DECIMAL64 = 9,
+  AU_LIST = 10,
+  AU_TIMESTAMP = 11,  // int64 microseconds since epoch
 };
 
 struct AuColDesc {
@@ -116,6 +118,7 @@ __device__ uint32_t hash_one(const AuColDesc& c, int64_t i, uint32_t seed) {
       return hash_int((uint32_t)((const int32_t*)c.data)[i], seed);
     case AU_INT64:
     case AU_DECIMAL64:
+    case AU_TIMESTAMP:
       return hash_long((uint64_t)((const int64_t*)c.data)[i], seed);
     case AU_FLOAT32: {
       float f = ((const float*)c.data)[i];
@@ -182,6 +185,7 @@ __device__ bool keys_equal(const AuColDesc* a_cols, int64_t ai,
         break;
       case AU_INT64:
       case AU_DECIMAL64:
+      case AU_TIMESTAMP:
         if (((const int64_t*)a.data)[ai] != ((const int64_t*)b.data)[bi]) return false;
         break;
       case AU_FLOAT32:
