@@ -566,18 +566,18 @@ class Executor:
         from .. import fused
 
         schema = {nm: c.dtype for nm, c in zip(batch.names, batch.columns)}
-        prog = fused.compile_exprs(exprs, schema)
-        self._fused_progs[id(node)] = (sig, prog)
-        return prog
+        progs = fused.compile_all(exprs, schema)
+        self._fused_progs[id(node)] = (sig, progs)
+        return progs
 
     def _eval_predicate(self, node, b):
         from ..exprs import eval_scope
 
-        prog = self._fused_prog(node, [node.predicate], b)
-        if prog is not None:
+        progs = self._fused_prog(node, [node.predicate], b)
+        if progs is not None:
             from .. import fused
 
-            c = fused.run(prog, b)[0]
+            c = fused.run(progs[0], b)[0]
         else:
             with eval_scope(b):
                 c = node.predicate.eval(b)
@@ -587,18 +587,25 @@ class Executor:
         return mask
 
     def _project_cols(self, node, b):
-        from ..exprs import eval_scope
-
-        prog = self._fused_prog(node, [a.expr for a in node.exprs], b)
-        if prog is None:
-            with eval_scope(b):
-                return [a.expr.eval(b) for a in node.exprs]
-        from .. import fused
+        from ..exprs import Col, eval_scope
 
         cols: List[Optional[Column]] = [None] * len(node.exprs)
-        fused_cols = fused.run(prog, b)
-        for k, i in enumerate(prog.expr_idx):
-            cols[i] = fused_cols[k]
+        nontrivial = []
+        for i, a in enumerate(node.exprs):
+            if isinstance(a.expr, Col):
+                # pass-through reference: no kernel, no copy
+                cols[i] = b.column(a.expr.name)
+            else:
+                nontrivial.append(i)
+        if nontrivial:
+            progs = self._fused_prog(node, [node.exprs[i].expr
+                                            for i in nontrivial], b)
+            for prog in progs or ():
+                from .. import fused
+
+                fused_cols = fused.run(prog, b)
+                for k, sub_i in enumerate(prog.expr_idx):
+                    cols[nontrivial[sub_i]] = fused_cols[k]
         if any(c is None for c in cols):
             with eval_scope(b):
                 for i, a in enumerate(node.exprs):

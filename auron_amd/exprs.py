@@ -294,6 +294,11 @@ def _parse_ts_micros(v: str):
     import datetime as _dt_mod
 
     t = v.strip().replace("T", " ")
+    if "." in t:
+        # py3.10 fromisoformat needs exactly 3 or 6 fractional digits
+        head, frac = t.rsplit(".", 1)
+        if frac.isdigit():
+            t = head + "." + (frac + "000000")[:6]
     try:
         if " " in t:
             d = _dt_mod.datetime.fromisoformat(t)

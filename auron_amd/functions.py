@@ -914,3 +914,150 @@ class MonotonicallyIncreasingId(Expr):
         n = batch.num_rows
         return Column(dtypes.int64,
                       base + torch.arange(n, dtype=torch.int64, device=batch.device))
+
+
+# ------------------------------------------------------------- timestamps
+def _ts_micros_col(e: Expr, batch) -> Column:
+    """Evaluate to a timestamp column (strings/dates coerce, spark_dates.rs
+    to_timestamp semantics)."""
+    c = e.eval(batch)
+    if c.dtype.code != dtypes.TIMESTAMP:
+        c = _cast_col(c, dtypes.timestamp)
+    return c
+
+
+_US_DAY = 86_400_000_000
+_US_HOUR = 3_600_000_000
+_US_MIN = 60_000_000
+_US_SEC = 1_000_000
+
+
+def _time_of_day_us(us: torch.Tensor) -> torch.Tensor:
+    return us - torch.div(us, _US_DAY, rounding_mode="floor") * _US_DAY
+
+
+@dataclass(eq=False)
+class Hour(Expr):
+    child: Expr
+
+    def eval(self, batch):
+        c = _ts_micros_col(self.child, batch)
+        tod = _time_of_day_us(c.data)
+        return Column(dtypes.int32, torch.div(tod, _US_HOUR, rounding_mode="floor").to(torch.int32), c.validity)
+
+
+@dataclass(eq=False)
+class Minute(Expr):
+    child: Expr
+
+    def eval(self, batch):
+        c = _ts_micros_col(self.child, batch)
+        tod = _time_of_day_us(c.data)
+        return Column(dtypes.int32, (torch.div(tod, _US_MIN, rounding_mode="floor") % 60).to(torch.int32), c.validity)
+
+
+@dataclass(eq=False)
+class Second(Expr):
+    child: Expr
+
+    def eval(self, batch):
+        c = _ts_micros_col(self.child, batch)
+        tod = _time_of_day_us(c.data)
+        return Column(dtypes.int32, (torch.div(tod, _US_SEC, rounding_mode="floor") % 60).to(torch.int32), c.validity)
+
+
+@dataclass(eq=False)
+class UnixTimestamp(Expr):
+    """unix_timestamp(ts|date|string) -> seconds since epoch (int64)."""
+    child: Expr
+
+    def eval(self, batch):
+        c = _ts_micros_col(self.child, batch)
+        return Column(dtypes.int64, torch.div(c.data, _US_SEC, rounding_mode="floor"), c.validity)
+
+
+@dataclass(eq=False)
+class FromUnixtime(Expr):
+    """from_unixtime(seconds) -> 'yyyy-MM-dd HH:mm:ss' string."""
+    child: Expr
+
+    def eval(self, batch):
+        c = _cast_col(self.child.eval(batch), dtypes.int64)
+        ts = Column(dtypes.timestamp, c.data * _US_SEC, c.validity)
+        return _cast_col(ts, dtypes.string)
+
+
+@dataclass(eq=False)
+class ToTimestamp(Expr):
+    child: Expr
+
+    def eval(self, batch):
+        return _ts_micros_col(self.child, batch)
+
+
+@dataclass(eq=False)
+class ToDate(Expr):
+    child: Expr
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        if c.dtype.code == dtypes.DATE32:
+            return c
+        if c.dtype.code == dtypes.TIMESTAMP:
+            return _cast_col(c, dtypes.date32)
+        return _cast_col(c, dtypes.date32)
+
+
+@dataclass(eq=False)
+class DateFormat(Expr):
+    """date_format(ts/date, java pattern) — host path, common subset of
+    the JDK pattern letters (spark_dates.rs date_format parity)."""
+    child: Expr
+    pattern: str
+
+    _MAP = [("yyyy", "%Y"), ("MM", "%m"), ("dd", "%d"), ("HH", "%H"),
+            ("mm", "%M"), ("ss", "%S"), ("EEEE", "%A"), ("EEE", "%a"),
+            ("MMMM", "%B"), ("MMM", "%b"), ("D", "%j"), ("yy", "%y")]
+
+    def eval(self, batch):
+        import datetime as _dt
+
+        c = _ts_micros_col(self.child, batch)
+        pat = self.pattern
+        for j, p in self._MAP:
+            pat = pat.replace(j, p)
+        epoch = _dt.datetime(1970, 1, 1)
+        vals = c.to_pylist()
+        out = [None if v is None else
+               (epoch + _dt.timedelta(microseconds=int(v))).strftime(pat)
+               for v in vals]
+        return Column.from_pylist(out, dtypes.string, str(batch.device))
+
+
+@dataclass(eq=False)
+class TruncTimestamp(Expr):
+    """date_trunc(unit, ts): YEAR/MONTH/DAY/HOUR/MINUTE/SECOND/WEEK."""
+    unit: str
+    child: Expr
+
+    def eval(self, batch):
+        c = _ts_micros_col(self.child, batch)
+        u = self.unit.lower()
+        us = c.data
+        if u in ("second", "minute", "hour", "day", "week"):
+            q = {"second": _US_SEC, "minute": _US_MIN, "hour": _US_HOUR,
+                 "day": _US_DAY, "week": 7 * _US_DAY}[u]
+            off = 4 * _US_DAY if u == "week" else 0  # 1970-01-01 is a Thursday
+            data = torch.div(us - off, q, rounding_mode="floor") * q + off
+            return Column(dtypes.timestamp, data, c.validity)
+        days = torch.div(us, _US_DAY, rounding_mode="floor")
+        y, m, d = _civil_from_days(days)
+        if u in ("year", "yyyy", "yy"):
+            m = torch.ones_like(m)
+            d = torch.ones_like(d)
+        elif u in ("month", "mon", "mm"):
+            d = torch.ones_like(d)
+        else:
+            raise ValueError(f"date_trunc unit {self.unit}")
+        data = _days_from_civil(y, m, d) * _US_DAY
+        return Column(dtypes.timestamp, data, c.validity)

@@ -75,9 +75,13 @@ def _is_d(dt: DataType) -> bool:
     return dt.code in (dtypes.FLOAT32, dtypes.FLOAT64)
 
 
+_MAX_COLS = 16   # EX_MAX_COLS (kernel-arg block)
+_MAX_OUTS = 8    # EX_MAX_OUTS
+
+
 class Prog:
     __slots__ = ("instr_np", "colnames", "out_dtypes", "max_depth",
-                 "expr_idx")
+                 "expr_idx", "_dev")
 
     def __init__(self, instr_np, colnames, out_dtypes, max_depth, expr_idx):
         self.instr_np = instr_np          # contiguous uint8 view of ExInstr[]
@@ -85,6 +89,16 @@ class Prog:
         self.out_dtypes = out_dtypes
         self.max_depth = max_depth
         self.expr_idx = expr_idx          # which caller exprs compiled
+        self._dev = {}                    # device -> uploaded instr tensor
+
+    def dev_instrs(self, device):
+        t = self._dev.get(device)
+        if t is None:
+            from .pinned import to_device
+
+            t = to_device(self.instr_np, device)
+            self._dev[device] = t
+        return t
 
 
 class _Compiler:
@@ -391,15 +405,23 @@ def compile_exprs(exprs: List, schema: Dict[str, DataType]) -> Optional[Prog]:
     out_dtypes: List[DataType] = []
     expr_idx: List[int] = []
     for i, e in enumerate(exprs):
+        if len(out_dtypes) >= _MAX_OUTS:
+            break
         mark_i, mark_d = len(comp.instrs), comp.depth
+        mark_c = len(comp.cols)
         try:
             dt = comp.emit(e)
             if dt.is_string or dt.code == dtypes.FLOAT32:
                 raise _Unsupported("output dtype")
+            if len(comp.cols) > _MAX_COLS:
+                raise _Unsupported("too many input columns")
             comp.op(OUT, len(out_dtypes), d=-1)
         except _Unsupported:
             del comp.instrs[mark_i:]
             comp.depth = mark_d
+            for nm in comp.cols[mark_c:]:
+                del comp.col_idx[nm]
+            del comp.cols[mark_c:]
             continue
         out_dtypes.append(dt)
         expr_idx.append(i)
@@ -412,10 +434,49 @@ def compile_exprs(exprs: List, schema: Dict[str, DataType]) -> Optional[Prog]:
                 comp.max_depth, expr_idx)
 
 
+def _host_descs(cols):
+    """AuColDesc[] as a HOST numpy array (device pointers inside); the
+    launch shim copies it into the kernel-argument block, so nothing is
+    uploaded per call."""
+    from . import native
+
+    arr = np.zeros(len(cols), dtype=native._DESC_DTYPE)
+    keep = []
+    for i, col in enumerate(cols):
+        data = col.data if col.data.is_contiguous() else col.data.contiguous()
+        keep.append(data)
+        arr[i]["data"] = data.data_ptr()
+        if col.validity is not None:
+            v = col.validity if col.validity.is_contiguous() else col.validity.contiguous()
+            keep.append(v)
+            arr[i]["validity"] = v.data_ptr()
+        arr[i]["dtype"] = col.dtype.code
+        arr[i]["scale"] = col.dtype.scale
+    return arr, keep
+
+
+def compile_all(exprs: List, schema: Dict[str, DataType]) -> Optional[List[Prog]]:
+    """Cover all compilable exprs with as many Progs as the kernel-arg
+    limits require (each <=8 outputs / <=16 input columns). Each Prog's
+    expr_idx refers to positions in the original `exprs` list."""
+    progs: List[Prog] = []
+    remaining = list(range(len(exprs)))
+    while remaining:
+        p = compile_exprs([exprs[i] for i in remaining], schema)
+        if p is None:
+            break
+        p.expr_idx = [remaining[j] for j in p.expr_idx]
+        progs.append(p)
+        covered = set(p.expr_idx)
+        remaining = [i for i in remaining if i not in covered]
+        if len(p.out_dtypes) < _MAX_OUTS:
+            break  # cap not hit: everything left is unsupported
+    return progs or None
+
+
 def run(prog: Prog, batch) -> List[Column]:
     """Execute a compiled program on a device batch -> output Columns."""
     from . import native
-    from .pinned import to_device
 
     device = batch.device
     n = batch.num_rows
@@ -428,12 +489,13 @@ def run(prog: Prog, batch) -> List[Column]:
         return outs
     lib = native.require()
     cols = [batch.column(nm) for nm in prog.colnames]
-    descs, keep = native.pack_descs(cols, device)
-    odescs, okeep = native.pack_descs(outs, device)
-    pdev = to_device(prog.instr_np, device)
+    descs, keep = _host_descs(cols)
+    odescs, okeep = _host_descs(outs)
+    pdev = prog.dev_instrs(device)
     n_instr = prog.instr_np.nbytes // 16
-    rc = lib.au_expr_exec(pdev.data_ptr(), n_instr, descs.data_ptr(),
-                          odescs.data_ptr(), len(outs),
+    rc = lib.au_expr_exec(pdev.data_ptr(), n_instr,
+                          descs.ctypes.data, len(cols),
+                          odescs.ctypes.data, len(outs),
                           max(prog.max_depth, 1), n,
                           native.stream_ptr(device))
     native.check(rc, "au_expr_exec")

@@ -70,8 +70,8 @@ def _try_load():
     lib.au_pq_copy_plain.argtypes = [c, ctypes.c_int, c, c, ctypes.c_int, i64, c]
     for f in ("au_pq_rle1", "au_pq_rle_idx", "au_pq_scatter", "au_pq_copy_plain"):
         getattr(lib, f).restype = ctypes.c_int
-    lib.au_expr_exec.argtypes = [c, ctypes.c_int, c, c, ctypes.c_int,
-                                 ctypes.c_int, i64, c]
+    lib.au_expr_exec.argtypes = [c, ctypes.c_int, c, ctypes.c_int, c,
+                                 ctypes.c_int, ctypes.c_int, i64, c]
     lib.au_expr_exec.restype = ctypes.c_int
     lib.au_part_hist.argtypes = [c, i64, i32, c, c]
     lib.au_part_scatter.argtypes = [c, i64, c, c, c, c]

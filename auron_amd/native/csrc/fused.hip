@@ -62,14 +62,24 @@ enum ExOp : int32_t {
 
 #define EX_BLOCK 256
 #define EX_MAX_INSTR 192
+#define EX_MAX_COLS 16
+#define EX_MAX_OUTS 8
+
+// column/output descriptors ride the kernel-argument block (copied by the
+// runtime at launch — no explicit H2D staging, no extra buffers to manage)
+struct ExParams {
+  AuColDesc cols[EX_MAX_COLS];
+  AuColDesc outs[EX_MAX_OUTS];
+};
 
 union ExVal { int64_t i; double d; };
 
 __global__ void __launch_bounds__(EX_BLOCK) k_expr_exec(
     const ExInstr* __restrict__ prog, int n_instr,
-    const AuColDesc* __restrict__ cols,
-    const AuColDesc* __restrict__ outs, int n_outs,
+    const ExParams p,
     int64_t n) {
+  const AuColDesc* __restrict__ cols = p.cols;
+  const AuColDesc* __restrict__ outs = p.outs;
   extern __shared__ int64_t stk[];  // [max_depth][EX_BLOCK]
   __shared__ ExInstr sprog[EX_MAX_INSTR];
   for (int i = threadIdx.x; i < n_instr; i += blockDim.x) sprog[i] = prog[i];
@@ -257,18 +267,26 @@ __global__ void __launch_bounds__(EX_BLOCK) k_expr_exec(
 #undef STK
 }
 
-AU_EXPORT int au_expr_exec(const void* prog, int n_instr,
-                           const void* cols,
-                           const void* outs, int n_outs,
+AU_EXPORT int au_expr_exec(const void* prog_dev, int n_instr,
+                           const void* host_cols, int n_cols,
+                           const void* host_outs, int n_outs,
                            int max_depth, int64_t n, void* stream) {
-  if (n <= 0 || n_instr <= 0 || n_instr > EX_MAX_INSTR) return n_instr > EX_MAX_INSTR ? 1001 : 0;
+  // host_cols/host_outs are HOST arrays of AuColDesc (whose data/validity
+  // members are device pointers); they are copied into the by-value
+  // kernel-arg block here, so the launch needs no descriptor upload
+  if (n <= 0 || n_instr <= 0) return 0;
+  if (n_instr > EX_MAX_INSTR || n_cols > EX_MAX_COLS || n_outs > EX_MAX_OUTS)
+    return 1001;
+  ExParams p{};
+  const AuColDesc* hc = (const AuColDesc*)host_cols;
+  const AuColDesc* ho = (const AuColDesc*)host_outs;
+  for (int i = 0; i < n_cols; i++) p.cols[i] = hc[i];
+  for (int i = 0; i < n_outs; i++) p.outs[i] = ho[i];
   int64_t g = (n + EX_BLOCK - 1) / EX_BLOCK;
   if (g > 2048) g = 2048;  // 256 CU x 8 XCD-filling grid-stride
   size_t lds = (size_t)max_depth * EX_BLOCK * sizeof(int64_t);
   hipLaunchKernelGGL(k_expr_exec, dim3((uint32_t)g), dim3(EX_BLOCK), lds,
                      (hipStream_t)stream,
-                     (const ExInstr*)prog, n_instr,
-                     (const AuColDesc*)cols,
-                     (const AuColDesc*)outs, n_outs, n);
+                     (const ExInstr*)prog_dev, n_instr, p, n);
   return (int)hipGetLastError();
 }

@@ -431,6 +431,11 @@ class Parser:
             return A.Str(t.text)
         if self.accept_kw("null"):
             return A.Null()
+        # typed literals: DATE '2000-01-01' / TIMESTAMP '2000-01-01 12:00:00'
+        if (self.cur.kind == "ident" and self.cur.text.lower() in
+                ("date", "timestamp") and self.peek().kind == "str"):
+            tn = self.advance().text.lower()
+            return A.CastE(A.Str(self.advance().text), tn)
         if self.accept_kw("case"):
             return self._case()
         if self.accept_kw("cast"):

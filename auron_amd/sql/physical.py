@@ -382,12 +382,13 @@ def prune_lquery(lq: LQuery):
 
 
 # ----------------------------------------------------------------- entry
-def sql_to_plan(sql_text: str, cat, session) -> P.PlanNode:
+def sql_to_plan(sql_text: str, cat, session, schemas=None) -> P.PlanNode:
     """Parse SQL, plan, prune, materialize CTEs/scalars, and return the
     executable physical plan (the engine-side product of the reference's
-    TaskDefinition handoff)."""
+    TaskDefinition handoff). `schemas` overrides the TPC-DS catalog with
+    {table: {column: DataType}} for ad-hoc tables."""
     ast_q = parse_sql(sql_text)
-    planner = Planner()
+    planner = Planner(catalog_schemas=schemas)
     lq = planner.plan(ast_q)
     prune_lquery(lq)
     for cte in planner.mat_ctes:

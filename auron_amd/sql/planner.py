@@ -1452,6 +1452,12 @@ class Planner:
                 return Literal((_dt.date.fromisoformat(e.value) - EPOCH).days,
                                dtypes.date32)
             return Cast(e, dtypes.date32)
+        if tn == "timestamp":
+            from ..exprs import _parse_ts_micros
+
+            if isinstance(e, Literal) and isinstance(e.value, str):
+                return Literal(_parse_ts_micros(e.value), dtypes.timestamp)
+            return Cast(e, dtypes.timestamp)
         if tn.startswith("decimal"):
             import re as _re
 
@@ -1553,7 +1559,119 @@ class Planner:
             return DatePart("month", conv(fc.args[0]))
         if name == "trim":
             return F.Trim(conv(fc.args[0]))
+        out = self._fn_extended(name, fc, conv)
+        if out is not None:
+            return out
         raise SqlError(f"unsupported function {name}")
+
+    def _fn_extended(self, name, fc, conv):
+        """Breadth routes beyond the TPC-DS core set (spark_dates.rs /
+        spark_strings.rs / spark math function names)."""
+        from ..exprs import Cast
+
+        a = fc.args
+        one = {"floor": F.Floor, "ceil": F.Ceil, "ceiling": F.Ceil,
+               "exp": F.Exp, "ln": F.Ln, "log10": F.Log10,
+               "sign": F.Sign, "signum": F.Sign, "isnan": F.IsNan,
+               "reverse": F.Reverse, "ascii": F.Ascii, "initcap": F.InitCap,
+               "quarter": F.Quarter, "dayofweek": F.DayOfWeek,
+               "weekofyear": F.WeekOfYear, "last_day": F.LastDay,
+               "hour": F.Hour, "minute": F.Minute, "second": F.Second,
+               "unix_timestamp": F.UnixTimestamp,
+               "from_unixtime": F.FromUnixtime,
+               "to_timestamp": F.ToTimestamp, "to_date": F.ToDate,
+               "size": F.ArraySize, "cardinality": F.ArraySize}
+        if name in one:
+            return one[name](conv(a[0]))
+        if name in ("day", "dayofmonth"):
+            from ..exprs import DatePart
+
+            return DatePart("day", conv(a[0]))
+        if name in ("pow", "power"):
+            return F.Pow(conv(a[0]), conv(a[1]))
+        if name == "greatest":
+            return F.Greatest([conv(x) for x in a])
+        if name == "least":
+            return F.Least([conv(x) for x in a])
+        if name in ("nvl", "ifnull"):
+            from ..exprs import Coalesce
+
+            return Coalesce([conv(a[0]), conv(a[1])])
+        if name == "nvl2":
+            return F.Nvl2(conv(a[0]), conv(a[1]), conv(a[2]))
+        if name == "if":
+            return F.If(conv(a[0]), conv(a[1]), conv(a[2]))
+        if name == "date_add":
+            return F.DateAdd(conv(a[0]), conv(a[1]))
+        if name == "date_sub":
+            return F.DateSub(conv(a[0]), conv(a[1]))
+        if name == "datediff":
+            return F.DateDiff(conv(a[0]), conv(a[1]))
+        if name == "add_months":
+            return F.AddMonths(conv(a[0]), _int_lit(conv(a[1])))
+        if name == "months_between":
+            return F.MonthsBetween(conv(a[0]), conv(a[1]))
+        if name == "next_day":
+            return F.NextDay(conv(a[0]), _str_lit(conv(a[1])))
+        if name == "trunc":
+            return F.TruncDate(conv(a[0]), _str_lit(conv(a[1])))
+        if name == "date_trunc":
+            return F.TruncTimestamp(_str_lit(conv(a[0])), conv(a[1]))
+        if name == "date_format":
+            return F.DateFormat(conv(a[0]), _str_lit(conv(a[1])))
+        if name == "replace":
+            return F.Replace(conv(a[0]), _str_lit(conv(a[1])),
+                             _str_lit(conv(a[2])) if len(a) > 2 else "")
+        if name == "lpad":
+            return F.LPad(conv(a[0]), _int_lit(conv(a[1])),
+                          _str_lit(conv(a[2])) if len(a) > 2 else " ")
+        if name == "rpad":
+            return F.RPad(conv(a[0]), _int_lit(conv(a[1])),
+                          _str_lit(conv(a[2])) if len(a) > 2 else " ")
+        if name == "ltrim":
+            return F.Trim(conv(a[0]), mode="leading")
+        if name == "rtrim":
+            return F.Trim(conv(a[0]), mode="trailing")
+        if name == "left":
+            return F.Left(conv(a[0]), _int_lit(conv(a[1])))
+        if name == "right":
+            return F.Right(conv(a[0]), _int_lit(conv(a[1])))
+        if name == "repeat":
+            return F.Repeat(conv(a[0]), _int_lit(conv(a[1])))  # scalar n
+        if name == "space":
+            return F.Space(conv(a[0]))
+        if name == "translate":
+            return F.Translate(conv(a[0]), _str_lit(conv(a[1])),
+                               _str_lit(conv(a[2])))
+        if name == "find_in_set":
+            return F.FindInSet(conv(a[0]), conv(a[1]))
+        if name == "split_part":
+            return F.SplitPart(conv(a[0]), _str_lit(conv(a[1])),
+                               _int_lit(conv(a[2])))
+        if name in ("instr", "position"):
+            return F.Instr(conv(a[0]), _str_lit(conv(a[1])))
+        if name == "locate":
+            return F.Instr(conv(a[1]), _str_lit(conv(a[0])))
+        if name == "bround":
+            return F.Bround(conv(a[0]),
+                            _int_lit(conv(a[1])) if len(a) > 1 else 0)
+        if name == "concat_ws":
+            return F.ConcatWs(_str_lit(conv(a[0])), [conv(x) for x in a[1:]])
+        if name == "get_json_object":
+            return F.GetJsonObject(conv(a[0]), _str_lit(conv(a[1])))
+        if name == "element_at":
+            return F.ElementAt(conv(a[0]), _int_lit(conv(a[1])))
+        if name == "array":
+            return F.MakeArray([conv(x) for x in a])
+        if name in ("double", "float"):
+            return Cast(conv(a[0]), dtypes.float64)
+        if name in ("int", "bigint"):
+            return Cast(conv(a[0]), dtypes.int64 if name == "bigint" else dtypes.int32)
+        if name == "string":
+            return Cast(conv(a[0]), dtypes.string)
+        if name == "xxhash64":
+            return F.XxHash64([conv(x) for x in a])
+        return None
 
     # --------------------------------------------- scalar subqueries
     def _scalar_subquery(self, q: A.Query, scope: Scope, ctx) -> Expr:
@@ -1789,6 +1907,12 @@ def _fold_const(e: Expr) -> Expr:
 
 def _is_null_lit(e: Expr) -> bool:
     return isinstance(e, Literal) and e.value is None
+
+
+def _str_lit(e: Expr) -> str:
+    if isinstance(e, Literal) and isinstance(e.value, str):
+        return e.value
+    raise SqlError("expected string literal argument")
 
 
 def _int_lit(e: Expr) -> int:

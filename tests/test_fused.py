@@ -56,11 +56,14 @@ def _battery():
 
 
 def test_compile_battery_cpu():
-    progs = fused.compile_exprs(_battery(), _schema())
+    progs = fused.compile_all(_battery(), _schema())
     assert progs is not None
-    assert len(progs.expr_idx) == len(_battery())
-    assert progs.max_depth <= 30
-    assert progs.instr_np.nbytes % 16 == 0
+    covered = sorted(i for p in progs for i in p.expr_idx)
+    assert covered == list(range(len(_battery())))
+    for p in progs:
+        assert p.max_depth <= 30
+        assert len(p.out_dtypes) <= 8
+        assert p.instr_np.nbytes % 16 == 0
 
 
 def test_compile_string_falls_back():
@@ -108,9 +111,13 @@ def test_fused_matches_eval_gpu():
     batch = _make_batch("cuda")
     exprs = _battery()
     schema = {nm: c.dtype for nm, c in zip(batch.names, batch.columns)}
-    prog = fused.compile_exprs(exprs, schema)
-    assert prog is not None and len(prog.expr_idx) == len(exprs)
-    got = fused.run(prog, batch)
+    progs = fused.compile_all(exprs, schema)
+    assert progs is not None
+    got = [None] * len(exprs)
+    for p in progs:
+        for k, i in enumerate(fused.run(p, batch)):
+            got[p.expr_idx[k]] = i
+    assert all(g is not None for g in got)
     with eval_scope(batch):
         want = [e.eval(batch) for e in exprs]
     torch.cuda.synchronize()

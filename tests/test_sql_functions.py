@@ -1,0 +1,137 @@
+"""SQL-surface scalar function routes (breadth beyond the TPC-DS core)
+and the timestamp dtype/cast matrix."""
+import os
+
+import pyarrow as pa
+import pyarrow.parquet as pq
+import pytest
+
+from auron_amd import AuronSession, dtypes
+from auron_amd.column import Column
+from auron_amd.exprs import _cast_col
+from auron_amd.plan import nodes as P
+from auron_amd.sql import sql_to_plan
+
+SCH = {"tt": {"x": dtypes.int64, "s": dtypes.string, "t": dtypes.int64,
+              "d": dtypes.date32}}
+
+
+@pytest.fixture()
+def env(tmp_path):
+    p = str(tmp_path / "p.parquet")
+    pq.write_table(pa.table({
+        "x": [1, 2, 3],
+        "s": ["a,b", "b", None],
+        "t": pa.array([1000000, 2000000, None], pa.int64()),
+        "d": pa.array([11016, 11017, 11023], pa.date32()),
+    }), p)
+
+    class Cat:
+        def scan(self, table, columns=None):
+            return P.ParquetScan([p], columns=columns)
+
+    return Cat(), AuronSession()
+
+
+def run(env, sql):
+    cat, s = env
+    return s.collect(sql_to_plan(sql, cat, s, schemas=SCH)).to_pydict()
+
+
+def test_math_routes(env):
+    out = run(env, "select floor(x/2.0) f, ceil(x/2.0) c, power(x,2) p, "
+                   "sign(x-2) sg, exp(0.0) e from tt order by f, c")
+    assert out["f"] == [0, 1, 1]
+    assert out["c"] == [1, 1, 2]
+    assert out["p"] == [1.0, 4.0, 9.0]
+    assert out["sg"] == [-1.0, 0.0, 1.0]
+
+
+def test_string_routes(env):
+    out = run(env, "select find_in_set('b', s) fis, lpad(s, 5, '*') lp, "
+                   "rpad(s, 4, '.') rp, reverse(s) rv, initcap(s) ic, "
+                   "repeat(s, 2) rr, instr(s, 'b') ib from tt")
+    assert out["fis"] == [2, 1, None]
+    assert out["lp"] == ["**a,b", "****b", None]
+    assert out["rp"] == ["a,b.", "b...", None]
+    assert out["rv"] == ["b,a", "b", None]
+    assert out["ib"] == [3, 1, None]
+
+
+def test_conditional_routes(env):
+    out = run(env, "select nvl(t, 0) n, nvl2(t, 1, 0) n2, "
+                   "if(x > 1, 'big', 'small') i, nullif(x, 2) nl "
+                   "from tt order by x")
+    assert out["n"] == [1000000, 2000000, 0]
+    assert out["n2"] == [1, 1, 0]
+    assert out["i"] == ["small", "big", "big"]
+    assert out["nl"] == [1, None, 3]
+
+
+def test_date_routes(env):
+    out = run(env, "select quarter(d) q, dayofweek(d) dw, day(d) dd, "
+                   "date_add(d, 5) da, datediff(date '2000-03-01', d) df "
+                   "from tt order by d")
+    # 11016 days = 2000-02-29 (leap day)
+    assert out["q"][0] == 1 and out["dd"][0] == 29
+    assert out["df"][0] == 1
+
+
+def test_timestamp_routes(env):
+    out = run(env, "select hour(to_timestamp('2001-03-04 05:06:07')) h, "
+                   "minute(to_timestamp('2001-03-04 05:06:07')) m, "
+                   "second(to_timestamp('2001-03-04 05:06:07')) sc, "
+                   "unix_timestamp(to_timestamp('1970-01-02 00:00:00')) u, "
+                   "from_unixtime(60) fu, "
+                   "date_format(to_timestamp('2001-03-04 05:06:07'), "
+                   "'yyyy/MM/dd') fm from tt limit 1")
+    assert out["h"] == [5] and out["m"] == [6] and out["sc"] == [7]
+    assert out["u"] == [86400]
+    assert out["fu"] == ["1970-01-01 00:01:00"]
+    assert out["fm"] == ["2001/03/04"]
+
+
+def test_timestamp_cast_matrix():
+    c = Column.from_pylist(["2001-03-04 05:06:07.25", "1969-12-31",
+                            "bogus", None], dtypes.string)
+    ts = _cast_col(c, dtypes.timestamp)
+    assert ts.validity.tolist() == [True, True, False, False]
+    # round-trip through string
+    back = _cast_col(ts, dtypes.string)
+    assert back.to_pylist()[0] == "2001-03-04 05:06:07.25"
+    # to date (floor) and to seconds (floor)
+    d = _cast_col(ts, dtypes.date32)
+    assert d.data[1].item() == -1  # 1969-12-31 is day -1
+    secs = _cast_col(ts, dtypes.int64)
+    assert secs.data[1].item() == -86400
+    # date -> timestamp -> float seconds
+    dd = Column.from_pylist([1], dtypes.date32)
+    t2 = _cast_col(dd, dtypes.timestamp)
+    assert t2.data[0].item() == 86_400_000_000
+    f = _cast_col(ts, dtypes.float64)
+    assert abs(f.data[0].item() - 983682367.25) < 1e-6
+
+
+def test_timestamp_arrow_roundtrip():
+    ts = Column.from_pylist([0, 86_400_000_000, None], dtypes.timestamp)
+    arr = ts.to_arrow()
+    assert pa.types.is_timestamp(arr.type)
+    back = Column.from_arrow(arr)
+    assert back.dtype.code == dtypes.TIMESTAMP
+    assert back.data.tolist()[:2] == [0, 86_400_000_000]
+
+
+def test_timestamp_group_and_join_keys():
+    # timestamps behave as first-class keys through the whole engine
+    from auron_amd.exprs import AggFunc, Aliased, Col
+
+    s = AuronSession()
+    ts = Column.from_pylist([0, 0, 86_400_000_000], dtypes.timestamp)
+    v = Column.from_pylist([1, 2, 3], dtypes.int64)
+    from auron_amd.column import RecordBatch
+
+    scan = P.MemoryScan([RecordBatch(["t", "v"], [ts, v])])
+    plan = P.HashAgg(scan, [Aliased(Col("t"), "t")],
+                     [AggFunc("sum", Col("v"), name="sv")], mode="complete")
+    out = s.collect(plan).to_pydict()
+    assert sorted(out["sv"]) == [3, 3]
