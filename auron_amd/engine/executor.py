@@ -1154,14 +1154,17 @@ class Executor:
     def _exec_HashAgg(self, node: P.HashAgg) -> List[RecordBatch]:
         from ..exprs import eval_scope
 
-        if node.mode == "partial" and AuronConf().get(AGG_STREAMING):
+        stream_ok = (AuronConf().get(AGG_STREAMING)
+                     and self._est_input_bytes(node.child)
+                     > self._stream_bytes_threshold())
+        if node.mode == "partial" and stream_ok:
             # streaming partial agg (agg_table.rs analogue): inputs are
             # consumed chunk-by-chunk, states re-merged when they shrink,
             # and the accumulated state registers with the memmgr so it
             # can spill under pressure — the whole input never has to be
             # resident at once
             return self._exec_hash_agg_partial_chunked(node)
-        if (node.mode == "complete" and AuronConf().get(AGG_STREAMING)
+        if (node.mode == "complete" and stream_ok
                 and all(a.fn in _SPLITTABLE_AGGS for a in node.aggs)
                 and (node.keys or self.ctx.world_size == 1)):
             # (keyless complete at W>1 keeps the one-shot path: its
@@ -1189,6 +1192,39 @@ class Executor:
         b = _concat(self.execute(node.child))
         with eval_scope(b):
             return self._hash_agg_body(node, b)
+
+    def _est_input_bytes(self, node: P.PlanNode) -> int:
+        """Static, rank-deterministic size estimate of a subtree's output
+        (decoded bytes). Drives the stream-vs-materialize choice: the
+        chunked/streaming machinery only pays for itself when the input
+        is a meaningful fraction of HBM, and the estimate must agree
+        across ranks (SPMD rule — no rank-local row counts)."""
+        if isinstance(node, P.ParquetScan):
+            try:
+                raw = sum(os.path.getsize(p) for p in node.paths)
+            except OSError:
+                return 1 << 62
+            # parquet on-disk -> decoded columns expansion factor
+            return 3 * raw // max(self.ctx.world_size, 1)
+        if isinstance(node, P.MemoryScan):
+            from ..memory import _batch_bytes
+
+            return sum(_batch_bytes(b) for b in node.batches)
+        if isinstance(node, P.Expand):
+            return len(node.projections) * self._est_input_bytes(node.child)
+        kids = node.children()
+        if not kids:
+            return 0
+        return sum(self._est_input_bytes(c) for c in kids)
+
+    def _stream_bytes_threshold(self) -> int:
+        env = os.environ.get("AURON_STREAM_BYTES")
+        if env is not None:
+            return int(env)
+        if self.ctx.device.type == "cuda":
+            _, total = torch.cuda.mem_get_info()
+            return int(total * 0.08)
+        return 1 << 30
 
     def _peek_stream(self, it, threshold: Optional[int] = None):
         """Pull from a batch iterator until `threshold` rows are buffered.

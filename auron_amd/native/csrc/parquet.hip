@@ -453,3 +453,25 @@ AU_EXPORT int au_pq_copy_plain(const void* pages_dev, int npages, const void* bu
                        npages, (const uint8_t*)buf, (uint64_t*)out, total);
   return (int)hipGetLastError();
 }
+
+// Host-side walk of a PLAIN BYTE_ARRAY data-page payload:
+// [u32 len][bytes][u32 len][bytes]... -> per-value (absolute offset, length).
+// Returns the number of values parsed, or -1 on overflow/corruption.
+// (arrow-rs does the equivalent offset materialization host-side too;
+// the variable-length walk is inherently serial.)
+AU_EXPORT int64_t au_host_plainba_parse(const uint8_t* buf, int64_t off,
+                                        int64_t len, int64_t* offs,
+                                        int32_t* lens, int64_t cap) {
+  int64_t pos = off, end = off + len, n = 0;
+  while (pos + 4 <= end) {
+    uint32_t l = (uint32_t)buf[pos] | ((uint32_t)buf[pos + 1] << 8) |
+                 ((uint32_t)buf[pos + 2] << 16) | ((uint32_t)buf[pos + 3] << 24);
+    pos += 4;
+    if ((int64_t)l > end - pos || n >= cap) return -1;
+    offs[n] = pos;
+    lens[n] = (int32_t)l;
+    pos += l;
+    n++;
+  }
+  return n;
+}

@@ -171,6 +171,10 @@ class ChunkPages:
     # per-page RLE index bit widths (host byte, cached so HBM-cache hits
     # never need the host buffer)
     idx_bws: Optional[list] = None
+    # lazy host-parsed PLAIN BYTE_ARRAY layout: per PRESENT value
+    # (absolute offset, byte length) in page order
+    plain_str_offs: Optional[np.ndarray] = None
+    plain_str_lens: Optional[np.ndarray] = None
 
     @property
     def is_dict(self) -> bool:
@@ -403,6 +407,32 @@ def parse_dict_strings(buf: np.ndarray, ck: ChunkPages) -> None:
         pos += 4 + ln
     ck.dict_str_offs = offs
     ck.dict_str_lens = lens
+
+
+def parse_plain_strings(buf: np.ndarray, ck: ChunkPages) -> None:
+    """Host walk of PLAIN BYTE_ARRAY page payloads -> per-present-value
+    (absolute offset, length) arrays (au_host_plainba_parse; the
+    variable-length [len][bytes] chain is inherently serial, like
+    arrow-rs's offset materialization)."""
+    if ck.plain_str_offs is not None:
+        return
+    from . import native
+
+    lib = native.host_lib()
+    total = sum(p.n_values for p in ck.pages)
+    offs = np.empty(total, dtype=np.int64)
+    lens = np.empty(total, dtype=np.int32)
+    base = buf.ctypes.data
+    n = 0
+    for p in ck.pages:
+        got = lib.au_host_plainba_parse(
+            base, p.values_off, p.values_len,
+            offs.ctypes.data + 8 * n, lens.ctypes.data + 4 * n, total - n)
+        if got < 0:
+            raise ValueError("corrupt PLAIN byte-array page")
+        n += got
+    ck.plain_str_offs = offs[:n]
+    ck.plain_str_lens = lens[:n].astype(np.int64)
 
 
 def _chunk_indices_np(buf: np.ndarray, ck: ChunkPages,
@@ -814,8 +844,8 @@ def read_columns_native(path: str, columns: List[str], device,
                 for (new_off, clen, nvals, _an, codec) in cm.chunks:
                     ck = parse_pages(buf, new_off, clen, nvals, cm.has_def,
                                      codec, extra)
-                    if ck is None or (cm.phys == "BYTE_ARRAY" and not ck.is_dict):
-                        good = False  # e.g. PLAIN byte-array pages -> host path
+                    if ck is None:
+                        good = False
                         break
                     cm.pages.append(ck)
                 if good:
@@ -859,8 +889,11 @@ def read_columns_native(path: str, columns: List[str], device,
             # cache-hit reads can run without a host buffer at all
             for cm in meta.cols:
                 for ck in cm.pages:
-                    if cm.phys == "BYTE_ARRAY" and ck.is_dict:
-                        parse_dict_strings(buf, ck)
+                    if cm.phys == "BYTE_ARRAY":
+                        if ck.is_dict:
+                            parse_dict_strings(buf, ck)
+                        else:
+                            parse_plain_strings(buf, ck)
                     if ck.is_dict and ck.idx_bws is None:
                         ck.idx_bws = [int(buf[p.values_off]) for p in ck.pages]
             _dbuf_cache_put((path, os.path.getmtime(path), tuple(columns)),
@@ -982,9 +1015,79 @@ def _decode_chunk_gpu_dict(dbuf, buf, ck: ChunkPages, num_values: int,
     return out, vmask
 
 
+def _decode_chunk_plain_strings(buf, dbuf, ck: ChunkPages, num_values: int,
+                                device, use_gpu: bool,
+                                chunk_nulls: bool) -> Column:
+    """PLAIN BYTE_ARRAY chunk -> string Column (host-parsed offsets,
+    device byte gather; the path pyarrow writers take when a column
+    overflows the dictionary-page limit — high-cardinality strings)."""
+    if not use_gpu:
+        parse_plain_strings(buf, ck)
+        has_def = ck.pages[0].def_off >= 0
+        validity = None
+        if has_def:
+            validity = np.zeros(num_values, dtype=np.uint8)
+            for p in ck.pages:
+                validity[p.row_start:p.row_start + p.n_values] = rle1_decode_np(
+                    buf[p.def_off:p.def_off + p.def_len], p.n_values)
+        vmask = validity.astype(bool) if validity is not None \
+            else np.ones(num_values, dtype=bool)
+        lens = np.zeros(num_values, dtype=np.int64)
+        lens[vmask] = ck.plain_str_lens
+        starts = np.zeros(num_values, dtype=np.int64)
+        starts[vmask] = ck.plain_str_offs
+        offsets = np.zeros(num_values + 1, dtype=np.int64)
+        np.cumsum(lens, out=offsets[1:])
+        data = np.empty(int(offsets[-1]), dtype=np.uint8)
+        pos = 0
+        for r in range(num_values):
+            ln = lens[r]
+            if ln:
+                data[pos:pos + ln] = buf[starts[r]:starts[r] + ln]
+                pos += ln
+        vt = None
+        if validity is not None and chunk_nulls and not validity.all():
+            vt = torch.from_numpy(validity.astype(bool))
+        return Column(dtypes.string, torch.from_numpy(data), vt,
+                      torch.from_numpy(offsets))
+    validity, prefix = _gpu_validity_prefix(dbuf, ck.pages, num_values, device,
+                                            chunk_nulls, ck.runs_np)
+    if ck.plain_str_offs is None:
+        parse_plain_strings(buf, ck)
+    offs_t = torch.from_numpy(ck.plain_str_offs).to(device, non_blocking=True)
+    lens_t = torch.from_numpy(ck.plain_str_lens).to(device, non_blocking=True)
+    if validity is not None:
+        vmask = validity.to(torch.bool)
+        pos = (prefix - 1).clamp(min=0)
+        p_idx = pos.clamp_(0, max(int(offs_t.numel()) - 1, 0))
+        zero = torch.zeros((), dtype=torch.int64, device=device)
+        lens = torch.where(vmask, lens_t[p_idx], zero)
+        starts = offs_t[p_idx]
+    else:
+        vmask = None
+        lens = lens_t[:num_values]
+        starts = offs_t[:num_values]
+    offsets = torch.zeros(num_values + 1, dtype=torch.int64, device=device)
+    torch.cumsum(lens, 0, out=offsets[1:])
+    total = int(offsets[-1].item())
+    if total:
+        row = torch.repeat_interleave(lens)
+        within = torch.arange(total, dtype=torch.int64, device=device) \
+            - offsets[:-1][row]
+        src = starts[row] + within
+        data = dbuf[src]
+    else:
+        data = torch.empty(0, dtype=torch.uint8, device=device)
+    vt = vmask if (vmask is not None and chunk_nulls) else None
+    return Column(dtypes.string, data, vt, offsets)
+
+
 def _decode_chunk_strings(buf, dbuf, ck: ChunkPages, num_values: int, device,
                           use_gpu: bool, chunk_nulls: bool) -> Column:
     """Dictionary-encoded BYTE_ARRAY chunk -> string Column."""
+    if not ck.is_dict:
+        return _decode_chunk_plain_strings(buf, dbuf, ck, num_values, device,
+                                           use_gpu, chunk_nulls)
     if not use_gpu:
         data, validity, offsets = decode_chunk_np_dict(buf, ck, num_values,
                                                        "BYTE_ARRAY")

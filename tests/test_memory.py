@@ -113,8 +113,14 @@ def test_chunked_partial_agg_spills_and_matches():
         [Aliased(col("k"), "k")],
         [AggFunc("sum", col("v"), name="s")], mode="final")
 
-    ctx = ExecContext(memmgr=MemManager(budget_bytes=64 << 10), batch_rows=2000)
-    out = Executor(ctx).collect(plan)
+    import os
+
+    os.environ["AURON_STREAM_BYTES"] = "0"  # force the streaming path
+    try:
+        ctx = ExecContext(memmgr=MemManager(budget_bytes=64 << 10), batch_rows=2000)
+        out = Executor(ctx).collect(plan)
+    finally:
+        del os.environ["AURON_STREAM_BYTES"]
     assert ctx.memmgr.metrics.get("spill_count", 0) > 0, ctx.memmgr.metrics
 
     ctx2 = ExecContext(memmgr=MemManager(budget_bytes=16 << 30))

@@ -153,31 +153,63 @@ def test_np_dict_decoder(dict_file):
     _check(cols, t)
 
 
-def test_plain_byte_array_falls_back(tmp_path):
-    t = pa.table({"s": pa.array(["aa", "bb"] * 100)})
-    p = tmp_path / "pba.parquet"
-    pq.write_table(t, str(p), compression="NONE", use_dictionary=False,
-                   data_page_version="1.0")
-    assert parquet_native.read_columns_native(str(p), ["s"], "cpu",
-                                              _np_only=True) is None
+@pytest.mark.parametrize("codec", ["NONE", "snappy", "zstd"])
+@pytest.mark.parametrize("pv", ["1.0", "2.0"])
+def test_plain_byte_array_native(tmp_path, codec, pv):
+    # the path pyarrow takes when a string column overflows the dictionary
+    # page limit (high-cardinality columns at SF>=10)
+    import random
+
+    rng = random.Random(11)
+    vals = ["".join(rng.choices("abcdefgh", k=rng.randint(0, 12)))
+            if rng.random() > 0.15 else None for _ in range(5000)]
+    t = pa.table({"s": pa.array(vals)})
+    p = str(tmp_path / "pba.parquet")
+    pq.write_table(t, p, compression=codec, use_dictionary=False,
+                   data_page_version=pv)
+    cols = parquet_native.read_columns_native(p, ["s"], "cpu", _np_only=True)
+    assert cols is not None
+    assert cols["s"].to_pylist() == vals
+
+
+@pytest.mark.gpu
+def test_plain_byte_array_gpu(tmp_path):
+    import random
+
+    from auron_amd import native
+
+    native.require()
+    rng = random.Random(12)
+    vals = ["".join(rng.choices("abcdefgh", k=rng.randint(0, 20)))
+            if rng.random() > 0.15 else None for _ in range(20000)]
+    t = pa.table({"s": pa.array(vals)})
+    for codec in ("NONE", "snappy"):
+        p = str(tmp_path / f"pba_{codec}.parquet")
+        pq.write_table(t, p, compression=codec, use_dictionary=False)
+        cols = parquet_native.read_columns_native(p, ["s"], "cuda")
+        assert cols is not None
+        assert cols["s"].to("cpu").to_pylist() == vals
 
 
 def test_partial_column_fallback(tmp_path):
-    """One dict column decodes natively while a PLAIN string column is
-    dropped for the host path — no wholesale failure."""
+    """Supported columns decode natively while an unsupported (nested
+    list) column is dropped for the host path — no wholesale failure."""
     n = 5000
     rng = np.random.default_rng(5)
     t = pa.table({
         "num": pa.array(rng.integers(0, 50, n)),
         "ps": pa.array([f"unique-{i}" for i in range(n)]),
+        "lst": pa.array([[1, 2]] * n),
     })
     p = str(tmp_path / "mix.parquet")
     pq.write_table(t, p, compression="NONE", data_page_version="1.0",
                    use_dictionary=["num"])
-    cols = parquet_native.read_columns_native(p, ["num", "ps"], "cpu",
-                                              _np_only=True)
-    assert cols is not None and "num" in cols and "ps" not in cols
+    ok, rest = parquet_native.split_supported(p, ["num", "ps", "lst"])
+    assert set(ok) == {"num", "ps"} and rest == ["lst"]
+    cols = parquet_native.read_columns_native(p, ok, "cpu", _np_only=True)
+    assert cols is not None and "num" in cols and "ps" in cols
     _check({"num": cols["num"]}, t.select(["num"]))
+    assert cols["ps"].to_pylist() == t["ps"].to_pylist()
 
 
 @pytest.mark.gpu
