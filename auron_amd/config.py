@@ -87,6 +87,42 @@ SORT_LIMIT_TOPK = _opt("spark.auron.sort.topK.enable", True, bool,
                        "ORDER BY+LIMIT keeps only the top K rows per batch")
 EXCHANGE_SAMPLE_ROWS = _opt("spark.auron.rangePartition.sampleRows", 4096, int,
                             "per-rank sample size for range-partition bounds")
+SCAN_CACHE_BYTES = _opt("spark.auron.scan.cache.maxBytes", 0, int,
+                        "per-query decoded-scan cache budget; 0 = auto "
+                        "(15% of HBM on GPU, 4 GB on CPU)",
+                        env="AURON_SCAN_CACHE_BYTES")
+STREAM_BYTES = _opt("spark.auron.stream.minInputBytes", -1, int,
+                    "estimated agg-input bytes above which operators "
+                    "stream instead of materializing; -1 = auto (8% of "
+                    "HBM on GPU, 1 GB on CPU), 0 = always stream",
+                    env="AURON_STREAM_BYTES")
+SCAN_HBM_CACHE = _opt("spark.auron.scan.hbmStagedCache.maxBytes", 0, int,
+                      "staged raw-page bytes kept device-resident across "
+                      "reads; 0 = auto (30% of HBM)",
+                      env="AURON_SCAN_HBM_CACHE")
+HBM_CACHE_RESERVE = _opt("spark.auron.scan.hbmStagedCache.reserveFraction",
+                         0.25, float,
+                         "free-HBM fraction the staged-bytes cache must "
+                         "never consume (evicts itself under pressure)")
+SCAN_PREFETCH = _opt("spark.auron.scan.streamPrefetch", 2, int,
+                     "files decoded ahead of the consumer on the "
+                     "streaming scan path")
+SCAN_IO_THREADS = _opt("spark.auron.scan.ioThreads", 8, int,
+                       "host page-read/staging thread-pool size",
+                       env="AURON_SCAN_IO_THREADS")
+STREAM_PEEK_FACTOR = _opt("spark.auron.stream.peekBatches", 8, int,
+                          "batch-size multiples buffered before the "
+                          "chunked agg commits to streaming")
+STREAM_JOIN_PROBE = _opt("spark.auron.stream.joinProbe.enable", True, bool,
+                         "probe build-right joins batch-by-batch on the "
+                         "streaming path (GPU only)",
+                         env="AURON_STREAM_JOIN_PROBE")
+SPILL_DIR = _opt("spark.auron.spill.dir", "", str,
+                 "directory for disk-tier spill files (empty = system tmp)",
+                 env="AURON_SPILL_DIR")
+MEM_BUDGET = _opt("spark.auron.memory.budgetBytes", 0, int,
+                  "absolute memmgr budget override; 0 = memoryFraction "
+                  "of device/host memory", env="AURON_MEM_BUDGET")
 
 
 class AuronConf:

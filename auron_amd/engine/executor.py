@@ -168,8 +168,11 @@ class Executor:
         if self._io_pool is None:
             from concurrent.futures import ThreadPoolExecutor
 
-            self._io_pool = ThreadPoolExecutor(max_workers=8,
-                                               thread_name_prefix="auron-io")
+            from ..config import SCAN_IO_THREADS
+
+            self._io_pool = ThreadPoolExecutor(
+                max_workers=AuronConf().get(SCAN_IO_THREADS),
+                thread_name_prefix="auron-io")
         return self._io_pool
 
     def cancel(self):
@@ -308,6 +311,9 @@ class Executor:
 
         from ..memory import _batch_bytes
 
+        from ..config import SCAN_PREFETCH
+
+        window = max(1, AuronConf().get(SCAN_PREFETCH))
         pool = self._scan_pool()
         pending: deque = deque()
         idx = 0
@@ -316,7 +322,7 @@ class Executor:
         retain_bytes = 0
         budget = self._scan_cache_budget() - self._scan_cache_bytes
         while idx < len(my_files) or pending:
-            while idx < len(my_files) and len(pending) < 2:
+            while idx < len(my_files) and len(pending) < window:
                 pending.append(pool.submit(
                     self._read_parquet_tolerant, my_files[idx],
                     node.columns, node.filters))
@@ -343,7 +349,10 @@ class Executor:
         cache), then each probe batch joins and is yielded immediately.
         Valid when every probe row is decided independently — inner/left
         (+ residual); other shapes fall back to the materializing path."""
+        from ..config import STREAM_JOIN_PROBE
+
         if (node.build_side != "right" or node.how not in ("inner", "left")
+                or not AuronConf().get(STREAM_JOIN_PROBE)
                 or (self.ctx.device.type != "cuda" and not os.environ.get(
                     "AURON_FORCE_STREAM_JOIN"))):
             # on CPU, hash_join runs the host reference impl, which would
@@ -441,9 +450,11 @@ class Executor:
         return RecordBatch.from_arrow(t, self.ctx.device)
 
     def _scan_cache_budget(self) -> int:
-        env = os.environ.get("AURON_SCAN_CACHE_BYTES")
-        if env is not None:
-            return int(env)
+        from ..config import SCAN_CACHE_BYTES
+
+        v = AuronConf().get(SCAN_CACHE_BYTES)
+        if v:
+            return v
         if self.ctx.device.type == "cuda":
             _, total = torch.cuda.mem_get_info()
             return int(total * 0.15)
@@ -1218,9 +1229,11 @@ class Executor:
         return sum(self._est_input_bytes(c) for c in kids)
 
     def _stream_bytes_threshold(self) -> int:
-        env = os.environ.get("AURON_STREAM_BYTES")
-        if env is not None:
-            return int(env)
+        from ..config import STREAM_BYTES
+
+        v = AuronConf().get(STREAM_BYTES)
+        if v >= 0:
+            return v
         if self.ctx.device.type == "cuda":
             _, total = torch.cuda.mem_get_info()
             return int(total * 0.08)
@@ -1233,7 +1246,9 @@ class Executor:
         take the cheaper one-shot path; the chunk/merge machinery only
         pays for itself when the input is genuinely large)."""
         if threshold is None:
-            threshold = 8 * self.ctx.batch_rows
+            from ..config import STREAM_PEEK_FACTOR
+
+            threshold = AuronConf().get(STREAM_PEEK_FACTOR) * self.ctx.batch_rows
         buf: List[RecordBatch] = []
         rows = 0
         for b in it:

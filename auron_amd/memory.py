@@ -155,7 +155,10 @@ class MemManager:
     def __init__(self, budget_bytes: Optional[int] = None, fraction: float = 0.8,
                  spill_dir: Optional[str] = None, host_budget_bytes: Optional[int] = None):
         if budget_bytes is None:
-            env = os.environ.get("AURON_MEM_BUDGET")
+            from .config import MEM_BUDGET, AuronConf
+
+            _mb = AuronConf().get(MEM_BUDGET)
+            env = str(_mb) if _mb else None
             if env is not None:
                 budget_bytes = int(env)
             elif torch.cuda.is_available():
@@ -169,6 +172,10 @@ class MemManager:
 
             host_budget_bytes = int(AuronConf().get(HOST_SPILL_BUDGET))
         self.host_budget = host_budget_bytes
+        if spill_dir is None:
+            from .config import SPILL_DIR, AuronConf
+
+            spill_dir = AuronConf().get(SPILL_DIR) or None
         self.spill_dir = spill_dir or tempfile.gettempdir()
         self._holders: List[BatchHolder] = []
         self._lock = threading.Lock()

@@ -540,9 +540,11 @@ _DBUF_BYTES = [0]
 
 
 def _dbuf_cache_cap() -> int:
-    env = os.environ.get("AURON_SCAN_HBM_CACHE")
-    if env is not None:
-        return int(env)
+    from .config import SCAN_HBM_CACHE, AuronConf
+
+    v = AuronConf().get(SCAN_HBM_CACHE)
+    if v:
+        return v
     if torch.cuda.is_available():
         _, total = torch.cuda.mem_get_info()
         return int(total * 0.30)
@@ -574,8 +576,10 @@ def _dbuf_cache_put(key, dbuf):
     # HBM would drop below 25% of the device, so big working sets
     # (SF>=100 joins/aggs) reclaim the cache before the allocator OOMs
     if torch.cuda.is_available():
+        from .config import HBM_CACHE_RESERVE, AuronConf
+
         free, total = torch.cuda.mem_get_info()
-        reserve = int(total * 0.25)
+        reserve = int(total * AuronConf().get(HBM_CACHE_RESERVE))
         while free - nbytes < reserve and _DBUF_CACHE:
             _, old = _DBUF_CACHE.popitem(last=False)
             _DBUF_BYTES[0] -= old.numel()
