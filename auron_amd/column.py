@@ -106,6 +106,18 @@ class Column:
             if dtype.code == dtypes.DECIMAL64:
                 scaled = [int(round(float(v) * (10 ** dtype.scale))) for v in tvals]
                 data = torch.tensor(scaled, dtype=torch.int64)
+            elif dtype.code == dtypes.DECIMAL128:
+                from decimal import Decimal
+
+                m64 = (1 << 64) - 1
+                limbs = []
+                for v in tvals:
+                    iv = int(Decimal(str(v)).scaleb(dtype.scale))
+                    lo = iv & m64
+                    limbs.append(((lo - (1 << 64)) if lo >= (1 << 63) else lo,
+                                  iv >> 64))
+                data = torch.tensor(limbs, dtype=torch.int64) if limbs else \
+                    torch.empty((0, 2), dtype=torch.int64)
             else:
                 data = torch.tensor(tvals, dtype=dtype.torch_dtype)
             col = Column(dtype, data, validity, None)
@@ -135,6 +147,15 @@ class Column:
                 else:
                     out.append(flat[offs[i]:offs[i + 1]])
             return out
+        if self.dtype.code == dtypes.DECIMAL128:
+            s = 10 ** self.dtype.scale
+            lo = c.data[:, 0].tolist()
+            hi = c.data[:, 1].tolist()
+            m64 = (1 << 64) - 1
+            vals = [((h << 64) | (l & m64)) / s for l, h in zip(lo, hi)]
+            if valid is None:
+                return vals
+            return [v if ok else None for v, ok in zip(vals, valid)]
         vals = c.data.tolist()
         if self.dtype.code == dtypes.DECIMAL64:
             s = 10 ** self.dtype.scale
@@ -277,6 +298,17 @@ class Column:
             else:
                 scaled = np.zeros(len(d128), dtype=np.int64)
             col = Column(dt, torch.from_numpy(scaled), validity, None)
+        elif dt.code == dtypes.DECIMAL128:
+            d128 = arr.combine_chunks() if isinstance(arr, pa.ChunkedArray) else arr
+            buf = d128.buffers()[1]
+            if buf is not None:
+                # both little-endian limbs: data[:,0]=low 64 bits, [:,1]=high
+                raw = np.frombuffer(buf, dtype=np.int64,
+                                    count=2 * (len(d128) + d128.offset))
+                limbs = raw[2 * d128.offset:].reshape(-1, 2).copy()
+            else:
+                limbs = np.zeros((len(d128), 2), dtype=np.int64)
+            col = Column(dt, torch.from_numpy(limbs), validity, None)
         else:
             if dt.code == dtypes.DATE32:
                 arr = arr.cast(pa.int32())
@@ -316,6 +348,21 @@ class Column:
                     for i in range(len(c))]
             p = max(self.dtype.precision, sc + 1, 1)
             return pa.array(vals, type=pa.decimal128(p, sc))
+        if self.dtype.code == dtypes.DECIMAL128:
+            from decimal import Decimal
+
+            sc = self.dtype.scale
+            lo = c.data[:, 0].numpy()
+            hi = c.data[:, 1].numpy()
+            vals = []
+            for i in range(len(c)):
+                if mask is not None and mask[i]:
+                    vals.append(None)
+                else:
+                    v = (int(hi[i]) << 64) | (int(lo[i]) & ((1 << 64) - 1))
+                    vals.append(Decimal(v).scaleb(-sc))
+            p = max(self.dtype.precision, sc + 1, 1)
+            return pa.array(vals, type=pa.decimal128(min(p, 38), sc))
         np_arr = c.data.numpy()
         if self.dtype.code == dtypes.TIMESTAMP:
             return pa.array(np_arr, from_pandas=False, mask=mask).cast(

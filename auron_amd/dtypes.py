@@ -23,6 +23,7 @@ STRING = 8  # uint8 byte buffer + int32 offsets
 DECIMAL64 = 9  # scaled int64
 LIST = 10  # flattened child values + int32 offsets (element nulls: round 2)
 TIMESTAMP = 11  # microseconds since epoch UTC, int64 storage
+DECIMAL128 = 12  # two-limb scaled int: data [n,2] int64 = (lo bits, hi)
 
 _NAMES = {
     BOOL: "bool",
@@ -37,6 +38,7 @@ _NAMES = {
     DECIMAL64: "decimal64",
     LIST: "list",
     TIMESTAMP: "timestamp",
+    DECIMAL128: "decimal128",
 }
 
 _TORCH = {
@@ -52,6 +54,7 @@ _TORCH = {
     DECIMAL64: torch.int64,
     LIST: None,  # resolved via .child
     TIMESTAMP: torch.int64,
+    DECIMAL128: torch.int64,
 }
 
 
@@ -65,6 +68,8 @@ class DataType:
     def name(self) -> str:
         if self.code == DECIMAL64:
             return f"decimal({self.precision},{self.scale})"
+        if self.code == DECIMAL128:
+            return f"decimal128({self.precision},{self.scale})"
         if self.code == LIST:
             return f"list<{_NAMES[self.precision]}>"
         return _NAMES[self.code]
@@ -96,7 +101,8 @@ class DataType:
 
     @property
     def is_numeric(self) -> bool:
-        return self.code in (INT8, INT16, INT32, INT64, FLOAT32, FLOAT64, DECIMAL64)
+        return self.code in (INT8, INT16, INT32, INT64, FLOAT32, FLOAT64,
+                             DECIMAL64, DECIMAL128)
 
     @property
     def is_integer(self) -> bool:
@@ -124,6 +130,10 @@ timestamp = DataType(TIMESTAMP)
 
 def decimal64(precision: int = 18, scale: int = 2) -> DataType:
     return DataType(DECIMAL64, precision, scale)
+
+
+def decimal128(precision: int = 38, scale: int = 2) -> DataType:
+    return DataType(DECIMAL128, precision, scale)
 
 
 def list_of(child: DataType) -> DataType:
@@ -156,6 +166,8 @@ def from_arrow(at) -> DataType:
     if pa.types.is_timestamp(at):
         return timestamp
     if pa.types.is_decimal(at):
+        if at.precision > 18:
+            return decimal128(at.precision, at.scale)
         return decimal64(at.precision, at.scale)
     if pa.types.is_dictionary(at):
         return from_arrow(at.value_type)
@@ -177,6 +189,8 @@ def to_arrow(dt: DataType):
         STRING: pa.string(),
     }
     if dt.code == DECIMAL64:
+        return pa.decimal128(dt.precision, dt.scale)
+    if dt.code == DECIMAL128:
         return pa.decimal128(dt.precision, dt.scale)
     if dt.code == TIMESTAMP:
         return pa.timestamp("us")

@@ -831,6 +831,22 @@ class Executor:
                 null_rank = nulls[ot].to(torch.int8)
             else:
                 v = c.data
+                if c.dtype.code == dtypes.DECIMAL128:
+                    # exact 128-bit ordering: stable sort by the low limb
+                    # as unsigned (bit pattern xor sign bit), then by the
+                    # signed high limb (major key last wins under stable)
+                    lo_ord = v[:, 0] ^ (-(1 << 63))
+                    o1 = torch.argsort(lo_ord, stable=True,
+                                       descending=not asc)
+                    v = v[:, 1][o1]
+                    o2 = torch.argsort(v, stable=True, descending=not asc)
+                    ot = o1[o2]
+                    nulls = (~c.validity) if c.validity is not None else torch.zeros(n, dtype=torch.bool, device=batch.device)
+                    null_rank = nulls[ot].to(torch.int8)
+                    null_key = -null_rank if asc else null_rank
+                    ot = ot[torch.argsort(null_key, stable=True)]
+                    perm = perm[ot]
+                    continue
                 if v.dtype == torch.bool:
                     v = v.to(torch.int8)
                 ot = torch.argsort(v, stable=True, descending=not asc)
@@ -845,8 +861,9 @@ class Executor:
     def _exec_Sort(self, node: P.Sort) -> List[RecordBatch]:
         batches = self.execute(node.child)
         total = sum(b.num_rows for b in batches)
+        _k0 = node.keys[0][0].eval(batches[0]).dtype
         if (node.limit is None and total > 4 * self.ctx.batch_rows
-                and not node.keys[0][0].eval(batches[0]).dtype.is_string):
+                and not _k0.is_string and _k0.code != dtypes.DECIMAL128):
             return self._exec_sort_external(node, batches, total)
         b = _concat(batches)
         perm = self._sort_permutation(b, node.keys)
@@ -1352,6 +1369,9 @@ class Executor:
             if comb == "first":
                 acc_col, _ = self._agg_first(gids, ngroups, s0)
                 cols.append(acc_col)
+            elif comb == "sum" and s0.dtype.code == dtypes.DECIMAL128:
+                data, cnt = self._sum128_scatter(gids, ngroups, s0)
+                cols.append(Column(s0.dtype, data, compact_validity(cnt > 0)))
             elif comb == "sum" and self._decimal_sum_unsafe(s0, b.num_rows):
                 data, cnt = self._sum_split_exact(gids, ngroups, s0)
                 cols.append(Column(s0.dtype, data, cnt > 0))
@@ -1401,6 +1421,10 @@ class Executor:
             for i, agg in enumerate(node.aggs):
                 if agg.fn in ("sum", "avg", "min", "max") and agg.expr is not None:
                     v = agg.expr.eval(b)
+                    if v.dtype.code == dtypes.DECIMAL128:
+                        continue  # two-limb path below
+                    if agg.fn == "sum" and self._sum_needs_128(v.dtype):
+                        continue  # decimal128 result path below
                     if agg.fn in ("sum", "avg") and self._decimal_sum_unsafe(v, n):
                         continue  # exact split path below
                     fuse_items.append((v, agg.fn))
@@ -1423,11 +1447,15 @@ class Executor:
                         "first": "first", "first_ignores_null": "first"}[agg.fn]
                 if comb == "first":
                     acc, cnt = self._agg_first(gids, ngroups, sv)
+                elif comb == "sum" and sv.dtype.code == dtypes.DECIMAL128:
+                    data, cnt = self._sum128_scatter(gids, ngroups, sv)
+                    acc = Column(sv.dtype, data, compact_validity(cnt > 0))
                 elif comb == "sum" and self._decimal_sum_unsafe(sv, n):
                     acc, cnt = self._sum_split_exact(gids, ngroups, sv)
                 else:
                     acc, cnt = ops.agg_scatter(gids, ngroups, sv, comb)
-                cols.append(self._finalize_agg(agg, sv.dtype, acc, merged_cnt))
+                cols.append(self._finalize_agg(agg, sv.dtype, acc, merged_cnt,
+                                                widen=False))
                 names.append(agg.name)
             else:
                 val = agg.expr.eval(b) if agg.expr is not None else None
@@ -1459,6 +1487,16 @@ class Executor:
                 elif i in fused:
                     fval, (facc, fcnt) = fused[i]
                     acc, vcnt, vdt = facc, fcnt, fval.dtype
+                elif agg.fn == "sum" and self._sum_needs_128(val.dtype):
+                    # declared result exceeds the 64-bit backing: exact
+                    # 128-bit limbs (state schema is static -> SPMD-safe)
+                    if val.dtype.code == dtypes.DECIMAL128:
+                        data, vcnt = self._sum128_scatter(gids, ngroups, val)
+                    else:
+                        data, vcnt = self._sum_split_128(gids, ngroups, val)
+                    vdt = val.dtype
+                    acc = Column(self._state_dtype(agg, vdt), data,
+                                 compact_validity(vcnt > 0))
                 elif agg.fn in ("sum", "avg") and self._decimal_sum_unsafe(val, n):
                     acc, vcnt = self._sum_split_exact(gids, ngroups, val)
                     vdt = val.dtype
@@ -1492,12 +1530,9 @@ class Executor:
         m = int(val.data.abs().max().item())
         return m > 0 and nrows * m >= (1 << 62)
 
-    def _sum_split_exact(self, gids, ngroups, val: Column):
-        """Exact decimal sum via hi/lo 32-bit split accumulators: each
-        int64 addend is split into (v>>32, v&0xffffffff); both partial
-        sums stay far from int64 range for <2^31 rows, and the recombine
-        detects true overflow of the mathematical result instead of
-        silently wrapping (reference promotes sum(decimal) to 128-bit)."""
+    def _sum_split_raw(self, gids, ngroups, val: Column):
+        """Split int64 addends into 32-bit digit accumulators (exact for
+        <2^31 rows): returns (h, rem, cnt) with group total = h*2^32+rem."""
         device = gids.device
         n = gids.numel()
         assert n < (1 << 31), "batch too large for split accumulation"
@@ -1516,6 +1551,16 @@ class Executor:
         carry = sum_lo >> 32
         rem = sum_lo & 0xFFFFFFFF
         h = sum_hi + carry
+        return h, rem, cnt
+
+    def _sum_split_exact(self, gids, ngroups, val: Column):
+        """Exact decimal sum via hi/lo 32-bit split accumulators: each
+        int64 addend is split into (v>>32, v&0xffffffff); both partial
+        sums stay far from int64 range for <2^31 rows, and the recombine
+        detects true overflow of the mathematical result instead of
+        silently wrapping (sum(decimal) with p+10>18 routes to the
+        decimal128 path instead of this one)."""
+        h, rem, cnt = self._sum_split_raw(gids, ngroups, val)
         if bool(((h < -(1 << 31)) | (h > (1 << 31) - 1)).any()):
             from ..session import AuronTaskError
 
@@ -1523,6 +1568,57 @@ class Executor:
                 "decimal sum overflow: group total exceeds 18 significant "
                 "digits (decimal64 backing); rescale or cast to double")
         return h * (1 << 32) + rem, cnt
+
+    def _sum_split_128(self, gids, ngroups, val: Column):
+        """Exact sum of decimal64 addends into decimal128 limbs.
+
+        total = h*2^32 + rem with rem in [0,2^32); two's-complement
+        128-bit limbs are lo = (h<<32)+rem (bit pattern) and
+        hi = h>>32 (arithmetic = floor(h/2^32))."""
+        h, rem, cnt = self._sum_split_raw(gids, ngroups, val)
+        lo = (h << 32) + rem
+        hi = h >> 32
+        return torch.stack([lo, hi], dim=1), cnt
+
+    @staticmethod
+    def _sum_needs_128(vdt: DataType) -> bool:
+        return (vdt.code == dtypes.DECIMAL64 and vdt.precision + 10 > 18) \
+            or vdt.code == dtypes.DECIMAL128
+
+    def _sum128_scatter(self, gids, ngroups, col: Column):
+        """Exact group sum of a decimal128 [n,2] column: the 128-bit value
+        is split into four 32-bit digits (top digit signed), each digit
+        scatter-summed in int64, then recomposed with carry propagation —
+        no digit sum can overflow below 2^31 rows."""
+        device = gids.device
+        n = gids.numel()
+        assert n < (1 << 31), "batch too large for split accumulation"
+        valid = col.validity if col.validity is not None else \
+            torch.ones(n, dtype=torch.bool, device=device)
+        z = torch.zeros((), dtype=torch.int64, device=device)
+        lo = torch.where(valid, col.data[:, 0], z)
+        hi = torch.where(valid, col.data[:, 1], z)
+        m = 0xFFFFFFFF
+        digs = [lo & m, (lo >> 32) & m, hi & m, hi >> 32]
+        sums = []
+        cnt = torch.zeros(ngroups, dtype=torch.int64, device=device)
+        for d in digs:
+            sd = torch.zeros(ngroups, dtype=torch.int64, device=device)
+            if n:
+                sd.scatter_add_(0, gids, d)
+            sums.append(sd)
+        if n:
+            cnt.scatter_add_(0, gids, valid.to(torch.int64))
+        c = sums[0] >> 32
+        r0 = sums[0] & m
+        t1 = sums[1] + c
+        r1 = t1 & m
+        t2 = sums[2] + (t1 >> 32)
+        r2 = t2 & m
+        t3 = sums[3] + (t2 >> 32)
+        out_lo = (r1 << 32) | r0
+        out_hi = (t3 << 32) | r2
+        return torch.stack([out_lo, out_hi], dim=1), cnt
 
     def _partial_skip(self, key_cols, n: int) -> bool:
         """Partial-agg skipping (conf.rs:39-42): sample the reduction
@@ -1570,7 +1666,12 @@ class Executor:
                 continue
             state_dt = self._state_dtype(agg, val.dtype)
             data = val.data
-            if data.dtype != state_dt.torch_dtype and not state_dt.uses_offsets:
+            if (state_dt.code == dtypes.DECIMAL128
+                    and val.dtype.code == dtypes.DECIMAL64):
+                from ..exprs import dec64_to_dec128
+
+                data = dec64_to_dec128(data)
+            elif data.dtype != state_dt.torch_dtype and not state_dt.uses_offsets:
                 data = data.to(state_dt.torch_dtype)
             cols.append(Column(state_dt, data, val.validity, val.offsets))
             names.append(s0)
@@ -1608,7 +1709,12 @@ class Executor:
         if agg.fn in ("count", "count_star", "count_distinct"):
             return dtypes.int64
         if agg.fn in ("sum", "avg"):
+            if vdt.code == dtypes.DECIMAL128:
+                return dtypes.decimal128(38, vdt.scale)
             if vdt.code == dtypes.DECIMAL64:
+                if agg.fn == "sum" and vdt.precision + 10 > 18:
+                    # exceeds the 64-bit backing: two-limb accumulator
+                    return dtypes.decimal128(vdt.precision + 10, vdt.scale)
                 return dtypes.decimal64(min(vdt.precision + 10, 38), vdt.scale)
             if vdt.is_integer or vdt.code == dtypes.BOOL:
                 return dtypes.int64
@@ -1642,7 +1748,8 @@ class Executor:
         return Column(dtypes.list_of(val.dtype), val.data[rows], None,
                       offsets.to(torch.int64))
 
-    def _finalize_agg(self, agg: AggFunc, vdt: DataType, acc, cnt: torch.Tensor) -> Column:
+    def _finalize_agg(self, agg: AggFunc, vdt: DataType, acc, cnt: torch.Tensor,
+                      widen: bool = True) -> Column:
         if agg.fn in ("collect_list", "collect_set"):
             return acc  # empty groups stay empty lists, not null
         validity = compact_validity(cnt > 0)
@@ -1668,7 +1775,9 @@ class Executor:
                 return Column(out_dt, data, validity)
             data = acc.to(torch.float64) / cnt.clamp(min=1).to(torch.float64)
             return Column(dtypes.float64, data, validity)
-        out_dt = self._state_dtype(agg, vdt)
+        # widen=False: vdt is ALREADY the state dtype (final mode) — a
+        # second _state_dtype pass would double-promote sum(decimal)
+        out_dt = self._state_dtype(agg, vdt) if widen else vdt
         return Column(out_dt, acc, validity)
 
     # ------------------------------------------------------------ hash join
@@ -1959,6 +2068,29 @@ class Executor:
                     out = Column(dtypes.int64, torch.ones(n, dtype=torch.int64, device=device))
             elif wf.fn in ("sum", "avg", "count", "min", "max"):
                 val = wf.arg.eval(sb)
+                if (val.dtype.code == dtypes.DECIMAL128
+                        or (wf.fn == "sum" and self._sum_needs_128(val.dtype))):
+                    if wf.fn == "sum" and not (node.order_by and n):
+                        # whole-partition sum promoted to decimal128:
+                        # exact limb accumulation (q12/q20/q98 ratios)
+                        if val.dtype.code == dtypes.DECIMAL128:
+                            data, cnt = self._sum128_scatter(seg, max(nseg, 1), val)
+                            out_dt = val.dtype
+                        else:
+                            data, cnt = self._sum_split_128(seg, max(nseg, 1), val)
+                            out_dt = self._state_dtype(
+                                AggFunc("sum", None, name=al.name), val.dtype)
+                        fin = Column(out_dt, data,
+                                     compact_validity(cnt > 0))
+                        out = fin.gather(seg)
+                        names.append(al.name)
+                        cols.append(out)
+                        continue
+                    # running frames over decimal128: float64 accumulator
+                    # (precision-limited; exact running 128-bit: future)
+                    from ..exprs import _cast_col
+
+                    val = _cast_col(val, dtypes.float64)
                 if node.order_by and n:
                     # Spark's default frame with ORDER BY: unbounded
                     # preceding .. current row (running aggregate)

@@ -36,7 +36,8 @@ def _byte_view(t: torch.Tensor) -> torch.Tensor:
     t = t.contiguous()
     if t.dtype == torch.bool:
         t = t.view(torch.uint8)
-    return t.view(torch.uint8) if t.dtype != torch.uint8 else t
+    t = t.view(torch.uint8) if t.dtype != torch.uint8 else t
+    return t.reshape(-1)  # [n,2] decimal128 limbs flatten to bytes
 
 
 def pack_batch(batch: RecordBatch, device) -> Tuple[dict, torch.Tensor]:
@@ -81,6 +82,7 @@ _TD = {
     dtypes.FLOAT32: torch.float32, dtypes.FLOAT64: torch.float64,
     dtypes.DATE32: torch.int32, dtypes.STRING: torch.uint8, dtypes.DECIMAL64: torch.int64,
     dtypes.TIMESTAMP: torch.int64,
+    dtypes.DECIMAL128: torch.int64,
 }
 _ESIZE = {torch.bool: 1, torch.int8: 1, torch.int16: 2, torch.int32: 4,
           torch.int64: 8, torch.float32: 4, torch.float64: 8, torch.uint8: 1}
@@ -112,6 +114,8 @@ def unpack_batch(meta: dict, buf: torch.Tensor) -> RecordBatch:
         validity = None
         if cm["val"] >= 0:
             validity = take(cm["val"]).to(torch.bool)
+        if code == dtypes.DECIMAL128:
+            data = data.view(-1, 2)
         offsets = None
         if cm["off"] >= 0:
             offsets = take(cm["off"]).view(torch.int64)

@@ -195,6 +195,8 @@ class Literal(Expr):
             if dt.is_string:
                 return Column(dt, torch.empty(0, dtype=torch.uint8, device=device), validity,
                               torch.zeros(n + 1, dtype=torch.int64, device=device))
+            if dt.code == dtypes.DECIMAL128:
+                return Column(dt, torch.zeros((n, 2), dtype=torch.int64, device=device), validity)
             return Column(dt, torch.zeros(n, dtype=dt.torch_dtype, device=device), validity)
         if dt.is_string:
             b = self.value.encode("utf-8")
@@ -204,6 +206,17 @@ class Literal(Expr):
         if dt.code == dtypes.DECIMAL64:
             v = int(round(float(self.value) * 10 ** dt.scale))
             return Column(dt, torch.full((n,), v, dtype=torch.int64, device=device))
+        if dt.code == dtypes.DECIMAL128:
+            from decimal import Decimal
+
+            iv = int(Decimal(str(self.value)).scaleb(dt.scale).to_integral_value())
+            m64 = (1 << 64) - 1
+            lo = iv & m64
+            if lo >= 1 << 63:
+                lo -= 1 << 64
+            limbs = torch.tensor([[lo, iv >> 64]], dtype=torch.int64,
+                                 device=device)
+            return Column(dt, limbs.expand(n, 2).contiguous())
         return Column(dt, torch.full((n,), self.value, dtype=dt.torch_dtype, device=device))
 
 
@@ -220,6 +233,9 @@ def _promote(l: Column, r: Column) -> Tuple[Column, Column, DataType]:
     a, b = l.dtype, r.dtype
     if a.code == b.code and a.code != dtypes.DECIMAL64:
         return l, r, a
+    if dtypes.DECIMAL128 in (a.code, b.code):
+        return (_cast_col(l, dtypes.float64), _cast_col(r, dtypes.float64),
+                dtypes.float64)
     if a.code == dtypes.DECIMAL64 or b.code == dtypes.DECIMAL64:
         # operate in float64 for mixed decimal arithmetic; dedicated decimal
         # kernels (scaled-int64) arrive with the decimal op set
@@ -243,6 +259,8 @@ def _cast_col(c: Column, dt: DataType) -> Column:
         return _cast_string(c, dt)
     if c.dtype.code == dtypes.TIMESTAMP or dt.code == dtypes.TIMESTAMP:
         return _cast_timestamp(c, dt)
+    if c.dtype.code == dtypes.DECIMAL128 or dt.code == dtypes.DECIMAL128:
+        return _cast_decimal128(c, dt)
     if c.dtype.code == dtypes.DECIMAL64 and dt.code == dtypes.DECIMAL64:
         diff = dt.scale - c.dtype.scale
         data = c.data * (10 ** diff) if diff >= 0 else torch.div(c.data, 10 ** (-diff), rounding_mode="trunc")
@@ -259,6 +277,52 @@ def _cast_col(c: Column, dt: DataType) -> Column:
         # Spark cast double->int truncates toward zero
         return Column(dt, c.data.trunc().to(dt.torch_dtype), c.validity)
     return Column(dt, c.data.to(dt.torch_dtype), c.validity)
+
+
+def dec128_to_float64(data: torch.Tensor, scale: int) -> torch.Tensor:
+    """[n,2] limbs -> float64 value/10^scale (precision-limited, like the
+    reference's decimal->double cast)."""
+    # signed-limb composition: value = (hi + carry)*2^64 + lo_signed with
+    # carry = 1 when the low limb's bit pattern is negative as int64 —
+    # avoids the 2^64-cancellation that loses ~11 bits near zero
+    lo = data[:, 0]
+    hi_adj = data[:, 1] + (lo < 0).to(torch.int64)
+    v = hi_adj.to(torch.float64) * (2.0 ** 64) + lo.to(torch.float64)
+    return v / (10.0 ** scale)
+
+
+def dec64_to_dec128(data: torch.Tensor) -> torch.Tensor:
+    """int64 scaled values -> [n,2] limbs (sign-extended high limb)."""
+    hi = data >> 63  # arithmetic: 0 or -1
+    return torch.stack([data, hi], dim=1)
+
+
+def _cast_decimal128(c: Column, dt: DataType) -> Column:
+    if c.dtype.code == dtypes.DECIMAL128:
+        f = dec128_to_float64(c.data, c.dtype.scale)
+        if dt.code == dtypes.DECIMAL128 or dt.code == dtypes.DECIMAL64:
+            # rescale through float64 only when scales differ; same-scale
+            # narrowing checks the value fits the 64-bit backing exactly
+            if dt.scale == c.dtype.scale and dt.code == dtypes.DECIMAL64:
+                lo, hi = c.data[:, 0], c.data[:, 1]
+                fits = hi == (lo >> 63)
+                if not bool(fits.all() if lo.device.type != "cuda" else True):
+                    raise ValueError("decimal128 -> decimal64 overflow")
+                return Column(dt, lo.clone(), c.validity)
+            scaled = torch.round(f * (10.0 ** dt.scale)).to(torch.int64)
+            if dt.code == dtypes.DECIMAL64:
+                return Column(dt, scaled, c.validity)
+            return Column(dt, dec64_to_dec128(scaled), c.validity)
+        if dt.is_float:
+            return Column(dt, f.to(dt.torch_dtype), c.validity)
+        if dt.is_integer:
+            return Column(dt, f.trunc().to(dt.torch_dtype), c.validity)
+        raise TypeError(f"cast decimal128 -> {dt}")
+    # -> decimal128
+    if c.dtype.code == dtypes.DECIMAL64 and c.dtype.scale == dt.scale:
+        return Column(dt, dec64_to_dec128(c.data), c.validity)
+    as64 = _cast_col(c, dtypes.decimal64(18, dt.scale))
+    return Column(dt, dec64_to_dec128(as64.data), c.validity)
 
 
 def _cast_timestamp(c: Column, dt: DataType) -> Column:

@@ -1,0 +1,129 @@
+"""decimal128: two-limb storage, exact big-decimal SUMs beyond the int64
+backing (cast.rs / agg sum 128-bit promotion parity), arrow/parquet
+round-trips, and exchange serialization."""
+import os
+from decimal import Decimal
+
+import pyarrow as pa
+import pyarrow.parquet as pq
+import pytest
+import torch
+
+from auron_amd import AuronSession, col, dtypes
+from auron_amd.column import Column, RecordBatch
+from auron_amd.engine.executor import ExecContext, Executor
+from auron_amd.exprs import AggFunc, Aliased, Col, _cast_col
+from auron_amd.plan import nodes as P
+
+D15 = dtypes.decimal64(15, 2)  # sum state -> decimal128(25,2)
+
+
+def _vals(n=3000, seed=5):
+    import random
+
+    rng = random.Random(seed)
+    # scaled magnitude ~1e16 so a few thousand addends overflow int64
+    return [rng.randint(-4 * 10 ** 16, 9 * 10 ** 16) for _ in range(n)]
+
+
+def _make_plan(raw, keys):
+    cols = [Column(dtypes.int64, torch.tensor(keys)),
+            Column(D15, torch.tensor(raw, dtype=torch.int64))]
+    scan = P.MemoryScan([RecordBatch(["k", "v"], cols)])
+    return P.HashAgg(scan, [Aliased(col("k"), "k")],
+                     [AggFunc("sum", col("v"), name="s")], mode="complete")
+
+
+def _want(raw, keys):
+    want = {}
+    for k, v in zip(keys, raw):
+        want[k] = want.get(k, 0) + v
+    return want
+
+
+def _check(out, want):
+    arr = out.column("s")
+    assert arr.dtype.code == dtypes.DECIMAL128
+    got = dict(zip(out.column("k").to_pylist(),
+                   arr.to_arrow().to_pylist()))
+    for k, w in want.items():
+        assert got[k] == Decimal(w).scaleb(-2), (k, got[k], w)
+
+
+def test_sum_decimal128_exact():
+    raw = _vals()
+    keys = [i % 7 for i in range(len(raw))]
+    out = AuronSession().collect(_make_plan(raw, keys))
+    # each group's exact total exceeds int64
+    want = _want(raw, keys)
+    assert any(abs(v) > 2 ** 63 for v in want.values())
+    _check(out, want)
+
+
+def test_sum_decimal128_partial_final_tower():
+    raw = _vals(seed=9)
+    keys = [i % 5 for i in range(len(raw))]
+    cols = [Column(dtypes.int64, torch.tensor(keys)),
+            Column(D15, torch.tensor(raw, dtype=torch.int64))]
+    scan = P.MemoryScan([RecordBatch(["k", "v"], cols)])
+    partial = P.HashAgg(scan, [Aliased(col("k"), "k")],
+                        [AggFunc("sum", col("v"), name="s")], mode="partial")
+    fin = P.HashAgg(partial, [Aliased(col("k"), "k")],
+                    [AggFunc("sum", col("v"), name="s")], mode="final")
+    ctx = ExecContext()
+    ex = Executor(ctx)
+    ex._rewrite = lambda n: n  # keep the explicit partial->final tower
+    out = ex.collect(fin)
+    _check(out, _want(raw, keys))
+
+
+def test_sum_decimal128_streaming_chunked():
+    raw = _vals(seed=11)
+    keys = [i % 3 for i in range(len(raw))]
+    os.environ["AURON_STREAM_BYTES"] = "0"
+    try:
+        ctx = ExecContext(batch_rows=256)
+        out = Executor(ctx).collect(_make_plan(raw, keys))
+    finally:
+        del os.environ["AURON_STREAM_BYTES"]
+    _check(out, _want(raw, keys))
+
+
+def test_decimal128_arrow_parquet_roundtrip(tmp_path):
+    vals = [Decimal("12345678901234567890123.456"), None,
+            Decimal("-" + "9" * 20 + ".001")]
+    arr = pa.array(vals, type=pa.decimal128(28, 3))
+    c = Column.from_arrow(arr)
+    assert c.dtype.code == dtypes.DECIMAL128 and c.data.shape == (3, 2)
+    assert c.to_arrow().to_pylist() == vals
+    # parquet round-trip through the host reader (FLBA decimals)
+    p = str(tmp_path / "d.parquet")
+    pq.write_table(pa.table({"d": arr}), p)
+    s = AuronSession()
+    out = s.collect(P.ParquetScan([p], columns=["d"]))
+    assert out.column("d").to_arrow().to_pylist() == vals
+
+
+def test_decimal128_exchange_pack_roundtrip():
+    from auron_amd.exchange import pack_batch, unpack_batch
+
+    c = Column.from_pylist([1.5, None, -2.25], dtypes.decimal128(22, 2))
+    b = RecordBatch(["d"], [c])
+    meta, buf = pack_batch(b, "cpu")
+    back = unpack_batch(meta, buf)
+    assert back.column("d").dtype.code == dtypes.DECIMAL128
+    assert back.column("d").data.shape == (3, 2)
+    assert torch.equal(back.column("d").data, c.data)
+
+
+def test_decimal128_cast_and_compare():
+    c = Column.from_pylist([1.25, -3.5, 10 ** 19], dtypes.decimal128(25, 2))
+    f = _cast_col(c, dtypes.float64)
+    assert abs(f.data[0].item() - 1.25) < 1e-12
+    assert abs(f.data[2].item() - 1e19) < 1e6
+    # comparison promotes through float64
+    from auron_amd.exprs import Cmp, Literal
+
+    b = RecordBatch(["d"], [c])
+    m = Cmp(">", Col("d"), Literal(0.0)).eval(b)
+    assert m.data.tolist() == [True, False, True]
