@@ -67,6 +67,8 @@ def _try_load():
     lib.au_host_rle1_parse.restype = i64
     lib.au_host_plainba_parse.argtypes = [c, i64, i64, c, c, i64]
     lib.au_host_plainba_parse.restype = i64
+    lib.au_host_delta_unpack.argtypes = [c, i64, i64, i64, ctypes.c_int, c]
+    lib.au_host_delta_unpack.restype = i64
     lib.au_pq_rle_idx.argtypes = [c, ctypes.c_int, c, c, c, c]
     lib.au_pq_scatter.argtypes = [c, ctypes.c_int, c, c, c, c, ctypes.c_int, i64, c]
     lib.au_pq_copy_plain.argtypes = [c, ctypes.c_int, c, c, ctypes.c_int, i64, c]

@@ -475,3 +475,78 @@ AU_EXPORT int64_t au_host_plainba_parse(const uint8_t* buf, int64_t off,
   }
   return n;
 }
+
+// Host-side DELTA_BINARY_PACKED decode (parquet encoding 5) into PLAIN
+// little-endian values (esize 4 or 8). The format is inherently serial
+// (varint headers, running prefix value); decoded pages then ride the
+// same device PLAIN path as everything else. Returns values written,
+// -1 on corruption.
+static inline int64_t du_uvarint(const uint8_t* b, int64_t end, int64_t* pos,
+                                 uint64_t* out) {
+  uint64_t v = 0; int shift = 0;
+  while (*pos < end) {
+    uint8_t x = b[(*pos)++];
+    v |= (uint64_t)(x & 0x7F) << shift;
+    if (!(x & 0x80)) { *out = v; return 0; }
+    shift += 7;
+  }
+  return -1;
+}
+
+AU_EXPORT int64_t au_host_delta_unpack(const uint8_t* buf, int64_t off,
+                                       int64_t len, int64_t n, int esize,
+                                       uint8_t* out) {
+  int64_t pos = off, end = off + len;
+  uint64_t block_size, nmini, total, ufirst;
+  if (du_uvarint(buf, end, &pos, &block_size)) return -1;
+  if (du_uvarint(buf, end, &pos, &nmini)) return -1;
+  if (du_uvarint(buf, end, &pos, &total)) return -1;
+  if (du_uvarint(buf, end, &pos, &ufirst)) return -1;
+  if (nmini == 0 || block_size % nmini) return -1;
+  // `n` is the output CAPACITY (page row count); the page encodes
+  // `total` values (= present values when the column has nulls)
+  if ((int64_t)total < n) n = (int64_t)total;
+  if (n <= 0) return 0;
+  int64_t per_mini = (int64_t)(block_size / nmini);
+  int64_t value = (int64_t)((ufirst >> 1) ^ -(int64_t)(ufirst & 1));
+  int64_t count = 0;
+  #define EMIT(v) do { \
+    if (esize == 4) { int32_t t = (int32_t)(v); __builtin_memcpy(out + 4 * count, &t, 4); } \
+    else { int64_t t = (v); __builtin_memcpy(out + 8 * count, &t, 8); } \
+    count++; } while (0)
+  EMIT(value);
+  while (count < n) {
+    uint64_t umin;
+    if (du_uvarint(buf, end, &pos, &umin)) return -1;
+    int64_t min_delta = (int64_t)((umin >> 1) ^ -(int64_t)(umin & 1));
+    if (pos + (int64_t)nmini > end) return -1;
+    const uint8_t* bws = buf + pos;
+    pos += nmini;
+    for (uint64_t mb = 0; mb < nmini && count < n; mb++) {
+      int bw = bws[mb];
+      if (bw > 64) return -1;
+      int64_t nbytes = (per_mini * bw + 7) / 8;
+      if (pos + nbytes > end) return -1;
+      // LSB-first bit unpack
+      int64_t bit = 0;
+      for (int64_t i = 0; i < per_mini && count < n; i++) {
+        uint64_t d = 0;
+        for (int got = 0; got < bw; ) {
+          int64_t byte_i = pos + (bit >> 3);
+          int within = (int)(bit & 7);
+          int take = 8 - within;
+          if (take > bw - got) take = bw - got;
+          uint64_t bits = ((uint64_t)buf[byte_i] >> within) & ((1ULL << take) - 1);
+          d |= bits << got;
+          got += take;
+          bit += take;
+        }
+        value += min_delta + (int64_t)d;
+        EMIT(value);
+      }
+      pos += nbytes;
+    }
+  }
+  #undef EMIT
+  return count;
+}

@@ -204,6 +204,14 @@ class _ExtraAlloc:
         self.cursor += n
         return dest
 
+    def delta(self, src: int, length: int, nv: int, esize: int) -> int:
+        """DELTA_BINARY_PACKED page payload -> PLAIN values at dest
+        (host decode, au_host_delta_unpack)."""
+        dest = self.cursor
+        self.jobs.append(("delta", src, length, dest, nv, esize))
+        self.cursor += nv * esize
+        return dest
+
 
 def _codec_decompress(codec: str, raw: bytes, ulen: int) -> bytes:
     import pyarrow as pa
@@ -214,7 +222,8 @@ def _codec_decompress(codec: str, raw: bytes, ulen: int) -> bytes:
 
 def parse_pages(buf: np.ndarray, chunk_off: int, chunk_len: int,
                 num_values: int, has_def: bool, codec: Optional[str] = None,
-                extra: Optional[_ExtraAlloc] = None) -> Optional[ChunkPages]:
+                extra: Optional[_ExtraAlloc] = None,
+                esize: int = 0) -> Optional[ChunkPages]:
     """Parse page headers of one column chunk (v1 + v2 data pages).
 
     Supports PLAIN data pages and RLE_DICTIONARY/PLAIN_DICTIONARY data
@@ -263,6 +272,8 @@ def parse_pages(buf: np.ndarray, chunk_off: int, chunk_len: int,
             enc = dph.get("encoding", 0)
             if enc in (2, 8):
                 enc = 8
+            elif enc == 5 and esize in (4, 8) and extra is not None:
+                pass  # DELTA_BINARY_PACKED: host-decoded to PLAIN below
             elif enc != 0:
                 return None
             nv = dph["num_values"]
@@ -284,6 +295,10 @@ def parse_pages(buf: np.ndarray, chunk_off: int, chunk_len: int,
                 def_off = page_data if dlen else -1
                 values_off = vals_raw
                 values_len = vals_clen
+            if enc == 5:
+                values_off = extra.delta(values_off, values_len, nv, esize)
+                values_len = nv * esize
+                enc = 0
             ck.pages.append(PageDesc(nv, def_off, dlen, values_off, values_len,
                                      row, enc))
             row += nv
@@ -301,6 +316,8 @@ def parse_pages(buf: np.ndarray, chunk_off: int, chunk_len: int,
             enc = 8
             if ck.dict_off < 0:
                 return None
+        elif enc == 5 and esize in (4, 8) and extra is not None:
+            pass  # DELTA_BINARY_PACKED: host-decoded to PLAIN below
         elif enc != 0:
             return None
         nv = dph["num_values"]
@@ -324,6 +341,11 @@ def parse_pages(buf: np.ndarray, chunk_off: int, chunk_len: int,
                                      row, enc))
         else:
             ck.pages.append(PageDesc(nv, -1, 0, page_data, page_len, row, enc))
+        if enc == 5:
+            p = ck.pages[-1]
+            dest = extra.delta(p.values_off, p.values_len, nv, esize)
+            ck.pages[-1] = PageDesc(p.n_values, p.def_off, p.def_len, dest,
+                                    nv * esize, p.row_start, 0)
         row += nv
         pos = page_data + page_len
     if row != num_values or not ck.pages:
@@ -824,6 +846,16 @@ def read_columns_native(path: str, columns: List[str], device,
                 buf[dest:dest + ulen] = np.frombuffer(
                     _codec_decompress(codec, bytes(buf[s0:s0 + clen]), ulen),
                     dtype=np.uint8)
+            elif job[0] == "delta":
+                from . import native
+
+                _, s0, length, dest, nv, esize = job
+                lib = native.host_lib()
+                got = lib.au_host_delta_unpack(
+                    buf.ctypes.data, s0, length, nv, esize,
+                    buf.ctypes.data + dest)
+                if got < 0 or got > nv:
+                    raise ValueError("corrupt DELTA_BINARY_PACKED page")
             else:
                 _, s0, n, dest = job
                 buf[dest:dest + n] = buf[s0:s0 + n]
@@ -846,8 +878,10 @@ def read_columns_native(path: str, columns: List[str], device,
                 good = True
                 j0 = len(extra.jobs)
                 for (new_off, clen, nvals, _an, codec) in cm.chunks:
+                    es = _PHYS_NP[cm.phys]().itemsize \
+                        if cm.phys in _PHYS_NP else 0
                     ck = parse_pages(buf, new_off, clen, nvals, cm.has_def,
-                                     codec, extra)
+                                     codec, extra, esize=es)
                     if ck is None:
                         good = False
                         break

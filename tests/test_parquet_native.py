@@ -219,3 +219,44 @@ def test_gpu_dict_decode(dict_file):
     assert cols is not None and set(cols) == set(t.schema.names)
     assert all(c.data.is_cuda for c in cols.values())
     _check({k: v.to("cpu") for k, v in cols.items()}, t)
+
+
+@pytest.mark.parametrize("codec", ["NONE", "zstd"])
+@pytest.mark.parametrize("pv", ["1.0", "2.0"])
+def test_delta_binary_packed_native(tmp_path, codec, pv):
+    import random
+
+    rng = random.Random(4)
+    xs = [rng.randint(-10 ** 9, 10 ** 9) if rng.random() > 0.1 else None
+          for _ in range(7000)]
+    ys = [rng.randint(0, 2 ** 40) for _ in range(7000)]
+    t = pa.table({"x": pa.array(xs, pa.int32()), "y": pa.array(ys, pa.int64())})
+    p = str(tmp_path / "delta.parquet")
+    pq.write_table(t, p, use_dictionary=False, compression=codec,
+                   column_encoding={"x": "DELTA_BINARY_PACKED",
+                                    "y": "DELTA_BINARY_PACKED"},
+                   data_page_version=pv)
+    cols = parquet_native.read_columns_native(p, ["x", "y"], "cpu",
+                                              _np_only=True)
+    assert cols is not None
+    assert cols["x"].to_pylist() == xs
+    assert cols["y"].to_pylist() == ys
+
+
+@pytest.mark.gpu
+def test_delta_binary_packed_gpu(tmp_path):
+    import random
+
+    from auron_amd import native
+
+    native.require()
+    rng = random.Random(6)
+    xs = [rng.randint(-10 ** 6, 10 ** 6) if rng.random() > 0.2 else None
+          for _ in range(30000)]
+    t = pa.table({"x": pa.array(xs, pa.int64())})
+    p = str(tmp_path / "dg.parquet")
+    pq.write_table(t, p, use_dictionary=False, compression="snappy",
+                   column_encoding={"x": "DELTA_BINARY_PACKED"})
+    cols = parquet_native.read_columns_native(p, ["x"], "cuda")
+    assert cols is not None
+    assert cols["x"].to("cpu").to_pylist() == xs
