@@ -6,13 +6,16 @@ footer via pyarrow, page headers via a minimal thrift-compact parser —
 and ships the raw column-chunk bytes to HBM once; gfx950 kernels decode
 RLE/bit-packed definition levels, RLE dictionary indices, and PLAIN
 values into device columns (csrc/parquet.hip). Unsupported shapes
-(compressed pages, nested, PLAIN byte-arrays) fall back to the pyarrow
+(nested lists/maps/structs, repeated levels) fall back to the pyarrow
 host path per column.
 
-Supported fast path: UNCOMPRESSED column chunks, data page v1,
-optional-level (max_def_level<=1) columns; PLAIN or RLE_DICTIONARY
-INT32/INT64/FLOAT/DOUBLE, and dictionary-encoded BYTE_ARRAY strings
-(dictionary page parsed host-side, per-row bytes gathered on device).
+Supported fast path: data pages v1+v2, UNCOMPRESSED/snappy/zstd/gzip
+chunks (host page decompression into the staging buffer's extra
+region), optional-level (max_def_level<=1) columns; PLAIN,
+RLE_DICTIONARY and DELTA_BINARY_PACKED (host delta unpack -> PLAIN)
+INT32/INT64/FLOAT/DOUBLE; BYTE_ARRAY strings both dictionary-encoded
+(dictionary page parsed host-side, per-row bytes gathered on device)
+and PLAIN (host offset walk, device byte gather).
 """
 from __future__ import annotations
 
