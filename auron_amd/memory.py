@@ -21,15 +21,17 @@ import torch
 from .column import Column, RecordBatch
 
 
-def _batch_bytes(b: RecordBatch) -> int:
-    total = 0
-    for c in b.columns:
-        total += c.data.numel() * c.data.element_size()
-        if c.validity is not None:
-            total += c.validity.numel()
-        if c.offsets is not None:
-            total += c.offsets.numel() * 8
+def _col_bytes(c: Column) -> int:
+    total = c.data.numel() * c.data.element_size()
+    if c.validity is not None:
+        total += c.validity.numel()
+    if c.offsets is not None:
+        total += c.offsets.numel() * 8
     return total
+
+
+def _batch_bytes(b: RecordBatch) -> int:
+    return sum(_col_bytes(c) for c in b.columns)
 
 
 class BatchHolder:
