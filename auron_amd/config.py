@@ -59,9 +59,15 @@ BROADCAST_CACHE_BYTES = _opt("spark.auron.broadcast.cache.maxBytes", 4 << 30, in
 FORCE_SHUFFLED_HASH_JOIN = _opt("spark.auron.forceShuffledHashJoin", False, bool,
                                 "never broadcast: hash-exchange both sides",
                                 env="AURON_FORCE_SHJ")
-EXPR_FUSION = _opt("spark.auron.expr.fusion.enable", True, bool,
+EXPR_FUSION = _opt("spark.auron.expr.fusion.enable", False, bool,
                    "compile project/filter scalar expression trees into one "
-                   "fused interpreter kernel launch",
+                   "fused interpreter kernel launch (k_expr_exec). Default "
+                   "off: the kernel itself measures ~24us/launch but the "
+                   "per-batch host dispatch costs more than the at::native "
+                   "launches it replaces at TPC-DS shapes (same-box A/B: "
+                   "SF=10 suite 8.24s on vs 6.52s off); worthwhile only for "
+                   "very wide scalar-heavy projections until the native "
+                   "plan walker absorbs the dispatch",
                    env="AURON_EXPR_FUSION")
 AGG_STREAMING = _opt("spark.auron.agg.streaming.enable", True, bool,
                      "chunked spill-capable partial aggregation",
