@@ -1061,3 +1061,91 @@ class TruncTimestamp(Expr):
             raise ValueError(f"date_trunc unit {self.unit}")
         data = _days_from_civil(y, m, d) * _US_DAY
         return Column(dtypes.timestamp, data, c.validity)
+
+
+# ----------------------------------------------------- decimal utilities
+@dataclass(eq=False)
+class UnscaledValue(Expr):
+    """spark_unscaled_value: decimal -> raw unscaled int64."""
+    child: Expr
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        if c.dtype.code == dtypes.DECIMAL64:
+            return Column(dtypes.int64, c.data.clone(), c.validity)
+        if c.dtype.code == dtypes.DECIMAL128:
+            lo, hi = c.data[:, 0], c.data[:, 1]
+            fits = hi == (lo >> 63)
+            v = c.validity if c.validity is not None else torch.ones(
+                len(c), dtype=torch.bool, device=c.device)
+            return Column(dtypes.int64, lo.clone(), compact_validity(v & fits))
+        raise TypeError("unscaled_value expects a decimal input")
+
+
+@dataclass(eq=False)
+class MakeDecimal(Expr):
+    """spark_make_decimal: unscaled int64 -> decimal(p,s) (inverse of
+    UnscaledValue; nulls for values whose digits exceed p)."""
+    child: Expr
+    precision: int = 18
+    scale: int = 2
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        data = c.data.to(torch.int64)
+        limit = 10 ** min(self.precision, 18)
+        ok = (data > -limit) & (data < limit)
+        v = c.validity
+        v = ok if v is None else (v & ok)
+        return Column(dtypes.decimal64(self.precision, self.scale), data,
+                      compact_validity(v))
+
+
+@dataclass(eq=False)
+class CheckOverflow(Expr):
+    """spark_check_overflow: null out decimal values whose digit count
+    exceeds the declared precision (non-ANSI overflow semantics)."""
+    child: Expr
+    precision: int = 18
+    scale: int = 2
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        if c.dtype.code == dtypes.DECIMAL64:
+            limit = 10 ** min(self.precision, 18)
+            ok = (c.data > -limit) & (c.data < limit)
+            v = c.validity
+            v = ok if v is None else (v & ok)
+            return Column(dtypes.decimal64(self.precision, c.dtype.scale),
+                          c.data, compact_validity(v))
+        return c
+
+
+# ------------------------------------------------------------- randomness
+@dataclass(eq=False)
+class Rand(Expr):
+    """rand([seed]): uniform [0,1) float64; per-partition generator seeded
+    by (seed, partition_id) like SparkRandn — deterministic per task."""
+    seed: int = 42
+
+    def eval(self, batch):
+        ctx = EVAL_CONTEXT.get()
+        g = torch.Generator(device="cpu")
+        g.manual_seed((self.seed << 16) ^ ctx.get("partition_id", 0)
+                      ^ ctx.get("batch_ordinal", 0))
+        data = torch.rand(batch.num_rows, dtype=torch.float64, generator=g)
+        return Column(dtypes.float64, data.to(batch.device))
+
+
+@dataclass(eq=False)
+class Randn(Expr):
+    """randn([seed]): standard normal float64 (SparkRandn parity)."""
+    seed: int = 42
+
+    def eval(self, batch):
+        ctx = EVAL_CONTEXT.get()
+        g = torch.Generator(device="cpu")
+        g.manual_seed((self.seed << 16) ^ ctx.get("partition_id", 0)
+                      ^ ctx.get("batch_ordinal", 0) ^ 0x5EED)
+        data = torch.randn(batch.num_rows, dtype=torch.float64, generator=g)
+        return Column(dtypes.float64, data.to(batch.device))

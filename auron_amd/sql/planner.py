@@ -1671,6 +1671,16 @@ class Planner:
             return Cast(conv(a[0]), dtypes.string)
         if name == "xxhash64":
             return F.XxHash64([conv(x) for x in a])
+        if name == "rand":
+            return F.Rand(_int_lit(conv(a[0])) if a else 42)
+        if name == "randn":
+            return F.Randn(_int_lit(conv(a[0])) if a else 42)
+        if name == "unscaled_value":
+            return F.UnscaledValue(conv(a[0]))
+        if name == "make_decimal":
+            p = _int_lit(conv(a[1])) if len(a) > 1 else 18
+            sc = _int_lit(conv(a[2])) if len(a) > 2 else 2
+            return F.MakeDecimal(conv(a[0]), p, sc)
         return None
 
     # --------------------------------------------- scalar subqueries

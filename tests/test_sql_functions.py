@@ -135,3 +135,11 @@ def test_timestamp_group_and_join_keys():
                      [AggFunc("sum", Col("v"), name="sv")], mode="complete")
     out = s.collect(plan).to_pydict()
     assert sorted(out["sv"]) == [3, 3]
+
+
+def test_decimal_utils_and_rand(env):
+    out = run(env, "select unscaled_value(cast(x as decimal(7,2))) uv, "
+                   "rand(5) r1, randn(5) r2 from tt order by uv")
+    assert out["uv"] == [100, 200, 300]
+    assert all(0.0 <= v < 1.0 for v in out["r1"])
+    assert len(set(out["r2"])) == 3  # distinct normals
