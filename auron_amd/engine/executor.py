@@ -2128,7 +2128,14 @@ class Executor:
                     from ..exprs import _cast_col
 
                     val = _cast_col(val, dtypes.float64)
-                if node.order_by and n:
+                bounded = (node.frame == "rows"
+                           and not (node.frame_lo is None
+                                    and node.frame_hi == 0))
+                if bounded and n:
+                    out = self._bounded_rows_window(
+                        wf.fn, val, seg, seg_start,
+                        node.frame_lo, node.frame_hi)
+                elif node.order_by and n:
                     # Spark's default frame with ORDER BY: unbounded
                     # preceding .. current row (running aggregate)
                     out = self._running_window(wf.fn, val, seg, seg_start)
@@ -2248,6 +2255,73 @@ class Executor:
         pend = torch.cat([pstart[1:] - 1,
                           torch.tensor([n - 1], dtype=torch.int64, device=device)])
         return pend[pg]
+
+    def _bounded_rows_window(self, fn: str, val: Column, seg: torch.Tensor,
+                             seg_start: torch.Tensor, lo, hi) -> Column:
+        """Explicit ROWS BETWEEN bounds (window_exec.rs frame processors):
+        sum/avg/count via segment-clamped prefix sums; min/max via a
+        shifted-compare chain over the (bounded) span."""
+        device = seg.device
+        n = seg.numel()
+        counts = torch.bincount(seg, minlength=int(seg.max().item()) + 1 if n else 1)
+        first = seg_start[seg]
+        last = first + counts[seg] - 1
+        i = torch.arange(n, dtype=torch.int64, device=device)
+        a = first if lo is None else torch.maximum(i + int(lo), first)
+        b = last if hi == "U" else torch.minimum(i + int(hi), last)
+        empty = a > b
+        valid = val.validity if val.validity is not None else \
+            torch.ones(n, dtype=torch.bool, device=device)
+        if fn in ("sum", "avg", "count"):
+            vd = val.data
+            if vd.dtype not in (torch.float64, torch.float32):
+                vd = vd.to(torch.int64)
+            x = torch.where(valid, vd, torch.zeros_like(vd))
+            S = torch.cumsum(x, 0)
+            C = torch.cumsum(valid.to(torch.int64), 0)
+
+            def rng(P, dt):
+                hi_v = P[b.clamp(min=0)]
+                lo_i = a - 1
+                lo_v = torch.where(lo_i >= 0, P[lo_i.clamp(min=0)],
+                                   torch.zeros((), dtype=P.dtype, device=device))
+                out = hi_v - lo_v
+                return torch.where(empty, torch.zeros_like(out), out)
+
+            cnt = rng(C, torch.int64)
+            if fn == "count":
+                return Column(dtypes.int64, cnt)
+            s = rng(S, vd.dtype)
+            if fn == "avg":
+                data = s.to(torch.float64) / cnt.clamp(min=1).to(torch.float64)
+                if val.dtype.code == dtypes.DECIMAL64:
+                    data = data / (10.0 ** val.dtype.scale)
+                return Column(dtypes.float64, data,
+                              compact_validity(cnt > 0))
+            out_dt = val.dtype if val.dtype.code == dtypes.DECIMAL64 else (
+                dtypes.float64 if val.dtype.is_float else dtypes.int64)
+            return Column(out_dt, s.to(out_dt.torch_dtype),
+                          compact_validity(cnt > 0))
+        # min/max: chain of shifted comparisons over the span
+        assert isinstance(lo, int) and isinstance(hi, int) and hi - lo < 1024, \
+            "bounded ROWS min/max supports spans < 1024"
+        cmp = torch.minimum if fn == "min" else torch.maximum
+        acc = None
+        acc_ok = torch.zeros(n, dtype=torch.bool, device=device)
+        for off in range(lo, hi + 1):
+            j = (i + off).clamp(0, max(n - 1, 0))
+            ok = (i + off >= first) & (i + off <= last) & valid[j]
+            v = val.data[j]
+            if acc is None:
+                acc = v.clone()
+                acc_ok = ok.clone()
+            else:
+                take_new = ok & (~acc_ok)
+                both = ok & acc_ok
+                acc = torch.where(take_new, v, acc)
+                acc = torch.where(both, cmp(acc, v), acc)
+                acc_ok = acc_ok | ok
+        return Column(val.dtype, acc, compact_validity(acc_ok))
 
     def _running_window(self, fn: str, val: Column, seg: torch.Tensor,
                         seg_start: torch.Tensor) -> Column:

@@ -192,6 +192,10 @@ class Window(PlanNode):
     order_by: List[Tuple[Expr, bool]]
     functions: List[Aliased]  # WindowFunc exprs aliased to output names
     frame: str = "range"
+    # explicit ROWS bounds: None = unbounded preceding, "U" = unbounded
+    # following, int = signed row offset; defaults = the SQL default frame
+    frame_lo: object = None
+    frame_hi: object = 0
 
     def children(self):
         return [self.child]

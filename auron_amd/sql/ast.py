@@ -66,7 +66,11 @@ class FuncCall(ANode):
 class WindowSpec(ANode):
     partition_by: List[ANode] = field(default_factory=list)
     order_by: List["OrderItem"] = field(default_factory=list)
-    frame: Optional[str] = None  # "rows" | "range" (only default bounds used)
+    frame: Optional[str] = None  # "rows" | "range"
+    # bounds: None = unbounded preceding, "U" = unbounded following,
+    # int = signed row offset (defaults = the SQL default frame)
+    frame_lo: object = None
+    frame_hi: object = 0
 
 
 @dataclass

@@ -517,14 +517,33 @@ class Parser:
             spec.order_by = self._order_by()
         if self.at_kw("rows", "range"):
             spec.frame = self.advance().text
-            # only the default frame (UNBOUNDED PRECEDING .. CURRENT ROW)
-            # appears in TPC-DS; parse and verify
+
+            def bound():
+                # -> None (unbounded preceding), "U" (unbounded following),
+                #    or signed row offset (preceding<0, current=0, following>0)
+                if self.accept_kw("unbounded"):
+                    if self.accept_kw("preceding"):
+                        return None
+                    self.expect_kw("following")
+                    return "U"
+                if self.accept_kw("current"):
+                    self.expect_kw("row")
+                    return 0
+                if self.cur.kind != "num":
+                    self.fail("expected frame bound")
+                k = int(self.advance().text)
+                if self.accept_kw("preceding"):
+                    return -k
+                self.expect_kw("following")
+                return k
+
             self.expect_kw("between")
-            self.expect_kw("unbounded")
-            self.expect_kw("preceding")
+            spec.frame_lo = bound()
             self.expect_kw("and")
-            self.expect_kw("current")
-            self.expect_kw("row")
+            spec.frame_hi = bound()
+            if spec.frame == "range" and not (
+                    spec.frame_lo is None and spec.frame_hi == 0):
+                self.fail("RANGE frames support only the default bounds")
         self.expect_op(")")
         return spec
 

@@ -181,7 +181,8 @@ class Lowering:
                 part = SINGLE
         order = [(self.subst_scalars(e), asc) for e, asc in node.order]
         funcs = [Aliased(wf, n) for wf, n in node.funcs]
-        return P.Window(child, partition, order, funcs, frame=node.frame), part
+        return P.Window(child, partition, order, funcs, frame=node.frame,
+                        frame_lo=node.frame_lo, frame_hi=node.frame_hi), part
 
     def _lower_LJoin(self, node: LJoin):
         lplan, lpart = self.lower(node.left)

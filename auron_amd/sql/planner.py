@@ -104,6 +104,8 @@ class LWindow(LNode):
     order: List[Tuple[Expr, bool]]
     funcs: List[Tuple[WindowFunc, str]]
     frame: str = "range"
+    frame_lo: object = None
+    frame_hi: object = 0
 
 
 @dataclass
@@ -1207,7 +1209,9 @@ class Planner:
                      for oi in spec.order_by]
             key = (tuple(expr_key(p) for p in part),
                    tuple((expr_key(e), a) for e, a in order),
-                   spec.frame or "range")
+                   spec.frame or "range",
+                   getattr(spec, "frame_lo", None),
+                   getattr(spec, "frame_hi", 0))
             groups.setdefault(key, []).append((fc, name, part, order))
         node = rel.node
         cols = list(rel.cols)
@@ -1221,7 +1225,8 @@ class Planner:
                 funcs.append((wf, name))
                 self.alloc.set_type(name, _window_dtype(fc, wf, self.alloc.types))
                 cols.append(RelCol(None, name, name, self.alloc.types.get(name)))
-            node = LWindow(node, part, order, funcs, frame)
+            node = LWindow(node, part, order, funcs, frame,
+                           key[3], key[4])
         out = Rel(node, cols, rel.est, rel.base_dim_only)
         return out, Scope([out], scope.parent)
 

@@ -143,3 +143,17 @@ def test_decimal_utils_and_rand(env):
     assert out["uv"] == [100, 200, 300]
     assert all(0.0 <= v < 1.0 for v in out["r1"])
     assert len(set(out["r2"])) == 3  # distinct normals
+
+
+def test_bounded_rows_frames(env):
+    out = run(env, "select x, "
+                   "sum(x) over (order by x rows between 1 preceding and "
+                   "1 following) s, "
+                   "min(x) over (order by x rows between 2 preceding and "
+                   "current row) mn, "
+                   "count(x) over (order by x rows between current row and "
+                   "unbounded following) cf "
+                   "from tt order by x")
+    assert out["s"] == [3, 6, 5]
+    assert out["mn"] == [1, 1, 1]
+    assert out["cf"] == [3, 2, 1]
