@@ -127,3 +127,44 @@ def test_decimal128_cast_and_compare():
     b = RecordBatch(["d"], [c])
     m = Cmp(">", Col("d"), Literal(0.0)).eval(b)
     assert m.data.tolist() == [True, False, True]
+
+
+@pytest.mark.gpu
+def test_sum_decimal128_gpu():
+    from auron_amd import native
+
+    native.require()
+    raw = _vals(seed=21)
+    keys = [i % 7 for i in range(len(raw))]
+    s = AuronSession(device="cuda:0")
+    cols = [Column(dtypes.int64, torch.tensor(keys)).to("cuda:0"),
+            Column(D15, torch.tensor(raw, dtype=torch.int64)).to("cuda:0")]
+    scan = P.MemoryScan([RecordBatch(["k", "v"], cols)])
+    plan = P.HashAgg(scan, [Aliased(col("k"), "k")],
+                     [AggFunc("sum", col("v"), name="s")], mode="complete")
+    out = s.collect(plan).to("cpu")
+    want = _want(raw, keys)
+    assert any(abs(v) > 2 ** 63 for v in want.values())
+    _check(out, want)
+
+
+@pytest.mark.gpu
+def test_timestamp_keys_and_bounded_frames_gpu():
+    from auron_amd import native
+    from auron_amd.exprs import Col, WindowFunc, Aliased as Al
+
+    native.require()
+    s = AuronSession(device="cuda:0")
+    ts = Column.from_pylist([0, 0, 86_400_000_000, 86_400_000_000],
+                            dtypes.timestamp).to("cuda:0")
+    v = Column.from_pylist([1, 2, 3, 4], dtypes.int64).to("cuda:0")
+    scan = P.MemoryScan([RecordBatch(["t", "v"], [ts, v])])
+    agg = P.HashAgg(scan, [Aliased(col("t"), "t")],
+                    [AggFunc("sum", col("v"), name="sv")], mode="complete")
+    out = s.collect(agg).to("cpu").to_pydict()
+    assert sorted(out["sv"]) == [3, 7]
+    win = P.Window(scan, [], [(Col("v"), True)],
+                   [Al(WindowFunc("sum", Col("v")), "s")],
+                   frame="rows", frame_lo=-1, frame_hi=1)
+    wout = s.collect(win).to("cpu").to_pydict()
+    assert wout["s"] == [3, 6, 9, 7]
