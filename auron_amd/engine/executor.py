@@ -145,7 +145,9 @@ class Executor:
         # per-top-level-execute scan cache: queries like q9/q28/q88 scan the
         # same (paths, columns) many times for independent subaggregates;
         # decode once per query (cleared at each top-level execute)
-        self._scan_cache: Dict[tuple, List[RecordBatch]] = {}
+        import collections as _collections
+
+        self._scan_cache: "_collections.OrderedDict" = _collections.OrderedDict()
         # compiled fused-expression programs per plan node (expr_fusion):
         # keyed by id(node) + input schema signature, cleared per query so
         # recycled ids can never resolve to a stale program
@@ -206,8 +208,6 @@ class Executor:
             raise TaskCancelled("task cancelled")
         top = len(self._child_time) == 1
         if top:  # top-level call = one query
-            self._scan_cache.clear()
-            self._scan_cache_bytes = 0
             self._fused_progs.clear()
         node = self._rewrite(node)
         name = type(node).__name__
@@ -293,7 +293,7 @@ class Executor:
     def _iter_ParquetScan(self, node: P.ParquetScan):
         key = (tuple(node.paths), repr(node.filters))
         want = list(node.columns) if node.columns else None
-        ent = self._scan_cache.get(key)
+        ent = self._scan_cache_get(key, node.paths)
         if (ent is not None and want is not None
                 and all(c in ent["cols"] for c in want)):
             for i in range(ent["n"]):
@@ -343,7 +343,7 @@ class Executor:
         if not got:
             yield from self._exec_ParquetScan(node)
         elif retain is not None:
-            self._scan_cache_put(key, retain)
+            self._scan_cache_put(key, node.paths, retain)
 
     def _iter_HashJoin(self, node: P.HashJoin):
         """Stream the probe side of a build-right join: the build relation
@@ -462,34 +462,71 @@ class Executor:
             return int(total * 0.15)
         return 4 << 30
 
-    def _scan_cache_put(self, key, batches: List[RecordBatch]):
-        """Merge decoded batches into the PER-COLUMN scan cache (within
-        budget). Keying by (paths, filters) and storing columns
-        individually means a query that re-scans the same table with
-        DIFFERENT column subsets (q9's 15 subaggregates, q14/q23's
-        channel arms) decodes each column once — file-per-batch
-        partitioning is deterministic, so columns from separate reads
-        align row-for-row."""
+    @staticmethod
+    def _scan_mtime_sig(paths):
+        try:
+            return (len(paths), os.path.getmtime(paths[0]),
+                    os.path.getmtime(paths[-1]))
+        except OSError:
+            return None
+
+    def _scan_cache_get(self, key, paths):
+        ent = self._scan_cache.get(key)
+        if ent is None:
+            return None
+        if ent["sig"] != self._scan_mtime_sig(paths):
+            self._scan_cache_bytes -= ent["bytes"]
+            del self._scan_cache[key]
+            return None
+        self._scan_cache.move_to_end(key)
+        return ent
+
+    def _scan_cache_put(self, key, paths, batches: List[RecordBatch]):
+        """Merge decoded batches into the PER-COLUMN scan cache (LRU,
+        byte-budgeted, persistent ACROSS queries — q9's 15 scalar
+        subqueries are separate top-level executes over the same table).
+        Keying by (paths, filters) and storing columns individually means
+        re-scans with DIFFERENT column subsets decode each column once;
+        file-per-batch partitioning is deterministic, so columns from
+        separate reads align row-for-row. Entries staleness-checked by
+        file mtimes; eviction keeps both the byte budget and a free-HBM
+        reserve."""
         from ..memory import _col_bytes
 
-        ent = self._scan_cache.setdefault(key, {"n": len(batches), "cols": {}})
-        if ent["n"] != len(batches):
-            return
+        sig = self._scan_mtime_sig(paths)
+        ent = self._scan_cache.get(key)
+        if ent is None or ent["sig"] != sig or ent["n"] != len(batches):
+            if ent is not None:
+                self._scan_cache_bytes -= ent["bytes"]
+            ent = {"n": len(batches), "cols": {}, "sig": sig, "bytes": 0}
+            self._scan_cache[key] = ent
         budget = self._scan_cache_budget()
+        free_ok = True
+        if self.ctx.device.type == "cuda":
+            free, total = torch.cuda.mem_get_info()
+            free_ok = free > total * 0.2
         for name in batches[0].names:
             if name in ent["cols"]:
                 continue
             col_list = [b.column(name) for b in batches]
             nb = sum(_col_bytes(c) for c in col_list)
-            if self._scan_cache_bytes + nb > budget:
+            while (self._scan_cache_bytes + nb > budget
+                   and len(self._scan_cache) > 1):
+                old_key, old = next(iter(self._scan_cache.items()))
+                if old_key == key:
+                    break
+                self._scan_cache_bytes -= old["bytes"]
+                del self._scan_cache[old_key]
+            if self._scan_cache_bytes + nb > budget or not free_ok:
                 continue
             ent["cols"][name] = col_list
+            ent["bytes"] += nb
             self._scan_cache_bytes += nb
 
     def _exec_ParquetScan(self, node: P.ParquetScan) -> List[RecordBatch]:
         key = (tuple(node.paths), repr(node.filters))
         want = list(node.columns) if node.columns else None
-        ent = self._scan_cache.get(key)
+        ent = self._scan_cache_get(key, node.paths)
         if (ent is not None and want is not None
                 and all(c in ent["cols"] for c in want)):
             return [RecordBatch(want, [ent["cols"][c][i] for c in want])
@@ -501,7 +538,7 @@ class Executor:
                                 filters=node.filters)
             out = self._exec_parquet_scan_uncached(sub)
             if len(out) == ent["n"]:
-                self._scan_cache_put(key, out)
+                self._scan_cache_put(key, node.paths, out)
                 got = {c: [b.column(c) for b in out] for c in missing}
                 return [RecordBatch(want,
                                     [(ent["cols"][c][i] if c in ent["cols"]
@@ -511,7 +548,7 @@ class Executor:
             # file set between reads): do a plain full decode
             return self._exec_parquet_scan_uncached(node)
         out = self._exec_parquet_scan_uncached(node)
-        self._scan_cache_put(key, out)
+        self._scan_cache_put(key, node.paths, out)
         return out
 
     def _exec_parquet_scan_uncached(self, node: P.ParquetScan) -> List[RecordBatch]:

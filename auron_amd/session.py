@@ -80,6 +80,8 @@ class AuronSession:
             parquet_native.dbuf_cache_clear()
             self.ctx.broadcast_cache.clear()
             self.ctx.broadcast_cache_bytes = 0
+            self.executor._scan_cache.clear()
+            self.executor._scan_cache_bytes = 0
             torch.cuda.empty_cache()
             return self.executor.collect(plan)
 
