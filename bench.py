@@ -153,6 +153,9 @@ def main():
                 "queries": qnames,
                 "sf": args.sf,
                 "parallelism": f"dp{world}",
+                "caches": "HBM staged-bytes + decoded-column buffer pool "
+                          "persist across queries/steps (buffer-pool "
+                          "semantics; warmup steps warm them)",
                 "per_query_s": {k: round(v, 4) for k, v in (per_q_last or {}).items()},
             },
         }
