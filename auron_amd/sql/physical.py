@@ -60,6 +60,11 @@ class Lowering:
         return walk_exprs(e, fn)
 
     def _exec_scalar(self, lq: LQuery) -> Literal:
+        # scalar-subquery plans are built after the main prune pass ran:
+        # prune here or the first subquery scan decodes EVERY column of
+        # its fact table (q9 at SF=100: 23 store_sales columns, ~20 s of
+        # host-side launch storm)
+        prune_lquery(lq)
         plan = self.lower_lquery(lq)
         batch = self.session.collect_all(plan)
         if batch.num_rows == 0:
