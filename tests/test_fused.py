@@ -156,3 +156,30 @@ def test_fused_empty_and_all_null_gpu():
     prog = fused.compile_exprs([Col("x") * Literal(2)], {"x": dtypes.int64})
     out = fused.run(prog, rb)[0]
     assert not bool(out.validity.any().cpu())
+
+
+@pytest.mark.gpu
+def test_executor_fused_filter_project_matches(monkeypatch):
+    """Integration: the executor's fused Filter/Project path (enabled via
+    env) produces identical results to the interpreted path."""
+    from auron_amd import AuronSession, col, native
+    from auron_amd.exprs import Aliased, Cmp, Literal
+    from auron_amd.plan import nodes as P
+
+    native.require()
+    batch = _make_batch("cuda", n=50000)
+    scan = P.MemoryScan([batch])
+    plan = P.Project(
+        P.Filter(scan, Cmp(">", Col("a"), Literal(0))),
+        [Aliased((Col("a") + Col("b")) * Literal(2), "y"),
+         Aliased(Col("p"), "p")])
+
+    def run():
+        s = AuronSession(device="cuda:0")
+        return s.collect(plan).to("cpu").to_pydict()
+
+    monkeypatch.setenv("AURON_EXPR_FUSION", "1")
+    got = run()
+    monkeypatch.setenv("AURON_EXPR_FUSION", "0")
+    want = run()
+    assert got == want
