@@ -676,6 +676,40 @@ class Like(Expr):
 
 
 @dataclass(eq=False)
+class StringStartsWith(Expr):
+    """ext-exprs StringStartsWith (vectorized prefix compare)."""
+    child: Expr
+    prefix: str
+
+    def eval(self, batch: RecordBatch) -> Column:
+        c = self.child.eval(batch)
+        return Column(dtypes.bool_, strings.startswith(c, self.prefix),
+                      c.validity)
+
+
+@dataclass(eq=False)
+class StringEndsWith(Expr):
+    child: Expr
+    suffix: str
+
+    def eval(self, batch: RecordBatch) -> Column:
+        c = self.child.eval(batch)
+        return Column(dtypes.bool_, strings.endswith(c, self.suffix),
+                      c.validity)
+
+
+@dataclass(eq=False)
+class StringContains(Expr):
+    child: Expr
+    needle: str
+
+    def eval(self, batch: RecordBatch) -> Column:
+        c = self.child.eval(batch)
+        return Column(dtypes.bool_, strings.contains(c, self.needle),
+                      c.validity)
+
+
+@dataclass(eq=False)
 class Substr(Expr):
     child: Expr
     start: int  # 1-based (SQL)

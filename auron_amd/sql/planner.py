@@ -1680,6 +1680,18 @@ class Planner:
             return F.Rand(_int_lit(conv(a[0])) if a else 42)
         if name == "randn":
             return F.Randn(_int_lit(conv(a[0])) if a else 42)
+        if name == "startswith":
+            from ..exprs import StringStartsWith
+
+            return StringStartsWith(conv(a[0]), _str_lit(conv(a[1])))
+        if name == "endswith":
+            from ..exprs import StringEndsWith
+
+            return StringEndsWith(conv(a[0]), _str_lit(conv(a[1])))
+        if name == "contains":
+            from ..exprs import StringContains
+
+            return StringContains(conv(a[0]), _str_lit(conv(a[1])))
         if name == "unscaled_value":
             return F.UnscaledValue(conv(a[0]))
         if name == "make_decimal":

@@ -157,3 +157,11 @@ def test_bounded_rows_frames(env):
     assert out["s"] == [3, 6, 5]
     assert out["mn"] == [1, 1, 1]
     assert out["cf"] == [3, 2, 1]
+
+
+def test_string_predicate_routes(env):
+    out = run(env, "select startswith(s, 'a') sw, endswith(s, 'b') ew, "
+                   "contains(s, ',') ct from tt")
+    assert out["sw"] == [True, False, None]
+    assert out["ew"] == [True, True, None]
+    assert out["ct"] == [True, False, None]
