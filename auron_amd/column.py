@@ -416,8 +416,60 @@ class RecordBatch:
 
     def gather(self, idx: torch.Tensor,
                may_have_negative: bool = False) -> "RecordBatch":
+        if self.device.type == "cuda" and len(self.columns) > 1:
+            fused = self._gather_fused(idx, may_have_negative)
+            if fused is not None:
+                return fused
         return RecordBatch(self.names,
                            [c.gather(idx, may_have_negative) for c in self.columns])
+
+    def _gather_fused(self, idx: torch.Tensor, may_have_negative: bool):
+        """All fixed-width columns in ONE kernel launch (k_multi_gather);
+        string/list columns keep the per-column path (k_bytes_gather).
+        Join/window/sort row gathers were one at::native index_select per
+        column — the top at::native entry of the final SF=10 profile."""
+        from . import native
+
+        if not native.available():
+            return None
+        import numpy as np
+
+        lib = native.lib()
+        idx = idx.to(self.device).to(torch.int64).contiguous()
+        n = int(idx.numel())
+        out_cols: list = [None] * len(self.columns)
+        descs = []
+        keep = [idx]
+        for ci, c in enumerate(self.columns):
+            if c.dtype.uses_offsets:
+                out_cols[ci] = c.gather(idx, may_have_negative)
+                continue
+            data = c.data if c.data.is_contiguous() else c.data.contiguous()
+            keep.append(data)
+            esize = data.element_size() * (2 if c.dtype.code == dtypes.DECIMAL128 else 1)
+            shape = (n, 2) if c.dtype.code == dtypes.DECIMAL128 else (n,)
+            dst = torch.empty(shape, dtype=data.dtype, device=self.device)
+            want_valid = c.validity is not None or may_have_negative
+            dv = torch.empty(n, dtype=torch.bool, device=self.device) \
+                if want_valid else None
+            sv = 0
+            if c.validity is not None:
+                v = c.validity if c.validity.is_contiguous() else c.validity.contiguous()
+                keep.append(v)
+                sv = v.data_ptr()
+            descs.append((data.data_ptr(), sv, dst.data_ptr(),
+                          dv.data_ptr() if dv is not None else 0, esize, 0))
+            out_cols[ci] = Column(c.dtype, dst, dv)
+        if descs and n:
+            arr = np.array(descs, dtype=np.uint64).reshape(-1, 6)
+            arr = np.ascontiguousarray(arr)
+            for lo in range(0, len(descs), 24):
+                sub = np.ascontiguousarray(arr[lo:lo + 24])
+                rc = lib.au_multi_gather(idx.data_ptr(), n, sub.ctypes.data,
+                                         sub.shape[0],
+                                         native.stream_ptr(self.device))
+                native.check(rc, "au_multi_gather")
+        return RecordBatch(self.names, out_cols)
 
     def filter(self, mask: torch.Tensor) -> "RecordBatch":
         idx = torch.nonzero(mask, as_tuple=False).flatten()

@@ -54,6 +54,8 @@ def _try_load():
     lib.au_join_fill.argtypes = [c, c, ctypes.c_int, i64, c, i64, c, c, c, c, c, c, ctypes.c_int, c]
     lib.au_pmod.argtypes = [c, i64, i32, c, c]
     lib.au_bytes_gather.argtypes = [c, c, c, c, i64, c]
+    lib.au_multi_gather.argtypes = [c, i64, c, ctypes.c_int, c]
+    lib.au_multi_gather.restype = ctypes.c_int
     lib.au_agg_scatter.argtypes = [c, i64, c, c, ctypes.c_int, ctypes.c_int,
                                    ctypes.c_int, c, c, i64, c]
     lib.au_agg_count.argtypes = [c, i64, c, c, c]

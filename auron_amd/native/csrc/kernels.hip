@@ -489,3 +489,65 @@ AU_EXPORT int au_bytes_gather(const uint8_t* src, const int64_t* src_start,
 
 // ------------------------------------------------------------------ version
 AU_EXPORT int au_abi_version() { return 1; }
+
+// ------------------------------------------------------- fused gather
+// One launch gathers ALL fixed-width columns of a batch (join/window/
+// sort/partition row gathers were one at::native index_select launch per
+// column — the top at::native entry of the final SF=10 profile).
+// Negative row index => null output (outer-join semantics).
+#define MG_MAX_COLS 24
+
+struct MgCol {
+  const void* src;
+  const uint8_t* src_valid;  // nullptr = all valid
+  void* dst;
+  uint8_t* dst_valid;        // nullptr = caller wants no validity
+  int64_t esize;             // 1/2/4/8/16
+  int64_t pad;
+};
+
+struct MgParams { MgCol cols[MG_MAX_COLS]; };
+
+__global__ void __launch_bounds__(256) k_multi_gather(
+    const int64_t* __restrict__ idx, int64_t n, MgParams p, int ncols) {
+  // blocks tile (col, row-chunk): consecutive blocks cover the same col
+  int64_t chunks_per_col = (n + 255) / 256;
+  for (int64_t b = blockIdx.x; b < chunks_per_col * ncols; b += gridDim.x) {
+    int c = (int)(b / chunks_per_col);
+    int64_t row = (b % chunks_per_col) * 256 + threadIdx.x;
+    if (row >= n) continue;
+    const MgCol col = p.cols[c];
+    int64_t i = idx[row];
+    bool ok = i >= 0;
+    int64_t src_i = ok ? i : 0;
+    bool v = ok && (!col.src_valid || col.src_valid[src_i]);
+    if (col.dst_valid) col.dst_valid[row] = v;
+    switch (col.esize) {
+      case 1: ((uint8_t*)col.dst)[row] = ok ? ((const uint8_t*)col.src)[src_i] : 0; break;
+      case 2: ((uint16_t*)col.dst)[row] = ok ? ((const uint16_t*)col.src)[src_i] : 0; break;
+      case 4: ((uint32_t*)col.dst)[row] = ok ? ((const uint32_t*)col.src)[src_i] : 0; break;
+      case 8: ((uint64_t*)col.dst)[row] = ok ? ((const uint64_t*)col.src)[src_i] : 0; break;
+      case 16: {
+        ulonglong2 z; z.x = 0; z.y = 0;
+        ((ulonglong2*)col.dst)[row] = ok ? ((const ulonglong2*)col.src)[src_i] : z;
+        break;
+      }
+      default: break;
+    }
+  }
+}
+
+AU_EXPORT int au_multi_gather(const void* idx, int64_t n,
+                              const void* host_cols, int ncols,
+                              void* stream) {
+  if (n <= 0 || ncols <= 0) return 0;
+  if (ncols > MG_MAX_COLS) return 1002;
+  MgParams p{};
+  const MgCol* hc = (const MgCol*)host_cols;
+  for (int i = 0; i < ncols; i++) p.cols[i] = hc[i];
+  int64_t chunks = ((n + 255) / 256) * ncols;
+  int64_t g = chunks > 4096 ? 4096 : chunks;
+  hipLaunchKernelGGL(k_multi_gather, dim3((uint32_t)g), dim3(256), 0,
+                     (hipStream_t)stream, (const int64_t*)idx, n, p, ncols);
+  return (int)hipGetLastError();
+}
