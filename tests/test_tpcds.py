@@ -88,7 +88,11 @@ def assert_result_matches(batch, df, qname=None):
             if v is None:
                 return "~"
             if isinstance(v, float):
-                return "~" if v != v else f"{v:.3e}"
+                if v != v:
+                    return "~"
+                if abs(v) < 1e12 and v == int(v):
+                    return int(v)  # pandas floats int-typed engine cols
+                return f"{v:.3e}"
             return v
 
         strip = lambda rows: collections.Counter(
