@@ -500,9 +500,25 @@ class RecordBatch:
 
     @staticmethod
     def from_arrow(table, device="cpu") -> "RecordBatch":
-        names = table.schema.names
-        cols = [Column.from_arrow(table.column(i), device) for i in range(len(names))]
-        return RecordBatch(list(names), cols)
+        import pyarrow as pa
+
+        names = []
+        cols = []
+        for i, name in enumerate(table.schema.names):
+            arr = table.column(i)
+            if pa.types.is_struct(arr.type):
+                # struct columns flatten into dotted leaf columns at the
+                # boundary (GetStructField then resolves to a plain column
+                # reference — ext-exprs GetIndexedField-on-struct parity)
+                carr = arr.combine_chunks() if isinstance(arr, pa.ChunkedArray) else arr
+                for fi in range(carr.type.num_fields):
+                    f = carr.type.field(fi)
+                    names.append(f"{name}.{f.name}")
+                    cols.append(Column.from_arrow(carr.field(fi), device))
+                continue
+            names.append(name)
+            cols.append(Column.from_arrow(arr, device))
+        return RecordBatch(names, cols)
 
     def to_arrow(self):
         import pyarrow as pa

@@ -63,3 +63,21 @@ def test_batch_filter():
     )
     out = b.filter(torch.tensor([True, False, True, False]))
     assert out.to_pydict() == {"x": [1, 3], "s": ["a", "c"]}
+
+
+def test_struct_columns_flatten_at_boundary():
+    """Struct columns flatten into dotted leaf columns at the arrow
+    boundary (GetStructField-on-struct reads become plain column refs)."""
+    import pyarrow as pa
+
+    from auron_amd.column import RecordBatch
+
+    t = pa.table({
+        "id": [1, 2],
+        "info": pa.array([{"a": 5, "b": "x"}, {"a": None, "b": "y"}],
+                         pa.struct([("a", pa.int64()), ("b", pa.string())])),
+    })
+    rb = RecordBatch.from_arrow(t)
+    assert rb.names == ["id", "info.a", "info.b"]
+    assert rb.column("info.a").to_pylist() == [5, None]
+    assert rb.column("info.b").to_pylist() == ["x", "y"]
