@@ -124,30 +124,78 @@ def unpack_batch(meta: dict, buf: torch.Tensor) -> RecordBatch:
     return RecordBatch(names, cols)
 
 
+def _schema_of(b: RecordBatch):
+    return [(n, (c.dtype.code, c.dtype.precision, c.dtype.scale))
+            for n, c in zip(b.names, b.columns)]
+
+
+def _meta_from_sizes(schema, row) -> Optional[dict]:
+    """Rebuild a pack_batch meta from the SPMD-identical schema plus one
+    rank's numeric size row [nrows, (data,val,off)*ncols]."""
+    if int(row[0]) < 0:
+        return None
+    cols = []
+    for i, (name, dt) in enumerate(schema):
+        cols.append({"name": name, "dtype": dt,
+                     "n": 0,  # unused by unpack
+                     "data": int(row[1 + 3 * i]),
+                     "val": int(row[2 + 3 * i]),
+                     "off": int(row[3 + 3 * i])})
+    return {"nrows": int(row[0]), "cols": cols}
+
+
 def all_to_all(batches_by_dest: List[Optional[RecordBatch]], device,
                group=None) -> List[RecordBatch]:
     """Send batches_by_dest[d] to rank d; return batches received (one per
-    source rank that sent a non-empty batch)."""
+    source rank that sent a non-empty batch).
+
+    Metadata rides ONE small int64 all_gather instead of a pickled
+    object collective: plans are SPMD-identical, so every rank derives
+    the same schema locally and only the per-dest segment byte sizes
+    vary by rank (weak-spot fix: the object gather was a host pickle
+    round-trip on every exchange, a latency tax at N=8)."""
     world = dist.get_world_size(group)
     rank = dist.get_rank(group)
     metas = []
     bufs = []
+    schema = None
     for d in range(world):
         b = batches_by_dest[d]
         if b is None:
             metas.append(None)
             bufs.append(torch.zeros(0, dtype=torch.uint8, device=device))
         else:
+            if schema is None:
+                schema = _schema_of(b)
             m, buf = pack_batch(b, device)
             metas.append(m)
             bufs.append(buf)
-    # ONE object collective carries both per-dest metas and buffer sizes
-    # (object all-gathers are host round-trips; halving them matters at N=8)
-    payload = (metas, [b.numel() for b in bufs])
-    gathered: List[Optional[tuple]] = [None] * world  # type: ignore
-    dist.all_gather_object(gathered, payload, group=group)
-    recv_metas = [gathered[s][0][rank] for s in range(world)]
-    recv_sizes = [int(gathered[s][1][rank]) for s in range(world)]
+    if schema is not None and all(m is not None for m in metas):
+        ncols = len(schema)
+        stride = 1 + 3 * ncols + 1  # nrows, sizes..., total buf bytes
+        comm_dev = device if dist.get_backend(group) == "nccl" else "cpu"
+        sz = torch.full((world, stride), -1, dtype=torch.int64)
+        for d, m in enumerate(metas):
+            sz[d, 0] = m["nrows"]
+            for i, cm in enumerate(m["cols"]):
+                sz[d, 1 + 3 * i] = cm["data"]
+                sz[d, 2 + 3 * i] = cm["val"]
+                sz[d, 3 + 3 * i] = cm["off"]
+            sz[d, -1] = bufs[d].numel()
+        sz = sz.reshape(-1).to(comm_dev)
+        outs = [torch.empty_like(sz) for _ in range(world)]
+        dist.all_gather(outs, sz, group=group)
+        rows = [o.cpu().reshape(world, stride)[rank] for o in outs]
+        recv_metas = [_meta_from_sizes(schema, r) for r in rows]
+        recv_sizes = [int(r[-1]) for r in rows]
+    else:
+        # a dest without a batch means the schema may be unknowable
+        # locally: fall back to the object collective
+        payload = (metas, [b.numel() for b in bufs])
+        gathered: List[Optional[tuple]] = [None] * world  # type: ignore
+        dist.all_gather_object(gathered, payload, group=group)
+        recv_metas = [gathered[s][0][rank] for s in range(world)]
+        recv_sizes = [int(gathered[s][1][rank]) for s in range(world)]
 
     backend = dist.get_backend(group)
     out_batches: List[RecordBatch] = []
