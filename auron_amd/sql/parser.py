@@ -35,7 +35,7 @@ KEYWORDS = {
     "limit", "as", "and", "or", "not", "in", "is", "null", "between", "like",
     "case", "when", "then", "else", "end", "cast", "exists", "union", "all",
     "intersect", "except", "join", "inner", "left", "right", "full", "outer",
-    "cross", "on", "with", "rollup", "grouping", "sets", "over", "partition",
+    "cross", "on", "with", "rollup", "cube", "grouping", "sets", "over", "partition",
     "rows", "range", "unbounded", "preceding", "following", "current", "row",
     "asc", "desc", "nulls", "first", "last", "interval", "substring", "for",
 }
@@ -229,6 +229,16 @@ class Parser:
                 self.expect_op("(")
                 group_by = self._expr_list()
                 self.expect_op(")")
+            elif self.accept_kw("cube"):
+                # CUBE(a,b,...) desugars to GROUPING SETS of all subsets
+                self.expect_op("(")
+                exprs = self._expr_list()
+                self.expect_op(")")
+                import itertools as _it
+
+                gsets = [list(sub)
+                         for r in range(len(exprs), -1, -1)
+                         for sub in _it.combinations(exprs, r)]
             elif self.accept_kw("grouping"):
                 self.expect_kw("sets")
                 self.expect_op("(")

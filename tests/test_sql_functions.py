@@ -165,3 +165,14 @@ def test_string_predicate_routes(env):
     assert out["sw"] == [True, False, None]
     assert out["ew"] == [True, True, None]
     assert out["ct"] == [True, False, None]
+
+
+def test_group_by_cube(env):
+    out = run(env, "select s, x, sum(x) sv, grouping(s) gs, grouping(x) gx "
+                   "from tt group by cube(s, x) order by gs, gx, s, x")
+    rows = list(zip(out["s"], out["x"], out["sv"], out["gs"], out["gx"]))
+    # 3 detail rows + per-s + per-x + grand total
+    assert (None, None, 6, 1, 1) in rows
+    assert ("b", None, 2, 0, 1) in rows
+    assert (None, 2, 2, 1, 0) in rows
+    assert len(rows) == 3 + 3 + 3 + 1
