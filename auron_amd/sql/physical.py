@@ -267,8 +267,25 @@ class Lowering:
             if a.expr is not None:
                 carry |= expr_cols(a.expr)
         carry_l = sorted(carry)
+        # a column that is BOTH a grouping key and an aggregate input must
+        # be carried under a fresh name, or the nulled key column shadows
+        # the aggregate's values in the rolled-up sets (CUBE(x) + sum(x))
+        key_names = {n for _, n in keys}
+        ren = {c: (f"__cv_{c}" if c in key_names else c) for c in carry_l}
+        if any(ren[c] != c for c in carry_l):
+            from .planner import walk_exprs
+
+            def _sub(e):
+                if isinstance(e, Col) and e.name in ren:
+                    return col(ren[e.name])
+                return e
+
+            for i, a in enumerate(aggs):
+                if a.expr is not None:
+                    aggs[i] = type(a)(a.fn, walk_exprs(a.expr, _sub),
+                                      a.distinct, a.name)
         pre_items = [Aliased(e, n) for (e, n) in keys] + \
-            [Aliased(col(c), c) for c in carry_l]
+            [Aliased(col(c), ren[c]) for c in carry_l]
         pre = P.Project(child, pre_items)
         nk = len(keys)
         projections = []
@@ -284,7 +301,7 @@ class Lowering:
                 gn = node.grp_names.get(n)
                 if gn:
                     proj.append(Aliased(lit(0 if mask[i] else 1), gn))
-            proj += [Aliased(col(c), c) for c in carry_l]
+            proj += [Aliased(col(ren[c]), ren[c]) for c in carry_l]
             projections.append(proj)
         expand = P.Expand(pre, projections)
         new_keys = [(col(n), n) for _, n in keys]
