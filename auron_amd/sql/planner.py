@@ -1261,17 +1261,30 @@ class Planner:
         by_alias = {n: e for e, n in outputs}
         out_keys = {expr_key(e): e for e, _ in outputs}
         scope = Scope([rel], outer)
+        def emit(e, oi):
+            # explicit NULLS FIRST/LAST that differs from the Spark
+            # default (asc->nulls first, desc->nulls last, which the
+            # executor implements) desugars into a leading IsNull key —
+            # no plan-node change needed
+            nf = getattr(oi, "nulls_first", None)
+            default_nf = oi.ascending
+            if nf is not None and nf != default_nf:
+                from ..exprs import IsNull
+
+                out.append((IsNull(e), not nf))
+            out.append((e, oi.ascending))
+
         for oi in order_items:
             e_ast = oi.expr
             if isinstance(e_ast, A.Num) and e_ast.is_int:
                 idx = int(e_ast.text) - 1
                 if not (0 <= idx < len(outputs)):
                     raise SqlError(f"ORDER BY ordinal {idx+1} out of range")
-                out.append((outputs[idx][0], oi.ascending))
+                emit(outputs[idx][0], oi)
                 continue
             if isinstance(e_ast, A.Ident) and e_ast.qualifier is None \
                     and e_ast.name.lower() in by_alias:
-                out.append((by_alias[e_ast.name.lower()], oi.ascending))
+                emit(by_alias[e_ast.name.lower()], oi)
                 continue
             try:
                 e = self.to_expr(e_ast, scope, ctx)
@@ -1281,7 +1294,7 @@ class Planner:
             k = expr_key(e)
             if k in out_keys:
                 e = out_keys[k]
-            out.append((e, oi.ascending))
+            emit(e, oi)
         return out
 
     # ----------------------------------------------------- expr conversion

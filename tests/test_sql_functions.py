@@ -176,3 +176,10 @@ def test_group_by_cube(env):
     assert ("b", None, 2, 0, 1) in rows
     assert (None, 2, 2, 1, 0) in rows
     assert len(rows) == 3 + 3 + 3 + 1
+
+
+def test_order_by_explicit_null_placement(env):
+    out = run(env, "select t from tt order by t nulls last")
+    assert out["t"] == [1000000, 2000000, None]
+    out = run(env, "select t from tt order by t desc nulls first")
+    assert out["t"] == [None, 2000000, 1000000]
