@@ -870,6 +870,22 @@ class Executor:
                 out.append(RecordBatch([a.name for a in proj], cols))
         return out
 
+    def _exec_Replicate(self, node: P.Replicate) -> List[RecordBatch]:
+        from ..exprs import eval_scope
+
+        out = []
+        for b in self.execute(node.child):
+            with eval_scope(b):
+                c = node.count.eval(b)
+            cnt = c.data.to(torch.int64).clamp(min=0)
+            if c.validity is not None:
+                cnt = torch.where(c.validity, cnt, torch.zeros_like(cnt))
+            idx = torch.repeat_interleave(
+                torch.arange(b.num_rows, dtype=torch.int64, device=b.device),
+                cnt)
+            out.append(b.gather(idx))
+        return out
+
     def _exec_Limit(self, node: P.Limit) -> List[RecordBatch]:
         remaining = node.n
         skip = node.offset

@@ -170,6 +170,17 @@ class Union(PlanNode):
         return list(self.inputs)
 
 
+@dataclass
+class Replicate(PlanNode):
+    """Emit each input row `count` times (INTERSECT ALL / EXCEPT ALL
+    multiset multiplicities)."""
+    child: PlanNode
+    count: Expr
+
+    def children(self):
+        return [self.child]
+
+
 @dataclass(eq=False)
 class Expand(PlanNode):
     """GROUPING SETS fan-out (expand_exec.rs)."""

@@ -194,7 +194,7 @@ class Parser:
             if op == "union":
                 op = "union_all" if self.accept_kw("all") else "union"
             elif self.accept_kw("all"):
-                self.fail(f"{op.upper()} ALL unsupported")
+                op = f"{op}_all"  # multiset INTERSECT ALL / EXCEPT ALL
             right = self._set_operand()
             left = A.SetOp(op, left, right)
         return left

@@ -22,9 +22,9 @@ from ..exprs import Aliased, Col, Expr, Literal, col, lit
 from ..plan import nodes as P
 from .parser import SqlError, parse_sql
 from .planner import (BROADCAST_ROWS, LAgg, LCTERef, LFilter, LJoin, LLimit,
-                      LNode, LProject, LQuery, LScan, LSort, LUnionAll,
-                      LWindow, MatCTE, Planner, ScalarSubqueryLit, expr_cols,
-                      expr_key, infer_dtype, walk_exprs)
+                      LNode, LProject, LQuery, LReplicate, LScan, LSort,
+                      LUnionAll, LWindow, MatCTE, Planner, ScalarSubqueryLit,
+                      expr_cols, expr_key, infer_dtype, walk_exprs)
 
 ANY = ("any",)
 SINGLE = ("single",)
@@ -147,6 +147,11 @@ class Lowering:
             if not kept:
                 part = ANY
         return P.Project(child, items), part
+
+    def _lower_LReplicate(self, node: LReplicate):
+        # replication is row-local: partitioning survives
+        child, part = self.lower(node.child)
+        return P.Replicate(child, self.subst_scalars(node.count)), part
 
     def _lower_LLimit(self, node: LLimit):
         child, part = self.lower(node.child)
@@ -339,6 +344,9 @@ def prune(node: LNode, needed: Set[str]):
         return
     if isinstance(node, LFilter):
         prune(node.child, needed | expr_cols(node.pred))
+        return
+    if isinstance(node, LReplicate):
+        prune(node.child, needed | expr_cols(node.count))
         return
     if isinstance(node, LProject):
         node.items = [(e, n) for e, n in node.items if n in needed] or node.items[:1]

@@ -109,6 +109,12 @@ class LWindow(LNode):
 
 
 @dataclass
+class LReplicate(LNode):
+    child: LNode
+    count: Expr  # per-row output multiplicity
+
+
+@dataclass
 class LUnionAll(LNode):
     children: List[LNode]
     # per-child projection onto the common output names
@@ -473,6 +479,28 @@ class Planner:
             [(e, n2) for (e, _), n2 in zip(routs, out_names)] + [(lit(1), side_name)],
         ]
         u = LUnionAll([lrel.node, rrel.node], projections)
+        cols = [RelCol(None, sn, en, self.alloc.types.get(en))
+                for sn, en in zip(sql_names, out_names)]
+        if s.op in ("intersect_all", "except_all"):
+            # multiset semantics: per-side counts, then replicate each
+            # distinct row min(c0,c1) / max(c0-c1,0) times
+            from .. import functions as F
+
+            c0 = self.alloc.fresh("__c0", dtypes.int64)
+            c1 = self.alloc.fresh("__c1", dtypes.int64)
+            agg = LAgg(u, [(col(n), n) for n in out_names],
+                       [AggFunc("sum", lit(1) - col(side_name), name=c0),
+                        AggFunc("sum", col(side_name), name=c1)])
+            if s.op == "intersect_all":
+                m = CaseWhen([(col(c0) < col(c1), col(c0))], col(c1))
+            else:
+                m = col(c0) - col(c1)
+            self.alloc.set_type(c0, dtypes.int64)
+            self.alloc.set_type(c1, dtypes.int64)
+            rep = LReplicate(LFilter(agg, m > lit(0)), m)
+            rel = Rel(rep, cols, max(lrel.est, rrel.est) / 4,
+                      lrel.base_dim_only and rrel.base_dim_only)
+            return rel, [(col(en), sn) for sn, en in zip(sql_names, out_names)]
         nmin = self.alloc.fresh("__smin", dtypes.int64)
         nmax = self.alloc.fresh("__smax", dtypes.int64)
         agg = LAgg(u, [(col(n), n) for n in out_names],
@@ -483,8 +511,6 @@ class Planner:
         else:  # except
             pred = (col(nmin) == 0) & (col(nmax) == 0)
         filt = LFilter(agg, pred)
-        cols = [RelCol(None, sn, en, self.alloc.types.get(en))
-                for sn, en in zip(sql_names, out_names)]
         rel = Rel(filt, cols, max(lrel.est, rrel.est) / 4,
                   lrel.base_dim_only and rrel.base_dim_only)
         return rel, [(col(en), sn) for sn, en in zip(sql_names, out_names)]

@@ -183,3 +183,11 @@ def test_order_by_explicit_null_placement(env):
     assert out["t"] == [1000000, 2000000, None]
     out = run(env, "select t from tt order by t desc nulls first")
     assert out["t"] == [None, 2000000, 1000000]
+
+
+def test_intersect_except_all(env):
+    # multiset semantics over the t column (1e6, 2e6, null)
+    out = run(env, "select x from tt intersect all select x from tt")
+    assert sorted(out["x"]) == [1, 2, 3]
+    out = run(env, "select x from tt except all select x from tt where x > 1")
+    assert sorted(out["x"]) == [1]
