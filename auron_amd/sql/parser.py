@@ -1,10 +1,11 @@
 """SQL lexer + recursive-descent parser for the Spark-SQL dialect used by
 the TPC-DS suite the reference ships (dev/auron-it tpcds-queries/q*.sql):
 SELECT/DISTINCT, comma + ANSI joins (left/right/full outer), WHERE,
-GROUP BY [ROLLUP]/GROUPING SETS, HAVING, window functions with
-PARTITION BY / ORDER BY / ROWS|RANGE frames, WITH CTEs, UNION [ALL] /
-INTERSECT / EXCEPT, scalar/IN/EXISTS subqueries, CASE, CAST, BETWEEN,
-LIKE, IN lists, INTERVAL arithmetic, ordinal ORDER BY references.
+GROUP BY [ROLLUP|CUBE]/GROUPING SETS, HAVING, window functions with
+PARTITION BY / ORDER BY / ROWS|RANGE frames (explicit ROWS bounds), WITH
+CTEs, UNION [ALL] / INTERSECT [ALL] / EXCEPT [ALL], scalar/IN/EXISTS
+subqueries, CASE, CAST, BETWEEN, LIKE, IN lists, INTERVAL arithmetic,
+ordinal ORDER BY references, NULLS FIRST/LAST, DATE/TIMESTAMP literals.
 """
 from __future__ import annotations
 
