@@ -1532,10 +1532,13 @@ class Planner:
             if isinstance(entry, tuple) and entry[0] == "stddev":
                 from ..exprs import Sqrt
 
-                _, s1, s2, cn = entry
+                _, s1, s2, cn, sqrt, pop = entry
                 n_f = Cast(col(cn), dtypes.float64)
-                var = (col(s2) - col(s1) * col(s1) / n_f) / (n_f - lit(1.0))
-                return CaseWhen([(col(cn) > 1, Sqrt(var))],
+                denom = n_f if pop else (n_f - lit(1.0))
+                var = (col(s2) - col(s1) * col(s1) / n_f) / denom
+                out = Sqrt(var) if sqrt else var
+                min_n = 0 if pop else 1
+                return CaseWhen([(col(cn) > min_n, out)],
                                 Literal(None, dtypes.float64))
             return col(entry)
         if name == "grouping":
@@ -1845,7 +1848,8 @@ def _conjuncts(node: A.ANode) -> List[A.ANode]:
     return [node]
 
 
-AGG_NAMES = {"sum", "avg", "min", "max", "count", "stddev_samp"}
+AGG_NAMES = {"sum", "avg", "min", "max", "count", "stddev_samp",
+             "stddev", "stddev_pop", "var_samp", "variance", "var_pop"}
 
 
 def _has_agg(sel: A.Select) -> bool:
@@ -2136,9 +2140,11 @@ def _register_agg(fc: A.FuncCall, planner: Planner, scope: Scope, ctx: AggCtx):
         return
     name = fc.name
     alloc = planner.alloc
-    if name == "stddev_samp":
-        # decomposed: sqrt((sum_sq - sum^2/n) / (n-1)); registered as three
-        # plain aggs; the replacement in to_expr synthesizes the expression
+    if name in ("stddev_samp", "stddev", "stddev_pop", "var_samp",
+                "variance", "var_pop"):
+        # decomposed moments: var = (sum_sq - sum^2/n) / (n-1 | n);
+        # registered as three plain aggs, the replacement in to_expr
+        # synthesizes sqrt()/plain per the variant
         arg = planner.to_expr(fc.args[0], scope, None)
         s1 = alloc.fresh("__sd_s1", dtypes.float64)
         s2 = alloc.fresh("__sd_s2", dtypes.float64)
@@ -2147,7 +2153,9 @@ def _register_agg(fc: A.FuncCall, planner: Planner, scope: Scope, ctx: AggCtx):
         ctx.aggs.append(AggFunc("sum", f, name=s1))
         ctx.aggs.append(AggFunc("sum", f * f, name=s2))
         ctx.aggs.append(AggFunc("count", arg, name=cn))
-        ctx.agg_map[key] = ("stddev", s1, s2, cn)
+        sqrt = name in ("stddev_samp", "stddev", "stddev_pop")
+        pop = name in ("stddev_pop", "var_pop")
+        ctx.agg_map[key] = ("stddev", s1, s2, cn, sqrt, pop)
         return
     if fc.star or (name == "count" and not fc.args):
         en = alloc.fresh("__cnt", dtypes.int64)

@@ -191,3 +191,17 @@ def test_intersect_except_all(env):
     assert sorted(out["x"]) == [1, 2, 3]
     out = run(env, "select x from tt except all select x from tt where x > 1")
     assert sorted(out["x"]) == [1]
+
+
+def test_variance_family(env):
+    import math
+    import statistics
+
+    out = run(env, "select stddev_samp(x) a, stddev_pop(x) b, var_samp(x) c, "
+                   "var_pop(x) d, variance(x) e from tt")
+    vals = [1.0, 2.0, 3.0]
+    assert math.isclose(out["a"][0], statistics.stdev(vals))
+    assert math.isclose(out["b"][0], statistics.pstdev(vals))
+    assert math.isclose(out["c"][0], statistics.variance(vals))
+    assert math.isclose(out["d"][0], statistics.pvariance(vals))
+    assert math.isclose(out["e"][0], statistics.variance(vals))
