@@ -1529,6 +1529,22 @@ class Planner:
             entry = ctx.agg_map.get(_agg_ast_key(fc))
             if entry is None:
                 raise SqlError(f"aggregate {name}() was not collected")
+            if isinstance(entry, tuple) and entry[0] == "covar":
+                from ..exprs import Sqrt
+
+                _, vname, sx, sy, sxy, sxx, syy, cn = entry
+                n_f = Cast(col(cn), dtypes.float64)
+                cxy = col(sxy) - col(sx) * col(sy) / n_f
+                if vname == "covar_pop":
+                    return CaseWhen([(col(cn) > 0, cxy / n_f)],
+                                    Literal(None, dtypes.float64))
+                if vname == "covar_samp":
+                    return CaseWhen([(col(cn) > 1, cxy / (n_f - lit(1.0)))],
+                                    Literal(None, dtypes.float64))
+                vx = col(sxx) - col(sx) * col(sx) / n_f
+                vy = col(syy) - col(sy) * col(sy) / n_f
+                return CaseWhen([(col(cn) > 0, cxy / Sqrt(vx * vy))],
+                                Literal(None, dtypes.float64))
             if isinstance(entry, tuple) and entry[0] == "stddev":
                 from ..exprs import Sqrt
 
@@ -1849,7 +1865,8 @@ def _conjuncts(node: A.ANode) -> List[A.ANode]:
 
 
 AGG_NAMES = {"sum", "avg", "min", "max", "count", "stddev_samp",
-             "stddev", "stddev_pop", "var_samp", "variance", "var_pop"}
+             "stddev", "stddev_pop", "var_samp", "variance", "var_pop",
+             "corr", "covar_samp", "covar_pop", "approx_count_distinct"}
 
 
 def _has_agg(sel: A.Select) -> bool:
@@ -2140,6 +2157,31 @@ def _register_agg(fc: A.FuncCall, planner: Planner, scope: Scope, ctx: AggCtx):
         return
     name = fc.name
     alloc = planner.alloc
+    if name == "approx_count_distinct":
+        # exact distinct count is a valid (and here cheap) implementation
+        arg = planner.to_expr(fc.args[0], scope, None)
+        en = alloc.fresh("__acntd", dtypes.int64)
+        ctx.aggs.append(AggFunc("count_distinct", arg, distinct=True, name=en))
+        ctx.agg_map[key] = en
+        return
+    if name in ("corr", "covar_samp", "covar_pop"):
+        # two-argument moments: covar = (Σxy - ΣxΣy/n) / (n-1 | n);
+        # corr = covar_pop / (stddev_pop(x) * stddev_pop(y))
+        x = planner.to_expr(fc.args[0], scope, None)
+        y = planner.to_expr(fc.args[1], scope, None)
+        fx = Cast(x, dtypes.float64)
+        fy = Cast(y, dtypes.float64)
+        names = [alloc.fresh(f"__cv_{i}", dtypes.float64) for i in range(5)]
+        cn = alloc.fresh("__cv_n", dtypes.int64)
+        both = CaseWhen([(Not(IsNull(x)) & Not(IsNull(y)), lit(1))], lit(None))
+        for nm, e in zip(names, [fx, fy, fx * fy, fx * fx, fy * fy]):
+            # null in EITHER argument drops the pair (SQL semantics)
+            ctx.aggs.append(AggFunc(
+                "sum", CaseWhen([(Not(IsNull(x)) & Not(IsNull(y)), e)],
+                                Literal(None, dtypes.float64)), name=nm))
+        ctx.aggs.append(AggFunc("count", both, name=cn))
+        ctx.agg_map[key] = ("covar", name, *names, cn)
+        return
     if name in ("stddev_samp", "stddev", "stddev_pop", "var_samp",
                 "variance", "var_pop"):
         # decomposed moments: var = (sum_sq - sum^2/n) / (n-1 | n);

@@ -205,3 +205,16 @@ def test_variance_family(env):
     assert math.isclose(out["c"][0], statistics.variance(vals))
     assert math.isclose(out["d"][0], statistics.pvariance(vals))
     assert math.isclose(out["e"][0], statistics.variance(vals))
+
+
+def test_corr_covar(env):
+    import math
+    import statistics
+
+    out = run(env, "select corr(x, x) c, covar_samp(x, x) cs, "
+                   "covar_pop(x, x) cp, approx_count_distinct(s) ad from tt")
+    vals = [1.0, 2.0, 3.0]
+    assert math.isclose(out["c"][0], 1.0)
+    assert math.isclose(out["cs"][0], statistics.variance(vals))
+    assert math.isclose(out["cp"][0], statistics.pvariance(vals))
+    assert out["ad"][0] == 2  # two distinct non-null strings
