@@ -1149,3 +1149,44 @@ class Randn(Expr):
                       ^ ctx.get("batch_ordinal", 0) ^ 0x5EED)
         data = torch.randn(batch.num_rows, dtype=torch.float64, generator=g)
         return Column(dtypes.float64, data.to(batch.device))
+
+
+@dataclass(eq=False)
+class ListQuantile(Expr):
+    """Exact interpolated quantile of each LIST row (Spark percentile
+    semantics; percentile_approx is implemented exactly). Sorting is one
+    device sort of (row, value) pairs — value-major argsort, then a
+    stable row-major argsort groups each row's values in order."""
+    child: Expr
+    p: float
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        assert c.dtype.is_list, "percentile over a collected list"
+        n = len(c)
+        device = c.device
+        offs = c.offsets
+        lens = offs[1:] - offs[:-1]
+        total = int(offs[-1].item()) if n else 0
+        if total == 0:
+            return Column(dtypes.float64,
+                          torch.zeros(n, dtype=torch.float64, device=device),
+                          torch.zeros(n, dtype=torch.bool, device=device))
+        vals = c.data.to(torch.float64)
+        row = torch.repeat_interleave(
+            torch.arange(n, dtype=torch.int64, device=device), lens)
+        o1 = torch.argsort(vals, stable=True)
+        o2 = torch.argsort(row[o1], stable=True)
+        sorted_vals = vals[o1][o2]  # row-major, values ascending per row
+        # interpolated index p*(len-1) within each row
+        pos = self.p * (lens.to(torch.float64) - 1).clamp(min=0)
+        lo = pos.floor().to(torch.int64)
+        hi = pos.ceil().to(torch.int64)
+        frac = pos - lo.to(torch.float64)
+        base = offs[:-1]
+        safe = lens > 0
+        lo_i = (base + lo).clamp(0, total - 1)
+        hi_i = (base + hi).clamp(0, total - 1)
+        out = sorted_vals[lo_i] * (1 - frac) + sorted_vals[hi_i] * frac
+        validity = compact_validity(safe) if bool((~safe).any()) else None
+        return Column(dtypes.float64, out, validity)

@@ -1529,6 +1529,11 @@ class Planner:
             entry = ctx.agg_map.get(_agg_ast_key(fc))
             if entry is None:
                 raise SqlError(f"aggregate {name}() was not collected")
+            if isinstance(entry, tuple) and entry[0] == "percentile":
+                from .. import functions as F
+
+                _, en, p = entry
+                return F.ListQuantile(col(en), p)
             if isinstance(entry, tuple) and entry[0] == "covar":
                 from ..exprs import Sqrt
 
@@ -1866,7 +1871,8 @@ def _conjuncts(node: A.ANode) -> List[A.ANode]:
 
 AGG_NAMES = {"sum", "avg", "min", "max", "count", "stddev_samp",
              "stddev", "stddev_pop", "var_samp", "variance", "var_pop",
-             "corr", "covar_samp", "covar_pop", "approx_count_distinct"}
+             "corr", "covar_samp", "covar_pop", "approx_count_distinct",
+             "percentile", "percentile_approx", "median"}
 
 
 def _has_agg(sel: A.Select) -> bool:
@@ -2157,6 +2163,22 @@ def _register_agg(fc: A.FuncCall, planner: Planner, scope: Scope, ctx: AggCtx):
         return
     name = fc.name
     alloc = planner.alloc
+    if name in ("percentile", "percentile_approx", "median"):
+        # exact interpolated percentile over a collected list (Spark
+        # percentile semantics; the approx variant is implemented exactly)
+        arg = planner.to_expr(fc.args[0], scope, None)
+        if name == "median":
+            p = 0.5
+        else:
+            pe = planner.to_expr(fc.args[1], scope, None)
+            if not isinstance(pe, Literal):
+                raise SqlError("percentile fraction must be a literal")
+            p = float(pe.value)
+        en = alloc.fresh("__pct", dtypes.list_of(dtypes.float64))
+        ctx.aggs.append(AggFunc("collect_list",
+                                Cast(arg, dtypes.float64), name=en))
+        ctx.agg_map[key] = ("percentile", en, p)
+        return
     if name == "approx_count_distinct":
         # exact distinct count is a valid (and here cheap) implementation
         arg = planner.to_expr(fc.args[0], scope, None)

@@ -218,3 +218,11 @@ def test_corr_covar(env):
     assert math.isclose(out["cs"][0], statistics.variance(vals))
     assert math.isclose(out["cp"][0], statistics.pvariance(vals))
     assert out["ad"][0] == 2  # two distinct non-null strings
+
+
+def test_percentile_median(env):
+    out = run(env, "select median(x) m, percentile(x, 0.5) p5, "
+                   "percentile_approx(x, 1.0) pmax from tt")
+    assert out["m"] == [2.0]
+    assert out["p5"] == [2.0]
+    assert out["pmax"] == [3.0]
