@@ -484,8 +484,6 @@ class Planner:
         if s.op in ("intersect_all", "except_all"):
             # multiset semantics: per-side counts, then replicate each
             # distinct row min(c0,c1) / max(c0-c1,0) times
-            from .. import functions as F
-
             c0 = self.alloc.fresh("__c0", dtypes.int64)
             c1 = self.alloc.fresh("__c1", dtypes.int64)
             agg = LAgg(u, [(col(n), n) for n in out_names],
@@ -1530,8 +1528,6 @@ class Planner:
             if entry is None:
                 raise SqlError(f"aggregate {name}() was not collected")
             if isinstance(entry, tuple) and entry[0] == "percentile":
-                from .. import functions as F
-
                 _, en, p = entry
                 return F.ListQuantile(col(en), p)
             if isinstance(entry, tuple) and entry[0] == "covar":
