@@ -1190,3 +1190,36 @@ class ListQuantile(Expr):
         out = sorted_vals[lo_i] * (1 - frac) + sorted_vals[hi_i] * frac
         validity = compact_validity(safe) if bool((~safe).any()) else None
         return Column(dtypes.float64, out, validity)
+
+
+_UNARY_MATH = {
+    "sin": torch.sin, "cos": torch.cos, "tan": torch.tan,
+    "asin": torch.asin, "acos": torch.acos, "atan": torch.atan,
+    "sinh": torch.sinh, "cosh": torch.cosh, "tanh": torch.tanh,
+    "log2": torch.log2, "cbrt": lambda x: torch.sign(x) * x.abs() ** (1.0 / 3),
+    "degrees": torch.rad2deg, "radians": torch.deg2rad,
+    "expm1": torch.expm1, "log1p": torch.log1p, "rint": torch.round,
+}
+
+
+@dataclass(eq=False)
+class UnaryMath(Expr):
+    """Generic float64 unary math (spark_math parity: sin/cos/.../cbrt)."""
+    fn: str
+    child: Expr
+
+    def eval(self, batch):
+        c = _f64(self.child, batch)
+        return Column(dtypes.float64, _UNARY_MATH[self.fn](c.data), c.validity)
+
+
+@dataclass(eq=False)
+class Atan2(Expr):
+    y: Expr
+    x: Expr
+
+    def eval(self, batch):
+        a = _f64(self.y, batch)
+        b = _f64(self.x, batch)
+        return Column(dtypes.float64, torch.atan2(a.data, b.data),
+                      combine_validity(a, b))

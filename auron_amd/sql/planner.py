@@ -1735,6 +1735,23 @@ class Planner:
             return Cast(conv(a[0]), dtypes.string)
         if name == "xxhash64":
             return F.XxHash64([conv(x) for x in a])
+        from ..functions import _UNARY_MATH
+
+        if name in _UNARY_MATH:
+            return F.UnaryMath(name, conv(a[0]))
+        if name == "atan2":
+            return F.Atan2(conv(a[0]), conv(a[1]))
+        if name == "pi":
+            from ..exprs import Literal as _L
+
+            return _L(3.141592653589793)
+        if name == "log":
+            # log(x) = ln; log(base, x) = ln(x)/ln(base)
+            if len(a) == 1:
+                return F.Ln(conv(a[0]))
+            from ..exprs import Arith as _Ar
+
+            return _Ar("/", F.Ln(conv(a[1])), F.Ln(conv(a[0])))
         if name == "rand":
             return F.Rand(_int_lit(conv(a[0])) if a else 42)
         if name == "randn":

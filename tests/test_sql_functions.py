@@ -226,3 +226,14 @@ def test_percentile_median(env):
     assert out["m"] == [2.0]
     assert out["p5"] == [2.0]
     assert out["pmax"] == [3.0]
+
+
+def test_math_breadth(env):
+    import math
+
+    out = run(env, "select sin(x) a, atan2(x, 1) b, log(2.0, 8.0) c, "
+                   "cbrt(27.0) d, tanh(0.0) e from tt order by a limit 1")
+    assert math.isclose(out["a"][0], math.sin(1))
+    assert math.isclose(out["c"][0], 3.0)
+    assert math.isclose(out["d"][0], 3.0)
+    assert out["e"][0] == 0.0
