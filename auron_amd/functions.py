@@ -1223,3 +1223,66 @@ class Atan2(Expr):
         b = _f64(self.x, batch)
         return Column(dtypes.float64, torch.atan2(a.data, b.data),
                       combine_validity(a, b))
+
+
+# ------------------------------------------------------------ regex (host)
+def _java_regex(p: str) -> str:
+    """Minimal Java->Python regex translation (the dialects agree on the
+    subset these functions see; host cold path like GetJsonObject)."""
+    return p
+
+
+@dataclass(eq=False)
+class RegexpExtract(Expr):
+    child: Expr
+    pattern: str
+    group: int = 1
+
+    def eval(self, batch):
+        import re as _re
+
+        rx = _re.compile(_java_regex(self.pattern))
+        c = self.child.eval(batch)
+        out = []
+        for v in c.to_pylist():
+            if v is None:
+                out.append(None)
+                continue
+            m = rx.search(v)
+            # Spark: no match -> empty string
+            out.append(m.group(self.group) if m and m.group(self.group)
+                       is not None else "")
+        return Column.from_pylist(out, dtypes.string, str(batch.device))
+
+
+@dataclass(eq=False)
+class RegexpReplace(Expr):
+    child: Expr
+    pattern: str
+    replacement: str
+
+    def eval(self, batch):
+        import re as _re
+
+        rx = _re.compile(_java_regex(self.pattern))
+        # Java $1 backrefs -> Python \1
+        rep = _re.sub(r"\$(\d+)", r"\\\1", self.replacement)
+        c = self.child.eval(batch)
+        out = [None if v is None else rx.sub(rep, v) for v in c.to_pylist()]
+        return Column.from_pylist(out, dtypes.string, str(batch.device))
+
+
+@dataclass(eq=False)
+class RLike(Expr):
+    child: Expr
+    pattern: str
+
+    def eval(self, batch):
+        import re as _re
+
+        rx = _re.compile(_java_regex(self.pattern))
+        c = self.child.eval(batch)
+        vals = c.to_pylist()
+        data = torch.tensor([v is not None and rx.search(v) is not None
+                             for v in vals], dtype=torch.bool)
+        return Column(dtypes.bool_, data.to(batch.device), c.validity)

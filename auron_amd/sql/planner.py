@@ -1752,6 +1752,14 @@ class Planner:
             from ..exprs import Arith as _Ar
 
             return _Ar("/", F.Ln(conv(a[1])), F.Ln(conv(a[0])))
+        if name == "regexp_extract":
+            g = _int_lit(conv(a[2])) if len(a) > 2 else 1
+            return F.RegexpExtract(conv(a[0]), _str_lit(conv(a[1])), g)
+        if name == "regexp_replace":
+            return F.RegexpReplace(conv(a[0]), _str_lit(conv(a[1])),
+                                   _str_lit(conv(a[2])))
+        if name == "rlike":
+            return F.RLike(conv(a[0]), _str_lit(conv(a[1])))
         if name == "rand":
             return F.Rand(_int_lit(conv(a[0])) if a else 42)
         if name == "randn":

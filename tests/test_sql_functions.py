@@ -237,3 +237,12 @@ def test_math_breadth(env):
     assert math.isclose(out["c"][0], 3.0)
     assert math.isclose(out["d"][0], 3.0)
     assert out["e"][0] == 0.0
+
+
+def test_regex_routes(env):
+    out = run(env, "select regexp_extract(s, '([ab]+)', 1) ex, "
+                   "regexp_replace(s, ',', '-') rp, rlike(s, '^a') rl "
+                   "from tt")
+    assert out["ex"] == ["a", "b", None]
+    assert out["rp"] == ["a-b", "b", None]
+    assert out["rl"] == [True, False, None]
