@@ -1286,3 +1286,79 @@ class RLike(Expr):
         data = torch.tensor([v is not None and rx.search(v) is not None
                              for v in vals], dtype=torch.bool)
         return Column(dtypes.bool_, data.to(batch.device), c.validity)
+
+
+@dataclass(eq=False)
+class ArrayContains(Expr):
+    """array_contains(list, value): vectorized flat compare + segment any."""
+    child: Expr
+    value: object
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        assert c.dtype.is_list
+        n = len(c)
+        device = c.device
+        offs = c.offsets
+        lens = offs[1:] - offs[:-1]
+        if int(offs[-1].item()) == 0:
+            return Column(dtypes.bool_,
+                          torch.zeros(n, dtype=torch.bool, device=device),
+                          c.validity)
+        row = torch.repeat_interleave(
+            torch.arange(n, dtype=torch.int64, device=device), lens)
+        if c.dtype.child.code == dtypes.DECIMAL64:
+            hit = c.data.to(torch.float64) / 10 ** c.dtype.scale == float(self.value)
+        else:
+            hit = c.data == self.value
+        out = torch.zeros(n, dtype=torch.bool, device=device)
+        out.scatter_reduce_(0, row, hit, reduce="amax", include_self=True)
+        return Column(dtypes.bool_, out.to(torch.bool), c.validity)
+
+
+@dataclass(eq=False)
+class SubstringIndex(Expr):
+    """substring_index(s, delim, count): Spark semantics (host path)."""
+    child: Expr
+    delim: str
+    count: int
+
+    def eval(self, batch):
+        c = self.child.eval(batch)
+        out = []
+        for v in c.to_pylist():
+            if v is None:
+                out.append(None)
+            elif self.count > 0:
+                out.append(self.delim.join(v.split(self.delim)[:self.count]))
+            elif self.count < 0:
+                out.append(self.delim.join(v.split(self.delim)[self.count:]))
+            else:
+                out.append("")
+        return Column.from_pylist(out, dtypes.string, str(batch.device))
+
+
+@dataclass(eq=False)
+class Levenshtein(Expr):
+    """levenshtein(a, b): edit distance (host DP, cold path)."""
+    left: Expr
+    right: Expr
+
+    def eval(self, batch):
+        a = self.left.eval(batch).to_pylist()
+        b = self.right.eval(batch).to_pylist()
+
+        def dist(x, y):
+            if x is None or y is None:
+                return None
+            prev = list(range(len(y) + 1))
+            for i, cx in enumerate(x, 1):
+                cur = [i]
+                for j, cy in enumerate(y, 1):
+                    cur.append(min(prev[j] + 1, cur[j - 1] + 1,
+                                   prev[j - 1] + (cx != cy)))
+                prev = cur
+            return prev[-1]
+
+        return Column.from_pylist([dist(x, y) for x, y in zip(a, b)],
+                                  dtypes.int32, str(batch.device))

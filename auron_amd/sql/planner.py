@@ -1752,6 +1752,17 @@ class Planner:
             from ..exprs import Arith as _Ar
 
             return _Ar("/", F.Ln(conv(a[1])), F.Ln(conv(a[0])))
+        if name == "array_contains":
+            from ..exprs import Literal as _L2
+
+            v = conv(a[1])
+            assert isinstance(v, _L2), "array_contains needs a literal value"
+            return F.ArrayContains(conv(a[0]), v.value)
+        if name == "substring_index":
+            return F.SubstringIndex(conv(a[0]), _str_lit(conv(a[1])),
+                                    _int_lit(conv(a[2])))
+        if name == "levenshtein":
+            return F.Levenshtein(conv(a[0]), conv(a[1]))
         if name == "regexp_extract":
             g = _int_lit(conv(a[2])) if len(a) > 2 else 1
             return F.RegexpExtract(conv(a[0]), _str_lit(conv(a[1])), g)

@@ -246,3 +246,12 @@ def test_regex_routes(env):
     assert out["ex"] == ["a", "b", None]
     assert out["rp"] == ["a-b", "b", None]
     assert out["rl"] == [True, False, None]
+
+
+def test_array_string_breadth(env):
+    out = run(env, "select array_contains(array(x, 2), 2) ac, "
+                   "substring_index(s, ',', 1) si, "
+                   "levenshtein(s, 'ab') lv from tt order by x")
+    assert out["ac"] == [True, True, True]
+    assert out["si"] == ["a", "b", None]
+    assert out["lv"] == [1, 1, None]  # 'a,b'->'ab' deletes one char
